@@ -1,0 +1,104 @@
+/* copr_gpu.h — C-ABI of the MI355X-native coprocessor batch-executor engine.
+ *
+ * This is the drop-in boundary described in DESIGN.md §1: it sits exactly where
+ * BatchDagHandler hands work to BatchExecutorsRunner
+ * (reference src/coprocessor/dag/mod.rs:157-196 -> runner.rs:673,840), expressed
+ * in the stable-C-ABI idiom of the reference's own plugin boundary
+ * (components/coprocessor_plugin_api/src/plugin_api.rs:21-42): no exceptions
+ * cross the ABI, plain pointers and sizes, engine owns what it allocates.
+ *
+ * The KV feed mirrors the pull side of the Storage trait
+ * (components/tidb_query_common/src/storage/mod.rs:32-79): the caller hands the
+ * scanned (key, value) byte stream; the engine copies it into HBM once and runs
+ * DAG requests against the resident region.
+ *
+ * Every compute entry point REQUIRES a HIP device; there is no CPU fallback.
+ */
+#ifndef COPR_GPU_H
+#define COPR_GPU_H
+
+#include "copr_types.h"
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef struct copr_engine copr_engine;   /* one per GPU (per process rank) */
+typedef struct copr_region copr_region;   /* one resident Region KV block   */
+
+typedef enum copr_status {
+  COPR_OK = 0,
+  COPR_ERR_NO_GPU = 1,          /* no HIP device / hip runtime failure */
+  COPR_ERR_INVALID_REQUEST = 2, /* descriptor validation (mirrors
+                                   ErrorInner::Evaluate mapping,
+                                   dag/mod.rs:219-258) */
+  COPR_ERR_UNSUPPORTED = 3,     /* sig/executor outside the native subset:
+                                   the Rust shim falls back to its CPU path */
+  COPR_ERR_STORAGE = 4,         /* malformed KV bytes (ErrorInner::Storage) */
+  COPR_ERR_OOM = 5,
+  COPR_ERR_INTERNAL = 6
+} copr_status;
+
+/* Last error message for this thread (valid until the next engine call). */
+const char *copr_last_error(void);
+
+/* ---- engine lifecycle ---- */
+copr_status copr_engine_create(int hip_device, copr_engine **out);
+void        copr_engine_destroy(copr_engine *);
+
+/* ---- region feed (Storage boundary) ----
+ * keys/vals are concatenated byte streams with offs[n] prefix offsets
+ * (offs[0] = 0, offs[n] = total bytes). Keys arrive raw — the MVCC
+ * memcomparable+ts envelope already stripped, as TikvStorage does
+ * (src/coprocessor/dag/storage_impl.rs:93). Copied to HBM; host buffers may be
+ * freed after return. */
+copr_status copr_region_create(copr_engine *,
+                               const uint8_t *keys, const uint64_t *key_offs,
+                               const uint8_t *vals, const uint64_t *val_offs,
+                               uint64_t n_kv, copr_region **out);
+void        copr_region_destroy(copr_region *);
+uint64_t    copr_region_num_kv(const copr_region *);
+
+/* ---- DAG execution (REQ_TYPE_DAG = 103, src/coprocessor/mod.rs:57) ----
+ * Runs the executor tree over the region; returns the SelectResponse chunk
+ * payload (datum-encoded rows). Free the result with copr_result_free. */
+copr_status copr_dag_run(copr_engine *, const CoprDagRequest *,
+                         copr_region *const *regions, uint32_t n_regions,
+                         CoprSelectResult *out);
+void        copr_result_free(CoprSelectResult *);
+
+/* ---- checksum (REQ_TYPE_CHECKSUM = 105; src/coprocessor/checksum.rs:59-114) ----
+ * CRC-64/XZ per KV over key||value, XOR-folded (order-independent). */
+copr_status copr_checksum(copr_engine *, copr_region *const *regions,
+                          uint32_t n_regions, uint64_t *checksum,
+                          uint64_t *total_kvs, uint64_t *total_bytes);
+
+/* ---- synthetic region generator (fixture factory) ----
+ * Host-side (OpenMP) generator of reference-format regions, mirroring
+ * test_coprocessor's fixture store (test_coprocessor/src/store.rs:83-91:
+ * table::encode_row_key + table::encode_row). Not part of the serving path —
+ * it exists so tests/bench feed identical bytes to engine and oracle.
+ * See DESIGN.md §6 for the RNG contract. */
+typedef struct CoprGenSpec {
+  int32_t  config_index;     /* BASELINE.json configs[] index (seeds the RNG) */
+  int64_t  table_id;
+  uint64_t n_rows;
+  uint64_t first_handle;     /* shard offset: rows get handles
+                                [first_handle, first_handle + n_rows) */
+  uint32_t n_cols;           /* schema per config (see copr_gen.cpp) */
+  int32_t  row_format;       /* 1 = row-v1 datums, 2 = row-v2 */
+} CoprGenSpec;
+
+typedef struct CoprGenOut {   /* host buffers owned by the generator */
+  uint8_t  *keys;  uint64_t *key_offs;
+  uint8_t  *vals;  uint64_t *val_offs;
+  uint64_t  n_kv;
+} CoprGenOut;
+
+copr_status copr_gen_region(const CoprGenSpec *, CoprGenOut *out);
+void        copr_gen_free(CoprGenOut *);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* COPR_GPU_H */
