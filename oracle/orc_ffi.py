@@ -1,0 +1,96 @@
+"""ctypes binding for the ORACLE library (test infrastructure ONLY).
+
+Importable only from tests/, __graft_entry__.smoke() and bench.py's
+cpu_baseline leg. Reuses the descriptor struct definitions from
+tikv_amd._ffi (pure ctypes declarations; no product library is loaded by
+importing them).
+"""
+import ctypes as C
+import os
+import sys
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+_ROOT = os.path.dirname(_HERE)
+if _ROOT not in sys.path:
+    sys.path.insert(0, _ROOT)
+
+from tikv_amd._ffi import CoprDagRequest  # noqa: E402
+
+
+class OrcResult(C.Structure):
+    _fields_ = [("data", C.POINTER(C.c_uint8)), ("data_len", C.c_uint64),
+                ("n_rows", C.c_uint64)]
+
+
+_lib = None
+
+
+def load_lib():
+    global _lib
+    if _lib is not None:
+        return _lib
+    lib = C.CDLL(os.path.join(_HERE, "liboracle.so"))
+    lib.orc_last_error.restype = C.c_char_p
+    lib.orc_dag_run.restype = C.c_int
+    lib.orc_dag_run.argtypes = [C.POINTER(CoprDagRequest),
+                                C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
+                                C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
+                                C.c_uint64, C.POINTER(OrcResult)]
+    lib.orc_result_free.argtypes = [C.POINTER(OrcResult)]
+    lib.orc_checksum.restype = C.c_int
+    lib.orc_checksum.argtypes = [C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
+                                 C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
+                                 C.c_uint64, C.POINTER(C.c_uint64),
+                                 C.POINTER(C.c_uint64), C.POINTER(C.c_uint64)]
+    lib.orc_crc64_xz.restype = C.c_uint64
+    lib.orc_crc64_xz.argtypes = [C.c_char_p, C.c_uint64]
+    lib.orc_test_memcmp_encode.restype = C.c_uint64
+    lib.orc_test_memcmp_encode.argtypes = [C.c_char_p, C.c_uint64, C.c_int, C.c_char_p]
+    lib.orc_test_memcmp_decode.restype = C.c_uint64
+    lib.orc_test_memcmp_decode.argtypes = [C.c_char_p, C.c_uint64, C.c_char_p,
+                                           C.POINTER(C.c_uint64)]
+    lib.orc_test_var_i64_encode.restype = C.c_uint64
+    lib.orc_test_var_i64_encode.argtypes = [C.c_int64, C.c_char_p]
+    lib.orc_test_var_i64_decode.restype = C.c_int
+    lib.orc_test_var_i64_decode.argtypes = [C.c_char_p, C.c_uint64,
+                                            C.POINTER(C.c_int64), C.POINTER(C.c_uint64)]
+    lib.orc_test_row_key.argtypes = [C.c_int64, C.c_int64, C.c_char_p]
+    lib.orc_test_int_handle.restype = C.c_int
+    lib.orc_test_int_handle.argtypes = [C.c_char_p, C.c_uint64, C.POINTER(C.c_int64)]
+    lib.orc_test_row_v2_col.restype = C.c_int
+    lib.orc_test_row_v2_col.argtypes = [C.c_char_p, C.c_uint64, C.c_int64,
+                                        C.c_int32, C.c_uint32, C.c_char_p,
+                                        C.POINTER(C.c_int)]
+    lib.orc_test_dec_add_encode.restype = C.c_int
+    lib.orc_test_dec_add_encode.argtypes = [C.c_char_p, C.c_uint64, C.c_char_p,
+                                            C.c_uint64, C.c_char_p]
+    lib.orc_test_dec_from_i64_encode.restype = C.c_int
+    lib.orc_test_dec_from_i64_encode.argtypes = [C.c_int64, C.c_char_p]
+    _lib = lib
+    return lib
+
+
+def dag_run(req, keys, key_offs, vals, val_offs, n_kv):
+    """Run the oracle pipeline. Buffer args are ctypes pointers (e.g. from
+    GenRegion) or bytes (auto-wrapped)."""
+    lib = load_lib()
+    res = OrcResult()
+    st = lib.orc_dag_run(C.byref(req), keys, key_offs, vals, val_offs, n_kv,
+                         C.byref(res))
+    if st != 0:
+        raise RuntimeError("oracle: %s" % lib.orc_last_error().decode())
+    data = C.string_at(res.data, res.data_len) if res.data_len else b""
+    n = res.n_rows
+    lib.orc_result_free(C.byref(res))
+    return data, n
+
+
+def checksum(keys, key_offs, vals, val_offs, n_kv):
+    lib = load_lib()
+    cs = C.c_uint64()
+    kvs = C.c_uint64()
+    byts = C.c_uint64()
+    st = lib.orc_checksum(keys, key_offs, vals, val_offs, n_kv,
+                          C.byref(cs), C.byref(kvs), C.byref(byts))
+    assert st == 0
+    return cs.value, kvs.value, byts.value

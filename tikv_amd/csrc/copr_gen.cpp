@@ -1,0 +1,249 @@
+/* copr_gen.cpp — synthetic Region generator (fixture factory; PRODUCT lib).
+ *
+ * Writes reference-format KV regions, mirroring the reference's own fixture
+ * writer (test_coprocessor/src/store.rs:83-91):
+ *   key  = table::encode_row_key (codec/table.rs:187-193)
+ *        = 't' || BE(table_id ^ SIGN) || "_r" || BE(handle ^ SIGN)
+ *   value= table::encode_row (codec/table.rs:166-184): repeated
+ *          [VAR_INT datum(col_id), datum(value)] in non-comparable form
+ *          (datum.rs write_datum, comparable=false: I64 -> VAR_INT zigzag
+ *          varint; Decimal -> DECIMAL flag + write_decimal; Bytes ->
+ *          COMPACT_BYTES).
+ *
+ * RNG contract: xoshiro256++ whose 256-bit state is seeded
+ * per row by a splitmix64 chain over (0xC0FFEE + config_index, handle), so
+ * generation is deterministic AND parallel. Schemas per BASELINE.json
+ * configs.
+ */
+#include "../../include/copr_gpu.h"
+#include "prod_decimal.h"
+
+#include <cstring>
+#include <cstdlib>
+#include <vector>
+#include <string>
+
+#ifdef _OPENMP
+#include <omp.h>
+#endif
+
+namespace {
+
+/* ---- encode primitives (cites as in header comment) ---- */
+static inline size_t enc_var_u64(uint8_t *buf, uint64_t v) {
+  size_t i = 0;
+  while (v >= 0x80) { buf[i++] = 0x80 | (v & 0x7f); v >>= 7; }
+  buf[i++] = (uint8_t)v;
+  return i;
+}
+static inline size_t enc_var_i64(uint8_t *buf, int64_t v) {
+  uint64_t uv = (uint64_t)v << 1;
+  if (v < 0) uv = ~uv;
+  return enc_var_u64(buf, uv);
+}
+static inline void enc_cmp_i64(uint8_t *buf, int64_t v) {
+  uint64_t u = (uint64_t)v ^ 0x8000000000000000ull;
+  for (int i = 7; i >= 0; i--) { buf[i] = (uint8_t)u; u >>= 8; }
+}
+static inline void enc_row_key(int64_t table_id, int64_t handle, uint8_t out[19]) {
+  out[0] = 't';
+  enc_cmp_i64(out + 1, table_id);
+  out[9] = '_'; out[10] = 'r';
+  enc_cmp_i64(out + 11, handle);
+}
+
+/* ---- RNG: splitmix64 (Vigna) + xoshiro256++ (Blackman/Vigna), public
+ * domain algorithms ---- */
+static inline uint64_t splitmix64(uint64_t &x) {
+  uint64_t z = (x += 0x9E3779B97F4A7C15ull);
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+  return z ^ (z >> 31);
+}
+struct Xo256 {
+  uint64_t s[4];
+  void seed(uint64_t config_seed, uint64_t row) {
+    uint64_t x = config_seed * 0x9E3779B97F4A7C15ull ^ (row + 0x243F6A8885A308D3ull);
+    for (int i = 0; i < 4; i++) s[i] = splitmix64(x);
+  }
+  static inline uint64_t rotl(uint64_t x, int k) { return (x << k) | (x >> (64 - k)); }
+  uint64_t next() {
+    uint64_t result = rotl(s[0] + s[3], 23) + s[0];
+    uint64_t t = s[1] << 17;
+    s[2] ^= s[0]; s[3] ^= s[1]; s[1] ^= s[2]; s[0] ^= s[3];
+    s[2] ^= t; s[3] = rotl(s[3], 45);
+    return result;
+  }
+};
+
+static inline int64_t uniform_pm(Xo256 &rng, int64_t bound) {
+  /* uniform in [-bound, bound] */
+  return (int64_t)(rng.next() % (uint64_t)(2 * bound + 1)) - bound;
+}
+
+/* ---- per-config row writers. Append value bytes to buf. ---- */
+struct GenCfg {
+  int32_t config_index;
+  uint64_t param;           /* cfg3: group count K */
+};
+
+/* cell helper: [VAR_INT col_id][datum] */
+static inline void put_cell_i64(std::string &buf, int64_t col_id, int64_t v) {
+  uint8_t tmp[24];
+  size_t n = 0;
+  tmp[n++] = 8;                       /* VAR_INT flag (col id datum) */
+  n += enc_var_i64(tmp + n, col_id);
+  tmp[n++] = 8;                       /* VAR_INT flag (value) */
+  n += enc_var_i64(tmp + n, v);
+  buf.append((const char *)tmp, n);
+}
+static inline void put_cell_decimal(std::string &buf, int64_t col_id,
+                                    int64_t scaled, uint8_t frac) {
+  uint8_t tmp[64];
+  size_t n = 0;
+  tmp[n++] = 8;
+  n += enc_var_i64(tmp + n, col_id);
+  tmp[n++] = 6;                       /* DECIMAL flag */
+  prod::PDec d = prod::pdec_from_scaled_i128(scaled, frac);
+  uint8_t prec, fr;
+  prod::pdec_prec_and_frac(d, &prec, &fr);
+  n += prod::pdec_encode(d, prec, fr, tmp + n);
+  buf.append((const char *)tmp, n);
+}
+static inline void put_cell_bytes(std::string &buf, int64_t col_id,
+                                  const uint8_t *data, size_t len) {
+  uint8_t tmp[16];
+  size_t n = 0;
+  tmp[n++] = 8;
+  n += enc_var_i64(tmp + n, col_id);
+  tmp[n++] = 2;                       /* COMPACT_BYTES flag */
+  n += enc_var_i64(tmp + n, (int64_t)len);
+  buf.append((const char *)tmp, n);
+  buf.append((const char *)data, len);
+}
+
+static bool write_row(const GenCfg &cfg, Xo256 &rng, std::string &buf) {
+  switch (cfg.config_index) {
+    case 0: {                          /* cfg1: 4 x i64, ids 1..4, ±1e9 */
+      for (int64_t c = 1; c <= 4; c++) put_cell_i64(buf, c, uniform_pm(rng, 1000000000));
+      return true;
+    }
+    case 1: {                          /* cfg2: 16 x i64, ids 1..16, ±1e9 */
+      for (int64_t c = 1; c <= 16; c++) put_cell_i64(buf, c, uniform_pm(rng, 1000000000));
+      return true;
+    }
+    case 2: {                          /* cfg3: i64 group, Decimal(12,2), VarBytes */
+      uint64_t k = cfg.param ? cfg.param : 64;
+      put_cell_i64(buf, 1, (int64_t)(rng.next() % k));
+      put_cell_decimal(buf, 2, uniform_pm(rng, 999999999999ll), 2);
+      uint8_t bytes[24];
+      size_t blen = 8 + rng.next() % 17;
+      for (size_t i = 0; i < blen; i++) {
+        double u = (double)(rng.next() >> 11) * (1.0 / 9007199254740992.0);
+        bytes[i] = (uint8_t)('a' + (int)(26.0 * u * u * u));  /* low-biased */
+      }
+      put_cell_bytes(buf, 3, bytes, blen);
+      return true;
+    }
+    case 3: {                          /* cfg4: opaque value 100..200 B */
+      size_t blen = 100 + rng.next() % 101;
+      uint8_t chunk[8];
+      for (size_t i = 0; i < blen; i += 8) {
+        uint64_t x = rng.next();
+        memcpy(chunk, &x, 8);
+        buf.append((const char *)chunk, (blen - i) < 8 ? (blen - i) : 8);
+      }
+      return true;
+    }
+    default:
+      return false;
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+copr_status copr_gen_region(const CoprGenSpec *spec, CoprGenOut *out) {
+  memset(out, 0, sizeof(*out));
+  GenCfg cfg{spec->config_index, 0};
+  /* for config_index==2 (cfg3), n_cols carries the group count K */
+  if (spec->config_index == 2) cfg.param = spec->n_cols ? spec->n_cols : 64;
+  uint64_t n = spec->n_rows;
+  uint64_t seed = 0xC0FFEEull + (uint64_t)spec->config_index;
+
+  uint64_t *key_offs = (uint64_t *)malloc((n + 1) * sizeof(uint64_t));
+  uint64_t *val_offs = (uint64_t *)malloc((n + 1) * sizeof(uint64_t));
+  uint8_t *keys = (uint8_t *)malloc(n * 19 + 1);
+  if (!key_offs || !val_offs || !keys) return COPR_ERR_OOM;
+  for (uint64_t i = 0; i <= n; i++) key_offs[i] = i * 19;
+
+  int T = 1;
+#ifdef _OPENMP
+  T = omp_get_max_threads();
+#endif
+  if ((uint64_t)T > n / 1024 + 1) T = (int)(n / 1024 + 1);
+  std::vector<std::string> chunk_buf(T);
+  std::vector<std::vector<uint32_t>> chunk_sizes(T);
+  uint64_t per = (n + T - 1) / T;
+  bool ok = true;
+
+#ifdef _OPENMP
+#pragma omp parallel for schedule(static, 1)
+#endif
+  for (int t = 0; t < T; t++) {
+    uint64_t lo = (uint64_t)t * per, hi = lo + per;
+    if (hi > n) hi = n;
+    if (lo >= hi) continue;
+    std::string &buf = chunk_buf[t];
+    buf.reserve((hi - lo) * 160);
+    chunk_sizes[t].resize(hi - lo);
+    Xo256 rng;
+    std::string row;
+    for (uint64_t i = lo; i < hi; i++) {
+      int64_t handle = (int64_t)(spec->first_handle + i);
+      enc_row_key(spec->table_id, handle, keys + i * 19);
+      rng.seed(seed, (uint64_t)handle);
+      row.clear();
+      if (!write_row(cfg, rng, row)) { ok = false; break; }
+      chunk_sizes[t][i - lo] = (uint32_t)row.size();
+      buf += row;
+    }
+  }
+  if (!ok) { free(key_offs); free(val_offs); free(keys); return COPR_ERR_INVALID_REQUEST; }
+
+  uint64_t total = 0;
+  std::vector<uint64_t> chunk_base(T);
+  for (int t = 0; t < T; t++) { chunk_base[t] = total; total += chunk_buf[t].size(); }
+  uint8_t *vals = (uint8_t *)malloc(total ? total : 1);
+  if (!vals) { free(key_offs); free(val_offs); free(keys); return COPR_ERR_OOM; }
+
+#ifdef _OPENMP
+#pragma omp parallel for schedule(static, 1)
+#endif
+  for (int t = 0; t < T; t++) {
+    if (!chunk_buf[t].empty())
+      memcpy(vals + chunk_base[t], chunk_buf[t].data(), chunk_buf[t].size());
+    uint64_t lo = (uint64_t)t * per, hi = lo + per;
+    if (hi > n) hi = n;
+    uint64_t off = chunk_base[t];
+    for (uint64_t i = lo; i < hi; i++) {
+      val_offs[i] = off;
+      off += chunk_sizes[t][i - lo];
+    }
+  }
+  val_offs[n] = total;
+
+  out->keys = keys; out->key_offs = key_offs;
+  out->vals = vals; out->val_offs = val_offs;
+  out->n_kv = n;
+  return COPR_OK;
+}
+
+void copr_gen_free(CoprGenOut *out) {
+  if (!out) return;
+  free(out->keys); free(out->key_offs); free(out->vals); free(out->val_offs);
+  memset(out, 0, sizeof(*out));
+}
+
+}  // extern "C"

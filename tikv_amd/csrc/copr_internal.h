@@ -1,0 +1,108 @@
+/* copr_internal.h — PRODUCT internal host<->kernel shared structures. */
+#ifndef COPR_INTERNAL_H
+#define COPR_INTERNAL_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+namespace copr {
+
+/* device-resident region */
+struct DevRegion {
+  uint8_t *d_keys = nullptr;
+  uint64_t *d_key_offs = nullptr;
+  uint8_t *d_vals = nullptr;
+  uint64_t *d_val_offs = nullptr;
+  uint64_t n_kv = 0;
+  uint64_t key_bytes = 0, val_bytes = 0;
+  uint32_t max_row_bytes = 0;      /* max value size over all rows */
+};
+
+/* compare kinds (order matches oracle CmpKind) */
+enum { CMP_LT = 0, CMP_LE, CMP_GT, CMP_GE, CMP_EQ, CMP_NE };
+
+/* device agg kinds */
+enum {
+  DAGG_COUNT_ROWS = 0,   /* count(const non-null): +1 per surviving row */
+  DAGG_COUNT_COL,        /* count(col): +1 per non-null */
+  DAGG_SUM_INT,          /* i128 sum of int col (covers sum/avg(int) after
+                            the Decimal rewrite: exact integer sum) */
+  DAGG_SUM_DEC,          /* i128 sum of decimal col scaled to target frac */
+};
+
+struct DevAggSpec {
+  int32_t kind;
+  int64_t col_id;        /* source column id (ignored for COUNT_ROWS) */
+  int32_t col_unsigned;
+  int32_t target_frac;   /* SUM_DEC: scale values to this frac */
+};
+
+#define COPR_MAX_AGGS 8
+#define COPR_MAX_OUT_COLS 20
+
+struct ScanPlan {
+  /* filter: cmp(col, const) over ints; has_filter 0 => keep all */
+  int32_t has_filter;
+  int64_t filter_col_id;
+  int32_t filter_cmp;
+  int64_t filter_const;
+  int32_t filter_col_unsigned;
+  int32_t filter_const_unsigned;
+  int32_t filter_const_null;     /* NULL const: predicate never true */
+
+  int32_t mode;                  /* 0 project, 1 simple agg, 2 hash agg */
+  int32_t n_aggs;
+  DevAggSpec aggs[COPR_MAX_AGGS];
+
+  /* hash agg (mode 2) */
+  int64_t group_col_id;
+  int32_t group_col_unsigned;
+  uint32_t table_size;           /* power of two */
+
+  /* project (mode 0): capture raw cell spans for these column ids */
+  int32_t n_out;
+  int64_t out_col_ids[COPR_MAX_OUT_COLS];
+  int32_t out_is_handle[COPR_MAX_OUT_COLS]; /* 1 => decoded int handle */
+
+  /* tiling */
+  uint32_t rows_per_tile;
+  uint32_t lds_bytes;            /* dynamic LDS per block */
+};
+
+/* simple-agg accumulators (device buffer, one per agg) */
+struct SimpleAggAcc {
+  unsigned long long cnt;
+  unsigned long long sum_lo;
+  unsigned long long sum_hi;     /* two's-complement high word */
+};
+
+/* hash-agg table (device): parallel arrays.
+ * keys[]: EMPTY sentinel = INT64_MIN bias — the real INT64_MIN key and the
+ * NULL key get dedicated accumulator blocks (reserved[0]=int64_min,
+ * reserved[1]=null). */
+struct HashAggTable {
+  long long *keys;               /* [table_size] */
+  SimpleAggAcc *accs;            /* [table_size * n_aggs] */
+  SimpleAggAcc *reserved;        /* [2 * n_aggs] */
+  unsigned long long *rsvd_seen; /* [2]: row counts for the 2 reserved keys */
+  unsigned int *error;           /* [0]=table full, [1]=parse error */
+  unsigned long long *n_groups;  /* occupied slot count */
+};
+
+/* project-mode outputs */
+struct ProjectOut {
+  unsigned long long *cells;     /* [n_rows * n_out]: off(44) | len(20);
+                                    len 0xFFFFF => missing column */
+  long long *handles;            /* [n_rows] if any handle col */
+  uint8_t *keep;                 /* [n_rows] */
+  unsigned int *error;
+};
+
+int dev_scan_launch(const ScanPlan &plan, const DevRegion &rgn,
+                    SimpleAggAcc *d_simple, const HashAggTable *ht,
+                    const ProjectOut *po, void *stream);
+int dev_crc64_launch(const DevRegion &rgn, const uint64_t *d_tables /*8*256*/,
+                     unsigned long long *d_xor, void *stream);
+
+}  // namespace copr
+#endif

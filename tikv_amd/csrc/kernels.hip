@@ -1,0 +1,621 @@
+/* kernels.hip — MI355X (gfx950/CDNA4) kernels for the coprocessor hot path.
+ *
+ * Every kernel here is HBM-bandwidth-bound integer/byte work (no MFMA — see
+ * DESIGN.md §4). The shared design center: a 256-thread workgroup stages the
+ * contiguous byte range of its row tile into LDS with coalesced uint4 loads,
+ * then each lane parses its row from LDS at random byte offsets. This turns
+ * the reference's per-KV branchy varint walk
+ * (table_scan_executor.rs:209-256 process_v1) into HBM-sequential traffic.
+ *
+ * Wavefront = 64 (CDNA4); grid-stride over row tiles so a launch covers the
+ * chip's 256 CUs across all 8 XCDs (consecutive tiles land on different XCDs,
+ * block b -> XCD b%8).
+ */
+#include <hip/hip_runtime.h>
+#include "copr_internal.h"
+
+namespace copr {
+
+#define THREADS 256u
+
+/* ---------------- LDS staging ----------------
+ * Copy [gbase, gbase+len) of src into lds (lds holds an aligned superset;
+ * returns the intra-LDS shift of gbase). Coalesced uint4 loads on the
+ * 16B-aligned floor. */
+__device__ static inline uint32_t stage_tile(const uint8_t *__restrict__ src,
+                                             uint64_t gbase, uint32_t len,
+                                             uint8_t *lds) {
+  uint64_t abase = gbase & ~15ull;
+  uint32_t shift = (uint32_t)(gbase - abase);
+  uint32_t total = (len + shift + 15u) & ~15u;
+  const uint4 *gs = (const uint4 *)(src + abase);
+  uint4 *ld = (uint4 *)lds;
+  for (uint32_t i = threadIdx.x; i * 16u < total; i += blockDim.x)
+    ld[i] = gs[i];
+  __syncthreads();
+  return shift;
+}
+
+/* ---------------- row-v1 parse from LDS ----------------
+ * datum split lengths per flag: datum.rs:1117-1155; varint: number.rs:445-520 */
+__device__ static inline bool d_var_u64(const uint8_t *p, uint32_t rem,
+                                        uint64_t *v, uint32_t *n) {
+  uint64_t val = 0;
+  uint32_t i = 0;
+  int shift = 0;
+  while (i < rem && i < 9u) {
+    uint64_t b = p[i];
+    val |= (b & 0x7f) << shift;
+    i++;
+    if (b < 0x80) { *v = val; *n = i; return true; }
+    shift += 7;
+  }
+  if (i == 9u && i < rem) {       /* 10th byte contributes 1 bit */
+    val |= (uint64_t)(p[9] & 0x01) << 63;
+    *v = val; *n = 10; return true;
+  }
+  return false;
+}
+__device__ static inline bool d_var_i64(const uint8_t *p, uint32_t rem,
+                                        int64_t *v, uint32_t *n) {
+  uint64_t uv;
+  if (!d_var_u64(p, rem, &uv, n)) return false;
+  uint64_t half = uv >> 1;
+  *v = (uv & 1) ? (int64_t)~half : (int64_t)half;
+  return true;
+}
+__device__ static inline uint64_t d_be_u64(const uint8_t *p) {
+  uint64_t v = 0;
+  #pragma unroll
+  for (int i = 0; i < 8; i++) v = (v << 8) | p[i];
+  return v;
+}
+
+/* decimal payload -> scaled int (prec<=18); mirrors read_decimal
+ * (decimal.rs:2204-2289) restricted to the scaled-i64 fast path.
+ * Returns consumed bytes or 0 on error. */
+__device__ static const uint32_t TEN_POW_D[10] = {1,10,100,1000,10000,100000,
+  1000000,10000000,100000000,1000000000};
+__device__ static const uint8_t DIG2B[10] = {0,1,1,2,2,3,3,4,4,4};
+
+__device__ static inline uint32_t d_decimal_scaled(const uint8_t *p, uint32_t rem,
+                                                   int64_t *scaled, int32_t *frac) {
+  if (rem < 3) return 0;
+  uint32_t prec = p[0], fr = p[1];
+  if (prec < fr || prec > 18) return 0;
+  uint32_t int_cnt = prec - fr;
+  uint32_t iw = int_cnt / 9, ld = int_cnt - iw * 9;
+  uint32_t fw = fr / 9, td = fr - fw * 9;
+  uint32_t need = 2 + iw * 4 + DIG2B[ld] + fw * 4 + DIG2B[td];
+  if (rem < need) return 0;
+  const uint8_t *q = p + 2;
+  uint32_t mask = (q[0] & 0x80) ? 0u : 0xFFFFFFFFu;
+  bool neg = mask != 0;
+  bool first = true;
+  uint64_t acc = 0;
+  auto rd_word = [&](uint32_t size) -> uint32_t {
+    uint8_t b0 = q[0];
+    if (first) { b0 ^= 0x80; first = false; }
+    uint32_t r;
+    switch (size) {
+      case 1: r = (uint32_t)(int32_t)(int8_t)b0; break;
+      case 2: r = (uint32_t)(((int32_t)(int8_t)b0 << 8) + (int32_t)q[1]); break;
+      case 3: r = (b0 & 128) ? ((255u << 24) | ((uint32_t)b0 << 16) |
+                                ((uint32_t)q[1] << 8) | q[2])
+                             : (((uint32_t)b0 << 16) | ((uint32_t)q[1] << 8) | q[2]);
+              break;
+      default: r = (uint32_t)(((int32_t)(int8_t)b0 << 24) + ((int32_t)q[1] << 16) +
+                              ((int32_t)q[2] << 8) + (int32_t)q[3]); break;
+    }
+    q += size;
+    return r;
+  };
+  if (ld) {
+    uint32_t w = rd_word(DIG2B[ld]) ^ mask;
+    if (w >= TEN_POW_D[ld + 1]) return 0;
+    acc = w;
+  }
+  for (uint32_t k = 0; k < iw; k++) {
+    uint32_t w = rd_word(4) ^ mask;
+    if (w > 999999999u) return 0;
+    acc = acc * 1000000000ull + w;
+  }
+  for (uint32_t k = 0; k < fw; k++) {
+    uint32_t w = rd_word(4) ^ mask;
+    if (w > 999999999u) return 0;
+    acc = acc * 1000000000ull + w;
+  }
+  if (td) {
+    uint32_t w = rd_word(DIG2B[td]) ^ mask;
+    if (w >= TEN_POW_D[td]) return 0;
+    acc = acc * TEN_POW_D[td] + w;
+  }
+  *scaled = neg ? -(int64_t)acc : (int64_t)acc;
+  *frac = (int32_t)fr;
+  return need;
+}
+
+/* length of the first datum (flag+payload) at p; 0 on error.
+ * decoded int value delivered for INT/UINT/VAR_INT/VAR_UINT; *is_null set for
+ * NIL. (split_datum, datum.rs:1117-1155) */
+struct CellView {
+  uint32_t len;        /* full datum length incl flag; 0 = error */
+  uint8_t flag;
+  bool is_null;
+  bool has_int;
+  int64_t ival;
+  bool has_dec;
+  int64_t dscaled; int32_t dfrac;
+};
+
+__device__ static inline void d_parse_datum(const uint8_t *p, uint32_t rem, CellView *cv) {
+  cv->len = 0; cv->is_null = false; cv->has_int = false; cv->has_dec = false;
+  if (rem == 0) return;
+  uint8_t flag = p[0];
+  cv->flag = flag;
+  const uint8_t *pl = p + 1;
+  uint32_t prem = rem - 1;
+  switch (flag) {
+    case 0:  /* NIL */
+      cv->is_null = true; cv->len = 1; return;
+    case 3:  /* INT: BE ^ sign */
+      if (prem < 8) return;
+      cv->has_int = true;
+      cv->ival = (int64_t)(d_be_u64(pl) ^ 0x8000000000000000ull);
+      cv->len = 9; return;
+    case 4:  /* UINT: BE */
+      if (prem < 8) return;
+      cv->has_int = true;
+      cv->ival = (int64_t)d_be_u64(pl);
+      cv->len = 9; return;
+    case 5: case 7:  /* FLOAT / DURATION: 8B */
+      if (prem < 8) return;
+      cv->len = 9; return;
+    case 8: {  /* VAR_INT */
+      int64_t v; uint32_t n;
+      if (!d_var_i64(pl, prem, &v, &n)) return;
+      cv->has_int = true; cv->ival = v; cv->len = 1 + n; return;
+    }
+    case 9: {  /* VAR_UINT */
+      uint64_t v; uint32_t n;
+      if (!d_var_u64(pl, prem, &v, &n)) return;
+      cv->has_int = true; cv->ival = (int64_t)v; cv->len = 1 + n; return;
+    }
+    case 2: {  /* COMPACT_BYTES: varint len + raw */
+      int64_t l; uint32_t n;
+      if (!d_var_i64(pl, prem, &l, &n)) return;
+      if (l < 0 || n + (uint64_t)l > prem) return;
+      cv->len = 1 + n + (uint32_t)l; return;
+    }
+    case 1: {  /* BYTES: 9-byte groups until marker != 0xFF */
+      uint32_t pos = 0;
+      for (;;) {
+        if (pos + 9 > prem) return;
+        uint8_t marker = pl[pos + 8];
+        pos += 9;
+        if (marker != 0xFF) {
+          if (0xFF - marker > 8) return;
+          break;
+        }
+      }
+      cv->len = 1 + pos; return;
+    }
+    case 6: {  /* DECIMAL */
+      int64_t sc; int32_t fr;
+      uint32_t n = d_decimal_scaled(pl, prem, &sc, &fr);
+      if (!n) {
+        /* could still be a valid >18-digit decimal: length-only fallback */
+        if (prem < 2) return;
+        uint32_t prec = pl[0], frc = pl[1];
+        if (prec < frc) return;
+        uint32_t int_cnt = prec - frc;
+        uint32_t iw = int_cnt / 9, ldg = int_cnt - iw * 9;
+        uint32_t fwc = frc / 9, tdg = frc - fwc * 9;
+        uint32_t need = 2 + iw * 4 + DIG2B[ldg] + fwc * 4 + DIG2B[tdg];
+        if (prem < need) return;
+        cv->len = 1 + need; return;
+      }
+      cv->has_dec = true; cv->dscaled = sc; cv->dfrac = fr;
+      cv->len = 1 + n; return;
+    }
+    default:
+      return;  /* JSON/VECTOR unsupported */
+  }
+}
+
+/* parse one row-v1 value; deliver each (col_id, cell offset, CellView) to the
+ * caller loop. Returns false on parse error. */
+#define ROW_FOREACH_BEGIN(vp, vlen)                                     \
+  {                                                                     \
+    uint32_t _pos = 0;                                                  \
+    if (!((vlen) == 0 || ((vlen) == 1 && (vp)[0] == 0))) {              \
+      while (_pos < (vlen)) {                                           \
+        if ((vp)[_pos] != 8) { parse_ok = false; break; }               \
+        _pos++;                                                         \
+        int64_t _cid; uint32_t _n;                                      \
+        if (!d_var_i64((vp) + _pos, (vlen) - _pos, &_cid, &_n)) {       \
+          parse_ok = false; break; }                                    \
+        _pos += _n;                                                     \
+        CellView _cv;                                                   \
+        d_parse_datum((vp) + _pos, (vlen) - _pos, &_cv);                \
+        if (_cv.len == 0) { parse_ok = false; break; }                  \
+        const uint32_t cell_off = _pos; (void)cell_off;                 \
+        const int64_t cell_id = _cid; const CellView &cell = _cv;
+
+#define ROW_FOREACH_END()                                               \
+        _pos += _cv.len;                                                \
+      }                                                                 \
+    }                                                                   \
+  }
+
+/* predicate eval: cmp_int with signed/unsigned variants
+ * (impl_compare.rs:66-160) */
+__device__ static inline int d_cmp_int(int64_t l, int64_t r, bool lu, bool ru) {
+  if (lu && ru) { uint64_t a = (uint64_t)l, b = (uint64_t)r; return a < b ? -1 : a > b ? 1 : 0; }
+  if (!lu && !ru) return l < r ? -1 : l > r ? 1 : 0;
+  if (lu && !ru) {
+    if (r < 0 || (uint64_t)l > 0x7FFFFFFFFFFFFFFFull) return 1;
+    return l < r ? -1 : l > r ? 1 : 0;
+  }
+  if (l < 0 || (uint64_t)r > 0x7FFFFFFFFFFFFFFFull) return -1;
+  return l < r ? -1 : l > r ? 1 : 0;
+}
+__device__ static inline bool d_cmp_res(int32_t kind, int ord) {
+  switch (kind) {
+    case CMP_LT: return ord < 0;
+    case CMP_LE: return ord <= 0;
+    case CMP_GT: return ord > 0;
+    case CMP_GE: return ord >= 0;
+    case CMP_EQ: return ord == 0;
+    default:     return ord != 0;
+  }
+}
+
+/* 128-bit signed accumulate via two u64 atomics (carry trick) */
+__device__ static inline void atomic_add_i128(unsigned long long *lo,
+                                              unsigned long long *hi, int64_t x) {
+  unsigned long long ux = (unsigned long long)x;
+  unsigned long long old = atomicAdd(lo, ux);
+  unsigned long long carry = (old + ux < old) ? 1ull : 0ull;
+  long long hi_add = (long long)carry + (x < 0 ? -1ll : 0ll);
+  if (hi_add) atomicAdd(hi, (unsigned long long)hi_add);
+}
+
+/* ---------------- the fused scan kernel ---------------- */
+__global__ void __launch_bounds__(THREADS)
+k_scan(ScanPlan plan,
+       const uint8_t *__restrict__ vals, const uint64_t *__restrict__ val_offs,
+       const uint8_t *__restrict__ keys, const uint64_t *__restrict__ key_offs,
+       uint64_t n_rows,
+       SimpleAggAcc *__restrict__ simple_acc,
+       HashAggTable ht,
+       ProjectOut po) {
+  extern __shared__ __attribute__((aligned(16))) uint8_t lds[];
+
+  const uint32_t rpt = plan.rows_per_tile;
+  const uint64_t n_tiles = (n_rows + rpt - 1) / rpt;
+
+  /* per-lane simple-agg accumulators (reduced per block at the end) */
+  unsigned long long l_cnt[COPR_MAX_AGGS];
+  long long l_lo[COPR_MAX_AGGS];
+  long long l_hi[COPR_MAX_AGGS];
+  for (int a = 0; a < COPR_MAX_AGGS; a++) { l_cnt[a] = 0; l_lo[a] = 0; l_hi[a] = 0; }
+
+  bool any_parse_err = false;
+
+  for (uint64_t tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
+    uint64_t row0 = tile * rpt;
+    uint64_t row1 = min(row0 + rpt, n_rows);
+    uint64_t gbase = val_offs[row0];
+    uint32_t tlen = (uint32_t)(val_offs[row1] - gbase);
+    __syncthreads();   /* previous tile's lanes done with LDS */
+    uint32_t shift = stage_tile(vals, gbase, tlen, lds);
+
+    uint64_t my_row = row0 + threadIdx.x;
+    if (my_row < row1) {
+      const uint8_t *vp = lds + shift + (uint32_t)(val_offs[my_row] - gbase);
+      uint32_t vlen = (uint32_t)(val_offs[my_row + 1] - val_offs[my_row]);
+      bool parse_ok = true;
+
+      /* single parse pass: collect everything the plan needs */
+      bool filt_found = false, filt_null = false; int64_t filt_v = 0;
+      bool grp_found = false, grp_null = false; int64_t grp_v = 0;
+      struct { bool found; bool null; int64_t iv; int64_t dsc; int32_t dfr;
+               bool has_dec; uint32_t off, len; } cols[COPR_MAX_AGGS];
+      for (int a = 0; a < plan.n_aggs; a++) cols[a] = {false, false, 0, 0, 0, false, 0, 0};
+      unsigned long long cell_pack[COPR_MAX_OUT_COLS];
+      for (int j = 0; j < plan.n_out; j++) cell_pack[j] = 0xFFFFFull; /* missing */
+
+      int needed = (plan.has_filter ? 1 : 0) + (plan.mode == 2 ? 1 : 0) + plan.n_out;
+      for (int a = 0; a < plan.n_aggs; a++)
+        if (plan.aggs[a].kind != DAGG_COUNT_ROWS) needed++;
+      int found = 0;
+
+      ROW_FOREACH_BEGIN(vp, vlen)
+        if (plan.has_filter && !filt_found && cell_id == plan.filter_col_id) {
+          filt_found = true;
+          if (cell.is_null) filt_null = true;
+          else if (cell.has_int) filt_v = cell.ival;
+          else parse_ok = false;
+          found++;
+        }
+        if (plan.mode == 2 && !grp_found && cell_id == plan.group_col_id) {
+          grp_found = true;
+          if (cell.is_null) grp_null = true;
+          else if (cell.has_int) grp_v = cell.ival;
+          else parse_ok = false;
+          found++;
+        }
+        for (int a = 0; a < plan.n_aggs; a++) {
+          if (plan.aggs[a].kind == DAGG_COUNT_ROWS || cols[a].found) continue;
+          if (cell_id == plan.aggs[a].col_id) {
+            cols[a].found = true;
+            cols[a].null = cell.is_null;
+            cols[a].iv = cell.ival;
+            cols[a].has_dec = cell.has_dec;
+            cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
+            if (!cell.is_null && !cell.has_int && !cell.has_dec) parse_ok = false;
+            found++;
+          }
+        }
+        for (int j = 0; j < plan.n_out; j++) {
+          if (plan.out_is_handle[j] || cell_pack[j] != 0xFFFFFull) continue;
+          if (cell_id == plan.out_col_ids[j]) {
+            uint64_t goff = val_offs[my_row] + cell_off;
+            cell_pack[j] = (goff << 20) | (cell.len & 0xFFFFFu);
+            found++;
+          }
+        }
+        if (found >= needed) break;
+      ROW_FOREACH_END()
+
+      if (!parse_ok) {
+        any_parse_err = true;
+      } else {
+        /* filter (missing column without default/NULL handling on device:
+           absent column value is NULL per default-fill — the host validated
+           that filter/agg/group columns are NULLable-default columns) */
+        bool keep = true;
+        if (plan.has_filter) {
+          if (plan.filter_const_null || !filt_found || filt_null) keep = false;
+          else keep = d_cmp_res(plan.filter_cmp,
+                                d_cmp_int(filt_v, plan.filter_const,
+                                          plan.filter_col_unsigned,
+                                          plan.filter_const_unsigned));
+        }
+        if (plan.mode == 0) {
+          po.keep[my_row] = keep ? 1 : 0;
+          if (keep) {
+            for (int j = 0; j < plan.n_out; j++) {
+              if (plan.out_is_handle[j]) {
+                const uint8_t *kp = keys + key_offs[my_row];
+                uint64_t h = d_be_u64(kp + 11) ^ 0x8000000000000000ull;
+                po.handles[my_row] = (long long)h;
+              } else {
+                po.cells[my_row * plan.n_out + j] = cell_pack[j];
+              }
+            }
+          }
+        } else if (keep) {
+          if (plan.mode == 1) {
+            for (int a = 0; a < plan.n_aggs; a++) {
+              const DevAggSpec &sp = plan.aggs[a];
+              if (sp.kind == DAGG_COUNT_ROWS) { l_cnt[a]++; continue; }
+              if (!cols[a].found || cols[a].null) continue;
+              if (sp.kind == DAGG_COUNT_COL) { l_cnt[a]++; continue; }
+              int64_t v;
+              if (sp.kind == DAGG_SUM_INT) v = cols[a].iv;
+              else {  /* SUM_DEC: rescale to target frac */
+                if (!cols[a].has_dec) { any_parse_err = true; continue; }
+                int d = sp.target_frac - cols[a].dfr;
+                if (d < 0 || d > 18) { any_parse_err = true; continue; }
+                int64_t scale = 1;
+                for (int t = 0; t < d; t++) scale *= 10;
+                v = cols[a].dsc * scale;
+              }
+              l_cnt[a]++;
+              /* local i128 accumulate */
+              unsigned long long old = (unsigned long long)l_lo[a];
+              unsigned long long nv = old + (unsigned long long)v;
+              l_hi[a] += (nv < old ? 1 : 0) + (v < 0 ? -1 : 0);
+              l_lo[a] = (long long)nv;
+            }
+          } else {
+            /* hash agg: find/claim slot */
+            SimpleAggAcc *acc_base = nullptr;
+            if (!grp_found || grp_null) {
+              atomicAdd(&ht.rsvd_seen[1], 1ull);
+              acc_base = ht.reserved + 1 * plan.n_aggs;
+            } else if (grp_v == (long long)0x8000000000000000ll) {
+              atomicAdd(&ht.rsvd_seen[0], 1ull);
+              acc_base = ht.reserved + 0 * plan.n_aggs;
+            } else {
+              uint64_t h = (uint64_t)grp_v * 0x9E3779B97F4A7C15ull;
+              h ^= h >> 29;
+              uint32_t mask = plan.table_size - 1u;
+              uint32_t slot = (uint32_t)(h & mask);
+              const unsigned long long EMPTY = 0x8000000000000000ull;
+              for (uint32_t probe = 0; ; probe++) {
+                if (probe > mask) { atomicOr(ht.error, 1u); break; }
+                unsigned long long cur =
+                    atomicCAS((unsigned long long *)&ht.keys[slot], EMPTY,
+                              (unsigned long long)grp_v);
+                if (cur == EMPTY) {
+                  atomicAdd(ht.n_groups, 1ull);
+                  acc_base = ht.accs + (uint64_t)slot * plan.n_aggs;
+                  break;
+                }
+                if (cur == (unsigned long long)grp_v) {
+                  acc_base = ht.accs + (uint64_t)slot * plan.n_aggs;
+                  break;
+                }
+                slot = (slot + 1) & mask;
+              }
+            }
+            if (acc_base) {
+              for (int a = 0; a < plan.n_aggs; a++) {
+                const DevAggSpec &sp = plan.aggs[a];
+                if (sp.kind == DAGG_COUNT_ROWS) { atomicAdd(&acc_base[a].cnt, 1ull); continue; }
+                if (!cols[a].found || cols[a].null) continue;
+                if (sp.kind == DAGG_COUNT_COL) { atomicAdd(&acc_base[a].cnt, 1ull); continue; }
+                int64_t v;
+                if (sp.kind == DAGG_SUM_INT) v = cols[a].iv;
+                else {
+                  if (!cols[a].has_dec) { any_parse_err = true; continue; }
+                  int d = sp.target_frac - cols[a].dfr;
+                  if (d < 0 || d > 18) { any_parse_err = true; continue; }
+                  int64_t scale = 1;
+                  for (int t = 0; t < d; t++) scale *= 10;
+                  v = cols[a].dsc * scale;
+                }
+                atomicAdd(&acc_base[a].cnt, 1ull);
+                atomic_add_i128(&acc_base[a].sum_lo, &acc_base[a].sum_hi, v);
+              }
+            }
+          }
+        }
+      }
+    }
+  }
+
+  /* block-level fold of simple-agg accumulators */
+  if (plan.mode == 1) {
+    for (int a = 0; a < plan.n_aggs; a++) {
+      /* wave reduce then one atomic per wave */
+      unsigned long long c = l_cnt[a];
+      long long lo = l_lo[a], hi = l_hi[a];
+      for (int off = 32; off > 0; off >>= 1) {
+        c += (unsigned long long)__shfl_down((long long)c, off, 64);
+        unsigned long long olo = (unsigned long long)lo;
+        unsigned long long plo = (unsigned long long)__shfl_down((long long)olo, off, 64);
+        long long phi = __shfl_down(hi, off, 64);
+        unsigned long long nlo = olo + plo;
+        hi += phi + (nlo < olo ? 1 : 0);
+        lo = (long long)nlo;
+      }
+      if ((threadIdx.x & 63u) == 0) {
+        if (c) atomicAdd(&simple_acc[a].cnt, c);
+        unsigned long long ulo = (unsigned long long)lo;
+        if (ulo | (unsigned long long)hi) {
+          unsigned long long old = atomicAdd(&simple_acc[a].sum_lo, ulo);
+          long long carry = (old + ulo < old) ? 1 : 0;
+          long long hi_add = hi + carry;
+          if (hi_add) atomicAdd(&simple_acc[a].sum_hi, (unsigned long long)hi_add);
+        }
+      }
+    }
+  }
+  if (any_parse_err) {
+    unsigned int *err = plan.mode == 0 ? po.error : (plan.mode == 2 ? ht.error : nullptr);
+    if (err) atomicOr(err + 1, 1u);
+    else if (simple_acc) atomicOr((unsigned int *)&simple_acc[COPR_MAX_AGGS].cnt, 1u);
+  }
+}
+
+/* ---------------- CRC-64/XZ kernel (checksum.rs:105-114) ----------------
+ * slice-by-8 tables staged in LDS (16 KiB); per-KV digest over key||value,
+ * XOR-fold per wave then one atomicXor per block. */
+__global__ void __launch_bounds__(THREADS)
+k_crc64(const uint8_t *__restrict__ vals, const uint64_t *__restrict__ val_offs,
+        const uint8_t *__restrict__ keys, const uint64_t *__restrict__ key_offs,
+        uint64_t n_rows, uint32_t rows_per_tile, uint32_t key_lds_bytes,
+        const uint64_t *__restrict__ g_tables,
+        unsigned long long *__restrict__ out_xor) {
+  extern __shared__ __attribute__((aligned(16))) uint8_t lds[];
+  uint64_t *tab = (uint64_t *)lds;                 /* 8*256*8 = 16 KiB */
+  uint8_t *key_lds = lds + 8 * 256 * 8;
+  uint8_t *val_lds = key_lds + key_lds_bytes;
+
+  for (uint32_t i = threadIdx.x; i < 2048; i += blockDim.x) tab[i] = g_tables[i];
+
+  const uint32_t rpt = rows_per_tile;
+  const uint64_t n_tiles = (n_rows + rpt - 1) / rpt;
+  unsigned long long acc = 0;
+
+  for (uint64_t tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
+    uint64_t row0 = tile * rpt;
+    uint64_t row1 = min(row0 + rpt, n_rows);
+    __syncthreads();
+    uint64_t kbase = key_offs[row0];
+    uint32_t kshift = stage_tile(keys, kbase, (uint32_t)(key_offs[row1] - kbase), key_lds);
+    uint64_t vbase = val_offs[row0];
+    uint32_t vshift = stage_tile(vals, vbase, (uint32_t)(val_offs[row1] - vbase), val_lds);
+
+    uint64_t my_row = row0 + threadIdx.x;
+    if (my_row < row1) {
+      uint64_t crc = ~0ull;
+      const uint8_t *kp = key_lds + kshift + (uint32_t)(key_offs[my_row] - kbase);
+      uint32_t klen = (uint32_t)(key_offs[my_row + 1] - key_offs[my_row]);
+      const uint8_t *vp = val_lds + vshift + (uint32_t)(val_offs[my_row] - vbase);
+      uint32_t vlen = (uint32_t)(val_offs[my_row + 1] - val_offs[my_row]);
+      /* key (short): bytewise */
+      for (uint32_t i = 0; i < klen; i++)
+        crc = tab[(uint32_t)((crc ^ kp[i]) & 0xFF)] ^ (crc >> 8);
+      /* value: slice-by-8 from LDS */
+      uint32_t i = 0;
+      for (; i + 8 <= vlen; i += 8) {
+        uint64_t x;
+        /* unaligned 8-byte read from LDS */
+        memcpy(&x, vp + i, 8);
+        crc ^= x;
+        crc = tab[7 * 256 + (uint32_t)(crc & 0xFF)] ^
+              tab[6 * 256 + (uint32_t)((crc >> 8) & 0xFF)] ^
+              tab[5 * 256 + (uint32_t)((crc >> 16) & 0xFF)] ^
+              tab[4 * 256 + (uint32_t)((crc >> 24) & 0xFF)] ^
+              tab[3 * 256 + (uint32_t)((crc >> 32) & 0xFF)] ^
+              tab[2 * 256 + (uint32_t)((crc >> 40) & 0xFF)] ^
+              tab[1 * 256 + (uint32_t)((crc >> 48) & 0xFF)] ^
+              tab[0 * 256 + (uint32_t)(crc >> 56)];
+      }
+      for (; i < vlen; i++)
+        crc = tab[(uint32_t)((crc ^ vp[i]) & 0xFF)] ^ (crc >> 8);
+      acc ^= ~crc;
+    }
+  }
+  /* wave XOR-fold, one atomic per wave */
+  for (int off = 32; off > 0; off >>= 1)
+    acc ^= (unsigned long long)__shfl_down((long long)acc, off, 64);
+  if ((threadIdx.x & 63u) == 0 && acc) atomicXor(out_xor, acc);
+}
+
+/* ---------------- launch wrappers ---------------- */
+int dev_scan_launch(const ScanPlan &plan, const DevRegion &rgn,
+                    SimpleAggAcc *d_simple, const HashAggTable *ht,
+                    const ProjectOut *po, void *stream) {
+  uint64_t n_tiles = (rgn.n_kv + plan.rows_per_tile - 1) / plan.rows_per_tile;
+  uint32_t grid = (uint32_t)(n_tiles < 4096 ? n_tiles : 4096);
+  if (grid == 0) grid = 1;
+  HashAggTable ht_v{};
+  ProjectOut po_v{};
+  if (ht) ht_v = *ht;
+  if (po) po_v = *po;
+  hipLaunchKernelGGL(k_scan, dim3(grid), dim3(THREADS), plan.lds_bytes,
+                     (hipStream_t)stream, plan,
+                     rgn.d_vals, rgn.d_val_offs, rgn.d_keys, rgn.d_key_offs,
+                     rgn.n_kv, d_simple, ht_v, po_v);
+  return (int)hipGetLastError();
+}
+
+int dev_crc64_launch(const DevRegion &rgn, const uint64_t *d_tables,
+                     unsigned long long *d_xor, void *stream) {
+  /* LDS: 16 KiB tables + key tile + value tile */
+  uint32_t rpt = 256;
+  uint32_t max_key = 64;   /* record keys are 19 B; allow some slack */
+  uint64_t max_tile_val = (uint64_t)rpt * (rgn.max_row_bytes + 16) + 32;
+  while (rpt > 64 && 16384 + rpt * max_key + max_tile_val > 160 * 1024 - 1024) {
+    rpt /= 2;
+    max_tile_val = (uint64_t)rpt * (rgn.max_row_bytes + 16) + 32;
+  }
+  uint32_t key_lds = rpt * max_key;
+  uint32_t lds_bytes = 16384 + key_lds + (uint32_t)max_tile_val;
+  uint64_t n_tiles = (rgn.n_kv + rpt - 1) / rpt;
+  uint32_t grid = (uint32_t)(n_tiles < 4096 ? n_tiles : 4096);
+  if (grid == 0) grid = 1;
+  hipLaunchKernelGGL(k_crc64, dim3(grid), dim3(THREADS), lds_bytes,
+                     (hipStream_t)stream,
+                     rgn.d_vals, rgn.d_val_offs, rgn.d_keys, rgn.d_key_offs,
+                     rgn.n_kv, rpt, key_lds, d_tables, d_xor);
+  return (int)hipGetLastError();
+}
+
+}  // namespace copr

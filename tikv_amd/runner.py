@@ -1,0 +1,307 @@
+"""High-level request builder + engine wrappers (test/bench orchestration).
+
+`DagSelect` mirrors the reference's request-builder fixture
+(test_coprocessor/src/dag.rs:21 DagSelect; ProductTable fixture.rs:25):
+it assembles the decoded descriptor structs of include/copr_types.h that the
+Rust shim would produce from a tipb::DagRequest.
+"""
+import ctypes as C
+
+from . import _ffi as F
+
+
+def field_type(tp=F.TP_LONGLONG, flag=0, flen=-1, decimal=-1, collate=63):
+    return F.CoprFieldType(tp, flag, flen, decimal, collate)
+
+
+class Col:
+    """Scan column descriptor (tipb::ColumnInfo subset)."""
+
+    def __init__(self, column_id, tp=F.TP_LONGLONG, flag=0, decimal=-1,
+                 pk_handle=False, default_val=None):
+        self.column_id = column_id
+        self.ft = field_type(tp, flag, decimal=decimal)
+        self.pk_handle = pk_handle
+        self.default_val = default_val
+
+
+class Expr:
+    """RPN expression builder."""
+
+    def __init__(self):
+        self.nodes = []
+        self._keep = []
+
+    def col(self, offset, ft=None):
+        n = F.CoprExprNode()
+        n.kind = F.EXPR_COLUMN_REF
+        n.i64_val = offset
+        if ft is not None:
+            n.ft = ft
+        self.nodes.append(n)
+        return self
+
+    def const_int(self, v, unsigned=False):
+        n = F.CoprExprNode()
+        n.kind = F.EXPR_CONST_UINT if unsigned else F.EXPR_CONST_INT
+        n.i64_val = C.c_int64(v).value if not unsigned else C.c_int64(v & (2**64 - 1) if v >= 0 else v).value
+        n.ft = field_type(F.TP_LONGLONG, F.FLAG_UNSIGNED if unsigned else 0)
+        self.nodes.append(n)
+        return self
+
+    def const_null(self):
+        n = F.CoprExprNode()
+        n.kind = F.EXPR_CONST_NULL
+        n.ft = field_type(F.TP_LONGLONG)
+        self.nodes.append(n)
+        return self
+
+    def func(self, sig, n_args=2, ft=None):
+        n = F.CoprExprNode()
+        n.kind = F.EXPR_SCALAR_FUNC
+        n.sig = sig
+        n.n_args = n_args
+        n.ft = ft if ft is not None else field_type(F.TP_LONGLONG)
+        self.nodes.append(n)
+        return self
+
+    def build(self):
+        arr = (F.CoprExprNode * len(self.nodes))(*self.nodes)
+        self._keep.append(arr)
+        return F.CoprExpr(arr, len(self.nodes)), self._keep
+
+
+class DagSelect:
+    """Builds a CoprDagRequest over a table-scan pipeline."""
+
+    def __init__(self, columns):
+        self.columns = columns
+        self._keep = []
+        self.executors = []
+        self.output_offsets = None
+        self._n_out_schema = len(columns)
+        self._agg_out_cols = 0
+        self._has_group = False
+
+        ex = F.CoprExecutor()
+        ex.kind = F.EXEC_TABLE_SCAN
+        cols = (F.CoprColumnInfo * len(columns))()
+        for i, c in enumerate(columns):
+            cols[i].column_id = c.column_id
+            cols[i].ft = c.ft
+            cols[i].pk_handle = 1 if c.pk_handle else 0
+            if c.default_val:
+                buf = (C.c_uint8 * len(c.default_val)).from_buffer_copy(c.default_val)
+                self._keep.append(buf)
+                cols[i].default_val = C.cast(buf, C.POINTER(C.c_uint8))
+                cols[i].default_val_len = len(c.default_val)
+        self._keep.append(cols)
+        ex.columns = cols
+        ex.n_columns = len(columns)
+        self.executors.append(ex)
+
+    def where(self, expr):
+        """Selection with one condition (an Expr)."""
+        e, keep = expr.build()
+        self._keep += keep
+        conds = (F.CoprExpr * 1)(e)
+        self._keep.append(conds)
+        ex = F.CoprExecutor()
+        ex.kind = F.EXEC_SELECTION
+        ex.conditions = conds
+        ex.n_conditions = 1
+        self.executors.append(ex)
+        return self
+
+    def _agg(self, kind, aggs, group_by=None):
+        ex = F.CoprExecutor()
+        ex.kind = kind
+        arr = (F.CoprAggDef * len(aggs))()
+        out_cols = 0
+        for i, (func, arg_expr, out_ft) in enumerate(aggs):
+            e, keep = arg_expr.build()
+            self._keep += keep
+            arr[i].func = func
+            arr[i].arg = e
+            arr[i].out_ft = out_ft
+            out_cols += 2 if func == F.AGG_AVG else 1
+        self._keep.append(arr)
+        ex.aggs = arr
+        ex.n_aggs = len(aggs)
+        if group_by is not None:
+            ge, keep = group_by.build()
+            self._keep += keep
+            garr = (F.CoprExpr * 1)(ge)
+            self._keep.append(garr)
+            ex.group_by = garr
+            ex.n_group_by = 1
+            out_cols += 1
+            self._has_group = True
+        self.executors.append(ex)
+        self._agg_out_cols = out_cols
+        self._n_out_schema = out_cols
+        return self
+
+    def simple_agg(self, aggs):
+        return self._agg(F.EXEC_SIMPLE_AGG, aggs)
+
+    def hash_agg(self, aggs, group_by):
+        return self._agg(F.EXEC_FAST_HASH_AGG, aggs, group_by)
+
+    def limit(self, n):
+        ex = F.CoprExecutor()
+        ex.kind = F.EXEC_LIMIT
+        ex.limit = n
+        self.executors.append(ex)
+        return self
+
+    def output(self, offsets):
+        self.output_offsets = list(offsets)
+        return self
+
+    def build(self):
+        if self.output_offsets is None:
+            self.output_offsets = list(range(self._n_out_schema))
+        exarr = (F.CoprExecutor * len(self.executors))(*self.executors)
+        offs = (C.c_uint32 * len(self.output_offsets))(*self.output_offsets)
+        self._keep += [exarr, offs]
+        req = F.CoprDagRequest()
+        req.executors = exarr
+        req.n_executors = len(self.executors)
+        req.output_offsets = offs
+        req.n_output_offsets = len(self.output_offsets)
+        self._req = req
+        return req
+
+
+# convenience agg-def helpers (out field types as TiDB would set them:
+# count -> LongLong; sum/avg over int/decimal -> NewDecimal)
+def count_star():
+    return (F.AGG_COUNT, Expr().const_int(1), field_type(F.TP_LONGLONG))
+
+
+def count_col(offset):
+    return (F.AGG_COUNT, Expr().col(offset), field_type(F.TP_LONGLONG))
+
+
+def sum_col(offset, decimal=-1):
+    return (F.AGG_SUM, Expr().col(offset), field_type(F.TP_NEWDECIMAL, decimal=decimal))
+
+
+def avg_col(offset, decimal=-1):
+    return (F.AGG_AVG, Expr().col(offset), field_type(F.TP_NEWDECIMAL, decimal=decimal))
+
+
+def cmp_col_const(offset, sig, const, unsigned_const=False):
+    return Expr().col(offset).const_int(const, unsigned_const).func(sig, 2)
+
+
+class GenRegion:
+    """Synthetic region (host buffers from the product generator)."""
+
+    def __init__(self, config_index, n_rows, table_id=1, first_handle=0,
+                 n_cols=0, row_format=1):
+        lib = F.load_lib()
+        spec = F.CoprGenSpec(config_index, table_id, n_rows, first_handle,
+                             n_cols, row_format)
+        self._out = F.CoprGenOut()
+        st = lib.copr_gen_region(C.byref(spec), C.byref(self._out))
+        if st != 0:
+            raise RuntimeError("generator failed: %d" % st)
+        self._lib = lib
+        self.n_kv = self._out.n_kv
+
+    @property
+    def keys(self):
+        return self._out.keys
+
+    @property
+    def key_offs(self):
+        return self._out.key_offs
+
+    @property
+    def vals(self):
+        return self._out.vals
+
+    @property
+    def val_offs(self):
+        return self._out.val_offs
+
+    def key_bytes(self):
+        return self._out.key_offs[self.n_kv]
+
+    def val_bytes(self):
+        return self._out.val_offs[self.n_kv]
+
+    def close(self):
+        if self._out.keys:
+            self._lib.copr_gen_free(C.byref(self._out))
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
+class Engine:
+    def __init__(self, device=0):
+        lib = F.load_lib()
+        h = C.c_void_p()
+        st = lib.copr_engine_create(device, C.byref(h))
+        if st != 0:
+            raise RuntimeError("copr_engine_create: %d (%s)" %
+                               (st, lib.copr_last_error().decode()))
+        self._lib = lib
+        self._h = h
+
+    def region(self, gen: GenRegion):
+        r = C.c_void_p()
+        st = self._lib.copr_region_create(self._h, gen.keys, gen.key_offs,
+                                          gen.vals, gen.val_offs, gen.n_kv,
+                                          C.byref(r))
+        if st != 0:
+            raise RuntimeError("copr_region_create: %d (%s)" %
+                               (st, self._lib.copr_last_error().decode()))
+        return Region(self, r)
+
+    def dag_run(self, req, regions):
+        arr = (C.c_void_p * len(regions))(*[r._h for r in regions])
+        res = F.CoprSelectResult()
+        st = self._lib.copr_dag_run(self._h, C.byref(req), arr, len(regions),
+                                    C.byref(res))
+        if st != 0:
+            raise RuntimeError("copr_dag_run: %d (%s)" %
+                               (st, self._lib.copr_last_error().decode()))
+        data = C.string_at(res.data, res.data_len) if res.data_len else b""
+        n_rows = res.n_rows
+        self._lib.copr_result_free(C.byref(res))
+        return data, n_rows
+
+    def checksum(self, regions):
+        arr = (C.c_void_p * len(regions))(*[r._h for r in regions])
+        cs = C.c_uint64()
+        kvs = C.c_uint64()
+        byts = C.c_uint64()
+        st = self._lib.copr_checksum(self._h, arr, len(regions), C.byref(cs),
+                                     C.byref(kvs), C.byref(byts))
+        if st != 0:
+            raise RuntimeError("copr_checksum: %d (%s)" %
+                               (st, self._lib.copr_last_error().decode()))
+        return cs.value, kvs.value, byts.value
+
+    def close(self):
+        if self._h:
+            self._lib.copr_engine_destroy(self._h)
+            self._h = None
+
+
+class Region:
+    def __init__(self, eng, h):
+        self._eng = eng
+        self._h = h
+
+    def close(self):
+        if self._h:
+            self._eng._lib.copr_region_destroy(self._h)
+            self._h = None
