@@ -445,6 +445,15 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
 
   std::vector<uint8_t> resp;
   uint64_t n_rows_out = 0;
+  /* HIP-event timing of the scan kernel(s), on the engine's own stream —
+     reported in summaries[0].time_processed_ns (the ExecSummary surface the
+     reference fills per executor slot, execute_stats.rs:43-76). bench.py's
+     roofline leg reads this. */
+  hipEvent_t ev_a = nullptr, ev_b = nullptr;
+  hipEventCreate(&ev_a);
+  hipEventCreate(&ev_b);
+  float kernel_ms = 0.0f;
+  bool timed = false;
 
   if (pl.sp.mode == 1 || pl.sp.mode == 2) {
     if (pl.sp.mode == 1) {
@@ -453,12 +462,15 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
       HIP_TRY(hipMalloc(&d_acc, sizeof(SimpleAggAcc) * (COPR_MAX_AGGS + 1)), "acc alloc");
       HIP_TRY(hipMemsetAsync(d_acc, 0, sizeof(SimpleAggAcc) * (COPR_MAX_AGGS + 1),
                              eng->stream), "acc memset");
+      hipEventRecord(ev_a, eng->stream);
       for (uint32_t rg = 0; rg < n_regions; rg++) {
         ScanPlan sp = pl.sp;
         pick_tiling(regions[rg]->dev, &sp);
         int e = dev_scan_launch(sp, regions[rg]->dev, d_acc, nullptr, nullptr, eng->stream);
         if (e) { hipFree(d_acc); return SET_ERR(COPR_ERR_INTERNAL, "scan launch failed"); }
       }
+      hipEventRecord(ev_b, eng->stream);
+      timed = true;
       SimpleAggAcc h_acc[COPR_MAX_AGGS + 1];
       hipError_t ce = hipMemcpyAsync(h_acc, d_acc, sizeof(h_acc), hipMemcpyDeviceToHost,
                                      eng->stream);
@@ -518,6 +530,7 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
           if (e == hipSuccess) e = hipStreamSynchronize(eng->stream);
           if (e != hipSuccess) { free_ht(); return SET_ERR(COPR_ERR_INTERNAL, "ht init"); }
         }
+        hipEventRecord(ev_a, eng->stream);
         for (uint32_t rg = 0; rg < n_regions; rg++) {
           ScanPlan sp = pl.sp;
           sp.table_size = tsize;
@@ -525,6 +538,8 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
           int le = dev_scan_launch(sp, regions[rg]->dev, nullptr, &ht, nullptr, eng->stream);
           if (le) { free_ht(); return SET_ERR(COPR_ERR_INTERNAL, "scan launch failed"); }
         }
+        hipEventRecord(ev_b, eng->stream);
+        timed = true;
         unsigned int h_err[2] = {0, 0};
         unsigned long long h_rsvd_seen[2];
         hipError_t ce = hipMemcpyAsync(h_err, ht.error, 8, hipMemcpyDeviceToHost, eng->stream);
@@ -663,14 +678,23 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
     }
   }
 
+  if (timed) {
+    hipEventSynchronize(ev_b);
+    hipEventElapsedTime(&kernel_ms, ev_a, ev_b);
+  }
+  hipEventDestroy(ev_a);
+  hipEventDestroy(ev_b);
   out->data = (uint8_t *)malloc(resp.size() ? resp.size() : 1);
   memcpy(out->data, resp.data(), resp.size());
   out->data_len = resp.size();
   out->n_rows = n_rows_out;
   out->summaries = (CoprExecSummary *)calloc(req->n_executors, sizeof(CoprExecSummary));
   out->n_summaries = req->n_executors;
-  if (out->n_summaries)
+  if (out->n_summaries) {
     out->summaries[out->n_summaries - 1].num_produced_rows = n_rows_out;
+    out->summaries[0].time_processed_ns = (uint64_t)(kernel_ms * 1e6);
+    out->summaries[0].num_iterations = 1;
+  }
   return COPR_OK;
 }
 

@@ -275,8 +275,9 @@ class Engine:
                                (st, self._lib.copr_last_error().decode()))
         data = C.string_at(res.data, res.data_len) if res.data_len else b""
         n_rows = res.n_rows
+        kernel_ns = res.summaries[0].time_processed_ns if res.n_summaries else 0
         self._lib.copr_result_free(C.byref(res))
-        return data, n_rows
+        return data, n_rows, kernel_ns
 
     def checksum(self, regions):
         arr = (C.c_void_p * len(regions))(*[r._h for r in regions])
