@@ -92,6 +92,11 @@ def main():
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
+    # torch first: initialize HIP device discovery before the engine's own
+    # runtime use in this process
+    import torch
+    have_cuda = torch.cuda.is_available()
+
     import tikv_amd
     from tikv_amd import _ffi as F
 
@@ -100,12 +105,11 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     dist = None
     if world > 1:
-        import torch
         import torch.distributed as tdist
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        backend = "nccl" if have_cuda else "gloo"
         tdist.init_process_group(backend=backend)
         dist = tdist
-        if torch.cuda.is_available():
+        if have_cuda:
             torch.cuda.set_device(local_rank)
 
     n_rows = args.rows
@@ -119,8 +123,7 @@ def main():
     req = build_request(tikv_amd, F)
     eng = tikv_amd.Engine(local_rank)
     rgn = eng.region(gen)
-    import torch
-    if torch.cuda.is_available():
+    if have_cuda:
         torch.cuda.set_device(local_rank)
 
     # algorithmic bytes per pass: every encoded value byte once + the
@@ -142,16 +145,19 @@ def main():
         dist.all_reduce(t)
         cnt = int(t.item())
 
-    # timed region
+    # timed region (each step already ends with a hipStreamSynchronize inside
+    # copr_dag_run; torch sync covers any torch-side stream)
     if dist:
         dist.barrier()
-    torch.cuda.synchronize()
+    if have_cuda:
+        torch.cuda.synchronize()
     kern_ns_total = 0
     t0 = time.perf_counter()
     for _ in range(args.steps):
         _, kns = step()
         kern_ns_total += kns
-    torch.cuda.synchronize()
+    if have_cuda:
+        torch.cuda.synchronize()
     if dist:
         dist.barrier()
     elapsed = time.perf_counter() - t0

@@ -131,6 +131,7 @@ namespace {
 
 struct HostPlan {
   ScanPlan sp{};
+  int filter_col_offset = -1;
   /* scan schema */
   std::vector<CoprColumnInfo> cols;
   std::vector<CoprFieldType> out_schema;
@@ -151,9 +152,46 @@ static bool et_int(int32_t tp) {
   }
 }
 
+/* decode an int datum (flag+payload) on host — used for default values.
+ * Mirrors decode_int_datum (datum_codec.rs:401-420). Returns 0 ok/1 null/-1 err. */
+static int host_decode_int_datum(const uint8_t *p, size_t len, int64_t *v) {
+  if (len == 0) return -1;
+  uint8_t flag = p[0];
+  p++; len--;
+  auto be64 = [](const uint8_t *q) {
+    uint64_t x = 0;
+    for (int i = 0; i < 8; i++) x = (x << 8) | q[i];
+    return x;
+  };
+  switch (flag) {
+    case 0: return 1;
+    case 3: if (len < 8) return -1; *v = (int64_t)(be64(p) ^ 0x8000000000000000ull); return 0;
+    case 4: if (len < 8) return -1; *v = (int64_t)be64(p); return 0;
+    case 8: case 9: {
+      uint64_t val = 0;
+      int shift = 0;
+      size_t i = 0;
+      while (i < len && p[i] >= 0x80 && i < 9) {
+        val |= (uint64_t)(p[i] & 0x7F) << shift;
+        shift += 7; i++;
+      }
+      if (i >= len) return -1;
+      val |= (uint64_t)p[i] << shift;
+      if (flag == 8) {
+        uint64_t half = val >> 1;
+        *v = (val & 1) ? (int64_t)~half : (int64_t)half;
+      } else {
+        *v = (int64_t)val;
+      }
+      return 0;
+    }
+    default: return -1;
+  }
+}
+
 /* pattern-match a selection condition: cmp(colref_int, const_int) (either
  * operand order). Mirrors LtInt-class dispatch (lib.rs:523, map_int_sig). */
-static copr_status match_filter(const CoprExpr &cond, const HostPlan &pl, ScanPlan *sp) {
+static copr_status match_filter(const CoprExpr &cond, HostPlan &pl, ScanPlan *sp) {
   if (cond.n_nodes != 3) return COPR_ERR_UNSUPPORTED;
   const CoprExprNode &a = cond.nodes[0], &b = cond.nodes[1], &f = cond.nodes[2];
   if (f.kind != COPR_EXPR_SCALAR_FUNC || f.n_args != 2) return COPR_ERR_UNSUPPORTED;
@@ -184,6 +222,7 @@ static copr_status match_filter(const CoprExpr &cond, const HostPlan &pl, ScanPl
   if (off >= pl.cols.size()) return COPR_ERR_INVALID_REQUEST;
   const CoprColumnInfo &ci = pl.cols[off];
   if (!et_int(ci.ft.tp) || ci.pk_handle) return COPR_ERR_UNSUPPORTED;
+  pl.filter_col_offset = (int)off;
   if (swapped) {
     /* const OP col  ==  col flip(OP) const */
     switch (cmp) {
@@ -196,6 +235,19 @@ static copr_status match_filter(const CoprExpr &cond, const HostPlan &pl, ScanPl
   }
   sp->has_filter = 1;
   sp->filter_col_id = ci.column_id;
+  /* default fill for a row missing this column (scan fills defaults before
+     the predicate decodes the column) */
+  if (ci.default_val && ci.default_val_len) {
+    int64_t dv;
+    int r = host_decode_int_datum(ci.default_val, ci.default_val_len, &dv);
+    if (r < 0) return COPR_ERR_INVALID_REQUEST;
+    sp->filter_missing_null = r == 1 ? 1 : 0;
+    sp->filter_missing_val = dv;
+  } else {
+    /* NULLable: missing -> NULL. (A missing NOT NULL column is a data
+       error in the reference; rows cannot legally lack it.) */
+    sp->filter_missing_null = 1;
+  }
   sp->filter_cmp = cmp;
   sp->filter_const = konst->i64_val;
   sp->filter_col_unsigned = (ci.ft.flag & COPR_FLAG_UNSIGNED) ? 1 : 0;
@@ -598,8 +650,13 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
       if (e == hipSuccess) e = hipMalloc((void **)&po.handles, n * 8);
       if (e == hipSuccess) e = hipMalloc((void **)&po.keep, n + 8);
       if (e == hipSuccess) e = hipMalloc((void **)&po.error, 8);
+      if (e == hipSuccess && pl.sp.has_filter) {
+        e = hipMalloc((void **)&po.filt_vals, n * 8 + 8);
+        if (e == hipSuccess) e = hipMalloc((void **)&po.filt_state, n + 8);
+      }
       auto free_po = [&]() {
         hipFree(po.cells); hipFree(po.handles); hipFree(po.keep); hipFree(po.error);
+        hipFree(po.filt_vals); hipFree(po.filt_state);
       };
       if (e != hipSuccess) { free_po(); return SET_ERR(COPR_ERR_OOM, "project alloc"); }
       hipMemsetAsync(po.keep, 0, n + 8, eng->stream);
@@ -620,6 +677,15 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
       if (ce == hipSuccess && any_handle) {
         h_handles.resize(n);
         ce = hipMemcpyAsync(h_handles.data(), po.handles, n * 8, hipMemcpyDeviceToHost, eng->stream);
+      }
+      std::vector<long long> h_fvals;
+      std::vector<uint8_t> h_fstate;
+      if (ce == hipSuccess && pl.sp.has_filter) {
+        h_fvals.resize(n);
+        h_fstate.resize(n);
+        ce = hipMemcpyAsync(h_fvals.data(), po.filt_vals, n * 8, hipMemcpyDeviceToHost, eng->stream);
+        if (ce == hipSuccess)
+          ce = hipMemcpyAsync(h_fstate.data(), po.filt_state, n, hipMemcpyDeviceToHost, eng->stream);
       }
       if (ce == hipSuccess)
         ce = hipMemcpyAsync(h_err, po.error, 8, hipMemcpyDeviceToHost, eng->stream);
@@ -654,6 +720,18 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
           if (pl.sp.out_is_handle[off]) {
             enc_datum_int(&resp, h_handles[i],
                           (pl.cols[off].ft.flag & COPR_FLAG_UNSIGNED) != 0);
+            continue;
+          }
+          if ((int)off == pl.filter_col_offset && pl.sp.has_filter) {
+            /* the predicate decoded this column in place, so its output is
+               the DECODED datum (ensure_columns_decoded -> lazy_column
+               Decoded encode, lazy_column.rs:165,242; vector.rs:372-383) */
+            uint8_t st8 = h_fstate[i];
+            bool uns = (pl.cols[off].ft.flag & COPR_FLAG_UNSIGNED) != 0;
+            if (st8 == 0) enc_datum_int(&resp, h_fvals[i], uns);
+            else if (st8 == 1) resp.push_back(0);
+            else if (pl.sp.filter_missing_null) resp.push_back(0);
+            else enc_datum_int(&resp, pl.sp.filter_missing_val, uns);
             continue;
           }
           unsigned long long cp = h_cells[i * pl.sp.n_out + off];

@@ -49,6 +49,10 @@ struct ScanPlan {
   int32_t filter_col_unsigned;
   int32_t filter_const_unsigned;
   int32_t filter_const_null;     /* NULL const: predicate never true */
+  /* value a row takes when the filter column is absent: the scan default
+     fill (table_scan_executor.rs:456-483) decoded */
+  int32_t filter_missing_null;   /* 1 => NULL */
+  int64_t filter_missing_val;
 
   int32_t mode;                  /* 0 project, 1 simple agg, 2 hash agg */
   int32_t n_aggs;
@@ -95,6 +99,11 @@ struct ProjectOut {
                                     len 0xFFFFF => missing column */
   long long *handles;            /* [n_rows] if any handle col */
   uint8_t *keep;                 /* [n_rows] */
+  /* the filter column is decoded in place by predicate eval, so its output
+     form is the DECODED datum (lazy_column.rs:165,242 + expr eval
+     ensure_columns_decoded): per-row value + state for the host encoder */
+  long long *filt_vals;          /* [n_rows] */
+  uint8_t *filt_state;           /* [n_rows]: 0 value, 1 NULL, 2 missing */
   unsigned int *error;
 };
 

@@ -377,7 +377,12 @@ k_scan(ScanPlan plan,
            that filter/agg/group columns are NULLable-default columns) */
         bool keep = true;
         if (plan.has_filter) {
-          if (plan.filter_const_null || !filt_found || filt_null) keep = false;
+          bool v_null = filt_null;
+          if (!filt_found) {
+            if (plan.filter_missing_null) v_null = true;
+            else { filt_v = plan.filter_missing_val; v_null = false; }
+          }
+          if (plan.filter_const_null || v_null) keep = false;
           else keep = d_cmp_res(plan.filter_cmp,
                                 d_cmp_int(filt_v, plan.filter_const,
                                           plan.filter_col_unsigned,
@@ -385,6 +390,10 @@ k_scan(ScanPlan plan,
         }
         if (plan.mode == 0) {
           po.keep[my_row] = keep ? 1 : 0;
+          if (plan.has_filter && po.filt_vals) {
+            po.filt_vals[my_row] = filt_v;
+            po.filt_state[my_row] = !filt_found ? 2 : (filt_null ? 1 : 0);
+          }
           if (keep) {
             for (int j = 0; j < plan.n_out; j++) {
               if (plan.out_is_handle[j]) {
