@@ -7,9 +7,15 @@
  * the reference's per-KV branchy varint walk
  * (table_scan_executor.rs:209-256 process_v1) into HBM-sequential traffic.
  *
+ * The aggregation kernels are templated on the aggregate count so every
+ * per-row state is register-resident (a dynamically indexed local array
+ * spills to scratch: 528 B/lane and ~5x wall in the first build of this
+ * kernel). The project kernel (row-returning scans) is separate and
+ * scratch-tolerant — it is not the hot path.
+ *
  * Wavefront = 64 (CDNA4); grid-stride over row tiles so a launch covers the
- * chip's 256 CUs across all 8 XCDs (consecutive tiles land on different XCDs,
- * block b -> XCD b%8).
+ * chip's 256 CUs across all 8 XCDs (consecutive tiles land on different
+ * XCDs, block b -> XCD b%8).
  */
 #include <hip/hip_runtime.h>
 #include "copr_internal.h"
@@ -72,8 +78,7 @@ __device__ static inline uint64_t d_be_u64(const uint8_t *p) {
 }
 
 /* decimal payload -> scaled int (prec<=18); mirrors read_decimal
- * (decimal.rs:2204-2289) restricted to the scaled-i64 fast path.
- * Returns consumed bytes or 0 on error. */
+ * (decimal.rs:2204-2289) restricted to the scaled-i64 fast path. */
 __device__ static const uint32_t TEN_POW_D[10] = {1,10,100,1000,10000,100000,
   1000000,10000000,100000000,1000000000};
 __device__ static const uint8_t DIG2B[10] = {0,1,1,2,2,3,3,4,4,4};
@@ -135,9 +140,7 @@ __device__ static inline uint32_t d_decimal_scaled(const uint8_t *p, uint32_t re
   return need;
 }
 
-/* length of the first datum (flag+payload) at p; 0 on error.
- * decoded int value delivered for INT/UINT/VAR_INT/VAR_UINT; *is_null set for
- * NIL. (split_datum, datum.rs:1117-1155) */
+/* one datum (flag+payload) view */
 struct CellView {
   uint32_t len;        /* full datum length incl flag; 0 = error */
   uint8_t flag;
@@ -204,7 +207,6 @@ __device__ static inline void d_parse_datum(const uint8_t *p, uint32_t rem, Cell
       int64_t sc; int32_t fr;
       uint32_t n = d_decimal_scaled(pl, prem, &sc, &fr);
       if (!n) {
-        /* could still be a valid >18-digit decimal: length-only fallback */
         if (prem < 2) return;
         uint32_t prec = pl[0], frc = pl[1];
         if (prec < frc) return;
@@ -223,8 +225,6 @@ __device__ static inline void d_parse_datum(const uint8_t *p, uint32_t rem, Cell
   }
 }
 
-/* parse one row-v1 value; deliver each (col_id, cell offset, CellView) to the
- * caller loop. Returns false on parse error. */
 #define ROW_FOREACH_BEGIN(vp, vlen)                                     \
   {                                                                     \
     uint32_t _pos = 0;                                                  \
@@ -248,8 +248,7 @@ __device__ static inline void d_parse_datum(const uint8_t *p, uint32_t rem, Cell
     }                                                                   \
   }
 
-/* predicate eval: cmp_int with signed/unsigned variants
- * (impl_compare.rs:66-160) */
+/* predicate eval (impl_compare.rs:66-160) */
 __device__ static inline int d_cmp_int(int64_t l, int64_t r, bool lu, bool ru) {
   if (lu && ru) { uint64_t a = (uint64_t)l, b = (uint64_t)r; return a < b ? -1 : a > b ? 1 : 0; }
   if (!lu && !ru) return l < r ? -1 : l > r ? 1 : 0;
@@ -271,6 +270,20 @@ __device__ static inline bool d_cmp_res(int32_t kind, int ord) {
   }
 }
 
+/* evaluate the plan's filter for one row's decoded filter-column state */
+__device__ static inline bool d_filter_keep(const ScanPlan &plan, bool found,
+                                            bool is_null, int64_t v) {
+  if (!plan.has_filter) return true;
+  if (!found) {
+    if (plan.filter_missing_null) is_null = true;
+    else { v = plan.filter_missing_val; is_null = false; }
+  }
+  if (plan.filter_const_null || is_null) return false;
+  return d_cmp_res(plan.filter_cmp,
+                   d_cmp_int(v, plan.filter_const,
+                             plan.filter_col_unsigned, plan.filter_const_unsigned));
+}
+
 /* 128-bit signed accumulate via two u64 atomics (carry trick) */
 __device__ static inline void atomic_add_i128(unsigned long long *lo,
                                               unsigned long long *hi, int64_t x) {
@@ -281,25 +294,36 @@ __device__ static inline void atomic_add_i128(unsigned long long *lo,
   if (hi_add) atomicAdd(hi, (unsigned long long)hi_add);
 }
 
-/* ---------------- the fused scan kernel ---------------- */
+/* ---------------- fused scan + filter + aggregate ----------------
+ * NAGGS is a compile-time bound so all per-row/per-lane state stays in
+ * registers. IS_HASH selects grouped aggregation. */
+struct AggColView { bool found, null, has_dec; int64_t iv, dsc; int32_t dfr; };
+
+template <int NAGGS, bool IS_HASH>
 __global__ void __launch_bounds__(THREADS)
-k_scan(ScanPlan plan,
-       const uint8_t *__restrict__ vals, const uint64_t *__restrict__ val_offs,
-       const uint8_t *__restrict__ keys, const uint64_t *__restrict__ key_offs,
-       uint64_t n_rows,
-       SimpleAggAcc *__restrict__ simple_acc,
-       HashAggTable ht,
-       ProjectOut po) {
+k_scan_agg(ScanPlan plan,
+           const uint8_t *__restrict__ vals, const uint64_t *__restrict__ val_offs,
+           uint64_t n_rows,
+           SimpleAggAcc *__restrict__ simple_acc,
+           HashAggTable ht) {
   extern __shared__ __attribute__((aligned(16))) uint8_t lds[];
 
   const uint32_t rpt = plan.rows_per_tile;
   const uint64_t n_tiles = (n_rows + rpt - 1) / rpt;
 
-  /* per-lane simple-agg accumulators (reduced per block at the end) */
-  unsigned long long l_cnt[COPR_MAX_AGGS];
-  long long l_lo[COPR_MAX_AGGS];
-  long long l_hi[COPR_MAX_AGGS];
-  for (int a = 0; a < COPR_MAX_AGGS; a++) { l_cnt[a] = 0; l_lo[a] = 0; l_hi[a] = 0; }
+  /* per-lane simple-agg accumulators (registers; folded per wave at end) */
+  unsigned long long l_cnt[NAGGS];
+  unsigned long long l_lo[NAGGS];
+  long long l_hi[NAGGS];
+  #pragma unroll
+  for (int a = 0; a < NAGGS; a++) { l_cnt[a] = 0; l_lo[a] = 0; l_hi[a] = 0; }
+
+  /* how many cells the parse must find before it can stop early
+     (process_v1 stops at decoded == columns_len, table_scan_executor.rs:223) */
+  int needed = (plan.has_filter ? 1 : 0) + (IS_HASH ? 1 : 0);
+  #pragma unroll
+  for (int a = 0; a < NAGGS; a++)
+    if (plan.aggs[a].kind != DAGG_COUNT_ROWS) needed++;
 
   bool any_parse_err = false;
 
@@ -317,18 +341,11 @@ k_scan(ScanPlan plan,
       uint32_t vlen = (uint32_t)(val_offs[my_row + 1] - val_offs[my_row]);
       bool parse_ok = true;
 
-      /* single parse pass: collect everything the plan needs */
       bool filt_found = false, filt_null = false; int64_t filt_v = 0;
       bool grp_found = false, grp_null = false; int64_t grp_v = 0;
-      struct { bool found; bool null; int64_t iv; int64_t dsc; int32_t dfr;
-               bool has_dec; uint32_t off, len; } cols[COPR_MAX_AGGS];
-      for (int a = 0; a < plan.n_aggs; a++) cols[a] = {false, false, 0, 0, 0, false, 0, 0};
-      unsigned long long cell_pack[COPR_MAX_OUT_COLS];
-      for (int j = 0; j < plan.n_out; j++) cell_pack[j] = 0xFFFFFull; /* missing */
-
-      int needed = (plan.has_filter ? 1 : 0) + (plan.mode == 2 ? 1 : 0) + plan.n_out;
-      for (int a = 0; a < plan.n_aggs; a++)
-        if (plan.aggs[a].kind != DAGG_COUNT_ROWS) needed++;
+      AggColView cols[NAGGS];
+      #pragma unroll
+      for (int a = 0; a < NAGGS; a++) cols[a] = {false, false, false, 0, 0, 0};
       int found = 0;
 
       ROW_FOREACH_BEGIN(vp, vlen)
@@ -339,14 +356,15 @@ k_scan(ScanPlan plan,
           else parse_ok = false;
           found++;
         }
-        if (plan.mode == 2 && !grp_found && cell_id == plan.group_col_id) {
+        if (IS_HASH && !grp_found && cell_id == plan.group_col_id) {
           grp_found = true;
           if (cell.is_null) grp_null = true;
           else if (cell.has_int) grp_v = cell.ival;
           else parse_ok = false;
           found++;
         }
-        for (int a = 0; a < plan.n_aggs; a++) {
+        #pragma unroll
+        for (int a = 0; a < NAGGS; a++) {
           if (plan.aggs[a].kind == DAGG_COUNT_ROWS || cols[a].found) continue;
           if (cell_id == plan.aggs[a].col_id) {
             cols[a].found = true;
@@ -357,6 +375,162 @@ k_scan(ScanPlan plan,
             if (!cell.is_null && !cell.has_int && !cell.has_dec) parse_ok = false;
             found++;
           }
+        }
+        if (found >= needed) break;
+      ROW_FOREACH_END()
+
+      if (!parse_ok) {
+        any_parse_err = true;
+      } else if (d_filter_keep(plan, filt_found, filt_null, filt_v)) {
+        SimpleAggAcc *acc_base = nullptr;
+        if (IS_HASH) {
+          if (!grp_found || grp_null) {
+            atomicAdd(&ht.rsvd_seen[1], 1ull);
+            acc_base = ht.reserved + 1 * NAGGS;
+          } else if (grp_v == (long long)0x8000000000000000ll) {
+            atomicAdd(&ht.rsvd_seen[0], 1ull);
+            acc_base = ht.reserved + 0 * NAGGS;
+          } else {
+            uint64_t h = (uint64_t)grp_v * 0x9E3779B97F4A7C15ull;
+            h ^= h >> 29;
+            uint32_t mask = plan.table_size - 1u;
+            uint32_t slot = (uint32_t)(h & mask);
+            const unsigned long long EMPTY = 0x8000000000000000ull;
+            for (uint32_t probe = 0; ; probe++) {
+              if (probe > mask) { atomicOr(ht.error, 1u); break; }
+              unsigned long long cur =
+                  atomicCAS((unsigned long long *)&ht.keys[slot], EMPTY,
+                            (unsigned long long)grp_v);
+              if (cur == EMPTY) {
+                atomicAdd(ht.n_groups, 1ull);
+                acc_base = ht.accs + (uint64_t)slot * NAGGS;
+                break;
+              }
+              if (cur == (unsigned long long)grp_v) {
+                acc_base = ht.accs + (uint64_t)slot * NAGGS;
+                break;
+              }
+              slot = (slot + 1) & mask;
+            }
+          }
+        }
+        #pragma unroll
+        for (int a = 0; a < NAGGS; a++) {
+          const DevAggSpec &sp = plan.aggs[a];
+          bool contribute;
+          int64_t v = 0;
+          if (sp.kind == DAGG_COUNT_ROWS) {
+            contribute = true;
+          } else if (!cols[a].found || cols[a].null) {
+            contribute = false;
+          } else if (sp.kind == DAGG_COUNT_COL) {
+            contribute = true;
+          } else if (sp.kind == DAGG_SUM_INT) {
+            contribute = true; v = cols[a].iv;
+          } else {  /* SUM_DEC */
+            int d = sp.target_frac - cols[a].dfr;
+            if (!cols[a].has_dec || d < 0 || d > 18) {
+              any_parse_err = true;
+              contribute = false;
+            } else {
+              int64_t scale = 1;
+              for (int t = 0; t < d; t++) scale *= 10;
+              v = cols[a].dsc * scale;
+              contribute = true;
+            }
+          }
+          if (!contribute) continue;
+          if (IS_HASH) {
+            if (acc_base) {
+              atomicAdd(&acc_base[a].cnt, 1ull);
+              if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC)
+                atomic_add_i128(&acc_base[a].sum_lo, &acc_base[a].sum_hi, v);
+            }
+          } else {
+            l_cnt[a]++;
+            if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC) {
+              unsigned long long old = l_lo[a];
+              unsigned long long nv = old + (unsigned long long)v;
+              l_hi[a] += (nv < old ? 1 : 0) + (v < 0 ? -1 : 0);
+              l_lo[a] = nv;
+            }
+          }
+        }
+      }
+    }
+  }
+
+  if (!IS_HASH) {
+    /* wave fold then one atomic per wave per agg */
+    #pragma unroll
+    for (int a = 0; a < NAGGS; a++) {
+      unsigned long long c = l_cnt[a];
+      unsigned long long lo = l_lo[a];
+      long long hi = l_hi[a];
+      for (int off = 32; off > 0; off >>= 1) {
+        c += (unsigned long long)__shfl_down((long long)c, off, 64);
+        unsigned long long plo = (unsigned long long)__shfl_down((long long)lo, off, 64);
+        long long phi = __shfl_down(hi, off, 64);
+        unsigned long long nlo = lo + plo;
+        hi += phi + (nlo < lo ? 1 : 0);
+        lo = nlo;
+      }
+      if ((threadIdx.x & 63u) == 0) {
+        if (c) atomicAdd(&simple_acc[a].cnt, c);
+        if (lo | (unsigned long long)hi) {
+          unsigned long long old = atomicAdd(&simple_acc[a].sum_lo, lo);
+          long long carry = (old + lo < old) ? 1 : 0;
+          long long hi_add = hi + carry;
+          if (hi_add) atomicAdd(&simple_acc[a].sum_hi, (unsigned long long)hi_add);
+        }
+      }
+    }
+  }
+  if (any_parse_err) {
+    if (IS_HASH) atomicOr(ht.error + 1, 1u);
+    else atomicOr((unsigned int *)&simple_acc[COPR_MAX_AGGS].cnt, 1u);
+  }
+}
+
+/* ---------------- project kernel (row-returning scans) ----------------
+ * Not the hot path: dynamic out-column indexing may spill; correctness and
+ * byte-identical output shape are what matter here. */
+__global__ void __launch_bounds__(THREADS)
+k_scan_project(ScanPlan plan,
+               const uint8_t *__restrict__ vals, const uint64_t *__restrict__ val_offs,
+               const uint8_t *__restrict__ keys, const uint64_t *__restrict__ key_offs,
+               uint64_t n_rows, ProjectOut po) {
+  extern __shared__ __attribute__((aligned(16))) uint8_t lds[];
+  const uint32_t rpt = plan.rows_per_tile;
+  const uint64_t n_tiles = (n_rows + rpt - 1) / rpt;
+  bool any_parse_err = false;
+
+  for (uint64_t tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
+    uint64_t row0 = tile * rpt;
+    uint64_t row1 = min(row0 + rpt, n_rows);
+    uint64_t gbase = val_offs[row0];
+    uint32_t tlen = (uint32_t)(val_offs[row1] - gbase);
+    __syncthreads();
+    uint32_t shift = stage_tile(vals, gbase, tlen, lds);
+
+    uint64_t my_row = row0 + threadIdx.x;
+    if (my_row < row1) {
+      const uint8_t *vp = lds + shift + (uint32_t)(val_offs[my_row] - gbase);
+      uint32_t vlen = (uint32_t)(val_offs[my_row + 1] - val_offs[my_row]);
+      bool parse_ok = true;
+      bool filt_found = false, filt_null = false; int64_t filt_v = 0;
+      unsigned long long cell_pack[COPR_MAX_OUT_COLS];
+      for (int j = 0; j < plan.n_out; j++) cell_pack[j] = 0xFFFFFull;
+      int needed = (plan.has_filter ? 1 : 0) + plan.n_out;
+      int found = 0;
+
+      ROW_FOREACH_BEGIN(vp, vlen)
+        if (plan.has_filter && !filt_found && cell_id == plan.filter_col_id) {
+          filt_found = true;
+          if (cell.is_null) filt_null = true;
+          else if (cell.has_int) filt_v = cell.ival;
+          else parse_ok = false;
+          found++;
         }
         for (int j = 0; j < plan.n_out; j++) {
           if (plan.out_is_handle[j] || cell_pack[j] != 0xFFFFFull) continue;
@@ -372,153 +546,27 @@ k_scan(ScanPlan plan,
       if (!parse_ok) {
         any_parse_err = true;
       } else {
-        /* filter (missing column without default/NULL handling on device:
-           absent column value is NULL per default-fill — the host validated
-           that filter/agg/group columns are NULLable-default columns) */
-        bool keep = true;
-        if (plan.has_filter) {
-          bool v_null = filt_null;
-          if (!filt_found) {
-            if (plan.filter_missing_null) v_null = true;
-            else { filt_v = plan.filter_missing_val; v_null = false; }
-          }
-          if (plan.filter_const_null || v_null) keep = false;
-          else keep = d_cmp_res(plan.filter_cmp,
-                                d_cmp_int(filt_v, plan.filter_const,
-                                          plan.filter_col_unsigned,
-                                          plan.filter_const_unsigned));
+        bool keep = d_filter_keep(plan, filt_found, filt_null, filt_v);
+        po.keep[my_row] = keep ? 1 : 0;
+        if (plan.has_filter && po.filt_vals) {
+          po.filt_vals[my_row] = filt_v;
+          po.filt_state[my_row] = !filt_found ? 2 : (filt_null ? 1 : 0);
         }
-        if (plan.mode == 0) {
-          po.keep[my_row] = keep ? 1 : 0;
-          if (plan.has_filter && po.filt_vals) {
-            po.filt_vals[my_row] = filt_v;
-            po.filt_state[my_row] = !filt_found ? 2 : (filt_null ? 1 : 0);
-          }
-          if (keep) {
-            for (int j = 0; j < plan.n_out; j++) {
-              if (plan.out_is_handle[j]) {
-                const uint8_t *kp = keys + key_offs[my_row];
-                uint64_t h = d_be_u64(kp + 11) ^ 0x8000000000000000ull;
-                po.handles[my_row] = (long long)h;
-              } else {
-                po.cells[my_row * plan.n_out + j] = cell_pack[j];
-              }
-            }
-          }
-        } else if (keep) {
-          if (plan.mode == 1) {
-            for (int a = 0; a < plan.n_aggs; a++) {
-              const DevAggSpec &sp = plan.aggs[a];
-              if (sp.kind == DAGG_COUNT_ROWS) { l_cnt[a]++; continue; }
-              if (!cols[a].found || cols[a].null) continue;
-              if (sp.kind == DAGG_COUNT_COL) { l_cnt[a]++; continue; }
-              int64_t v;
-              if (sp.kind == DAGG_SUM_INT) v = cols[a].iv;
-              else {  /* SUM_DEC: rescale to target frac */
-                if (!cols[a].has_dec) { any_parse_err = true; continue; }
-                int d = sp.target_frac - cols[a].dfr;
-                if (d < 0 || d > 18) { any_parse_err = true; continue; }
-                int64_t scale = 1;
-                for (int t = 0; t < d; t++) scale *= 10;
-                v = cols[a].dsc * scale;
-              }
-              l_cnt[a]++;
-              /* local i128 accumulate */
-              unsigned long long old = (unsigned long long)l_lo[a];
-              unsigned long long nv = old + (unsigned long long)v;
-              l_hi[a] += (nv < old ? 1 : 0) + (v < 0 ? -1 : 0);
-              l_lo[a] = (long long)nv;
-            }
-          } else {
-            /* hash agg: find/claim slot */
-            SimpleAggAcc *acc_base = nullptr;
-            if (!grp_found || grp_null) {
-              atomicAdd(&ht.rsvd_seen[1], 1ull);
-              acc_base = ht.reserved + 1 * plan.n_aggs;
-            } else if (grp_v == (long long)0x8000000000000000ll) {
-              atomicAdd(&ht.rsvd_seen[0], 1ull);
-              acc_base = ht.reserved + 0 * plan.n_aggs;
+        if (keep) {
+          for (int j = 0; j < plan.n_out; j++) {
+            if (plan.out_is_handle[j]) {
+              const uint8_t *kp = keys + key_offs[my_row];
+              uint64_t h = d_be_u64(kp + 11) ^ 0x8000000000000000ull;
+              po.handles[my_row] = (long long)h;
             } else {
-              uint64_t h = (uint64_t)grp_v * 0x9E3779B97F4A7C15ull;
-              h ^= h >> 29;
-              uint32_t mask = plan.table_size - 1u;
-              uint32_t slot = (uint32_t)(h & mask);
-              const unsigned long long EMPTY = 0x8000000000000000ull;
-              for (uint32_t probe = 0; ; probe++) {
-                if (probe > mask) { atomicOr(ht.error, 1u); break; }
-                unsigned long long cur =
-                    atomicCAS((unsigned long long *)&ht.keys[slot], EMPTY,
-                              (unsigned long long)grp_v);
-                if (cur == EMPTY) {
-                  atomicAdd(ht.n_groups, 1ull);
-                  acc_base = ht.accs + (uint64_t)slot * plan.n_aggs;
-                  break;
-                }
-                if (cur == (unsigned long long)grp_v) {
-                  acc_base = ht.accs + (uint64_t)slot * plan.n_aggs;
-                  break;
-                }
-                slot = (slot + 1) & mask;
-              }
-            }
-            if (acc_base) {
-              for (int a = 0; a < plan.n_aggs; a++) {
-                const DevAggSpec &sp = plan.aggs[a];
-                if (sp.kind == DAGG_COUNT_ROWS) { atomicAdd(&acc_base[a].cnt, 1ull); continue; }
-                if (!cols[a].found || cols[a].null) continue;
-                if (sp.kind == DAGG_COUNT_COL) { atomicAdd(&acc_base[a].cnt, 1ull); continue; }
-                int64_t v;
-                if (sp.kind == DAGG_SUM_INT) v = cols[a].iv;
-                else {
-                  if (!cols[a].has_dec) { any_parse_err = true; continue; }
-                  int d = sp.target_frac - cols[a].dfr;
-                  if (d < 0 || d > 18) { any_parse_err = true; continue; }
-                  int64_t scale = 1;
-                  for (int t = 0; t < d; t++) scale *= 10;
-                  v = cols[a].dsc * scale;
-                }
-                atomicAdd(&acc_base[a].cnt, 1ull);
-                atomic_add_i128(&acc_base[a].sum_lo, &acc_base[a].sum_hi, v);
-              }
+              po.cells[my_row * plan.n_out + j] = cell_pack[j];
             }
           }
         }
       }
     }
   }
-
-  /* block-level fold of simple-agg accumulators */
-  if (plan.mode == 1) {
-    for (int a = 0; a < plan.n_aggs; a++) {
-      /* wave reduce then one atomic per wave */
-      unsigned long long c = l_cnt[a];
-      long long lo = l_lo[a], hi = l_hi[a];
-      for (int off = 32; off > 0; off >>= 1) {
-        c += (unsigned long long)__shfl_down((long long)c, off, 64);
-        unsigned long long olo = (unsigned long long)lo;
-        unsigned long long plo = (unsigned long long)__shfl_down((long long)olo, off, 64);
-        long long phi = __shfl_down(hi, off, 64);
-        unsigned long long nlo = olo + plo;
-        hi += phi + (nlo < olo ? 1 : 0);
-        lo = (long long)nlo;
-      }
-      if ((threadIdx.x & 63u) == 0) {
-        if (c) atomicAdd(&simple_acc[a].cnt, c);
-        unsigned long long ulo = (unsigned long long)lo;
-        if (ulo | (unsigned long long)hi) {
-          unsigned long long old = atomicAdd(&simple_acc[a].sum_lo, ulo);
-          long long carry = (old + ulo < old) ? 1 : 0;
-          long long hi_add = hi + carry;
-          if (hi_add) atomicAdd(&simple_acc[a].sum_hi, (unsigned long long)hi_add);
-        }
-      }
-    }
-  }
-  if (any_parse_err) {
-    unsigned int *err = plan.mode == 0 ? po.error : (plan.mode == 2 ? ht.error : nullptr);
-    if (err) atomicOr(err + 1, 1u);
-    else if (simple_acc) atomicOr((unsigned int *)&simple_acc[COPR_MAX_AGGS].cnt, 1u);
-  }
+  if (any_parse_err) atomicOr(po.error + 1, 1u);
 }
 
 /* ---------------- CRC-64/XZ kernel (checksum.rs:105-114) ----------------
@@ -557,14 +605,11 @@ k_crc64(const uint8_t *__restrict__ vals, const uint64_t *__restrict__ val_offs,
       uint32_t klen = (uint32_t)(key_offs[my_row + 1] - key_offs[my_row]);
       const uint8_t *vp = val_lds + vshift + (uint32_t)(val_offs[my_row] - vbase);
       uint32_t vlen = (uint32_t)(val_offs[my_row + 1] - val_offs[my_row]);
-      /* key (short): bytewise */
       for (uint32_t i = 0; i < klen; i++)
         crc = tab[(uint32_t)((crc ^ kp[i]) & 0xFF)] ^ (crc >> 8);
-      /* value: slice-by-8 from LDS */
       uint32_t i = 0;
       for (; i + 8 <= vlen; i += 8) {
         uint64_t x;
-        /* unaligned 8-byte read from LDS */
         memcpy(&x, vp + i, 8);
         crc ^= x;
         crc = tab[7 * 256 + (uint32_t)(crc & 0xFF)] ^
@@ -581,35 +626,52 @@ k_crc64(const uint8_t *__restrict__ vals, const uint64_t *__restrict__ val_offs,
       acc ^= ~crc;
     }
   }
-  /* wave XOR-fold, one atomic per wave */
   for (int off = 32; off > 0; off >>= 1)
     acc ^= (unsigned long long)__shfl_down((long long)acc, off, 64);
   if ((threadIdx.x & 63u) == 0 && acc) atomicXor(out_xor, acc);
 }
 
 /* ---------------- launch wrappers ---------------- */
+template <bool IS_HASH>
+static int launch_agg(const ScanPlan &plan, const DevRegion &rgn,
+                      SimpleAggAcc *d_simple, HashAggTable ht, hipStream_t s,
+                      uint32_t grid) {
+  #define CASE(N)                                                            \
+    hipLaunchKernelGGL((k_scan_agg<N, IS_HASH>), dim3(grid), dim3(THREADS),  \
+                       plan.lds_bytes, s, plan, rgn.d_vals, rgn.d_val_offs,  \
+                       rgn.n_kv, d_simple, ht)
+  switch (plan.n_aggs) {
+    case 1: CASE(1); break;
+    case 2: CASE(2); break;
+    case 3: CASE(3); break;
+    case 4: CASE(4); break;
+    default: CASE(COPR_MAX_AGGS); break;
+  }
+  #undef CASE
+  return (int)hipGetLastError();
+}
+
 int dev_scan_launch(const ScanPlan &plan, const DevRegion &rgn,
                     SimpleAggAcc *d_simple, const HashAggTable *ht,
                     const ProjectOut *po, void *stream) {
   uint64_t n_tiles = (rgn.n_kv + plan.rows_per_tile - 1) / plan.rows_per_tile;
   uint32_t grid = (uint32_t)(n_tiles < 4096 ? n_tiles : 4096);
   if (grid == 0) grid = 1;
-  HashAggTable ht_v{};
-  ProjectOut po_v{};
-  if (ht) ht_v = *ht;
-  if (po) po_v = *po;
-  hipLaunchKernelGGL(k_scan, dim3(grid), dim3(THREADS), plan.lds_bytes,
-                     (hipStream_t)stream, plan,
-                     rgn.d_vals, rgn.d_val_offs, rgn.d_keys, rgn.d_key_offs,
-                     rgn.n_kv, d_simple, ht_v, po_v);
+  hipStream_t s = (hipStream_t)stream;
+  if (plan.mode == 1)
+    return launch_agg<false>(plan, rgn, d_simple, HashAggTable{}, s, grid);
+  if (plan.mode == 2)
+    return launch_agg<true>(plan, rgn, nullptr, *ht, s, grid);
+  hipLaunchKernelGGL(k_scan_project, dim3(grid), dim3(THREADS), plan.lds_bytes,
+                     s, plan, rgn.d_vals, rgn.d_val_offs, rgn.d_keys,
+                     rgn.d_key_offs, rgn.n_kv, *po);
   return (int)hipGetLastError();
 }
 
 int dev_crc64_launch(const DevRegion &rgn, const uint64_t *d_tables,
                      unsigned long long *d_xor, void *stream) {
-  /* LDS: 16 KiB tables + key tile + value tile */
   uint32_t rpt = 256;
-  uint32_t max_key = 64;   /* record keys are 19 B; allow some slack */
+  uint32_t max_key = 64;
   uint64_t max_tile_val = (uint64_t)rpt * (rgn.max_row_bytes + 16) + 32;
   while (rpt > 64 && 16384 + rpt * max_key + max_tile_val > 160 * 1024 - 1024) {
     rpt /= 2;
