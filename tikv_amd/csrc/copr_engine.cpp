@@ -404,17 +404,19 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
  * tile fits the LDS budget (Guideline: 256-thread blocks; <=64 KiB tile
  * keeps >=2 blocks/CU of occupancy on the 160 KiB LDS). */
 static void pick_tiling(const DevRegion &rgn, ScanPlan *sp) {
-  const uint32_t LDS_BUDGET = 64 * 1024;
+  uint32_t budget = 64 * 1024;          /* >=2 blocks/CU on 160 KiB LDS */
+  if (const char *e = getenv("COPR_LDS_BUDGET")) budget = (uint32_t)atoi(e);
   uint32_t per_row = rgn.max_row_bytes + 1;
-  uint32_t rows = 256;
-  while (rows > 64 && (uint64_t)rows * per_row + 64 > LDS_BUDGET) rows /= 2;
-  if ((uint64_t)rows * per_row + 64 > LDS_BUDGET) {
+  uint32_t rows = 1024;                 /* up to 4 rows per lane per tile */
+  if (const char *e = getenv("COPR_ROWS_PER_TILE")) rows = (uint32_t)atoi(e);
+  while (rows > 64 && (uint64_t)rows * per_row + 96 > budget) rows /= 2;
+  if ((uint64_t)rows * per_row + 96 > budget) {
     /* giant rows: single-row tiles with exact size */
     rows = 1;
   }
   sp->rows_per_tile = rows;
-  uint64_t lds = (uint64_t)rows * per_row + 64;
-  if (rows == 1) lds = (uint64_t)rgn.max_row_bytes + 64;
+  uint64_t lds = (uint64_t)rows * per_row + 96;
+  if (rows == 1) lds = (uint64_t)rgn.max_row_bytes + 96;
   sp->lds_bytes = (uint32_t)lds;
 }
 

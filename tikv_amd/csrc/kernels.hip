@@ -37,21 +37,22 @@ __device__ static inline uint32_t stage_tile(const uint8_t *__restrict__ src,
   const uint4 *gs = (const uint4 *)(src + abase);
   uint4 *ld = (uint4 *)lds;
   uint32_t n16 = total >> 4;
-  /* batched: issue up to 8 independent loads per lane, then write — the HBM
-     latency (~900 cyc) is paid once per batch instead of once per element */
-  for (uint32_t base = 0; base < n16; base += blockDim.x * 8u) {
+  /* batched: 8 UNCONDITIONAL independent loads per lane in the main body
+     (a per-element bounds branch makes hipcc wait vmcnt(0) per element —
+     cdna_hip_programming §6 trap (c)), then a simple tail loop */
+  uint32_t per_round = blockDim.x * 8u;
+  uint32_t base = 0;
+  for (; base + per_round <= n16; base += per_round) {
     uint4 tmp[8];
     #pragma unroll
-    for (int j = 0; j < 8; j++) {
-      uint32_t i = base + threadIdx.x + (uint32_t)j * blockDim.x;
-      if (i < n16) tmp[j] = gs[i];
-    }
+    for (int j = 0; j < 8; j++)
+      tmp[j] = gs[base + threadIdx.x + (uint32_t)j * blockDim.x];
     #pragma unroll
-    for (int j = 0; j < 8; j++) {
-      uint32_t i = base + threadIdx.x + (uint32_t)j * blockDim.x;
-      if (i < n16) ld[i] = tmp[j];
-    }
+    for (int j = 0; j < 8; j++)
+      ld[base + threadIdx.x + (uint32_t)j * blockDim.x] = tmp[j];
   }
+  for (uint32_t i = base + threadIdx.x; i < n16; i += blockDim.x)
+    ld[i] = gs[i];
   __syncthreads();
   return shift;
 }
@@ -427,8 +428,7 @@ k_scan_agg(ScanPlan plan,
     __syncthreads();   /* previous tile's lanes done with LDS */
     uint32_t shift = stage_tile(vals, gbase, tlen, lds);
 
-    uint64_t my_row = row0 + threadIdx.x;
-    if (my_row < row1) {
+    for (uint64_t my_row = row0 + threadIdx.x; my_row < row1; my_row += blockDim.x) {
       const uint8_t *vp = lds + shift + (uint32_t)(val_offs[my_row] - gbase);
       uint32_t vlen = (uint32_t)(val_offs[my_row + 1] - val_offs[my_row]);
       bool parse_ok = true;
