@@ -615,8 +615,7 @@ k_scan_project(ScanPlan plan,
     __syncthreads();
     uint32_t shift = stage_tile(vals, gbase, tlen, lds);
 
-    uint64_t my_row = row0 + threadIdx.x;
-    if (my_row < row1) {
+    for (uint64_t my_row = row0 + threadIdx.x; my_row < row1; my_row += blockDim.x) {
       const uint8_t *vp = lds + shift + (uint32_t)(val_offs[my_row] - gbase);
       uint32_t vlen = (uint32_t)(val_offs[my_row + 1] - val_offs[my_row]);
       bool parse_ok = true;
