@@ -404,6 +404,7 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
  * tile fits the LDS budget (Guideline: 256-thread blocks; <=64 KiB tile
  * keeps >=2 blocks/CU of occupancy on the 160 KiB LDS). */
 static void pick_tiling(const DevRegion &rgn, ScanPlan *sp) {
+  sp->diag_stage_only = getenv("COPR_DIAG_STAGE_ONLY") ? 1 : 0;
   uint32_t budget = 64 * 1024;          /* >=2 blocks/CU on 160 KiB LDS */
   if (const char *e = getenv("COPR_LDS_BUDGET")) budget = (uint32_t)atoi(e);
   uint32_t per_row = rgn.max_row_bytes + 1;

@@ -68,6 +68,9 @@ struct ScanPlan {
   int64_t out_col_ids[COPR_MAX_OUT_COLS];
   int32_t out_is_handle[COPR_MAX_OUT_COLS]; /* 1 => decoded int handle */
 
+  /* diagnostics: 1 = stage tiles but skip the parse (bandwidth ceiling probe;
+     COPR_DIAG_STAGE_ONLY=1; results are garbage, never used in tests) */
+  int32_t diag_stage_only;
   /* tiling */
   uint32_t rows_per_tile;
   uint32_t lds_bytes;            /* dynamic LDS per block */

@@ -427,6 +427,10 @@ k_scan_agg(ScanPlan plan,
     uint32_t tlen = (uint32_t)(val_offs[row1] - gbase);
     __syncthreads();   /* previous tile's lanes done with LDS */
     uint32_t shift = stage_tile(vals, gbase, tlen, lds);
+    if (plan.diag_stage_only) {
+      if (lds[shift] == 0xA5u && threadIdx.x == 1023u) l_cnt[0]++;  /* keep the stage live */
+      continue;
+    }
 
     for (uint64_t my_row = row0 + threadIdx.x; my_row < row1; my_row += blockDim.x) {
       const uint8_t *vp = lds + shift + (uint32_t)(val_offs[my_row] - gbase);
