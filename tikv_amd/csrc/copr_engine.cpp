@@ -97,9 +97,10 @@ extern "C" copr_status copr_region_create(copr_engine *eng,
   r->h_key_offs.assign(key_offs, key_offs + n_kv + 1);
   r->h_val_offs.assign(val_offs, val_offs + n_kv + 1);
 
-  /* +64 slack: LDS staging reads 16B-aligned past the end */
+  /* +2 KiB slack: LDS staging reads 16B-aligned past the end, and the glds
+     pipeline streams whole 1 KiB chunks */
   auto alloc_copy = [&](void **dst, const void *src, uint64_t bytes) -> hipError_t {
-    hipError_t e = hipMalloc(dst, bytes + 64);
+    hipError_t e = hipMalloc(dst, bytes + 2048);
     if (e != hipSuccess) return e;
     return hipMemcpy(*dst, src, bytes, hipMemcpyHostToDevice);
   };
@@ -405,9 +406,37 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
  * keeps >=2 blocks/CU of occupancy on the 160 KiB LDS). */
 static void pick_tiling(const DevRegion &rgn, ScanPlan *sp) {
   sp->diag_stage_only = getenv("COPR_DIAG_STAGE_ONLY") ? 1 : 0;
+  sp->use_pipe = 0;
+  uint32_t per_row = rgn.max_row_bytes + 1;
+
+  if (!getenv("COPR_NO_PIPE")) {
+    /* glds double-buffer pipeline: 2 x (offs slab + vals slab), 1 KiB
+       granular. Target <= ~120 KiB so at least one more block can coexist
+       on smaller shapes; >= 2 blocks/CU needs <= 76 KiB. */
+    uint32_t budget = 76 * 1024;
+    if (const char *e = getenv("COPR_PIPE_LDS_BUDGET")) budget = (uint32_t)atoi(e);
+    uint32_t rows = 1024;
+    if (const char *e = getenv("COPR_ROWS_PER_TILE")) rows = (uint32_t)atoi(e);
+    while (rows > 64) {
+      uint32_t os = (((rows + 1) * 8) + 1023u) & ~1023u;
+      uint32_t vs = ((rows * per_row + 15u) + 1023u) & ~1023u;
+      if (2 * (os + vs) <= budget) break;
+      rows /= 2;
+    }
+    uint32_t os = (((rows + 1) * 8) + 1023u) & ~1023u;
+    uint32_t vs = ((rows * per_row + 15u) + 1023u) & ~1023u;
+    if (rows >= 64 && 2 * (os + vs) <= 160 * 1024 - 2048) {
+      sp->use_pipe = 1;
+      sp->rows_per_tile = rows;
+      sp->offs_slab = os;
+      sp->vals_slab = vs;
+      sp->lds_bytes = 2 * (os + vs);
+      return;
+    }
+  }
+
   uint32_t budget = 64 * 1024;          /* >=2 blocks/CU on 160 KiB LDS */
   if (const char *e = getenv("COPR_LDS_BUDGET")) budget = (uint32_t)atoi(e);
-  uint32_t per_row = rgn.max_row_bytes + 1;
   uint32_t rows = 1024;                 /* up to 4 rows per lane per tile */
   if (const char *e = getenv("COPR_ROWS_PER_TILE")) rows = (uint32_t)atoi(e);
   while (rows > 64 && (uint64_t)rows * per_row + 96 > budget) rows /= 2;

@@ -74,6 +74,12 @@ struct ScanPlan {
   /* tiling */
   uint32_t rows_per_tile;
   uint32_t lds_bytes;            /* dynamic LDS per block */
+  /* glds pipeline (k_scan_agg_pipe): per-buffer slab sizes, both 1 KiB
+     multiples; lds_bytes = 2 * (offs_slab + vals_slab). use_pipe 0 falls
+     back to the single-buffer kernel. */
+  int32_t use_pipe;
+  uint32_t offs_slab;
+  uint32_t vals_slab;
 };
 
 /* simple-agg accumulators (device buffer, one per agg) */

@@ -389,11 +389,74 @@ __device__ static inline void atomic_add_i128(unsigned long long *lo,
 
 /* ---------------- fused scan + filter + aggregate ----------------
  * NAGGS is a compile-time bound so all per-row/per-lane state stays in
- * registers. IS_HASH selects grouped aggregation. */
+ * registers. IS_HASH selects grouped aggregation. NLOADS = staged uint4
+ * loads per lane per tile (compile-time so the load batch is unconditional
+ * and register-resident): the block software-pipelines tiles — while lanes
+ * parse tile t from LDS, the loads for tile t+1 are already in flight into
+ * registers (async-STAGE split, cdna_hip_programming G15). */
 struct AggColView { bool found, null, has_dec; int64_t iv, dsc; int32_t dfr; };
 
-template <int NAGGS, bool IS_HASH>
-__global__ void __launch_bounds__(THREADS)
+struct TileInfo {
+  uint64_t row0, row1, gbase;
+  uint32_t shift, n16;
+};
+
+template <int NLOADS>
+__device__ static inline void tile_info(const uint64_t *__restrict__ val_offs,
+                                        uint64_t n_rows, uint32_t rpt,
+                                        uint64_t tile, TileInfo *ti) {
+  ti->row0 = tile * rpt;
+  ti->row1 = min(ti->row0 + rpt, n_rows);
+  uint64_t gb = val_offs[ti->row0];
+  uint64_t ge = val_offs[ti->row1];
+  uint64_t abase = gb & ~15ull;
+  ti->gbase = gb;
+  ti->shift = (uint32_t)(gb - abase);
+  ti->n16 = (uint32_t)(((ge - gb) + ti->shift + 15ull) >> 4);
+}
+
+/* the staging buffer is NLOADS NAMED uint4s (not an array): hipcc's
+ * allocator spills a local array with a long live range to scratch even
+ * with VGPR budget to spare */
+template <int NLOADS>
+struct StageRegs {
+  uint4 r0, r1, r2, r3, r4, r5, r6, r7, r8, r9, r10, r11, r12, r13, r14, r15;
+};
+
+#define STAGE_EACH(OP) \
+  OP(0) OP(1) OP(2) OP(3) OP(4) OP(5) OP(6) OP(7) \
+  OP(8) OP(9) OP(10) OP(11) OP(12) OP(13) OP(14) OP(15)
+
+template <int NLOADS>
+__device__ static inline void tile_load(const uint8_t *__restrict__ vals,
+                                        const TileInfo &ti, StageRegs<NLOADS> &R) {
+  const uint4 *gs = (const uint4 *)(vals + (ti.gbase & ~15ull));
+  uint32_t last = ti.n16 ? ti.n16 - 1u : 0u;
+  /* clamped, UNCONDITIONAL loads (a per-element branch makes hipcc wait
+     vmcnt(0) per element — §6 trap (c)); duplicates hit L2 */
+  #define COPR_LOADJ(J) \
+    if constexpr (J < NLOADS) \
+      R.r##J = gs[min(threadIdx.x + (uint32_t)J * THREADS, last)];
+  STAGE_EACH(COPR_LOADJ)
+  #undef COPR_LOADJ
+}
+
+template <int NLOADS>
+__device__ static inline void tile_store(uint8_t *lds, const TileInfo &ti,
+                                         const StageRegs<NLOADS> &R) {
+  uint4 *ld = (uint4 *)lds;
+  uint32_t last = ti.n16 ? ti.n16 - 1u : 0u;
+  /* clamped stores: the clamped duplicate rewrites the same element with
+     the same value */
+  #define COPR_STOREJ(J) \
+    if constexpr (J < NLOADS) \
+      ld[min(threadIdx.x + (uint32_t)J * THREADS, last)] = R.r##J;
+  STAGE_EACH(COPR_STOREJ)
+  #undef COPR_STOREJ
+}
+
+template <int NAGGS, bool IS_HASH, int NLOADS>
+__global__ void __launch_bounds__(THREADS, 2)
 k_scan_agg(ScanPlan plan,
            const uint8_t *__restrict__ vals, const uint64_t *__restrict__ val_offs,
            uint64_t n_rows,
@@ -404,15 +467,12 @@ k_scan_agg(ScanPlan plan,
   const uint32_t rpt = plan.rows_per_tile;
   const uint64_t n_tiles = (n_rows + rpt - 1) / rpt;
 
-  /* per-lane simple-agg accumulators (registers; folded per wave at end) */
   unsigned long long l_cnt[NAGGS];
   unsigned long long l_lo[NAGGS];
   long long l_hi[NAGGS];
   #pragma unroll
   for (int a = 0; a < NAGGS; a++) { l_cnt[a] = 0; l_lo[a] = 0; l_hi[a] = 0; }
 
-  /* how many cells the parse must find before it can stop early
-     (process_v1 stops at decoded == columns_len, table_scan_executor.rs:223) */
   int needed = (plan.has_filter ? 1 : 0) + (IS_HASH ? 1 : 0);
   #pragma unroll
   for (int a = 0; a < NAGGS; a++)
@@ -420,20 +480,32 @@ k_scan_agg(ScanPlan plan,
 
   bool any_parse_err = false;
 
-  for (uint64_t tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
-    uint64_t row0 = tile * rpt;
-    uint64_t row1 = min(row0 + rpt, n_rows);
-    uint64_t gbase = val_offs[row0];
-    uint32_t tlen = (uint32_t)(val_offs[row1] - gbase);
-    __syncthreads();   /* previous tile's lanes done with LDS */
-    uint32_t shift = stage_tile(vals, gbase, tlen, lds);
+  StageRegs<NLOADS> regs;
+  TileInfo cur, nxt;
+  uint64_t tile = blockIdx.x;
+  if (tile < n_tiles) {
+    tile_info<NLOADS>(val_offs, n_rows, rpt, tile, &cur);
+    tile_load<NLOADS>(vals, cur, regs);
+  }
+  for (; tile < n_tiles; tile += gridDim.x) {
+    __syncthreads();                    /* previous parse done: LDS free */
+    tile_store<NLOADS>(lds, cur, regs);
+    __syncthreads();                    /* staged tile visible */
+    /* UNCONDITIONALLY load the next tile (clamped: the last iteration
+       re-loads its own tile from L2) so `regs` is never conditionally
+       defined — a merge of partially-defined paths makes the allocator
+       spill the whole staging array to scratch */
+    uint64_t next = min(tile + gridDim.x, n_tiles - 1);
+    tile_info<NLOADS>(val_offs, n_rows, rpt, next, &nxt);
+    tile_load<NLOADS>(vals, nxt, regs);     /* in flight during the parse */
     if (plan.diag_stage_only) {
-      if (lds[shift] == 0xA5u && threadIdx.x == 1023u) l_cnt[0]++;  /* keep the stage live */
+      if (lds[cur.shift] == 0xA5u && threadIdx.x == 1023u) l_cnt[0]++;
+      cur = nxt;
       continue;
     }
 
-    for (uint64_t my_row = row0 + threadIdx.x; my_row < row1; my_row += blockDim.x) {
-      const uint8_t *vp = lds + shift + (uint32_t)(val_offs[my_row] - gbase);
+    for (uint64_t my_row = cur.row0 + threadIdx.x; my_row < cur.row1; my_row += blockDim.x) {
+      const uint8_t *vp = lds + cur.shift + (uint32_t)(val_offs[my_row] - cur.gbase);
       uint32_t vlen = (uint32_t)(val_offs[my_row + 1] - val_offs[my_row]);
       bool parse_ok = true;
 
@@ -504,15 +576,15 @@ k_scan_agg(ScanPlan plan,
             const unsigned long long EMPTY = 0x8000000000000000ull;
             for (uint32_t probe = 0; ; probe++) {
               if (probe > mask) { atomicOr(ht.error, 1u); break; }
-              unsigned long long cur =
+              unsigned long long curk =
                   atomicCAS((unsigned long long *)&ht.keys[slot], EMPTY,
                             (unsigned long long)grp_v);
-              if (cur == EMPTY) {
+              if (curk == EMPTY) {
                 atomicAdd(ht.n_groups, 1ull);
                 acc_base = ht.accs + (uint64_t)slot * NAGGS;
                 break;
               }
-              if (cur == (unsigned long long)grp_v) {
+              if (curk == (unsigned long long)grp_v) {
                 acc_base = ht.accs + (uint64_t)slot * NAGGS;
                 break;
               }
@@ -564,10 +636,266 @@ k_scan_agg(ScanPlan plan,
         }
       }
     }
+    cur = nxt;
   }
 
   if (!IS_HASH) {
-    /* wave fold then one atomic per wave per agg */
+    #pragma unroll
+    for (int a = 0; a < NAGGS; a++) {
+      unsigned long long c = l_cnt[a];
+      unsigned long long lo = l_lo[a];
+      long long hi = l_hi[a];
+      for (int off = 32; off > 0; off >>= 1) {
+        c += (unsigned long long)__shfl_down((long long)c, off, 64);
+        unsigned long long plo = (unsigned long long)__shfl_down((long long)lo, off, 64);
+        long long phi = __shfl_down(hi, off, 64);
+        unsigned long long nlo = lo + plo;
+        hi += phi + (nlo < lo ? 1 : 0);
+        lo = nlo;
+      }
+      if ((threadIdx.x & 63u) == 0) {
+        if (c) atomicAdd(&simple_acc[a].cnt, c);
+        if (lo | (unsigned long long)hi) {
+          unsigned long long old = atomicAdd(&simple_acc[a].sum_lo, lo);
+          long long carry = (old + lo < old) ? 1 : 0;
+          long long hi_add = hi + carry;
+          if (hi_add) atomicAdd(&simple_acc[a].sum_hi, (unsigned long long)hi_add);
+        }
+      }
+    }
+  }
+  if (any_parse_err) {
+    if (IS_HASH) atomicOr(ht.error + 1, 1u);
+    else atomicOr((unsigned int *)&simple_acc[COPR_MAX_AGGS].cnt, 1u);
+  }
+}
+
+/* ---------------- glds-pipelined scan + filter + aggregate ----------------
+ * Two LDS buffers; tile t+1's bytes (offsets slab + values slab) stream
+ * HBM->LDS via __builtin_amdgcn_global_load_lds WHILE lanes parse tile t:
+ * issue(next) -> parse(cur) -> s_waitcnt vmcnt(0) -> barrier -> swap. The
+ * parse touches no vector global loads (per-row offsets come from the LDS
+ * offs slab), so hipcc inserts no early vmcnt(0) that would drain the DMA
+ * queue (cdna_hip_programming §5 pipelining-across-barriers / §6 G15). */
+template <int NAGGS, bool IS_HASH>
+__global__ void __launch_bounds__(THREADS, 2)
+k_scan_agg_pipe(ScanPlan plan,
+                const uint8_t *__restrict__ vals,
+                const uint64_t *__restrict__ val_offs, uint64_t n_rows,
+                SimpleAggAcc *__restrict__ simple_acc, HashAggTable ht) {
+  extern __shared__ __attribute__((aligned(16))) uint8_t lds[];
+  const uint32_t rpt = plan.rows_per_tile;
+  const uint64_t n_tiles = (n_rows + rpt - 1) / rpt;
+  const uint32_t OS = plan.offs_slab;
+  const uint32_t BUFSZ = OS + plan.vals_slab;
+  const uint32_t wave = threadIdx.x >> 6, lane = threadIdx.x & 63u;
+  const uint32_t nwaves = THREADS / 64u;
+
+  unsigned long long l_cnt[NAGGS];
+  unsigned long long l_lo[NAGGS];
+  long long l_hi[NAGGS];
+  #pragma unroll
+  for (int a = 0; a < NAGGS; a++) { l_cnt[a] = 0; l_lo[a] = 0; l_hi[a] = 0; }
+
+  int needed = (plan.has_filter ? 1 : 0) + (IS_HASH ? 1 : 0);
+  #pragma unroll
+  for (int a = 0; a < NAGGS; a++)
+    if (plan.aggs[a].kind != DAGG_COUNT_ROWS) needed++;
+
+  bool any_parse_err = false;
+
+  auto issue_tile = [&](uint64_t tile, uint8_t *b) {
+    uint64_t row0 = tile * rpt;
+    uint64_t row1 = min(row0 + rpt, n_rows);
+    /* read the tile bounds FIRST: their compiler-inserted vmcnt(0) must not
+       land between glds issues (it would drain the DMA queue) */
+    uint64_t gb = val_offs[row0];
+    uint64_t ge = val_offs[row1];
+    /* offsets slab: val_offs[row0 .. row1] raw (1 KiB chunks, whole-chunk
+       over-read stays inside the +2 KiB region slack) */
+    const uint8_t *osrc = (const uint8_t *)(val_offs + row0);
+    uint32_t obytes = (uint32_t)((row1 - row0 + 1) * 8);
+    uint32_t oc = (obytes + 1023u) >> 10;
+    for (uint32_t c = wave; c < oc; c += nwaves) {
+      uint32_t off = (c << 10) + lane * 16u;
+      __builtin_amdgcn_global_load_lds((const uint32_t *)(osrc + off),
+                                       (uint32_t *)(b + off), 16, 0, 0);
+    }
+    /* values slab, 16-aligned from the tile's aligned base */
+    uint64_t abase = gb & ~15ull;
+    uint32_t tbytes = (uint32_t)(ge - abase);
+    uint32_t vc = (tbytes + 1023u) >> 10;
+    const uint8_t *vsrc = vals + abase;
+    uint8_t *bv = b + OS;
+    for (uint32_t c = wave; c < vc; c += nwaves) {
+      uint32_t off = (c << 10) + lane * 16u;
+      __builtin_amdgcn_global_load_lds((const uint32_t *)(vsrc + off),
+                                       (uint32_t *)(bv + off), 16, 0, 0);
+    }
+  };
+
+  uint64_t tile = blockIdx.x;
+  uint32_t bsel = 0;
+  if (tile < n_tiles) issue_tile(tile, lds);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  for (; tile < n_tiles; tile += gridDim.x) {
+    uint8_t *b = lds + bsel * BUFSZ;
+    uint8_t *bn = lds + (bsel ^ 1) * BUFSZ;
+    uint64_t next = tile + gridDim.x;
+    if (next < n_tiles) issue_tile(next, bn);   /* in flight during parse */
+
+    uint64_t row0 = tile * rpt;
+    uint64_t row1 = min(row0 + rpt, n_rows);
+    const uint64_t *loffs = (const uint64_t *)b;
+    uint64_t gb = loffs[0];
+    uint32_t shift = (uint32_t)(gb & 15ull);
+    const uint8_t *bv = b + OS;
+
+    if (!plan.diag_stage_only) {
+    for (uint64_t my_row = row0 + threadIdx.x; my_row < row1; my_row += blockDim.x) {
+      uint32_t r = (uint32_t)(my_row - row0);
+      uint64_t o0 = loffs[r], o1 = loffs[r + 1];
+      const uint8_t *vp = bv + shift + (uint32_t)(o0 - gb);
+      uint32_t vlen = (uint32_t)(o1 - o0);
+      bool parse_ok = true;
+
+      bool filt_found = false, filt_null = false; int64_t filt_v = 0;
+      bool grp_found = false, grp_null = false; int64_t grp_v = 0;
+      AggColView cols[NAGGS];
+      #pragma unroll
+      for (int a = 0; a < NAGGS; a++) cols[a] = {false, false, false, 0, 0, 0};
+      int found = 0;
+
+      if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
+        uint32_t pos = 0;
+        while (pos < vlen) {
+          int64_t cell_id;
+          uint32_t cell_off;
+          CellView cell;
+          if (!next_cell(vp, vlen, &pos, &cell_id, &cell_off, &cell)) {
+            parse_ok = false;
+            break;
+          }
+          if (plan.has_filter && !filt_found && cell_id == plan.filter_col_id) {
+            filt_found = true;
+            if (cell.is_null) filt_null = true;
+            else if (cell.has_int) filt_v = cell.ival;
+            else parse_ok = false;
+            found++;
+          }
+          if (IS_HASH && !grp_found && cell_id == plan.group_col_id) {
+            grp_found = true;
+            if (cell.is_null) grp_null = true;
+            else if (cell.has_int) grp_v = cell.ival;
+            else parse_ok = false;
+            found++;
+          }
+          #pragma unroll
+          for (int a = 0; a < NAGGS; a++) {
+            if (plan.aggs[a].kind == DAGG_COUNT_ROWS || cols[a].found) continue;
+            if (cell_id == plan.aggs[a].col_id) {
+              cols[a].found = true;
+              cols[a].null = cell.is_null;
+              cols[a].iv = cell.ival;
+              cols[a].has_dec = cell.has_dec;
+              cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
+              if (!cell.is_null && !cell.has_int && !cell.has_dec) parse_ok = false;
+              found++;
+            }
+          }
+          if (found >= needed) break;
+        }
+      }
+
+      if (!parse_ok) {
+        any_parse_err = true;
+      } else if (d_filter_keep(plan, filt_found, filt_null, filt_v)) {
+        SimpleAggAcc *acc_base = nullptr;
+        if (IS_HASH) {
+          if (!grp_found || grp_null) {
+            atomicAdd(&ht.rsvd_seen[1], 1ull);
+            acc_base = ht.reserved + 1 * NAGGS;
+          } else if (grp_v == (long long)0x8000000000000000ll) {
+            atomicAdd(&ht.rsvd_seen[0], 1ull);
+            acc_base = ht.reserved + 0 * NAGGS;
+          } else {
+            uint64_t h = (uint64_t)grp_v * 0x9E3779B97F4A7C15ull;
+            h ^= h >> 29;
+            uint32_t mask = plan.table_size - 1u;
+            uint32_t slot = (uint32_t)(h & mask);
+            const unsigned long long EMPTY = 0x8000000000000000ull;
+            for (uint32_t probe = 0; ; probe++) {
+              if (probe > mask) { atomicOr(ht.error, 1u); break; }
+              unsigned long long curk =
+                  atomicCAS((unsigned long long *)&ht.keys[slot], EMPTY,
+                            (unsigned long long)grp_v);
+              if (curk == EMPTY) {
+                atomicAdd(ht.n_groups, 1ull);
+                acc_base = ht.accs + (uint64_t)slot * NAGGS;
+                break;
+              }
+              if (curk == (unsigned long long)grp_v) {
+                acc_base = ht.accs + (uint64_t)slot * NAGGS;
+                break;
+              }
+              slot = (slot + 1) & mask;
+            }
+          }
+        }
+        #pragma unroll
+        for (int a = 0; a < NAGGS; a++) {
+          const DevAggSpec &sp = plan.aggs[a];
+          bool contribute;
+          int64_t v = 0;
+          if (sp.kind == DAGG_COUNT_ROWS) {
+            contribute = true;
+          } else if (!cols[a].found || cols[a].null) {
+            contribute = false;
+          } else if (sp.kind == DAGG_COUNT_COL) {
+            contribute = true;
+          } else if (sp.kind == DAGG_SUM_INT) {
+            contribute = true; v = cols[a].iv;
+          } else {  /* SUM_DEC */
+            int d = sp.target_frac - cols[a].dfr;
+            if (!cols[a].has_dec || d < 0 || d > 18) {
+              any_parse_err = true;
+              contribute = false;
+            } else {
+              int64_t scale = 1;
+              for (int t = 0; t < d; t++) scale *= 10;
+              v = cols[a].dsc * scale;
+              contribute = true;
+            }
+          }
+          if (!contribute) continue;
+          if (IS_HASH) {
+            if (acc_base) {
+              atomicAdd(&acc_base[a].cnt, 1ull);
+              if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC)
+                atomic_add_i128(&acc_base[a].sum_lo, &acc_base[a].sum_hi, v);
+            }
+          } else {
+            l_cnt[a]++;
+            if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC) {
+              unsigned long long old = l_lo[a];
+              unsigned long long nv = old + (unsigned long long)v;
+              l_hi[a] += (nv < old ? 1 : 0) + (v < 0 ? -1 : 0);
+              l_lo[a] = nv;
+            }
+          }
+        }
+      }
+    }
+    }  /* !diag */
+
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    bsel ^= 1;
+  }
+
+  if (!IS_HASH) {
     #pragma unroll
     for (int a = 0; a < NAGGS; a++) {
       unsigned long long c = l_cnt[a];
@@ -737,14 +1065,14 @@ k_crc64(const uint8_t *__restrict__ vals, const uint64_t *__restrict__ val_offs,
 }
 
 /* ---------------- launch wrappers ---------------- */
-template <bool IS_HASH>
-static int launch_agg(const ScanPlan &plan, const DevRegion &rgn,
-                      SimpleAggAcc *d_simple, HashAggTable ht, hipStream_t s,
-                      uint32_t grid) {
+template <bool IS_HASH, int NLOADS>
+static int launch_agg2(const ScanPlan &plan, const DevRegion &rgn,
+                       SimpleAggAcc *d_simple, HashAggTable ht, hipStream_t s,
+                       uint32_t grid) {
   #define CASE(N)                                                            \
-    hipLaunchKernelGGL((k_scan_agg<N, IS_HASH>), dim3(grid), dim3(THREADS),  \
-                       plan.lds_bytes, s, plan, rgn.d_vals, rgn.d_val_offs,  \
-                       rgn.n_kv, d_simple, ht)
+    hipLaunchKernelGGL((k_scan_agg<N, IS_HASH, NLOADS>), dim3(grid),         \
+                       dim3(THREADS), plan.lds_bytes, s, plan, rgn.d_vals,   \
+                       rgn.d_val_offs, rgn.n_kv, d_simple, ht)
   switch (plan.n_aggs) {
     case 1: CASE(1); break;
     case 2: CASE(2); break;
@@ -754,6 +1082,37 @@ static int launch_agg(const ScanPlan &plan, const DevRegion &rgn,
   }
   #undef CASE
   return (int)hipGetLastError();
+}
+
+template <bool IS_HASH>
+static int launch_agg_pipe(const ScanPlan &plan, const DevRegion &rgn,
+                           SimpleAggAcc *d_simple, HashAggTable ht,
+                           hipStream_t s, uint32_t grid) {
+  #define CASE(N)                                                            \
+    hipLaunchKernelGGL((k_scan_agg_pipe<N, IS_HASH>), dim3(grid),            \
+                       dim3(THREADS), plan.lds_bytes, s, plan, rgn.d_vals,   \
+                       rgn.d_val_offs, rgn.n_kv, d_simple, ht)
+  switch (plan.n_aggs) {
+    case 1: CASE(1); break;
+    case 2: CASE(2); break;
+    case 3: CASE(3); break;
+    case 4: CASE(4); break;
+    default: CASE(COPR_MAX_AGGS); break;
+  }
+  #undef CASE
+  return (int)hipGetLastError();
+}
+
+template <bool IS_HASH>
+static int launch_agg(const ScanPlan &plan, const DevRegion &rgn,
+                      SimpleAggAcc *d_simple, HashAggTable ht, hipStream_t s,
+                      uint32_t grid) {
+  if (plan.use_pipe)
+    return launch_agg_pipe<IS_HASH>(plan, rgn, d_simple, ht, s, grid);
+  uint32_t nl = (plan.lds_bytes + 16 * THREADS - 1) / (16 * THREADS);
+  if (nl <= 4) return launch_agg2<IS_HASH, 4>(plan, rgn, d_simple, ht, s, grid);
+  if (nl <= 9) return launch_agg2<IS_HASH, 9>(plan, rgn, d_simple, ht, s, grid);
+  return launch_agg2<IS_HASH, 16>(plan, rgn, d_simple, ht, s, grid);
 }
 
 int dev_scan_launch(const ScanPlan &plan, const DevRegion &rgn,
