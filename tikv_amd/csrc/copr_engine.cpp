@@ -407,6 +407,7 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
 static void pick_tiling(const DevRegion &rgn, ScanPlan *sp) {
   sp->diag_stage_only = getenv("COPR_DIAG_STAGE_ONLY") ? 1 : 0;
   sp->use_pipe = 0;
+  sp->glds_nt = getenv("COPR_GLDS_NT") ? 1 : 0;
   uint32_t per_row = rgn.max_row_bytes + 1;
 
   if (!getenv("COPR_NO_PIPE")) {

@@ -78,6 +78,7 @@ struct ScanPlan {
      multiples; lds_bytes = 2 * (offs_slab + vals_slab). use_pipe 0 falls
      back to the single-buffer kernel. */
   int32_t use_pipe;
+  int32_t glds_nt;               /* nt (aux=2) on the values stream */
   uint32_t offs_slab;
   uint32_t vals_slab;
 };

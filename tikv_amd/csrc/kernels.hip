@@ -480,32 +480,20 @@ k_scan_agg(ScanPlan plan,
 
   bool any_parse_err = false;
 
-  StageRegs<NLOADS> regs;
-  TileInfo cur, nxt;
-  uint64_t tile = blockIdx.x;
-  if (tile < n_tiles) {
-    tile_info<NLOADS>(val_offs, n_rows, rpt, tile, &cur);
-    tile_load<NLOADS>(vals, cur, regs);
-  }
-  for (; tile < n_tiles; tile += gridDim.x) {
+  for (uint64_t tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
+    uint64_t row0 = tile * rpt;
+    uint64_t row1 = min(row0 + rpt, n_rows);
+    uint64_t gbase = val_offs[row0];
+    uint32_t tlen = (uint32_t)(val_offs[row1] - gbase);
     __syncthreads();                    /* previous parse done: LDS free */
-    tile_store<NLOADS>(lds, cur, regs);
-    __syncthreads();                    /* staged tile visible */
-    /* UNCONDITIONALLY load the next tile (clamped: the last iteration
-       re-loads its own tile from L2) so `regs` is never conditionally
-       defined — a merge of partially-defined paths makes the allocator
-       spill the whole staging array to scratch */
-    uint64_t next = min(tile + gridDim.x, n_tiles - 1);
-    tile_info<NLOADS>(val_offs, n_rows, rpt, next, &nxt);
-    tile_load<NLOADS>(vals, nxt, regs);     /* in flight during the parse */
+    uint32_t shift = stage_tile(vals, gbase, tlen, lds);
     if (plan.diag_stage_only) {
-      if (lds[cur.shift] == 0xA5u && threadIdx.x == 1023u) l_cnt[0]++;
-      cur = nxt;
+      if (lds[shift] == 0xA5u && threadIdx.x == 1023u) l_cnt[0]++;
       continue;
     }
 
-    for (uint64_t my_row = cur.row0 + threadIdx.x; my_row < cur.row1; my_row += blockDim.x) {
-      const uint8_t *vp = lds + cur.shift + (uint32_t)(val_offs[my_row] - cur.gbase);
+    for (uint64_t my_row = row0 + threadIdx.x; my_row < row1; my_row += blockDim.x) {
+      const uint8_t *vp = lds + shift + (uint32_t)(val_offs[my_row] - gbase);
       uint32_t vlen = (uint32_t)(val_offs[my_row + 1] - val_offs[my_row]);
       bool parse_ok = true;
 
@@ -636,7 +624,6 @@ k_scan_agg(ScanPlan plan,
         }
       }
     }
-    cur = nxt;
   }
 
   if (!IS_HASH) {
@@ -727,10 +714,19 @@ k_scan_agg_pipe(ScanPlan plan,
     uint32_t vc = (tbytes + 1023u) >> 10;
     const uint8_t *vsrc = vals + abase;
     uint8_t *bv = b + OS;
-    for (uint32_t c = wave; c < vc; c += nwaves) {
-      uint32_t off = (c << 10) + lane * 16u;
-      __builtin_amdgcn_global_load_lds((const uint32_t *)(vsrc + off),
-                                       (uint32_t *)(bv + off), 16, 0, 0);
+    if (plan.glds_nt) {
+      for (uint32_t c = wave; c < vc; c += nwaves) {
+        uint32_t off = (c << 10) + lane * 16u;
+        /* aux=2 (nt): each byte is read once by one CU (streaming) */
+        __builtin_amdgcn_global_load_lds((const uint32_t *)(vsrc + off),
+                                         (uint32_t *)(bv + off), 16, 0, 2);
+      }
+    } else {
+      for (uint32_t c = wave; c < vc; c += nwaves) {
+        uint32_t off = (c << 10) + lane * 16u;
+        __builtin_amdgcn_global_load_lds((const uint32_t *)(vsrc + off),
+                                         (uint32_t *)(bv + off), 16, 0, 0);
+      }
     }
   };
 
