@@ -420,12 +420,12 @@ static void pick_tiling(const DevRegion &rgn, ScanPlan *sp) {
     if (const char *e = getenv("COPR_ROWS_PER_TILE")) rows = (uint32_t)atoi(e);
     while (rows > 64) {
       uint32_t os = (((rows + 1) * 8) + 1023u) & ~1023u;
-      uint32_t vs = ((rows * per_row + 15u) + 1023u) & ~1023u;
+      uint32_t vs = ((rows * per_row + 31u) + 1023u) & ~1023u;
       if (2 * (os + vs) <= budget) break;
       rows /= 2;
     }
     uint32_t os = (((rows + 1) * 8) + 1023u) & ~1023u;
-    uint32_t vs = ((rows * per_row + 15u) + 1023u) & ~1023u;
+    uint32_t vs = ((rows * per_row + 31u) + 1023u) & ~1023u;
     if (rows >= 64 && 2 * (os + vs) <= 160 * 1024 - 2048) {
       sp->use_pipe = 1;
       sp->rows_per_tile = rows;

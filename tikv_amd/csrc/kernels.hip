@@ -664,7 +664,7 @@ k_scan_agg(ScanPlan plan,
  * parse touches no vector global loads (per-row offsets come from the LDS
  * offs slab), so hipcc inserts no early vmcnt(0) that would drain the DMA
  * queue (cdna_hip_programming §5 pipelining-across-barriers / §6 G15). */
-template <int NAGGS, bool IS_HASH>
+template <int NAGGS, bool IS_HASH, bool FASTFC = false>
 __global__ void __launch_bounds__(THREADS, 2)
 k_scan_agg_pipe(ScanPlan plan,
                 const uint8_t *__restrict__ vals,
@@ -750,6 +750,88 @@ k_scan_agg_pipe(ScanPlan plan,
     const uint8_t *bv = b + OS;
 
     if (!plan.diag_stage_only) {
+    if (FASTFC) {
+      /* specialized: one int filter column + count(*) — the cfg2 shape.
+         Streaming 16-byte register window over the row: ~1 ds_read_b64 pair
+         per two cells, no value extraction for non-target cells. */
+      const int64_t FCID = plan.filter_col_id;
+      unsigned long long cnt = 0;
+      for (uint64_t my_row = row0 + threadIdx.x; my_row < row1; my_row += blockDim.x) {
+        uint32_t r = (uint32_t)(my_row - row0);
+        uint64_t o0 = loffs[r], o1 = loffs[r + 1];
+        const uint8_t *vp = bv + shift + (uint32_t)(o0 - gb);
+        uint32_t vlen = (uint32_t)(o1 - o0);
+        bool found = false, fnull = false, ok = true;
+        int64_t fv = 0;
+        if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
+          uintptr_t base = (uintptr_t)vp;
+          uintptr_t wabs = ~(uintptr_t)0;   /* window invalid */
+          uint64_t wlo = 0, whi = 0;
+          uint32_t pos = 0;
+          while (pos < vlen) {
+            uintptr_t ua = base + pos;
+            if (ua - wabs > 8) {            /* reload 16B window */
+              wabs = ua & ~(uintptr_t)7;
+              const uint64_t *q = (const uint64_t *)wabs;
+              wlo = q[0];
+              whi = q[1];
+            }
+            uint32_t sh = (uint32_t)(ua - wabs) * 8u;
+            uint64_t x;
+            if (sh == 0) x = wlo;
+            else if (sh == 64) x = whi;
+            else x = (wlo >> sh) | (whi << (64 - sh));
+            if ((x & 0xFF) != 8) { ok = false; break; }
+            uint32_t b1 = (uint32_t)(x >> 8) & 0xFF;
+            uint32_t dflag = (uint32_t)(x >> 16) & 0xFF;
+            if (b1 < 0x80 && (dflag == 8 || dflag == 9)) {
+              uint64_t m = x >> 24;
+              uint64_t stops = ~m & 0x8080808080ull;
+              if (stops) {
+                uint32_t half = b1 >> 1;
+                int64_t cid = (b1 & 1) ? (int64_t)(~(uint64_t)half) : (int64_t)half;
+                uint32_t n = ((uint32_t)__ffsll((long long)stops)) >> 3;
+                if (cid == FCID) {
+                  uint64_t vm = m & ((n == 5) ? 0xFFFFFFFFFFull
+                                              : ((1ull << (8 * n)) - 1));
+                  uint64_t uv = (vm & 0x7f) | ((vm >> 8) & 0x7f) << 7 |
+                                ((vm >> 16) & 0x7f) << 14 |
+                                ((vm >> 24) & 0x7f) << 21 |
+                                ((vm >> 32) & 0x7f) << 28;
+                  if (dflag == 8) {
+                    uint64_t h2 = uv >> 1;
+                    fv = (uv & 1) ? (int64_t)~h2 : (int64_t)h2;
+                  } else {
+                    fv = (int64_t)uv;
+                  }
+                  found = true;
+                  break;
+                }
+                pos += 3 + n;
+                continue;
+              }
+            }
+            /* uncommon cell: generic parser for this cell */
+            {
+              int64_t cid;
+              uint32_t cell_off;
+              CellView cell;
+              if (!next_cell(vp, vlen, &pos, &cid, &cell_off, &cell)) { ok = false; break; }
+              if (cid == FCID) {
+                if (cell.is_null) fnull = true;
+                else if (cell.has_int) fv = cell.ival;
+                else ok = false;
+                found = true;
+                break;
+              }
+            }
+          }
+        }
+        if (!ok) any_parse_err = true;
+        else if (d_filter_keep(plan, found, fnull, fv)) cnt++;
+      }
+      l_cnt[0] += cnt;
+    } else {
     for (uint64_t my_row = row0 + threadIdx.x; my_row < row1; my_row += blockDim.x) {
       uint32_t r = (uint32_t)(my_row - row0);
       uint64_t o0 = loffs[r], o1 = loffs[r + 1];
@@ -884,6 +966,7 @@ k_scan_agg_pipe(ScanPlan plan,
         }
       }
     }
+    }  /* FASTFC else */
     }  /* !diag */
 
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -1084,6 +1167,13 @@ template <bool IS_HASH>
 static int launch_agg_pipe(const ScanPlan &plan, const DevRegion &rgn,
                            SimpleAggAcc *d_simple, HashAggTable ht,
                            hipStream_t s, uint32_t grid) {
+  if (!IS_HASH && plan.n_aggs == 1 && plan.aggs[0].kind == DAGG_COUNT_ROWS &&
+      plan.has_filter) {
+    hipLaunchKernelGGL((k_scan_agg_pipe<1, false, true>), dim3(grid),
+                       dim3(THREADS), plan.lds_bytes, s, plan, rgn.d_vals,
+                       rgn.d_val_offs, rgn.n_kv, d_simple, ht);
+    return (int)hipGetLastError();
+  }
   #define CASE(N)                                                            \
     hipLaunchKernelGGL((k_scan_agg_pipe<N, IS_HASH>), dim3(grid),            \
                        dim3(THREADS), plan.lds_bytes, s, plan, rgn.d_vals,   \
