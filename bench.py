@@ -40,9 +40,9 @@ def load_oracle():
     return mod
 
 
-def build_request(tikv_amd, F):
+def build_request(tikv_amd, F, filter_offset=3):
     cols = [tikv_amd.Col(i) for i in range(1, 17)]
-    sel = tikv_amd.cmp_col_const(3, F.SIG_LT_INT, FILTER_K)
+    sel = tikv_amd.cmp_col_const(filter_offset, F.SIG_LT_INT, FILTER_K)
     return (tikv_amd.DagSelect(cols).where(sel)
             .simple_agg([tikv_amd.count_star()]).build())
 
@@ -89,6 +89,8 @@ def main():
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--rows", type=int, default=N_ROWS_DEFAULT,
                     help="rows per GPU (dev override; BASELINE value default)")
+    ap.add_argument("--filter-offset", type=int, default=3,
+                    help="dev: which column offset the predicate filters")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
@@ -120,7 +122,7 @@ def main():
     log("generated in %.1fs (%.2f GB values)" %
         (time.perf_counter() - t0, gen.val_bytes() / 1e9))
 
-    req = build_request(tikv_amd, F)
+    req = build_request(tikv_amd, F, args.filter_offset)
     eng = tikv_amd.Engine(local_rank)
     rgn = eng.region(gen)
     if have_cuda:
