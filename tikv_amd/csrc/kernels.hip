@@ -691,13 +691,12 @@ k_scan_agg_pipe(ScanPlan plan,
 
   bool any_parse_err = false;
 
-  auto issue_tile = [&](uint64_t tile, uint8_t *b) {
+  /* issue the HBM->LDS DMA for a tile whose bounds (gb, ge) are ALREADY in
+     registers — no loads between glds issues, so no compiler vmcnt lands in
+     the middle of the queue */
+  auto issue_tile = [&](uint64_t tile, uint8_t *b, uint64_t gb, uint64_t ge) {
     uint64_t row0 = tile * rpt;
     uint64_t row1 = min(row0 + rpt, n_rows);
-    /* read the tile bounds FIRST: their compiler-inserted vmcnt(0) must not
-       land between glds issues (it would drain the DMA queue) */
-    uint64_t gb = val_offs[row0];
-    uint64_t ge = val_offs[row1];
     /* offsets slab: val_offs[row0 .. row1] raw (1 KiB chunks, whole-chunk
        over-read stays inside the +2 KiB region slack) */
     const uint8_t *osrc = (const uint8_t *)(val_offs + row0);
@@ -729,18 +728,33 @@ k_scan_agg_pipe(ScanPlan plan,
       }
     }
   };
+  auto bounds_of = [&](uint64_t tile, uint64_t *gb, uint64_t *ge) {
+    uint64_t row0 = tile * rpt;
+    uint64_t row1 = min(row0 + rpt, n_rows);
+    *gb = val_offs[row0];
+    *ge = val_offs[row1];
+  };
 
   uint64_t tile = blockIdx.x;
   uint32_t bsel = 0;
-  if (tile < n_tiles) issue_tile(tile, lds);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
+  uint64_t nb_gb = 0, nb_ge = 0;        /* bounds of tile+gridDim, in flight */
+  if (tile < n_tiles) {
+    uint64_t gb, ge;
+    bounds_of(tile, &gb, &ge);
+    issue_tile(tile, lds, gb, ge);
+    bounds_of(min(tile + gridDim.x, n_tiles - 1), &nb_gb, &nb_ge);
+  }
 
   for (; tile < n_tiles; tile += gridDim.x) {
+    /* ONE drain per tile: current tile's DMA (issued last iteration, landed
+       during the previous parse) + the prefetched next-tile bounds */
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
     uint8_t *b = lds + bsel * BUFSZ;
     uint8_t *bn = lds + (bsel ^ 1) * BUFSZ;
     uint64_t next = tile + gridDim.x;
-    if (next < n_tiles) issue_tile(next, bn);   /* in flight during parse */
+    if (next < n_tiles) issue_tile(next, bn, nb_gb, nb_ge);
+    bounds_of(min(next + gridDim.x, n_tiles - 1), &nb_gb, &nb_ge);
 
     uint64_t row0 = tile * rpt;
     uint64_t row1 = min(row0 + rpt, n_rows);
@@ -968,9 +982,6 @@ k_scan_agg_pipe(ScanPlan plan,
     }
     }  /* FASTFC else */
     }  /* !diag */
-
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
     bsel ^= 1;
   }
 
