@@ -1016,6 +1016,149 @@ k_scan_agg_pipe(ScanPlan plan,
   }
 }
 
+/* ---------------- direct-window filter+count kernel ----------------
+ * For selective scans (one int predicate column + count(*)), skip LDS
+ * staging entirely: each lane loads an aligned 64 B register window at its
+ * row's start and parses cells in registers. The reference's own parse
+ * stops at the last requested column (table_scan_executor.rs:223), so the
+ * row tail is never touched — HBM traffic drops to the touched prefix
+ * cachelines. 8 waves/SIMD hide the load latency; no barriers, no LDS. */
+__global__ void __launch_bounds__(THREADS)
+k_scan_fc_direct(ScanPlan plan,
+                 const uint8_t *__restrict__ vals,
+                 const uint64_t *__restrict__ val_offs, uint64_t n_rows,
+                 SimpleAggAcc *__restrict__ simple_acc) {
+  const int64_t FCID = plan.filter_col_id;
+  unsigned long long cnt = 0;
+  bool any_err = false;
+
+  for (uint64_t row = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       row < n_rows; row += (uint64_t)gridDim.x * blockDim.x) {
+    uint64_t o0 = val_offs[row];
+    uint64_t o1 = val_offs[row + 1];
+    uint32_t vlen = (uint32_t)(o1 - o0);
+    bool found = false, fnull = false, ok = true;
+    int64_t fv = 0;
+    if (!(vlen == 0)) {
+      /* aligned 64 B register window over the row prefix */
+      const uint4 *g = (const uint4 *)(vals + (o0 & ~15ull));
+      uint4 w0 = g[0], w1 = g[1], w2 = g[2], w3 = g[3];
+      uint64_t q0 = ((uint64_t)w0.y << 32) | w0.x, q1 = ((uint64_t)w0.w << 32) | w0.z;
+      uint64_t q2 = ((uint64_t)w1.y << 32) | w1.x, q3 = ((uint64_t)w1.w << 32) | w1.z;
+      uint64_t q4 = ((uint64_t)w2.y << 32) | w2.x, q5 = ((uint64_t)w2.w << 32) | w2.z;
+      uint64_t q6 = ((uint64_t)w3.y << 32) | w3.x, q7 = ((uint64_t)w3.w << 32) | w3.z;
+      uint32_t sh0 = (uint32_t)(o0 & 15ull);
+      uint32_t wcap = 64 - sh0;                 /* window bytes available */
+      /* byte pos -> 8-byte value via a small select tree */
+      auto win8 = [&](uint32_t pos) -> uint64_t {
+        uint32_t p = sh0 + pos;
+        uint32_t i = p >> 3;
+        uint32_t sh = (p & 7u) * 8u;
+        uint64_t lo, hi;
+        switch (i) {
+          case 0: lo = q0; hi = q1; break;
+          case 1: lo = q1; hi = q2; break;
+          case 2: lo = q2; hi = q3; break;
+          case 3: lo = q3; hi = q4; break;
+          case 4: lo = q4; hi = q5; break;
+          case 5: lo = q5; hi = q6; break;
+          default: lo = q6; hi = q7; break;
+        }
+        if (sh == 0) return lo;
+        return (lo >> sh) | (hi << (64 - sh));
+      };
+      uint32_t pos = 0;
+      uint32_t first = win8(0) & 0xFF;
+      if (vlen == 1 && first == 0) {
+        /* row with no columns */
+      } else {
+        while (pos < vlen) {
+          if (pos + 8 > wcap || pos + 8 > vlen) {
+            /* window exhausted or near row end: generic global-memory path
+               for the rest of this row (rare; also covers malformed rows) */
+            const uint8_t *vp = vals + o0;
+            while (pos < vlen) {
+              int64_t cid;
+              uint32_t cell_off;
+              CellView cell;
+              if (!next_cell(vp, vlen, &pos, &cid, &cell_off, &cell)) { ok = false; break; }
+              if (cid == FCID) {
+                if (cell.is_null) fnull = true;
+                else if (cell.has_int) fv = cell.ival;
+                else ok = false;
+                found = true;
+                break;
+              }
+            }
+            break;
+          }
+          uint64_t x = win8(pos);
+          if ((x & 0xFF) != 8) { ok = false; break; }
+          uint32_t b1 = (uint32_t)(x >> 8) & 0xFF;
+          uint32_t dflag = (uint32_t)(x >> 16) & 0xFF;
+          if (b1 < 0x80 && (dflag == 8 || dflag == 9)) {
+            uint64_t m = x >> 24;
+            uint64_t stops = ~m & 0x8080808080ull;
+            if (stops) {
+              uint32_t half = b1 >> 1;
+              int64_t cid = (b1 & 1) ? (int64_t)(~(uint64_t)half) : (int64_t)half;
+              uint32_t n = ((uint32_t)__ffsll((long long)stops)) >> 3;
+              if (cid == FCID) {
+                uint64_t vm = m & ((n == 5) ? 0xFFFFFFFFFFull
+                                            : ((1ull << (8 * n)) - 1));
+                uint64_t uv = (vm & 0x7f) | ((vm >> 8) & 0x7f) << 7 |
+                              ((vm >> 16) & 0x7f) << 14 |
+                              ((vm >> 24) & 0x7f) << 21 |
+                              ((vm >> 32) & 0x7f) << 28;
+                if (dflag == 8) {
+                  uint64_t h2 = uv >> 1;
+                  fv = (uv & 1) ? (int64_t)~h2 : (int64_t)h2;
+                } else {
+                  fv = (int64_t)uv;
+                }
+                found = true;
+                break;
+              }
+              pos += 3 + n;
+              continue;
+            }
+          }
+          if (b1 < 0x80 && dflag == 0) {
+            uint32_t half = b1 >> 1;
+            int64_t cid = (b1 & 1) ? (int64_t)(~(uint64_t)half) : (int64_t)half;
+            if (cid == FCID) { fnull = true; found = true; break; }
+            pos += 3;
+            continue;
+          }
+          /* uncommon cell: generic parse of this one cell from global */
+          {
+            const uint8_t *vp = vals + o0;
+            int64_t cid;
+            uint32_t cell_off;
+            CellView cell;
+            if (!next_cell(vp, vlen, &pos, &cid, &cell_off, &cell)) { ok = false; break; }
+            if (cid == FCID) {
+              if (cell.is_null) fnull = true;
+              else if (cell.has_int) fv = cell.ival;
+              else ok = false;
+              found = true;
+              break;
+            }
+          }
+        }
+      }
+    }
+    if (!ok) any_err = true;
+    else if (d_filter_keep(plan, found, fnull, fv)) cnt++;
+  }
+
+  /* wave fold + one atomic per wave */
+  for (int off = 32; off > 0; off >>= 1)
+    cnt += (unsigned long long)__shfl_down((long long)cnt, off, 64);
+  if ((threadIdx.x & 63u) == 0 && cnt) atomicAdd(&simple_acc[0].cnt, cnt);
+  if (any_err) atomicOr((unsigned int *)&simple_acc[COPR_MAX_AGGS].cnt, 1u);
+}
+
 /* ---------------- project kernel (row-returning scans) ----------------
  * Not the hot path: dynamic out-column indexing may spill; correctness and
  * byte-identical output shape are what matter here. */
@@ -1180,6 +1323,14 @@ static int launch_agg_pipe(const ScanPlan &plan, const DevRegion &rgn,
                            hipStream_t s, uint32_t grid) {
   if (!IS_HASH && plan.n_aggs == 1 && plan.aggs[0].kind == DAGG_COUNT_ROWS &&
       plan.has_filter) {
+    if (!getenv("COPR_NO_DIRECT")) {
+      uint64_t n_blk = (rgn.n_kv + THREADS - 1) / THREADS;
+      uint32_t dgrid = (uint32_t)(n_blk < 8192 ? n_blk : 8192);
+      if (dgrid == 0) dgrid = 1;
+      hipLaunchKernelGGL(k_scan_fc_direct, dim3(dgrid), dim3(THREADS), 0, s,
+                         plan, rgn.d_vals, rgn.d_val_offs, rgn.n_kv, d_simple);
+      return (int)hipGetLastError();
+    }
     hipLaunchKernelGGL((k_scan_agg_pipe<1, false, true>), dim3(grid),
                        dim3(THREADS), plan.lds_bytes, s, plan, rgn.d_vals,
                        rgn.d_val_offs, rgn.n_kv, d_simple, ht);
