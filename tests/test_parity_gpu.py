@@ -202,3 +202,32 @@ def test_empty_region(engine):
         assert o == g
     finally:
         gen.close()
+
+
+def test_filter_on_late_column_parity(engine):
+    """filter column beyond the fast windows (exercises fallback parse)."""
+    gen = tikv_amd.GenRegion(config_index=1, n_rows=40000, table_id=1)
+    try:
+        cols = [tikv_amd.Col(i) for i in range(1, 17)]
+        sel = tikv_amd.cmp_col_const(15, F.SIG_LT_INT, 0)
+        req = (tikv_amd.DagSelect(cols).where(sel)
+               .simple_agg([tikv_amd.count_star()]).build())
+        (o, on), (g, gn) = run_both(req, gen, engine)
+        assert on == gn == 1
+        assert o == g
+    finally:
+        gen.close()
+
+
+def test_filter_mid_column_parity(engine):
+    gen = tikv_amd.GenRegion(config_index=1, n_rows=40000, table_id=1)
+    try:
+        cols = [tikv_amd.Col(i) for i in range(1, 17)]
+        sel = tikv_amd.cmp_col_const(8, F.SIG_GE_INT, 250_000_000)
+        req = (tikv_amd.DagSelect(cols).where(sel)
+               .simple_agg([tikv_amd.count_star()]).build())
+        (o, on), (g, gn) = run_both(req, gen, engine)
+        assert on == gn == 1
+        assert o == g
+    finally:
+        gen.close()

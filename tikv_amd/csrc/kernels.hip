@@ -1323,7 +1323,7 @@ static int launch_agg_pipe(const ScanPlan &plan, const DevRegion &rgn,
                            hipStream_t s, uint32_t grid) {
   if (!IS_HASH && plan.n_aggs == 1 && plan.aggs[0].kind == DAGG_COUNT_ROWS &&
       plan.has_filter) {
-    if (!getenv("COPR_NO_DIRECT")) {
+    if (getenv("COPR_DIRECT")) {
       uint64_t n_blk = (rgn.n_kv + THREADS - 1) / THREADS;
       uint32_t dgrid = (uint32_t)(n_blk < 8192 ? n_blk : 8192);
       if (dgrid == 0) dgrid = 1;
