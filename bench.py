@@ -69,6 +69,19 @@ def cpu_baseline_leg(gen, req, target_seconds=12.0):
     return sample / dt, "%d rows of the same region, 1 thread" % sample
 
 
+def cpu_checksum_leg(gen, target_seconds=12.0):
+    orc = load_oracle()
+    probe = min(200_000, gen.n_kv)
+    t0 = time.perf_counter()
+    orc.checksum(gen.keys, gen.key_offs, gen.vals, gen.val_offs, probe)
+    dt = time.perf_counter() - t0
+    sample = int(min(gen.n_kv, max(probe, probe / dt * target_seconds)))
+    t0 = time.perf_counter()
+    orc.checksum(gen.keys, gen.key_offs, gen.vals, gen.val_offs, sample)
+    dt = time.perf_counter() - t0
+    return sample / dt, "%d KV pairs of the same region, 1 thread" % sample
+
+
 def read_traffic():
     """Per-launch HBM bytes measured offline with rocprofv3 --pmc (FETCH_SIZE
     x2 gfx950 correction + WRITE_SIZE, per MI355X_MICROARCH.md §HBM), stored
@@ -194,6 +207,8 @@ def main():
     rows_per_sec = total_rows / elapsed
     ms_per_step = elapsed / args.steps * 1e3
     kern_s = kern_ns_total / 1e9 / args.steps
+    if args.workload == "cfg4":
+        kern_s = elapsed / args.steps     # checksum call is synchronous
     achieved_gbs = algo_bytes / kern_s / 1e9 if kern_s > 0 else 0.0
 
     result = None
@@ -202,7 +217,10 @@ def main():
         cpu = None
         if world == 1 and not args.no_cpu_baseline:
             log("cpu baseline (oracle) ...")
-            cpu_rps, sample = cpu_baseline_leg(gen, req)
+            if args.workload == "cfg4":
+                cpu_rps, sample = cpu_checksum_leg(gen)
+            else:
+                cpu_rps, sample = cpu_baseline_leg(gen, req)
             cpu = {"value": cpu_rps, "unit": "rows/s", "cores": 1,
                    "kind": "port", "sample": sample}
         result = {
