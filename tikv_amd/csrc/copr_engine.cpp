@@ -404,13 +404,13 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
 /* tiling: pick rows_per_tile from the region's max row size so the staged
  * tile fits the LDS budget (Guideline: 256-thread blocks; <=64 KiB tile
  * keeps >=2 blocks/CU of occupancy on the 160 KiB LDS). */
-static void pick_tiling(const DevRegion &rgn, ScanPlan *sp) {
+static void pick_tiling(const DevRegion &rgn, ScanPlan *sp, bool force_nopipe = false) {
   sp->diag_stage_only = getenv("COPR_DIAG_STAGE_ONLY") ? 1 : 0;
   sp->use_pipe = 0;
   sp->glds_nt = getenv("COPR_GLDS_NT") ? 1 : 0;
   uint32_t per_row = rgn.max_row_bytes + 1;
 
-  if (!getenv("COPR_NO_PIPE")) {
+  if (!force_nopipe && !getenv("COPR_NO_PIPE")) {
     /* glds double-buffer pipeline: 2 x (offs slab + vals slab), 1 KiB
        granular. Target <= ~120 KiB so at least one more block can coexist
        on smaller shapes; >= 2 blocks/CU needs <= 76 KiB. */
@@ -619,7 +619,21 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
         for (uint32_t rg = 0; rg < n_regions; rg++) {
           ScanPlan sp = pl.sp;
           sp.table_size = tsize;
-          pick_tiling(regions[rg]->dev, &sp);
+          /* hash mode runs the single-buffer kernel with a per-block LDS
+             pre-aggregation table appended after the tile region */
+          pick_tiling(regions[rg]->dev, &sp, /*force_nopipe=*/true);
+          if (sp.lds_bytes > 70 * 1024) {
+            /* shrink the tile so table + tile keep >=2 blocks/CU */
+            uint32_t per_row = regions[rg]->dev.max_row_bytes + 1;
+            uint32_t rows = sp.rows_per_tile;
+            while (rows > 64 && (uint64_t)rows * per_row + 96 > 48 * 1024) rows /= 2;
+            sp.rows_per_tile = rows;
+            sp.lds_bytes = (uint32_t)((uint64_t)rows * per_row + 96);
+          }
+          sp.lds_agg_slots = 256;
+          sp.lds_agg_off = (sp.lds_bytes + 15u) & ~15u;
+          sp.lds_bytes = sp.lds_agg_off +
+                         sp.lds_agg_slots * (8 + (uint32_t)sp.n_aggs * (uint32_t)sizeof(SimpleAggAcc));
           int le = dev_scan_launch(sp, regions[rg]->dev, nullptr, &ht, nullptr, eng->stream);
           if (le) { free_ht(); return SET_ERR(COPR_ERR_INTERNAL, "scan launch failed"); }
         }

@@ -81,6 +81,11 @@ struct ScanPlan {
   int32_t glds_nt;               /* nt (aux=2) on the values stream */
   uint32_t offs_slab;
   uint32_t vals_slab;
+  /* hash agg: per-block LDS pre-aggregation table (0 = disabled).
+     Low-cardinality GROUP BY otherwise serializes on a handful of global
+     atomic addresses. */
+  uint32_t lds_agg_slots;        /* power of two */
+  uint32_t lds_agg_off;          /* byte offset of the table in LDS */
 };
 
 /* simple-agg accumulators (device buffer, one per agg) */
