@@ -1458,7 +1458,13 @@ int dev_crc64_launch(const DevRegion &rgn, const uint64_t *d_tables,
   uint32_t rpt = 256;
   uint32_t max_key = 64;
   uint64_t max_tile_val = (uint64_t)rpt * (rgn.max_row_bytes + 16) + 32;
-  while (rpt > 64 && 16384 + rpt * max_key + max_tile_val > 160 * 1024 - 1024) {
+  /* keep tables+keys+values <= ~52 KiB so >=3 blocks/CU stay resident —
+     the per-row CRC chain is latency-bound and needs wave parallelism */
+  while (rpt > 32 && 16384 + rpt * max_key + max_tile_val > 52 * 1024) {
+    rpt /= 2;
+    max_tile_val = (uint64_t)rpt * (rgn.max_row_bytes + 16) + 32;
+  }
+  while (rpt > 1 && 16384 + rpt * max_key + max_tile_val > 158 * 1024) {
     rpt /= 2;
     max_tile_val = (uint64_t)rpt * (rgn.max_row_bytes + 16) + 32;
   }
