@@ -56,6 +56,18 @@ copr_status copr_region_create(copr_engine *,
                                const uint8_t *keys, const uint64_t *key_offs,
                                const uint8_t *vals, const uint64_t *val_offs,
                                uint64_t n_kv, copr_region **out);
+
+/* MVCC variant: the feed is the raw write-CF stream (sorted user-key asc /
+ * commit_ts desc; key = memcomparable(user_key)||BE(~commit_ts), value =
+ * WriteRef bytes — txn_types/src/write.rs:296-361) and the engine's version
+ * filter kernel materializes the visible rows at read_ts on device
+ * (forward.rs:440-515 semantics; Put values must be short values —
+ * default-CF lookup is COPR_ERR_UNSUPPORTED). */
+copr_status copr_region_create_mvcc(copr_engine *,
+                                    const uint8_t *keys, const uint64_t *key_offs,
+                                    const uint8_t *vals, const uint64_t *val_offs,
+                                    uint64_t n_kv, uint64_t read_ts,
+                                    copr_region **out);
 void        copr_region_destroy(copr_region *);
 uint64_t    copr_region_num_kv(const copr_region *);
 
@@ -97,6 +109,10 @@ typedef struct CoprGenOut {   /* host buffers owned by the generator */
 
 copr_status copr_gen_region(const CoprGenSpec *, CoprGenOut *out);
 void        copr_gen_free(CoprGenOut *);
+
+/* test/debug: copy a resident region back to host buffers (owned by the
+ * engine allocator; free with copr_gen_free) */
+copr_status copr_region_dump(copr_engine *, copr_region *, CoprGenOut *out);
 
 #ifdef __cplusplus
 }

@@ -60,6 +60,20 @@ int orc_test_dec_add_encode(const uint8_t *a, uint64_t alen,
                             const uint8_t *b, uint64_t blen, uint8_t *out);
 int orc_test_dec_from_i64_encode(int64_t v, uint8_t *out);
 
+/* ---- MVCC write-CF version filter (test-side restatement of
+ * forward.rs:433-515 LatestKvPolicy + write.rs:296-361 WriteRef::parse,
+ * for records without gc_fence/last_change tags) ---- */
+typedef struct OrcRegion {
+  uint8_t *keys; uint64_t *key_offs;
+  uint8_t *vals; uint64_t *val_offs;
+  uint64_t n_kv;
+} OrcRegion;
+/* 0 ok; 1 malformed; 2 unsupported (gc_fence / default-CF value) */
+int  orc_mvcc_filter(const uint8_t *keys, const uint64_t *key_offs,
+                     const uint8_t *vals, const uint64_t *val_offs,
+                     uint64_t n_kv, uint64_t read_ts, OrcRegion *out);
+void orc_region_free(OrcRegion *);
+
 #ifdef __cplusplus
 }
 #endif

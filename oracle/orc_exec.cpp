@@ -1126,4 +1126,151 @@ int orc_test_dec_from_i64_encode(int64_t v, uint8_t *out) {
   return (int)dec_encode(d, prec, frac, out);
 }
 
+/* ---- MVCC write-CF filter ----
+ * Key: memcomparable(user_key) || BE(~commit_ts) (types.rs:152-161).
+ * Value: [type][varint start_ts][tags 'v' len sv | 'R' | 'F' u64 |
+ * 'l' u64+varint | 'S' varint; unknown tag stops] (write.rs:296-361).
+ * Visibility (forward.rs:440-515): newest version with commit_ts <= read_ts;
+ * Put -> row (short value; default-CF lookup unsupported here), Delete ->
+ * skip key, Lock/Rollback (no last_change tag -> LastChange::Unknown,
+ * types.rs:721-731) -> next older version. gc_fence unsupported. */
+static bool parse_write_rec(const uint8_t *v, size_t len, char *type,
+                            const uint8_t **sv, size_t *sv_len,
+                            uint64_t *gc_fence, int *lc_not_exist) {
+  *sv = nullptr; *sv_len = 0; *gc_fence = 0; *lc_not_exist = 0;
+  if (len < 1) return false;
+  char t = (char)v[0];
+  if (t != 'P' && t != 'D' && t != 'L' && t != 'R') return false;
+  *type = t;
+  size_t p = 1;
+  uint64_t sts; size_t n;
+  if (!decode_var_u64(v + p, len - p, &sts, &n)) return false;
+  p += n;
+  while (p < len) {
+    uint8_t tag = v[p++];
+    switch (tag) {
+      case 'v': {
+        if (p >= len) return false;
+        uint8_t l = v[p++];
+        if (p + l > len) return false;
+        *sv = v + p; *sv_len = l;
+        p += l;
+        break;
+      }
+      case 'R': break;                         /* overlapped rollback flag */
+      case 'F': {                              /* gc fence: u64 BE */
+        if (p + 8 > len) return false;
+        uint64_t f = 0;
+        for (int b = 0; b < 8; b++) f = (f << 8) | v[p + b];
+        *gc_fence = f;
+        p += 8;
+        break;
+      }
+      case 'l': {                              /* last_change ts + versions */
+        if (p + 8 > len) return false;
+        uint64_t lts = 0;
+        for (int b = 0; b < 8; b++) lts = (lts << 8) | v[p + b];
+        p += 8;
+        uint64_t vers; size_t nn;
+        if (!decode_var_u64(v + p, len - p, &vers, &nn)) return false;
+        p += nn;
+        /* LastChange::from_parts (types.rs:721-731): ts==0 && vers>0 =>
+           NotExist => key invisible; otherwise iterating older versions is
+           semantically equivalent to the reference's seek optimization */
+        if (lts == 0 && vers > 0) *lc_not_exist = 1;
+        break;
+      }
+      case 'S': {
+        uint64_t vv; size_t nn;
+        if (!decode_var_u64(v + p, len - p, &vv, &nn)) return false;
+        p += nn;
+        break;
+      }
+      default:
+        return true;                           /* unknown tag stops parse */
+    }
+  }
+  return true;
+}
+
+extern "C" int orc_mvcc_filter(const uint8_t *keys, const uint64_t *key_offs,
+                               const uint8_t *vals, const uint64_t *val_offs,
+                               uint64_t n_kv, uint64_t read_ts, OrcRegion *out) {
+  std::vector<uint8_t> okeys, ovals;
+  std::vector<uint64_t> okoffs{0}, ovoffs{0};
+  auto uenc = [&](uint64_t i, size_t *len) -> const uint8_t * {
+    size_t kl = (size_t)(key_offs[i + 1] - key_offs[i]);
+    if (kl < 9) return nullptr;
+    *len = kl - 8;
+    return keys + key_offs[i];
+  };
+  auto commit_ts = [&](uint64_t i) -> uint64_t {
+    const uint8_t *p = keys + key_offs[i + 1] - 8;
+    uint64_t d = 0;
+    for (int b = 0; b < 8; b++) d = (d << 8) | p[b];
+    return ~d;
+  };
+  uint64_t i = 0;
+  while (i < n_kv) {
+    size_t ulen;
+    const uint8_t *u = uenc(i, &ulen);
+    if (!u) return 1;
+    /* walk this key group */
+    uint64_t j = i;
+    while (j < n_kv) {
+      size_t ul2;
+      const uint8_t *u2 = uenc(j, &ul2);
+      if (!u2 || ul2 != ulen || memcmp(u, u2, ulen) != 0) break;
+      uint64_t ts = commit_ts(j);
+      if (ts > read_ts) { j++; continue; }
+      char type; const uint8_t *sv; size_t svl; uint64_t fence; int lc_ne;
+      if (!parse_write_rec(vals + val_offs[j],
+                           (size_t)(val_offs[j + 1] - val_offs[j]),
+                           &type, &sv, &svl, &fence, &lc_ne))
+        return 1;
+      /* gc fence pointing within read_ts: key invisible
+         (write.rs:425-442 + forward.rs:444-446 break None) */
+      if (fence != 0 && fence <= read_ts) break;
+      if (type == 'P') {
+        if (!sv) return 2;                     /* default-CF value */
+        std::vector<uint8_t> raw;
+        if (!memcmp_decode(u, ulen, &raw)) return 1;
+        okeys.insert(okeys.end(), raw.begin(), raw.end());
+        okoffs.push_back(okeys.size());
+        ovals.insert(ovals.end(), sv, sv + svl);
+        ovoffs.push_back(ovals.size());
+        break;
+      }
+      if (type == 'D') break;
+      if (lc_ne) break;                        /* LastChange::NotExist */
+      j++;                                     /* Lock/Rollback: older */
+    }
+    /* advance to the next user key */
+    while (i < n_kv) {
+      size_t ul2;
+      const uint8_t *u2 = uenc(i, &ul2);
+      if (!u2) return 1;
+      if (ul2 != ulen || memcmp(u, u2, ulen) != 0) break;
+      i++;
+    }
+  }
+  uint64_t n = okoffs.size() - 1;
+  out->n_kv = n;
+  out->keys = (uint8_t *)malloc(okeys.size() ? okeys.size() : 1);
+  memcpy(out->keys, okeys.data(), okeys.size());
+  out->vals = (uint8_t *)malloc(ovals.size() ? ovals.size() : 1);
+  memcpy(out->vals, ovals.data(), ovals.size());
+  out->key_offs = (uint64_t *)malloc((n + 1) * 8);
+  memcpy(out->key_offs, okoffs.data(), (n + 1) * 8);
+  out->val_offs = (uint64_t *)malloc((n + 1) * 8);
+  memcpy(out->val_offs, ovoffs.data(), (n + 1) * 8);
+  return 0;
+}
+
+extern "C" void orc_region_free(OrcRegion *r) {
+  if (!r) return;
+  free(r->keys); free(r->key_offs); free(r->vals); free(r->val_offs);
+  memset(r, 0, sizeof(*r));
+}
+
 }  // extern "C"

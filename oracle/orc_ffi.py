@@ -22,6 +22,12 @@ class OrcResult(C.Structure):
                 ("n_rows", C.c_uint64)]
 
 
+class OrcRegion(C.Structure):
+    _fields_ = [("keys", C.POINTER(C.c_uint8)), ("key_offs", C.POINTER(C.c_uint64)),
+                ("vals", C.POINTER(C.c_uint8)), ("val_offs", C.POINTER(C.c_uint64)),
+                ("n_kv", C.c_uint64)]
+
+
 _lib = None
 
 
@@ -42,6 +48,11 @@ def load_lib():
                                  C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
                                  C.c_uint64, C.POINTER(C.c_uint64),
                                  C.POINTER(C.c_uint64), C.POINTER(C.c_uint64)]
+    lib.orc_mvcc_filter.restype = C.c_int
+    lib.orc_mvcc_filter.argtypes = [C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
+                                    C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
+                                    C.c_uint64, C.c_uint64, C.POINTER(OrcRegion)]
+    lib.orc_region_free.argtypes = [C.POINTER(OrcRegion)]
     lib.orc_crc64_xz.restype = C.c_uint64
     lib.orc_crc64_xz.argtypes = [C.c_char_p, C.c_uint64]
     lib.orc_test_memcmp_encode.restype = C.c_uint64
@@ -94,3 +105,20 @@ def checksum(keys, key_offs, vals, val_offs, n_kv):
                           C.byref(cs), C.byref(kvs), C.byref(byts))
     assert st == 0
     return cs.value, kvs.value, byts.value
+
+
+def mvcc_filter(keys, key_offs, vals, val_offs, n_kv, read_ts):
+    """Run the oracle MVCC filter; returns (keys, key_offs, vals, val_offs, n)."""
+    lib = load_lib()
+    out = OrcRegion()
+    st = lib.orc_mvcc_filter(keys, key_offs, vals, val_offs, n_kv, read_ts,
+                             C.byref(out))
+    if st != 0:
+        raise RuntimeError("orc_mvcc_filter: %d" % st)
+    n = out.n_kv
+    koffs = [out.key_offs[i] for i in range(n + 1)]
+    voffs = [out.val_offs[i] for i in range(n + 1)]
+    keysb = C.string_at(out.keys, koffs[-1]) if koffs[-1] else b""
+    valsb = C.string_at(out.vals, voffs[-1]) if voffs[-1] else b""
+    lib.orc_region_free(C.byref(out))
+    return keysb, koffs, valsb, voffs, n

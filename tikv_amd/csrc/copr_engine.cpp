@@ -118,6 +118,84 @@ extern "C" copr_status copr_region_create(copr_engine *eng,
   return COPR_OK;
 }
 
+extern "C" copr_status copr_region_create_mvcc(copr_engine *eng,
+                                               const uint8_t *keys,
+                                               const uint64_t *key_offs,
+                                               const uint8_t *vals,
+                                               const uint64_t *val_offs,
+                                               uint64_t n_kv, uint64_t read_ts,
+                                               copr_region **out) {
+  if (!eng) return SET_ERR(COPR_ERR_INVALID_REQUEST, "null engine");
+  HIP_TRY(hipSetDevice(eng->device), "hipSetDevice");
+  /* upload raw write-CF arrays (temporary) */
+  uint8_t *rk = nullptr, *rv = nullptr;
+  uint64_t *rko = nullptr, *rvo = nullptr;
+  hipError_t e = hipSuccess;
+  auto up = [&](void **dst, const void *srcp, uint64_t bytes) {
+    if (e != hipSuccess) return;
+    e = hipMalloc(dst, bytes + 2048);
+    if (e == hipSuccess) e = hipMemcpy(*dst, srcp, bytes, hipMemcpyHostToDevice);
+  };
+  up((void **)&rk, keys, key_offs[n_kv]);
+  up((void **)&rko, key_offs, (n_kv + 1) * 8);
+  up((void **)&rv, vals, val_offs[n_kv]);
+  up((void **)&rvo, val_offs, (n_kv + 1) * 8);
+  auto free_raw = [&]() { hipFree(rk); hipFree(rko); hipFree(rv); hipFree(rvo); };
+  if (e != hipSuccess) {
+    free_raw();
+    return SET_ERR(e == hipErrorOutOfMemory ? COPR_ERR_OOM : COPR_ERR_INTERNAL,
+                   "mvcc upload failed");
+  }
+  DevRegion dr{};
+  int unsup = 0;
+  int rc = dev_mvcc_build(rk, rko, rv, rvo, n_kv, read_ts, &dr, &unsup,
+                          eng->stream);
+  free_raw();
+  if (rc == -3) return SET_ERR(COPR_ERR_UNSUPPORTED, "mvcc: default-CF value");
+  if (rc == -2) return SET_ERR(COPR_ERR_OOM, "mvcc build oom");
+  if (rc != 0) return SET_ERR(COPR_ERR_STORAGE, "malformed write-CF entry");
+  copr_region *r = new copr_region();
+  r->eng = eng;
+  r->dev = dr;
+  r->h_key_offs.resize(dr.n_kv + 1);
+  r->h_val_offs.resize(dr.n_kv + 1);
+  hipError_t ce = hipMemcpy(r->h_key_offs.data(), dr.d_key_offs,
+                            (dr.n_kv + 1) * 8, hipMemcpyDeviceToHost);
+  if (ce == hipSuccess)
+    ce = hipMemcpy(r->h_val_offs.data(), dr.d_val_offs, (dr.n_kv + 1) * 8,
+                   hipMemcpyDeviceToHost);
+  if (ce != hipSuccess) {
+    copr_region_destroy(r);
+    return SET_ERR(COPR_ERR_INTERNAL, "mvcc offs readback");
+  }
+  *out = r;
+  return COPR_OK;
+}
+
+extern "C" copr_status copr_region_dump(copr_engine *eng, copr_region *r,
+                                        CoprGenOut *out) {
+  if (!eng || !r) return SET_ERR(COPR_ERR_INVALID_REQUEST, "null arg");
+  HIP_TRY(hipSetDevice(eng->device), "hipSetDevice");
+  memset(out, 0, sizeof(*out));
+  uint64_t n = r->dev.n_kv;
+  out->n_kv = n;
+  out->keys = (uint8_t *)malloc(r->dev.key_bytes + 1);
+  out->vals = (uint8_t *)malloc(r->dev.val_bytes + 1);
+  out->key_offs = (uint64_t *)malloc((n + 1) * 8);
+  out->val_offs = (uint64_t *)malloc((n + 1) * 8);
+  if (!out->keys || !out->vals || !out->key_offs || !out->val_offs)
+    return SET_ERR(COPR_ERR_OOM, "dump alloc");
+  HIP_TRY(hipMemcpy(out->keys, r->dev.d_keys, r->dev.key_bytes,
+                    hipMemcpyDeviceToHost), "dump keys");
+  HIP_TRY(hipMemcpy(out->vals, r->dev.d_vals, r->dev.val_bytes,
+                    hipMemcpyDeviceToHost), "dump vals");
+  HIP_TRY(hipMemcpy(out->key_offs, r->dev.d_key_offs, (n + 1) * 8,
+                    hipMemcpyDeviceToHost), "dump koffs");
+  HIP_TRY(hipMemcpy(out->val_offs, r->dev.d_val_offs, (n + 1) * 8,
+                    hipMemcpyDeviceToHost), "dump voffs");
+  return COPR_OK;
+}
+
 extern "C" void copr_region_destroy(copr_region *r) {
   if (!r) return;
   hipFree(r->dev.d_keys); hipFree(r->dev.d_key_offs);

@@ -122,6 +122,19 @@ static inline void put_cell_bytes(std::string &buf, int64_t col_id,
   buf.append((const char *)data, len);
 }
 
+/* memcomparable bytes (codec/src/byte.rs:67-101): 8-byte groups + marker */
+static inline void enc_memcmp(std::string &out, const uint8_t *src, size_t len) {
+  size_t full = len / 8;
+  for (size_t g = 0; g < full; g++) {
+    out.append((const char *)src + g * 8, 8);
+    out.push_back((char)0xFF);
+  }
+  size_t rem = len - full * 8;
+  out.append((const char *)src + full * 8, rem);
+  out.append(8 - rem, '\0');
+  out.push_back((char)(0xFF - (8 - rem)));
+}
+
 static bool write_row(const GenCfg &cfg, Xo256 &rng, std::string &buf) {
   switch (cfg.config_index) {
     case 0: {                          /* cfg1: 4 x i64, ids 1..4, ±1e9 */
@@ -162,6 +175,131 @@ static bool write_row(const GenCfg &cfg, Xo256 &rng, std::string &buf) {
 
 }  // namespace
 
+static copr_status copr_gen_region_mvcc(const CoprGenSpec *spec, CoprGenOut *out);
+
+
+/* MVCC write-CF wrap (row_format=3): each logical row becomes 1..3 versions
+ * of a write-CF entry, sorted user-key asc / commit_ts desc:
+ *   key   = memcomparable(record_key) || BE(~commit_ts)
+ *           (txn_types/src/types.rs:152-161, Key::from_raw + append_ts)
+ *   value = [type 'P'|'D'|'L'|'R'][varint start_ts]
+ *           [optional 'v' len short_value]   (txn_types/src/write.rs:296-361)
+ * Reader semantics target read_ts = COPR_MVCC_READ_TS (default 1000):
+ * newest version with commit_ts <= read_ts; Put -> visible (short value),
+ * Delete -> key invisible, Lock/Rollback -> look older
+ * (forward.rs:433-515). */
+static copr_status copr_gen_region_mvcc(const CoprGenSpec *spec, CoprGenOut *out) {
+  GenCfg cfg{spec->config_index, 0};
+  if (spec->config_index == 2) cfg.param = spec->n_cols ? spec->n_cols : 64;
+  uint64_t n = spec->n_rows;
+  uint64_t seed = 0xC0FFEEull + (uint64_t)spec->config_index + 0x77ull;
+
+  int T = 1;
+#ifdef _OPENMP
+  T = omp_get_max_threads();
+#endif
+  if ((uint64_t)T > n / 1024 + 1) T = (int)(n / 1024 + 1);
+  std::vector<std::string> kchunk(T), vchunk(T);
+  std::vector<std::vector<uint32_t>> ksz(T), vsz(T);
+  uint64_t per = (n + T - 1) / T;
+
+#ifdef _OPENMP
+#pragma omp parallel for schedule(static, 1)
+#endif
+  for (int t = 0; t < T; t++) {
+    uint64_t lo = (uint64_t)t * per, hi = lo + per;
+    if (hi > n) hi = n;
+    if (lo >= hi) continue;
+    Xo256 rng;
+    std::string row;
+    uint8_t rk[19];
+    for (uint64_t i = lo; i < hi; i++) {
+      int64_t handle = (int64_t)(spec->first_handle + i);
+      enc_row_key(spec->table_id, handle, rk);
+      rng.seed(seed, (uint64_t)handle);
+      uint32_t nver = 1 + (uint32_t)(rng.next() % 3);
+      uint64_t ts = 900 + rng.next() % 200;           /* newest; read_ts=1000 */
+      for (uint32_t v = 0; v < nver; v++) {
+        /* key */
+        size_t k0 = kchunk[t].size();
+        enc_memcmp(kchunk[t], rk, 19);
+        uint8_t tsb[8];
+        uint64_t desc = ~ts;
+        for (int b = 7; b >= 0; b--) { tsb[b] = (uint8_t)desc; desc >>= 8; }
+        kchunk[t].append((const char *)tsb, 8);
+        ksz[t].push_back((uint32_t)(kchunk[t].size() - k0));
+        /* value */
+        size_t v0 = vchunk[t].size();
+        uint32_t pick = (uint32_t)(rng.next() % 10);
+        char type = pick < 7 ? 'P' : pick == 7 ? 'D' : pick == 8 ? 'R' : 'L';
+        vchunk[t].push_back(type);
+        uint8_t tmp[10];
+        size_t nn = enc_var_u64(tmp, ts ? ts - 1 : 0);  /* start_ts */
+        vchunk[t].append((const char *)tmp, nn);
+        if (type == 'P') {
+          row.clear();
+          if (!write_row(cfg, rng, row)) row.clear();
+          if (row.size() > 255) row.resize(255);       /* short-value cap */
+          vchunk[t].push_back('v');
+          vchunk[t].push_back((char)(uint8_t)row.size());
+          vchunk[t] += row;
+          if ((rng.next() & 15) == 0) {
+            /* gc fence tag: 0 (valid) or a ts below read_ts (invalidates) */
+            uint64_t fence = (rng.next() & 1) ? 0 : 500;
+            vchunk[t].push_back('F');
+            for (int b = 7; b >= 0; b--) vchunk[t].push_back((char)(uint8_t)(fence >> (8 * b)));
+          }
+        } else if (type == 'R' && (rng.next() & 3) == 0) {
+          /* protected rollback short value (write.rs:35) */
+          vchunk[t].push_back('v');
+          vchunk[t].push_back((char)1);
+          vchunk[t].push_back('p');
+        } else if ((type == 'L' || type == 'R') && (rng.next() & 7) == 0) {
+          /* last_change NotExist: BE(0) + varint(1) (types.rs:703-718) */
+          vchunk[t].push_back('l');
+          for (int b = 0; b < 8; b++) vchunk[t].push_back('\0');
+          vchunk[t].push_back((char)1);
+        }
+        vsz[t].push_back((uint32_t)(vchunk[t].size() - v0));
+        ts -= 1 + rng.next() % 100;
+      }
+    }
+  }
+
+  uint64_t total_e = 0, ktot = 0, vtot = 0;
+  std::vector<uint64_t> ebase(T), kbase(T), vbase(T);
+  for (int t = 0; t < T; t++) {
+    ebase[t] = total_e; total_e += ksz[t].size();
+    kbase[t] = ktot; ktot += kchunk[t].size();
+    vbase[t] = vtot; vtot += vchunk[t].size();
+  }
+  uint8_t *keys = (uint8_t *)malloc(ktot ? ktot : 1);
+  uint8_t *vals = (uint8_t *)malloc(vtot ? vtot : 1);
+  uint64_t *key_offs = (uint64_t *)malloc((total_e + 1) * 8);
+  uint64_t *val_offs = (uint64_t *)malloc((total_e + 1) * 8);
+  if (!keys || !vals || !key_offs || !val_offs) return COPR_ERR_OOM;
+
+#ifdef _OPENMP
+#pragma omp parallel for schedule(static, 1)
+#endif
+  for (int t = 0; t < T; t++) {
+    if (!kchunk[t].empty()) memcpy(keys + kbase[t], kchunk[t].data(), kchunk[t].size());
+    if (!vchunk[t].empty()) memcpy(vals + vbase[t], vchunk[t].data(), vchunk[t].size());
+    uint64_t ko = kbase[t], vo = vbase[t], e = ebase[t];
+    for (size_t j = 0; j < ksz[t].size(); j++) {
+      key_offs[e] = ko; val_offs[e] = vo;
+      ko += ksz[t][j]; vo += vsz[t][j];
+      e++;
+    }
+  }
+  key_offs[total_e] = ktot;
+  val_offs[total_e] = vtot;
+  out->keys = keys; out->key_offs = key_offs;
+  out->vals = vals; out->val_offs = val_offs;
+  out->n_kv = total_e;
+  return COPR_OK;
+}
+
 extern "C" {
 
 copr_status copr_gen_region(const CoprGenSpec *spec, CoprGenOut *out) {
@@ -171,6 +309,8 @@ copr_status copr_gen_region(const CoprGenSpec *spec, CoprGenOut *out) {
   if (spec->config_index == 2) cfg.param = spec->n_cols ? spec->n_cols : 64;
   uint64_t n = spec->n_rows;
   uint64_t seed = 0xC0FFEEull + (uint64_t)spec->config_index;
+
+  if (spec->row_format == 3) return copr_gen_region_mvcc(spec, out);
 
   uint64_t *key_offs = (uint64_t *)malloc((n + 1) * sizeof(uint64_t));
   uint64_t *val_offs = (uint64_t *)malloc((n + 1) * sizeof(uint64_t));

@@ -127,6 +127,12 @@ int dev_scan_launch(const ScanPlan &plan, const DevRegion &rgn,
                     const ProjectOut *po, void *stream);
 int dev_crc64_launch(const DevRegion &rgn, const uint64_t *d_tables /*8*256*/,
                      unsigned long long *d_xor, void *stream);
+/* MVCC write-CF filter: builds a visible-row DevRegion from raw write-CF
+ * arrays already on device. 0 ok, -1 malformed, -2 oom, -3 unsupported. */
+int dev_mvcc_build(const uint8_t *d_keys, const uint64_t *d_ko,
+                   const uint8_t *d_vals, const uint64_t *d_vo, uint64_t n,
+                   uint64_t read_ts, DevRegion *out, int *unsupported,
+                   void *stream);
 
 }  // namespace copr
 #endif

@@ -18,6 +18,7 @@
  * XCDs, block b -> XCD b%8).
  */
 #include <hip/hip_runtime.h>
+#include <hipcub/hipcub.hpp>
 #include "copr_internal.h"
 
 namespace copr {
@@ -1368,6 +1369,289 @@ k_crc64(const uint8_t *__restrict__ vals, const uint64_t *__restrict__ val_offs,
   for (int off = 32; off > 0; off >>= 1)
     acc ^= (unsigned long long)__shfl_down((long long)acc, off, 64);
   if ((threadIdx.x & 63u) == 0 && acc) atomicXor(out_xor, acc);
+}
+
+
+/* ---------------- MVCC write-CF version filter ----------------
+ * Input: sorted write-CF entries (user-key asc, commit_ts desc):
+ *   key   = memcomparable(user_key) || BE(~commit_ts)  (types.rs:152-161)
+ *   value = [type][varint start_ts][tags 'v','R','F','l','S']
+ *            (write.rs:296-361)
+ * Output: the visible (raw user_key, short_value) stream — exactly what
+ * TikvStorage hands the scan executors (storage_impl.rs:93).
+ * Rule (forward.rs:440-515): newest version with commit_ts <= read_ts; a
+ * gc fence in (0, read_ts] makes the key invisible (write.rs:425-442);
+ * Put -> emit short value (default-CF lookup unsupported: error),
+ * Delete -> skip key, Lock/Rollback -> older version unless
+ * LastChange::NotExist (types.rs:721-731). */
+__device__ static inline bool d_parse_write_rec(const uint8_t *v, uint32_t len,
+                                                char *type, uint32_t *sv_off,
+                                                uint32_t *sv_len,
+                                                uint64_t *fence, int *lc_ne) {
+  *sv_off = 0; *sv_len = 0; *fence = 0; *lc_ne = 0;
+  if (len < 1) return false;
+  char t = (char)v[0];
+  if (t != 'P' && t != 'D' && t != 'L' && t != 'R') return false;
+  *type = t;
+  uint32_t p = 1;
+  uint64_t sts; uint32_t n;
+  if (!d_var_u64(v + p, len - p, &sts, &n)) return false;
+  p += n;
+  while (p < len) {
+    uint8_t tag = v[p++];
+    if (tag == 'v') {
+      if (p >= len) return false;
+      uint8_t l = v[p++];
+      if (p + l > len) return false;
+      *sv_off = p; *sv_len = l;
+      p += l;
+    } else if (tag == 'R') {
+    } else if (tag == 'F') {
+      if (p + 8 > len) return false;
+      *fence = d_be_u64(v + p);
+      p += 8;
+    } else if (tag == 'l') {
+      if (p + 8 > len) return false;
+      uint64_t lts = d_be_u64(v + p);
+      p += 8;
+      uint64_t vers; uint32_t nn;
+      if (!d_var_u64(v + p, len - p, &vers, &nn)) return false;
+      p += nn;
+      if (lts == 0 && vers > 0) *lc_ne = 1;
+    } else if (tag == 'S') {
+      uint64_t vv; uint32_t nn;
+      if (!d_var_u64(v + p, len - p, &vv, &nn)) return false;
+      p += nn;
+    } else {
+      break;                                   /* unknown tag stops parse */
+    }
+  }
+  return true;
+}
+
+__device__ static inline bool d_ukey_eq(const uint8_t *a, uint32_t alen,
+                                        const uint8_t *b, uint32_t blen) {
+  if (alen != blen) return false;
+  uint32_t i = 0;
+  for (; i + 8 <= alen; i += 8) {
+    uint64_t xa, xb;
+    memcpy(&xa, a + i, 8);
+    memcpy(&xb, b + i, 8);
+    if (xa != xb) return false;
+  }
+  for (; i < alen; i++)
+    if (a[i] != b[i]) return false;
+  return true;
+}
+
+__global__ void __launch_bounds__(THREADS)
+k_mvcc_flags(const uint8_t *__restrict__ keys, const uint64_t *__restrict__ ko,
+             const uint8_t *__restrict__ vals, const uint64_t *__restrict__ vo,
+             uint64_t n, uint64_t read_ts,
+             uint8_t *__restrict__ vis, uint32_t *__restrict__ ksz,
+             uint32_t *__restrict__ vsz, unsigned int *__restrict__ err) {
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (uint64_t)gridDim.x * blockDim.x) {
+    vis[i] = 0; ksz[i] = 0; vsz[i] = 0;
+    uint32_t klen = (uint32_t)(ko[i + 1] - ko[i]);
+    if (klen < 17 || ((klen - 8) % 9) != 0) { atomicOr(err, 1u); continue; }
+    const uint8_t *k = keys + ko[i];
+    uint32_t ulen = klen - 8;
+    uint64_t ts = ~d_be_u64(k + ulen);
+    if (ts > read_ts) continue;
+    char type; uint32_t sv_off, sv_len; uint64_t fence; int lc_ne;
+    if (!d_parse_write_rec(vals + vo[i], (uint32_t)(vo[i + 1] - vo[i]),
+                           &type, &sv_off, &sv_len, &fence, &lc_ne)) {
+      atomicOr(err, 1u);
+      continue;
+    }
+    if (type != 'P') continue;
+    if (fence != 0 && fence <= read_ts) continue;
+    /* every newer same-key version with ts <= read_ts must be a valid
+       Lock/Rollback that keeps iterating */
+    bool emit = true;
+    for (uint64_t j = i; j-- > 0;) {
+      uint32_t klj = (uint32_t)(ko[j + 1] - ko[j]);
+      if (klj < 17) { atomicOr(err, 1u); emit = false; break; }
+      const uint8_t *kj = keys + ko[j];
+      if (!d_ukey_eq(k, ulen, kj, klj - 8)) break;
+      uint64_t tsj = ~d_be_u64(kj + klj - 8);
+      if (tsj > read_ts) continue;
+      char tj; uint32_t so, sl; uint64_t fj; int lnj;
+      if (!d_parse_write_rec(vals + vo[j], (uint32_t)(vo[j + 1] - vo[j]),
+                             &tj, &so, &sl, &fj, &lnj)) {
+        atomicOr(err, 1u);
+        emit = false;
+        break;
+      }
+      if (fj != 0 && fj <= read_ts) { emit = false; break; }
+      if (tj != 'L' && tj != 'R') { emit = false; break; }
+      if (lnj) { emit = false; break; }
+    }
+    if (!emit) continue;
+    if (sv_len == 0 && sv_off == 0) { atomicOr(err, 2u); continue; } /* default CF */
+    /* decoded user key length from the memcomparable groups */
+    uint32_t groups = ulen / 9;
+    uint8_t marker = k[ulen - 1];
+    uint32_t pad = 0xFFu - marker;
+    if (pad > 8) { atomicOr(err, 1u); continue; }
+    vis[i] = 1;
+    ksz[i] = (groups - 1) * 8 + (8 - pad);
+    vsz[i] = sv_len;
+  }
+}
+
+__global__ void __launch_bounds__(THREADS)
+k_mvcc_widen(const uint8_t *__restrict__ vis, const uint32_t *__restrict__ ksz,
+             const uint32_t *__restrict__ vsz, uint64_t n,
+             uint64_t *__restrict__ v64, uint64_t *__restrict__ k64,
+             uint64_t *__restrict__ s64) {
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (uint64_t)gridDim.x * blockDim.x) {
+    v64[i] = vis[i];
+    k64[i] = ksz[i];
+    s64[i] = vsz[i];
+  }
+}
+
+__global__ void __launch_bounds__(THREADS)
+k_mvcc_gather(const uint8_t *__restrict__ keys, const uint64_t *__restrict__ ko,
+              const uint8_t *__restrict__ vals, const uint64_t *__restrict__ vo,
+              uint64_t n,
+              const uint8_t *__restrict__ vis, const uint32_t *__restrict__ ksz,
+              const uint32_t *__restrict__ vsz,
+              const uint64_t *__restrict__ idx_sc,
+              const uint64_t *__restrict__ kb_sc,
+              const uint64_t *__restrict__ vb_sc,
+              uint8_t *__restrict__ out_keys, uint64_t *__restrict__ out_ko,
+              uint8_t *__restrict__ out_vals, uint64_t *__restrict__ out_vo,
+              unsigned long long *__restrict__ max_row) {
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (uint64_t)gridDim.x * blockDim.x) {
+    if (!vis[i]) continue;
+    uint64_t pos = idx_sc[i], kout = kb_sc[i], vout = vb_sc[i];
+    const uint8_t *k = keys + ko[i];
+    uint32_t dklen = ksz[i];
+    /* memcomparable decode: strip the marker byte of each 9-byte group */
+    uint32_t g = 0, w = 0;
+    while (w < dklen) {
+      uint32_t take = dklen - w < 8 ? dklen - w : 8;
+      for (uint32_t b = 0; b < take; b++)
+        out_keys[kout + w + b] = k[g * 9 + b];
+      w += take;
+      g++;
+    }
+    char type; uint32_t sv_off, sv_len; uint64_t fence; int lc_ne;
+    d_parse_write_rec(vals + vo[i], (uint32_t)(vo[i + 1] - vo[i]),
+                      &type, &sv_off, &sv_len, &fence, &lc_ne);
+    const uint8_t *sv = vals + vo[i] + sv_off;
+    for (uint32_t b = 0; b < sv_len; b++) out_vals[vout + b] = sv[b];
+    out_ko[pos] = kout;
+    out_vo[pos] = vout;
+    atomicMax(max_row, (unsigned long long)sv_len);
+  }
+}
+
+int dev_mvcc_build(const uint8_t *d_keys, const uint64_t *d_ko,
+                   const uint8_t *d_vals, const uint64_t *d_vo, uint64_t n,
+                   uint64_t read_ts, DevRegion *out, int *unsupported,
+                   void *stream) {
+  *unsupported = 0;
+  hipStream_t s = (hipStream_t)stream;
+  uint8_t *vis = nullptr;
+  uint32_t *ksz = nullptr, *vsz = nullptr;
+  unsigned int *err = nullptr;
+  uint64_t *v64 = nullptr, *k64 = nullptr, *s64 = nullptr;
+  uint64_t *idx_sc = nullptr, *kb_sc = nullptr, *vb_sc = nullptr;
+  unsigned long long *d_maxrow = nullptr;
+  void *tmp = nullptr;
+  size_t tmp_bytes = 0;
+  hipError_t e = hipSuccess;
+  auto freeall = [&]() {
+    hipFree(vis); hipFree(ksz); hipFree(vsz); hipFree(err);
+    hipFree(v64); hipFree(k64); hipFree(s64);
+    hipFree(idx_sc); hipFree(kb_sc); hipFree(vb_sc);
+    hipFree(d_maxrow); hipFree(tmp);
+  };
+  uint64_t na = n ? n : 1;
+  if (e == hipSuccess) e = hipMalloc(&vis, na);
+  if (e == hipSuccess) e = hipMalloc(&ksz, na * 4);
+  if (e == hipSuccess) e = hipMalloc(&vsz, na * 4);
+  if (e == hipSuccess) e = hipMalloc(&err, 4);
+  if (e == hipSuccess) e = hipMalloc(&v64, na * 8);
+  if (e == hipSuccess) e = hipMalloc(&k64, na * 8);
+  if (e == hipSuccess) e = hipMalloc(&s64, na * 8);
+  if (e == hipSuccess) e = hipMalloc(&idx_sc, na * 8);
+  if (e == hipSuccess) e = hipMalloc(&kb_sc, na * 8);
+  if (e == hipSuccess) e = hipMalloc(&vb_sc, na * 8);
+  if (e == hipSuccess) e = hipMalloc(&d_maxrow, 8);
+  if (e != hipSuccess) { freeall(); return -2; }
+  hipMemsetAsync(err, 0, 4, s);
+  hipMemsetAsync(d_maxrow, 0, 8, s);
+  uint32_t grid = (uint32_t)(((n + THREADS - 1) / THREADS) < 8192
+                                 ? ((n + THREADS - 1) / THREADS) : 8192);
+  if (grid == 0) grid = 1;
+  hipLaunchKernelGGL(k_mvcc_flags, dim3(grid), dim3(THREADS), 0, s,
+                     d_keys, d_ko, d_vals, d_vo, n, read_ts, vis, ksz, vsz, err);
+  hipLaunchKernelGGL(k_mvcc_widen, dim3(grid), dim3(THREADS), 0, s,
+                     vis, ksz, vsz, n, v64, k64, s64);
+  unsigned int h_err = 0;
+  hipMemcpyAsync(&h_err, err, 4, hipMemcpyDeviceToHost, s);
+  if (hipStreamSynchronize(s) != hipSuccess) { freeall(); return -1; }
+  if (h_err & 1) { freeall(); return -1; }
+  if (h_err & 2) { freeall(); *unsupported = 1; return -3; }
+  /* exclusive sums */
+  hipcub::DeviceScan::ExclusiveSum(nullptr, tmp_bytes, v64, idx_sc, (int)n, s);
+  if (hipMalloc(&tmp, tmp_bytes ? tmp_bytes : 16) != hipSuccess) { freeall(); return -2; }
+  hipcub::DeviceScan::ExclusiveSum(tmp, tmp_bytes, v64, idx_sc, (int)n, s);
+  hipcub::DeviceScan::ExclusiveSum(tmp, tmp_bytes, k64, kb_sc, (int)n, s);
+  hipcub::DeviceScan::ExclusiveSum(tmp, tmp_bytes, s64, vb_sc, (int)n, s);
+  /* totals = last scan + last size */
+  uint64_t t_idx = 0, t_kb = 0, t_vb = 0, l_idx = 0, l_kb = 0, l_vb = 0;
+  if (n) {
+    hipMemcpyAsync(&t_idx, idx_sc + n - 1, 8, hipMemcpyDeviceToHost, s);
+    hipMemcpyAsync(&t_kb, kb_sc + n - 1, 8, hipMemcpyDeviceToHost, s);
+    hipMemcpyAsync(&t_vb, vb_sc + n - 1, 8, hipMemcpyDeviceToHost, s);
+    hipMemcpyAsync(&l_idx, v64 + n - 1, 8, hipMemcpyDeviceToHost, s);
+    hipMemcpyAsync(&l_kb, k64 + n - 1, 8, hipMemcpyDeviceToHost, s);
+    hipMemcpyAsync(&l_vb, s64 + n - 1, 8, hipMemcpyDeviceToHost, s);
+  }
+  if (hipStreamSynchronize(s) != hipSuccess) { freeall(); return -1; }
+  uint64_t n_vis = t_idx + l_idx, ktot = t_kb + l_kb, vtot = t_vb + l_vb;
+  /* outputs (region-owned; +2 KiB slack like copr_region_create) */
+  uint8_t *o_keys = nullptr, *o_vals = nullptr;
+  uint64_t *o_ko = nullptr, *o_vo = nullptr;
+  e = hipMalloc(&o_keys, ktot + 2048);
+  if (e == hipSuccess) e = hipMalloc(&o_vals, vtot + 2048);
+  if (e == hipSuccess) e = hipMalloc(&o_ko, (n_vis + 1) * 8 + 64);
+  if (e == hipSuccess) e = hipMalloc(&o_vo, (n_vis + 1) * 8 + 64);
+  if (e != hipSuccess) {
+    hipFree(o_keys); hipFree(o_vals); hipFree(o_ko); hipFree(o_vo);
+    freeall();
+    return -2;
+  }
+  hipLaunchKernelGGL(k_mvcc_gather, dim3(grid), dim3(THREADS), 0, s,
+                     d_keys, d_ko, d_vals, d_vo, n, vis, ksz, vsz,
+                     idx_sc, kb_sc, vb_sc, o_keys, o_ko, o_vals, o_vo, d_maxrow);
+  hipMemcpyAsync(o_ko + n_vis, &ktot, 8, hipMemcpyHostToDevice, s);
+  hipMemcpyAsync(o_vo + n_vis, &vtot, 8, hipMemcpyHostToDevice, s);
+  unsigned long long h_maxrow = 0;
+  hipMemcpyAsync(&h_maxrow, d_maxrow, 8, hipMemcpyDeviceToHost, s);
+  if (hipStreamSynchronize(s) != hipSuccess) {
+    hipFree(o_keys); hipFree(o_vals); hipFree(o_ko); hipFree(o_vo);
+    freeall();
+    return -1;
+  }
+  freeall();
+  out->d_keys = o_keys;
+  out->d_key_offs = o_ko;
+  out->d_vals = o_vals;
+  out->d_val_offs = o_vo;
+  out->n_kv = n_vis;
+  out->key_bytes = ktot;
+  out->val_bytes = vtot;
+  out->max_row_bytes = (uint32_t)h_maxrow;
+  return 0;
 }
 
 /* ---------------- launch wrappers ---------------- */

@@ -265,6 +265,32 @@ class Engine:
                                (st, self._lib.copr_last_error().decode()))
         return Region(self, r)
 
+    def region_mvcc(self, gen: GenRegion, read_ts):
+        """Build a visible-row region from a raw write-CF stream (the
+        device MVCC version filter)."""
+        r = C.c_void_p()
+        st = self._lib.copr_region_create_mvcc(
+            self._h, gen.keys, gen.key_offs, gen.vals, gen.val_offs,
+            gen.n_kv, read_ts, C.byref(r))
+        if st != 0:
+            raise RuntimeError("copr_region_create_mvcc: %d (%s)" %
+                               (st, self._lib.copr_last_error().decode()))
+        return Region(self, r)
+
+    def dump_region(self, region):
+        """test/debug: host copies of a resident region."""
+        out = F.CoprGenOut()
+        st = self._lib.copr_region_dump(self._h, region._h, C.byref(out))
+        if st != 0:
+            raise RuntimeError("copr_region_dump: %d" % st)
+        n = out.n_kv
+        koffs = [out.key_offs[i] for i in range(n + 1)]
+        voffs = [out.val_offs[i] for i in range(n + 1)]
+        keys = C.string_at(out.keys, koffs[-1]) if koffs[-1] else b""
+        vals = C.string_at(out.vals, voffs[-1]) if voffs[-1] else b""
+        self._lib.copr_gen_free(C.byref(out))
+        return keys, koffs, vals, voffs, n
+
     def dag_run(self, req, regions):
         arr = (C.c_void_p * len(regions))(*[r._h for r in regions])
         res = F.CoprSelectResult()
