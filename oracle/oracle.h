@@ -21,6 +21,8 @@ typedef struct OrcResult {
   uint8_t  *data;       /* datum-encoded response rows (runner.rs:1188) */
   uint64_t  data_len;
   uint64_t  n_rows;
+  uint64_t  resume_row; /* paging resume point (row index the scan stopped
+                           at; UINT64_MAX = drained) — runner.rs:915-943 */
 } OrcResult;
 
 /* 0 = ok; nonzero = error (message via orc_last_error) */

@@ -808,7 +808,8 @@ static bool run_pipeline(const CoprDagRequest *req,
                          const uint8_t *keys, const uint64_t *key_offs,
                          const uint8_t *vals, const uint64_t *val_offs,
                          uint64_t n_kv, std::vector<uint8_t> *resp,
-                         uint64_t *n_out_rows) {
+                         uint64_t *n_out_rows, uint64_t *resume_row) {
+  *resume_row = UINT64_MAX;
   Pipeline pl;
   if (!build_pipeline(req, &pl)) return false;
 
@@ -889,6 +890,12 @@ static bool run_pipeline(const CoprDagRequest *req,
         out_rows++;
       }
       emitted_rows += take;
+      /* paging: stop at the batch boundary where accumulated OUTPUT rows
+         reach paging_size (runner.rs:917-921 record_all >= paging_size) */
+      if (req->paging_size && emitted_rows >= req->paging_size && cursor < n_kv) {
+        *resume_row = cursor;
+        break;
+      }
     } else if (!batch.logical_rows.empty()) {
       /* evaluate agg args + group keys over the batch */
       size_t n_logical = batch.logical_rows.size();
@@ -1030,13 +1037,15 @@ int orc_dag_run(const CoprDagRequest *req,
                 const uint8_t *vals, const uint64_t *val_offs,
                 uint64_t n_kv, OrcResult *out) {
   std::vector<uint8_t> resp;
-  uint64_t n_rows = 0;
-  if (!run_pipeline(req, keys, key_offs, vals, val_offs, n_kv, &resp, &n_rows))
+  uint64_t n_rows = 0, resume = UINT64_MAX;
+  if (!run_pipeline(req, keys, key_offs, vals, val_offs, n_kv, &resp, &n_rows,
+                    &resume))
     return 1;
   out->data = (uint8_t *)malloc(resp.size() ? resp.size() : 1);
   memcpy(out->data, resp.data(), resp.size());
   out->data_len = resp.size();
   out->n_rows = n_rows;
+  out->resume_row = resume;
   return 0;
 }
 

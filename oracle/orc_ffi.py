@@ -19,7 +19,7 @@ from tikv_amd._ffi import CoprDagRequest  # noqa: E402
 
 class OrcResult(C.Structure):
     _fields_ = [("data", C.POINTER(C.c_uint8)), ("data_len", C.c_uint64),
-                ("n_rows", C.c_uint64)]
+                ("n_rows", C.c_uint64), ("resume_row", C.c_uint64)]
 
 
 class OrcRegion(C.Structure):
@@ -81,7 +81,7 @@ def load_lib():
     return lib
 
 
-def dag_run(req, keys, key_offs, vals, val_offs, n_kv):
+def dag_run(req, keys, key_offs, vals, val_offs, n_kv, with_resume=False):
     """Run the oracle pipeline. Buffer args are ctypes pointers (e.g. from
     GenRegion) or bytes (auto-wrapped)."""
     lib = load_lib()
@@ -92,7 +92,10 @@ def dag_run(req, keys, key_offs, vals, val_offs, n_kv):
         raise RuntimeError("oracle: %s" % lib.orc_last_error().decode())
     data = C.string_at(res.data, res.data_len) if res.data_len else b""
     n = res.n_rows
+    resume = res.resume_row
     lib.orc_result_free(C.byref(res))
+    if with_resume:
+        return data, n, resume
     return data, n
 
 

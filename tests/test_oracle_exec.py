@@ -312,3 +312,26 @@ def test_oracle_on_generated_cfg1():
         assert data == d_int(1000) + bytes([6]) + out.raw[:nd]
     finally:
         g.close()
+
+
+def test_oracle_paging_resume():
+    cols = [tikv_amd.Col(1)]
+    rows = [(int_handle_key(1, i), row_v1([(1, d_varint(i))])) for i in range(5000)]
+    req = (tikv_amd.DagSelect(cols)
+           .where(tikv_amd.cmp_col_const(0, F.SIG_GE_INT, 0))
+           .paging(100).build())
+    orc = _orc()
+    kb, ko, vb, vo, n = make_region(rows)
+    data, nrows, resume = orc.dag_run(
+        req, C.cast(kb, C.POINTER(C.c_uint8)), ko,
+        C.cast(vb, C.POINTER(C.c_uint8)), vo, n, with_resume=True)
+    # ladder 32,64,128 -> 224 rows scanned/output at the stopping boundary
+    assert nrows == 224 and resume == 224
+    # drained case: paging larger than output
+    req2 = (tikv_amd.DagSelect(cols)
+            .where(tikv_amd.cmp_col_const(0, F.SIG_GE_INT, 0))
+            .paging(10000).build())
+    data, nrows, resume = orc.dag_run(
+        req2, C.cast(kb, C.POINTER(C.c_uint8)), ko,
+        C.cast(vb, C.POINTER(C.c_uint8)), vo, n, with_resume=True)
+    assert nrows == 5000 and resume == 2**64 - 1

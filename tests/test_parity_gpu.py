@@ -231,3 +231,29 @@ def test_filter_mid_column_parity(engine):
         assert o == g
     finally:
         gen.close()
+
+
+def test_paging_resume_parity(engine):
+    """paging: stop at the batch-ladder boundary where output rows reach
+    paging_size; resume point identical to the oracle (runner.rs:917-943)."""
+    import importlib.util
+    gen = tikv_amd.GenRegion(config_index=0, n_rows=30000, table_id=1)
+    orc = _orc()
+    try:
+        cols = [tikv_amd.Col(i) for i in range(1, 5)]
+        sel = tikv_amd.cmp_col_const(1, F.SIG_GT_INT, 0)   # ~50% selectivity
+        req = (tikv_amd.DagSelect(cols).where(sel).paging(500)
+               .output([0, 1]).build())
+        o_data, o_n, o_resume = orc.dag_run(req, gen.keys, gen.key_offs,
+                                            gen.vals, gen.val_offs, gen.n_kv,
+                                            with_resume=True)
+        rgn = engine.region(gen)
+        try:
+            g_data, g_n, _, g_resume = engine.dag_run(req, [rgn], with_resume=True)
+        finally:
+            rgn.close()
+        assert o_n == g_n and o_resume == g_resume
+        assert o_resume != 2**64 - 1          # stopped early
+        assert o_data == g_data
+    finally:
+        gen.close()

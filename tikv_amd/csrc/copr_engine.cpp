@@ -827,8 +827,26 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
         HIP_TRY(hipMemcpy(h_vals.data(), r->dev.d_vals, r->dev.val_bytes,
                           hipMemcpyDeviceToHost), "vals D2H");
       }
+      /* paging (runner.rs:917-943): walk the reference's batch ladder
+         (32 -> x2 -> 1024) over the keep flags and stop at the first batch
+         boundary where accumulated OUTPUT rows reach paging_size */
+      uint64_t scan_end = n;
+      if (req->paging_size && n_regions == 1) {
+        uint64_t bs = 32, pos = 0, outcnt = 0;
+        while (pos < n) {
+          uint64_t e2 = std::min(n, pos + bs);
+          for (uint64_t i = pos; i < e2; i++) outcnt += h_keep[i] ? 1 : 0;
+          pos = e2;
+          if (outcnt >= req->paging_size) break;
+          if (bs < 1024) bs *= 2;
+        }
+        if (pos < n) {
+          scan_end = pos;
+          out->resume_row = pos;
+        }
+      }
       std::vector<uint8_t> rowbuf;
-      for (uint64_t i = 0; i < n && n_rows_out < pl.limit; i++) {
+      for (uint64_t i = 0; i < scan_end && n_rows_out < pl.limit; i++) {
         if (!h_keep[i]) continue;
         const uint8_t *vbase = nullptr;
         if (whole) vbase = h_vals.data();

@@ -155,6 +155,10 @@ class DagSelect:
         self.executors.append(ex)
         return self
 
+    def paging(self, size):
+        self._paging = size
+        return self
+
     def output(self, offsets):
         self.output_offsets = list(offsets)
         return self
@@ -170,6 +174,7 @@ class DagSelect:
         req.n_executors = len(self.executors)
         req.output_offsets = offs
         req.n_output_offsets = len(self.output_offsets)
+        req.paging_size = getattr(self, "_paging", 0)
         self._req = req
         return req
 
@@ -291,7 +296,7 @@ class Engine:
         self._lib.copr_gen_free(C.byref(out))
         return keys, koffs, vals, voffs, n
 
-    def dag_run(self, req, regions):
+    def dag_run(self, req, regions, with_resume=False):
         arr = (C.c_void_p * len(regions))(*[r._h for r in regions])
         res = F.CoprSelectResult()
         st = self._lib.copr_dag_run(self._h, C.byref(req), arr, len(regions),
@@ -302,7 +307,10 @@ class Engine:
         data = C.string_at(res.data, res.data_len) if res.data_len else b""
         n_rows = res.n_rows
         kernel_ns = res.summaries[0].time_processed_ns if res.n_summaries else 0
+        resume = res.resume_row
         self._lib.copr_result_free(C.byref(res))
+        if with_resume:
+            return data, n_rows, kernel_ns, resume
         return data, n_rows, kernel_ns
 
     def checksum(self, regions):
