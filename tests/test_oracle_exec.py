@@ -335,3 +335,22 @@ def test_oracle_paging_resume():
         req2, C.cast(kb, C.POINTER(C.c_uint8)), ko,
         C.cast(vb, C.POINTER(C.c_uint8)), vo, n, with_resume=True)
     assert nrows == 5000 and resume == 2**64 - 1
+
+
+def test_oracle_row_v2_generated():
+    """generator row_format=2 parses through the oracle scan pipeline."""
+    g = tikv_amd.GenRegion(config_index=0, n_rows=3000, table_id=1, row_format=2)
+    try:
+        cols = [tikv_amd.Col(i) for i in range(1, 5)]
+        req = tikv_amd.DagSelect(cols).simple_agg(
+            [tikv_amd.count_star(), tikv_amd.count_col(0)]).build()
+        orc = _orc()
+        data, n = orc.dag_run(req, g.keys, g.key_offs, g.vals, g.val_offs, g.n_kv)
+        assert n == 1
+        total = int.from_bytes(data[1:9], "big") ^ (1 << 63)
+        nonnull = int.from_bytes(data[10:18], "big") ^ (1 << 63)
+        assert total == 3000
+        assert 0 < nonnull <= 3000   # ~1/64 of col1 values are NULL
+        assert nonnull < 3000        # with 3000 rows some NULLs occur w.h.p.
+    finally:
+        g.close()

@@ -257,3 +257,38 @@ def test_paging_resume_parity(engine):
         assert o_data == g_data
     finally:
         gen.close()
+
+
+def test_row_v2_count_filter_parity(engine):
+    """row-v2 values (generator row_format=2): device v2 decode vs oracle."""
+    gen = tikv_amd.GenRegion(config_index=1, n_rows=120000, table_id=1,
+                             row_format=2)
+    try:
+        cols = [tikv_amd.Col(i) for i in range(1, 17)]
+        sel = tikv_amd.cmp_col_const(3, F.SIG_LT_INT, -800_000_000)
+        req = (tikv_amd.DagSelect(cols).where(sel)
+               .simple_agg([tikv_amd.count_star(), tikv_amd.sum_col(0),
+                            tikv_amd.avg_col(5)]).build())
+        (o, on), (g, gn) = run_both(req, gen, engine)
+        assert on == gn == 1
+        assert o == g
+    finally:
+        gen.close()
+
+
+def test_row_v2_hash_agg_parity(engine):
+    gen = tikv_amd.GenRegion(config_index=2, n_rows=100000, table_id=1,
+                             n_cols=64, row_format=2)
+    try:
+        cols = [tikv_amd.Col(1),
+                tikv_amd.Col(2, tp=F.TP_NEWDECIMAL, decimal=2),
+                tikv_amd.Col(3, tp=F.TP_VARCHAR)]
+        req = tikv_amd.DagSelect(cols).hash_agg(
+            [tikv_amd.count_star(), tikv_amd.sum_col(1, decimal=2),
+             tikv_amd.count_col(2)],
+            tikv_amd.Expr().col(0)).build()
+        (o, on), (g, gn) = run_both(req, gen, engine)
+        assert on == gn
+        assert sorted(split_rows(o, 4)) == sorted(split_rows(g, 4))
+    finally:
+        gen.close()

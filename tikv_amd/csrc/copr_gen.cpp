@@ -173,6 +173,84 @@ static bool write_row(const GenCfg &cfg, Xo256 &rng, std::string &buf) {
   }
 }
 
+/* row v2 writer (small layout): header [128][flags][nn u16le][null u16le],
+ * sorted u8 ids (non-null then null), u16le end-offsets, then cell payloads
+ * (codec/row/v2/row_slice.rs:76-168; cell encodings compat_v1.rs:12-38:
+ * ints little-endian minimal width — 1/2/4/8, negatives 8). */
+static void v2_int_cell(std::string &cells, int64_t v) {
+  if (v >= 0 && v <= 0xFF) cells.push_back((char)(uint8_t)v);
+  else if (v >= 0 && v <= 0xFFFF) {
+    cells.push_back((char)(uint8_t)v); cells.push_back((char)(uint8_t)(v >> 8));
+  } else if (v >= 0 && v <= 0xFFFFFFFFll) {
+    for (int b = 0; b < 4; b++) cells.push_back((char)(uint8_t)(v >> (8 * b)));
+  } else {
+    for (int b = 0; b < 8; b++) cells.push_back((char)(uint8_t)(v >> (8 * b)));
+  }
+}
+
+static bool write_row_v2(const GenCfg &cfg, Xo256 &rng, std::string &buf) {
+  /* cells in id order; occasional NULL column for coverage */
+  uint8_t nn_ids[20], null_ids[20];
+  uint32_t nn = 0, nl = 0;
+  std::string cells;
+  uint16_t ends[20];
+  auto add_int = [&](uint8_t id, int64_t v) {
+    if ((rng.next() & 63) == 0) { null_ids[nl++] = id; return; }
+    nn_ids[nn] = id;
+    v2_int_cell(cells, v);
+    ends[nn] = (uint16_t)cells.size();
+    nn++;
+  };
+  switch (cfg.config_index) {
+    case 0:
+      for (uint8_t c = 1; c <= 4; c++) add_int(c, uniform_pm(rng, 1000000000));
+      break;
+    case 1:
+      for (uint8_t c = 1; c <= 16; c++) add_int(c, uniform_pm(rng, 1000000000));
+      break;
+    case 2: {
+      uint64_t k = cfg.param ? cfg.param : 64;
+      add_int(1, (int64_t)(rng.next() % k));
+      if ((rng.next() & 63) == 0) { null_ids[nl++] = 2; }
+      else {
+        nn_ids[nn] = 2;
+        prod::PDec d = prod::pdec_from_scaled_i128(uniform_pm(rng, 999999999999ll), 2);
+        uint8_t prec, fr, tmp[48];
+        prod::pdec_prec_and_frac(d, &prec, &fr);
+        size_t m = prod::pdec_encode(d, prec, fr, tmp);
+        cells.append((const char *)tmp, m);
+        ends[nn] = (uint16_t)cells.size();
+        nn++;
+      }
+      nn_ids[nn] = 3;
+      uint8_t bytes[24];
+      size_t blen = 8 + rng.next() % 17;
+      for (size_t i = 0; i < blen; i++) {
+        double u = (double)(rng.next() >> 11) * (1.0 / 9007199254740992.0);
+        bytes[i] = (uint8_t)('a' + (int)(26.0 * u * u * u));
+      }
+      cells.append((const char *)bytes, blen);
+      ends[nn] = (uint16_t)cells.size();
+      nn++;
+      break;
+    }
+    default:
+      return false;
+  }
+  buf.push_back((char)128);
+  buf.push_back(0);                                 /* flags: small, no cksum */
+  buf.push_back((char)(uint8_t)nn); buf.push_back(0);
+  buf.push_back((char)(uint8_t)nl); buf.push_back(0);
+  for (uint32_t i = 0; i < nn; i++) buf.push_back((char)nn_ids[i]);
+  for (uint32_t i = 0; i < nl; i++) buf.push_back((char)null_ids[i]);
+  for (uint32_t i = 0; i < nn; i++) {
+    buf.push_back((char)(uint8_t)ends[i]);
+    buf.push_back((char)(uint8_t)(ends[i] >> 8));
+  }
+  buf += cells;
+  return true;
+}
+
 }  // namespace
 
 static copr_status copr_gen_region_mvcc(const CoprGenSpec *spec, CoprGenOut *out);
@@ -345,7 +423,9 @@ copr_status copr_gen_region(const CoprGenSpec *spec, CoprGenOut *out) {
       enc_row_key(spec->table_id, handle, keys + i * 19);
       rng.seed(seed, (uint64_t)handle);
       row.clear();
-      if (!write_row(cfg, rng, row)) { ok = false; break; }
+      bool wok = spec->row_format == 2 ? write_row_v2(cfg, rng, row)
+                                       : write_row(cfg, rng, row);
+      if (!wok) { ok = false; break; }
       chunk_sizes[t][i - lo] = (uint32_t)row.size();
       buf += row;
     }

@@ -342,6 +342,163 @@ __device__ static inline bool next_cell(const uint8_t *vp, uint32_t vlen,
   return true;
 }
 
+
+struct AggColView { bool found, null, has_dec; int64_t iv, dsc; int32_t dfr; };
+
+/* ---------------- row v2 (codec/row/v2/row_slice.rs:76-168) ---------------- */
+struct V2Row {
+  bool big;
+  uint32_t nn, nl;
+  const uint8_t *ids, *nulls, *offs, *vals;
+  uint32_t vbytes;
+};
+
+__device__ static inline bool d_v2_parse(const uint8_t *p, uint32_t len, V2Row *r) {
+  if (len < 6 || p[0] != 128) return false;
+  uint8_t flags = p[1];
+  r->big = (flags & 1) != 0;
+  bool ck = (flags & 2) != 0;
+  r->nn = (uint32_t)p[2] | ((uint32_t)p[3] << 8);
+  r->nl = (uint32_t)p[4] | ((uint32_t)p[5] << 8);
+  uint32_t idw = r->big ? 4u : 1u, offw = r->big ? 4u : 2u;
+  uint32_t pos = 6;
+  if (pos + idw * (r->nn + r->nl) + offw * r->nn > len) return false;
+  r->ids = p + pos; pos += idw * r->nn;
+  r->nulls = p + pos; pos += idw * r->nl;
+  r->offs = p + pos; pos += offw * r->nn;
+  r->vals = p + pos;
+  r->vbytes = len - pos;
+  if (ck) {                        /* checksum trailer after the last value */
+    uint32_t vend = 0;
+    if (r->nn) {
+      const uint8_t *o = r->offs + offw * (r->nn - 1);
+      vend = r->big ? ((uint32_t)o[0] | ((uint32_t)o[1] << 8) |
+                       ((uint32_t)o[2] << 16) | ((uint32_t)o[3] << 24))
+                    : ((uint32_t)o[0] | ((uint32_t)o[1] << 8));
+    }
+    if (vend > r->vbytes) return false;
+    r->vbytes = vend;
+  }
+  return true;
+}
+
+/* search_in_non_null_ids / search_in_null_ids (row_slice.rs:125-196):
+ * 1 = found (start,end set), 0 = in null list, -1 = absent */
+__device__ static inline int d_v2_find(const V2Row &r, int64_t cid,
+                                       uint32_t *s, uint32_t *e) {
+  if (cid <= 0) return -1;
+  if (!r.big && cid > 255) return -1;
+  if (r.big && cid > 0xFFFFFFFFll) return -1;
+  uint32_t target = (uint32_t)cid;
+  uint32_t offw = r.big ? 4u : 2u;
+  int lo = 0, hi = (int)r.nn - 1;
+  while (lo <= hi) {
+    int mid = (lo + hi) >> 1;
+    uint32_t v = r.big ? ((uint32_t)r.ids[4 * mid] | ((uint32_t)r.ids[4 * mid + 1] << 8) |
+                          ((uint32_t)r.ids[4 * mid + 2] << 16) | ((uint32_t)r.ids[4 * mid + 3] << 24))
+                       : r.ids[mid];
+    if (v == target) {
+      const uint8_t *o = r.offs + offw * mid;
+      uint32_t end = r.big ? ((uint32_t)o[0] | ((uint32_t)o[1] << 8) |
+                              ((uint32_t)o[2] << 16) | ((uint32_t)o[3] << 24))
+                           : ((uint32_t)o[0] | ((uint32_t)o[1] << 8));
+      uint32_t st = 0;
+      if (mid > 0) {
+        const uint8_t *po = r.offs + offw * (mid - 1);
+        st = r.big ? ((uint32_t)po[0] | ((uint32_t)po[1] << 8) |
+                      ((uint32_t)po[2] << 16) | ((uint32_t)po[3] << 24))
+                   : ((uint32_t)po[0] | ((uint32_t)po[1] << 8));
+      }
+      if (st > end || end > r.vbytes) return -1;
+      *s = st; *e = end;
+      return 1;
+    }
+    if (v < target) lo = mid + 1; else hi = mid - 1;
+  }
+  lo = 0; hi = (int)r.nl - 1;
+  while (lo <= hi) {
+    int mid = (lo + hi) >> 1;
+    uint32_t v = r.big ? ((uint32_t)r.nulls[4 * mid] | ((uint32_t)r.nulls[4 * mid + 1] << 8) |
+                          ((uint32_t)r.nulls[4 * mid + 2] << 16) | ((uint32_t)r.nulls[4 * mid + 3] << 24))
+                       : r.nulls[mid];
+    if (v == target) return 0;
+    if (v < target) lo = mid + 1; else hi = mid - 1;
+  }
+  return -1;
+}
+
+/* v2 int cell: little-endian 1/2/4/8 bytes, signed sign-extends
+ * (compat_v1.rs:12-38) */
+__device__ static inline bool d_v2_int(const uint8_t *c, uint32_t n, bool uns,
+                                       int64_t *v) {
+  uint64_t u = 0;
+  switch (n) {
+    case 1: u = c[0]; if (!uns) u = (uint64_t)(int64_t)(int8_t)c[0]; break;
+    case 2: u = (uint32_t)c[0] | ((uint32_t)c[1] << 8);
+            if (!uns) u = (uint64_t)(int64_t)(int16_t)u; break;
+    case 4: u = (uint32_t)c[0] | ((uint32_t)c[1] << 8) |
+                ((uint32_t)c[2] << 16) | ((uint32_t)c[3] << 24);
+            if (!uns) u = (uint64_t)(int64_t)(int32_t)u; break;
+    case 8:
+      #pragma unroll
+      for (int b = 0; b < 8; b++) u |= (uint64_t)c[b] << (8 * b);
+      break;
+    default: return false;
+  }
+  *v = (int64_t)u;
+  return true;
+}
+
+template <int NAGGS, bool IS_HASH>
+__device__ static inline bool d_v2_collect(const ScanPlan &plan,
+                                           const uint8_t *vp, uint32_t vlen,
+                                           bool *filt_found, bool *filt_null,
+                                           int64_t *filt_v, bool *grp_found,
+                                           bool *grp_null, int64_t *grp_v,
+                                           AggColView (&cols)[NAGGS]) {
+  V2Row r;
+  if (!d_v2_parse(vp, vlen, &r)) return false;
+  if (plan.has_filter) {
+    uint32_t s, e;
+    int st = d_v2_find(r, plan.filter_col_id, &s, &e);
+    if (st >= 0) {
+      *filt_found = true;
+      if (st == 0) *filt_null = true;
+      else if (!d_v2_int(r.vals + s, e - s, plan.filter_col_unsigned, filt_v))
+        return false;
+    }
+  }
+  if (IS_HASH) {
+    uint32_t s, e;
+    int st = d_v2_find(r, plan.group_col_id, &s, &e);
+    if (st >= 0) {
+      *grp_found = true;
+      if (st == 0) *grp_null = true;
+      else if (!d_v2_int(r.vals + s, e - s, plan.group_col_unsigned, grp_v))
+        return false;
+    }
+  }
+  #pragma unroll
+  for (int a = 0; a < NAGGS; a++) {
+    const DevAggSpec &sp = plan.aggs[a];
+    if (sp.kind == DAGG_COUNT_ROWS) continue;
+    uint32_t s, e;
+    int st = d_v2_find(r, sp.col_id, &s, &e);
+    if (st < 0) continue;
+    cols[a].found = true;
+    if (st == 0) { cols[a].null = true; continue; }
+    if (sp.kind == DAGG_SUM_DEC) {
+      int64_t sc; int32_t fr;
+      if (!d_decimal_scaled(r.vals + s, e - s, &sc, &fr)) return false;
+      cols[a].has_dec = true; cols[a].dsc = sc; cols[a].dfr = fr;
+    } else if (sp.kind == DAGG_SUM_INT) {
+      if (!d_v2_int(r.vals + s, e - s, sp.col_unsigned, &cols[a].iv)) return false;
+    }
+    /* COUNT_COL: found/null is all that matters */
+  }
+  return true;
+}
+
 /* predicate eval (impl_compare.rs:66-160) */
 __device__ static inline int d_cmp_int(int64_t l, int64_t r, bool lu, bool ru) {
   if (lu && ru) { uint64_t a = (uint64_t)l, b = (uint64_t)r; return a < b ? -1 : a > b ? 1 : 0; }
@@ -395,7 +552,6 @@ __device__ static inline void atomic_add_i128(unsigned long long *lo,
  * and register-resident): the block software-pipelines tiles — while lanes
  * parse tile t from LDS, the loads for tile t+1 are already in flight into
  * registers (async-STAGE split, cdna_hip_programming G15). */
-struct AggColView { bool found, null, has_dec; int64_t iv, dsc; int32_t dfr; };
 
 struct TileInfo {
   uint64_t row0, row1, gbase;
@@ -524,6 +680,12 @@ k_scan_agg(ScanPlan plan,
       int found = 0;
 
       if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
+        if (vp[0] == 128) {          /* row v2 */
+        parse_ok = d_v2_collect<NAGGS, IS_HASH>(plan, vp, vlen,
+                                                &filt_found, &filt_null, &filt_v,
+                                                &grp_found, &grp_null, &grp_v,
+                                                cols);
+        } else {
         uint32_t pos = 0;
         while (pos < vlen) {
           int64_t cell_id;
@@ -562,6 +724,7 @@ k_scan_agg(ScanPlan plan,
           }
           if (found >= needed) break;
         }
+        }  /* v1/v2 */
       }
 
       if (!parse_ok) {
@@ -852,6 +1015,21 @@ k_scan_agg_pipe(ScanPlan plan,
         bool found = false, fnull = false, ok = true;
         int64_t fv = 0;
         if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
+          if (vp[0] == 128) {        /* row v2: direct column lookup */
+            V2Row r;
+            if (!d_v2_parse(vp, vlen, &r)) ok = false;
+            else {
+              uint32_t s, e;
+              int st = d_v2_find(r, FCID, &s, &e);
+              if (st >= 0) {
+                found = true;
+                if (st == 0) fnull = true;
+                else if (!d_v2_int(r.vals + s, e - s,
+                                   plan.filter_col_unsigned, &fv))
+                  ok = false;
+              }
+            }
+          } else {
           uintptr_t base = (uintptr_t)vp;
           uintptr_t wabs = ~(uintptr_t)0;   /* window invalid */
           uint64_t wlo = 0, whi = 0;
@@ -914,6 +1092,7 @@ k_scan_agg_pipe(ScanPlan plan,
               }
             }
           }
+          }  /* v1/v2 */
         }
         if (!ok) any_parse_err = true;
         else if (d_filter_keep(plan, found, fnull, fv)) cnt++;
@@ -935,6 +1114,12 @@ k_scan_agg_pipe(ScanPlan plan,
       int found = 0;
 
       if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
+        if (vp[0] == 128) {          /* row v2 */
+        parse_ok = d_v2_collect<NAGGS, IS_HASH>(plan, vp, vlen,
+                                                &filt_found, &filt_null, &filt_v,
+                                                &grp_found, &grp_null, &grp_v,
+                                                cols);
+        } else {
         uint32_t pos = 0;
         while (pos < vlen) {
           int64_t cell_id;
@@ -973,6 +1158,7 @@ k_scan_agg_pipe(ScanPlan plan,
           }
           if (found >= needed) break;
         }
+        }  /* v1/v2 */
       }
 
       if (!parse_ok) {
