@@ -442,6 +442,19 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
         }
         oa.in_kind = ds.kind;
         break;
+      case COPR_AGG_MAX: case COPR_AGG_MIN:
+        if (!ci || !et_int(ci->ft.tp))
+          return SET_ERR(COPR_ERR_UNSUPPORTED, "max/min native only over int");
+        ds.kind = ad.func == COPR_AGG_MAX ? DAGG_MAX_INT : DAGG_MIN_INT;
+        oa.in_kind = ds.kind;
+        break;
+      case COPR_AGG_BIT_AND: case COPR_AGG_BIT_OR: case COPR_AGG_BIT_XOR:
+        if (!ci || !et_int(ci->ft.tp))
+          return SET_ERR(COPR_ERR_UNSUPPORTED, "bit ops native only over int");
+        ds.kind = ad.func == COPR_AGG_BIT_AND ? DAGG_BIT_AND
+                  : ad.func == COPR_AGG_BIT_OR ? DAGG_BIT_OR : DAGG_BIT_XOR;
+        oa.in_kind = ds.kind;
+        break;
       default:
         return SET_ERR(COPR_ERR_UNSUPPORTED, "agg func not yet native");
     }
@@ -569,6 +582,25 @@ static void encode_agg_row(const HostPlan &pl, const SimpleAggAcc *accs,
     if (oa.func == COPR_AGG_COUNT) {
       bool uns = (pl.out_schema[oc].flag & COPR_FLAG_UNSIGNED) != 0;
       enc_datum_int(&(*cols_out)[oc], (int64_t)ac.cnt, uns);
+      oc++;
+    } else if (oa.func == COPR_AGG_MAX || oa.func == COPR_AGG_MIN) {
+      /* undo the fold transform (d_fold_xform): MIN folded max over ~biased */
+      bool uns = (pl.out_schema[oc].flag & COPR_FLAG_UNSIGNED) != 0;
+      if (ac.cnt == 0) enc_datum_null(&(*cols_out)[oc]);
+      else {
+        unsigned long long b = ac.sum_lo;
+        if (oa.func == COPR_AGG_MIN) b = ~b;
+        bool col_uns = pl.sp.aggs[oa.dev_idx].col_unsigned != 0;
+        if (!col_uns) b ^= 0x8000000000000000ull;
+        enc_datum_int(&(*cols_out)[oc], (int64_t)b, uns);
+      }
+      oc++;
+    } else if (oa.func == COPR_AGG_BIT_AND || oa.func == COPR_AGG_BIT_OR ||
+               oa.func == COPR_AGG_BIT_XOR) {
+      bool uns = (pl.out_schema[oc].flag & COPR_FLAG_UNSIGNED) != 0;
+      unsigned long long b = ac.sum_lo;
+      if (oa.func == COPR_AGG_BIT_AND) b = ~b;   /* OR of complements */
+      enc_datum_int(&(*cols_out)[oc], (int64_t)b, uns);
       oc++;
     } else {
       if (oa.func == COPR_AGG_AVG) {

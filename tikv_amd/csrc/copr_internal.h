@@ -28,6 +28,18 @@ enum {
   DAGG_SUM_INT,          /* i128 sum of int col (covers sum/avg(int) after
                             the Decimal rewrite: exact integer sum) */
   DAGG_SUM_DEC,          /* i128 sum of decimal col scaled to target frac */
+  /* fold aggregates over int columns; all ride sum_lo as a u64 fold with
+     an order-preserving transform so the identity is 0:
+       MAX: fold max over (v ^ SIGN-bias)      (cnt gates NULL)
+       MIN: fold max over ~(v ^ SIGN-bias)
+       BIT_AND: fold or over ~v  (result = ~acc; empty input -> ~0,
+                impl_bit_op.rs AND identity)
+       BIT_OR / BIT_XOR: fold or/xor over v */
+  DAGG_MAX_INT,
+  DAGG_MIN_INT,
+  DAGG_BIT_AND,
+  DAGG_BIT_OR,
+  DAGG_BIT_XOR,
 };
 
 struct DevAggSpec {

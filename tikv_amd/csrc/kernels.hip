@@ -491,7 +491,7 @@ __device__ static inline bool d_v2_collect(const ScanPlan &plan,
       int64_t sc; int32_t fr;
       if (!d_decimal_scaled(r.vals + s, e - s, &sc, &fr)) return false;
       cols[a].has_dec = true; cols[a].dsc = sc; cols[a].dfr = fr;
-    } else if (sp.kind == DAGG_SUM_INT) {
+    } else if (sp.kind == DAGG_SUM_INT || d_is_fold(sp.kind)) {
       if (!d_v2_int(r.vals + s, e - s, sp.col_unsigned, &cols[a].iv)) return false;
     }
     /* COUNT_COL: found/null is all that matters */
@@ -534,6 +534,26 @@ __device__ static inline bool d_filter_keep(const ScanPlan &plan, bool found,
                    d_cmp_int(v, plan.filter_const,
                              plan.filter_col_unsigned, plan.filter_const_unsigned));
 }
+
+
+/* fold-aggregate transforms (kinds DAGG_MAX_INT..DAGG_BIT_XOR): map the
+ * value into a u64 whose fold identity is 0 */
+__device__ static inline unsigned long long d_fold_xform(int32_t kind,
+                                                         int64_t v, bool uns) {
+  unsigned long long b = (unsigned long long)v;
+  if (!uns) b ^= 0x8000000000000000ull;       /* order-preserving bias */
+  switch (kind) {
+    case DAGG_MAX_INT: return b;
+    case DAGG_MIN_INT: return ~b;
+    case DAGG_BIT_AND: return ~(unsigned long long)v;
+    case DAGG_BIT_XOR: return (unsigned long long)v;
+    default:           return (unsigned long long)v;   /* BIT_OR */
+  }
+}
+__device__ static inline bool d_is_fold(int32_t kind) {
+  return kind >= DAGG_MAX_INT && kind <= DAGG_BIT_XOR;
+}
+__device__ static inline bool d_is_xor(int32_t kind) { return kind == DAGG_BIT_XOR; }
 
 /* 128-bit signed accumulate via two u64 atomics (carry trick) */
 __device__ static inline void atomic_add_i128(unsigned long long *lo,
@@ -792,7 +812,7 @@ k_scan_agg(ScanPlan plan,
             contribute = false;
           } else if (sp.kind == DAGG_COUNT_COL) {
             contribute = true;
-          } else if (sp.kind == DAGG_SUM_INT) {
+          } else if (sp.kind == DAGG_SUM_INT || d_is_fold(sp.kind)) {
             contribute = true; v = cols[a].iv;
           } else {  /* SUM_DEC */
             int d = sp.target_frac - cols[a].dfr;
@@ -812,6 +832,13 @@ k_scan_agg(ScanPlan plan,
               atomicAdd(&acc_base[a].cnt, 1ull);
               if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC)
                 atomic_add_i128(&acc_base[a].sum_lo, &acc_base[a].sum_hi, v);
+              else if (d_is_fold(sp.kind)) {
+                unsigned long long b = d_fold_xform(sp.kind, v, sp.col_unsigned);
+                if (d_is_xor(sp.kind)) atomicXor(&acc_base[a].sum_lo, b);
+                else if (sp.kind == DAGG_MAX_INT || sp.kind == DAGG_MIN_INT)
+                  atomicMax(&acc_base[a].sum_lo, b);
+                else atomicOr(&acc_base[a].sum_lo, b);
+              }
             }
           } else {
             l_cnt[a]++;
@@ -820,6 +847,12 @@ k_scan_agg(ScanPlan plan,
               unsigned long long nv = old + (unsigned long long)v;
               l_hi[a] += (nv < old ? 1 : 0) + (v < 0 ? -1 : 0);
               l_lo[a] = nv;
+            } else if (d_is_fold(sp.kind)) {
+              unsigned long long b = d_fold_xform(sp.kind, v, sp.col_unsigned);
+              if (d_is_xor(sp.kind)) l_lo[a] ^= b;
+              else if (sp.kind == DAGG_MAX_INT || sp.kind == DAGG_MIN_INT)
+                l_lo[a] = l_lo[a] > b ? l_lo[a] : b;
+              else l_lo[a] |= b;
             }
           }
         }
@@ -852,7 +885,17 @@ k_scan_agg(ScanPlan plan,
       #pragma unroll
       for (int a = 0; a < NAGGS; a++) {
         const SimpleAggAcc &la = laccs[s * NAGGS + a];
+        const int32_t kind = plan.aggs[a].kind;
         if (la.cnt) atomicAdd(&gacc[a].cnt, la.cnt);
+        if (d_is_fold(kind)) {
+          if (la.sum_lo) {
+            if (d_is_xor(kind)) atomicXor(&gacc[a].sum_lo, la.sum_lo);
+            else if (kind == DAGG_MAX_INT || kind == DAGG_MIN_INT)
+              atomicMax(&gacc[a].sum_lo, la.sum_lo);
+            else atomicOr(&gacc[a].sum_lo, la.sum_lo);
+          }
+          continue;
+        }
         if (la.sum_lo | la.sum_hi) {
           unsigned long long old = atomicAdd(&gacc[a].sum_lo, la.sum_lo);
           long long carry = (old + la.sum_lo < old) ? 1 : 0;
@@ -866,9 +909,31 @@ k_scan_agg(ScanPlan plan,
   if (!IS_HASH) {
     #pragma unroll
     for (int a = 0; a < NAGGS; a++) {
+      const int32_t kind = plan.aggs[a].kind;
       unsigned long long c = l_cnt[a];
       unsigned long long lo = l_lo[a];
       long long hi = l_hi[a];
+      if (d_is_fold(kind)) {
+        for (int off = 32; off > 0; off >>= 1) {
+          c += (unsigned long long)__shfl_down((long long)c, off, 64);
+          unsigned long long plo =
+              (unsigned long long)__shfl_down((long long)lo, off, 64);
+          if (d_is_xor(kind)) lo ^= plo;
+          else if (kind == DAGG_MAX_INT || kind == DAGG_MIN_INT)
+            lo = lo > plo ? lo : plo;
+          else lo |= plo;
+        }
+        if ((threadIdx.x & 63u) == 0) {
+          if (c) atomicAdd(&simple_acc[a].cnt, c);
+          if (lo) {
+            if (d_is_xor(kind)) atomicXor(&simple_acc[a].sum_lo, lo);
+            else if (kind == DAGG_MAX_INT || kind == DAGG_MIN_INT)
+              atomicMax(&simple_acc[a].sum_lo, lo);
+            else atomicOr(&simple_acc[a].sum_lo, lo);
+          }
+        }
+        continue;
+      }
       for (int off = 32; off > 0; off >>= 1) {
         c += (unsigned long long)__shfl_down((long long)c, off, 64);
         unsigned long long plo = (unsigned long long)__shfl_down((long long)lo, off, 64);
@@ -1207,7 +1272,7 @@ k_scan_agg_pipe(ScanPlan plan,
             contribute = false;
           } else if (sp.kind == DAGG_COUNT_COL) {
             contribute = true;
-          } else if (sp.kind == DAGG_SUM_INT) {
+          } else if (sp.kind == DAGG_SUM_INT || d_is_fold(sp.kind)) {
             contribute = true; v = cols[a].iv;
           } else {  /* SUM_DEC */
             int d = sp.target_frac - cols[a].dfr;
@@ -1227,6 +1292,13 @@ k_scan_agg_pipe(ScanPlan plan,
               atomicAdd(&acc_base[a].cnt, 1ull);
               if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC)
                 atomic_add_i128(&acc_base[a].sum_lo, &acc_base[a].sum_hi, v);
+              else if (d_is_fold(sp.kind)) {
+                unsigned long long b = d_fold_xform(sp.kind, v, sp.col_unsigned);
+                if (d_is_xor(sp.kind)) atomicXor(&acc_base[a].sum_lo, b);
+                else if (sp.kind == DAGG_MAX_INT || sp.kind == DAGG_MIN_INT)
+                  atomicMax(&acc_base[a].sum_lo, b);
+                else atomicOr(&acc_base[a].sum_lo, b);
+              }
             }
           } else {
             l_cnt[a]++;
@@ -1235,6 +1307,12 @@ k_scan_agg_pipe(ScanPlan plan,
               unsigned long long nv = old + (unsigned long long)v;
               l_hi[a] += (nv < old ? 1 : 0) + (v < 0 ? -1 : 0);
               l_lo[a] = nv;
+            } else if (d_is_fold(sp.kind)) {
+              unsigned long long b = d_fold_xform(sp.kind, v, sp.col_unsigned);
+              if (d_is_xor(sp.kind)) l_lo[a] ^= b;
+              else if (sp.kind == DAGG_MAX_INT || sp.kind == DAGG_MIN_INT)
+                l_lo[a] = l_lo[a] > b ? l_lo[a] : b;
+              else l_lo[a] |= b;
             }
           }
         }
@@ -1248,9 +1326,31 @@ k_scan_agg_pipe(ScanPlan plan,
   if (!IS_HASH) {
     #pragma unroll
     for (int a = 0; a < NAGGS; a++) {
+      const int32_t kind = plan.aggs[a].kind;
       unsigned long long c = l_cnt[a];
       unsigned long long lo = l_lo[a];
       long long hi = l_hi[a];
+      if (d_is_fold(kind)) {
+        for (int off = 32; off > 0; off >>= 1) {
+          c += (unsigned long long)__shfl_down((long long)c, off, 64);
+          unsigned long long plo =
+              (unsigned long long)__shfl_down((long long)lo, off, 64);
+          if (d_is_xor(kind)) lo ^= plo;
+          else if (kind == DAGG_MAX_INT || kind == DAGG_MIN_INT)
+            lo = lo > plo ? lo : plo;
+          else lo |= plo;
+        }
+        if ((threadIdx.x & 63u) == 0) {
+          if (c) atomicAdd(&simple_acc[a].cnt, c);
+          if (lo) {
+            if (d_is_xor(kind)) atomicXor(&simple_acc[a].sum_lo, lo);
+            else if (kind == DAGG_MAX_INT || kind == DAGG_MIN_INT)
+              atomicMax(&simple_acc[a].sum_lo, lo);
+            else atomicOr(&simple_acc[a].sum_lo, lo);
+          }
+        }
+        continue;
+      }
       for (int off = 32; off > 0; off >>= 1) {
         c += (unsigned long long)__shfl_down((long long)c, off, 64);
         unsigned long long plo = (unsigned long long)__shfl_down((long long)lo, off, 64);

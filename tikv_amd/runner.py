@@ -197,6 +197,18 @@ def avg_col(offset, decimal=-1):
     return (F.AGG_AVG, Expr().col(offset), field_type(F.TP_NEWDECIMAL, decimal=decimal))
 
 
+def max_col(offset, tp=F.TP_LONGLONG, flag=0):
+    return (F.AGG_MAX, Expr().col(offset), field_type(tp, flag))
+
+
+def min_col(offset, tp=F.TP_LONGLONG, flag=0):
+    return (F.AGG_MIN, Expr().col(offset), field_type(tp, flag))
+
+
+def bit_op(func, offset):
+    return (func, Expr().col(offset), field_type(F.TP_LONGLONG, F.FLAG_UNSIGNED))
+
+
 def cmp_col_const(offset, sig, const, unsigned_const=False):
     return Expr().col(offset).const_int(const, unsigned_const).func(sig, 2)
 
