@@ -343,6 +343,26 @@ __device__ static inline bool next_cell(const uint8_t *vp, uint32_t vlen,
 }
 
 
+/* fold-aggregate transforms (kinds DAGG_MAX_INT..DAGG_BIT_XOR): map the
+ * value into a u64 whose fold identity is 0 */
+__device__ static inline unsigned long long d_fold_xform(int32_t kind,
+                                                         int64_t v, bool uns) {
+  unsigned long long b = (unsigned long long)v;
+  if (!uns) b ^= 0x8000000000000000ull;       /* order-preserving bias */
+  switch (kind) {
+    case DAGG_MAX_INT: return b;
+    case DAGG_MIN_INT: return ~b;
+    case DAGG_BIT_AND: return ~(unsigned long long)v;
+    case DAGG_BIT_XOR: return (unsigned long long)v;
+    default:           return (unsigned long long)v;   /* BIT_OR */
+  }
+}
+__device__ static inline bool d_is_fold(int32_t kind) {
+  return kind >= DAGG_MAX_INT && kind <= DAGG_BIT_XOR;
+}
+__device__ static inline bool d_is_xor(int32_t kind) { return kind == DAGG_BIT_XOR; }
+
+
 struct AggColView { bool found, null, has_dec; int64_t iv, dsc; int32_t dfr; };
 
 /* ---------------- row v2 (codec/row/v2/row_slice.rs:76-168) ---------------- */
@@ -535,25 +555,6 @@ __device__ static inline bool d_filter_keep(const ScanPlan &plan, bool found,
                              plan.filter_col_unsigned, plan.filter_const_unsigned));
 }
 
-
-/* fold-aggregate transforms (kinds DAGG_MAX_INT..DAGG_BIT_XOR): map the
- * value into a u64 whose fold identity is 0 */
-__device__ static inline unsigned long long d_fold_xform(int32_t kind,
-                                                         int64_t v, bool uns) {
-  unsigned long long b = (unsigned long long)v;
-  if (!uns) b ^= 0x8000000000000000ull;       /* order-preserving bias */
-  switch (kind) {
-    case DAGG_MAX_INT: return b;
-    case DAGG_MIN_INT: return ~b;
-    case DAGG_BIT_AND: return ~(unsigned long long)v;
-    case DAGG_BIT_XOR: return (unsigned long long)v;
-    default:           return (unsigned long long)v;   /* BIT_OR */
-  }
-}
-__device__ static inline bool d_is_fold(int32_t kind) {
-  return kind >= DAGG_MAX_INT && kind <= DAGG_BIT_XOR;
-}
-__device__ static inline bool d_is_xor(int32_t kind) { return kind == DAGG_BIT_XOR; }
 
 /* 128-bit signed accumulate via two u64 atomics (carry trick) */
 __device__ static inline void atomic_add_i128(unsigned long long *lo,
