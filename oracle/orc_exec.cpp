@@ -392,8 +392,53 @@ struct ScanState {
   std::vector<CoprColumnInfo> cols;
   std::vector<int> handle_indices;              /* pk_handle columns */
   std::unordered_map<int64_t, int> col_id_index;
+  bool index_scan = false;                      /* BatchIndexScanExecutor */
   uint64_t next_kv = 0;                         /* scan cursor */
 };
+
+/* IndexScan process_kv_pair (index_scan_executor.rs:373-560, old-collation
+ * local path): key = 't'||tid||'_i'||index_id|| comparable datums ||
+ * [int-handle datum]; columns are POSITIONAL raw comparable datum slices
+ * (extract_columns_from_datum_format :504-517); the int handle comes from
+ * the key tail for non-unique indexes (decode_int_handle_from_key :460-481)
+ * or the 8-byte BE value for unique ones (:417-422). Values longer than 9
+ * bytes (new encoding) are out of scope here. */
+static bool scan_process_kv_index(ScanState &st, const uint8_t *key, size_t klen,
+                                  const uint8_t *val, size_t vlen, Batch *batch) {
+  if (klen < 19 || key[0] != 't' || key[9] != '_' || key[10] != 'i')
+    FAIL("not an index key");
+  if (vlen > 9) FAIL("new index value encoding not supported");
+  const uint8_t *p = key + 19;
+  size_t rem = klen - 19;
+  size_t ncols = st.cols.size();
+  size_t n_idx_cols = 0;
+  for (auto &c : st.cols) if (!c.pk_handle) n_idx_cols++;
+  size_t ci = 0;
+  for (size_t i = 0; i < ncols && ci < n_idx_cols; i++) {
+    if (st.cols[i].pk_handle) continue;
+    size_t dlen;
+    if (!split_datum(p, rem, &dlen)) FAIL("bad index key datum");
+    batch->cols[i].raw_vals.emplace_back((const char *)p, dlen);
+    p += dlen; rem -= dlen;
+    ci++;
+  }
+  for (size_t i = 0; i < ncols; i++) {
+    if (!st.cols[i].pk_handle) continue;
+    int64_t handle;
+    if (rem > 0) {
+      if (rem < 9 || (p[0] != 3 && p[0] != 4)) FAIL("bad index handle datum");
+      handle = p[0] == 3 ? decode_comparable_i64(p + 1)
+                         : (int64_t)decode_comparable_u64(p + 1);
+    } else {
+      if (vlen < 8) FAIL("unique index value too short for handle");
+      uint64_t u = 0;
+      for (int b = 0; b < 8; b++) u = (u << 8) | val[b];   /* plain BE u64 */
+      handle = (int64_t)u;
+    }
+    batch->cols[i].dec.push_int(handle);
+  }
+  return true;
+}
 
 /* process_kv_pair (+_v1/_v2): table_scan_executor.rs:375-485,209-291 */
 static bool scan_process_kv(ScanState &st, const uint8_t *key, size_t klen,
@@ -732,7 +777,9 @@ static bool build_pipeline(const CoprDagRequest *req, Pipeline *pl) {
   pl->req = req;
   if (req->n_executors == 0) FAIL("empty executor list");
   const CoprExecutor &first = req->executors[0];
-  if (first.kind != COPR_EXEC_TABLE_SCAN) FAIL("first executor must be a scan");
+  if (first.kind != COPR_EXEC_TABLE_SCAN && first.kind != COPR_EXEC_INDEX_SCAN)
+    FAIL("first executor must be a scan");
+  pl->scan.index_scan = first.kind == COPR_EXEC_INDEX_SCAN;
   pl->scan_exec = &first;
   for (uint32_t i = 0; i < first.n_columns; i++) {
     const CoprColumnInfo &ci = first.columns[i];
@@ -846,11 +893,16 @@ static bool run_pipeline(const CoprDagRequest *req,
     }
     for (uint64_t k = 0; k < n; k++) {
       uint64_t idx = cursor + k;
-      if (!scan_process_kv(pl.scan, keys + key_offs[idx],
-                           (size_t)(key_offs[idx + 1] - key_offs[idx]),
-                           vals + val_offs[idx],
-                           (size_t)(val_offs[idx + 1] - val_offs[idx]), &batch))
-        return false;
+      bool okk = pl.scan.index_scan
+          ? scan_process_kv_index(pl.scan, keys + key_offs[idx],
+                                  (size_t)(key_offs[idx + 1] - key_offs[idx]),
+                                  vals + val_offs[idx],
+                                  (size_t)(val_offs[idx + 1] - val_offs[idx]), &batch)
+          : scan_process_kv(pl.scan, keys + key_offs[idx],
+                            (size_t)(key_offs[idx + 1] - key_offs[idx]),
+                            vals + val_offs[idx],
+                            (size_t)(val_offs[idx + 1] - val_offs[idx]), &batch);
+      if (!okk) return false;
     }
     cursor += n;
     batch.logical_rows.resize(n);

@@ -354,3 +354,57 @@ def test_oracle_row_v2_generated():
         assert nonnull < 3000        # with 3000 rows some NULLs occur w.h.p.
     finally:
         g.close()
+
+
+def idx_key(table_id, index_id, datums, handle=None):
+    def cmp64(x):
+        return ((x ^ (1 << 63)) & (2**64 - 1)).to_bytes(8, "big")
+    k = b"t" + cmp64(table_id) + b"_i" + cmp64(index_id)
+    for d in datums:
+        k += d
+    if handle is not None:
+        k += bytes([3]) + cmp64(handle)
+    return k
+
+
+def test_oracle_index_scan():
+    """non-unique index: positional comparable datums + handle from key."""
+    cols = [tikv_amd.Col(1), tikv_amd.Col(2), tikv_amd.Col(-1, pk_handle=True)]
+    rows = []
+    for h, (a, b) in enumerate([(5, 100), (7, -3), (5, 42)]):
+        rows.append((idx_key(1, 1, [d_int(a), d_int(b)], handle=h), b"0"))
+    req = tikv_amd.DagSelect(cols, index=True).build()
+    data, n = run_oracle(req, make_region(rows))
+    assert n == 3
+    expect = (d_int(5) + d_int(100) + d_int(0) +
+              d_int(7) + d_int(-3) + d_int(1) +
+              d_int(5) + d_int(42) + d_int(2))
+    assert data == expect
+    # unique index: handle from the 8-byte BE value
+    rows_u = [(idx_key(1, 1, [d_int(9), d_int(1)]),
+               (12345).to_bytes(8, "big"))]
+    data, n = run_oracle(req, make_region(rows_u))
+    assert n == 1
+    assert data == d_int(9) + d_int(1) + d_int(12345)
+
+
+def test_oracle_index_agg():
+    cols = [tikv_amd.Col(1), tikv_amd.Col(2), tikv_amd.Col(-1, pk_handle=True)]
+    rows = []
+    for h, (a, b) in enumerate([(5, 100), (7, -3), (5, 42), (7, 1)]):
+        rows.append((idx_key(1, 1, [d_int(a), d_int(b)], handle=h), b"0"))
+    req = (tikv_amd.DagSelect(cols, index=True)
+           .where(tikv_amd.cmp_col_const(0, F.SIG_GT_INT, 0))
+           .hash_agg([tikv_amd.count_star(), tikv_amd.sum_col(1)],
+                     tikv_amd.Expr().col(0)).build())
+    data, n = run_oracle(req, make_region(rows))
+    assert n == 2
+    lib = _orc().load_lib()
+
+    def dec(v):
+        out = C.create_string_buffer(64)
+        nd = lib.orc_test_dec_from_i64_encode(v, out)
+        return bytes([6]) + out.raw[:nd]
+
+    got = set(split_datum_rows(data, 3))
+    assert got == {d_int(2) + dec(142) + d_int(5), d_int(2) + dec(-2) + d_int(7)}

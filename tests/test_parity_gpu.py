@@ -325,3 +325,37 @@ def test_max_min_bit_aggregates_parity(engine):
         assert sorted(split_rows(o, 4)) == sorted(split_rows(g, 4))
     finally:
         gen.close()
+
+
+def test_cfg5_index_scan_parity(engine):
+    """cfg5 shape: IndexScan + Selection + HashAgg over index-key datums."""
+    gen = tikv_amd.GenRegion(config_index=4, n_rows=150000, table_id=1)
+    try:
+        cols = [tikv_amd.Col(1), tikv_amd.Col(2),
+                tikv_amd.Col(-1, pk_handle=True)]
+        sel = tikv_amd.cmp_col_const(1, F.SIG_GE_INT, 0)   # amount >= 0
+        req = (tikv_amd.DagSelect(cols, index=True).where(sel)
+               .hash_agg([tikv_amd.count_star(), tikv_amd.sum_col(1),
+                          tikv_amd.max_col(2)],
+                         tikv_amd.Expr().col(0)).build())
+        (o, on), (g, gn) = run_both(req, gen, engine)
+        assert on == gn
+        assert sorted(split_rows(o, 4)) == sorted(split_rows(g, 4))
+    finally:
+        gen.close()
+
+
+def test_cfg5_index_simple_agg_parity(engine):
+    gen = tikv_amd.GenRegion(config_index=4, n_rows=100000, table_id=1)
+    try:
+        cols = [tikv_amd.Col(1), tikv_amd.Col(2),
+                tikv_amd.Col(-1, pk_handle=True)]
+        sel = tikv_amd.cmp_col_const(0, F.SIG_LT_INT, 300)  # ~10% of [0,3000)
+        req = (tikv_amd.DagSelect(cols, index=True).where(sel)
+               .simple_agg([tikv_amd.count_star(), tikv_amd.sum_col(1),
+                            tikv_amd.min_col(2), tikv_amd.max_col(0)]).build())
+        (o, on), (g, gn) = run_both(req, gen, engine)
+        assert on == gn == 1
+        assert o == g
+    finally:
+        gen.close()

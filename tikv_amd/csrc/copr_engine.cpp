@@ -88,12 +88,15 @@ extern "C" copr_status copr_region_create(copr_engine *eng,
   r->dev.n_kv = n_kv;
   r->dev.key_bytes = key_offs[n_kv];
   r->dev.val_bytes = val_offs[n_kv];
-  uint32_t max_row = 0;
+  uint32_t max_row = 0, max_key = 0;
   for (uint64_t i = 0; i < n_kv; i++) {
     uint64_t l = val_offs[i + 1] - val_offs[i];
     if (l > max_row) max_row = (uint32_t)l;
+    uint64_t kl = key_offs[i + 1] - key_offs[i];
+    if (kl > max_key) max_key = (uint32_t)kl;
   }
   r->dev.max_row_bytes = max_row;
+  r->dev.max_key_bytes = max_key;
   r->h_key_offs.assign(key_offs, key_offs + n_kv + 1);
   r->h_val_offs.assign(val_offs, val_offs + n_kv + 1);
 
@@ -300,7 +303,8 @@ static copr_status match_filter(const CoprExpr &cond, HostPlan &pl, ScanPlan *sp
   size_t off = (size_t)colref->i64_val;
   if (off >= pl.cols.size()) return COPR_ERR_INVALID_REQUEST;
   const CoprColumnInfo &ci = pl.cols[off];
-  if (!et_int(ci.ft.tp) || ci.pk_handle) return COPR_ERR_UNSUPPORTED;
+  if (!et_int(ci.ft.tp)) return COPR_ERR_UNSUPPORTED;
+  if (ci.pk_handle && !pl.sp.index_mode) return COPR_ERR_UNSUPPORTED;
   pl.filter_col_offset = (int)off;
   if (swapped) {
     /* const OP col  ==  col flip(OP) const */
@@ -339,11 +343,22 @@ static copr_status match_filter(const CoprExpr &cond, HostPlan &pl, ScanPlan *sp
 static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
   if (req->n_executors == 0) return SET_ERR(COPR_ERR_INVALID_REQUEST, "empty executors");
   const CoprExecutor &scan = req->executors[0];
-  if (scan.kind != COPR_EXEC_TABLE_SCAN)
-    return SET_ERR(COPR_ERR_UNSUPPORTED, "first executor must be TABLE_SCAN");
+  if (scan.kind != COPR_EXEC_TABLE_SCAN && scan.kind != COPR_EXEC_INDEX_SCAN)
+    return SET_ERR(COPR_ERR_UNSUPPORTED, "first executor must be a scan");
   pl->cols.assign(scan.columns, scan.columns + scan.n_columns);
   ScanPlan &sp = pl->sp;
   memset(&sp, 0, sizeof(sp));
+  if (scan.kind == COPR_EXEC_INDEX_SCAN) {
+    /* positional plan: rewrite each column's "id" to its datum position in
+       the index key; the handle column sits at position n_idx_cols */
+    sp.index_mode = 1;
+    int pos = 0;
+    for (auto &ci : pl->cols)
+      if (!ci.pk_handle) ci.column_id = pos++;
+    sp.index_n_cols = pos;
+    for (auto &ci : pl->cols)
+      if (ci.pk_handle) ci.column_id = pos;
+  }
 
   const CoprExecutor *agg = nullptr;
   for (uint32_t e = 1; e < req->n_executors; e++) {
@@ -371,6 +386,8 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
   }
 
   if (!agg) {
+    if (sp.index_mode)
+      return SET_ERR(COPR_ERR_UNSUPPORTED, "index scan project not yet native");
     sp.mode = 0;
     sp.n_out = (int32_t)pl->cols.size();
     if (sp.n_out > COPR_MAX_OUT_COLS)
@@ -407,7 +424,7 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
       size_t off = (size_t)an.i64_val;
       if (off >= pl->cols.size()) return SET_ERR(COPR_ERR_INVALID_REQUEST, "bad col offset");
       ci = &pl->cols[off];
-      if (ci->pk_handle)
+      if (ci->pk_handle && !sp.index_mode)
         return SET_ERR(COPR_ERR_UNSUPPORTED, "agg over handle column not yet supported");
     }
     switch (ad.func) {
@@ -482,7 +499,7 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
     size_t off = (size_t)ge.nodes[0].i64_val;
     if (off >= pl->cols.size()) return SET_ERR(COPR_ERR_INVALID_REQUEST, "bad group offset");
     const CoprColumnInfo &ci = pl->cols[off];
-    if (!et_int(ci.ft.tp) || ci.pk_handle)
+    if (!et_int(ci.ft.tp) || (ci.pk_handle && !sp.index_mode))
       return SET_ERR(COPR_ERR_UNSUPPORTED, "group-by type not yet native");
     sp.group_col_id = ci.column_id;
     sp.group_col_unsigned = (ci.ft.flag & COPR_FLAG_UNSIGNED) ? 1 : 0;
@@ -499,7 +516,7 @@ static void pick_tiling(const DevRegion &rgn, ScanPlan *sp, bool force_nopipe = 
   sp->diag_stage_only = getenv("COPR_DIAG_STAGE_ONLY") ? 1 : 0;
   sp->use_pipe = 0;
   sp->glds_nt = getenv("COPR_GLDS_NT") ? 1 : 0;
-  uint32_t per_row = rgn.max_row_bytes + 1;
+  uint32_t per_row = (sp->index_mode ? rgn.max_key_bytes : rgn.max_row_bytes) + 1;
 
   if (!force_nopipe && !getenv("COPR_NO_PIPE")) {
     /* glds double-buffer pipeline: 2 x (offs slab + vals slab), 1 KiB

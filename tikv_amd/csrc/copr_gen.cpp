@@ -158,6 +158,10 @@ static bool write_row(const GenCfg &cfg, Xo256 &rng, std::string &buf) {
       put_cell_bytes(buf, 3, bytes, blen);
       return true;
     }
+    case 4: {                          /* cfg5: handled by the key writer */
+      buf.push_back('0');              /* old-encoding non-unique value */
+      return true;
+    }
     case 3: {                          /* cfg4: opaque value 100..200 B */
       size_t blen = 100 + rng.next() % 101;
       uint8_t chunk[8];
@@ -392,9 +396,10 @@ copr_status copr_gen_region(const CoprGenSpec *spec, CoprGenOut *out) {
 
   uint64_t *key_offs = (uint64_t *)malloc((n + 1) * sizeof(uint64_t));
   uint64_t *val_offs = (uint64_t *)malloc((n + 1) * sizeof(uint64_t));
-  uint8_t *keys = (uint8_t *)malloc(n * 19 + 1);
+  uint64_t klen = spec->config_index == 4 ? 46 : 19;
+  uint8_t *keys = (uint8_t *)malloc(n * klen + 1);
   if (!key_offs || !val_offs || !keys) return COPR_ERR_OOM;
-  for (uint64_t i = 0; i <= n; i++) key_offs[i] = i * 19;
+  for (uint64_t i = 0; i <= n; i++) key_offs[i] = i * klen;
 
   int T = 1;
 #ifdef _OPENMP
@@ -420,8 +425,27 @@ copr_status copr_gen_region(const CoprGenSpec *spec, CoprGenOut *out) {
     std::string row;
     for (uint64_t i = lo; i < hi; i++) {
       int64_t handle = (int64_t)(spec->first_handle + i);
-      enc_row_key(spec->table_id, handle, keys + i * 19);
       rng.seed(seed, (uint64_t)handle);
+      if (spec->config_index == 4) {
+        /* cfg5: TPCC-order-line-shaped secondary index entry (index_id 1,
+           non-unique): 't'||tid||'_i'||BE(1^S)|| INT datums (ol_w_d i64 in
+           [0,3000), amount i64 +-1e6) || INT handle
+           (index_scan_executor.rs key format; encode_index_seek_key,
+           table.rs:229-235) */
+        uint8_t *k = keys + i * 46;
+        k[0] = 't';
+        enc_cmp_i64(k + 1, spec->table_id);
+        k[9] = '_'; k[10] = 'i';
+        enc_cmp_i64(k + 11, 1);
+        k[19] = 3;  /* INT_FLAG */
+        enc_cmp_i64(k + 20, (int64_t)(rng.next() % 3000));
+        k[28] = 3;
+        enc_cmp_i64(k + 29, uniform_pm(rng, 1000000));
+        k[37] = 3;
+        enc_cmp_i64(k + 38, handle);
+      } else {
+        enc_row_key(spec->table_id, handle, keys + i * 19);
+      }
       row.clear();
       bool wok = spec->row_format == 2 ? write_row_v2(cfg, rng, row)
                                        : write_row(cfg, rng, row);

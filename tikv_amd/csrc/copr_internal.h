@@ -16,6 +16,7 @@ struct DevRegion {
   uint64_t n_kv = 0;
   uint64_t key_bytes = 0, val_bytes = 0;
   uint32_t max_row_bytes = 0;      /* max value size over all rows */
+  uint32_t max_key_bytes = 0;      /* max key size (index scans parse keys) */
 };
 
 /* compare kinds (order matches oracle CmpKind) */
@@ -67,6 +68,12 @@ struct ScanPlan {
   int64_t filter_missing_val;
 
   int32_t mode;                  /* 0 project, 1 simple agg, 2 hash agg */
+  /* index scan (BatchIndexScanExecutor): the parsed stream is the KEY
+     stream; columns are POSITIONAL comparable datums after the 19-byte
+     prefix, with the int handle as the trailing datum (position
+     index_n_cols). col ids in this plan are positions. */
+  int32_t index_mode;
+  int32_t index_n_cols;
   int32_t n_aggs;
   DevAggSpec aggs[COPR_MAX_AGGS];
 

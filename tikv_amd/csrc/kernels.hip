@@ -519,6 +519,64 @@ __device__ static inline bool d_v2_collect(const ScanPlan &plan,
   return true;
 }
 
+
+/* IndexScan positional walk (index_scan_executor.rs:504-560 old-collation
+ * local path): vp = the whole index KEY; comparable datums start at byte 19;
+ * cell "ids" are positions 0..n-1, with the trailing int-handle datum at
+ * position index_n_cols. */
+template <int NAGGS, bool IS_HASH>
+__device__ static inline bool d_index_collect(const ScanPlan &plan,
+                                              const uint8_t *vp, uint32_t vlen,
+                                              bool *filt_found, bool *filt_null,
+                                              int64_t *filt_v, bool *grp_found,
+                                              bool *grp_null, int64_t *grp_v,
+                                              AggColView (&cols)[NAGGS],
+                                              int64_t *handle_out,
+                                              bool *handle_found) {
+  if (vlen < 19 || vp[0] != 't' || vp[9] != '_' || vp[10] != 'i') return false;
+  uint32_t pos = 19;
+  int64_t ci = 0;
+  *handle_found = false;
+  while (pos < vlen) {
+    CellView cell;
+    d_parse_datum(vp + pos, vlen - pos, &cell);
+    if (cell.len == 0) return false;
+    const int64_t cell_id = ci;
+    if (cell_id == (int64_t)plan.index_n_cols) {   /* trailing handle datum */
+      if (!cell.has_int) return false;
+      *handle_out = cell.ival;
+      *handle_found = true;
+    }
+    if (plan.has_filter && !*filt_found && cell_id == plan.filter_col_id) {
+      *filt_found = true;
+      if (cell.is_null) *filt_null = true;
+      else if (cell.has_int) *filt_v = cell.ival;
+      else return false;
+    }
+    if (IS_HASH && !*grp_found && cell_id == plan.group_col_id) {
+      *grp_found = true;
+      if (cell.is_null) *grp_null = true;
+      else if (cell.has_int) *grp_v = cell.ival;
+      else return false;
+    }
+    #pragma unroll
+    for (int a = 0; a < NAGGS; a++) {
+      if (plan.aggs[a].kind == DAGG_COUNT_ROWS || cols[a].found) continue;
+      if (cell_id == plan.aggs[a].col_id) {
+        cols[a].found = true;
+        cols[a].null = cell.is_null;
+        cols[a].iv = cell.ival;
+        cols[a].has_dec = cell.has_dec;
+        cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
+        if (!cell.is_null && !cell.has_int && !cell.has_dec) return false;
+      }
+    }
+    pos += cell.len;
+    ci++;
+  }
+  return true;
+}
+
 /* predicate eval (impl_compare.rs:66-160) */
 __device__ static inline int d_cmp_int(int64_t l, int64_t r, bool lu, bool ru) {
   if (lu && ru) { uint64_t a = (uint64_t)l, b = (uint64_t)r; return a < b ? -1 : a > b ? 1 : 0; }
@@ -700,7 +758,14 @@ k_scan_agg(ScanPlan plan,
       for (int a = 0; a < NAGGS; a++) cols[a] = {false, false, false, 0, 0, 0};
       int found = 0;
 
-      if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
+      if (plan.index_mode) {
+        int64_t hval = 0; bool hfound = false;
+        parse_ok = d_index_collect<NAGGS, IS_HASH>(plan, vp, vlen,
+                                                   &filt_found, &filt_null,
+                                                   &filt_v, &grp_found,
+                                                   &grp_null, &grp_v, cols,
+                                                   &hval, &hfound);
+      } else if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
         if (vp[0] == 128) {          /* row v2 */
         parse_ok = d_v2_collect<NAGGS, IS_HASH>(plan, vp, vlen,
                                                 &filt_found, &filt_null, &filt_v,
@@ -1179,7 +1244,14 @@ k_scan_agg_pipe(ScanPlan plan,
       for (int a = 0; a < NAGGS; a++) cols[a] = {false, false, false, 0, 0, 0};
       int found = 0;
 
-      if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
+      if (plan.index_mode) {
+        int64_t hval = 0; bool hfound = false;
+        parse_ok = d_index_collect<NAGGS, IS_HASH>(plan, vp, vlen,
+                                                   &filt_found, &filt_null,
+                                                   &filt_v, &grp_found,
+                                                   &grp_null, &grp_v, cols,
+                                                   &hval, &hfound);
+      } else if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
         if (vp[0] == 128) {          /* row v2 */
         parse_ok = d_v2_collect<NAGGS, IS_HASH>(plan, vp, vlen,
                                                 &filt_found, &filt_null, &filt_v,
@@ -1965,8 +2037,8 @@ template <bool IS_HASH>
 static int launch_agg_pipe(const ScanPlan &plan, const DevRegion &rgn,
                            SimpleAggAcc *d_simple, HashAggTable ht,
                            hipStream_t s, uint32_t grid) {
-  if (!IS_HASH && plan.n_aggs == 1 && plan.aggs[0].kind == DAGG_COUNT_ROWS &&
-      plan.has_filter) {
+  if (!IS_HASH && !plan.index_mode && plan.n_aggs == 1 &&
+      plan.aggs[0].kind == DAGG_COUNT_ROWS && plan.has_filter) {
     if (getenv("COPR_DIRECT")) {
       uint64_t n_blk = (rgn.n_kv + THREADS - 1) / THREADS;
       uint32_t dgrid = (uint32_t)(n_blk < 8192 ? n_blk : 8192);
@@ -2014,10 +2086,18 @@ int dev_scan_launch(const ScanPlan &plan, const DevRegion &rgn,
   uint32_t grid = (uint32_t)(n_tiles < 4096 ? n_tiles : 4096);
   if (grid == 0) grid = 1;
   hipStream_t s = (hipStream_t)stream;
+  DevRegion r2 = rgn;
+  if (plan.index_mode) {
+    /* index scans parse the KEY stream: swap it into the streamed slot */
+    r2.d_vals = rgn.d_keys;
+    r2.d_val_offs = rgn.d_key_offs;
+    r2.val_bytes = rgn.key_bytes;
+    r2.max_row_bytes = rgn.max_key_bytes;
+  }
   if (plan.mode == 1)
-    return launch_agg<false>(plan, rgn, d_simple, HashAggTable{}, s, grid);
+    return launch_agg<false>(plan, r2, d_simple, HashAggTable{}, s, grid);
   if (plan.mode == 2)
-    return launch_agg<true>(plan, rgn, nullptr, *ht, s, grid);
+    return launch_agg<true>(plan, r2, nullptr, *ht, s, grid);
   hipLaunchKernelGGL(k_scan_project, dim3(grid), dim3(THREADS), plan.lds_bytes,
                      s, plan, rgn.d_vals, rgn.d_val_offs, rgn.d_keys,
                      rgn.d_key_offs, rgn.n_kv, *po);

@@ -72,9 +72,9 @@ class Expr:
 
 
 class DagSelect:
-    """Builds a CoprDagRequest over a table-scan pipeline."""
+    """Builds a CoprDagRequest over a table- or index-scan pipeline."""
 
-    def __init__(self, columns):
+    def __init__(self, columns, index=False):
         self.columns = columns
         self._keep = []
         self.executors = []
@@ -84,7 +84,7 @@ class DagSelect:
         self._has_group = False
 
         ex = F.CoprExecutor()
-        ex.kind = F.EXEC_TABLE_SCAN
+        ex.kind = F.EXEC_INDEX_SCAN if index else F.EXEC_TABLE_SCAN
         cols = (F.CoprColumnInfo * len(columns))()
         for i, c in enumerate(columns):
             cols[i].column_id = c.column_id
