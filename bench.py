@@ -105,8 +105,9 @@ def main():
     ap.add_argument("--filter-offset", type=int, default=3,
                     help="dev: which column offset the predicate filters")
     ap.add_argument("--no-cpu-baseline", action="store_true")
-    ap.add_argument("--workload", default="cfg2", choices=["cfg2", "cfg3", "cfg4"],
-                    help="cfg2 is the contract workload; cfg3/cfg4 are secondary lines")
+    ap.add_argument("--workload", default="cfg2",
+                    choices=["cfg2", "cfg3", "cfg4", "cfg5"],
+                    help="cfg2 is the contract workload; others are secondary lines")
     args = ap.parse_args()
 
     # torch first: initialize HIP device discovery before the engine's own
@@ -126,11 +127,9 @@ def main():
         backend = "nccl" if have_cuda else "gloo"
         tdist.init_process_group(backend=backend)
         dist = tdist
-        if have_cuda:
-            torch.cuda.set_device(local_rank)
 
     n_rows = args.rows
-    cfg_index = {"cfg2": 1, "cfg3": 2, "cfg4": 3}[args.workload]
+    cfg_index = {"cfg2": 1, "cfg3": 2, "cfg4": 3, "cfg5": 4}[args.workload]
     log("rank %d/%d: generating %d rows (%s shape)" % (rank, world, n_rows, args.workload))
     t0 = time.perf_counter()
     gen = tikv_amd.GenRegion(config_index=cfg_index, n_rows=n_rows, table_id=1,
@@ -148,12 +147,21 @@ def main():
         req = tikv_amd.DagSelect(cols).hash_agg(
             [tikv_amd.count_star(), tikv_amd.sum_col(1, decimal=2),
              tikv_amd.avg_col(0)], tikv_amd.Expr().col(0)).build()
+    elif args.workload == "cfg5":
+        cols = [tikv_amd.Col(1), tikv_amd.Col(2),
+                tikv_amd.Col(-1, pk_handle=True)]
+        sel = tikv_amd.cmp_col_const(1, F.SIG_GE_INT, 0)
+        req = (tikv_amd.DagSelect(cols, index=True).where(sel)
+               .hash_agg([tikv_amd.count_star(), tikv_amd.sum_col(1)],
+                         tikv_amd.Expr().col(0)).build())
     else:
         req = None
-    eng = tikv_amd.Engine(local_rank)
+    n_dev = torch.cuda.device_count() if have_cuda else 1
+    dev = local_rank % max(n_dev, 1)
+    eng = tikv_amd.Engine(dev)
     rgn = eng.region(gen)
     if have_cuda:
-        torch.cuda.set_device(local_rank)
+        torch.cuda.set_device(dev)
 
     # algorithmic bytes per pass: every encoded value byte once + the
     # val_offs the kernel reads (8 B per row boundary). Keys are only read
@@ -161,13 +169,16 @@ def main():
     algo_bytes = gen.val_bytes() + 8 * (gen.n_kv + 1)
     if args.workload == "cfg4":
         algo_bytes += gen.key_bytes() + 8 * (gen.n_kv + 1)
+    elif args.workload == "cfg5":
+        # index scans parse the KEY stream
+        algo_bytes = gen.key_bytes() + 8 * (gen.n_kv + 1)
 
     def step():
         if args.workload == "cfg4":
             cs, kvs, byts = eng.checksum([rgn])
             return cs, 0
         data, n, kns = eng.dag_run(req, [rgn])
-        if args.workload == "cfg3":
+        if args.workload in ("cfg3", "cfg5"):
             return n, kns
         return parse_count(data), kns
 
@@ -241,6 +252,7 @@ def main():
                     "cfg2": "cfg2: 100M-row i64x16 TableScan + Selection(col3<k, 10%) + count(*)",
                     "cfg3": "cfg3: mixed i64/Decimal/VarBytes TableScan + HashAgg(count,sum,avg BY col0, K=64)",
                     "cfg4": "cfg4: CRC64-XOR checksum over KV pairs",
+                    "cfg5": "cfg5: secondary IndexScan + Selection + HashAgg",
                 }[args.workload],
                 "rows_per_gpu": n_rows,
                 "selectivity": 0.1,
