@@ -124,7 +124,9 @@ def main():
     dist = None
     if world > 1:
         import torch.distributed as tdist
-        backend = "nccl" if have_cuda else "gloo"
+        # RCCL needs one device per rank; oversubscribed test boxes use gloo
+        n_dev0 = torch.cuda.device_count() if have_cuda else 0
+        backend = "nccl" if have_cuda and world <= n_dev0 else "gloo"
         tdist.init_process_group(backend=backend)
         dist = tdist
 
