@@ -226,7 +226,12 @@ def main():
 
     result = None
     if rank == 0:
-        traffic = read_traffic()
+        # the PMC-measured traffic applies to the cfg2 kernel; scale by rows
+        traffic = None
+        if args.workload == "cfg2":
+            t = read_traffic()
+            if t is not None:
+                traffic = t * (n_rows / N_ROWS_DEFAULT)
         cpu = None
         if world == 1 and not args.no_cpu_baseline:
             log("cpu baseline (oracle) ...")
