@@ -1449,6 +1449,179 @@ k_scan_agg_pipe(ScanPlan plan,
   }
 }
 
+
+
+/* ---------------- 3-buffer counted-wait pipeline (filter+count) ----------------
+ * Keeps TWO tiles' DMA in flight while parsing a third: per tile each wave
+ * issues EXACTLY KW glds (clamped duplicates pad the count), the drain is a
+ * counted s_waitcnt vmcnt(2*KW) (prefetch stays in flight across the RAW
+ * barrier — cdna_hip_programming §5 "what does break it" / glds-span rows),
+ * and the tile bounds are read with scalar s_load so they never enter the
+ * vmcnt queue. */
+__device__ static inline uint64_t s_load_u64(const uint64_t *p) {
+  uint64_t v;
+  asm volatile("s_load_dwordx2 %0, %1, 0x0\n\ts_waitcnt lgkmcnt(0)"
+               : "=s"(v) : "s"(p));
+  return v;
+}
+
+#define THREADS3 512u
+template <int KW>
+__global__ void __launch_bounds__(THREADS3, 2)
+k_scan_fc_pipe3(ScanPlan plan,
+                const uint8_t *__restrict__ vals,
+                const uint64_t *__restrict__ val_offs, uint64_t n_rows,
+                SimpleAggAcc *__restrict__ simple_acc) {
+  extern __shared__ __attribute__((aligned(16))) uint8_t lds[];
+  const uint32_t rpt = plan.rows_per_tile;
+  const uint64_t n_tiles = (n_rows + rpt - 1) / rpt;
+  const uint32_t OS = plan.offs_slab;
+  const uint32_t BUFSZ = OS + plan.vals_slab;
+  const uint32_t wave = threadIdx.x >> 6, lane = threadIdx.x & 63u;
+  const uint32_t nwaves = THREADS3 / 64u;
+  const int64_t FCID = plan.filter_col_id;
+
+  unsigned long long cnt = 0;
+  bool any_parse_err = false;
+
+  /* issue EXACTLY KW glds; the buffer SLOT comes from the unclamped request
+     (a clamped duplicate must land in the future slot, never a live one) */
+  auto issue_tile = [&](uint64_t tile_req) {
+    uint8_t *b = lds + (uint32_t)(tile_req / gridDim.x % 3ull) * BUFSZ;
+    uint64_t t = min(tile_req, n_tiles - 1);
+    uint64_t row0 = t * rpt;
+    uint64_t row1 = min(row0 + rpt, n_rows);
+    uint64_t gb = s_load_u64(val_offs + row0);
+    uint64_t ge = s_load_u64(val_offs + row1);
+    const uint8_t *osrc = (const uint8_t *)(val_offs + row0);
+    uint32_t obytes = (uint32_t)((row1 - row0 + 1) * 8);
+    uint32_t oc = (obytes + 1023u) >> 10;
+    uint64_t abase = gb & ~15ull;
+    uint32_t tbytes = (uint32_t)(ge - abase);
+    uint32_t vc = (tbytes + 1023u) >> 10;
+    const uint8_t *vsrc = vals + abase;
+    uint8_t *bv = b + OS;
+    uint32_t vc_last = vc ? vc - 1 : 0;
+    #pragma unroll
+    for (int j = 0; j < KW; j++) {
+      uint32_t c = wave + (uint32_t)j * nwaves;
+      if (c < oc) {
+        uint32_t off = (c << 10) + lane * 16u;
+        __builtin_amdgcn_global_load_lds((const uint32_t *)(osrc + off),
+                                         (uint32_t *)(b + off), 16, 0, 0);
+      } else {
+        uint32_t cv = min(c - oc, vc_last);
+        uint32_t off = (cv << 10) + lane * 16u;
+        __builtin_amdgcn_global_load_lds((const uint32_t *)(vsrc + off),
+                                         (uint32_t *)(bv + off), 16, 0, 0);
+      }
+    }
+  };
+
+  uint64_t tile = blockIdx.x;
+  if (tile >= n_tiles) return;
+  issue_tile(tile);
+  issue_tile(tile + gridDim.x);
+
+  for (; tile < n_tiles; tile += gridDim.x) {
+    /* all waves done reading the buffer the next issue overwrites */
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    issue_tile(tile + 2ull * gridDim.x);
+    /* drain everything but the newest two tiles' DMA */
+    asm volatile("s_waitcnt vmcnt(%0)" :: "n"(2 * KW) : "memory");
+    __builtin_amdgcn_s_barrier();
+
+    uint8_t *b = lds + (uint32_t)(tile / gridDim.x % 3ull) * BUFSZ;
+    uint64_t row0 = tile * rpt;
+    uint64_t row1 = min(row0 + rpt, n_rows);
+    const uint64_t *loffs = (const uint64_t *)b;
+    uint64_t gb = loffs[0];
+    uint32_t shift = (uint32_t)(gb & 15ull);
+    const uint8_t *bv = b + OS;
+
+    for (uint64_t my_row = row0 + threadIdx.x; my_row < row1; my_row += blockDim.x) {
+      uint32_t r = (uint32_t)(my_row - row0);
+      uint64_t o0 = loffs[r], o1 = loffs[r + 1];
+      const uint8_t *vp = bv + shift + (uint32_t)(o0 - gb);
+      uint32_t vlen = (uint32_t)(o1 - o0);
+      bool found = false, fnull = false, ok = true;
+      int64_t fv = 0;
+      if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
+        uintptr_t base = (uintptr_t)vp;
+        uintptr_t wabs = ~(uintptr_t)0;
+        uint64_t wlo = 0, whi = 0;
+        uint32_t pos = 0;
+        while (pos < vlen) {
+          uintptr_t ua = base + pos;
+          if (ua - wabs > 8) {
+            wabs = ua & ~(uintptr_t)7;
+            const uint64_t *q = (const uint64_t *)wabs;
+            wlo = q[0];
+            whi = q[1];
+          }
+          uint32_t sh = (uint32_t)(ua - wabs) * 8u;
+          uint64_t x;
+          if (sh == 0) x = wlo;
+          else if (sh == 64) x = whi;
+          else x = (wlo >> sh) | (whi << (64 - sh));
+          if ((x & 0xFF) != 8) { ok = false; break; }
+          uint32_t b1 = (uint32_t)(x >> 8) & 0xFF;
+          uint32_t dflag = (uint32_t)(x >> 16) & 0xFF;
+          if (b1 < 0x80 && (dflag == 8 || dflag == 9)) {
+            uint64_t m = x >> 24;
+            uint64_t stops = ~m & 0x8080808080ull;
+            if (stops) {
+              uint32_t half = b1 >> 1;
+              int64_t cid = (b1 & 1) ? (int64_t)(~(uint64_t)half) : (int64_t)half;
+              uint32_t n = ((uint32_t)__ffsll((long long)stops)) >> 3;
+              if (cid == FCID) {
+                uint64_t vm = m & ((n == 5) ? 0xFFFFFFFFFFull
+                                            : ((1ull << (8 * n)) - 1));
+                uint64_t uv = (vm & 0x7f) | ((vm >> 8) & 0x7f) << 7 |
+                              ((vm >> 16) & 0x7f) << 14 |
+                              ((vm >> 24) & 0x7f) << 21 |
+                              ((vm >> 32) & 0x7f) << 28;
+                if (dflag == 8) {
+                  uint64_t h2 = uv >> 1;
+                  fv = (uv & 1) ? (int64_t)~h2 : (int64_t)h2;
+                } else {
+                  fv = (int64_t)uv;
+                }
+                found = true;
+                break;
+              }
+              pos += 3 + n;
+              continue;
+            }
+          }
+          {
+            int64_t cid;
+            uint32_t cell_off;
+            CellView cell;
+            if (!next_cell(vp, vlen, &pos, &cid, &cell_off, &cell)) { ok = false; break; }
+            if (cid == FCID) {
+              if (cell.is_null) fnull = true;
+              else if (cell.has_int) fv = cell.ival;
+              else ok = false;
+              found = true;
+              break;
+            }
+          }
+        }
+      }
+      if (!ok) any_parse_err = true;
+      else if (d_filter_keep(plan, found, fnull, fv)) cnt++;
+    }
+  }
+
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  for (int off = 32; off > 0; off >>= 1)
+    cnt += (unsigned long long)__shfl_down((long long)cnt, off, 64);
+  if ((threadIdx.x & 63u) == 0 && cnt) atomicAdd(&simple_acc[0].cnt, cnt);
+  if (any_parse_err) atomicOr((unsigned int *)&simple_acc[COPR_MAX_AGGS].cnt, 1u);
+}
+
 /* ---------------- direct-window filter+count kernel ----------------
  * For selective scans (one int predicate column + count(*)), skip LDS
  * staging entirely: each lane loads an aligned 64 B register window at its
@@ -2046,6 +2219,28 @@ static int launch_agg_pipe(const ScanPlan &plan, const DevRegion &rgn,
       hipLaunchKernelGGL(k_scan_fc_direct, dim3(dgrid), dim3(THREADS), 0, s,
                          plan, rgn.d_vals, rgn.d_val_offs, rgn.n_kv, d_simple);
       return (int)hipGetLastError();
+    }
+    if (!getenv("COPR_NO_PIPE3")) {
+      /* 3-buffer variant: chunks = offs + values per tile, split over 4
+         waves; KW = per-wave issue count; LDS = 3 buffers */
+      uint32_t oc = (((plan.rows_per_tile + 1) * 8) + 1023u) >> 10;
+      uint32_t vc = plan.vals_slab >> 10;
+      uint32_t kw = (oc + vc + 7u) / 8u;        /* 512-thread = 8 waves */
+      ScanPlan p3 = plan;
+      p3.lds_bytes = 3 * (plan.offs_slab + plan.vals_slab);
+      if (p3.lds_bytes <= 160 * 1024 - 2048) {
+        #define P3(KWV)                                                      \
+          hipLaunchKernelGGL((k_scan_fc_pipe3<KWV>), dim3(grid),             \
+                             dim3(512), p3.lds_bytes, s, p3, rgn.d_vals,     \
+                             rgn.d_val_offs, rgn.n_kv, d_simple)
+        if (kw <= 6) { P3(6); return (int)hipGetLastError(); }
+        if (kw <= 8) { P3(8); return (int)hipGetLastError(); }
+        if (kw <= 10) { P3(10); return (int)hipGetLastError(); }
+        if (kw <= 12) { P3(12); return (int)hipGetLastError(); }
+        if (kw <= 16) { P3(16); return (int)hipGetLastError(); }
+        if (kw <= 24) { P3(24); return (int)hipGetLastError(); }
+        #undef P3
+      }
     }
     hipLaunchKernelGGL((k_scan_agg_pipe<1, false, true>), dim3(grid),
                        dim3(THREADS), plan.lds_bytes, s, plan, rgn.d_vals,
