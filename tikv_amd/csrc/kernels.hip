@@ -1459,9 +1459,13 @@ k_scan_agg_pipe(ScanPlan plan,
  * and the tile bounds are read with scalar s_load so they never enter the
  * vmcnt queue. */
 __device__ static inline uint64_t s_load_u64(const uint64_t *p) {
+  /* force the (uniform) address into SGPRs so s_load always selects */
+  uint32_t lo = __builtin_amdgcn_readfirstlane((uint32_t)(uintptr_t)p);
+  uint32_t hi = __builtin_amdgcn_readfirstlane((uint32_t)((uintptr_t)p >> 32));
+  uint64_t addr = ((uint64_t)hi << 32) | lo;
   uint64_t v;
   asm volatile("s_load_dwordx2 %0, %1, 0x0\n\ts_waitcnt lgkmcnt(0)"
-               : "=s"(v) : "s"(p));
+               : "=s"(v) : "s"(addr));
   return v;
 }
 
