@@ -2224,7 +2224,7 @@ static int launch_agg_pipe(const ScanPlan &plan, const DevRegion &rgn,
                          plan, rgn.d_vals, rgn.d_val_offs, rgn.n_kv, d_simple);
       return (int)hipGetLastError();
     }
-    if (!getenv("COPR_NO_PIPE3")) {
+    if (getenv("COPR_PIPE3")) {
       /* 3-buffer variant: chunks = offs + values per tile, split over 4
          waves; KW = per-wave issue count; LDS = 3 buffers */
       uint32_t oc = (((plan.rows_per_tile + 1) * 8) + 1023u) >> 10;
