@@ -1626,6 +1626,200 @@ k_scan_fc_pipe3(ScanPlan plan,
   if (any_parse_err) atomicOr((unsigned int *)&simple_acc[COPR_MAX_AGGS].cnt, 1u);
 }
 
+
+/* ---------------- loader-wave ring pipeline (filter+count) ----------------
+ * 512-thread block: wave 7 is a dedicated LOADER streaming 64-row slots
+ * (offs chunk + value chunks) into a DEPTH-deep LDS ring with counted
+ * vmcnt throttling; waves 0..6 are CONSUMERS, each parsing whole slots
+ * (one row per lane). All synchronization is block-local LDS flags
+ * (monotonic slot numbers), so DMA issue never couples to parse barriers —
+ * the ldsdma-engine shape of cdna_hip_programming §5.6 / megakernel rows.
+ * KW = glds per slot (fixed by clamped duplicates). */
+template <int KW, int DEPTH>
+__global__ void __launch_bounds__(512, 2)
+k_scan_fc_ring(ScanPlan plan,
+               const uint8_t *__restrict__ vals,
+               const uint64_t *__restrict__ val_offs, uint64_t n_rows,
+               SimpleAggAcc *__restrict__ simple_acc) {
+  extern __shared__ __attribute__((aligned(16))) uint8_t lds[];
+  const uint32_t SLOT_ROWS = 64;
+  const uint32_t OS = 1024;                       /* offs slab: 65*8 -> 1 chunk */
+  const uint32_t SLOT_BYTES = OS + (uint32_t)(KW - 1) * 1024u;
+  /* ring slots first, then 2*DEPTH flag words */
+  volatile uint32_t *ready = (volatile uint32_t *)(lds + DEPTH * SLOT_BYTES);
+  volatile uint32_t *consumed = ready + DEPTH;
+
+  const uint32_t wave = threadIdx.x >> 6, lane = threadIdx.x & 63u;
+  const uint64_t n_slots = (n_rows + SLOT_ROWS - 1) / SLOT_ROWS;
+
+  /* init flags */
+  if (threadIdx.x < DEPTH) {
+    ready[threadIdx.x] = 0;
+    consumed[threadIdx.x] = 0;
+  }
+  __syncthreads();
+
+  unsigned long long cnt = 0;
+  bool any_parse_err = false;
+  const int64_t FCID = plan.filter_col_id;
+
+  if (wave == 7) {
+    /* ---- loader ---- */
+    uint32_t inflight = 0;
+    uint64_t j = 0;                               /* block-local slot round */
+    for (uint64_t slot = blockIdx.x; ; slot += gridDim.x, j++) {
+      bool live = slot < n_slots;
+      if (!live && inflight == 0) break;
+      if (live) {
+        uint32_t pos = (uint32_t)(j % DEPTH);
+        /* wait until the slot's previous round was consumed */
+        if (j >= DEPTH) {
+          uint32_t want = (uint32_t)(j - DEPTH + 1);
+          while (consumed[pos] < want) __builtin_amdgcn_s_sleep(8);
+        }
+        uint8_t *b = lds + pos * SLOT_BYTES;
+        uint64_t row0 = slot * SLOT_ROWS;
+        uint64_t row1 = min(row0 + SLOT_ROWS, n_rows);
+        uint64_t gb = s_load_u64(val_offs + row0);
+        uint64_t ge = s_load_u64(val_offs + row1);
+        const uint8_t *osrc = (const uint8_t *)(val_offs + row0);
+        uint64_t abase = gb & ~15ull;
+        uint32_t vc = ((uint32_t)(ge - abase) + 1023u) >> 10;
+        uint32_t vc_last = vc ? vc - 1 : 0;
+        const uint8_t *vsrc = vals + abase;
+        uint8_t *bv = b + OS;
+        /* KW glds: chunk 0 = offs, 1..KW-1 = value chunks (clamped) */
+        __builtin_amdgcn_global_load_lds((const uint32_t *)(osrc + lane * 16u),
+                                         (uint32_t *)(b + lane * 16u), 16, 0, 0);
+        #pragma unroll
+        for (int c = 0; c < KW - 1; c++) {
+          uint32_t cv = min((uint32_t)c, vc_last);
+          uint32_t off = (cv << 10) + lane * 16u;
+          __builtin_amdgcn_global_load_lds((const uint32_t *)(vsrc + off),
+                                           (uint32_t *)(bv + off), 16, 0, 0);
+        }
+        inflight++;
+      }
+      /* drain to <= 2 slots in flight (or everything on the tail) */
+      if (inflight > 2 || (!live && inflight)) {
+        if (live) {
+          asm volatile("s_waitcnt vmcnt(%0)" :: "n"(2 * KW) : "memory");
+          /* slot of round j-2 has landed */
+          uint32_t done_pos = (uint32_t)((j - 2) % DEPTH);
+          ready[done_pos] = (uint32_t)(j - 2 + 1);
+          inflight--;
+        } else {
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+          while (inflight) {
+            uint32_t done_pos = (uint32_t)((j - inflight) % DEPTH);
+            ready[done_pos] = (uint32_t)(j - inflight + 1);
+            inflight--;
+          }
+        }
+      }
+    }
+  } else {
+    /* ---- consumers: wave w takes block-rounds j with j % 7 == w ---- */
+    for (uint64_t j = wave; ; j += 7) {
+      uint64_t slot = blockIdx.x + j * gridDim.x;
+      if (slot >= n_slots) break;
+      uint32_t pos = (uint32_t)(j % DEPTH);
+      uint32_t want = (uint32_t)(j + 1);
+      while (ready[pos] < want) __builtin_amdgcn_s_sleep(8);
+      asm volatile("" ::: "memory");  /* data reads stay after the flag */
+      const uint8_t *b = lds + pos * SLOT_BYTES;
+      const uint64_t *loffs = (const uint64_t *)b;
+      uint64_t row0 = slot * SLOT_ROWS;
+      uint64_t row1 = min(row0 + SLOT_ROWS, n_rows);
+      uint64_t gb = loffs[0];
+      uint32_t shift = (uint32_t)(gb & 15ull);
+      const uint8_t *bv = b + OS;
+      uint64_t my_row = row0 + lane;
+      if (my_row < row1) {
+        uint32_t r = (uint32_t)(my_row - row0);
+        uint64_t o0 = loffs[r], o1 = loffs[r + 1];
+        const uint8_t *vp = bv + shift + (uint32_t)(o0 - gb);
+        uint32_t vlen = (uint32_t)(o1 - o0);
+        bool found = false, fnull = false, ok = true;
+        int64_t fv = 0;
+        if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
+          uintptr_t base = (uintptr_t)vp;
+          uintptr_t wabs = ~(uintptr_t)0;
+          uint64_t wlo = 0, whi = 0;
+          uint32_t posb = 0;
+          while (posb < vlen) {
+            uintptr_t ua = base + posb;
+            if (ua - wabs > 8) {
+              wabs = ua & ~(uintptr_t)7;
+              const uint64_t *q = (const uint64_t *)wabs;
+              wlo = q[0];
+              whi = q[1];
+            }
+            uint32_t sh = (uint32_t)(ua - wabs) * 8u;
+            uint64_t x;
+            if (sh == 0) x = wlo;
+            else if (sh == 64) x = whi;
+            else x = (wlo >> sh) | (whi << (64 - sh));
+            if ((x & 0xFF) != 8) { ok = false; break; }
+            uint32_t b1 = (uint32_t)(x >> 8) & 0xFF;
+            uint32_t dflag = (uint32_t)(x >> 16) & 0xFF;
+            if (b1 < 0x80 && (dflag == 8 || dflag == 9)) {
+              uint64_t m = x >> 24;
+              uint64_t stops = ~m & 0x8080808080ull;
+              if (stops) {
+                uint32_t half = b1 >> 1;
+                int64_t cid = (b1 & 1) ? (int64_t)(~(uint64_t)half) : (int64_t)half;
+                uint32_t n = ((uint32_t)__ffsll((long long)stops)) >> 3;
+                if (cid == FCID) {
+                  uint64_t vm = m & ((n == 5) ? 0xFFFFFFFFFFull
+                                              : ((1ull << (8 * n)) - 1));
+                  uint64_t uv = (vm & 0x7f) | ((vm >> 8) & 0x7f) << 7 |
+                                ((vm >> 16) & 0x7f) << 14 |
+                                ((vm >> 24) & 0x7f) << 21 |
+                                ((vm >> 32) & 0x7f) << 28;
+                  if (dflag == 8) {
+                    uint64_t h2 = uv >> 1;
+                    fv = (uv & 1) ? (int64_t)~h2 : (int64_t)h2;
+                  } else {
+                    fv = (int64_t)uv;
+                  }
+                  found = true;
+                  break;
+                }
+                posb += 3 + n;
+                continue;
+              }
+            }
+            {
+              int64_t cid;
+              uint32_t cell_off;
+              CellView cell;
+              if (!next_cell(vp, vlen, &posb, &cid, &cell_off, &cell)) { ok = false; break; }
+              if (cid == FCID) {
+                if (cell.is_null) fnull = true;
+                else if (cell.has_int) fv = cell.ival;
+                else ok = false;
+                found = true;
+                break;
+              }
+            }
+          }
+        }
+        if (!ok) any_parse_err = true;
+        else if (d_filter_keep(plan, found, fnull, fv)) cnt++;
+      }
+      /* done with the slot: release it to the loader (whole wave writes the
+         same value — benign) */
+      if (lane == 0) consumed[pos] = want;
+    }
+  }
+
+  for (int off = 32; off > 0; off >>= 1)
+    cnt += (unsigned long long)__shfl_down((long long)cnt, off, 64);
+  if ((threadIdx.x & 63u) == 0 && cnt) atomicAdd(&simple_acc[0].cnt, cnt);
+  if (any_parse_err) atomicOr((unsigned int *)&simple_acc[COPR_MAX_AGGS].cnt, 1u);
+}
+
 /* ---------------- direct-window filter+count kernel ----------------
  * For selective scans (one int predicate column + count(*)), skip LDS
  * staging entirely: each lane loads an aligned 64 B register window at its
@@ -2223,6 +2417,22 @@ static int launch_agg_pipe(const ScanPlan &plan, const DevRegion &rgn,
       hipLaunchKernelGGL(k_scan_fc_direct, dim3(dgrid), dim3(THREADS), 0, s,
                          plan, rgn.d_vals, rgn.d_val_offs, rgn.n_kv, d_simple);
       return (int)hipGetLastError();
+    }
+    if (getenv("COPR_RING")) {
+      /* KW = 1 offs chunk + value chunks for a 64-row slot */
+      uint32_t vc = ((64u * rgn.max_row_bytes + 16u) + 1023u) >> 10;
+      uint32_t kw = 1 + vc;
+      ScanPlan pr = plan;
+      #define RG(KWV, D)                                                     \
+        pr.lds_bytes = D * (1024u + (KWV - 1) * 1024u) + 2 * D * 4 + 64;     \
+        hipLaunchKernelGGL((k_scan_fc_ring<KWV, D>), dim3(grid), dim3(512),  \
+                           pr.lds_bytes, s, pr, rgn.d_vals, rgn.d_val_offs,  \
+                           rgn.n_kv, d_simple)
+      if (kw <= 6) { RG(6, 6); return (int)hipGetLastError(); }
+      if (kw <= 10) { RG(10, 5); return (int)hipGetLastError(); }
+      if (kw <= 14) { RG(14, 4); return (int)hipGetLastError(); }
+      if (kw <= 24) { RG(24, 4); return (int)hipGetLastError(); }
+      #undef RG
     }
     if (getenv("COPR_PIPE3")) {
       /* 3-buffer variant: chunks = offs + values per tile, split over 4
