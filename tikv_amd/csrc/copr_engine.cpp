@@ -117,6 +117,7 @@ extern "C" copr_status copr_region_create(copr_engine *eng,
     copr_region_destroy(r);
     return e == hipErrorOutOfMemory ? COPR_ERR_OOM : COPR_ERR_INTERNAL;
   }
+  if (!getenv("COPR_NO_DIR")) dev_celldir_build(r->dev, eng->stream);
   *out = r;
   return COPR_OK;
 }
@@ -171,6 +172,7 @@ extern "C" copr_status copr_region_create_mvcc(copr_engine *eng,
     copr_region_destroy(r);
     return SET_ERR(COPR_ERR_INTERNAL, "mvcc offs readback");
   }
+  if (!getenv("COPR_NO_DIR")) dev_celldir_build(r->dev, eng->stream);
   *out = r;
   return COPR_OK;
 }
@@ -203,6 +205,7 @@ extern "C" void copr_region_destroy(copr_region *r) {
   if (!r) return;
   hipFree(r->dev.d_keys); hipFree(r->dev.d_key_offs);
   hipFree(r->dev.d_vals); hipFree(r->dev.d_val_offs);
+  if (r->dev.d_celldir) hipFree(r->dev.d_celldir);
   delete r;
 }
 
@@ -677,6 +680,11 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
       hipEventRecord(ev_a, eng->stream);
       for (uint32_t rg = 0; rg < n_regions; rg++) {
         ScanPlan sp = pl.sp;
+        sp.dir_plane = nullptr;
+        if (sp.has_filter && !sp.index_mode && regions[rg]->dev.d_celldir &&
+            sp.filter_col_id >= 1 && sp.filter_col_id <= 16)
+          sp.dir_plane = regions[rg]->dev.d_celldir +
+                         (uint64_t)(sp.filter_col_id - 1) * regions[rg]->dev.n_kv;
         pick_tiling(regions[rg]->dev, &sp);
         int e = dev_scan_launch(sp, regions[rg]->dev, d_acc, nullptr, nullptr, eng->stream);
         if (e) { hipFree(d_acc); return SET_ERR(COPR_ERR_INTERNAL, "scan launch failed"); }

@@ -17,6 +17,14 @@ struct DevRegion {
   uint64_t key_bytes = 0, val_bytes = 0;
   uint32_t max_row_bytes = 0;      /* max value size over all rows */
   uint32_t max_key_bytes = 0;      /* max key size (index scans parse keys) */
+  /* cell directory: 16 column-major planes of n_kv bytes, built once at
+     region ingest. dir[(c)*n_kv + row] = byte offset of the row-v1 cell
+     whose column id is c+1 (the id-flag byte), 0xFF = column absent,
+     0xFE = row not representable (v2/empty/oversized/ids outside 1..16)
+     -> sequential walk. Removes the per-row sequential cell walk from the
+     filter parse, the measured wall once DMA fully overlaps (the same idea
+     row-v2 bakes into its format as the in-row offsets array). */
+  uint8_t *d_celldir = nullptr;
 };
 
 /* compare kinds (order matches oracle CmpKind) */
@@ -87,6 +95,9 @@ struct ScanPlan {
   int64_t out_col_ids[COPR_MAX_OUT_COLS];
   int32_t out_is_handle[COPR_MAX_OUT_COLS]; /* 1 => decoded int handle */
 
+  /* filter column's cell-directory plane (d_celldir + (col-1)*n_kv) or
+     null; consumed by the FASTFC pipe kernel only */
+  const uint8_t *dir_plane;
   /* diagnostics: 1 = stage tiles but skip the parse (bandwidth ceiling probe;
      COPR_DIAG_STAGE_ONLY=1; results are garbage, never used in tests) */
   int32_t diag_stage_only;
@@ -148,6 +159,7 @@ int dev_crc64_launch(const DevRegion &rgn, const uint64_t *d_tables /*8*256*/,
                      unsigned long long *d_xor, void *stream);
 /* MVCC write-CF filter: builds a visible-row DevRegion from raw write-CF
  * arrays already on device. 0 ok, -1 malformed, -2 oom, -3 unsupported. */
+int dev_celldir_build(DevRegion &rgn, hipStream_t s);
 int dev_mvcc_build(const uint8_t *d_keys, const uint64_t *d_ko,
                    const uint8_t *d_vals, const uint64_t *d_vo, uint64_t n,
                    uint64_t read_ts, DevRegion *out, int *unsupported,

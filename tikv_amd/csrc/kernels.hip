@@ -1137,15 +1137,20 @@ k_scan_agg_pipe(ScanPlan plan,
          Streaming 16-byte register window over the row: ~1 ds_read_b64 pair
          per two cells, no value extraction for non-target cells. */
       const int64_t FCID = plan.filter_col_id;
+      const uint8_t *DIRP = plan.dir_plane;
       unsigned long long cnt = 0;
       for (uint64_t my_row = row0 + threadIdx.x; my_row < row1; my_row += blockDim.x) {
+        /* directory byte first: global load latency overlaps the LDS reads */
+        uint32_t dir8 = DIRP ? (uint32_t)DIRP[my_row] : 0xFEu;
         uint32_t r = (uint32_t)(my_row - row0);
         uint64_t o0 = loffs[r], o1 = loffs[r + 1];
         const uint8_t *vp = bv + shift + (uint32_t)(o0 - gb);
         uint32_t vlen = (uint32_t)(o1 - o0);
         bool found = false, fnull = false, ok = true;
         int64_t fv = 0;
-        if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
+        if (dir8 == 0xFFu) {
+          /* directory: filter column absent in this row -> default fill */
+        } else if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
           if (vp[0] == 128) {        /* row v2: direct column lookup */
             V2Row r;
             if (!d_v2_parse(vp, vlen, &r)) ok = false;
@@ -1164,7 +1169,9 @@ k_scan_agg_pipe(ScanPlan plan,
           uintptr_t base = (uintptr_t)vp;
           uintptr_t wabs = ~(uintptr_t)0;   /* window invalid */
           uint64_t wlo = 0, whi = 0;
-          uint32_t pos = 0;
+          /* directory hit: jump straight to the filter column's cell (the
+             while body then matches on its first iteration) */
+          uint32_t pos = dir8 < 0xFEu ? dir8 : 0u;
           while (pos < vlen) {
             uintptr_t ua = base + pos;
             if (ua - wabs > 8) {            /* reload 16B window */
@@ -1626,6 +1633,137 @@ k_scan_fc_pipe3(ScanPlan plan,
   if (any_parse_err) atomicOr((unsigned int *)&simple_acc[COPR_MAX_AGGS].cnt, 1u);
 }
 
+
+
+/* ---------------- cell-directory ingest ----------------
+ * One-time pass at region load: per row, walk the row-v1 cells byte-wise
+ * from GLOBAL memory (codec::row walk, components/tidb_query_datum_codec)
+ * and record each small-id cell's start offset into 16 column-major byte
+ * planes. All per-row state lives in two packed u64s (no local arrays ->
+ * no scratch). */
+__device__ static inline uint32_t d_dec_bin_size(uint32_t prec, uint32_t frac) {
+  const uint32_t dig2b[10] = {0, 1, 1, 2, 2, 3, 3, 4, 4, 4};
+  uint32_t intg = prec - frac;
+  return (intg / 9) * 4 + dig2b[intg % 9] + (frac / 9) * 4 + dig2b[frac % 9];
+}
+
+__global__ void __launch_bounds__(256)
+k_build_celldir(const uint8_t *__restrict__ vals,
+                const uint64_t *__restrict__ val_offs, uint64_t n_rows,
+                uint8_t *__restrict__ dir) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t row = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       row < n_rows; row += stride) {
+    uint64_t o0 = val_offs[row], o1 = val_offs[row + 1];
+    const uint8_t *vp = vals + o0;
+    uint32_t vlen = (uint32_t)(o1 - o0);
+    uint64_t plo = ~0ull, phi = ~0ull;       /* 16 x 0xFF (absent) */
+    bool bad = false;
+    if (vlen == 0 || (vlen == 1 && vp[0] == 0) || vp[0] == 128) {
+      bad = true;                            /* empty / v2: generic path */
+    } else {
+      uint32_t pos = 0;
+      while (pos < vlen && !bad) {
+        uint32_t cell_start = pos;
+        if (pos + 2 > vlen || vp[pos] != 8) { bad = true; break; }
+        uint32_t b1 = vp[pos + 1];
+        int64_t cid;
+        uint32_t p = pos + 2;
+        if (b1 < 0x80) {
+          uint32_t half = b1 >> 1;
+          cid = (b1 & 1) ? (int64_t)(~(uint64_t)half) : (int64_t)half;
+        } else {                              /* multi-byte zigzag id */
+          uint64_t uv = (uint64_t)(b1 & 0x7F);
+          uint32_t sh = 7;
+          bool done = false;
+          while (p < vlen && sh < 70) {
+            uint8_t by = vp[p++];
+            uv |= (uint64_t)(by & 0x7F) << sh;
+            sh += 7;
+            if (!(by & 0x80)) { done = true; break; }
+          }
+          if (!done) { bad = true; break; }
+          uint64_t half = uv >> 1;
+          cid = (uv & 1) ? (int64_t)(~half) : (int64_t)half;
+        }
+        if (p >= vlen) { bad = true; break; }
+        uint8_t f = vp[p];                    /* value datum flag */
+        uint32_t adv = 0;
+        switch (f) {
+          case 0: adv = 1; break;             /* NIL */
+          case 3: case 4: case 5: case 7:     /* INT/UINT/FLOAT/DURATION */
+            adv = 9; break;
+          case 8: case 9: {                   /* VAR_INT / VAR_UINT */
+            uint32_t q = p + 1, n = 0;
+            while (q < vlen && n < 10 && (vp[q] & 0x80)) { q++; n++; }
+            if (q >= vlen) { bad = true; }
+            adv = q + 1 - p;
+            break;
+          }
+          case 1: {                           /* BYTES (memcomparable) */
+            uint32_t q = p + 1;
+            for (;;) {
+              if (q + 9 > vlen) { bad = true; break; }
+              uint8_t m = vp[q + 8];
+              q += 9;
+              if (m != 0xFF) break;
+            }
+            adv = q - p;
+            break;
+          }
+          case 2: {                           /* COMPACT_BYTES */
+            uint32_t q = p + 1;
+            uint64_t uv = 0; uint32_t sh = 0; bool done = false;
+            while (q < vlen && sh < 70) {
+              uint8_t by = vp[q++];
+              uv |= (uint64_t)(by & 0x7F) << sh;
+              sh += 7;
+              if (!(by & 0x80)) { done = true; break; }
+            }
+            int64_t n = (uv & 1) ? (int64_t)(~(uv >> 1)) : (int64_t)(uv >> 1);
+            if (!done || n < 0) { bad = true; break; }
+            adv = (q - p) + (uint32_t)n;
+            break;
+          }
+          case 6: {                           /* DECIMAL: [prec][frac][bin] */
+            if (p + 3 > vlen) { bad = true; break; }
+            adv = 3 + d_dec_bin_size(vp[p + 1], vp[p + 2]);
+            break;
+          }
+          default: bad = true; break;
+        }
+        if (bad) break;
+        pos = p + adv;
+        if (pos > vlen) { bad = true; break; }
+        if (cid >= 1 && cid <= 16) {
+          uint32_t c = (uint32_t)(cid - 1);
+          uint64_t v = (cell_start < 254) ? (uint64_t)cell_start : 0xFEull;
+          uint32_t sh = (c & 7) * 8;
+          if (c < 8) plo = (plo & ~(0xFFull << sh)) | (v << sh);
+          else       phi = (phi & ~(0xFFull << sh)) | (v << sh);
+        }
+      }
+    }
+    if (bad) { plo = 0xFEFEFEFEFEFEFEFEull; phi = plo; }
+    #pragma unroll
+    for (uint32_t c = 0; c < 16; c++) {
+      uint64_t w = c < 8 ? plo : phi;
+      dir[(uint64_t)c * n_rows + row] = (uint8_t)(w >> ((c & 7) * 8));
+    }
+  }
+}
+
+int dev_celldir_build(DevRegion &rgn, hipStream_t s) {
+  if (!rgn.n_kv) return 0;
+  hipError_t e = hipMalloc((void **)&rgn.d_celldir, rgn.n_kv * 16 + 2048);
+  if (e != hipSuccess) { rgn.d_celldir = nullptr; return 0; /* optional */ }
+  uint32_t grid = (uint32_t)min((rgn.n_kv + 255) / 256, (uint64_t)16384);
+  hipLaunchKernelGGL(k_build_celldir, dim3(grid), dim3(256), 0, s,
+                     rgn.d_vals, rgn.d_val_offs, rgn.n_kv, rgn.d_celldir);
+  e = hipStreamSynchronize(s);
+  if (e != hipSuccess) { hipFree(rgn.d_celldir); rgn.d_celldir = nullptr; }
+  return 0;
+}
 
 /* ---------------- loader-wave ring pipeline (filter+count) ----------------
  * 512-thread block: wave 7 is a dedicated LOADER streaming 64-row slots
@@ -2422,10 +2560,16 @@ static int launch_agg_pipe(const ScanPlan &plan, const DevRegion &rgn,
       /* KW = 1 offs chunk + value chunks for a 64-row slot */
       uint32_t vc = ((64u * rgn.max_row_bytes + 16u) + 1023u) >> 10;
       uint32_t kw = 1 + vc;
+      /* persistent-style grid: ~3 blocks/CU resident so each block's ring
+         reaches steady state (hundreds of rounds), not 3-4 rounds */
+      uint64_t n_slots = (rgn.n_kv + 63) / 64;
+      uint32_t rgrid = (uint32_t)min(n_slots, (uint64_t)2304);
+      if (const char *e = getenv("COPR_RING_GRID"))
+        rgrid = (uint32_t)min(n_slots, (uint64_t)atoi(e));
       ScanPlan pr = plan;
       #define RG(KWV, D)                                                     \
         pr.lds_bytes = D * (1024u + (KWV - 1) * 1024u) + 2 * D * 4 + 64;     \
-        hipLaunchKernelGGL((k_scan_fc_ring<KWV, D>), dim3(grid), dim3(512),  \
+        hipLaunchKernelGGL((k_scan_fc_ring<KWV, D>), dim3(rgrid), dim3(512), \
                            pr.lds_bytes, s, pr, rgn.d_vals, rgn.d_val_offs,  \
                            rgn.n_kv, d_simple)
       if (kw <= 6) { RG(6, 6); return (int)hipGetLastError(); }
