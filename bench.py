@@ -166,9 +166,13 @@ def main():
         torch.cuda.set_device(dev)
 
     # algorithmic bytes per pass: every encoded value byte once + the
-    # val_offs the kernel reads (8 B per row boundary). Keys are only read
-    # by the checksum (cfg4).
+    # val_offs the kernel reads (8 B per row boundary) + the filter column's
+    # cell-directory plane (1 B per row, built at ingest; COPR_NO_DIR drops
+    # it and the kernel's read of it together). Keys are only read by the
+    # checksum (cfg4).
     algo_bytes = gen.val_bytes() + 8 * (gen.n_kv + 1)
+    if not os.environ.get("COPR_NO_DIR") and args.workload == "cfg2":
+        algo_bytes += gen.n_kv
     if args.workload == "cfg4":
         algo_bytes += gen.key_bytes() + 8 * (gen.n_kv + 1)
     elif args.workload == "cfg5":

@@ -343,6 +343,29 @@ static copr_status match_filter(const CoprExpr &cond, HostPlan &pl, ScanPlan *sp
   return COPR_OK;
 }
 
+
+/* directory usable for the generic collect path? every column the plan
+   touches must be a small id with a plane (1..16) */
+static void wire_celldir(ScanPlan *sp, const DevRegion &dev) {
+  sp->dir_plane = nullptr;
+  sp->celldir = nullptr;
+  sp->celldir_n = 0;
+  if (sp->index_mode || !dev.d_celldir) return;
+  if (sp->has_filter && sp->filter_col_id >= 1 && sp->filter_col_id <= 16)
+    sp->dir_plane = dev.d_celldir +
+                    (uint64_t)(sp->filter_col_id - 1) * dev.n_kv;
+  bool ok = true;
+  if (sp->has_filter && (sp->filter_col_id < 1 || sp->filter_col_id > 16))
+    ok = false;
+  if (sp->mode == 2 && (sp->group_col_id < 1 || sp->group_col_id > 16))
+    ok = false;
+  for (int a = 0; a < sp->n_aggs && ok; a++)
+    if (sp->aggs[a].kind != DAGG_COUNT_ROWS &&
+        (sp->aggs[a].col_id < 1 || sp->aggs[a].col_id > 16))
+      ok = false;
+  if (ok) { sp->celldir = dev.d_celldir; sp->celldir_n = dev.n_kv; }
+}
+
 static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
   if (req->n_executors == 0) return SET_ERR(COPR_ERR_INVALID_REQUEST, "empty executors");
   const CoprExecutor &scan = req->executors[0];
@@ -680,11 +703,7 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
       hipEventRecord(ev_a, eng->stream);
       for (uint32_t rg = 0; rg < n_regions; rg++) {
         ScanPlan sp = pl.sp;
-        sp.dir_plane = nullptr;
-        if (sp.has_filter && !sp.index_mode && regions[rg]->dev.d_celldir &&
-            sp.filter_col_id >= 1 && sp.filter_col_id <= 16)
-          sp.dir_plane = regions[rg]->dev.d_celldir +
-                         (uint64_t)(sp.filter_col_id - 1) * regions[rg]->dev.n_kv;
+        wire_celldir(&sp, regions[rg]->dev);
         pick_tiling(regions[rg]->dev, &sp);
         int e = dev_scan_launch(sp, regions[rg]->dev, d_acc, nullptr, nullptr, eng->stream);
         if (e) { hipFree(d_acc); return SET_ERR(COPR_ERR_INTERNAL, "scan launch failed"); }
@@ -754,6 +773,7 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
         for (uint32_t rg = 0; rg < n_regions; rg++) {
           ScanPlan sp = pl.sp;
           sp.table_size = tsize;
+          wire_celldir(&sp, regions[rg]->dev);
           /* hash mode runs the single-buffer kernel with a per-block LDS
              pre-aggregation table appended after the tile region */
           pick_tiling(regions[rg]->dev, &sp, /*force_nopipe=*/true);

@@ -98,6 +98,10 @@ struct ScanPlan {
   /* filter column's cell-directory plane (d_celldir + (col-1)*n_kv) or
      null; consumed by the FASTFC pipe kernel only */
   const uint8_t *dir_plane;
+  /* full directory base + rows for the generic collect path (hash agg /
+     multi-agg): host sets it only when every needed col id is in 1..16 */
+  const uint8_t *celldir;
+  uint64_t celldir_n;
   /* diagnostics: 1 = stage tiles but skip the parse (bandwidth ceiling probe;
      COPR_DIAG_STAGE_ONLY=1; results are garbage, never used in tests) */
   int32_t diag_stage_only;

@@ -766,7 +766,70 @@ k_scan_agg(ScanPlan plan,
                                                    &grp_null, &grp_v, cols,
                                                    &hval, &hfound);
       } else if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
-        if (vp[0] == 128) {          /* row v2 */
+        /* directory fast path: one direct next_cell per needed column
+           instead of a sequential walk over every cell */
+        bool dir_done = false;
+        if (plan.celldir && vp[0] != 128) {
+          const uint8_t *db = plan.celldir;
+          const uint64_t dn = plan.celldir_n;
+          uint32_t d_f = 0xFFu, d_g = 0xFFu, d_a[NAGGS];
+          bool seq = false;
+          if (plan.has_filter)
+            d_f = db[(uint64_t)(plan.filter_col_id - 1) * dn + my_row];
+          if (IS_HASH)
+            d_g = db[(uint64_t)(plan.group_col_id - 1) * dn + my_row];
+          #pragma unroll
+          for (int a = 0; a < NAGGS; a++) {
+            d_a[a] = 0xFFu;
+            if (plan.aggs[a].kind != DAGG_COUNT_ROWS)
+              d_a[a] = db[(uint64_t)(plan.aggs[a].col_id - 1) * dn + my_row];
+          }
+          seq = (d_f == 0xFEu) | (d_g == 0xFEu);
+          #pragma unroll
+          for (int a = 0; a < NAGGS; a++) seq |= (d_a[a] == 0xFEu);
+          if (!seq) {
+            dir_done = true;
+            int64_t cid; uint32_t coff; CellView cell; uint32_t pos;
+            if (plan.has_filter && d_f != 0xFFu) {
+              pos = d_f;
+              if (next_cell(vp, vlen, &pos, &cid, &coff, &cell) &&
+                  cid == plan.filter_col_id) {
+                filt_found = true;
+                if (cell.is_null) filt_null = true;
+                else if (cell.has_int) filt_v = cell.ival;
+                else parse_ok = false;
+              } else parse_ok = false;
+            }
+            if (IS_HASH && d_g != 0xFFu) {
+              pos = d_g;
+              if (next_cell(vp, vlen, &pos, &cid, &coff, &cell) &&
+                  cid == plan.group_col_id) {
+                grp_found = true;
+                if (cell.is_null) grp_null = true;
+                else if (cell.has_int) grp_v = cell.ival;
+                else parse_ok = false;
+              } else parse_ok = false;
+            }
+            #pragma unroll
+            for (int a = 0; a < NAGGS; a++) {
+              if (plan.aggs[a].kind == DAGG_COUNT_ROWS || d_a[a] == 0xFFu)
+                continue;
+              pos = d_a[a];
+              if (next_cell(vp, vlen, &pos, &cid, &coff, &cell) &&
+                  cid == plan.aggs[a].col_id) {
+                cols[a].found = true;
+                cols[a].null = cell.is_null;
+                cols[a].iv = cell.ival;
+                cols[a].has_dec = cell.has_dec;
+                cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
+                if (!cell.is_null && !cell.has_int && !cell.has_dec)
+                  parse_ok = false;
+              } else parse_ok = false;
+            }
+          }
+        }
+        if (dir_done) {
+        } else if (vp[0] == 128) {   /* row v2 */
         parse_ok = d_v2_collect<NAGGS, IS_HASH>(plan, vp, vlen,
                                                 &filt_found, &filt_null, &filt_v,
                                                 &grp_found, &grp_null, &grp_v,
@@ -1259,7 +1322,70 @@ k_scan_agg_pipe(ScanPlan plan,
                                                    &grp_null, &grp_v, cols,
                                                    &hval, &hfound);
       } else if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
-        if (vp[0] == 128) {          /* row v2 */
+        /* directory fast path: one direct next_cell per needed column
+           instead of a sequential walk over every cell */
+        bool dir_done = false;
+        if (plan.celldir && vp[0] != 128) {
+          const uint8_t *db = plan.celldir;
+          const uint64_t dn = plan.celldir_n;
+          uint32_t d_f = 0xFFu, d_g = 0xFFu, d_a[NAGGS];
+          bool seq = false;
+          if (plan.has_filter)
+            d_f = db[(uint64_t)(plan.filter_col_id - 1) * dn + my_row];
+          if (IS_HASH)
+            d_g = db[(uint64_t)(plan.group_col_id - 1) * dn + my_row];
+          #pragma unroll
+          for (int a = 0; a < NAGGS; a++) {
+            d_a[a] = 0xFFu;
+            if (plan.aggs[a].kind != DAGG_COUNT_ROWS)
+              d_a[a] = db[(uint64_t)(plan.aggs[a].col_id - 1) * dn + my_row];
+          }
+          seq = (d_f == 0xFEu) | (d_g == 0xFEu);
+          #pragma unroll
+          for (int a = 0; a < NAGGS; a++) seq |= (d_a[a] == 0xFEu);
+          if (!seq) {
+            dir_done = true;
+            int64_t cid; uint32_t coff; CellView cell; uint32_t pos;
+            if (plan.has_filter && d_f != 0xFFu) {
+              pos = d_f;
+              if (next_cell(vp, vlen, &pos, &cid, &coff, &cell) &&
+                  cid == plan.filter_col_id) {
+                filt_found = true;
+                if (cell.is_null) filt_null = true;
+                else if (cell.has_int) filt_v = cell.ival;
+                else parse_ok = false;
+              } else parse_ok = false;
+            }
+            if (IS_HASH && d_g != 0xFFu) {
+              pos = d_g;
+              if (next_cell(vp, vlen, &pos, &cid, &coff, &cell) &&
+                  cid == plan.group_col_id) {
+                grp_found = true;
+                if (cell.is_null) grp_null = true;
+                else if (cell.has_int) grp_v = cell.ival;
+                else parse_ok = false;
+              } else parse_ok = false;
+            }
+            #pragma unroll
+            for (int a = 0; a < NAGGS; a++) {
+              if (plan.aggs[a].kind == DAGG_COUNT_ROWS || d_a[a] == 0xFFu)
+                continue;
+              pos = d_a[a];
+              if (next_cell(vp, vlen, &pos, &cid, &coff, &cell) &&
+                  cid == plan.aggs[a].col_id) {
+                cols[a].found = true;
+                cols[a].null = cell.is_null;
+                cols[a].iv = cell.ival;
+                cols[a].has_dec = cell.has_dec;
+                cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
+                if (!cell.is_null && !cell.has_int && !cell.has_dec)
+                  parse_ok = false;
+              } else parse_ok = false;
+            }
+          }
+        }
+        if (dir_done) {
+        } else if (vp[0] == 128) {   /* row v2 */
         parse_ok = d_v2_collect<NAGGS, IS_HASH>(plan, vp, vlen,
                                                 &filt_found, &filt_null, &filt_v,
                                                 &grp_found, &grp_null, &grp_v,
