@@ -170,6 +170,11 @@ typedef struct CoprExecutor {
   uint32_t          n_aggs;
   /* LIMIT / TOPN */
   uint64_t limit;
+  /* TOPN: order-by expressions (top_n_executor.rs) with per-expression
+     descending flags (NULL sorts first; desc reverses the whole order) */
+  const CoprExpr *order_by;
+  const int32_t  *order_desc;
+  uint32_t        n_order_by;
 } CoprExecutor;
 
 typedef struct CoprDagRequest {

@@ -769,7 +769,17 @@ struct Pipeline {
   std::vector<const CoprExecutor *> rest;
   /* agg runtime */
   const CoprExecutor *agg_exec = nullptr;   /* at most one agg node supported */
+  bool stream_agg = false;    /* BatchStreamAggregationExecutor: groups are
+                                 CONTIGUOUS RUNS of equal keys in input order
+                                 (stream_aggr_executor.rs:108-117); no re-merge
+                                 of non-adjacent equal keys */
   std::vector<AggDefRt> agg_defs;
+  /* BatchTopNExecutor (top_n_executor.rs): keep the n smallest rows under
+     the order-by comparator (NULL sorts first; desc reverses the whole
+     order incl. NULL placement), emitted in sorted order after the source
+     drains. Ties are emitted in source-row order (the reference's heap is
+     unstable on ties; stable-by-arrival is a deterministic refinement). */
+  const CoprExecutor *topn_exec = nullptr;
   uint64_t limit = UINT64_MAX;
 };
 
@@ -799,6 +809,16 @@ static bool build_pipeline(const CoprDagRequest *req, Pipeline *pl) {
         if (pl->agg_exec) FAIL("only one aggregation node supported");
         pl->agg_exec = &ex;
         break;
+      case COPR_EXEC_STREAM_AGG:
+        if (pl->agg_exec) FAIL("only one aggregation node supported");
+        pl->agg_exec = &ex;
+        pl->stream_agg = true;
+        break;
+      case COPR_EXEC_TOPN:
+        if (pl->topn_exec) FAIL("only one TopN node supported");
+        if (ex.n_order_by == 0) FAIL("TopN needs order-by expressions");
+        pl->topn_exec = &ex;
+        break;
       case COPR_EXEC_LIMIT:
         pl->limit = ex.limit;
         break;
@@ -809,6 +829,7 @@ static bool build_pipeline(const CoprDagRequest *req, Pipeline *pl) {
   /* out schema: no agg -> scan schema; agg -> agg outputs then group-bys
      (util/aggr_executor.rs:137-147 prepares schema: aggr outputs first,
       then FastHashAgg group_by ft appended — fast_hash_aggr_executor.rs:272) */
+  if (pl->agg_exec && pl->topn_exec) FAIL("TopN below/above agg unsupported");
   if (!pl->agg_exec) {
     pl->out_schema = pl->scan_schema;
   } else {
@@ -877,10 +898,42 @@ static bool run_pipeline(const CoprDagRequest *req,
   uint64_t out_rows = 0;
   uint64_t emitted_rows = 0;
 
+  /* TopN collection: per surviving row, the order keys + the row's encoded
+     output cells (top_n_executor.rs keeps whole rows in its heap; we keep
+     all candidates and partial-sort at drain -- same results) */
+  struct TopRow {
+    std::vector<int64_t> kv;
+    std::vector<uint8_t> knul;
+    std::vector<uint8_t> cells;
+    uint64_t arrival;
+  };
+  std::vector<TopRow> top_rows;
+  std::vector<uint8_t> top_desc;      /* per order-by expr */
+  std::vector<uint8_t> top_uns;
+  if (pl.topn_exec) {
+    for (uint32_t ob = 0; ob < pl.topn_exec->n_order_by; ob++) {
+      top_desc.push_back(pl.topn_exec->order_desc
+                             ? (uint8_t)(pl.topn_exec->order_desc[ob] != 0) : 0);
+      const CoprExprNode &last =
+          pl.topn_exec->order_by[ob].nodes[pl.topn_exec->order_by[ob].n_nodes - 1];
+      CoprFieldType oft = last.ft;
+      if (last.kind == COPR_EXPR_COLUMN_REF)
+        oft = pl.scan_schema[(size_t)last.i64_val];
+      ET oet;
+      if (!et_of_tp(oft.tp, &oet) || oet != ET::Int)
+        FAIL("TopN order-by supports int columns");
+      top_uns.push_back((uint8_t)is_unsigned(oft));
+    }
+  }
+
+  /* stream agg: run boundary = key differs from the previous row's */
+  bool sa_have_prev = false;
+  GroupKey sa_prev;
+
   /* runner loop: batch 32 -> x2 -> 1024 (runner.rs:39,51,986) */
   uint64_t batch_size = 32;
   uint64_t cursor = 0;
-  while (cursor < n_kv && emitted_rows < pl.limit) {
+  while (cursor < n_kv && (pl.topn_exec || emitted_rows < pl.limit)) {
     uint64_t n = std::min<uint64_t>(batch_size, n_kv - cursor);
     Batch batch;
     batch.cols.resize(pl.scan.cols.size());
@@ -927,7 +980,41 @@ static bool run_pipeline(const CoprDagRequest *req,
       }
     }
 
-    if (!pl.agg_exec) {
+    if (pl.topn_exec) {
+      if (!batch.logical_rows.empty()) {
+        size_t n_logical = batch.logical_rows.size();
+        std::vector<StackEntry> okeys(pl.topn_exec->n_order_by);
+        for (uint32_t ob = 0; ob < pl.topn_exec->n_order_by; ob++)
+          if (!eval_rpn(pl.topn_exec->order_by[ob], &batch, pl.scan_schema,
+                        &okeys[ob]))
+            return false;
+        for (uint32_t li = 0; li < n_logical; li++) {
+          TopRow tr;
+          tr.arrival = top_rows.size();
+          for (uint32_t ob = 0; ob < pl.topn_exec->n_order_by; ob++) {
+            const StackEntry &v = okeys[ob];
+            if (v.is_scalar) {
+              tr.knul.push_back((uint8_t)v.s_null);
+              tr.kv.push_back(v.s_i);
+            } else {
+              tr.knul.push_back((uint8_t)(v.vec.nulls[li] != 0));
+              tr.kv.push_back(v.vec.nulls[li] ? 0 : v.vec.i[li]);
+            }
+          }
+          uint32_t pi = batch.logical_rows[li];
+          std::vector<uint8_t> cells;
+          for (uint32_t oo = 0; oo < req->n_output_offsets; oo++) {
+            uint32_t off = req->output_offsets[oo];
+            if (off >= batch.cols.size()) FAIL("output offset out of range");
+            if (!encode_output_cell(batch.cols[off], pi, pl.out_schema[off],
+                                    &cells))
+              return false;
+          }
+          tr.cells = std::move(cells);
+          top_rows.push_back(std::move(tr));
+        }
+      }
+    } else if (!pl.agg_exec) {
       /* stream rows straight to the response (with LIMIT) */
       uint64_t take = std::min<uint64_t>(batch.logical_rows.size(),
                                          pl.limit - emitted_rows);
@@ -999,8 +1086,23 @@ static bool run_pipeline(const CoprDagRequest *req,
               }
             }
           }
-          auto it = groups.find(key);
           size_t gi;
+          if (pl.stream_agg) {
+            /* contiguous-run grouping: new state whenever the key changes */
+            if (!sa_have_prev || !(key == sa_prev)) {
+              gi = group_keys_in_order.size();
+              group_keys_in_order.push_back(key);
+              size_t base = states.size();
+              states.resize(base + pl.agg_defs.size());
+              for (size_t a = 0; a < pl.agg_defs.size(); a++)
+                agg_state_init(&states[base + a], pl.agg_defs[a]);
+              sa_have_prev = true;
+              sa_prev = key;
+            } else {
+              gi = group_keys_in_order.size() - 1;
+            }
+          } else {
+          auto it = groups.find(key);
           if (it == groups.end()) {
             gi = group_keys_in_order.size();
             groups.emplace(key, gi);
@@ -1011,6 +1113,7 @@ static bool run_pipeline(const CoprDagRequest *req,
               agg_state_init(&states[base + a], pl.agg_defs[a]);
           } else {
             gi = it->second;
+          }
           }
           for (size_t a = 0; a < pl.agg_defs.size(); a++)
             if (!agg_update(&states[gi * pl.agg_defs.size() + a], pl.agg_defs[a],
@@ -1024,7 +1127,34 @@ static bool run_pipeline(const CoprDagRequest *req,
     if (batch_size > 1024) batch_size = 1024;
   }
 
-  if (pl.agg_exec) {
+  if (pl.topn_exec) {
+    std::vector<size_t> order(top_rows.size());
+    for (size_t i = 0; i < order.size(); i++) order[i] = i;
+    std::sort(order.begin(), order.end(), [&](size_t x, size_t y) {
+      const TopRow &a = top_rows[x], &b = top_rows[y];
+      for (size_t k = 0; k < a.kv.size(); k++) {
+        int c;
+        if (a.knul[k] != b.knul[k]) c = a.knul[k] ? -1 : 1;   /* NULL first */
+        else if (a.knul[k]) c = 0;
+        else if (top_uns[k]) {
+          uint64_t ua = (uint64_t)a.kv[k], ub = (uint64_t)b.kv[k];
+          c = ua < ub ? -1 : (ua > ub ? 1 : 0);
+        } else {
+          c = a.kv[k] < b.kv[k] ? -1 : (a.kv[k] > b.kv[k] ? 1 : 0);
+        }
+        if (top_desc[k]) c = -c;
+        if (c) return c < 0;
+      }
+      return a.arrival < b.arrival;    /* stable tie-break */
+    });
+    uint64_t take = std::min<uint64_t>(
+        std::min<uint64_t>(pl.topn_exec->limit, pl.limit), order.size());
+    for (uint64_t i = 0; i < take; i++) {
+      const TopRow &tr = top_rows[order[i]];
+      resp->insert(resp->end(), tr.cells.begin(), tr.cells.end());
+      out_rows++;
+    }
+  } else if (pl.agg_exec) {
     /* drain: iterate groups, push states (fast_hash_aggr_executor.rs:393) */
     size_t n_groups = simple_agg ? 1 : group_keys_in_order.size();
     size_t n_out_cols = n_agg_out_cols + (simple_agg ? 0 : 1);

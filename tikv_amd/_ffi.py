@@ -43,7 +43,10 @@ class CoprExecutor(C.Structure):
                 ("conditions", C.POINTER(CoprExpr)), ("n_conditions", C.c_uint32),
                 ("group_by", C.POINTER(CoprExpr)), ("n_group_by", C.c_uint32),
                 ("aggs", C.POINTER(CoprAggDef)), ("n_aggs", C.c_uint32),
-                ("limit", C.c_uint64)]
+                ("limit", C.c_uint64),
+                ("order_by", C.POINTER(CoprExpr)),
+                ("order_desc", C.POINTER(C.c_int32)),
+                ("n_order_by", C.c_uint32)]
 
 
 class CoprDagRequest(C.Structure):

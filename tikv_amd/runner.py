@@ -148,6 +148,34 @@ class DagSelect:
     def hash_agg(self, aggs, group_by):
         return self._agg(F.EXEC_FAST_HASH_AGG, aggs, group_by)
 
+    def stream_agg(self, aggs, group_by):
+        """BatchStreamAggregationExecutor: groups are contiguous runs of
+        equal keys in input order (stream_aggr_executor.rs:108-117)."""
+        return self._agg(F.EXEC_STREAM_AGG, aggs, group_by)
+
+    def topn(self, order_by, n, desc=False):
+        """BatchTopNExecutor (top_n_executor.rs): n smallest rows under the
+        order-by comparator (NULL first; desc reverses), emitted sorted.
+        order_by: Expr or [(Expr, desc_bool), ...]."""
+        ex = F.CoprExecutor()
+        ex.kind = F.EXEC_TOPN
+        ex.limit = n
+        if not isinstance(order_by, list):
+            order_by = [(order_by, desc)]
+        earr = (F.CoprExpr * len(order_by))()
+        darr = (C.c_int32 * len(order_by))()
+        for i, (expr, d) in enumerate(order_by):
+            e, keep = expr.build()
+            self._keep += keep
+            earr[i] = e
+            darr[i] = 1 if d else 0
+        self._keep += [earr, darr]
+        ex.order_by = earr
+        ex.order_desc = darr
+        ex.n_order_by = len(order_by)
+        self.executors.append(ex)
+        return self
+
     def limit(self, n):
         ex = F.CoprExecutor()
         ex.kind = F.EXEC_LIMIT
