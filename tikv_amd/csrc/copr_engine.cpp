@@ -225,7 +225,12 @@ struct HostPlan {
   std::vector<OutAgg> out_aggs;
   bool has_agg = false;
   bool hash_agg = false;
+  bool stream_agg = false;
   CoprFieldType group_ft{};
+  /* TopN (top_n_executor.rs): single int order-by column */
+  bool has_topn = false;
+  uint64_t topn_n = 0;
+  int topn_desc = 0;
   uint64_t limit = UINT64_MAX;
 };
 
@@ -357,7 +362,8 @@ static void wire_celldir(ScanPlan *sp, const DevRegion &dev) {
   bool ok = true;
   if (sp->has_filter && (sp->filter_col_id < 1 || sp->filter_col_id > 16))
     ok = false;
-  if (sp->mode == 2 && (sp->group_col_id < 1 || sp->group_col_id > 16))
+  if ((sp->mode == 2 || sp->mode == 3 || sp->group_col_id != 0) &&
+      (sp->group_col_id < 1 || sp->group_col_id > 16))
     ok = false;
   for (int a = 0; a < sp->n_aggs && ok; a++)
     if (sp->aggs[a].kind != DAGG_COUNT_ROWS &&
@@ -403,6 +409,35 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
         agg = &ex;
         pl->hash_agg = ex.kind == COPR_EXEC_FAST_HASH_AGG;
         break;
+      case COPR_EXEC_STREAM_AGG:
+        if (agg) return SET_ERR(COPR_ERR_UNSUPPORTED, "one aggregation supported");
+        agg = &ex;
+        pl->hash_agg = true;          /* same group-by plan shape */
+        pl->stream_agg = true;
+        break;
+      case COPR_EXEC_TOPN: {
+        if (pl->has_topn) return SET_ERR(COPR_ERR_UNSUPPORTED, "one TopN supported");
+        if (sp.index_mode)
+          return SET_ERR(COPR_ERR_UNSUPPORTED, "TopN over index scan not yet native");
+        if (ex.n_order_by != 1)
+          return SET_ERR(COPR_ERR_UNSUPPORTED, "one order-by expression supported");
+        const CoprExpr &oe = ex.order_by[0];
+        if (oe.n_nodes != 1 || oe.nodes[0].kind != COPR_EXPR_COLUMN_REF)
+          return SET_ERR(COPR_ERR_UNSUPPORTED, "order-by must be a column");
+        size_t off = (size_t)oe.nodes[0].i64_val;
+        if (off >= pl->cols.size())
+          return SET_ERR(COPR_ERR_INVALID_REQUEST, "bad order-by offset");
+        const CoprColumnInfo &ci = pl->cols[off];
+        if (!et_int(ci.ft.tp) || ci.pk_handle)
+          return SET_ERR(COPR_ERR_UNSUPPORTED, "order-by type not yet native");
+        pl->has_topn = true;
+        pl->topn_n = ex.limit;
+        pl->topn_desc = ex.order_desc ? (ex.order_desc[0] != 0) : 0;
+        /* the extract pass reuses the group-col slot for the order column */
+        sp.group_col_id = ci.column_id;
+        sp.group_col_unsigned = (ci.ft.flag & COPR_FLAG_UNSIGNED) ? 1 : 0;
+        break;
+      }
       case COPR_EXEC_LIMIT:
         pl->limit = ex.limit;
         break;
@@ -411,6 +446,8 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
     }
   }
 
+  if (agg && pl->has_topn)
+    return SET_ERR(COPR_ERR_UNSUPPORTED, "TopN with aggregation unsupported");
   if (!agg) {
     if (sp.index_mode)
       return SET_ERR(COPR_ERR_UNSUPPORTED, "index scan project not yet native");
@@ -428,7 +465,7 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
 
   /* aggregation plan */
   pl->has_agg = true;
-  sp.mode = pl->hash_agg ? 2 : 1;
+  sp.mode = pl->stream_agg ? 3 : (pl->hash_agg ? 2 : 1);
   if (agg->n_aggs == 0 || agg->n_aggs > COPR_MAX_AGGS)
     return SET_ERR(COPR_ERR_UNSUPPORTED, "agg count out of range");
   sp.n_aggs = 0;
@@ -693,6 +730,117 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
   float kernel_ms = 0.0f;
   bool timed = false;
 
+  if (pl.has_topn) {
+    /* ---- TopN: winner selection on device, then the regular project path
+       over a gathered sub-region (rows already in output order) ---- */
+    hipEventDestroy(ev_a); hipEventDestroy(ev_b);
+    if (n_regions != 1)
+      return SET_ERR(COPR_ERR_UNSUPPORTED, "TopN supports one region per request");
+    copr_region *r = regions[0];
+    ScanPlan sp = pl.sp;
+    wire_celldir(&sp, r->dev);
+    pick_tiling(r->dev, &sp, /*force_nopipe=*/true);
+    uint64_t take_n = pl.topn_n < pl.limit ? pl.topn_n : pl.limit;
+    std::vector<uint32_t> winners;
+    int rc = dev_topn_select(sp, r->dev, take_n, pl.topn_desc, eng->stream,
+                             &winners);
+    if (rc == -3) return SET_ERR(COPR_ERR_STORAGE, "row parse error on device");
+    if (rc == -2) return SET_ERR(COPR_ERR_OOM, "topn temp alloc");
+    if (rc) return SET_ERR(COPR_ERR_INTERNAL, "topn selection failed");
+    if (winners.empty()) {
+      out->data = (uint8_t *)malloc(1);
+      out->data_len = 0;
+      out->n_rows = 0;
+      out->summaries = (CoprExecSummary *)calloc(req->n_executors,
+                                                 sizeof(CoprExecSummary));
+      out->n_summaries = req->n_executors;
+      return COPR_OK;
+    }
+    DevRegion sub{};
+    rc = dev_subregion_build(r->dev, winners.data(), winners.size(), &sub,
+                             eng->stream);
+    if (rc)
+      return SET_ERR(rc == -2 ? COPR_ERR_OOM : COPR_ERR_INTERNAL,
+                     "subregion build failed");
+    copr_region tmp;
+    tmp.eng = eng;
+    tmp.dev = sub;
+    auto free_sub = [&]() {
+      hipFree(sub.d_keys); hipFree(sub.d_key_offs);
+      hipFree(sub.d_vals); hipFree(sub.d_val_offs);
+      if (sub.d_celldir) hipFree(sub.d_celldir);
+    };
+    tmp.h_key_offs.resize(sub.n_kv + 1);
+    tmp.h_val_offs.resize(sub.n_kv + 1);
+    hipError_t ce = hipMemcpy(tmp.h_key_offs.data(), sub.d_key_offs,
+                              (sub.n_kv + 1) * 8, hipMemcpyDeviceToHost);
+    if (ce == hipSuccess)
+      ce = hipMemcpy(tmp.h_val_offs.data(), sub.d_val_offs, (sub.n_kv + 1) * 8,
+                     hipMemcpyDeviceToHost);
+    if (ce != hipSuccess) {
+      free_sub();
+      return SET_ERR(COPR_ERR_INTERNAL, "subregion offs readback");
+    }
+    uint32_t mr = 0, mk = 0;
+    for (uint64_t i = 0; i < sub.n_kv; i++) {
+      uint64_t l = tmp.h_val_offs[i + 1] - tmp.h_val_offs[i];
+      if (l > mr) mr = (uint32_t)l;
+      uint64_t kl = tmp.h_key_offs[i + 1] - tmp.h_key_offs[i];
+      if (kl > mk) mk = (uint32_t)kl;
+    }
+    tmp.dev.max_row_bytes = mr;
+    tmp.dev.max_key_bytes = mk;
+    /* plain project request over the sub-region: the scan node + the
+       original output offsets (paging not applicable -- <= n rows) */
+    CoprExecutor scan_ex = req->executors[0];
+    scan_ex.conditions = nullptr; scan_ex.n_conditions = 0;
+    CoprDagRequest preq{};
+    preq.executors = &scan_ex;
+    preq.n_executors = 1;
+    preq.output_offsets = req->output_offsets;
+    preq.n_output_offsets = req->n_output_offsets;
+    preq.flags = req->flags;
+    preq.div_precision_increment = req->div_precision_increment;
+    copr_region *rp = &tmp;
+    copr_status st2 = copr_dag_run(eng, &preq, &rp, 1, out);
+    free_sub();
+    return st2;
+  }
+
+  if (pl.sp.mode == 3) {
+    /* ---- stream agg: contiguous-run grouping in input order ---- */
+    if (n_regions != 1)
+      return SET_ERR(COPR_ERR_UNSUPPORTED,
+                     "stream agg supports one region per request");
+    copr_region *r = regions[0];
+    ScanPlan sp = pl.sp;
+    wire_celldir(&sp, r->dev);
+    pick_tiling(r->dev, &sp, /*force_nopipe=*/true);
+    std::vector<SimpleAggAcc> h_accs;
+    std::vector<long long> h_gk;
+    std::vector<uint8_t> h_gs;
+    hipEventRecord(ev_a, eng->stream);
+    int n_seg = dev_stream_agg(sp, r->dev, eng->stream, &h_accs, &h_gk, &h_gs);
+    hipEventRecord(ev_b, eng->stream);
+    timed = true;
+    if (n_seg == -3)
+      return SET_ERR(COPR_ERR_STORAGE, "row parse error on device");
+    if (n_seg == -2) return SET_ERR(COPR_ERR_OOM, "stream agg temp alloc");
+    if (n_seg < 0) return SET_ERR(COPR_ERR_INTERNAL, "stream agg failed");
+    size_t n_out_cols = pl.out_schema.size();
+    uint64_t take = (uint64_t)n_seg < pl.limit ? (uint64_t)n_seg : pl.limit;
+    for (uint64_t g = 0; g < take; g++) {
+      std::vector<std::vector<uint8_t>> cols(n_out_cols);
+      encode_agg_row(pl, &h_accs[g * sp.n_aggs], true, h_gs[g] == 1, h_gk[g],
+                     &cols);
+      for (uint32_t oo = 0; oo < req->n_output_offsets; oo++) {
+        uint32_t off = req->output_offsets[oo];
+        if (off < n_out_cols)
+          resp.insert(resp.end(), cols[off].begin(), cols[off].end());
+      }
+      n_rows_out++;
+    }
+  } else
   if (pl.sp.mode == 1 || pl.sp.mode == 2) {
     if (pl.sp.mode == 1) {
       /* ---- simple agg: one accumulator set across all regions ---- */

@@ -164,6 +164,14 @@ int dev_crc64_launch(const DevRegion &rgn, const uint64_t *d_tables /*8*256*/,
 /* MVCC write-CF filter: builds a visible-row DevRegion from raw write-CF
  * arrays already on device. 0 ok, -1 malformed, -2 oom, -3 unsupported. */
 int dev_celldir_build(DevRegion &rgn, hipStream_t s);
+int dev_stream_agg(const ScanPlan &plan, const DevRegion &rgn, void *stream,
+                   std::vector<SimpleAggAcc> *h_accs,
+                   std::vector<long long> *h_gk, std::vector<uint8_t> *h_gs);
+int dev_topn_select(const ScanPlan &plan, const DevRegion &rgn,
+                    uint64_t topn_n, int desc, void *stream,
+                    std::vector<uint32_t> *winners);
+int dev_subregion_build(const DevRegion &src, const uint32_t *h_rows,
+                        uint64_t m, DevRegion *out, void *stream);
 int dev_mvcc_build(const uint8_t *d_keys, const uint64_t *d_ko,
                    const uint8_t *d_vals, const uint64_t *d_vo, uint64_t n,
                    uint64_t read_ts, DevRegion *out, int *unsupported,

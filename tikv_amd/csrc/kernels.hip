@@ -2758,6 +2758,669 @@ static int launch_agg(const ScanPlan &plan, const DevRegion &rgn,
   return launch_agg2<IS_HASH, 16>(plan, rgn, d_simple, ht, s, grid);
 }
 
+
+/* ================= extract / stream-agg / TopN =================
+ * Shared extract pass: per row write (keep/key-null/key) for the plan's
+ * group (or order-by) column plus per-agg contribution values; the
+ * stream-agg and TopN pipelines below consume these columnar arrays.
+ * StreamAgg: groups are contiguous runs of equal keys in input order
+ * (stream_aggr_executor.rs:108-117). TopN: n smallest under the order
+ * comparator, NULL first, desc reverses (top_n_executor.rs). */
+struct ExtractOut {
+  uint8_t *st;       /* 0 filtered out; 1 kept NULL key; 2 kept value key */
+  int64_t *gkey;
+  int64_t *av;       /* [NAGGS][n] contribution values */
+  uint8_t *as_;      /* [NAGGS][n] 1 = contributes */
+};
+
+template <int NAGGS>
+__global__ void __launch_bounds__(256)
+k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
+               const uint64_t *__restrict__ val_offs, uint64_t n_rows,
+               ExtractOut eo, unsigned int *__restrict__ err) {
+  extern __shared__ __attribute__((aligned(16))) uint8_t lds[];
+  const uint32_t rpt = plan.rows_per_tile;
+  const uint64_t n_tiles = (n_rows + rpt - 1) / rpt;
+  bool any_err = false;
+
+  for (uint64_t tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
+    uint64_t row0 = tile * rpt;
+    uint64_t row1 = min(row0 + rpt, n_rows);
+    uint64_t gbase = val_offs[row0];
+    uint32_t tlen = (uint32_t)(val_offs[row1] - gbase);
+    __syncthreads();
+    uint32_t shift = stage_tile(vals, gbase, tlen, lds);
+
+    for (uint64_t my_row = row0 + threadIdx.x; my_row < row1;
+         my_row += blockDim.x) {
+      const uint8_t *vp = lds + shift + (uint32_t)(val_offs[my_row] - gbase);
+      uint32_t vlen = (uint32_t)(val_offs[my_row + 1] - val_offs[my_row]);
+      bool parse_ok = true;
+      bool filt_found = false, filt_null = false; int64_t filt_v = 0;
+      bool grp_found = false, grp_null = false; int64_t grp_v = 0;
+      AggColView cols[NAGGS > 0 ? NAGGS : 1];
+      #pragma unroll
+      for (int a = 0; a < NAGGS; a++) cols[a] = {false, false, false, 0, 0, 0};
+
+      if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
+        bool dir_done = false;
+        if (plan.celldir && vp[0] != 128) {
+          const uint8_t *db = plan.celldir;
+          const uint64_t dn = plan.celldir_n;
+          uint32_t d_f = 0xFFu, d_g = 0xFFu, d_a[NAGGS > 0 ? NAGGS : 1];
+          bool seq = false;
+          if (plan.has_filter)
+            d_f = db[(uint64_t)(plan.filter_col_id - 1) * dn + my_row];
+          d_g = db[(uint64_t)(plan.group_col_id - 1) * dn + my_row];
+          #pragma unroll
+          for (int a = 0; a < NAGGS; a++) {
+            d_a[a] = 0xFFu;
+            if (plan.aggs[a].kind != DAGG_COUNT_ROWS)
+              d_a[a] = db[(uint64_t)(plan.aggs[a].col_id - 1) * dn + my_row];
+          }
+          seq = (d_f == 0xFEu) | (d_g == 0xFEu);
+          #pragma unroll
+          for (int a = 0; a < NAGGS; a++) seq |= (d_a[a] == 0xFEu);
+          if (!seq) {
+            dir_done = true;
+            int64_t cid; uint32_t coff; CellView cell; uint32_t pos;
+            if (plan.has_filter && d_f != 0xFFu) {
+              pos = d_f;
+              if (next_cell(vp, vlen, &pos, &cid, &coff, &cell) &&
+                  cid == plan.filter_col_id) {
+                filt_found = true;
+                if (cell.is_null) filt_null = true;
+                else if (cell.has_int) filt_v = cell.ival;
+                else parse_ok = false;
+              } else parse_ok = false;
+            }
+            if (d_g != 0xFFu) {
+              pos = d_g;
+              if (next_cell(vp, vlen, &pos, &cid, &coff, &cell) &&
+                  cid == plan.group_col_id) {
+                grp_found = true;
+                if (cell.is_null) grp_null = true;
+                else if (cell.has_int) grp_v = cell.ival;
+                else parse_ok = false;
+              } else parse_ok = false;
+            }
+            #pragma unroll
+            for (int a = 0; a < NAGGS; a++) {
+              if (plan.aggs[a].kind == DAGG_COUNT_ROWS || d_a[a] == 0xFFu)
+                continue;
+              pos = d_a[a];
+              if (next_cell(vp, vlen, &pos, &cid, &coff, &cell) &&
+                  cid == plan.aggs[a].col_id) {
+                cols[a].found = true;
+                cols[a].null = cell.is_null;
+                cols[a].iv = cell.ival;
+                cols[a].has_dec = cell.has_dec;
+                cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
+                if (!cell.is_null && !cell.has_int && !cell.has_dec)
+                  parse_ok = false;
+              } else parse_ok = false;
+            }
+          }
+        }
+        if (dir_done) {
+        } else if (vp[0] == 128) {
+          parse_ok = d_v2_collect<(NAGGS > 0 ? NAGGS : 1), true>(
+              plan, vp, vlen, &filt_found, &filt_null, &filt_v, &grp_found,
+              &grp_null, &grp_v, cols);
+        } else {
+          uint32_t pos = 0;
+          int needed = (plan.has_filter ? 1 : 0) + 1;
+          #pragma unroll
+          for (int a = 0; a < NAGGS; a++)
+            if (plan.aggs[a].kind != DAGG_COUNT_ROWS) needed++;
+          int found = 0;
+          while (pos < vlen) {
+            int64_t cell_id;
+            uint32_t cell_off;
+            CellView cell;
+            if (!next_cell(vp, vlen, &pos, &cell_id, &cell_off, &cell)) {
+              parse_ok = false;
+              break;
+            }
+            if (plan.has_filter && !filt_found &&
+                cell_id == plan.filter_col_id) {
+              filt_found = true;
+              if (cell.is_null) filt_null = true;
+              else if (cell.has_int) filt_v = cell.ival;
+              else parse_ok = false;
+              found++;
+            }
+            if (!grp_found && cell_id == plan.group_col_id) {
+              grp_found = true;
+              if (cell.is_null) grp_null = true;
+              else if (cell.has_int) grp_v = cell.ival;
+              else parse_ok = false;
+              found++;
+            }
+            #pragma unroll
+            for (int a = 0; a < NAGGS; a++) {
+              if (plan.aggs[a].kind == DAGG_COUNT_ROWS || cols[a].found)
+                continue;
+              if (cell_id == plan.aggs[a].col_id) {
+                cols[a].found = true;
+                cols[a].null = cell.is_null;
+                cols[a].iv = cell.ival;
+                cols[a].has_dec = cell.has_dec;
+                cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
+                if (!cell.is_null && !cell.has_int && !cell.has_dec)
+                  parse_ok = false;
+                found++;
+              }
+            }
+            if (found >= needed) break;
+          }
+        }
+      }
+
+      uint8_t s = 0;
+      if (!parse_ok) {
+        any_err = true;
+      } else if (d_filter_keep(plan, filt_found, filt_null, filt_v)) {
+        s = (grp_found && !grp_null) ? 2 : 1;
+      }
+      eo.st[my_row] = s;
+      eo.gkey[my_row] = grp_v;
+      #pragma unroll
+      for (int a = 0; a < NAGGS; a++) {
+        const DevAggSpec &sp = plan.aggs[a];
+        uint8_t contribute = 0;
+        int64_t v = 0;
+        if (s) {
+          if (sp.kind == DAGG_COUNT_ROWS) {
+            contribute = 1;
+          } else if (!cols[a].found || cols[a].null) {
+          } else if (sp.kind == DAGG_COUNT_COL) {
+            contribute = 1;
+          } else if (sp.kind == DAGG_SUM_INT || d_is_fold(sp.kind)) {
+            contribute = 1; v = cols[a].iv;
+          } else {  /* SUM_DEC */
+            int d = sp.target_frac - cols[a].dfr;
+            if (!cols[a].has_dec || d < 0 || d > 18) {
+              any_err = true;
+            } else {
+              int64_t scale = 1;
+              for (int t = 0; t < d; t++) scale *= 10;
+              v = cols[a].dsc * scale;
+              contribute = 1;
+            }
+          }
+        }
+        eo.as_[(uint64_t)a * n_rows + my_row] = contribute;
+        eo.av[(uint64_t)a * n_rows + my_row] = v;
+      }
+    }
+  }
+  if (any_err) atomicOr(err, 1u);
+}
+
+__global__ static void k_st_keep_flags(const uint8_t *st, uint32_t *f,
+                                       uint64_t n, uint8_t want_lo,
+                                       uint8_t want_hi) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) f[i] = (st[i] >= want_lo && st[i] <= want_hi) ? 1u : 0u;
+}
+
+/* gather kept rows' key/state (+ per-agg) into compacted arrays */
+__global__ static void k_compact_rows(const uint8_t *st, const int64_t *gkey,
+                                      const int64_t *av, const uint8_t *as_,
+                                      const uint64_t *pos, uint64_t n,
+                                      uint64_t m, int n_aggs, int64_t *ck,
+                                      uint8_t *cs, int64_t *cav, uint8_t *cas,
+                                      uint32_t *crow) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n || !st[i]) return;
+  uint64_t p = pos[i];
+  ck[p] = gkey[i];
+  cs[p] = st[i];
+  if (crow) crow[p] = (uint32_t)i;
+  for (int a = 0; a < n_aggs; a++) {
+    cav[(uint64_t)a * m + p] = av[(uint64_t)a * n + i];
+    cas[(uint64_t)a * m + p] = as_[(uint64_t)a * n + i];
+  }
+}
+
+__global__ static void k_run_flags(const int64_t *ck, const uint8_t *cs,
+                                   uint32_t *f, uint64_t m) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= m) return;
+  f[i] = (i == 0 || ck[i] != ck[i - 1] || cs[i] != cs[i - 1]) ? 1u : 0u;
+}
+
+/* per-compacted-row aggregate update into the run's accumulator slots */
+__global__ static void k_run_update(const int64_t *ck, const uint8_t *cs,
+                                    const int64_t *cav, const uint8_t *cas,
+                                    const uint32_t *segid, uint64_t m,
+                                    ScanPlan plan, SimpleAggAcc *accs,
+                                    long long *gk, uint8_t *gs) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= m) return;
+  uint32_t seg = segid[i] - 1;
+  if (i == 0 || segid[i] != segid[i - 1]) {
+    gk[seg] = ck[i];
+    gs[seg] = cs[i];
+  }
+  SimpleAggAcc *base = accs + (uint64_t)seg * plan.n_aggs;
+  for (int a = 0; a < plan.n_aggs; a++) {
+    if (!cas[(uint64_t)a * m + i]) continue;
+    const DevAggSpec &sp = plan.aggs[a];
+    int64_t v = cav[(uint64_t)a * m + i];
+    atomicAdd(&base[a].cnt, 1ull);
+    if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC)
+      atomic_add_i128(&base[a].sum_lo, &base[a].sum_hi, v);
+    else if (d_is_fold(sp.kind)) {
+      unsigned long long b = d_fold_xform(sp.kind, v, sp.col_unsigned);
+      if (d_is_xor(sp.kind)) atomicXor(&base[a].sum_lo, b);
+      else if (sp.kind == DAGG_MAX_INT || sp.kind == DAGG_MIN_INT)
+        atomicMax(&base[a].sum_lo, b);
+      else atomicOr(&base[a].sum_lo, b);
+    }
+  }
+}
+
+__global__ static void k_sort_keys(const int64_t *ck, const uint8_t *cs,
+                                   const uint64_t *pos, const uint32_t *crow,
+                                   uint64_t m, int uns, int desc,
+                                   unsigned long long *skey, uint32_t *srow) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= m) return;
+  unsigned long long k = (unsigned long long)ck[i];
+  if (!uns) k ^= 0x8000000000000000ull;
+  if (desc) k = ~k;
+  skey[i] = k;
+  srow[i] = crow[i];
+  (void)cs; (void)pos;
+}
+
+__global__ static void k_gather_kr(const uint8_t *st, const int64_t *gkey,
+                                   const uint64_t *pos, const uint32_t *f,
+                                   uint64_t n, int64_t *ck, uint32_t *crow) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n || !f[i]) return;
+  uint64_t p = pos[i];
+  ck[p] = gkey[i];
+  crow[p] = (uint32_t)i;
+}
+
+static int extract_launch(const ScanPlan &plan, const DevRegion &rgn,
+                          const ExtractOut &eo, unsigned int *d_err,
+                          hipStream_t s) {
+  uint64_t n_tiles = (rgn.n_kv + plan.rows_per_tile - 1) / plan.rows_per_tile;
+  uint32_t grid = (uint32_t)(n_tiles < 4096 ? n_tiles : 4096);
+  if (grid == 0) grid = 1;
+  #define XCASE(NA)                                                           \
+    case NA:                                                                  \
+      hipLaunchKernelGGL((k_scan_extract<NA>), dim3(grid), dim3(256),         \
+                         plan.lds_bytes, s, plan, rgn.d_vals, rgn.d_val_offs, \
+                         rgn.n_kv, eo, d_err);                                \
+      break
+  switch (plan.n_aggs) {
+    XCASE(0); XCASE(1); XCASE(2); XCASE(3); XCASE(4);
+    default: XCASE(COPR_MAX_AGGS); break;
+  }
+  #undef XCASE
+  return (int)hipGetLastError();
+}
+
+/* Stream aggregation: extract -> compact kept rows -> run boundaries ->
+ * per-run accumulators. Returns n_runs (>=0) or -1 internal, -2 oom,
+ * -3 row parse error. Results are copied into the host vectors in run
+ * order. */
+int dev_stream_agg(const ScanPlan &plan, const DevRegion &rgn, void *stream,
+                   std::vector<SimpleAggAcc> *h_accs,
+                   std::vector<long long> *h_gk, std::vector<uint8_t> *h_gs) {
+  hipStream_t s = (hipStream_t)stream;
+  uint64_t n = rgn.n_kv;
+  if (!n) return 0;
+  uint8_t *st = nullptr, *as_ = nullptr, *cs = nullptr, *cas = nullptr;
+  uint8_t *gs = nullptr;
+  int64_t *gkey = nullptr, *av = nullptr, *ck = nullptr, *cav = nullptr;
+  long long *gk = nullptr;
+  uint32_t *f32 = nullptr, *segid = nullptr;
+  uint64_t *pos = nullptr;
+  unsigned int *d_err = nullptr;
+  SimpleAggAcc *accs = nullptr;
+  void *tmp = nullptr;
+  size_t tmpb = 0;
+  auto freeall = [&]() {
+    hipFree(st); hipFree(as_); hipFree(cs); hipFree(cas); hipFree(gs);
+    hipFree(gkey); hipFree(av); hipFree(ck); hipFree(cav); hipFree(gk);
+    hipFree(f32); hipFree(segid); hipFree(pos); hipFree(d_err);
+    hipFree(accs); hipFree(tmp);
+  };
+  int na = plan.n_aggs;
+  hipError_t e = hipSuccess;
+  if (e == hipSuccess) e = hipMalloc(&st, n);
+  if (e == hipSuccess) e = hipMalloc(&gkey, n * 8);
+  if (e == hipSuccess) e = hipMalloc(&av, (uint64_t)na * n * 8 + 8);
+  if (e == hipSuccess) e = hipMalloc(&as_, (uint64_t)na * n + 8);
+  if (e == hipSuccess) e = hipMalloc(&d_err, 4);
+  if (e == hipSuccess) e = hipMemsetAsync(d_err, 0, 4, s);
+  if (e != hipSuccess) { freeall(); return -2; }
+  ExtractOut eo{st, gkey, av, as_};
+  if (extract_launch(plan, rgn, eo, d_err, s)) { freeall(); return -1; }
+  unsigned int h_err = 0;
+  e = hipMemcpyAsync(&h_err, d_err, 4, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  if (e != hipSuccess) { freeall(); return -1; }
+  if (h_err) { freeall(); return -3; }
+
+  uint32_t blocks = (uint32_t)((n + 255) / 256);
+  if (e == hipSuccess) e = hipMalloc(&f32, n * 4 + 4);
+  if (e == hipSuccess) e = hipMalloc(&pos, n * 8 + 8);
+  if (e != hipSuccess) { freeall(); return -2; }
+  hipLaunchKernelGGL(k_st_keep_flags, dim3(blocks), dim3(256), 0, s, st, f32,
+                     n, 1, 2);
+  hipcub::DeviceScan::ExclusiveSum(nullptr, tmpb, f32, pos, (int)n, s);
+  if (hipMalloc(&tmp, tmpb) != hipSuccess) { freeall(); return -2; }
+  hipcub::DeviceScan::ExclusiveSum(tmp, tmpb, f32, pos, (int)n, s);
+  uint64_t m = 0;
+  uint32_t last_f = 0;
+  e = hipMemcpyAsync(&m, pos + (n - 1), 8, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess)
+    e = hipMemcpyAsync(&last_f, f32 + (n - 1), 4, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  if (e != hipSuccess) { freeall(); return -1; }
+  m += last_f;
+  if (!m) { freeall(); return 0; }
+
+  uint64_t ma = m;
+  if (e == hipSuccess) e = hipMalloc(&ck, ma * 8);
+  if (e == hipSuccess) e = hipMalloc(&cs, ma);
+  if (e == hipSuccess) e = hipMalloc(&cav, (uint64_t)na * ma * 8 + 8);
+  if (e == hipSuccess) e = hipMalloc(&cas, (uint64_t)na * ma + 8);
+  if (e != hipSuccess) { freeall(); return -2; }
+  hipLaunchKernelGGL(k_compact_rows, dim3(blocks), dim3(256), 0, s, st, gkey,
+                     av, as_, pos, n, m, na, ck, cs, cav, cas,
+                     (uint32_t *)nullptr);
+
+  uint32_t mblocks = (uint32_t)((m + 255) / 256);
+  if (hipMalloc(&segid, ma * 4) != hipSuccess) { freeall(); return -2; }
+  hipLaunchKernelGGL(k_run_flags, dim3(mblocks), dim3(256), 0, s, ck, cs, f32,
+                     m);
+  size_t tmpb2 = 0;
+  hipcub::DeviceScan::InclusiveSum(nullptr, tmpb2, f32, segid, (int)m, s);
+  if (tmpb2 > tmpb) {
+    hipFree(tmp); tmp = nullptr;
+    if (hipMalloc(&tmp, tmpb2) != hipSuccess) { freeall(); return -2; }
+    tmpb = tmpb2;
+  }
+  hipcub::DeviceScan::InclusiveSum(tmp, tmpb2, f32, segid, (int)m, s);
+  uint32_t n_seg = 0;
+  e = hipMemcpyAsync(&n_seg, segid + (m - 1), 4, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  if (e != hipSuccess) { freeall(); return -1; }
+
+  if (e == hipSuccess) e = hipMalloc(&accs, (uint64_t)n_seg * na *
+                                                sizeof(SimpleAggAcc));
+  if (e == hipSuccess) e = hipMalloc(&gk, (uint64_t)n_seg * 8);
+  if (e == hipSuccess) e = hipMalloc(&gs, n_seg);
+  if (e == hipSuccess)
+    e = hipMemsetAsync(accs, 0, (uint64_t)n_seg * na * sizeof(SimpleAggAcc), s);
+  if (e != hipSuccess) { freeall(); return -2; }
+  hipLaunchKernelGGL(k_run_update, dim3(mblocks), dim3(256), 0, s, ck, cs,
+                     cav, cas, segid, m, plan, accs, gk, gs);
+  h_accs->resize((size_t)n_seg * na);
+  h_gk->resize(n_seg);
+  h_gs->resize(n_seg);
+  e = hipMemcpyAsync(h_accs->data(), accs,
+                     (uint64_t)n_seg * na * sizeof(SimpleAggAcc),
+                     hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess)
+    e = hipMemcpyAsync(h_gk->data(), gk, (uint64_t)n_seg * 8,
+                       hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess)
+    e = hipMemcpyAsync(h_gs->data(), gs, n_seg, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  freeall();
+  if (e != hipSuccess) return -1;
+  return (int)n_seg;
+}
+
+/* TopN winner selection: extract the order column, split kept rows into
+ * NULL-key and value-key lists (both in source order), radix-sort the
+ * value list (stable -> ties keep arrival order), and take the first n
+ * per the NULL-first / desc rules. Returns 0 (winners filled), -1
+ * internal, -2 oom, -3 parse error. */
+int dev_topn_select(const ScanPlan &plan, const DevRegion &rgn,
+                    uint64_t topn_n, int desc, void *stream,
+                    std::vector<uint32_t> *winners) {
+  hipStream_t s = (hipStream_t)stream;
+  uint64_t n = rgn.n_kv;
+  winners->clear();
+  if (!n || !topn_n) return 0;
+  uint8_t *st = nullptr, *cs = nullptr, *cas = nullptr;
+  int64_t *gkey = nullptr, *ck = nullptr, *cav = nullptr;
+  uint32_t *f32 = nullptr, *crow = nullptr, *srow = nullptr, *srow2 = nullptr;
+  uint64_t *pos = nullptr;
+  unsigned long long *skey = nullptr, *skey2 = nullptr;
+  unsigned int *d_err = nullptr;
+  void *tmp = nullptr;
+  size_t tmpb = 0;
+  auto freeall = [&]() {
+    hipFree(st); hipFree(cs); hipFree(cas); hipFree(gkey); hipFree(ck);
+    hipFree(cav); hipFree(f32); hipFree(crow); hipFree(srow); hipFree(srow2);
+    hipFree(pos); hipFree(skey); hipFree(skey2); hipFree(d_err); hipFree(tmp);
+  };
+  hipError_t e = hipSuccess;
+  if (e == hipSuccess) e = hipMalloc(&st, n);
+  if (e == hipSuccess) e = hipMalloc(&gkey, n * 8);
+  if (e == hipSuccess) e = hipMalloc(&cav, 16);   /* NAGGS=0 stubs */
+  if (e == hipSuccess) e = hipMalloc(&cas, 16);
+  if (e == hipSuccess) e = hipMalloc(&d_err, 4);
+  if (e == hipSuccess) e = hipMemsetAsync(d_err, 0, 4, s);
+  if (e != hipSuccess) { freeall(); return -2; }
+  ExtractOut eo{st, gkey, (int64_t *)cav, cas};
+  ScanPlan p0 = plan;
+  p0.n_aggs = 0;
+  if (extract_launch(p0, rgn, eo, d_err, s)) { freeall(); return -1; }
+  unsigned int h_err = 0;
+  e = hipMemcpyAsync(&h_err, d_err, 4, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  if (e != hipSuccess) { freeall(); return -1; }
+  if (h_err) { freeall(); return -3; }
+
+  uint32_t blocks = (uint32_t)((n + 255) / 256);
+  if (e == hipSuccess) e = hipMalloc(&f32, n * 4 + 4);
+  if (e == hipSuccess) e = hipMalloc(&pos, n * 8 + 8);
+  if (e != hipSuccess) { freeall(); return -2; }
+
+  /* pass A: NULL-key rows (st==1); pass B: value rows (st==2) */
+  std::vector<uint32_t> a_rows, b_rows_sorted;
+  for (int pass = 0; pass < 2; pass++) {
+    uint8_t want = pass == 0 ? 1 : 2;
+    hipLaunchKernelGGL(k_st_keep_flags, dim3(blocks), dim3(256), 0, s, st,
+                       f32, n, want, want);
+    size_t tb = 0;
+    hipcub::DeviceScan::ExclusiveSum(nullptr, tb, f32, pos, (int)n, s);
+    if (tb > tmpb) {
+      hipFree(tmp); tmp = nullptr;
+      if (hipMalloc(&tmp, tb) != hipSuccess) { freeall(); return -2; }
+      tmpb = tb;
+    }
+    hipcub::DeviceScan::ExclusiveSum(tmp, tmpb, f32, pos, (int)n, s);
+    uint64_t m = 0;
+    uint32_t lf = 0;
+    e = hipMemcpyAsync(&m, pos + (n - 1), 8, hipMemcpyDeviceToHost, s);
+    if (e == hipSuccess)
+      e = hipMemcpyAsync(&lf, f32 + (n - 1), 4, hipMemcpyDeviceToHost, s);
+    if (e == hipSuccess) e = hipStreamSynchronize(s);
+    if (e != hipSuccess) { freeall(); return -1; }
+    m += lf;
+    if (!m) continue;
+    /* gather (key,row); reuse a tiny "state" compact with want-only rows */
+    if (ck) { hipFree(ck); ck = nullptr; }
+    if (cs) { hipFree(cs); cs = nullptr; }
+    if (crow) { hipFree(crow); crow = nullptr; }
+    if (e == hipSuccess) e = hipMalloc(&ck, m * 8);
+    if (e == hipSuccess) e = hipMalloc(&cs, m);
+    if (e == hipSuccess) e = hipMalloc(&crow, m * 4);
+    if (e != hipSuccess) { freeall(); return -2; }
+    hipLaunchKernelGGL(k_gather_kr, dim3(blocks), dim3(256), 0, s, st, gkey,
+                       pos, f32, n, ck, crow);
+    if (pass == 0) {
+      a_rows.resize(m);
+      e = hipMemcpyAsync(a_rows.data(), crow, m * 4, hipMemcpyDeviceToHost, s);
+      if (e == hipSuccess) e = hipStreamSynchronize(s);
+      if (e != hipSuccess) { freeall(); return -1; }
+      /* only the first topn_n can matter */
+      if (a_rows.size() > topn_n) a_rows.resize(topn_n);
+    } else {
+      uint32_t mb = (uint32_t)((m + 255) / 256);
+      if (e == hipSuccess) e = hipMalloc(&skey, m * 8);
+      if (e == hipSuccess) e = hipMalloc(&skey2, m * 8);
+      if (e == hipSuccess) e = hipMalloc(&srow, m * 4);
+      if (e == hipSuccess) e = hipMalloc(&srow2, m * 4);
+      if (e != hipSuccess) { freeall(); return -2; }
+      hipLaunchKernelGGL(k_sort_keys, dim3(mb), dim3(256), 0, s, ck, cs, pos,
+                         crow, m, plan.group_col_unsigned, desc, skey, srow);
+      size_t tb = 0;
+      hipcub::DeviceRadixSort::SortPairs(nullptr, tb, skey, skey2, srow,
+                                         srow2, (int)m, 0, 64, s);
+      if (tb > tmpb) {
+        hipFree(tmp); tmp = nullptr;
+        if (hipMalloc(&tmp, tb) != hipSuccess) { freeall(); return -2; }
+        tmpb = tb;
+      }
+      hipcub::DeviceRadixSort::SortPairs(tmp, tmpb, skey, skey2, srow, srow2,
+                                         (int)m, 0, 64, s);
+      uint64_t take = m < topn_n ? m : topn_n;
+      b_rows_sorted.resize(take);
+      e = hipMemcpyAsync(b_rows_sorted.data(), srow2, take * 4,
+                         hipMemcpyDeviceToHost, s);
+      if (e == hipSuccess) e = hipStreamSynchronize(s);
+      if (e != hipSuccess) { freeall(); return -1; }
+    }
+  }
+  freeall();
+  /* order: asc -> NULLs first; desc -> NULLs last */
+  if (!desc) {
+    for (uint32_t r : a_rows) {
+      if (winners->size() >= topn_n) break;
+      winners->push_back(r);
+    }
+    for (uint32_t r : b_rows_sorted) {
+      if (winners->size() >= topn_n) break;
+      winners->push_back(r);
+    }
+  } else {
+    for (uint32_t r : b_rows_sorted) {
+      if (winners->size() >= topn_n) break;
+      winners->push_back(r);
+    }
+    for (uint32_t r : a_rows) {
+      if (winners->size() >= topn_n) break;
+      winners->push_back(r);
+    }
+  }
+  return 0;
+}
+
+/* gather the given rows (in order) of a region into a new DevRegion */
+__global__ static void k_sub_sizes(const uint64_t *ko, const uint64_t *vo,
+                                   const uint32_t *rows, uint64_t m,
+                                   uint64_t *klen, uint64_t *vlen) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= m) return;
+  uint32_t r = rows[i];
+  klen[i] = ko[r + 1] - ko[r];
+  vlen[i] = vo[r + 1] - vo[r];
+}
+
+__global__ static void k_sub_copy(const uint8_t *src_k, const uint64_t *src_ko,
+                                  const uint8_t *src_v, const uint64_t *src_vo,
+                                  const uint32_t *rows, uint64_t m,
+                                  const uint64_t *dst_ko,
+                                  const uint64_t *dst_vo, uint8_t *dst_k,
+                                  uint8_t *dst_v) {
+  uint64_t i = blockIdx.x;               /* one block per row */
+  if (i >= m) return;
+  uint32_t r = rows[i];
+  uint64_t ks = src_ko[r], kl = src_ko[r + 1] - ks, kd = dst_ko[i];
+  for (uint64_t b = threadIdx.x; b < kl; b += blockDim.x)
+    dst_k[kd + b] = src_k[ks + b];
+  uint64_t vs = src_vo[r], vl = src_vo[r + 1] - vs, vd = dst_vo[i];
+  for (uint64_t b = threadIdx.x; b < vl; b += blockDim.x)
+    dst_v[vd + b] = src_v[vs + b];
+}
+
+int dev_subregion_build(const DevRegion &src, const uint32_t *h_rows,
+                        uint64_t m, DevRegion *out, void *stream) {
+  hipStream_t s = (hipStream_t)stream;
+  *out = DevRegion{};
+  out->n_kv = m;
+  uint32_t *d_rows = nullptr;
+  uint64_t *klen = nullptr, *vlen = nullptr;
+  void *tmp = nullptr;
+  size_t tmpb = 0;
+  hipError_t e = hipSuccess;
+  auto fail = [&](int rc) {
+    hipFree(d_rows); hipFree(klen); hipFree(vlen); hipFree(tmp);
+    hipFree(out->d_keys); hipFree(out->d_key_offs);
+    hipFree(out->d_vals); hipFree(out->d_val_offs);
+    *out = DevRegion{};
+    return rc;
+  };
+  if (!m) {
+    if (hipMalloc(&out->d_key_offs, 8 + 2048) != hipSuccess) return -2;
+    if (hipMalloc(&out->d_val_offs, 8 + 2048) != hipSuccess) return fail(-2);
+    if (hipMalloc(&out->d_keys, 2048) != hipSuccess) return fail(-2);
+    if (hipMalloc(&out->d_vals, 2048) != hipSuccess) return fail(-2);
+    hipMemsetAsync(out->d_key_offs, 0, 8, s);
+    hipMemsetAsync(out->d_val_offs, 0, 8, s);
+    hipStreamSynchronize(s);
+    return 0;
+  }
+  if (e == hipSuccess) e = hipMalloc(&d_rows, m * 4);
+  if (e == hipSuccess) e = hipMalloc(&klen, (m + 1) * 8);
+  if (e == hipSuccess) e = hipMalloc(&vlen, (m + 1) * 8);
+  if (e == hipSuccess) e = hipMalloc(&out->d_key_offs, (m + 1) * 8 + 2048);
+  if (e == hipSuccess) e = hipMalloc(&out->d_val_offs, (m + 1) * 8 + 2048);
+  if (e == hipSuccess)
+    e = hipMemcpyAsync(d_rows, h_rows, m * 4, hipMemcpyHostToDevice, s);
+  if (e != hipSuccess) return fail(-2);
+  uint32_t blocks = (uint32_t)((m + 255) / 256);
+  hipLaunchKernelGGL(k_sub_sizes, dim3(blocks), dim3(256), 0, s,
+                     src.d_key_offs, src.d_val_offs, d_rows, m, klen, vlen);
+  hipMemsetAsync(klen + m, 0, 8, s);
+  hipMemsetAsync(vlen + m, 0, 8, s);
+  hipcub::DeviceScan::ExclusiveSum(nullptr, tmpb, klen, out->d_key_offs,
+                                   (int)(m + 1), s);
+  if (hipMalloc(&tmp, tmpb) != hipSuccess) return fail(-2);
+  hipcub::DeviceScan::ExclusiveSum(tmp, tmpb, klen, out->d_key_offs,
+                                   (int)(m + 1), s);
+  hipcub::DeviceScan::ExclusiveSum(tmp, tmpb, vlen, out->d_val_offs,
+                                   (int)(m + 1), s);
+  uint64_t kb = 0, vb = 0;
+  e = hipMemcpyAsync(&kb, out->d_key_offs + m, 8, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess)
+    e = hipMemcpyAsync(&vb, out->d_val_offs + m, 8, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  if (e != hipSuccess) return fail(-1);
+  out->key_bytes = kb;
+  out->val_bytes = vb;
+  if (e == hipSuccess) e = hipMalloc(&out->d_keys, kb + 2048);
+  if (e == hipSuccess) e = hipMalloc(&out->d_vals, vb + 2048);
+  if (e != hipSuccess) return fail(-2);
+  hipLaunchKernelGGL(k_sub_copy, dim3((uint32_t)m), dim3(64), 0, s, src.d_keys,
+                     src.d_key_offs, src.d_vals, src.d_val_offs, d_rows, m,
+                     out->d_key_offs, out->d_val_offs, out->d_keys,
+                     out->d_vals);
+  e = hipStreamSynchronize(s);
+  hipFree(d_rows); hipFree(klen); hipFree(vlen); hipFree(tmp);
+  if (e != hipSuccess) {
+    hipFree(out->d_keys); hipFree(out->d_key_offs);
+    hipFree(out->d_vals); hipFree(out->d_val_offs);
+    *out = DevRegion{};
+    return -1;
+  }
+  return 0;
+}
+
 int dev_scan_launch(const ScanPlan &plan, const DevRegion &rgn,
                     SimpleAggAcc *d_simple, const HashAggTable *ht,
                     const ProjectOut *po, void *stream) {
