@@ -41,6 +41,10 @@ struct copr_engine {
   int device = 0;
   hipStream_t stream = nullptr;
   uint64_t *d_crc_tables = nullptr;   /* 8*256 u64, built lazily */
+  /* internal channel for the TopN sub-region project: column OFFSET the
+     order expression decoded in place (response encodes it decoded,
+     lazy_column.rs:165,242); -1 = none */
+  int dec_col_off = -1;
 };
 
 struct copr_region {
@@ -231,6 +235,7 @@ struct HostPlan {
   bool has_topn = false;
   uint64_t topn_n = 0;
   int topn_desc = 0;
+  int topn_off = -1;
   uint64_t limit = UINT64_MAX;
 };
 
@@ -431,6 +436,7 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
         if (!et_int(ci.ft.tp) || ci.pk_handle)
           return SET_ERR(COPR_ERR_UNSUPPORTED, "order-by type not yet native");
         pl->has_topn = true;
+        pl->topn_off = (int)off;
         pl->topn_n = ex.limit;
         pl->topn_desc = ex.order_desc ? (ex.order_desc[0] != 0) : 0;
         /* the extract pass reuses the group-col slot for the order column */
@@ -718,6 +724,27 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
   copr_status st = build_plan(req, &pl);
   if (st != COPR_OK) return st;
 
+  if (eng->dec_col_off >= 0 && !pl.has_agg && !pl.has_topn &&
+      !pl.sp.has_filter && (size_t)eng->dec_col_off < pl.cols.size()) {
+    /* TopN sub-region project: the order column was decoded in place by
+       the order expression -> output it in decoded form */
+    const CoprColumnInfo &ci = pl.cols[eng->dec_col_off];
+    pl.filter_col_offset = eng->dec_col_off;
+    pl.sp.has_filter = 1;
+    pl.sp.filter_decode_only = 1;
+    pl.sp.filter_col_id = ci.column_id;
+    pl.sp.filter_col_unsigned = (ci.ft.flag & COPR_FLAG_UNSIGNED) ? 1 : 0;
+    if (ci.default_val && ci.default_val_len) {
+      int64_t dv;
+      int r = host_decode_int_datum(ci.default_val, ci.default_val_len, &dv);
+      if (r < 0) return SET_ERR(COPR_ERR_INVALID_REQUEST, "bad default");
+      pl.sp.filter_missing_null = r == 1 ? 1 : 0;
+      pl.sp.filter_missing_val = dv;
+    } else {
+      pl.sp.filter_missing_null = 1;
+    }
+  }
+
   std::vector<uint8_t> resp;
   uint64_t n_rows_out = 0;
   /* HIP-event timing of the scan kernel(s), on the engine's own stream —
@@ -802,7 +829,9 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
     preq.flags = req->flags;
     preq.div_precision_increment = req->div_precision_increment;
     copr_region *rp = &tmp;
+    eng->dec_col_off = pl.topn_off;
     copr_status st2 = copr_dag_run(eng, &preq, &rp, 1, out);
+    eng->dec_col_off = -1;
     free_sub();
     return st2;
   }

@@ -74,6 +74,11 @@ struct ScanPlan {
      fill (table_scan_executor.rs:456-483) decoded */
   int32_t filter_missing_null;   /* 1 => NULL */
   int64_t filter_missing_val;
+  /* 1 = keep every row but still export filt_vals/filt_state: used for a
+     column an upstream expression decoded in place (e.g. the TopN order
+     column) so the response encodes it in DECODED form
+     (lazy_column.rs:165,242) */
+  int32_t filter_decode_only;
 
   int32_t mode;                  /* 0 project, 1 simple agg, 2 hash agg */
   /* index scan (BatchIndexScanExecutor): the parsed stream is the KEY

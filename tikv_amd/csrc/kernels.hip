@@ -602,6 +602,7 @@ __device__ static inline bool d_cmp_res(int32_t kind, int ord) {
 /* evaluate the plan's filter for one row's decoded filter-column state */
 __device__ static inline bool d_filter_keep(const ScanPlan &plan, bool found,
                                             bool is_null, int64_t v) {
+  if (plan.filter_decode_only) return true;
   if (!plan.has_filter) return true;
   if (!found) {
     if (plan.filter_missing_null) is_null = true;
