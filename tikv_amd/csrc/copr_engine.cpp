@@ -588,30 +588,32 @@ static void pick_tiling(const DevRegion &rgn, ScanPlan *sp, bool force_nopipe = 
   uint32_t per_row = (sp->index_mode ? rgn.max_key_bytes : rgn.max_row_bytes) + 1;
 
   if (!force_nopipe && !getenv("COPR_NO_PIPE")) {
-    /* glds double-buffer pipeline: 2 x (offs slab + vals slab), 1 KiB
-       granular. Target <= ~120 KiB so at least one more block can coexist
-       on smaller shapes; >= 2 blocks/CU needs <= 76 KiB. */
-    uint32_t budget = 76 * 1024;
+    /* glds double-buffer pipeline: 2 x (offs slab + vals slab [+ dir
+       slab]), 1 KiB granular. >= 2 blocks/CU needs <= ~78 KiB. */
+    uint32_t ds = (sp->dir_plane && !getenv("COPR_NO_DIR_SLAB")) ? 1024u : 0u;
+    uint32_t budget = 76 * 1024 + 2 * ds;
     if (const char *e = getenv("COPR_PIPE_LDS_BUDGET")) budget = (uint32_t)atoi(e);
     uint32_t rows = 1024;
     if (const char *e = getenv("COPR_ROWS_PER_TILE")) rows = (uint32_t)atoi(e);
     while (rows > 64) {
       uint32_t os = (((rows + 1) * 8) + 1023u) & ~1023u;
       uint32_t vs = ((rows * per_row + 31u) + 1023u) & ~1023u;
-      if (2 * (os + vs) <= budget) break;
+      if (2 * (os + vs + ds) <= budget) break;
       rows /= 2;
     }
     uint32_t os = (((rows + 1) * 8) + 1023u) & ~1023u;
     uint32_t vs = ((rows * per_row + 31u) + 1023u) & ~1023u;
-    if (rows >= 64 && 2 * (os + vs) <= 160 * 1024 - 2048) {
+    if (rows >= 64 && 2 * (os + vs + ds) <= 160 * 1024 - 2048) {
       sp->use_pipe = 1;
       sp->rows_per_tile = rows;
       sp->offs_slab = os;
       sp->vals_slab = vs;
-      sp->lds_bytes = 2 * (os + vs);
+      sp->dir_slab = ds;
+      sp->lds_bytes = 2 * (os + vs + ds);
       return;
     }
   }
+  sp->dir_slab = 0;
 
   uint32_t budget = 64 * 1024;          /* >=2 blocks/CU on 160 KiB LDS */
   if (const char *e = getenv("COPR_LDS_BUDGET")) budget = (uint32_t)atoi(e);

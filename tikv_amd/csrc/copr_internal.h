@@ -120,6 +120,11 @@ struct ScanPlan {
   int32_t glds_nt;               /* nt (aux=2) on the values stream */
   uint32_t offs_slab;
   uint32_t vals_slab;
+  /* 1 KiB per buffer staging the tile's slice of the filter column's
+     directory plane (dir_plane != null): the per-row dir byte is the head
+     of the parse dependency chain, so it rides the DMA instead of being a
+     per-row global load */
+  uint32_t dir_slab;
   /* hash agg: per-block LDS pre-aggregation table (0 = disabled).
      Low-cardinality GROUP BY otherwise serializes on a handful of global
      atomic addresses. */

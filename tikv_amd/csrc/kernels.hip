@@ -1106,7 +1106,9 @@ k_scan_agg_pipe(ScanPlan plan,
   const uint32_t rpt = plan.rows_per_tile;
   const uint64_t n_tiles = (n_rows + rpt - 1) / rpt;
   const uint32_t OS = plan.offs_slab;
-  const uint32_t BUFSZ = OS + plan.vals_slab;
+  const uint32_t DS = plan.dir_slab;          /* dir slice after the values */
+  const uint32_t BUFSZ = OS + plan.vals_slab + DS;
+  const uint32_t DOFF = OS + plan.vals_slab;
   const uint32_t wave = threadIdx.x >> 6, lane = threadIdx.x & 63u;
   const uint32_t nwaves = THREADS / 64u;
 
@@ -1159,6 +1161,14 @@ k_scan_agg_pipe(ScanPlan plan,
                                          (uint32_t *)(bv + off), 16, 0, 0);
       }
     }
+    if (DS && wave == nwaves - 1) {
+      /* tile's directory slice: rpt <= 1024 bytes = one 1 KiB chunk
+         (row0 is rpt-aligned and the plane has tail slack) */
+      const uint8_t *dsrc = plan.dir_plane + row0;
+      __builtin_amdgcn_global_load_lds((const uint32_t *)(dsrc + lane * 16u),
+                                       (uint32_t *)(b + DOFF + lane * 16u),
+                                       16, 0, 0);
+    }
   };
   auto bounds_of = [&](uint64_t tile, uint64_t *gb, uint64_t *ge) {
     uint64_t row0 = tile * rpt;
@@ -1202,10 +1212,11 @@ k_scan_agg_pipe(ScanPlan plan,
          per two cells, no value extraction for non-target cells. */
       const int64_t FCID = plan.filter_col_id;
       const uint8_t *DIRP = plan.dir_plane;
+      const uint8_t *b_dir = b + DOFF;
       unsigned long long cnt = 0;
       for (uint64_t my_row = row0 + threadIdx.x; my_row < row1; my_row += blockDim.x) {
-        /* directory byte first: global load latency overlaps the LDS reads */
-        uint32_t dir8 = DIRP ? (uint32_t)DIRP[my_row] : 0xFEu;
+        uint32_t dir8 = DS ? (uint32_t)b_dir[my_row - row0]
+                           : (DIRP ? (uint32_t)DIRP[my_row] : 0xFEu);
         uint32_t r = (uint32_t)(my_row - row0);
         uint64_t o0 = loffs[r], o1 = loffs[r + 1];
         const uint8_t *vp = bv + shift + (uint32_t)(o0 - gb);
