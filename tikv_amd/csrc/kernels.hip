@@ -2012,17 +2012,21 @@ k_scan_fc_ring(ScanPlan plan,
       const uint8_t *bv = b + OS;
       uint64_t my_row = row0 + lane;
       if (my_row < row1) {
+        uint32_t dir8 = plan.dir_plane ? (uint32_t)plan.dir_plane[my_row]
+                                       : 0xFEu;
         uint32_t r = (uint32_t)(my_row - row0);
         uint64_t o0 = loffs[r], o1 = loffs[r + 1];
         const uint8_t *vp = bv + shift + (uint32_t)(o0 - gb);
         uint32_t vlen = (uint32_t)(o1 - o0);
         bool found = false, fnull = false, ok = true;
         int64_t fv = 0;
-        if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
+        if (dir8 == 0xFFu) {
+          /* directory: filter column absent -> default fill */
+        } else if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
           uintptr_t base = (uintptr_t)vp;
           uintptr_t wabs = ~(uintptr_t)0;
           uint64_t wlo = 0, whi = 0;
-          uint32_t posb = 0;
+          uint32_t posb = dir8 < 0xFEu ? dir8 : 0u;
           while (posb < vlen) {
             uintptr_t ua = base + posb;
             if (ua - wabs > 8) {
