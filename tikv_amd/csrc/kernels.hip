@@ -1976,13 +1976,13 @@ k_scan_fc_ring(ScanPlan plan,
         }
         inflight++;
       }
-      /* drain to <= 2 slots in flight (or everything on the tail) */
-      if (inflight > 2 || (!live && inflight)) {
+      /* drain to <= 3 slots in flight (or everything on the tail) */
+      if (inflight > 3 || (!live && inflight)) {
         if (live) {
-          asm volatile("s_waitcnt vmcnt(%0)" :: "n"(2 * KW) : "memory");
-          /* slot of round j-2 has landed */
-          uint32_t done_pos = (uint32_t)((j - 2) % DEPTH);
-          ready[done_pos] = (uint32_t)(j - 2 + 1);
+          asm volatile("s_waitcnt vmcnt(%0)" :: "n"(3 * KW) : "memory");
+          /* slot of round j-3 has landed */
+          uint32_t done_pos = (uint32_t)((j - 3) % DEPTH);
+          ready[done_pos] = (uint32_t)(j - 3 + 1);
           inflight--;
         } else {
           asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
