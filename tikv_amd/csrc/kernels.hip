@@ -1927,6 +1927,8 @@ k_scan_fc_ring(ScanPlan plan,
 
   const uint32_t wave = threadIdx.x >> 6, lane = threadIdx.x & 63u;
   const uint64_t n_slots = (n_rows + SLOT_ROWS - 1) / SLOT_ROWS;
+  /* loader in-flight depth in slots (vmcnt immediate caps at 63) */
+  constexpr int INF = (3 * KW <= 60) ? 3 : 2;
 
   /* init flags */
   if (threadIdx.x < DEPTH) {
@@ -1976,13 +1978,13 @@ k_scan_fc_ring(ScanPlan plan,
         }
         inflight++;
       }
-      /* drain to <= 3 slots in flight (or everything on the tail) */
-      if (inflight > 3 || (!live && inflight)) {
+      /* drain to <= INF slots in flight (or everything on the tail) */
+      if (inflight > (uint32_t)INF || (!live && inflight)) {
         if (live) {
-          asm volatile("s_waitcnt vmcnt(%0)" :: "n"(3 * KW) : "memory");
-          /* slot of round j-3 has landed */
-          uint32_t done_pos = (uint32_t)((j - 3) % DEPTH);
-          ready[done_pos] = (uint32_t)(j - 3 + 1);
+          asm volatile("s_waitcnt vmcnt(%0)" :: "n"(INF * KW) : "memory");
+          /* slot of round j-INF has landed */
+          uint32_t done_pos = (uint32_t)((j - INF) % DEPTH);
+          ready[done_pos] = (uint32_t)(j - INF + 1);
           inflight--;
         } else {
           asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
