@@ -194,3 +194,138 @@ def test_stream_agg_gpu_parity(engine):
             rgn.close()
     finally:
         g.close()
+
+
+# ---- bytes group keys (SlowHashAggregationImpl) ------------------------
+def var_bytes_cell(col_id, payload):
+    return (b"\x08" + var_i64(col_id) + b"\x02" + var_i64(len(payload))
+            + payload)
+
+
+def region_of_mixed(rows):
+    """rows: list of dicts {col_id: int | bytes | None}"""
+    keys = b"".join(row_key(i) for i in range(len(rows)))
+    vals = b""
+    vo = [0]
+    for r in rows:
+        v = b""
+        for cid in sorted(r):
+            x = r[cid]
+            if x is None:
+                v += cell_null(cid)
+            elif isinstance(x, bytes):
+                v += var_bytes_cell(cid, x)
+            else:
+                v += cell_int(cid, x)
+        vals += v
+        vo.append(len(vals))
+    ko = [19 * i for i in range(len(rows) + 1)]
+    kb = (C.c_uint8 * max(len(keys), 1)).from_buffer_copy(keys or b"\0")
+    vb = (C.c_uint8 * max(len(vals), 1)).from_buffer_copy(vals or b"\0")
+    return (C.cast(kb, C.POINTER(C.c_uint8)), (C.c_uint64 * len(ko))(*ko),
+            C.cast(vb, C.POINTER(C.c_uint8)), (C.c_uint64 * len(vo))(*vo),
+            len(rows), (kb, vb))
+
+
+def split_rows(data, n_cols):
+    """split a datum-encoded response into per-row byte strings (ints,
+    NILs and compact-bytes only)."""
+    rows = []
+    p = 0
+    while p < len(data):
+        start = p
+        for _ in range(n_cols):
+            f = data[p]
+            if f == 0:
+                p += 1
+            elif f in (3, 4, 5):
+                p += 9
+            elif f == 2:
+                q = p + 1
+                uv, sh = 0, 0
+                while True:
+                    b = data[q]
+                    q += 1
+                    uv |= (b & 0x7F) << sh
+                    sh += 7
+                    if not (b & 0x80):
+                        break
+                n = uv >> 1
+                p = q + n
+            else:
+                raise AssertionError("unexpected datum flag %d" % f)
+        rows.append(bytes(data[start:p]))
+    return rows
+
+
+def test_bytes_group_oracle():
+    orc = _orc()
+    rows = [{1: 1, 2: b"aa"}, {1: 2, 2: b"bb"}, {1: 3, 2: b"aa"},
+            {1: 4, 2: None}, {1: 5, 2: b"aa"}, {1: 6, 2: None},
+            {1: 7, 2: b""}]
+    k, ko, v, vo, n, keep = region_of_mixed(rows)
+    cols = [tikv_amd.Col(1), tikv_amd.Col(2, tp=F.TP_VARCHAR)]
+    req = (tikv_amd.DagSelect(cols)
+           .hash_agg([tikv_amd.count_star(), tikv_amd.max_col(0)],
+                     tikv_amd.Expr().col(1)).build())
+    data, nrows = orc.dag_run(req, k, ko, v, vo, n)
+    assert nrows == 4  # aa, bb, NULL, ""
+    got = set(split_rows(data, 3))
+    want = set()
+    for cnt, mx, key in ((3, 5, b"aa"), (1, 2, b"bb"), (2, 6, None),
+                         (1, 7, b"")):
+        row = b"\x03" + ((cnt ^ (1 << 63)).to_bytes(8, "big"))
+        row += b"\x03" + ((mx ^ (1 << 63)).to_bytes(8, "big"))
+        if key is None:
+            row += b"\x00"
+        else:
+            row += b"\x02" + var_i64(len(key)) + key
+        want.add(row)
+    assert got == want
+
+
+@pytest.mark.gpu
+def test_bytes_group_gpu_parity(engine):
+    orc = _orc()
+    # hand-built low-cardinality case (repeats + NULL + empty)
+    rows = []
+    keys = [b"k1", b"key-two", b"", None, b"k1", b"x" * 40]
+    for i in range(3000):
+        rows.append({1: i, 2: keys[i % len(keys)]})
+    k, ko, v, vo, n, keep = region_of_mixed(rows)
+    cols = [tikv_amd.Col(1), tikv_amd.Col(2, tp=F.TP_VARCHAR)]
+    req = (tikv_amd.DagSelect(cols)
+           .hash_agg([tikv_amd.count_star(), tikv_amd.max_col(0)],
+                     tikv_amd.Expr().col(1)).build())
+    od, orows = orc.dag_run(req, k, ko, v, vo, n)
+    import ctypes as CT
+    kb = CT.cast((CT.c_uint8 * max(ko[-1], 1)).from_buffer_copy(
+        CT.string_at(k, ko[-1]) or b"\0"), CT.POINTER(CT.c_uint8))
+    rgn = engine.region_raw(k, ko, v, vo, n)
+    try:
+        gd, grows, _ = engine.dag_run(req, [rgn])
+        assert grows == orows
+        assert set(split_rows(gd, 3)) == set(split_rows(od, 3))
+    finally:
+        rgn.close()
+    # generated cfg2 region: random varbytes (mostly distinct keys) + filter
+    g = tikv_amd.GenRegion(config_index=2, n_rows=60001, table_id=5)
+    try:
+        rgn2 = engine.region(g)
+        try:
+            cols2 = [tikv_amd.Col(1),
+                     tikv_amd.Col(2, tp=F.TP_NEWDECIMAL, decimal=2),
+                     tikv_amd.Col(3, tp=F.TP_VARCHAR)]
+            sel = tikv_amd.cmp_col_const(0, F.SIG_GT_INT, 0)
+            req2 = (tikv_amd.DagSelect(cols2).where(sel)
+                    .hash_agg([tikv_amd.count_star(), tikv_amd.max_col(0)],
+                              tikv_amd.Expr().col(2)).build())
+            gd2, gr2, _ = engine.dag_run(req2, [rgn2])
+            od2, or2 = orc.dag_run(req2, g.keys, g.key_offs, g.vals,
+                                   g.val_offs, g.n_kv)
+            assert gr2 == or2
+            assert set(split_rows(gd2, 3)) == set(split_rows(od2, 3))
+        finally:
+            rgn2.close()
+    finally:
+        g.close()

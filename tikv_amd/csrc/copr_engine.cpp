@@ -367,7 +367,8 @@ static void wire_celldir(ScanPlan *sp, const DevRegion &dev) {
   bool ok = true;
   if (sp->has_filter && (sp->filter_col_id < 1 || sp->filter_col_id > 16))
     ok = false;
-  if ((sp->mode == 2 || sp->mode == 3 || sp->group_col_id != 0) &&
+  if ((sp->mode == 2 || sp->mode == 3 || sp->mode == 4 ||
+       sp->group_col_id != 0) &&
       (sp->group_col_id < 1 || sp->group_col_id > 16))
     ok = false;
   for (int a = 0; a < sp->n_aggs && ok; a++)
@@ -568,8 +569,17 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
     size_t off = (size_t)ge.nodes[0].i64_val;
     if (off >= pl->cols.size()) return SET_ERR(COPR_ERR_INVALID_REQUEST, "bad group offset");
     const CoprColumnInfo &ci = pl->cols[off];
-    if (!et_int(ci.ft.tp) || (ci.pk_handle && !sp.index_mode))
+    bool grp_bytes = ci.ft.tp == COPR_TP_VARCHAR || ci.ft.tp == COPR_TP_STRING ||
+                     ci.ft.tp == COPR_TP_VARSTRING || ci.ft.tp == COPR_TP_BLOB;
+    if ((!et_int(ci.ft.tp) && !grp_bytes) || (ci.pk_handle && !sp.index_mode))
       return SET_ERR(COPR_ERR_UNSUPPORTED, "group-by type not yet native");
+    if (grp_bytes) {
+      /* SlowHashAggregationImpl: encoded-bytes group keys
+         (slow_hash_aggr_executor.rs:220,285) */
+      if (sp.index_mode)
+        return SET_ERR(COPR_ERR_UNSUPPORTED, "bytes group over index scan");
+      sp.mode = 4;
+    }
     sp.group_col_id = ci.column_id;
     sp.group_col_unsigned = (ci.ft.flag & COPR_FLAG_UNSIGNED) ? 1 : 0;
     pl->group_ft = ci.ft;
@@ -864,6 +874,59 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
       std::vector<std::vector<uint8_t>> cols(n_out_cols);
       encode_agg_row(pl, &h_accs[g * sp.n_aggs], true, h_gs[g] == 1, h_gk[g],
                      &cols);
+      for (uint32_t oo = 0; oo < req->n_output_offsets; oo++) {
+        uint32_t off = req->output_offsets[oo];
+        if (off < n_out_cols)
+          resp.insert(resp.end(), cols[off].begin(), cols[off].end());
+      }
+      n_rows_out++;
+    }
+  } else if (pl.sp.mode == 4) {
+    /* ---- bytes-key hash agg ---- */
+    if (n_regions != 1)
+      return SET_ERR(COPR_ERR_UNSUPPORTED,
+                     "bytes hash agg supports one region per request");
+    copr_region *r = regions[0];
+    ScanPlan sp = pl.sp;
+    wire_celldir(&sp, r->dev);
+    pick_tiling(r->dev, &sp, /*force_nopipe=*/true);
+    std::vector<SimpleAggAcc> h_accs;
+    std::vector<uint64_t> h_kofs;
+    std::vector<uint32_t> h_klen;
+    std::vector<uint8_t> h_kst;
+    hipEventRecord(ev_a, eng->stream);
+    int n_seg = dev_bytes_agg(sp, r->dev, eng->stream, &h_accs, &h_kofs,
+                              &h_klen, &h_kst);
+    hipEventRecord(ev_b, eng->stream);
+    timed = true;
+    if (n_seg == -3)
+      return SET_ERR(COPR_ERR_STORAGE, "row parse error on device");
+    if (n_seg == -2) return SET_ERR(COPR_ERR_OOM, "bytes agg temp alloc");
+    if (n_seg < 0) return SET_ERR(COPR_ERR_INTERNAL, "bytes agg failed");
+    size_t n_out_cols = pl.out_schema.size();
+    uint64_t take = (uint64_t)n_seg < pl.limit ? (uint64_t)n_seg : pl.limit;
+    std::vector<uint8_t> keybuf;
+    for (uint64_t g = 0; g < take; g++) {
+      std::vector<std::vector<uint8_t>> cols(n_out_cols);
+      encode_agg_row(pl, &h_accs[g * sp.n_aggs], false, false, 0, &cols);
+      std::vector<uint8_t> &gcol = cols[n_out_cols - 1];
+      if (h_kst[g] == 1) {
+        gcol.push_back(0);                       /* NIL */
+      } else {
+        /* COMPACT_BYTES datum of the group payload (datum_codec.rs:290-294) */
+        gcol.push_back(2);
+        uint64_t uv = ((uint64_t)h_klen[g]) << 1;   /* zigzag, non-negative */
+        while (uv >= 0x80) { gcol.push_back((uint8_t)(uv | 0x80)); uv >>= 7; }
+        gcol.push_back((uint8_t)uv);
+        keybuf.resize(h_klen[g]);
+        if (h_klen[g]) {
+          hipError_t ce2 = hipMemcpy(keybuf.data(), r->dev.d_vals + h_kofs[g],
+                                     h_klen[g], hipMemcpyDeviceToHost);
+          if (ce2 != hipSuccess)
+            return SET_ERR(COPR_ERR_INTERNAL, "group key readback");
+        }
+        gcol.insert(gcol.end(), keybuf.begin(), keybuf.end());
+      }
       for (uint32_t oo = 0; oo < req->n_output_offsets; oo++) {
         uint32_t off = req->output_offsets[oo];
         if (off < n_out_cols)

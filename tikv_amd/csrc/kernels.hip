@@ -475,7 +475,8 @@ __device__ static inline bool d_v2_collect(const ScanPlan &plan,
                                            bool *filt_found, bool *filt_null,
                                            int64_t *filt_v, bool *grp_found,
                                            bool *grp_null, int64_t *grp_v,
-                                           AggColView (&cols)[NAGGS]) {
+                                           AggColView (&cols)[NAGGS],
+                                           bool parse_grp = true) {
   V2Row r;
   if (!d_v2_parse(vp, vlen, &r)) return false;
   if (plan.has_filter) {
@@ -488,7 +489,7 @@ __device__ static inline bool d_v2_collect(const ScanPlan &plan,
         return false;
     }
   }
-  if (IS_HASH) {
+  if (IS_HASH && parse_grp) {
     uint32_t s, e;
     int st = d_v2_find(r, plan.group_col_id, &s, &e);
     if (st >= 0) {
@@ -2789,7 +2790,37 @@ struct ExtractOut {
   int64_t *gkey;
   int64_t *av;       /* [NAGGS][n] contribution values */
   uint8_t *as_;      /* [NAGGS][n] 1 = contributes */
+  /* bytes group keys (SlowHashAggregationImpl, slow_hash_aggr_executor.rs:
+     220,285): the group column's DECODED payload span in the region's
+     value stream + a 64-bit FNV-1a fingerprint (grouping is verified
+     byte-exact downstream, the hash only orders candidates) */
+  uint64_t *ghash;
+  uint64_t *gofs;
+  uint32_t *glen;
 };
+
+__device__ static inline uint64_t d_fnv1a(const uint8_t *p, uint32_t n) {
+  uint64_t h = 1469598103934665603ull;
+  for (uint32_t i = 0; i < n; i++) h = (h ^ p[i]) * 1099511628211ull;
+  return h;
+}
+
+/* payload span of a COMPACT_BYTES datum at cell_off (flag 2 + varint len +
+   raw payload; byte.rs:518-530). Returns false for other encodings. */
+__device__ static inline bool d_compact_bytes_span(const uint8_t *vp,
+                                                   uint32_t vlen,
+                                                   uint32_t cell_off,
+                                                   uint32_t *pofs,
+                                                   uint32_t *plen) {
+  if (cell_off >= vlen || vp[cell_off] != 2) return false;
+  int64_t n;
+  uint32_t nb;
+  if (!d_var_i64(vp + cell_off + 1, vlen - cell_off - 1, &n, &nb)) return false;
+  if (n < 0 || cell_off + 1 + nb + (uint64_t)n > vlen) return false;
+  *pofs = cell_off + 1 + nb;
+  *plen = (uint32_t)n;
+  return true;
+}
 
 template <int NAGGS>
 __global__ void __launch_bounds__(256)
@@ -2814,6 +2845,8 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
       const uint8_t *vp = lds + shift + (uint32_t)(val_offs[my_row] - gbase);
       uint32_t vlen = (uint32_t)(val_offs[my_row + 1] - val_offs[my_row]);
       bool parse_ok = true;
+      const bool GBYTES = eo.ghash != nullptr;
+      uint64_t g_h = 0, g_o = 0; uint32_t g_l = 0;
       bool filt_found = false, filt_null = false; int64_t filt_v = 0;
       bool grp_found = false, grp_null = false; int64_t grp_v = 0;
       AggColView cols[NAGGS > 0 ? NAGGS : 1];
@@ -2858,7 +2891,14 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
                   cid == plan.group_col_id) {
                 grp_found = true;
                 if (cell.is_null) grp_null = true;
-                else if (cell.has_int) grp_v = cell.ival;
+                else if (GBYTES) {
+                  uint32_t po, plen2;
+                  if (d_compact_bytes_span(vp, vlen, coff, &po, &plen2)) {
+                    g_h = d_fnv1a(vp + po, plen2);
+                    g_o = val_offs[my_row] + po;
+                    g_l = plen2;
+                  } else parse_ok = false;
+                } else if (cell.has_int) grp_v = cell.ival;
                 else parse_ok = false;
               } else parse_ok = false;
             }
@@ -2884,7 +2924,25 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
         } else if (vp[0] == 128) {
           parse_ok = d_v2_collect<(NAGGS > 0 ? NAGGS : 1), true>(
               plan, vp, vlen, &filt_found, &filt_null, &filt_v, &grp_found,
-              &grp_null, &grp_v, cols);
+              &grp_null, &grp_v, cols, /*parse_grp=*/!GBYTES);
+          if (parse_ok && GBYTES) {
+            V2Row r2;
+            if (d_v2_parse(vp, vlen, &r2)) {
+              uint32_t s2, e2;
+              int vst = d_v2_find(r2, plan.group_col_id, &s2, &e2);
+              if (vst >= 0) {
+                grp_found = true;
+                if (vst == 0) grp_null = true;
+                else {
+                  /* v2 cells hold the raw payload already */
+                  const uint8_t *pp = r2.vals + s2;
+                  g_h = d_fnv1a(pp, e2 - s2);
+                  g_o = val_offs[my_row] + (uint32_t)(pp - vp);
+                  g_l = e2 - s2;
+                }
+              }
+            } else parse_ok = false;
+          }
         } else {
           uint32_t pos = 0;
           int needed = (plan.has_filter ? 1 : 0) + 1;
@@ -2911,7 +2969,14 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
             if (!grp_found && cell_id == plan.group_col_id) {
               grp_found = true;
               if (cell.is_null) grp_null = true;
-              else if (cell.has_int) grp_v = cell.ival;
+              else if (GBYTES) {
+                uint32_t po, plen2;
+                if (d_compact_bytes_span(vp, vlen, cell_off, &po, &plen2)) {
+                  g_h = d_fnv1a(vp + po, plen2);
+                  g_o = val_offs[my_row] + po;
+                  g_l = plen2;
+                } else parse_ok = false;
+              } else if (cell.has_int) grp_v = cell.ival;
               else parse_ok = false;
               found++;
             }
@@ -2943,6 +3008,11 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
       }
       eo.st[my_row] = s;
       eo.gkey[my_row] = grp_v;
+      if (GBYTES) {
+        eo.ghash[my_row] = s == 2 ? g_h : 0;
+        eo.gofs[my_row] = g_o;
+        eo.glen[my_row] = s == 2 ? g_l : 0;
+      }
       #pragma unroll
       for (int a = 0; a < NAGGS; a++) {
         const DevAggSpec &sp = plan.aggs[a];
@@ -3119,7 +3189,7 @@ int dev_stream_agg(const ScanPlan &plan, const DevRegion &rgn, void *stream,
   if (e == hipSuccess) e = hipMalloc(&d_err, 4);
   if (e == hipSuccess) e = hipMemsetAsync(d_err, 0, 4, s);
   if (e != hipSuccess) { freeall(); return -2; }
-  ExtractOut eo{st, gkey, av, as_};
+  ExtractOut eo{st, gkey, av, as_, nullptr, nullptr, nullptr};
   if (extract_launch(plan, rgn, eo, d_err, s)) { freeall(); return -1; }
   unsigned int h_err = 0;
   e = hipMemcpyAsync(&h_err, d_err, 4, hipMemcpyDeviceToHost, s);
@@ -3232,7 +3302,7 @@ int dev_topn_select(const ScanPlan &plan, const DevRegion &rgn,
   if (e == hipSuccess) e = hipMalloc(&d_err, 4);
   if (e == hipSuccess) e = hipMemsetAsync(d_err, 0, 4, s);
   if (e != hipSuccess) { freeall(); return -2; }
-  ExtractOut eo{st, gkey, (int64_t *)cav, cas};
+  ExtractOut eo{st, gkey, (int64_t *)cav, cas, nullptr, nullptr, nullptr};
   ScanPlan p0 = plan;
   p0.n_aggs = 0;
   if (extract_launch(p0, rgn, eo, d_err, s)) { freeall(); return -1; }
@@ -3336,6 +3406,230 @@ int dev_topn_select(const ScanPlan &plan, const DevRegion &rgn,
     }
   }
   return 0;
+}
+
+
+/* ---- bytes-group hash agg pipeline (SlowHashAggregationImpl) ---- */
+__global__ static void k_compact_bytes(const uint8_t *st, const uint64_t *gh,
+                                       const uint64_t *go, const uint32_t *gl,
+                                       const int64_t *av, const uint8_t *as_,
+                                       const uint64_t *pos, uint64_t n,
+                                       uint64_t m, int n_aggs, uint64_t *ch,
+                                       uint64_t *co, uint32_t *cl, uint8_t *cs,
+                                       int64_t *cav, uint8_t *cas,
+                                       uint32_t *idx) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n || !st[i]) return;
+  uint64_t p = pos[i];
+  ch[p] = gh[i];
+  co[p] = go[i];
+  cl[p] = gl[i];
+  cs[p] = st[i];
+  idx[p] = (uint32_t)p;
+  for (int a = 0; a < n_aggs; a++) {
+    cav[(uint64_t)a * m + p] = av[(uint64_t)a * n + i];
+    cas[(uint64_t)a * m + p] = as_[(uint64_t)a * n + i];
+  }
+}
+
+/* boundary flags over the hash-sorted permutation: a new group starts when
+ * state/hash differ, or the hash ties but the payload bytes differ
+ * (grouping stays byte-exact; the hash only clusters candidates) */
+__global__ static void k_bytes_boundary(const uint64_t *ch, const uint64_t *co,
+                                        const uint32_t *cl, const uint8_t *cs,
+                                        const uint32_t *perm,
+                                        const uint8_t *vals, uint64_t m,
+                                        uint32_t *f) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= m) return;
+  if (i == 0) { f[0] = 1; return; }
+  uint32_t a = perm[i], b = perm[i - 1];
+  uint32_t nf = 0;
+  if (cs[a] != cs[b]) nf = 1;
+  else if (cs[a] == 2) {
+    if (ch[a] != ch[b] || cl[a] != cl[b]) nf = 1;
+    else {
+      const uint8_t *pa = vals + co[a], *pb = vals + co[b];
+      for (uint32_t t = 0; t < cl[a]; t++)
+        if (pa[t] != pb[t]) { nf = 1; break; }
+    }
+  }
+  f[i] = nf;
+}
+
+__global__ static void k_bytes_update(const uint64_t *co, const uint32_t *cl,
+                                      const uint8_t *cs, const int64_t *cav,
+                                      const uint8_t *cas,
+                                      const uint32_t *perm,
+                                      const uint32_t *segid, uint64_t m,
+                                      ScanPlan plan, SimpleAggAcc *accs,
+                                      uint64_t *rofs, uint32_t *rlen,
+                                      uint8_t *rst) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= m) return;
+  uint32_t seg = segid[i] - 1;
+  uint32_t a0 = perm[i];
+  if (i == 0 || segid[i] != segid[i - 1]) {
+    rofs[seg] = co[a0];
+    rlen[seg] = cl[a0];
+    rst[seg] = cs[a0];
+  }
+  SimpleAggAcc *base = accs + (uint64_t)seg * plan.n_aggs;
+  for (int a = 0; a < plan.n_aggs; a++) {
+    if (!cas[(uint64_t)a * m + a0]) continue;
+    const DevAggSpec &sp = plan.aggs[a];
+    int64_t v = cav[(uint64_t)a * m + a0];
+    atomicAdd(&base[a].cnt, 1ull);
+    if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC)
+      atomic_add_i128(&base[a].sum_lo, &base[a].sum_hi, v);
+    else if (d_is_fold(sp.kind)) {
+      unsigned long long b = d_fold_xform(sp.kind, v, sp.col_unsigned);
+      if (d_is_xor(sp.kind)) atomicXor(&base[a].sum_lo, b);
+      else if (sp.kind == DAGG_MAX_INT || sp.kind == DAGG_MIN_INT)
+        atomicMax(&base[a].sum_lo, b);
+      else atomicOr(&base[a].sum_lo, b);
+    }
+  }
+}
+
+/* Bytes-key hash aggregation. Returns n_groups (>=0), -1 internal, -2 oom,
+ * -3 parse error. Group keys returned as (ofs,len) spans into the region's
+ * value stream + the representative state byte (1 = NULL group). */
+int dev_bytes_agg(const ScanPlan &plan, const DevRegion &rgn, void *stream,
+                  std::vector<SimpleAggAcc> *h_accs,
+                  std::vector<uint64_t> *h_kofs, std::vector<uint32_t> *h_klen,
+                  std::vector<uint8_t> *h_kst) {
+  hipStream_t s = (hipStream_t)stream;
+  uint64_t n = rgn.n_kv;
+  if (!n) return 0;
+  uint8_t *st = nullptr, *as_ = nullptr, *cs = nullptr, *cas = nullptr;
+  uint8_t *rst = nullptr;
+  int64_t *gkey = nullptr, *av = nullptr, *cav = nullptr;
+  uint64_t *gh = nullptr, *go = nullptr, *ch = nullptr, *co = nullptr;
+  uint64_t *ch2 = nullptr, *rofs = nullptr, *pos = nullptr;
+  uint32_t *gl = nullptr, *cl = nullptr, *f32 = nullptr, *segid = nullptr;
+  uint32_t *idx = nullptr, *perm = nullptr, *rlen = nullptr;
+  unsigned int *d_err = nullptr;
+  SimpleAggAcc *accs = nullptr;
+  void *tmp = nullptr;
+  size_t tmpb = 0;
+  auto freeall = [&]() {
+    hipFree(st); hipFree(as_); hipFree(cs); hipFree(cas); hipFree(rst);
+    hipFree(gkey); hipFree(av); hipFree(cav);
+    hipFree(gh); hipFree(go); hipFree(ch); hipFree(co); hipFree(ch2);
+    hipFree(rofs); hipFree(pos); hipFree(gl); hipFree(cl); hipFree(f32);
+    hipFree(segid); hipFree(idx); hipFree(perm); hipFree(rlen);
+    hipFree(d_err); hipFree(accs); hipFree(tmp);
+  };
+  int na = plan.n_aggs;
+  hipError_t e = hipSuccess;
+  if (e == hipSuccess) e = hipMalloc(&st, n);
+  if (e == hipSuccess) e = hipMalloc(&gkey, n * 8);
+  if (e == hipSuccess) e = hipMalloc(&gh, n * 8);
+  if (e == hipSuccess) e = hipMalloc(&go, n * 8);
+  if (e == hipSuccess) e = hipMalloc(&gl, n * 4);
+  if (e == hipSuccess) e = hipMalloc(&av, (uint64_t)na * n * 8 + 8);
+  if (e == hipSuccess) e = hipMalloc(&as_, (uint64_t)na * n + 8);
+  if (e == hipSuccess) e = hipMalloc(&d_err, 4);
+  if (e == hipSuccess) e = hipMemsetAsync(d_err, 0, 4, s);
+  if (e != hipSuccess) { freeall(); return -2; }
+  ExtractOut eo{st, gkey, av, as_, gh, go, gl};
+  if (extract_launch(plan, rgn, eo, d_err, s)) { freeall(); return -1; }
+  unsigned int h_err = 0;
+  e = hipMemcpyAsync(&h_err, d_err, 4, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  if (e != hipSuccess) { freeall(); return -1; }
+  if (h_err) { freeall(); return -3; }
+
+  uint32_t blocks = (uint32_t)((n + 255) / 256);
+  if (e == hipSuccess) e = hipMalloc(&f32, n * 4 + 4);
+  if (e == hipSuccess) e = hipMalloc(&pos, n * 8 + 8);
+  if (e != hipSuccess) { freeall(); return -2; }
+  hipLaunchKernelGGL(k_st_keep_flags, dim3(blocks), dim3(256), 0, s, st, f32,
+                     n, 1, 2);
+  hipcub::DeviceScan::ExclusiveSum(nullptr, tmpb, f32, pos, (int)n, s);
+  if (hipMalloc(&tmp, tmpb) != hipSuccess) { freeall(); return -2; }
+  hipcub::DeviceScan::ExclusiveSum(tmp, tmpb, f32, pos, (int)n, s);
+  uint64_t m = 0;
+  uint32_t lf = 0;
+  e = hipMemcpyAsync(&m, pos + (n - 1), 8, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess)
+    e = hipMemcpyAsync(&lf, f32 + (n - 1), 4, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  if (e != hipSuccess) { freeall(); return -1; }
+  m += lf;
+  if (!m) { freeall(); return 0; }
+
+  if (e == hipSuccess) e = hipMalloc(&ch, m * 8);
+  if (e == hipSuccess) e = hipMalloc(&ch2, m * 8);
+  if (e == hipSuccess) e = hipMalloc(&co, m * 8);
+  if (e == hipSuccess) e = hipMalloc(&cl, m * 4);
+  if (e == hipSuccess) e = hipMalloc(&cs, m);
+  if (e == hipSuccess) e = hipMalloc(&idx, m * 4);
+  if (e == hipSuccess) e = hipMalloc(&perm, m * 4);
+  if (e == hipSuccess) e = hipMalloc(&cav, (uint64_t)na * m * 8 + 8);
+  if (e == hipSuccess) e = hipMalloc(&cas, (uint64_t)na * m + 8);
+  if (e != hipSuccess) { freeall(); return -2; }
+  hipLaunchKernelGGL(k_compact_bytes, dim3(blocks), dim3(256), 0, s, st, gh,
+                     go, gl, av, as_, pos, n, m, na, ch, co, cl, cs, cav, cas,
+                     idx);
+  size_t tb = 0;
+  hipcub::DeviceRadixSort::SortPairs(nullptr, tb, ch, ch2, idx, perm, (int)m,
+                                     0, 64, s);
+  if (tb > tmpb) {
+    hipFree(tmp); tmp = nullptr;
+    if (hipMalloc(&tmp, tb) != hipSuccess) { freeall(); return -2; }
+    tmpb = tb;
+  }
+  hipcub::DeviceRadixSort::SortPairs(tmp, tmpb, ch, ch2, idx, perm, (int)m, 0,
+                                     64, s);
+  uint32_t mblocks = (uint32_t)((m + 255) / 256);
+  if (hipMalloc(&segid, m * 4) != hipSuccess) { freeall(); return -2; }
+  hipLaunchKernelGGL(k_bytes_boundary, dim3(mblocks), dim3(256), 0, s, ch, co,
+                     cl, cs, perm, rgn.d_vals, m, f32);
+  size_t tb2 = 0;
+  hipcub::DeviceScan::InclusiveSum(nullptr, tb2, f32, segid, (int)m, s);
+  if (tb2 > tmpb) {
+    hipFree(tmp); tmp = nullptr;
+    if (hipMalloc(&tmp, tb2) != hipSuccess) { freeall(); return -2; }
+    tmpb = tb2;
+  }
+  hipcub::DeviceScan::InclusiveSum(tmp, tmpb, f32, segid, (int)m, s);
+  uint32_t n_seg = 0;
+  e = hipMemcpyAsync(&n_seg, segid + (m - 1), 4, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  if (e != hipSuccess) { freeall(); return -1; }
+
+  if (e == hipSuccess)
+    e = hipMalloc(&accs, (uint64_t)n_seg * na * sizeof(SimpleAggAcc));
+  if (e == hipSuccess) e = hipMalloc(&rofs, (uint64_t)n_seg * 8);
+  if (e == hipSuccess) e = hipMalloc(&rlen, (uint64_t)n_seg * 4);
+  if (e == hipSuccess) e = hipMalloc(&rst, n_seg);
+  if (e == hipSuccess)
+    e = hipMemsetAsync(accs, 0, (uint64_t)n_seg * na * sizeof(SimpleAggAcc), s);
+  if (e != hipSuccess) { freeall(); return -2; }
+  hipLaunchKernelGGL(k_bytes_update, dim3(mblocks), dim3(256), 0, s, co, cl,
+                     cs, cav, cas, perm, segid, m, plan, accs, rofs, rlen,
+                     rst);
+  h_accs->resize((size_t)n_seg * na);
+  h_kofs->resize(n_seg);
+  h_klen->resize(n_seg);
+  h_kst->resize(n_seg);
+  e = hipMemcpyAsync(h_accs->data(), accs,
+                     (uint64_t)n_seg * na * sizeof(SimpleAggAcc),
+                     hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess)
+    e = hipMemcpyAsync(h_kofs->data(), rofs, (uint64_t)n_seg * 8,
+                       hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess)
+    e = hipMemcpyAsync(h_klen->data(), rlen, (uint64_t)n_seg * 4,
+                       hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess)
+    e = hipMemcpyAsync(h_kst->data(), rst, n_seg, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  freeall();
+  if (e != hipSuccess) return -1;
+  return (int)n_seg;
 }
 
 /* gather the given rows (in order) of a region into a new DevRegion */

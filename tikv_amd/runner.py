@@ -310,6 +310,16 @@ class Engine:
                                (st, self._lib.copr_last_error().decode()))
         return Region(self, r)
 
+    def region_raw(self, keys, key_offs, vals, val_offs, n_kv):
+        """Region from explicit host buffers (ctypes pointers/arrays)."""
+        r = C.c_void_p()
+        st = self._lib.copr_region_create(self._h, keys, key_offs, vals,
+                                          val_offs, n_kv, C.byref(r))
+        if st != 0:
+            raise RuntimeError("copr_region_create: %d (%s)" %
+                               (st, self._lib.copr_last_error().decode()))
+        return Region(self, r)
+
     def region_mvcc(self, gen: GenRegion, read_ts):
         """Build a visible-row region from a raw write-CF stream (the
         device MVCC version filter)."""
