@@ -905,6 +905,15 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
     if (n_seg < 0) return SET_ERR(COPR_ERR_INTERNAL, "bytes agg failed");
     size_t n_out_cols = pl.out_schema.size();
     uint64_t take = (uint64_t)n_seg < pl.limit ? (uint64_t)n_seg : pl.limit;
+    /* one bulk readback of the value stream beats per-group copies when the
+       region is small or the group count is large */
+    std::vector<uint8_t> all_vals;
+    if (n_seg > 64 && r->dev.val_bytes <= (256ull << 20)) {
+      all_vals.resize(r->dev.val_bytes);
+      if (hipMemcpy(all_vals.data(), r->dev.d_vals, r->dev.val_bytes,
+                    hipMemcpyDeviceToHost) != hipSuccess)
+        return SET_ERR(COPR_ERR_INTERNAL, "vals readback");
+    }
     std::vector<uint8_t> keybuf;
     for (uint64_t g = 0; g < take; g++) {
       std::vector<std::vector<uint8_t>> cols(n_out_cols);
@@ -918,14 +927,20 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
         uint64_t uv = ((uint64_t)h_klen[g]) << 1;   /* zigzag, non-negative */
         while (uv >= 0x80) { gcol.push_back((uint8_t)(uv | 0x80)); uv >>= 7; }
         gcol.push_back((uint8_t)uv);
-        keybuf.resize(h_klen[g]);
-        if (h_klen[g]) {
-          hipError_t ce2 = hipMemcpy(keybuf.data(), r->dev.d_vals + h_kofs[g],
-                                     h_klen[g], hipMemcpyDeviceToHost);
-          if (ce2 != hipSuccess)
-            return SET_ERR(COPR_ERR_INTERNAL, "group key readback");
+        if (!all_vals.empty()) {
+          gcol.insert(gcol.end(), all_vals.data() + h_kofs[g],
+                      all_vals.data() + h_kofs[g] + h_klen[g]);
+        } else {
+          keybuf.resize(h_klen[g]);
+          if (h_klen[g]) {
+            hipError_t ce2 = hipMemcpy(keybuf.data(),
+                                       r->dev.d_vals + h_kofs[g], h_klen[g],
+                                       hipMemcpyDeviceToHost);
+            if (ce2 != hipSuccess)
+              return SET_ERR(COPR_ERR_INTERNAL, "group key readback");
+          }
+          gcol.insert(gcol.end(), keybuf.begin(), keybuf.end());
         }
-        gcol.insert(gcol.end(), keybuf.begin(), keybuf.end());
       }
       for (uint32_t oo = 0; oo < req->n_output_offsets; oo++) {
         uint32_t off = req->output_offsets[oo];
