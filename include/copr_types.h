@@ -189,6 +189,14 @@ typedef struct CoprDagRequest {
   int32_t  div_precision_increment;
   /* paging (runner.rs:92-126): 0 = disabled */
   uint64_t paging_size;
+  /* response encoding (tipb EncodeType, runner.rs:1188-1225):
+     0 = TypeDefault (datum rows), 1 = TypeChunk (column-oriented chunks:
+     per batch, per output column: u32le length, u32le null_cnt,
+     [bitmap ceil(len/8) if null_cnt>0, bit set = NOT NULL],
+     [i64le var_offsets x (len+1) for var-size], data -- chunk/column.rs:
+     1052-1071; ints 8B LE, decimals the 40B struct dump
+     decimal.rs:2135-2142, bytes raw payload) */
+  int32_t encode_type;
 } CoprDagRequest;
 
 /* ---- results ---- */

@@ -757,6 +757,182 @@ struct GroupKeyHash {
   }
 };
 
+
+/* ---------------- TypeChunk response encoding ----------------
+ * chunk/column.rs:41-71 (container choice by field type), :1052-1071
+ * (wire layout), decimal.rs:2135-2142 (decimal = 40 B struct dump),
+ * runner.rs:1188-1225 (one Chunk per executor batch). The pipeline always
+ * produces TypeDefault datum rows internally; this post-pass re-encodes
+ * them column-wise per chunk. */
+struct ChunkCol {
+  bool fixed = true;
+  uint32_t flen = 8;
+  uint32_t length = 0, null_cnt = 0;
+  std::vector<uint8_t> bitmap, data;
+  std::vector<int64_t> offsets{0};
+  bool init(const CoprFieldType &ft) {
+    switch (ft.tp) {
+      case COPR_TP_TINY: case COPR_TP_SHORT: case COPR_TP_INT24:
+      case COPR_TP_LONG: case COPR_TP_LONGLONG: case COPR_TP_YEAR:
+      case COPR_TP_DOUBLE: case COPR_TP_DURATION:
+        fixed = true; flen = 8; return true;
+      case COPR_TP_NEWDECIMAL:
+        fixed = true; flen = 40; return true;
+      case COPR_TP_VARCHAR: case COPR_TP_STRING: case COPR_TP_VARSTRING:
+      case COPR_TP_BLOB:
+        fixed = false; return true;
+      default:
+        return false;
+    }
+  }
+  void bit(bool notnull) {
+    if ((length & 7) == 0) bitmap.push_back(0);
+    if (notnull) bitmap[length >> 3] |= (uint8_t)(1u << (length & 7));
+    else null_cnt++;
+  }
+  void app_null() {
+    bit(false);
+    if (fixed) data.insert(data.end(), flen, 0);
+    else offsets.push_back(offsets.back());
+    length++;
+  }
+  void app_fixed(const uint8_t *p, uint32_t n) {
+    bit(true);
+    data.insert(data.end(), p, p + n);
+    if (n < flen) data.insert(data.end(), flen - n, 0);
+    length++;
+  }
+  void app_var(const uint8_t *p, size_t n) {
+    bit(true);
+    data.insert(data.end(), p, p + n);
+    offsets.push_back(offsets.back() + (int64_t)n);
+    length++;
+  }
+  void flush(std::vector<uint8_t> *out) {
+    uint8_t w[8];
+    auto u32le = [&](uint32_t v) {
+      w[0] = v; w[1] = v >> 8; w[2] = v >> 16; w[3] = v >> 24;
+      out->insert(out->end(), w, w + 4);
+    };
+    u32le(length);
+    u32le(null_cnt);
+    if (null_cnt > 0)
+      out->insert(out->end(), bitmap.begin(), bitmap.end());
+    if (!fixed) {
+      for (int64_t v : offsets) {
+        uint64_t u = (uint64_t)v;
+        for (int i = 0; i < 8; i++) w[i] = (uint8_t)(u >> (8 * i));
+        out->insert(out->end(), w, w + 8);
+      }
+    }
+    out->insert(out->end(), data.begin(), data.end());
+  }
+};
+
+/* parse ONE datum at p and append it to the chunk column; returns bytes
+   consumed, 0 on error (datum.rs flags; from_raw_datums semantics) */
+static size_t chunk_append_datum(ChunkCol *c, const uint8_t *p, size_t rem) {
+  if (!rem) return 0;
+  uint8_t flag = p[0];
+  uint8_t tmp[8];
+  switch (flag) {
+    case 0:
+      c->app_null();
+      return 1;
+    case 3: case 4: {                    /* comparable int/uint */
+      if (rem < 9) return 0;
+      uint64_t u = 0;
+      for (int i = 0; i < 8; i++) u = (u << 8) | p[1 + i];
+      if (flag == 3) u ^= 0x8000000000000000ull;
+      for (int i = 0; i < 8; i++) tmp[i] = (uint8_t)(u >> (8 * i));
+      c->app_fixed(tmp, 8);
+      return 9;
+    }
+    case 5: {                            /* comparable f64 */
+      if (rem < 9) return 0;
+      uint64_t u = 0;
+      for (int i = 0; i < 8; i++) u = (u << 8) | p[1 + i];
+      if (u & 0x8000000000000000ull) u &= 0x7FFFFFFFFFFFFFFFull;
+      else u = ~u;
+      for (int i = 0; i < 8; i++) tmp[i] = (uint8_t)(u >> (8 * i));
+      c->app_fixed(tmp, 8);
+      return 9;
+    }
+    case 8: case 9: {                    /* var int / var uint */
+      uint64_t uv = 0;
+      size_t n = 0;
+      if (!decode_var_u64(p + 1, rem - 1, &uv, &n)) return 0;
+      uint64_t u = uv;
+      if (flag == 8) {
+        uint64_t half = uv >> 1;
+        u = (uv & 1) ? ~half : half;
+      }
+      for (int i = 0; i < 8; i++) tmp[i] = (uint8_t)(u >> (8 * i));
+      c->app_fixed(tmp, 8);
+      return 1 + n;
+    }
+    case 6: {                            /* decimal -> 40 B struct dump */
+      Decimal d;
+      size_t used = 0;
+      if (!dec_decode(p + 1, rem - 1, &d, &used)) return 0;
+      uint8_t buf[40];
+      buf[0] = d.int_cnt; buf[1] = d.frac_cnt; buf[2] = d.result_frac_cnt;
+      buf[3] = d.negative ? 1 : 0;
+      for (int i = 0; i < 9; i++) {
+        uint32_t w = d.word_buf[i];
+        buf[4 + 4 * i] = (uint8_t)w;
+        buf[5 + 4 * i] = (uint8_t)(w >> 8);
+        buf[6 + 4 * i] = (uint8_t)(w >> 16);
+        buf[7 + 4 * i] = (uint8_t)(w >> 24);
+      }
+      c->app_fixed(buf, 40);
+      return 1 + used;
+    }
+    case 2: {                            /* compact bytes */
+      int64_t n;
+      size_t nb;
+      if (!decode_var_i64(p + 1, rem - 1, &n, &nb)) return 0;
+      if (n < 0 || 1 + nb + (uint64_t)n > rem) return 0;
+      c->app_var(p + 1 + nb, (size_t)n);
+      return 1 + (size_t)nb + (size_t)n;
+    }
+    case 1: {                            /* memcomparable bytes */
+      std::vector<uint8_t> out;
+      size_t used = memcmp_decode(p + 1, rem - 1, &out);
+      if (!used) return 0;
+      c->app_var(out.data(), out.size());
+      return 1 + used;
+    }
+    default:
+      return 0;
+  }
+}
+
+/* re-encode a TypeDefault datum-row response into TypeChunk chunks */
+static bool chunk_encode_post(const std::vector<uint8_t> &datum_resp,
+                              const std::vector<uint64_t> &rows_per_chunk,
+                              const std::vector<CoprFieldType> &out_fts,
+                              std::vector<uint8_t> *out) {
+  size_t p = 0;
+  size_t nc = out_fts.size();
+  for (uint64_t nrows : rows_per_chunk) {
+    if (!nrows) continue;
+    std::vector<ChunkCol> cols(nc);
+    for (size_t c = 0; c < nc; c++)
+      if (!cols[c].init(out_fts[c])) return false;
+    for (uint64_t r = 0; r < nrows; r++) {
+      for (size_t c = 0; c < nc; c++) {
+        size_t used = chunk_append_datum(&cols[c], datum_resp.data() + p,
+                                         datum_resp.size() - p);
+        if (!used) return false;
+        p += used;
+      }
+    }
+    for (size_t c = 0; c < nc; c++) cols[c].flush(out);
+  }
+  return p == datum_resp.size();
+}
+
 /* ---------------- the pipeline ---------------- */
 struct Pipeline {
   const CoprDagRequest *req;
@@ -897,6 +1073,9 @@ static bool run_pipeline(const CoprDagRequest *req,
 
   uint64_t out_rows = 0;
   uint64_t emitted_rows = 0;
+  /* chunk segmentation for TypeChunk (runner.rs:1188-1225: one chunk per
+     executor batch; drains emit in 1024-row chunks) */
+  std::vector<uint64_t> chunk_rows;
 
   /* TopN collection: per surviving row, the order keys + the row's encoded
      output cells (top_n_executor.rs keeps whole rows in its heap; we keep
@@ -1029,6 +1208,7 @@ static bool run_pipeline(const CoprDagRequest *req,
         out_rows++;
       }
       emitted_rows += take;
+      if (take) chunk_rows.push_back(take);
       /* paging: stop at the batch boundary where accumulated OUTPUT rows
          reach paging_size (runner.rs:917-921 record_all >= paging_size) */
       if (req->paging_size && emitted_rows >= req->paging_size && cursor < n_kv) {
@@ -1154,6 +1334,11 @@ static bool run_pipeline(const CoprDagRequest *req,
       resp->insert(resp->end(), tr.cells.begin(), tr.cells.end());
       out_rows++;
     }
+    for (uint64_t left = take; left; ) {
+      uint64_t b = std::min<uint64_t>(left, 1024);
+      chunk_rows.push_back(b);
+      left -= b;
+    }
   } else if (pl.agg_exec) {
     /* drain: iterate groups, push states (fast_hash_aggr_executor.rs:393) */
     size_t n_groups = simple_agg ? 1 : group_keys_in_order.size();
@@ -1193,6 +1378,11 @@ static bool run_pipeline(const CoprDagRequest *req,
     }
     /* encode drained rows via output_offsets */
     uint64_t take = std::min<uint64_t>(n_groups, pl.limit);
+    for (uint64_t left = take; left; ) {
+      uint64_t b = std::min<uint64_t>(left, 1024);
+      chunk_rows.push_back(b);
+      left -= b;
+    }
     for (uint64_t g = 0; g < take; g++) {
       for (uint32_t oo = 0; oo < req->n_output_offsets; oo++) {
         uint32_t off = req->output_offsets[oo];
@@ -1205,6 +1395,15 @@ static bool run_pipeline(const CoprDagRequest *req,
     }
   }
 
+  if (req->encode_type == 1) {
+    std::vector<CoprFieldType> fts;
+    for (uint32_t oo = 0; oo < req->n_output_offsets; oo++)
+      fts.push_back(pl.out_schema[req->output_offsets[oo]]);
+    std::vector<uint8_t> chunked;
+    if (!chunk_encode_post(*resp, chunk_rows, fts, &chunked))
+      FAIL("TypeChunk encode failed");
+    resp->swap(chunked);
+  }
   *n_out_rows = out_rows;
   return true;
 }

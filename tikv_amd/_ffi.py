@@ -54,7 +54,7 @@ class CoprDagRequest(C.Structure):
                 ("output_offsets", C.POINTER(C.c_uint32)),
                 ("n_output_offsets", C.c_uint32),
                 ("flags", C.c_uint64), ("div_precision_increment", C.c_int32),
-                ("paging_size", C.c_uint64)]
+                ("paging_size", C.c_uint64), ("encode_type", C.c_int32)]
 
 
 class CoprExecSummary(C.Structure):

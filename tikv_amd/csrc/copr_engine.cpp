@@ -724,6 +724,190 @@ static void encode_agg_row(const HostPlan &pl, const SimpleAggAcc *accs,
 
 }  // namespace
 
+
+/* ---------------- TypeChunk response encoding (engine side) ----------------
+ * chunk/column.rs:41-71,1052-1071; decimal.rs:2135-2142; one chunk per
+ * executor batch (runner.rs:1188-1225). Independent product restatement of
+ * the same wire format the oracle pins. */
+namespace {
+
+struct EChunkCol {
+  bool fixed = true;
+  uint32_t flen = 8;
+  uint32_t length = 0, null_cnt = 0;
+  std::vector<uint8_t> bitmap, data;
+  std::vector<int64_t> offsets{0};
+  bool init(const CoprFieldType &ft) {
+    switch (ft.tp) {
+      case COPR_TP_TINY: case COPR_TP_SHORT: case COPR_TP_INT24:
+      case COPR_TP_LONG: case COPR_TP_LONGLONG: case COPR_TP_YEAR:
+      case COPR_TP_DOUBLE: case COPR_TP_DURATION:
+        fixed = true; flen = 8; return true;
+      case COPR_TP_NEWDECIMAL:
+        fixed = true; flen = 40; return true;
+      case COPR_TP_VARCHAR: case COPR_TP_STRING: case COPR_TP_VARSTRING:
+      case COPR_TP_BLOB:
+        fixed = false; return true;
+      default:
+        return false;
+    }
+  }
+  void bit(bool notnull) {
+    if ((length & 7) == 0) bitmap.push_back(0);
+    if (notnull) bitmap[length >> 3] |= (uint8_t)(1u << (length & 7));
+    else null_cnt++;
+  }
+  void app_null() {
+    bit(false);
+    if (fixed) data.insert(data.end(), flen, 0);
+    else offsets.push_back(offsets.back());
+    length++;
+  }
+  void app_fixed(const uint8_t *p, uint32_t n) {
+    bit(true);
+    data.insert(data.end(), p, p + n);
+    if (n < flen) data.insert(data.end(), flen - n, 0);
+    length++;
+  }
+  void app_var(const uint8_t *p, size_t n) {
+    bit(true);
+    data.insert(data.end(), p, p + n);
+    offsets.push_back(offsets.back() + (int64_t)n);
+    length++;
+  }
+  void flush(std::vector<uint8_t> *out) {
+    uint8_t w[8];
+    auto u32le = [&](uint32_t v) {
+      w[0] = (uint8_t)v; w[1] = (uint8_t)(v >> 8);
+      w[2] = (uint8_t)(v >> 16); w[3] = (uint8_t)(v >> 24);
+      out->insert(out->end(), w, w + 4);
+    };
+    u32le(length);
+    u32le(null_cnt);
+    if (null_cnt > 0) out->insert(out->end(), bitmap.begin(), bitmap.end());
+    if (!fixed) {
+      for (int64_t v : offsets) {
+        uint64_t u = (uint64_t)v;
+        for (int i = 0; i < 8; i++) w[i] = (uint8_t)(u >> (8 * i));
+        out->insert(out->end(), w, w + 8);
+      }
+    }
+    out->insert(out->end(), data.begin(), data.end());
+  }
+};
+
+static bool e_var_u64(const uint8_t *p, size_t rem, uint64_t *v, size_t *n) {
+  uint64_t uv = 0;
+  size_t i = 0;
+  int sh = 0;
+  while (i < rem && i < 10) {
+    uint8_t b = p[i++];
+    uv |= (uint64_t)(b & 0x7F) << sh;
+    sh += 7;
+    if (!(b & 0x80)) { *v = uv; *n = i; return true; }
+  }
+  return false;
+}
+
+static size_t e_chunk_append_datum(EChunkCol *c, const uint8_t *p,
+                                   size_t rem) {
+  if (!rem) return 0;
+  uint8_t flag = p[0];
+  uint8_t tmp[8];
+  switch (flag) {
+    case 0:
+      c->app_null();
+      return 1;
+    case 3: case 4: {
+      if (rem < 9) return 0;
+      uint64_t u = 0;
+      for (int i = 0; i < 8; i++) u = (u << 8) | p[1 + i];
+      if (flag == 3) u ^= 0x8000000000000000ull;
+      for (int i = 0; i < 8; i++) tmp[i] = (uint8_t)(u >> (8 * i));
+      c->app_fixed(tmp, 8);
+      return 9;
+    }
+    case 5: {
+      if (rem < 9) return 0;
+      uint64_t u = 0;
+      for (int i = 0; i < 8; i++) u = (u << 8) | p[1 + i];
+      if (u & 0x8000000000000000ull) u &= 0x7FFFFFFFFFFFFFFFull;
+      else u = ~u;
+      for (int i = 0; i < 8; i++) tmp[i] = (uint8_t)(u >> (8 * i));
+      c->app_fixed(tmp, 8);
+      return 9;
+    }
+    case 8: case 9: {
+      uint64_t uv;
+      size_t n;
+      if (!e_var_u64(p + 1, rem - 1, &uv, &n)) return 0;
+      uint64_t u = uv;
+      if (flag == 8) {
+        uint64_t half = uv >> 1;
+        u = (uv & 1) ? ~half : half;
+      }
+      for (int i = 0; i < 8; i++) tmp[i] = (uint8_t)(u >> (8 * i));
+      c->app_fixed(tmp, 8);
+      return 1 + n;
+    }
+    case 6: {
+      prod::PDec d;
+      uint8_t rf = 0;
+      size_t used = 0;
+      if (!prod::pdec_decode(p + 1, rem - 1, &d, &rf, &used)) return 0;
+      uint8_t buf[40];
+      buf[0] = d.int_cnt; buf[1] = d.frac_cnt; buf[2] = rf;
+      buf[3] = d.negative ? 1 : 0;
+      for (int i = 0; i < 9; i++) {
+        uint32_t w = d.word_buf[i];
+        buf[4 + 4 * i] = (uint8_t)w;
+        buf[5 + 4 * i] = (uint8_t)(w >> 8);
+        buf[6 + 4 * i] = (uint8_t)(w >> 16);
+        buf[7 + 4 * i] = (uint8_t)(w >> 24);
+      }
+      c->app_fixed(buf, 40);
+      return 1 + used;
+    }
+    case 2: {
+      uint64_t uv;
+      size_t nb;
+      if (!e_var_u64(p + 1, rem - 1, &uv, &nb)) return 0;
+      int64_t n = (uv & 1) ? (int64_t)(~(uv >> 1)) : (int64_t)(uv >> 1);
+      if (n < 0 || 1 + nb + (uint64_t)n > rem) return 0;
+      c->app_var(p + 1 + nb, (size_t)n);
+      return 1 + nb + (size_t)n;
+    }
+    default:
+      return 0;   /* memcomparable bytes not produced by these paths */
+  }
+}
+
+static bool e_chunk_encode_post(const std::vector<uint8_t> &datum_resp,
+                                const std::vector<uint64_t> &rows_per_chunk,
+                                const std::vector<CoprFieldType> &out_fts,
+                                std::vector<uint8_t> *out) {
+  size_t p = 0;
+  size_t nc = out_fts.size();
+  for (uint64_t nrows : rows_per_chunk) {
+    if (!nrows) continue;
+    std::vector<EChunkCol> cols(nc);
+    for (size_t c = 0; c < nc; c++)
+      if (!cols[c].init(out_fts[c])) return false;
+    for (uint64_t r = 0; r < nrows; r++) {
+      for (size_t c = 0; c < nc; c++) {
+        size_t used = e_chunk_append_datum(&cols[c], datum_resp.data() + p,
+                                           datum_resp.size() - p);
+        if (!used) return false;
+        p += used;
+      }
+    }
+    for (size_t c = 0; c < nc; c++) cols[c].flush(out);
+  }
+  return p == datum_resp.size();
+}
+
+}  // namespace
+
 /* ---------------- DAG run ---------------- */
 extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
                                     copr_region *const *regions, uint32_t n_regions,
@@ -759,6 +943,9 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
 
   std::vector<uint8_t> resp;
   uint64_t n_rows_out = 0;
+  /* TypeChunk segmentation (one chunk per executor batch; drains in
+     1024-row chunks) */
+  std::vector<uint64_t> chunk_rows;
   /* HIP-event timing of the scan kernel(s), on the engine's own stream —
      reported in summaries[0].time_processed_ns (the ExecSummary surface the
      reference fills per executor slot, execute_stats.rs:43-76). bench.py's
@@ -841,10 +1028,30 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
     preq.flags = req->flags;
     preq.div_precision_increment = req->div_precision_increment;
     copr_region *rp = &tmp;
+    preq.encode_type = 0;                 /* post-encode below if chunked */
     eng->dec_col_off = pl.topn_off;
     copr_status st2 = copr_dag_run(eng, &preq, &rp, 1, out);
     eng->dec_col_off = -1;
     free_sub();
+    if (st2 == COPR_OK && req->encode_type == 1) {
+      std::vector<uint8_t> dat(out->data, out->data + out->data_len);
+      std::vector<uint64_t> chunks;
+      for (uint64_t left = out->n_rows; left;) {
+        uint64_t b = left < 1024 ? left : 1024;
+        chunks.push_back(b);
+        left -= b;
+      }
+      std::vector<CoprFieldType> fts;
+      for (uint32_t oo = 0; oo < req->n_output_offsets; oo++)
+        fts.push_back(pl.out_schema[req->output_offsets[oo]]);
+      std::vector<uint8_t> chunked;
+      if (!e_chunk_encode_post(dat, chunks, fts, &chunked))
+        return SET_ERR(COPR_ERR_UNSUPPORTED, "TypeChunk encode failed");
+      free(out->data);
+      out->data = (uint8_t *)malloc(chunked.size() ? chunked.size() : 1);
+      memcpy(out->data, chunked.data(), chunked.size());
+      out->data_len = chunked.size();
+    }
     return st2;
   }
 
@@ -1179,6 +1386,23 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
           out->resume_row = pos;
         }
       }
+      if (req->encode_type == 1) {
+        /* chunk segmentation mirrors the batch ladder over this region's
+           keep flags, capped by the remaining LIMIT */
+        uint64_t bs = 32, pos2 = 0, limit_left =
+            pl.limit == UINT64_MAX ? UINT64_MAX : pl.limit - n_rows_out;
+        while (pos2 < scan_end && limit_left) {
+          uint64_t e2 = std::min(scan_end, pos2 + bs);
+          uint64_t cnt2 = 0;
+          for (uint64_t i = pos2; i < e2 && cnt2 < limit_left; i++)
+            cnt2 += h_keep[i] ? 1 : 0;
+          if (cnt2 > limit_left) cnt2 = limit_left;
+          if (cnt2) chunk_rows.push_back(cnt2);
+          if (limit_left != UINT64_MAX) limit_left -= cnt2;
+          pos2 = e2;
+          if (bs < 1024) bs *= 2;
+        }
+      }
       std::vector<uint8_t> rowbuf;
       for (uint64_t i = 0; i < scan_end && n_rows_out < pl.limit; i++) {
         if (!h_keep[i]) continue;
@@ -1239,6 +1463,26 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
   }
   hipEventDestroy(ev_a);
   hipEventDestroy(ev_b);
+  if (req->encode_type == 1) {
+    if (chunk_rows.empty() && n_rows_out) {
+      for (uint64_t left = n_rows_out; left;) {      /* drain: 1024/chunk */
+        uint64_t b = left < 1024 ? left : 1024;
+        chunk_rows.push_back(b);
+        left -= b;
+      }
+    }
+    std::vector<CoprFieldType> fts;
+    for (uint32_t oo = 0; oo < req->n_output_offsets; oo++) {
+      uint32_t off = req->output_offsets[oo];
+      if (off >= pl.out_schema.size())
+        return SET_ERR(COPR_ERR_INVALID_REQUEST, "bad output offset");
+      fts.push_back(pl.out_schema[off]);
+    }
+    std::vector<uint8_t> chunked;
+    if (!e_chunk_encode_post(resp, chunk_rows, fts, &chunked))
+      return SET_ERR(COPR_ERR_UNSUPPORTED, "TypeChunk encode failed");
+    resp.swap(chunked);
+  }
   out->data = (uint8_t *)malloc(resp.size() ? resp.size() : 1);
   memcpy(out->data, resp.data(), resp.size());
   out->data_len = resp.size();

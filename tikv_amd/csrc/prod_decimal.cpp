@@ -260,4 +260,80 @@ bool pdec_decode_scaled(const uint8_t *p0, size_t len, int64_t *scaled,
   return true;
 }
 
+
+/* full read_decimal (decimal.rs:2204-2289): parse a [prec][frac] encoded
+ * decimal into the in-memory word representation, with the reference's
+ * normalization (leading zero words reduce int_cnt; trailing digits scale
+ * up to a full word; all-zero collapses to zero). Used by the TypeChunk
+ * response encoder, which dumps this struct (decimal.rs:2135-2142). */
+bool pdec_decode(const uint8_t *p0, size_t len, PDec *d, uint8_t *result_frac,
+                 size_t *consumed) {
+  const uint8_t *p = p0;
+  size_t rem = len;
+  if (rem < 3) return false;
+  uint8_t prec = p[0], frac_cnt = p[1];
+  p += 2; rem -= 2;
+  if (prec < frac_cnt) return false;
+  int int_cnt = prec - frac_cnt;
+  int int_word_cnt = int_cnt / DIGITS_PER_WORD;
+  int leading_digits = int_cnt - int_word_cnt * DIGITS_PER_WORD;
+  int frac_word_cnt = frac_cnt / DIGITS_PER_WORD;
+  int trailing_digits = frac_cnt - frac_word_cnt * DIGITS_PER_WORD;
+  int tot_words = int_word_cnt + (leading_digits ? 1 : 0) + frac_word_cnt +
+                  (trailing_digits ? 1 : 0);
+  if (tot_words > 9) return false;
+  uint32_t mask = (p[0] & 0x80) ? 0 : 0xFFFFFFFFu;
+  *d = PDec{};
+  d->int_cnt = (uint8_t)int_cnt;
+  d->frac_cnt = frac_cnt;
+  d->negative = mask != 0;
+  bool is_first = true;
+  int word_idx = 0;
+  if (leading_digits > 0) {
+    uint32_t w;
+    if (!read_word(p, rem, DIG_2_BYTES[leading_digits], is_first, &w))
+      return false;
+    w ^= mask;
+    if (w >= TEN_POW[leading_digits + 1]) return false;
+    d->word_buf[word_idx] = w;
+    if (w != 0) word_idx++;
+    else d->int_cnt -= (uint8_t)leading_digits;
+  }
+  for (int k = 0; k < int_word_cnt; k++) {
+    uint32_t w;
+    if (!read_word(p, rem, 4, is_first, &w)) return false;
+    w ^= mask;
+    if (w > WORD_BASE - 1) return false;
+    d->word_buf[word_idx] = w;
+    if (word_idx > 0 || w != 0) word_idx++;
+    else d->int_cnt -= (uint8_t)DIGITS_PER_WORD;
+  }
+  for (int k = 0; k < frac_word_cnt; k++) {
+    uint32_t w;
+    if (!read_word(p, rem, 4, is_first, &w)) return false;
+    w ^= mask;
+    if (w > WORD_BASE - 1) return false;
+    d->word_buf[word_idx++] = w;
+  }
+  if (trailing_digits > 0) {
+    uint32_t w;
+    if (!read_word(p, rem, DIG_2_BYTES[trailing_digits], is_first, &w))
+      return false;
+    w ^= mask;
+    unsigned long long v =
+        (unsigned long long)w * TEN_POW[DIGITS_PER_WORD - trailing_digits];
+    if (v > WORD_BASE - 1) return false;
+    d->word_buf[word_idx] = (uint32_t)v;
+  }
+  if (d->int_cnt == 0 && d->frac_cnt == 0) {
+    /* Decimal::zero() (decimal.rs:996) */
+    *d = PDec{};
+    d->int_cnt = 1;
+    d->negative = false;
+  }
+  *result_frac = frac_cnt;
+  *consumed = (size_t)(p - p0);
+  return true;
+}
+
 }  // namespace prod

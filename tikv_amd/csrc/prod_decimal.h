@@ -42,5 +42,10 @@ bool pdec_decode_scaled(const uint8_t *p, size_t len, int64_t *scaled,
                         uint8_t *frac_out, size_t *consumed);
 bool pdec_encoded_len(const uint8_t *p, size_t len, size_t *elen);
 
+/* full read_decimal (decimal.rs:2204-2289) into the word representation;
+ * result_frac = the encoded frac count (result_frac_cnt in the struct dump) */
+bool pdec_decode(const uint8_t *p, size_t len, PDec *d, uint8_t *result_frac,
+                 size_t *consumed);
+
 }  // namespace prod
 #endif

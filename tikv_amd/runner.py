@@ -187,6 +187,11 @@ class DagSelect:
         self._paging = size
         return self
 
+    def chunked(self):
+        """TypeChunk response encoding (tipb EncodeType::TypeChunk)."""
+        self._encode_type = 1
+        return self
+
     def output(self, offsets):
         self.output_offsets = list(offsets)
         return self
@@ -203,6 +208,7 @@ class DagSelect:
         req.output_offsets = offs
         req.n_output_offsets = len(self.output_offsets)
         req.paging_size = getattr(self, "_paging", 0)
+        req.encode_type = getattr(self, "_encode_type", 0)
         self._req = req
         return req
 
