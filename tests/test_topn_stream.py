@@ -458,3 +458,78 @@ def test_chunked_gpu_parity(engine):
             rgn.close()
     finally:
         g.close()
+
+
+# ---- FIRST aggregate ----------------------------------------------------
+def test_first_oracle():
+    orc = _orc()
+    seq = [(1, 10), (1, None), (2, None), (2, 7), (None, 3)]
+    rows = [{1: g, 2: v} for g, v in seq]
+    k, ko, v, vo, n, keep = region_of(rows)
+    cols = [tikv_amd.Col(1), tikv_amd.Col(2)]
+    req = (tikv_amd.DagSelect(cols)
+           .stream_agg([tikv_amd.first_col(1), tikv_amd.count_star()],
+                       tikv_amd.Expr().col(0)).build())
+    data, nrows = orc.dag_run(req, k, ko, v, vo, n)
+    assert nrows == 3   # runs: [1,1], [2,2], [None]
+    got = []
+    p = 0
+    for _ in range(3):
+        if data[p] == 0:
+            fv, p = None, p + 1
+        else:
+            fv, p = dec_int(data, p)
+        cnt, p = dec_int(data, p)
+        if data[p] == 0:
+            key, p = None, p + 1
+        else:
+            key, p = dec_int(data, p)
+        got.append((fv, cnt, key))
+    # first of run 1 = 10; run 2 first value is NULL; NULL-key run first = 3
+    assert got == [(10, 2, 1), (None, 2, 2), (3, 1, None)]
+
+
+def test_first_simple_oracle():
+    orc = _orc()
+    rows = [{1: v} for v in [None, 5, 6]]
+    k, ko, v, vo, n, keep = region_of(rows)
+    cols = [tikv_amd.Col(1)]
+    req = (tikv_amd.DagSelect(cols)
+           .simple_agg([tikv_amd.first_col(0), tikv_amd.count_star()])
+           .build())
+    data, nrows = orc.dag_run(req, k, ko, v, vo, n)
+    assert nrows == 1
+    assert data[0] == 0          # first row's value is NULL
+    cnt, p = dec_int(data, 1)
+    assert cnt == 3 and p == len(data)
+
+
+@pytest.mark.gpu
+def test_first_gpu_parity(engine):
+    orc = _orc()
+    g = tikv_amd.GenRegion(config_index=0, n_rows=50001, table_id=5)
+    try:
+        rgn = engine.region(g)
+        try:
+            cols = [tikv_amd.Col(i) for i in range(1, 5)]
+            # stream agg with FIRST
+            req = (tikv_amd.DagSelect(cols)
+                   .stream_agg([tikv_amd.first_col(1), tikv_amd.count_star()],
+                               tikv_amd.Expr().col(0)).build())
+            gd, gr, _ = engine.dag_run(req, [rgn])
+            od, orows = orc.dag_run(req, g.keys, g.key_offs, g.vals,
+                                    g.val_offs, g.n_kv)
+            assert (gr, gd) == (orows, od)
+            # simple agg with FIRST (reroutes through the one-run stream)
+            sel = tikv_amd.cmp_col_const(2, F.SIG_GT_INT, 0)
+            req2 = (tikv_amd.DagSelect(cols).where(sel)
+                    .simple_agg([tikv_amd.first_col(1), tikv_amd.count_star(),
+                                 tikv_amd.max_col(3)]).build())
+            gd2, gr2, _ = engine.dag_run(req2, [rgn])
+            od2, or2 = orc.dag_run(req2, g.keys, g.key_offs, g.vals,
+                                   g.val_offs, g.n_kv)
+            assert (gr2, gd2) == (or2, od2)
+        finally:
+            rgn.close()
+    finally:
+        g.close()

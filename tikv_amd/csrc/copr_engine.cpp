@@ -230,6 +230,7 @@ struct HostPlan {
   bool has_agg = false;
   bool hash_agg = false;
   bool stream_agg = false;
+  bool simple_as_stream = false;   /* simple agg with FIRST: one-run stream */
   CoprFieldType group_ft{};
   /* TopN (top_n_executor.rs): single int order-by column */
   bool has_topn = false;
@@ -535,6 +536,12 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
         ds.kind = ad.func == COPR_AGG_MAX ? DAGG_MAX_INT : DAGG_MIN_INT;
         oa.in_kind = ds.kind;
         break;
+      case COPR_AGG_FIRST:
+        if (!ci || !et_int(ci->ft.tp))
+          return SET_ERR(COPR_ERR_UNSUPPORTED, "first native only over int");
+        ds.kind = DAGG_FIRST;
+        oa.in_kind = DAGG_FIRST;
+        break;
       case COPR_AGG_BIT_AND: case COPR_AGG_BIT_OR: case COPR_AGG_BIT_XOR:
         if (!ci || !et_int(ci->ft.tp))
           return SET_ERR(COPR_ERR_UNSUPPORTED, "bit ops native only over int");
@@ -560,6 +567,9 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
     }
     pl->out_schema.push_back(ad.out_ft);
   }
+  bool any_first = false;
+  for (int a = 0; a < sp.n_aggs; a++)
+    if (sp.aggs[a].kind == DAGG_FIRST) any_first = true;
   if (pl->hash_agg) {
     if (agg->n_group_by != 1)
       return SET_ERR(COPR_ERR_UNSUPPORTED, "exactly one group-by expr");
@@ -584,6 +594,17 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
     sp.group_col_unsigned = (ci.ft.flag & COPR_FLAG_UNSIGNED) ? 1 : 0;
     pl->group_ft = ci.ft;
     pl->out_schema.push_back(ci.ft);
+    if (any_first && sp.mode == 2)
+      return SET_ERR(COPR_ERR_UNSUPPORTED,
+                     "FIRST over int-key hash agg not yet native "
+                     "(stream/bytes grouping supports it)");
+  } else if (any_first) {
+    /* simple agg with FIRST reroutes through a single-run stream pass
+       (group col that never matches -> one NULL-key run) */
+    sp.mode = 3;
+    sp.group_col_id = INT64_MIN + 1;
+    sp.group_col_unsigned = 0;
+    pl->simple_as_stream = true;
   }
   return COPR_OK;
 }
@@ -680,6 +701,12 @@ static void encode_agg_row(const HostPlan &pl, const SimpleAggAcc *accs,
     if (oa.func == COPR_AGG_COUNT) {
       bool uns = (pl.out_schema[oc].flag & COPR_FLAG_UNSIGNED) != 0;
       enc_datum_int(&(*cols_out)[oc], (int64_t)ac.cnt, uns);
+      oc++;
+    } else if (oa.func == COPR_AGG_FIRST) {
+      /* first row's value, NULL included (impl_first.rs) */
+      bool uns = (pl.out_schema[oc].flag & COPR_FLAG_UNSIGNED) != 0;
+      if (ac.cnt == 0 || ac.sum_hi) enc_datum_null(&(*cols_out)[oc]);
+      else enc_datum_int(&(*cols_out)[oc], (int64_t)ac.sum_lo, uns);
       oc++;
     } else if (oa.func == COPR_AGG_MAX || oa.func == COPR_AGG_MIN) {
       /* undo the fold transform (d_fold_xform): MIN folded max over ~biased */
@@ -1076,11 +1103,18 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
     if (n_seg == -2) return SET_ERR(COPR_ERR_OOM, "stream agg temp alloc");
     if (n_seg < 0) return SET_ERR(COPR_ERR_INTERNAL, "stream agg failed");
     size_t n_out_cols = pl.out_schema.size();
+    if (pl.simple_as_stream && n_seg == 0) {
+      /* empty input: the reference still emits the one aggregate row */
+      h_accs.assign(sp.n_aggs, SimpleAggAcc{});
+      h_gk.assign(1, 0);
+      h_gs.assign(1, 1);
+      n_seg = 1;
+    }
     uint64_t take = (uint64_t)n_seg < pl.limit ? (uint64_t)n_seg : pl.limit;
     for (uint64_t g = 0; g < take; g++) {
       std::vector<std::vector<uint8_t>> cols(n_out_cols);
-      encode_agg_row(pl, &h_accs[g * sp.n_aggs], true, h_gs[g] == 1, h_gk[g],
-                     &cols);
+      encode_agg_row(pl, &h_accs[g * sp.n_aggs], !pl.simple_as_stream,
+                     h_gs[g] == 1, h_gk[g], &cols);
       for (uint32_t oo = 0; oo < req->n_output_offsets; oo++) {
         uint32_t off = req->output_offsets[oo];
         if (off < n_out_cols)

@@ -49,6 +49,11 @@ enum {
   DAGG_BIT_AND,
   DAGG_BIT_OR,
   DAGG_BIT_XOR,
+  /* FIRST (impl_first.rs): the first row's value, NULL included. Handled
+     by the segmented pipelines only (simple agg reroutes through a
+     single-run stream pass): acc.cnt = 1 once set, sum_lo = value,
+     sum_hi = 1 when the first row's value is NULL. */
+  DAGG_FIRST,
 };
 
 struct DevAggSpec {

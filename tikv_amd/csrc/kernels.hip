@@ -3019,7 +3019,11 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
         uint8_t contribute = 0;
         int64_t v = 0;
         if (s) {
-          if (sp.kind == DAGG_COUNT_ROWS) {
+          if (sp.kind == DAGG_FIRST) {
+            /* first row's value, NULL (or missing->NULL) included */
+            contribute = (cols[a].found && !cols[a].null) ? 1 : 2;
+            v = cols[a].iv;
+          } else if (sp.kind == DAGG_COUNT_ROWS) {
             contribute = 1;
           } else if (!cols[a].found || cols[a].null) {
           } else if (sp.kind == DAGG_COUNT_COL) {
@@ -3094,8 +3098,19 @@ __global__ static void k_run_update(const int64_t *ck, const uint8_t *cs,
   }
   SimpleAggAcc *base = accs + (uint64_t)seg * plan.n_aggs;
   for (int a = 0; a < plan.n_aggs; a++) {
-    if (!cas[(uint64_t)a * m + i]) continue;
     const DevAggSpec &sp = plan.aggs[a];
+    uint8_t ct = cas[(uint64_t)a * m + i];
+    if (sp.kind == DAGG_FIRST) {
+      /* runs are contiguous in input order: the boundary row IS the
+         group's first row */
+      if (i == 0 || segid[i] != segid[i - 1]) {
+        base[a].cnt = 1;
+        base[a].sum_lo = (unsigned long long)cav[(uint64_t)a * m + i];
+        base[a].sum_hi = ct == 2 ? 1 : 0;
+      }
+      continue;
+    }
+    if (!ct) continue;
     int64_t v = cav[(uint64_t)a * m + i];
     atomicAdd(&base[a].cnt, 1ull);
     if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC)
@@ -3476,8 +3491,13 @@ __global__ static void k_bytes_update(const uint64_t *co, const uint32_t *cl,
   }
   SimpleAggAcc *base = accs + (uint64_t)seg * plan.n_aggs;
   for (int a = 0; a < plan.n_aggs; a++) {
-    if (!cas[(uint64_t)a * m + a0]) continue;
     const DevAggSpec &sp = plan.aggs[a];
+    if (sp.kind == DAGG_FIRST) {
+      /* min source (compacted) index via max of its complement */
+      atomicMax(&base[a].cnt, ~(unsigned long long)a0);
+      continue;
+    }
+    if (!cas[(uint64_t)a * m + a0]) continue;
     int64_t v = cav[(uint64_t)a * m + a0];
     atomicAdd(&base[a].cnt, 1ull);
     if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC)
@@ -3490,6 +3510,35 @@ __global__ static void k_bytes_update(const uint64_t *co, const uint32_t *cl,
       else atomicOr(&base[a].sum_lo, b);
     }
   }
+}
+
+__global__ static void k_bytes_first_fix(const int64_t *cav,
+                                         const uint8_t *cas,
+                                         const uint32_t *perm,
+                                         const uint32_t *segid, uint64_t m,
+                                         ScanPlan plan,
+                                         SimpleAggAcc *accs) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= m) return;
+  uint32_t seg = segid[i] - 1;
+  uint32_t a0 = perm[i];
+  SimpleAggAcc *base = accs + (uint64_t)seg * plan.n_aggs;
+  for (int a = 0; a < plan.n_aggs; a++) {
+    if (plan.aggs[a].kind != DAGG_FIRST) continue;
+    if (base[a].cnt == ~(unsigned long long)a0) {
+      base[a].sum_lo = (unsigned long long)cav[(uint64_t)a * m + a0];
+      base[a].sum_hi = cas[(uint64_t)a * m + a0] == 2 ? 1 : 0;
+    }
+  }
+}
+
+__global__ static void k_bytes_first_norm(uint64_t n_seg, ScanPlan plan,
+                                          SimpleAggAcc *accs) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n_seg) return;
+  SimpleAggAcc *base = accs + i * plan.n_aggs;
+  for (int a = 0; a < plan.n_aggs; a++)
+    if (plan.aggs[a].kind == DAGG_FIRST) base[a].cnt = 1;
 }
 
 /* Bytes-key hash aggregation. Returns n_groups (>=0), -1 internal, -2 oom,
@@ -3611,6 +3660,17 @@ int dev_bytes_agg(const ScanPlan &plan, const DevRegion &rgn, void *stream,
   hipLaunchKernelGGL(k_bytes_update, dim3(mblocks), dim3(256), 0, s, co, cl,
                      cs, cav, cas, perm, segid, m, plan, accs, rofs, rlen,
                      rst);
+  bool has_first = false;
+  for (int a = 0; a < na; a++)
+    if (plan.aggs[a].kind == DAGG_FIRST) has_first = true;
+  if (has_first) {
+    hipLaunchKernelGGL(k_bytes_first_fix, dim3(mblocks), dim3(256), 0, s,
+                       cav, cas, perm, segid, m, plan, accs);
+    uint32_t sblocks = (uint32_t)((n_seg + 255) / 256);
+    if (sblocks)
+      hipLaunchKernelGGL(k_bytes_first_norm, dim3(sblocks), dim3(256), 0, s,
+                         (uint64_t)n_seg, plan, accs);
+  }
   h_accs->resize((size_t)n_seg * na);
   h_kofs->resize(n_seg);
   h_klen->resize(n_seg);

@@ -239,6 +239,10 @@ def min_col(offset, tp=F.TP_LONGLONG, flag=0):
     return (F.AGG_MIN, Expr().col(offset), field_type(tp, flag))
 
 
+def first_col(offset, tp=F.TP_LONGLONG, flag=0):
+    return (F.AGG_FIRST, Expr().col(offset), field_type(tp, flag))
+
+
 def bit_op(func, offset):
     return (func, Expr().col(offset), field_type(F.TP_LONGLONG, F.FLAG_UNSIGNED))
 
