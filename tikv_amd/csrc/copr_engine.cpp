@@ -752,6 +752,88 @@ static void encode_agg_row(const HostPlan &pl, const SimpleAggAcc *accs,
 }  // namespace
 
 
+
+/* row-v2 raw cell -> v1 datum re-encode (compat_v1.rs:28-126): the
+   engine-side restatement used when project outputs come from v2 rows */
+static bool e_v2_cell_to_datum(int32_t tp, uint32_t ft_flag,
+                               const uint8_t *cell, size_t n,
+                               std::vector<uint8_t> *out) {
+  auto le_u64 = [&](uint64_t *u) {
+    if (n != 1 && n != 2 && n != 4 && n != 8) return false;
+    uint64_t v = 0;
+    for (size_t i = 0; i < n; i++) v |= (uint64_t)cell[i] << (8 * i);
+    *u = v;
+    return true;
+  };
+  auto le_i64 = [&](int64_t *iv) {
+    uint64_t v;
+    if (!le_u64(&v)) return false;
+    switch (n) {
+      case 1: *iv = (int8_t)v; break;
+      case 2: *iv = (int16_t)v; break;
+      case 4: *iv = (int32_t)v; break;
+      default: *iv = (int64_t)v; break;
+    }
+    return true;
+  };
+  auto be8 = [&](uint64_t u) {
+    for (int i = 7; i >= 0; i--) out->push_back((uint8_t)(u >> (8 * i)));
+  };
+  switch (tp) {
+    case COPR_TP_TINY: case COPR_TP_SHORT: case COPR_TP_INT24:
+    case COPR_TP_LONG: case COPR_TP_LONGLONG: {
+      if (ft_flag & COPR_FLAG_UNSIGNED) {
+        uint64_t u;
+        if (!le_u64(&u)) return false;
+        out->push_back(4);
+        be8(u);
+      } else {
+        int64_t iv;
+        if (!le_i64(&iv)) return false;
+        out->push_back(3);
+        be8((uint64_t)iv ^ 0x8000000000000000ull);
+      }
+      return true;
+    }
+    case COPR_TP_FLOAT: case COPR_TP_DOUBLE:
+      out->push_back(5);
+      out->insert(out->end(), cell, cell + n);
+      return true;
+    case COPR_TP_VARCHAR: case COPR_TP_VARSTRING: case COPR_TP_STRING:
+    case COPR_TP_BLOB: {
+      out->push_back(2);
+      uint64_t uv = (uint64_t)n << 1;
+      while (uv >= 0x80) { out->push_back((uint8_t)(uv | 0x80)); uv >>= 7; }
+      out->push_back((uint8_t)uv);
+      out->insert(out->end(), cell, cell + n);
+      return true;
+    }
+    case COPR_TP_NEWDECIMAL:
+      out->push_back(6);
+      out->insert(out->end(), cell, cell + n);
+      return true;
+    case COPR_TP_DURATION: {
+      int64_t iv;
+      if (!le_i64(&iv)) return false;
+      out->push_back(7);
+      be8((uint64_t)iv ^ 0x8000000000000000ull);
+      return true;
+    }
+    case COPR_TP_DATE: case COPR_TP_DATETIME: case COPR_TP_TIMESTAMP: {
+      uint64_t u;
+      if (!le_u64(&u)) return false;
+      out->push_back(4);
+      be8(u);
+      return true;
+    }
+    case COPR_TP_NULL:
+      out->push_back(0);
+      return true;
+    default:
+      return false;
+  }
+}
+
 /* ---------------- TypeChunk response encoding (engine side) ----------------
  * chunk/column.rs:41-71,1052-1071; decimal.rs:2135-2142; one chunk per
  * executor batch (runner.rs:1188-1225). Independent product restatement of
@@ -1481,9 +1563,21 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
               resp.push_back(0);
             else
               return SET_ERR(COPR_ERR_STORAGE, "missing NOT NULL column");
+          } else if (clen == 0xFFFFEu) {
+            resp.push_back(0);                 /* explicit v2 NULL cell */
           } else {
             uint64_t goff = cp >> 20;
-            resp.insert(resp.end(), vbase + goff, vbase + goff + clen);
+            uint64_t rs = r->h_val_offs[i];
+            uint64_t rl = r->h_val_offs[i + 1] - rs;
+            if (rl > 1 && vbase[rs] == 128) {
+              /* v2 row: raw payload -> datum re-encode */
+              const CoprColumnInfo &ci = pl.cols[off];
+              if (!e_v2_cell_to_datum(ci.ft.tp, ci.ft.flag, vbase + goff,
+                                      clen, &resp))
+                return SET_ERR(COPR_ERR_UNSUPPORTED, "v2 cell re-encode");
+            } else {
+              resp.insert(resp.end(), vbase + goff, vbase + goff + clen);
+            }
           }
         }
         n_rows_out++;

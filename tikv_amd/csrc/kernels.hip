@@ -2277,6 +2277,39 @@ k_scan_project(ScanPlan plan,
       int needed = (plan.has_filter ? 1 : 0) + plan.n_out;
       int found = 0;
 
+      if (vlen > 1 && vp[0] == 128) {
+        /* row v2: direct column lookup; value cells are RAW payloads, the
+           host re-encodes them as datums (compat_v1.rs:28-126). 0xFFFFE
+           marks an explicit NULL cell (v1 NULLs are 1-byte datums with a
+           capturable span; v2 NULLs have no bytes). */
+        V2Row r2;
+        if (!d_v2_parse(vp, vlen, &r2)) {
+          parse_ok = false;
+        } else {
+          if (plan.has_filter) {
+            uint32_t s2, e2;
+            int vst = d_v2_find(r2, plan.filter_col_id, &s2, &e2);
+            if (vst >= 0) {
+              filt_found = true;
+              if (vst == 0) filt_null = true;
+              else if (!d_v2_int(r2.vals + s2, e2 - s2,
+                                 plan.filter_col_unsigned, &filt_v))
+                parse_ok = false;
+            }
+          }
+          for (int j = 0; j < plan.n_out && parse_ok; j++) {
+            if (plan.out_is_handle[j]) continue;
+            uint32_t s2, e2;
+            int vst = d_v2_find(r2, plan.out_col_ids[j], &s2, &e2);
+            if (vst < 0) continue;                 /* missing: default fill */
+            if (vst == 0) { cell_pack[j] = 0xFFFFEull; continue; }
+            uint64_t goff = val_offs[my_row] + (uint32_t)(r2.vals + s2 - vp);
+            uint32_t clen = e2 - s2;
+            if (clen >= 0xFFFFEu) { parse_ok = false; break; }
+            cell_pack[j] = (goff << 20) | clen;
+          }
+        }
+      } else {
       ROW_FOREACH_BEGIN(vp, vlen)
         if (plan.has_filter && !filt_found && cell_id == plan.filter_col_id) {
           filt_found = true;
@@ -2295,6 +2328,7 @@ k_scan_project(ScanPlan plan,
         }
         if (found >= needed) break;
       ROW_FOREACH_END()
+      }
 
       if (!parse_ok) {
         any_parse_err = true;
