@@ -565,3 +565,29 @@ def test_project_row_v2_gpu_parity(engine):
             rgn.close()
     finally:
         g.close()
+
+
+@pytest.mark.gpu
+def test_topn_filter_col_in_output_gpu(engine):
+    """selection + TopN where BOTH the filter column and the order column
+    appear in the output: both must encode in decoded form."""
+    orc = _orc()
+    g = tikv_amd.GenRegion(config_index=1, n_rows=80001, table_id=5)
+    try:
+        rgn = engine.region(g)
+        try:
+            cols = [tikv_amd.Col(i) for i in range(1, 17)]
+            sel = tikv_amd.cmp_col_const(3, F.SIG_GT_INT, -900000000)
+            for desc in (False, True):
+                req = (tikv_amd.DagSelect(cols).where(sel)
+                       .topn(tikv_amd.Expr().col(6), 64, desc=desc)
+                       .output([3, 6, 11]).build())
+                gd, gr, _ = engine.dag_run(req, [rgn])
+                od, orows = orc.dag_run(req, g.keys, g.key_offs, g.vals,
+                                        g.val_offs, g.n_kv)
+                assert gr == orows
+                assert gd == od
+        finally:
+            rgn.close()
+    finally:
+        g.close()

@@ -43,8 +43,10 @@ struct copr_engine {
   uint64_t *d_crc_tables = nullptr;   /* 8*256 u64, built lazily */
   /* internal channel for the TopN sub-region project: column OFFSET the
      order expression decoded in place (response encodes it decoded,
-     lazy_column.rs:165,242); -1 = none */
+     lazy_column.rs:165,242); -1 = none. dec2 = the original request's
+     filter column when distinct from the order column. */
   int dec_col_off = -1;
+  int dec2_col_off = -1;
 };
 
 struct copr_region {
@@ -237,6 +239,7 @@ struct HostPlan {
   uint64_t topn_n = 0;
   int topn_desc = 0;
   int topn_off = -1;
+  int dec2_col_offset = -1;
   uint64_t limit = UINT64_MAX;
 };
 
@@ -1049,6 +1052,23 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
       pl.sp.filter_missing_null = 1;
     }
   }
+  if (eng->dec2_col_off >= 0 && !pl.has_agg && !pl.has_topn &&
+      (size_t)eng->dec2_col_off < pl.cols.size() &&
+      eng->dec2_col_off != eng->dec_col_off) {
+    const CoprColumnInfo &ci = pl.cols[eng->dec2_col_off];
+    pl.dec2_col_offset = eng->dec2_col_off;
+    pl.sp.dec2_col_id = ci.column_id;
+    pl.sp.dec2_col_unsigned = (ci.ft.flag & COPR_FLAG_UNSIGNED) ? 1 : 0;
+    if (ci.default_val && ci.default_val_len) {
+      int64_t dv;
+      int r = host_decode_int_datum(ci.default_val, ci.default_val_len, &dv);
+      if (r < 0) return SET_ERR(COPR_ERR_INVALID_REQUEST, "bad default");
+      pl.sp.dec2_missing_null = r == 1 ? 1 : 0;
+      pl.sp.dec2_missing_val = dv;
+    } else {
+      pl.sp.dec2_missing_null = 1;
+    }
+  }
 
   std::vector<uint8_t> resp;
   uint64_t n_rows_out = 0;
@@ -1139,8 +1159,12 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
     copr_region *rp = &tmp;
     preq.encode_type = 0;                 /* post-encode below if chunked */
     eng->dec_col_off = pl.topn_off;
+    eng->dec2_col_off =
+        (pl.sp.has_filter && pl.filter_col_offset >= 0 &&
+         pl.filter_col_offset != pl.topn_off) ? pl.filter_col_offset : -1;
     copr_status st2 = copr_dag_run(eng, &preq, &rp, 1, out);
     eng->dec_col_off = -1;
+    eng->dec2_col_off = -1;
     free_sub();
     if (st2 == COPR_OK && req->encode_type == 1) {
       std::vector<uint8_t> dat(out->data, out->data + out->data_len);
@@ -1436,9 +1460,14 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
         e = hipMalloc((void **)&po.filt_vals, n * 8 + 8);
         if (e == hipSuccess) e = hipMalloc((void **)&po.filt_state, n + 8);
       }
+      if (e == hipSuccess && pl.sp.dec2_col_id) {
+        e = hipMalloc((void **)&po.dec2_vals, n * 8 + 8);
+        if (e == hipSuccess) e = hipMalloc((void **)&po.dec2_state, n + 8);
+      }
       auto free_po = [&]() {
         hipFree(po.cells); hipFree(po.handles); hipFree(po.keep); hipFree(po.error);
         hipFree(po.filt_vals); hipFree(po.filt_state);
+        hipFree(po.dec2_vals); hipFree(po.dec2_state);
       };
       if (e != hipSuccess) { free_po(); return SET_ERR(COPR_ERR_OOM, "project alloc"); }
       hipMemsetAsync(po.keep, 0, n + 8, eng->stream);
@@ -1468,6 +1497,15 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
         ce = hipMemcpyAsync(h_fvals.data(), po.filt_vals, n * 8, hipMemcpyDeviceToHost, eng->stream);
         if (ce == hipSuccess)
           ce = hipMemcpyAsync(h_fstate.data(), po.filt_state, n, hipMemcpyDeviceToHost, eng->stream);
+      }
+      std::vector<long long> h_d2vals;
+      std::vector<uint8_t> h_d2state;
+      if (ce == hipSuccess && pl.sp.dec2_col_id) {
+        h_d2vals.resize(n);
+        h_d2state.resize(n);
+        ce = hipMemcpyAsync(h_d2vals.data(), po.dec2_vals, n * 8, hipMemcpyDeviceToHost, eng->stream);
+        if (ce == hipSuccess)
+          ce = hipMemcpyAsync(h_d2state.data(), po.dec2_state, n, hipMemcpyDeviceToHost, eng->stream);
       }
       if (ce == hipSuccess)
         ce = hipMemcpyAsync(h_err, po.error, 8, hipMemcpyDeviceToHost, eng->stream);
@@ -1537,6 +1575,15 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
           if (pl.sp.out_is_handle[off]) {
             enc_datum_int(&resp, h_handles[i],
                           (pl.cols[off].ft.flag & COPR_FLAG_UNSIGNED) != 0);
+            continue;
+          }
+          if ((int)off == pl.dec2_col_offset && pl.sp.dec2_col_id) {
+            uint8_t st8 = h_d2state[i];
+            bool uns = (pl.cols[off].ft.flag & COPR_FLAG_UNSIGNED) != 0;
+            if (st8 == 0) enc_datum_int(&resp, h_d2vals[i], uns);
+            else if (st8 == 1) resp.push_back(0);
+            else if (pl.sp.dec2_missing_null) resp.push_back(0);
+            else enc_datum_int(&resp, pl.sp.dec2_missing_val, uns);
             continue;
           }
           if ((int)off == pl.filter_col_offset && pl.sp.has_filter) {

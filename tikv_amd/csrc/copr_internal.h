@@ -84,6 +84,13 @@ struct ScanPlan {
      column) so the response encodes it in DECODED form
      (lazy_column.rs:165,242) */
   int32_t filter_decode_only;
+  /* second expression-decoded column for the TopN sub-region project (the
+     original request's FILTER column when it also appears in the output):
+     captured like the filter channel, output in decoded form. 0 = none. */
+  int64_t dec2_col_id;
+  int32_t dec2_col_unsigned;
+  int32_t dec2_missing_null;
+  int64_t dec2_missing_val;
 
   int32_t mode;                  /* 0 project, 1 simple agg, 2 hash agg */
   /* index scan (BatchIndexScanExecutor): the parsed stream is the KEY
@@ -168,6 +175,8 @@ struct ProjectOut {
      ensure_columns_decoded): per-row value + state for the host encoder */
   long long *filt_vals;          /* [n_rows] */
   uint8_t *filt_state;           /* [n_rows]: 0 value, 1 NULL, 2 missing */
+  long long *dec2_vals;          /* [n_rows] (dec2_col_id != 0) */
+  uint8_t *dec2_state;
   unsigned int *error;
 };
 

@@ -2272,9 +2272,12 @@ k_scan_project(ScanPlan plan,
       uint32_t vlen = (uint32_t)(val_offs[my_row + 1] - val_offs[my_row]);
       bool parse_ok = true;
       bool filt_found = false, filt_null = false; int64_t filt_v = 0;
+      bool d2_found = false, d2_null = false;
+      int64_t d2_v = 0;
       unsigned long long cell_pack[COPR_MAX_OUT_COLS];
       for (int j = 0; j < plan.n_out; j++) cell_pack[j] = 0xFFFFFull;
-      int needed = (plan.has_filter ? 1 : 0) + plan.n_out;
+      int needed = (plan.has_filter ? 1 : 0) + (plan.dec2_col_id ? 1 : 0) +
+                   plan.n_out;
       int found = 0;
 
       if (vlen > 1 && vp[0] == 128) {
@@ -2294,6 +2297,17 @@ k_scan_project(ScanPlan plan,
               if (vst == 0) filt_null = true;
               else if (!d_v2_int(r2.vals + s2, e2 - s2,
                                  plan.filter_col_unsigned, &filt_v))
+                parse_ok = false;
+            }
+          }
+          if (plan.dec2_col_id) {
+            uint32_t s2, e2;
+            int vst = d_v2_find(r2, plan.dec2_col_id, &s2, &e2);
+            if (vst >= 0) {
+              d2_found = true;
+              if (vst == 0) d2_null = true;
+              else if (!d_v2_int(r2.vals + s2, e2 - s2,
+                                 plan.dec2_col_unsigned, &d2_v))
                 parse_ok = false;
             }
           }
@@ -2318,6 +2332,13 @@ k_scan_project(ScanPlan plan,
           else parse_ok = false;
           found++;
         }
+        if (plan.dec2_col_id && !d2_found && cell_id == plan.dec2_col_id) {
+          d2_found = true;
+          if (cell.is_null) d2_null = true;
+          else if (cell.has_int) d2_v = cell.ival;
+          else parse_ok = false;
+          found++;
+        }
         for (int j = 0; j < plan.n_out; j++) {
           if (plan.out_is_handle[j] || cell_pack[j] != 0xFFFFFull) continue;
           if (cell_id == plan.out_col_ids[j]) {
@@ -2338,6 +2359,10 @@ k_scan_project(ScanPlan plan,
         if (plan.has_filter && po.filt_vals) {
           po.filt_vals[my_row] = filt_v;
           po.filt_state[my_row] = !filt_found ? 2 : (filt_null ? 1 : 0);
+        }
+        if (plan.dec2_col_id && po.dec2_vals) {
+          po.dec2_vals[my_row] = d2_v;
+          po.dec2_state[my_row] = !d2_found ? 2 : (d2_null ? 1 : 0);
         }
         if (keep) {
           for (int j = 0; j < plan.n_out; j++) {
