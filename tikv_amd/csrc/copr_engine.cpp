@@ -1334,9 +1334,15 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
       n_rows_out = 1;
     } else {
       /* ---- hash agg ---- */
+      /* pre-size by the total row count: avoids a doomed first pass on
+         near-distinct group keys; the device-side compaction keeps the
+         readback proportional to n_groups either way */
       uint64_t total_rows = 0;
-      for (uint32_t rg = 0; rg < n_regions; rg++) total_rows += regions[rg]->dev.n_kv;
+      for (uint32_t rg = 0; rg < n_regions; rg++)
+        total_rows += regions[rg]->dev.n_kv;
       uint32_t tsize = 1u << 16;
+      while (tsize < (1u << 27) && (uint64_t)tsize * 4 < total_rows)
+        tsize <<= 1;
       for (int attempt = 0; attempt < 4; attempt++) {
         HashAggTable ht{};
         hipError_t e = hipSuccess;
@@ -1412,17 +1418,19 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
           continue;
         }
         if (h_err[1]) { free_ht(); return SET_ERR(COPR_ERR_STORAGE, "row parse error on device"); }
-        /* copy back table */
-        std::vector<long long> h_keys(tsize);
-        std::vector<SimpleAggAcc> h_accs((size_t)tsize * pl.sp.n_aggs);
+        /* device-side compaction, then copy back only occupied slots */
+        std::vector<long long> h_keys;
+        std::vector<SimpleAggAcc> h_accs;
         std::vector<SimpleAggAcc> h_rsvd(2 * pl.sp.n_aggs);
-        ce = hipMemcpy(h_keys.data(), ht.keys, keys_b, hipMemcpyDeviceToHost);
-        if (ce == hipSuccess) ce = hipMemcpy(h_accs.data(), ht.accs, accs_b, hipMemcpyDeviceToHost);
-        if (ce == hipSuccess) ce = hipMemcpy(h_rsvd.data(), ht.reserved,
-                                             2 * pl.sp.n_aggs * sizeof(SimpleAggAcc),
-                                             hipMemcpyDeviceToHost);
+        int ng = dev_ht_compact(ht, tsize, pl.sp.n_aggs, eng->stream,
+                                &h_keys, &h_accs);
+        ce = hipMemcpy(h_rsvd.data(), ht.reserved,
+                       2 * pl.sp.n_aggs * sizeof(SimpleAggAcc),
+                       hipMemcpyDeviceToHost);
         free_ht();
-        if (ce != hipSuccess) return SET_ERR(COPR_ERR_INTERNAL, hipGetErrorString(ce));
+        if (ng == -2) return SET_ERR(COPR_ERR_OOM, "table compact alloc");
+        if (ng < 0 || ce != hipSuccess)
+          return SET_ERR(COPR_ERR_INTERNAL, "table compact failed");
         /* encode rows: occupied slots + reserved groups */
         size_t n_out_cols = pl.out_schema.size();
         auto emit_group = [&](const SimpleAggAcc *accs, bool gnull, int64_t gkey) {
@@ -1435,10 +1443,8 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
           }
           n_rows_out++;
         };
-        for (uint32_t s = 0; s < tsize; s++) {
-          if (h_keys[s] == (long long)0x8000000000000000ll) continue;
-          emit_group(&h_accs[(size_t)s * pl.sp.n_aggs], false, h_keys[s]);
-        }
+        for (size_t s = 0; s < h_keys.size(); s++)
+          emit_group(&h_accs[s * pl.sp.n_aggs], false, h_keys[s]);
         if (h_rsvd_seen[0]) emit_group(&h_rsvd[0], false, INT64_MIN);
         if (h_rsvd_seen[1]) emit_group(&h_rsvd[pl.sp.n_aggs], true, 0);
         break;

@@ -194,6 +194,9 @@ int dev_stream_agg(const ScanPlan &plan, const DevRegion &rgn, void *stream,
 int dev_topn_select(const ScanPlan &plan, const DevRegion &rgn,
                     uint64_t topn_n, int desc, void *stream,
                     std::vector<uint32_t> *winners);
+int dev_ht_compact(const HashAggTable &ht, uint32_t tsize, int n_aggs,
+                   void *stream, std::vector<long long> *h_keys,
+                   std::vector<SimpleAggAcc> *h_accs);
 int dev_bytes_agg(const ScanPlan &plan, const DevRegion &rgn, void *stream,
                   std::vector<SimpleAggAcc> *h_accs,
                   std::vector<uint64_t> *h_kofs, std::vector<uint32_t> *h_klen,

@@ -3483,6 +3483,82 @@ int dev_topn_select(const ScanPlan &plan, const DevRegion &rgn,
 }
 
 
+
+/* compact a global hash table's occupied slots (keys + accumulator rows)
+ * into dense arrays so the host readback scales with n_groups, not with
+ * the table size */
+__global__ static void k_ht_flags(const long long *keys, uint32_t tsize,
+                                  uint32_t *f) {
+  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < tsize)
+    f[i] = keys[i] != (long long)0x8000000000000000ll ? 1u : 0u;
+}
+
+__global__ static void k_ht_gather(const long long *keys,
+                                   const SimpleAggAcc *accs, uint32_t tsize,
+                                   const uint64_t *pos, int n_aggs,
+                                   long long *ck, SimpleAggAcc *caccs) {
+  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= tsize || keys[i] == (long long)0x8000000000000000ll) return;
+  uint64_t p = pos[i];
+  ck[p] = keys[i];
+  for (int a = 0; a < n_aggs; a++)
+    caccs[p * n_aggs + a] = accs[(uint64_t)i * n_aggs + a];
+}
+
+/* returns n_groups (>=0) with host vectors filled, or -1/-2 */
+int dev_ht_compact(const HashAggTable &ht, uint32_t tsize, int n_aggs,
+                   void *stream, std::vector<long long> *h_keys,
+                   std::vector<SimpleAggAcc> *h_accs) {
+  hipStream_t s = (hipStream_t)stream;
+  uint32_t *f = nullptr;
+  uint64_t *pos = nullptr;
+  long long *ck = nullptr;
+  SimpleAggAcc *caccs = nullptr;
+  void *tmp = nullptr;
+  size_t tmpb = 0;
+  auto freeall = [&]() {
+    hipFree(f); hipFree(pos); hipFree(ck); hipFree(caccs); hipFree(tmp);
+  };
+  hipError_t e = hipSuccess;
+  if (e == hipSuccess) e = hipMalloc(&f, (uint64_t)tsize * 4 + 4);
+  if (e == hipSuccess) e = hipMalloc(&pos, (uint64_t)tsize * 8 + 8);
+  if (e != hipSuccess) { freeall(); return -2; }
+  uint32_t blocks = (tsize + 255) / 256;
+  hipLaunchKernelGGL(k_ht_flags, dim3(blocks), dim3(256), 0, s, ht.keys,
+                     tsize, f);
+  hipcub::DeviceScan::ExclusiveSum(nullptr, tmpb, f, pos, (int)tsize, s);
+  if (hipMalloc(&tmp, tmpb) != hipSuccess) { freeall(); return -2; }
+  hipcub::DeviceScan::ExclusiveSum(tmp, tmpb, f, pos, (int)tsize, s);
+  uint64_t g = 0;
+  uint32_t lf = 0;
+  e = hipMemcpyAsync(&g, pos + (tsize - 1), 8, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess)
+    e = hipMemcpyAsync(&lf, f + (tsize - 1), 4, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  if (e != hipSuccess) { freeall(); return -1; }
+  g += lf;
+  h_keys->resize(g);
+  h_accs->resize(g * n_aggs);
+  if (g) {
+    if (e == hipSuccess) e = hipMalloc(&ck, g * 8);
+    if (e == hipSuccess)
+      e = hipMalloc(&caccs, g * n_aggs * sizeof(SimpleAggAcc));
+    if (e != hipSuccess) { freeall(); return -2; }
+    hipLaunchKernelGGL(k_ht_gather, dim3(blocks), dim3(256), 0, s, ht.keys,
+                       ht.accs, tsize, pos, n_aggs, ck, caccs);
+    e = hipMemcpyAsync(h_keys->data(), ck, g * 8, hipMemcpyDeviceToHost, s);
+    if (e == hipSuccess)
+      e = hipMemcpyAsync(h_accs->data(), caccs,
+                         g * n_aggs * sizeof(SimpleAggAcc),
+                         hipMemcpyDeviceToHost, s);
+    if (e == hipSuccess) e = hipStreamSynchronize(s);
+  }
+  freeall();
+  if (e != hipSuccess) return -1;
+  return (int)g;
+}
+
 /* ---- bytes-group hash agg pipeline (SlowHashAggregationImpl) ---- */
 __global__ static void k_compact_bytes(const uint8_t *st, const uint64_t *gh,
                                        const uint64_t *go, const uint32_t *gl,
