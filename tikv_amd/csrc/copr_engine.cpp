@@ -1334,15 +1334,12 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
       n_rows_out = 1;
     } else {
       /* ---- hash agg ---- */
-      /* pre-size by the total row count: avoids a doomed first pass on
-         near-distinct group keys; the device-side compaction keeps the
-         readback proportional to n_groups either way */
-      uint64_t total_rows = 0;
-      for (uint32_t rg = 0; rg < n_regions; rg++)
-        total_rows += regions[rg]->dev.n_kv;
+      /* start small and grow x16 on overflow: GB-scale tables cost more
+         in per-call allocation and init than a retry pass saves (measured:
+         pre-sizing by row count regressed low-cardinality workloads 16x).
+         The device-side compaction keeps the readback proportional to
+         n_groups when the table does grow. */
       uint32_t tsize = 1u << 16;
-      while (tsize < (1u << 27) && (uint64_t)tsize * 4 < total_rows)
-        tsize <<= 1;
       for (int attempt = 0; attempt < 4; attempt++) {
         HashAggTable ht{};
         hipError_t e = hipSuccess;
