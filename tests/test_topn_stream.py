@@ -591,3 +591,38 @@ def test_topn_filter_col_in_output_gpu(engine):
             rgn.close()
     finally:
         g.close()
+
+
+@pytest.mark.gpu
+def test_chunked_paging_and_misc_aggs_gpu(engine):
+    """paging + TypeChunk coexist (chunks follow the ladder up to the
+    resume point); avg/bit aggregates through the chunk encoder."""
+    orc = _orc()
+    g = tikv_amd.GenRegion(config_index=1, n_rows=70001, table_id=5)
+    try:
+        rgn = engine.region(g)
+        try:
+            cols = [tikv_amd.Col(i) for i in range(1, 17)]
+            sel = tikv_amd.cmp_col_const(3, F.SIG_GT_INT, 0)
+            req = (tikv_amd.DagSelect(cols).where(sel).paging(500)
+                   .output([1, 5]).chunked().build())
+            gd, gr, gresume = engine.dag_run(req, [rgn], with_resume=True)
+            od, orows, oresume = orc.dag_run(req, g.keys, g.key_offs, g.vals,
+                                             g.val_offs, g.n_kv,
+                                             with_resume=True)
+            assert (gr, gresume) == (orows, oresume)
+            assert gd == od
+            # avg + bit ops, chunked simple agg
+            req2 = (tikv_amd.DagSelect(cols)
+                    .simple_agg([tikv_amd.avg_col(2),
+                                 tikv_amd.bit_op(F.AGG_BIT_XOR, 4),
+                                 tikv_amd.bit_op(F.AGG_BIT_AND, 4)])
+                    .chunked().build())
+            gd2, gr2, _ = engine.dag_run(req2, [rgn])
+            od2, or2 = orc.dag_run(req2, g.keys, g.key_offs, g.vals,
+                                   g.val_offs, g.n_kv)
+            assert (gr2, gd2) == (or2, od2)
+        finally:
+            rgn.close()
+    finally:
+        g.close()
