@@ -606,7 +606,7 @@ def test_chunked_paging_and_misc_aggs_gpu(engine):
             sel = tikv_amd.cmp_col_const(3, F.SIG_GT_INT, 0)
             req = (tikv_amd.DagSelect(cols).where(sel).paging(500)
                    .output([1, 5]).chunked().build())
-            gd, gr, gresume = engine.dag_run(req, [rgn], with_resume=True)
+            gd, gr, _, gresume = engine.dag_run(req, [rgn], with_resume=True)
             od, orows, oresume = orc.dag_run(req, g.keys, g.key_offs, g.vals,
                                              g.val_offs, g.n_kv,
                                              with_resume=True)
