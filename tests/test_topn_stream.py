@@ -626,3 +626,59 @@ def test_chunked_paging_and_misc_aggs_gpu(engine):
             rgn.close()
     finally:
         g.close()
+
+
+# ---- two ANDed selection conditions ------------------------------------
+def test_two_conditions_oracle():
+    orc = _orc()
+    rows = [{1: a, 2: b} for a, b in
+            [(5, 1), (15, 1), (5, 9), (None, 1), (7, None), (12, 3)]]
+    k, ko, v, vo, n, keep = region_of(rows)
+    cols = [tikv_amd.Col(1), tikv_amd.Col(2)]
+    req = (tikv_amd.DagSelect(cols)
+           .where(tikv_amd.cmp_col_const(0, F.SIG_LT_INT, 10),
+                  tikv_amd.cmp_col_const(1, F.SIG_LT_INT, 5))
+           .simple_agg([tikv_amd.count_star()]).build())
+    data, nrows = orc.dag_run(req, k, ko, v, vo, n)
+    cnt, p = dec_int(data, 0)
+    assert cnt == 1 and nrows == 1   # only (5,1)
+
+
+@pytest.mark.gpu
+def test_two_conditions_gpu_parity(engine):
+    orc = _orc()
+    g = tikv_amd.GenRegion(config_index=1, n_rows=120001, table_id=5)
+    try:
+        rgn = engine.region(g)
+        try:
+            cols = [tikv_amd.Col(i) for i in range(1, 17)]
+            c1 = tikv_amd.cmp_col_const(3, F.SIG_GT_INT, -500000000)
+            c2 = tikv_amd.cmp_col_const(7, F.SIG_LE_INT, 250000000)
+            # count with both conjuncts (generic collect path)
+            req = (tikv_amd.DagSelect(cols).where(c1, c2)
+                   .simple_agg([tikv_amd.count_star(),
+                                tikv_amd.sum_col(1)]).build())
+            gd, gr, _ = engine.dag_run(req, [rgn])
+            od, orows = orc.dag_run(req, g.keys, g.key_offs, g.vals,
+                                    g.val_offs, g.n_kv)
+            assert (gr, gd) == (orows, od)
+            # project: BOTH predicate columns in the output, decoded form
+            req2 = (tikv_amd.DagSelect(cols).where(c1, c2)
+                    .output([3, 7, 10]).build())
+            gd2, gr2, _ = engine.dag_run(req2, [rgn])
+            od2, or2 = orc.dag_run(req2, g.keys, g.key_offs, g.vals,
+                                   g.val_offs, g.n_kv)
+            assert (gr2, gd2) == (or2, od2)
+            # hash agg with two conjuncts
+            req3 = (tikv_amd.DagSelect(cols).where(c1, c2)
+                    .hash_agg([tikv_amd.count_star()],
+                              tikv_amd.Expr().col(5)).build())
+            gd3, gr3, _ = engine.dag_run(req3, [rgn])
+            od3, or3 = orc.dag_run(req3, g.keys, g.key_offs, g.vals,
+                                   g.val_offs, g.n_kv)
+            assert gr3 == or3
+            assert set(split_rows(gd3, 2)) == set(split_rows(od3, 2))
+        finally:
+            rgn.close()
+    finally:
+        g.close()

@@ -240,6 +240,7 @@ struct HostPlan {
   int topn_desc = 0;
   int topn_off = -1;
   int dec2_col_offset = -1;
+  int filter2_col_offset = -1;
   uint64_t limit = UINT64_MAX;
 };
 
@@ -371,6 +372,8 @@ static void wire_celldir(ScanPlan *sp, const DevRegion &dev) {
   bool ok = true;
   if (sp->has_filter && (sp->filter_col_id < 1 || sp->filter_col_id > 16))
     ok = false;
+  if (sp->filter2_on && (sp->filter2_col_id < 1 || sp->filter2_col_id > 16))
+    ok = false;
   if ((sp->mode == 2 || sp->mode == 3 || sp->mode == 4 ||
        sp->group_col_id != 0) &&
       (sp->group_col_id < 1 || sp->group_col_id > 16))
@@ -407,11 +410,33 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
     const CoprExecutor &ex = req->executors[e];
     switch (ex.kind) {
       case COPR_EXEC_SELECTION: {
-        if (sp.has_filter || ex.n_conditions != 1)
-          return SET_ERR(COPR_ERR_UNSUPPORTED, "one selection condition supported");
+        /* conditions are ANDed (selection_executor.rs:86); up to two int
+           cmp(col, const) conjuncts run natively */
+        if (sp.has_filter || ex.n_conditions < 1 || ex.n_conditions > 2)
+          return SET_ERR(COPR_ERR_UNSUPPORTED,
+                         "up to two selection conditions supported");
         copr_status st = match_filter(ex.conditions[0], *pl, &sp);
         if (st != COPR_OK)
           return SET_ERR(st, "unsupported selection condition shape");
+        if (ex.n_conditions == 2) {
+          ScanPlan s2{};
+          HostPlan p2;
+          p2.cols = pl->cols;
+          p2.sp.index_mode = sp.index_mode;
+          st = match_filter(ex.conditions[1], p2, &s2);
+          if (st != COPR_OK)
+            return SET_ERR(st, "unsupported selection condition shape");
+          sp.filter2_on = 1;
+          sp.filter2_col_id = s2.filter_col_id;
+          sp.filter2_cmp = s2.filter_cmp;
+          sp.filter2_const = s2.filter_const;
+          sp.filter2_col_unsigned = s2.filter_col_unsigned;
+          sp.filter2_const_unsigned = s2.filter_const_unsigned;
+          sp.filter2_const_null = s2.filter_const_null;
+          sp.filter2_missing_null = s2.filter_missing_null;
+          sp.filter2_missing_val = s2.filter_missing_val;
+          pl->filter2_col_offset = p2.filter_col_offset;
+        }
         break;
       }
       case COPR_EXEC_SIMPLE_AGG: case COPR_EXEC_FAST_HASH_AGG:
@@ -1051,6 +1076,16 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
     } else {
       pl.sp.filter_missing_null = 1;
     }
+  }
+  if (pl.sp.filter2_on && pl.sp.mode == 0 && !pl.has_topn &&
+      pl.filter2_col_offset >= 0) {
+    /* expression-decoded output for the second predicate column */
+    const CoprColumnInfo &ci = pl.cols[pl.filter2_col_offset];
+    pl.dec2_col_offset = pl.filter2_col_offset;
+    pl.sp.dec2_col_id = ci.column_id;
+    pl.sp.dec2_col_unsigned = (ci.ft.flag & COPR_FLAG_UNSIGNED) ? 1 : 0;
+    pl.sp.dec2_missing_null = pl.sp.filter2_missing_null;
+    pl.sp.dec2_missing_val = pl.sp.filter2_missing_val;
   }
   if (eng->dec2_col_off >= 0 && !pl.has_agg && !pl.has_topn &&
       (size_t)eng->dec2_col_off < pl.cols.size() &&

@@ -79,6 +79,19 @@ struct ScanPlan {
      fill (table_scan_executor.rs:456-483) decoded */
   int32_t filter_missing_null;   /* 1 => NULL */
   int64_t filter_missing_val;
+  /* second ANDed selection conjunct (selection_executor.rs:86 ANDs all
+     conditions); same shape as the first filter. FASTFC and the dir
+     single-plane shortcut require a single conjunct, so two-filter
+     requests take the generic collect path. */
+  int32_t filter2_on;
+  int64_t filter2_col_id;
+  int32_t filter2_cmp;
+  int64_t filter2_const;
+  int32_t filter2_col_unsigned;
+  int32_t filter2_const_unsigned;
+  int32_t filter2_const_null;
+  int32_t filter2_missing_null;
+  int64_t filter2_missing_val;
   /* 1 = keep every row but still export filt_vals/filt_state: used for a
      column an upstream expression decoded in place (e.g. the TopN order
      column) so the response encodes it in DECODED form

@@ -476,7 +476,10 @@ __device__ static inline bool d_v2_collect(const ScanPlan &plan,
                                            int64_t *filt_v, bool *grp_found,
                                            bool *grp_null, int64_t *grp_v,
                                            AggColView (&cols)[NAGGS],
-                                           bool parse_grp = true) {
+                                           bool parse_grp = true,
+                                           bool *f2_found = nullptr,
+                                           bool *f2_null = nullptr,
+                                           int64_t *f2_v = nullptr) {
   V2Row r;
   if (!d_v2_parse(vp, vlen, &r)) return false;
   if (plan.has_filter) {
@@ -486,6 +489,16 @@ __device__ static inline bool d_v2_collect(const ScanPlan &plan,
       *filt_found = true;
       if (st == 0) *filt_null = true;
       else if (!d_v2_int(r.vals + s, e - s, plan.filter_col_unsigned, filt_v))
+        return false;
+    }
+  }
+  if (plan.filter2_on && f2_found) {
+    uint32_t s, e;
+    int st = d_v2_find(r, plan.filter2_col_id, &s, &e);
+    if (st >= 0) {
+      *f2_found = true;
+      if (st == 0) *f2_null = true;
+      else if (!d_v2_int(r.vals + s, e - s, plan.filter2_col_unsigned, f2_v))
         return false;
     }
   }
@@ -616,6 +629,20 @@ __device__ static inline bool d_filter_keep(const ScanPlan &plan, bool found,
 }
 
 
+__device__ static inline bool d_filter2_keep(const ScanPlan &plan,
+                                             bool found, bool is_null,
+                                             int64_t v) {
+  if (!plan.filter2_on) return true;
+  if (!found) {
+    if (plan.filter2_missing_null) is_null = true;
+    else { v = plan.filter2_missing_val; is_null = false; }
+  }
+  if (plan.filter2_const_null || is_null) return false;
+  return d_cmp_res(plan.filter2_cmp,
+                   d_cmp_int(v, plan.filter2_const, plan.filter2_col_unsigned,
+                             plan.filter2_const_unsigned));
+}
+
 /* 128-bit signed accumulate via two u64 atomics (carry trick) */
 __device__ static inline void atomic_add_i128(unsigned long long *lo,
                                               unsigned long long *hi, int64_t x) {
@@ -711,7 +738,7 @@ k_scan_agg(ScanPlan plan,
   #pragma unroll
   for (int a = 0; a < NAGGS; a++) { l_cnt[a] = 0; l_lo[a] = 0; l_hi[a] = 0; }
 
-  int needed = (plan.has_filter ? 1 : 0) + (IS_HASH ? 1 : 0);
+  int needed = (plan.has_filter ? 1 : 0) + (plan.filter2_on ? 1 : 0) + (IS_HASH ? 1 : 0);
   #pragma unroll
   for (int a = 0; a < NAGGS; a++)
     if (plan.aggs[a].kind != DAGG_COUNT_ROWS) needed++;
@@ -754,6 +781,7 @@ k_scan_agg(ScanPlan plan,
       bool parse_ok = true;
 
       bool filt_found = false, filt_null = false; int64_t filt_v = 0;
+      bool f2_found = false, f2_null = false; int64_t f2_v = 0;
       bool grp_found = false, grp_null = false; int64_t grp_v = 0;
       AggColView cols[NAGGS];
       #pragma unroll
@@ -774,10 +802,12 @@ k_scan_agg(ScanPlan plan,
         if (plan.celldir && vp[0] != 128) {
           const uint8_t *db = plan.celldir;
           const uint64_t dn = plan.celldir_n;
-          uint32_t d_f = 0xFFu, d_g = 0xFFu, d_a[NAGGS];
+          uint32_t d_f = 0xFFu, d_f2 = 0xFFu, d_g = 0xFFu, d_a[NAGGS];
           bool seq = false;
           if (plan.has_filter)
             d_f = db[(uint64_t)(plan.filter_col_id - 1) * dn + my_row];
+          if (plan.filter2_on)
+            d_f2 = db[(uint64_t)(plan.filter2_col_id - 1) * dn + my_row];
           if (IS_HASH)
             d_g = db[(uint64_t)(plan.group_col_id - 1) * dn + my_row];
           #pragma unroll
@@ -786,7 +816,7 @@ k_scan_agg(ScanPlan plan,
             if (plan.aggs[a].kind != DAGG_COUNT_ROWS)
               d_a[a] = db[(uint64_t)(plan.aggs[a].col_id - 1) * dn + my_row];
           }
-          seq = (d_f == 0xFEu) | (d_g == 0xFEu);
+          seq = (d_f == 0xFEu) | (d_f2 == 0xFEu) | (d_g == 0xFEu);
           #pragma unroll
           for (int a = 0; a < NAGGS; a++) seq |= (d_a[a] == 0xFEu);
           if (!seq) {
@@ -799,6 +829,16 @@ k_scan_agg(ScanPlan plan,
                 filt_found = true;
                 if (cell.is_null) filt_null = true;
                 else if (cell.has_int) filt_v = cell.ival;
+                else parse_ok = false;
+              } else parse_ok = false;
+            }
+            if (plan.filter2_on && d_f2 != 0xFFu) {
+              pos = d_f2;
+              if (next_cell(vp, vlen, &pos, &cid, &coff, &cell) &&
+                  cid == plan.filter2_col_id) {
+                f2_found = true;
+                if (cell.is_null) f2_null = true;
+                else if (cell.has_int) f2_v = cell.ival;
                 else parse_ok = false;
               } else parse_ok = false;
             }
@@ -835,7 +875,8 @@ k_scan_agg(ScanPlan plan,
         parse_ok = d_v2_collect<NAGGS, IS_HASH>(plan, vp, vlen,
                                                 &filt_found, &filt_null, &filt_v,
                                                 &grp_found, &grp_null, &grp_v,
-                                                cols);
+                                                cols, true,
+                                                &f2_found, &f2_null, &f2_v);
         } else {
         uint32_t pos = 0;
         while (pos < vlen) {
@@ -850,6 +891,13 @@ k_scan_agg(ScanPlan plan,
             filt_found = true;
             if (cell.is_null) filt_null = true;
             else if (cell.has_int) filt_v = cell.ival;
+            else parse_ok = false;
+            found++;
+          }
+          if (plan.filter2_on && !f2_found && cell_id == plan.filter2_col_id) {
+            f2_found = true;
+            if (cell.is_null) f2_null = true;
+            else if (cell.has_int) f2_v = cell.ival;
             else parse_ok = false;
             found++;
           }
@@ -880,7 +928,8 @@ k_scan_agg(ScanPlan plan,
 
       if (!parse_ok) {
         any_parse_err = true;
-      } else if (d_filter_keep(plan, filt_found, filt_null, filt_v)) {
+      } else if (d_filter_keep(plan, filt_found, filt_null, filt_v) &&
+                 d_filter2_keep(plan, f2_found, f2_null, f2_v)) {
         SimpleAggAcc *acc_base = nullptr;
         if (IS_HASH) {
           if (!grp_found || grp_null) {
@@ -1119,7 +1168,7 @@ k_scan_agg_pipe(ScanPlan plan,
   #pragma unroll
   for (int a = 0; a < NAGGS; a++) { l_cnt[a] = 0; l_lo[a] = 0; l_hi[a] = 0; }
 
-  int needed = (plan.has_filter ? 1 : 0) + (IS_HASH ? 1 : 0);
+  int needed = (plan.has_filter ? 1 : 0) + (plan.filter2_on ? 1 : 0) + (IS_HASH ? 1 : 0);
   #pragma unroll
   for (int a = 0; a < NAGGS; a++)
     if (plan.aggs[a].kind != DAGG_COUNT_ROWS) needed++;
@@ -1321,6 +1370,7 @@ k_scan_agg_pipe(ScanPlan plan,
       bool parse_ok = true;
 
       bool filt_found = false, filt_null = false; int64_t filt_v = 0;
+      bool f2_found = false, f2_null = false; int64_t f2_v = 0;
       bool grp_found = false, grp_null = false; int64_t grp_v = 0;
       AggColView cols[NAGGS];
       #pragma unroll
@@ -1341,10 +1391,12 @@ k_scan_agg_pipe(ScanPlan plan,
         if (plan.celldir && vp[0] != 128) {
           const uint8_t *db = plan.celldir;
           const uint64_t dn = plan.celldir_n;
-          uint32_t d_f = 0xFFu, d_g = 0xFFu, d_a[NAGGS];
+          uint32_t d_f = 0xFFu, d_f2 = 0xFFu, d_g = 0xFFu, d_a[NAGGS];
           bool seq = false;
           if (plan.has_filter)
             d_f = db[(uint64_t)(plan.filter_col_id - 1) * dn + my_row];
+          if (plan.filter2_on)
+            d_f2 = db[(uint64_t)(plan.filter2_col_id - 1) * dn + my_row];
           if (IS_HASH)
             d_g = db[(uint64_t)(plan.group_col_id - 1) * dn + my_row];
           #pragma unroll
@@ -1353,7 +1405,7 @@ k_scan_agg_pipe(ScanPlan plan,
             if (plan.aggs[a].kind != DAGG_COUNT_ROWS)
               d_a[a] = db[(uint64_t)(plan.aggs[a].col_id - 1) * dn + my_row];
           }
-          seq = (d_f == 0xFEu) | (d_g == 0xFEu);
+          seq = (d_f == 0xFEu) | (d_f2 == 0xFEu) | (d_g == 0xFEu);
           #pragma unroll
           for (int a = 0; a < NAGGS; a++) seq |= (d_a[a] == 0xFEu);
           if (!seq) {
@@ -1366,6 +1418,16 @@ k_scan_agg_pipe(ScanPlan plan,
                 filt_found = true;
                 if (cell.is_null) filt_null = true;
                 else if (cell.has_int) filt_v = cell.ival;
+                else parse_ok = false;
+              } else parse_ok = false;
+            }
+            if (plan.filter2_on && d_f2 != 0xFFu) {
+              pos = d_f2;
+              if (next_cell(vp, vlen, &pos, &cid, &coff, &cell) &&
+                  cid == plan.filter2_col_id) {
+                f2_found = true;
+                if (cell.is_null) f2_null = true;
+                else if (cell.has_int) f2_v = cell.ival;
                 else parse_ok = false;
               } else parse_ok = false;
             }
@@ -1402,7 +1464,8 @@ k_scan_agg_pipe(ScanPlan plan,
         parse_ok = d_v2_collect<NAGGS, IS_HASH>(plan, vp, vlen,
                                                 &filt_found, &filt_null, &filt_v,
                                                 &grp_found, &grp_null, &grp_v,
-                                                cols);
+                                                cols, true,
+                                                &f2_found, &f2_null, &f2_v);
         } else {
         uint32_t pos = 0;
         while (pos < vlen) {
@@ -1417,6 +1480,13 @@ k_scan_agg_pipe(ScanPlan plan,
             filt_found = true;
             if (cell.is_null) filt_null = true;
             else if (cell.has_int) filt_v = cell.ival;
+            else parse_ok = false;
+            found++;
+          }
+          if (plan.filter2_on && !f2_found && cell_id == plan.filter2_col_id) {
+            f2_found = true;
+            if (cell.is_null) f2_null = true;
+            else if (cell.has_int) f2_v = cell.ival;
             else parse_ok = false;
             found++;
           }
@@ -1447,7 +1517,8 @@ k_scan_agg_pipe(ScanPlan plan,
 
       if (!parse_ok) {
         any_parse_err = true;
-      } else if (d_filter_keep(plan, filt_found, filt_null, filt_v)) {
+      } else if (d_filter_keep(plan, filt_found, filt_null, filt_v) &&
+                 d_filter2_keep(plan, f2_found, f2_null, f2_v)) {
         SimpleAggAcc *acc_base = nullptr;
         if (IS_HASH) {
           if (!grp_found || grp_null) {
@@ -2354,7 +2425,8 @@ k_scan_project(ScanPlan plan,
       if (!parse_ok) {
         any_parse_err = true;
       } else {
-        bool keep = d_filter_keep(plan, filt_found, filt_null, filt_v);
+        bool keep = d_filter_keep(plan, filt_found, filt_null, filt_v) &&
+                    d_filter2_keep(plan, d2_found, d2_null, d2_v);
         po.keep[my_row] = keep ? 1 : 0;
         if (plan.has_filter && po.filt_vals) {
           po.filt_vals[my_row] = filt_v;
@@ -2751,7 +2823,8 @@ static int launch_agg_pipe(const ScanPlan &plan, const DevRegion &rgn,
                            SimpleAggAcc *d_simple, HashAggTable ht,
                            hipStream_t s, uint32_t grid) {
   if (!IS_HASH && !plan.index_mode && plan.n_aggs == 1 &&
-      plan.aggs[0].kind == DAGG_COUNT_ROWS && plan.has_filter) {
+      plan.aggs[0].kind == DAGG_COUNT_ROWS && plan.has_filter &&
+      !plan.filter2_on) {
     if (getenv("COPR_DIRECT")) {
       uint64_t n_blk = (rgn.n_kv + THREADS - 1) / THREADS;
       uint32_t dgrid = (uint32_t)(n_blk < 8192 ? n_blk : 8192);
@@ -2907,6 +2980,7 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
       const bool GBYTES = eo.ghash != nullptr;
       uint64_t g_h = 0, g_o = 0; uint32_t g_l = 0;
       bool filt_found = false, filt_null = false; int64_t filt_v = 0;
+      bool f2_found = false, f2_null = false; int64_t f2_v = 0;
       bool grp_found = false, grp_null = false; int64_t grp_v = 0;
       AggColView cols[NAGGS > 0 ? NAGGS : 1];
       #pragma unroll
@@ -2917,10 +2991,13 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
         if (plan.celldir && vp[0] != 128) {
           const uint8_t *db = plan.celldir;
           const uint64_t dn = plan.celldir_n;
-          uint32_t d_f = 0xFFu, d_g = 0xFFu, d_a[NAGGS > 0 ? NAGGS : 1];
+          uint32_t d_f = 0xFFu, d_f2 = 0xFFu, d_g = 0xFFu,
+                   d_a[NAGGS > 0 ? NAGGS : 1];
           bool seq = false;
           if (plan.has_filter)
             d_f = db[(uint64_t)(plan.filter_col_id - 1) * dn + my_row];
+          if (plan.filter2_on)
+            d_f2 = db[(uint64_t)(plan.filter2_col_id - 1) * dn + my_row];
           d_g = db[(uint64_t)(plan.group_col_id - 1) * dn + my_row];
           #pragma unroll
           for (int a = 0; a < NAGGS; a++) {
@@ -2928,7 +3005,7 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
             if (plan.aggs[a].kind != DAGG_COUNT_ROWS)
               d_a[a] = db[(uint64_t)(plan.aggs[a].col_id - 1) * dn + my_row];
           }
-          seq = (d_f == 0xFEu) | (d_g == 0xFEu);
+          seq = (d_f == 0xFEu) | (d_f2 == 0xFEu) | (d_g == 0xFEu);
           #pragma unroll
           for (int a = 0; a < NAGGS; a++) seq |= (d_a[a] == 0xFEu);
           if (!seq) {
@@ -2941,6 +3018,16 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
                 filt_found = true;
                 if (cell.is_null) filt_null = true;
                 else if (cell.has_int) filt_v = cell.ival;
+                else parse_ok = false;
+              } else parse_ok = false;
+            }
+            if (plan.filter2_on && d_f2 != 0xFFu) {
+              pos = d_f2;
+              if (next_cell(vp, vlen, &pos, &cid, &coff, &cell) &&
+                  cid == plan.filter2_col_id) {
+                f2_found = true;
+                if (cell.is_null) f2_null = true;
+                else if (cell.has_int) f2_v = cell.ival;
                 else parse_ok = false;
               } else parse_ok = false;
             }
@@ -2983,7 +3070,8 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
         } else if (vp[0] == 128) {
           parse_ok = d_v2_collect<(NAGGS > 0 ? NAGGS : 1), true>(
               plan, vp, vlen, &filt_found, &filt_null, &filt_v, &grp_found,
-              &grp_null, &grp_v, cols, /*parse_grp=*/!GBYTES);
+              &grp_null, &grp_v, cols, /*parse_grp=*/!GBYTES,
+              &f2_found, &f2_null, &f2_v);
           if (parse_ok && GBYTES) {
             V2Row r2;
             if (d_v2_parse(vp, vlen, &r2)) {
@@ -3004,7 +3092,7 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
           }
         } else {
           uint32_t pos = 0;
-          int needed = (plan.has_filter ? 1 : 0) + 1;
+          int needed = (plan.has_filter ? 1 : 0) + (plan.filter2_on ? 1 : 0) + 1;
           #pragma unroll
           for (int a = 0; a < NAGGS; a++)
             if (plan.aggs[a].kind != DAGG_COUNT_ROWS) needed++;
@@ -3022,6 +3110,14 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
               filt_found = true;
               if (cell.is_null) filt_null = true;
               else if (cell.has_int) filt_v = cell.ival;
+              else parse_ok = false;
+              found++;
+            }
+            if (plan.filter2_on && !f2_found &&
+                cell_id == plan.filter2_col_id) {
+              f2_found = true;
+              if (cell.is_null) f2_null = true;
+              else if (cell.has_int) f2_v = cell.ival;
               else parse_ok = false;
               found++;
             }
@@ -3062,7 +3158,8 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
       uint8_t s = 0;
       if (!parse_ok) {
         any_err = true;
-      } else if (d_filter_keep(plan, filt_found, filt_null, filt_v)) {
+      } else if (d_filter_keep(plan, filt_found, filt_null, filt_v) &&
+                 d_filter2_keep(plan, f2_found, f2_null, f2_v)) {
         s = (grp_found && !grp_null) ? 2 : 1;
       }
       eo.st[my_row] = s;

@@ -100,16 +100,20 @@ class DagSelect:
         ex.n_columns = len(columns)
         self.executors.append(ex)
 
-    def where(self, expr):
-        """Selection with one condition (an Expr)."""
-        e, keep = expr.build()
-        self._keep += keep
-        conds = (F.CoprExpr * 1)(e)
+    def where(self, *exprs):
+        """Selection; multiple conditions are ANDed
+        (selection_executor.rs:86)."""
+        built = []
+        for expr in exprs:
+            e, keep = expr.build()
+            self._keep += keep
+            built.append(e)
+        conds = (F.CoprExpr * len(built))(*built)
         self._keep.append(conds)
         ex = F.CoprExecutor()
         ex.kind = F.EXEC_SELECTION
         ex.conditions = conds
-        ex.n_conditions = 1
+        ex.n_conditions = len(built)
         self.executors.append(ex)
         return self
 
