@@ -682,3 +682,68 @@ def test_two_conditions_gpu_parity(engine):
             rgn.close()
     finally:
         g.close()
+
+
+def test_chunk_wire_format_golden():
+    """hand-computed TypeChunk bytes (chunk/column.rs:1052-1071): pins the
+    wire layout independently of the python decoder used elsewhere."""
+    orc = _orc()
+    rows = [{1: 5, 2: b"ab"}, {1: None, 2: b"c"}, {1: -2, 2: b"ab"}]
+    k, ko, v, vo, n, keep = region_of_mixed(rows)
+    cols = [tikv_amd.Col(1), tikv_amd.Col(2, tp=F.TP_VARCHAR)]
+    req = tikv_amd.DagSelect(cols).chunked().build()
+    data, nrows = orc.dag_run(req, k, ko, v, vo, n)
+    assert nrows == 3
+    # one chunk (first ladder batch of 32 covers all 3 rows), two columns
+    exp = bytearray()
+    # col 0: i64 fixed; row1 NULL -> bitmap present, bit set = NOT null
+    exp += (3).to_bytes(4, "little") + (1).to_bytes(4, "little")
+    exp.append(0b00000101)
+    exp += (5).to_bytes(8, "little", signed=True)
+    exp += (0).to_bytes(8, "little")
+    exp += (-2).to_bytes(8, "little", signed=True)
+    # col 1: var-size, no nulls -> no bitmap; offsets i64le x4; payloads
+    exp += (3).to_bytes(4, "little") + (0).to_bytes(4, "little")
+    for off in (0, 2, 3, 5):
+        exp += off.to_bytes(8, "little")
+    exp += b"abcab"
+    assert data == bytes(exp)
+
+
+@pytest.mark.gpu
+def test_executor_edge_cases_gpu(engine):
+    """empty-result edges across the new executors."""
+    orc = _orc()
+    g = tikv_amd.GenRegion(config_index=1, n_rows=5001, table_id=5)
+    try:
+        rgn = engine.region(g)
+        try:
+            cols = [tikv_amd.Col(i) for i in range(1, 17)]
+            never = tikv_amd.cmp_col_const(3, F.SIG_GT_INT, 2 * 10**9)
+            reqs = [
+                # topn with nothing surviving
+                (tikv_amd.DagSelect(cols).where(never)
+                 .topn(tikv_amd.Expr().col(4), 10).build()),
+                # stream agg with nothing surviving
+                (tikv_amd.DagSelect(cols).where(never)
+                 .stream_agg([tikv_amd.count_star()],
+                             tikv_amd.Expr().col(0)).build()),
+                # simple agg + FIRST with nothing surviving (one zero row)
+                (tikv_amd.DagSelect(cols).where(never)
+                 .simple_agg([tikv_amd.first_col(1),
+                              tikv_amd.count_star()]).build()),
+                # topn n larger than the row count
+                (tikv_amd.DagSelect(cols)
+                 .topn(tikv_amd.Expr().col(4), 100000).output([2]).build()),
+                # limit 0
+                (tikv_amd.DagSelect(cols).limit(0).build()),
+            ]
+            for req in reqs:
+                gd, gr, _ = engine.dag_run(req, [rgn])
+                od, orows = orc.dag_run(req, g.keys, g.key_offs, g.vals,
+                                        g.val_offs, g.n_kv)
+                assert (gr, gd) == (orows, od)
+        finally:
+            rgn.close()
+    finally:
+        g.close()
