@@ -747,3 +747,82 @@ def test_executor_edge_cases_gpu(engine):
             rgn.close()
     finally:
         g.close()
+
+
+# ---- general RPN selection predicates ----------------------------------
+def test_rpn_oracle_shapes():
+    orc = _orc()
+    rows = [{1: a, 2: b} for a, b in
+            [(5, 3), (10, -4), (None, 7), (2, None), (-6, -6), (0, 0)]]
+    k, ko, v, vo, n, keep = region_of(rows)
+    cols = [tikv_amd.Col(1), tikv_amd.Col(2)]
+
+    def count_with(sel):
+        req = (tikv_amd.DagSelect(cols).where(sel)
+               .simple_agg([tikv_amd.count_star()]).build())
+        data, nrows = orc.dag_run(req, k, ko, v, vo, n)
+        return dec_int(data, 0)[0]
+
+    # col0 + col1 > 0
+    e = (tikv_amd.Expr().col(0).col(1).func(F.SIG_PLUS_INT)
+         .const_int(0).func(F.SIG_GT_INT))
+    assert count_with(e) == 2       # (5,3)=8, (10,-4)=6
+    # NOT(col0 < col1)
+    e = (tikv_amd.Expr().col(0).col(1).func(F.SIG_LT_INT)
+         .func(F.SIG_UNARY_NOT, 1))
+    assert count_with(e) == 4       # (5,3), (10,-4), (-6,-6), (0,0)
+    # col0 IS NULL
+    e = tikv_amd.Expr().col(0).func(F.SIG_INT_IS_NULL, 1)
+    assert count_with(e) == 1
+    # col0 > 0 OR col1 > 0
+    e = (tikv_amd.Expr().col(0).const_int(0).func(F.SIG_GT_INT)
+         .col(1).const_int(0).func(F.SIG_GT_INT).func(F.SIG_LOGICAL_OR))
+    assert count_with(e) == 4       # (5,3), (10,-4), (None,7), (2,None)
+
+
+@pytest.mark.gpu
+def test_rpn_gpu_parity(engine):
+    orc = _orc()
+    g = tikv_amd.GenRegion(config_index=1, n_rows=90001, table_id=5)
+    try:
+        rgn = engine.region(g)
+        try:
+            cols = [tikv_amd.Col(i) for i in range(1, 17)]
+            sels = [
+                # col3 + col7 > 0 (two columns, arithmetic)
+                (tikv_amd.Expr().col(2).col(6).func(F.SIG_PLUS_INT)
+                 .const_int(0).func(F.SIG_GT_INT)),
+                # NOT(col3 < col7) (column-vs-column compare)
+                (tikv_amd.Expr().col(2).col(6).func(F.SIG_LT_INT)
+                 .func(F.SIG_UNARY_NOT, 1)),
+                # range on ONE column through the RPN path: a < col3 AND
+                # col3 < b (same column twice -> one capture slot)
+                (tikv_amd.Expr()
+                 .const_int(-600000000).col(2).func(F.SIG_LT_INT)
+                 .col(2).const_int(600000000).func(F.SIG_LT_INT)
+                 .func(F.SIG_LOGICAL_AND)),
+                # col3 IS TRUE
+                tikv_amd.Expr().col(2).func(F.SIG_INT_IS_TRUE, 1),
+            ]
+            for sel in sels:
+                req = (tikv_amd.DagSelect(cols).where(sel)
+                       .simple_agg([tikv_amd.count_star(),
+                                    tikv_amd.sum_col(5)]).build())
+                gd, gr, _ = engine.dag_run(req, [rgn])
+                od, orows = orc.dag_run(req, g.keys, g.key_offs, g.vals,
+                                        g.val_offs, g.n_kv)
+                assert (gr, gd) == (orows, od)
+            # project with an RPN predicate: both referenced columns in the
+            # output, decoded
+            sel = (tikv_amd.Expr().col(2).col(6).func(F.SIG_PLUS_INT)
+                   .const_int(0).func(F.SIG_GT_INT))
+            req = (tikv_amd.DagSelect(cols).where(sel)
+                   .output([2, 6, 12]).build())
+            gd, gr, _ = engine.dag_run(req, [rgn])
+            od, orows = orc.dag_run(req, g.keys, g.key_offs, g.vals,
+                                    g.val_offs, g.n_kv)
+            assert (gr, gd) == (orows, od)
+        finally:
+            rgn.close()
+    finally:
+        g.close()

@@ -385,6 +385,140 @@ static void wire_celldir(ScanPlan *sp, const DevRegion &dev) {
   if (ok) { sp->celldir = dev.d_celldir; sp->celldir_n = dev.n_kv; }
 }
 
+
+/* general selection translation: flatten the (ANDed) condition programs
+   into a DevRpnNode program over <=2 distinct int columns. Falls back from
+   the fixed cmp fast path; mirrors the node set orc_exec eval_rpn accepts
+   for ints (expr_eval.rs:205,264). */
+static copr_status translate_rpn(const CoprExecutor &ex, HostPlan *pl,
+                                 ScanPlan *sp) {
+  struct Slot { int64_t col_id; int off; };
+  Slot slots[2];
+  int n_slots = 0;
+  int n = 0;
+  int depth = 0;
+  DevRpnNode prog[COPR_MAX_RPN];
+  auto push = [&](const DevRpnNode &nd, int darg) -> bool {
+    if (n >= COPR_MAX_RPN) return false;
+    depth += darg;
+    if (depth > 4 || depth < 1) return false;
+    prog[n++] = nd;
+    return true;
+  };
+  for (uint32_t c = 0; c < ex.n_conditions; c++) {
+    const CoprExpr &cond = ex.conditions[c];
+    for (uint32_t i = 0; i < cond.n_nodes; i++) {
+      const CoprExprNode &en = cond.nodes[i];
+      DevRpnNode nd{};
+      switch (en.kind) {
+        case COPR_EXPR_COLUMN_REF: {
+          size_t off = (size_t)en.i64_val;
+          if (off >= pl->cols.size()) return COPR_ERR_INVALID_REQUEST;
+          const CoprColumnInfo &ci = pl->cols[off];
+          if (!et_int(ci.ft.tp) || ci.pk_handle) return COPR_ERR_UNSUPPORTED;
+          int s = -1;
+          for (int t = 0; t < n_slots; t++)
+            if (slots[t].col_id == ci.column_id) s = t;
+          if (s < 0) {
+            if (n_slots == 2) return COPR_ERR_UNSUPPORTED;
+            slots[n_slots] = {ci.column_id, (int)off};
+            s = n_slots++;
+          }
+          nd.kind = 0;
+          nd.slot = s;
+          nd.uns = (ci.ft.flag & COPR_FLAG_UNSIGNED) ? 1 : 0;
+          if (!push(nd, +1)) return COPR_ERR_UNSUPPORTED;
+          break;
+        }
+        case COPR_EXPR_CONST_INT:
+        case COPR_EXPR_CONST_UINT:
+          nd.kind = 1;
+          nd.cval = en.i64_val;
+          nd.uns = en.kind == COPR_EXPR_CONST_UINT ? 1 : 0;
+          if (!push(nd, +1)) return COPR_ERR_UNSUPPORTED;
+          break;
+        case COPR_EXPR_CONST_NULL:
+          nd.kind = 2;
+          if (!push(nd, +1)) return COPR_ERR_UNSUPPORTED;
+          break;
+        case COPR_EXPR_SCALAR_FUNC: {
+          int na;
+          switch (en.sig) {
+            case COPR_SIG_LT_INT: case COPR_SIG_LE_INT: case COPR_SIG_GT_INT:
+            case COPR_SIG_GE_INT: case COPR_SIG_EQ_INT: case COPR_SIG_NE_INT:
+            case COPR_SIG_LOGICAL_AND: case COPR_SIG_LOGICAL_OR:
+            case COPR_SIG_PLUS_INT: case COPR_SIG_MINUS_INT:
+            case COPR_SIG_MULTIPLY_INT:
+              na = 2;
+              break;
+            case COPR_SIG_UNARY_NOT: case COPR_SIG_INT_IS_NULL:
+            case COPR_SIG_INT_IS_TRUE: case COPR_SIG_INT_IS_FALSE:
+              na = 1;
+              break;
+            default:
+              return COPR_ERR_UNSUPPORTED;
+          }
+          if (en.n_args != na) return COPR_ERR_INVALID_REQUEST;
+          nd.kind = 3;
+          nd.sig = en.sig;
+          nd.uns = (en.ft.flag & COPR_FLAG_UNSIGNED) ? 1 : 0;
+          if (!push(nd, 1 - na)) return COPR_ERR_UNSUPPORTED;
+          break;
+        }
+        default:
+          return COPR_ERR_UNSUPPORTED;
+      }
+    }
+    if (c > 0) {
+      /* conditions AND together (selection_executor.rs:86) */
+      DevRpnNode nd{};
+      nd.kind = 3;
+      nd.sig = COPR_SIG_LOGICAL_AND;
+      if (!push(nd, -1)) return COPR_ERR_UNSUPPORTED;
+    }
+  }
+  if (depth != 1 || n == 0) return COPR_ERR_INVALID_REQUEST;
+  /* capture channels + missing fill (scan default fill precedes the
+     predicate, table_scan_executor.rs:456-483) */
+  auto fill = [&](const CoprColumnInfo &ci, int32_t *mnull, int64_t *mval)
+      -> copr_status {
+    if (ci.default_val && ci.default_val_len) {
+      int64_t dv;
+      int r = host_decode_int_datum(ci.default_val, ci.default_val_len, &dv);
+      if (r < 0) return COPR_ERR_INVALID_REQUEST;
+      *mnull = r == 1 ? 1 : 0;
+      *mval = dv;
+    } else {
+      *mnull = 1;
+    }
+    return COPR_OK;
+  };
+  if (n_slots >= 1) {
+    const CoprColumnInfo &ci = pl->cols[slots[0].off];
+    sp->has_filter = 1;
+    sp->filter_col_id = ci.column_id;
+    sp->filter_col_unsigned = (ci.ft.flag & COPR_FLAG_UNSIGNED) ? 1 : 0;
+    copr_status st = fill(ci, &sp->filter_missing_null,
+                          &sp->filter_missing_val);
+    if (st != COPR_OK) return st;
+    pl->filter_col_offset = slots[0].off;
+  }
+  if (n_slots == 2) {
+    const CoprColumnInfo &ci = pl->cols[slots[1].off];
+    sp->filter2_on = 1;
+    sp->filter2_col_id = ci.column_id;
+    sp->filter2_col_unsigned = (ci.ft.flag & COPR_FLAG_UNSIGNED) ? 1 : 0;
+    copr_status st = fill(ci, &sp->filter2_missing_null,
+                          &sp->filter2_missing_val);
+    if (st != COPR_OK) return st;
+    pl->filter2_col_offset = slots[1].off;
+  }
+  sp->rpn_on = 1;
+  sp->rpn_n = n;
+  for (int i = 0; i < n; i++) sp->rpn[i] = prog[i];
+  return COPR_OK;
+}
+
 static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
   if (req->n_executors == 0) return SET_ERR(COPR_ERR_INVALID_REQUEST, "empty executors");
   const CoprExecutor &scan = req->executors[0];
@@ -410,22 +544,39 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
     const CoprExecutor &ex = req->executors[e];
     switch (ex.kind) {
       case COPR_EXEC_SELECTION: {
-        /* conditions are ANDed (selection_executor.rs:86); up to two int
-           cmp(col, const) conjuncts run natively */
-        if (sp.has_filter || ex.n_conditions < 1 || ex.n_conditions > 2)
-          return SET_ERR(COPR_ERR_UNSUPPORTED,
-                         "up to two selection conditions supported");
-        copr_status st = match_filter(ex.conditions[0], *pl, &sp);
-        if (st != COPR_OK)
-          return SET_ERR(st, "unsupported selection condition shape");
+        /* conditions are ANDed (selection_executor.rs:86): up to two
+           cmp(col, const) conjuncts take the fixed fast path; other int
+           RPN shapes over <=2 distinct columns run through the device
+           RPN evaluator */
+        if (sp.has_filter || sp.rpn_on || ex.n_conditions < 1)
+          return SET_ERR(COPR_ERR_UNSUPPORTED, "one selection node supported");
+        ScanPlan saved = sp;
+        int saved_off = pl->filter_col_offset;
+        copr_status st = ex.n_conditions <= 2
+                             ? match_filter(ex.conditions[0], *pl, &sp)
+                             : COPR_ERR_UNSUPPORTED;
+        if (st != COPR_OK) {
+          sp = saved;
+          pl->filter_col_offset = saved_off;
+          st = translate_rpn(ex, pl, &sp);
+          if (st != COPR_OK)
+            return SET_ERR(st, "unsupported selection condition shape");
+          break;
+        }
         if (ex.n_conditions == 2) {
           ScanPlan s2{};
           HostPlan p2;
           p2.cols = pl->cols;
           p2.sp.index_mode = sp.index_mode;
           st = match_filter(ex.conditions[1], p2, &s2);
-          if (st != COPR_OK)
-            return SET_ERR(st, "unsupported selection condition shape");
+          if (st != COPR_OK) {
+            sp = saved;
+            pl->filter_col_offset = saved_off;
+            st = translate_rpn(ex, pl, &sp);
+            if (st != COPR_OK)
+              return SET_ERR(st, "unsupported selection condition shape");
+            break;
+          }
           sp.filter2_on = 1;
           sp.filter2_col_id = s2.filter_col_id;
           sp.filter2_cmp = s2.filter_cmp;

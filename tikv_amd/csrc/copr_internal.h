@@ -4,6 +4,9 @@
 
 #include <stdint.h>
 #include <stddef.h>
+#include <vector>
+
+#include "../../include/copr_types.h"   /* scalar sig / tp enums */
 
 namespace copr {
 
@@ -56,6 +59,18 @@ enum {
   DAGG_FIRST,
 };
 
+/* flattened RPN node for the device predicate evaluator (mirrors the
+   oracle's eval_rpn over RpnExpression — expr_eval.rs:205,264; int-typed
+   columns only; column refs resolve to capture slots 0/1 at plan time) */
+#define COPR_MAX_RPN 16
+struct DevRpnNode {
+  int32_t kind;   /* 0 col slot, 1 const int, 2 const NULL, 3 scalar func */
+  int32_t sig;    /* kind 3: CoprScalarSig */
+  int32_t slot;   /* kind 0: capture slot (0 = filter chan, 1 = filter2) */
+  int32_t uns;    /* value unsignedness (col ft / const kind / func out ft) */
+  int64_t cval;   /* kind 1 */
+};
+
 struct DevAggSpec {
   int32_t kind;
   int64_t col_id;        /* source column id (ignored for COUNT_ROWS) */
@@ -92,6 +107,13 @@ struct ScanPlan {
   int32_t filter2_const_null;
   int32_t filter2_missing_null;
   int64_t filter2_missing_val;
+  /* general selection predicate: when rpn_on, the keep decision comes
+     from evaluating this program over the two captured columns instead of
+     the fixed cmp shapes (the capture channels and missing-fill fields
+     above are reused; their cmp fields are ignored) */
+  int32_t rpn_on;
+  int32_t rpn_n;
+  DevRpnNode rpn[COPR_MAX_RPN];
   /* 1 = keep every row but still export filt_vals/filt_state: used for a
      column an upstream expression decoded in place (e.g. the TopN order
      column) so the response encodes it in DECODED form

@@ -643,6 +643,125 @@ __device__ static inline bool d_filter2_keep(const ScanPlan &plan,
                              plan.filter2_const_unsigned));
 }
 
+/* evaluate the plan's RPN predicate over the two captured columns.
+ * Mirrors orc_exec eval_rpn node-for-node (NULL-aware logical ops
+ * impl_op.rs; signed overflow on int arithmetic errors the request).
+ * Returns keep; *err set on overflow / malformed program. */
+__device__ static inline bool d_rpn_keep(const ScanPlan &plan,
+                                         bool f1_found, bool f1_null,
+                                         int64_t f1_v, bool f2_found,
+                                         bool f2_null, int64_t f2_v,
+                                         bool *err) {
+  int64_t sv[4];
+  uint8_t sn[4], su[4];
+  int sp = 0;
+  /* missing-column fill, same as the fixed filter path */
+  if (!f1_found) {
+    if (plan.filter_missing_null) f1_null = true;
+    else { f1_v = plan.filter_missing_val; f1_null = false; }
+  }
+  if (!f2_found) {
+    if (plan.filter2_missing_null) f2_null = true;
+    else { f2_v = plan.filter2_missing_val; f2_null = false; }
+  }
+  for (int i = 0; i < plan.rpn_n; i++) {
+    const DevRpnNode &nd = plan.rpn[i];
+    if (nd.kind == 0) {
+      if (sp >= 4) { *err = true; return false; }
+      sv[sp] = nd.slot ? f2_v : f1_v;
+      sn[sp] = nd.slot ? (f2_null ? 1 : 0) : (f1_null ? 1 : 0);
+      su[sp] = (uint8_t)nd.uns;
+      sp++;
+    } else if (nd.kind == 1) {
+      if (sp >= 4) { *err = true; return false; }
+      sv[sp] = nd.cval; sn[sp] = 0; su[sp] = (uint8_t)nd.uns;
+      sp++;
+    } else if (nd.kind == 2) {
+      if (sp >= 4) { *err = true; return false; }
+      sv[sp] = 0; sn[sp] = 1; su[sp] = 0;
+      sp++;
+    } else {
+      int na = (nd.sig == COPR_SIG_UNARY_NOT ||
+                nd.sig == COPR_SIG_INT_IS_NULL ||
+                nd.sig == COPR_SIG_INT_IS_TRUE ||
+                nd.sig == COPR_SIG_INT_IS_FALSE) ? 1 : 2;
+      if (sp < na) { *err = true; return false; }
+      int b = sp - na;
+      bool n0 = sn[b] != 0, n1 = na == 2 && sn[b + 1] != 0;
+      int64_t a0 = sv[b], a1 = na == 2 ? sv[b + 1] : 0;
+      int64_t rv = 0;
+      uint8_t rn = 0;
+      switch (nd.sig) {
+        case COPR_SIG_LT_INT: case COPR_SIG_LE_INT: case COPR_SIG_GT_INT:
+        case COPR_SIG_GE_INT: case COPR_SIG_EQ_INT: case COPR_SIG_NE_INT: {
+          if (n0 || n1) { rn = 1; break; }
+          int cmp = nd.sig == COPR_SIG_LT_INT ? CMP_LT :
+                    nd.sig == COPR_SIG_LE_INT ? CMP_LE :
+                    nd.sig == COPR_SIG_GT_INT ? CMP_GT :
+                    nd.sig == COPR_SIG_GE_INT ? CMP_GE :
+                    nd.sig == COPR_SIG_EQ_INT ? CMP_EQ : CMP_NE;
+          rv = d_cmp_res(cmp, d_cmp_int(a0, a1, su[b], su[b + 1])) ? 1 : 0;
+          break;
+        }
+        case COPR_SIG_LOGICAL_AND: {
+          bool t0f = !n0 && a0 == 0, t1f = !n1 && a1 == 0;
+          if (t0f || t1f) rv = 0;
+          else if (n0 || n1) rn = 1;
+          else rv = 1;
+          break;
+        }
+        case COPR_SIG_LOGICAL_OR: {
+          bool t0 = !n0 && a0 != 0, t1 = !n1 && a1 != 0;
+          if (t0 || t1) rv = 1;
+          else if (n0 || n1) rn = 1;
+          else rv = 0;
+          break;
+        }
+        case COPR_SIG_UNARY_NOT:
+          if (n0) rn = 1;
+          else rv = a0 == 0 ? 1 : 0;
+          break;
+        case COPR_SIG_PLUS_INT: case COPR_SIG_MINUS_INT:
+        case COPR_SIG_MULTIPLY_INT: {
+          if (n0 || n1) { rn = 1; break; }
+          bool ovf;
+          long long res;
+          if (nd.sig == COPR_SIG_PLUS_INT)
+            ovf = __builtin_add_overflow((long long)a0, (long long)a1, &res);
+          else if (nd.sig == COPR_SIG_MINUS_INT)
+            ovf = __builtin_sub_overflow((long long)a0, (long long)a1, &res);
+          else
+            ovf = __builtin_mul_overflow((long long)a0, (long long)a1, &res);
+          if (ovf) { *err = true; return false; }
+          rv = res;
+          break;
+        }
+        case COPR_SIG_INT_IS_NULL: rv = n0 ? 1 : 0; break;
+        case COPR_SIG_INT_IS_TRUE: rv = (!n0 && a0 != 0) ? 1 : 0; break;
+        case COPR_SIG_INT_IS_FALSE: rv = (!n0 && a0 == 0) ? 1 : 0; break;
+        default: *err = true; return false;
+      }
+      sp = b;
+      sv[sp] = rv; sn[sp] = rn; su[sp] = (uint8_t)nd.uns;
+      sp++;
+    }
+  }
+  if (sp != 1) { *err = true; return false; }
+  return sn[0] == 0 && sv[0] != 0;   /* as_mysql_bool */
+}
+
+/* combined keep decision over the two capture channels */
+__device__ static inline bool d_keep2(const ScanPlan &plan, bool f1_found,
+                                      bool f1_null, int64_t f1_v,
+                                      bool f2_found, bool f2_null,
+                                      int64_t f2_v, bool *err) {
+  if (plan.rpn_on)
+    return d_rpn_keep(plan, f1_found, f1_null, f1_v, f2_found, f2_null, f2_v,
+                      err);
+  return d_filter_keep(plan, f1_found, f1_null, f1_v) &&
+         d_filter2_keep(plan, f2_found, f2_null, f2_v);
+}
+
 /* 128-bit signed accumulate via two u64 atomics (carry trick) */
 __device__ static inline void atomic_add_i128(unsigned long long *lo,
                                               unsigned long long *hi, int64_t x) {
@@ -928,8 +1047,11 @@ k_scan_agg(ScanPlan plan,
 
       if (!parse_ok) {
         any_parse_err = true;
-      } else if (d_filter_keep(plan, filt_found, filt_null, filt_v) &&
-                 d_filter2_keep(plan, f2_found, f2_null, f2_v)) {
+      } else if (bool ke = false;
+                 d_keep2(plan, filt_found, filt_null, filt_v, f2_found,
+                         f2_null, f2_v, &ke)
+                     ? true
+                     : (ke ? (any_parse_err = true, false) : false)) {
         SimpleAggAcc *acc_base = nullptr;
         if (IS_HASH) {
           if (!grp_found || grp_null) {
@@ -1517,8 +1639,11 @@ k_scan_agg_pipe(ScanPlan plan,
 
       if (!parse_ok) {
         any_parse_err = true;
-      } else if (d_filter_keep(plan, filt_found, filt_null, filt_v) &&
-                 d_filter2_keep(plan, f2_found, f2_null, f2_v)) {
+      } else if (bool ke = false;
+                 d_keep2(plan, filt_found, filt_null, filt_v, f2_found,
+                         f2_null, f2_v, &ke)
+                     ? true
+                     : (ke ? (any_parse_err = true, false) : false)) {
         SimpleAggAcc *acc_base = nullptr;
         if (IS_HASH) {
           if (!grp_found || grp_null) {
@@ -2425,8 +2550,10 @@ k_scan_project(ScanPlan plan,
       if (!parse_ok) {
         any_parse_err = true;
       } else {
-        bool keep = d_filter_keep(plan, filt_found, filt_null, filt_v) &&
-                    d_filter2_keep(plan, d2_found, d2_null, d2_v);
+        bool kerr = false;
+        bool keep = d_keep2(plan, filt_found, filt_null, filt_v, d2_found,
+                            d2_null, d2_v, &kerr);
+        if (kerr) { any_parse_err = true; keep = false; }
         po.keep[my_row] = keep ? 1 : 0;
         if (plan.has_filter && po.filt_vals) {
           po.filt_vals[my_row] = filt_v;
@@ -2824,7 +2951,7 @@ static int launch_agg_pipe(const ScanPlan &plan, const DevRegion &rgn,
                            hipStream_t s, uint32_t grid) {
   if (!IS_HASH && !plan.index_mode && plan.n_aggs == 1 &&
       plan.aggs[0].kind == DAGG_COUNT_ROWS && plan.has_filter &&
-      !plan.filter2_on) {
+      !plan.filter2_on && !plan.rpn_on) {
     if (getenv("COPR_DIRECT")) {
       uint64_t n_blk = (rgn.n_kv + THREADS - 1) / THREADS;
       uint32_t dgrid = (uint32_t)(n_blk < 8192 ? n_blk : 8192);
@@ -3158,8 +3285,11 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
       uint8_t s = 0;
       if (!parse_ok) {
         any_err = true;
-      } else if (d_filter_keep(plan, filt_found, filt_null, filt_v) &&
-                 d_filter2_keep(plan, f2_found, f2_null, f2_v)) {
+      } else if (bool ke = false;
+                 d_keep2(plan, filt_found, filt_null, filt_v, f2_found,
+                         f2_null, f2_v, &ke)
+                     ? true
+                     : (ke ? (any_err = true, false) : false)) {
         s = (grp_found && !grp_null) ? 2 : 1;
       }
       eo.st[my_row] = s;
