@@ -826,3 +826,72 @@ def test_rpn_gpu_parity(engine):
             rgn.close()
     finally:
         g.close()
+
+
+def test_chunk_decimal_struct_golden():
+    """40-byte decimal struct dump layout (decimal.rs:2135-2142 dump of the
+    read_decimal result), hand-computed for 123.45 and -0.07."""
+    orc = _orc()
+    lib = orc.load_lib()
+    # datum-encode 123.45 via the oracle codec test hook: scaled 12345,frac 2
+    import ctypes as CT
+    # build rows with decimal cells through the encode hook
+    buf = CT.create_string_buffer(48)
+    st = lib.orc_test_dec_from_i64_encode(123, buf)   # 123 as decimal
+    assert st >= 0
+    # instead pin via a full pipeline: single row {1: decimal}, project
+    # hand-encode the datum: [prec=5][frac=2] comparable words
+    # 123.45: int word 123 (1 leading word of 3 digits -> 2 bytes), frac 45
+    # (2 trailing digits -> 1 byte); positive -> first byte |= 0x80
+    enc = bytes([5, 2, 0x80, 123, 45])
+    row = b"\x08" + var_i64(1) + b"\x06" + enc
+    keys = row_key(0)
+    kb = (C.c_uint8 * len(keys)).from_buffer_copy(keys)
+    vb = (C.c_uint8 * len(row)).from_buffer_copy(row)
+    ko = (C.c_uint64 * 2)(0, len(keys))
+    vo = (C.c_uint64 * 2)(0, len(row))
+    cols = [tikv_amd.Col(1, tp=F.TP_NEWDECIMAL, decimal=2)]
+    req = tikv_amd.DagSelect(cols).chunked().build()
+    data, nrows = orc.dag_run(req, C.cast(kb, C.POINTER(C.c_uint8)), ko,
+                              C.cast(vb, C.POINTER(C.c_uint8)), vo, 1)
+    assert nrows == 1
+    exp = bytearray()
+    exp += (1).to_bytes(4, "little") + (0).to_bytes(4, "little")
+    exp += bytes([3, 2, 2, 0])            # int_cnt, frac_cnt, result_frac, neg
+    words = [123, 450000000] + [0] * 7    # 45 scaled to a full 9-digit word
+    for w in words:
+        exp += w.to_bytes(4, "little")
+    assert data == bytes(exp)
+
+
+def test_rpn_overflow_errors():
+    """BIGINT overflow in a predicate errors the request (both sides)."""
+    orc = _orc()
+    rows = [{1: 2**62}]
+    k, ko, v, vo, n, keep = region_of(rows)
+    cols = [tikv_amd.Col(1)]
+    sel = (tikv_amd.Expr().col(0).const_int(2**62).func(F.SIG_PLUS_INT)
+           .const_int(0).func(F.SIG_GT_INT))
+    req = (tikv_amd.DagSelect(cols).where(sel)
+           .simple_agg([tikv_amd.count_star()]).build())
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError):
+        orc.dag_run(req, k, ko, v, vo, n)
+
+
+@pytest.mark.gpu
+def test_rpn_overflow_errors_gpu(engine):
+    rows = [{1: 2**62}]
+    k, ko, v, vo, n, keep = region_of(rows)
+    rgn = engine.region_raw(k, ko, v, vo, n)
+    try:
+        cols = [tikv_amd.Col(1)]
+        sel = (tikv_amd.Expr().col(0).const_int(2**62).func(F.SIG_PLUS_INT)
+               .const_int(0).func(F.SIG_GT_INT))
+        req = (tikv_amd.DagSelect(cols).where(sel)
+               .simple_agg([tikv_amd.count_star()]).build())
+        import pytest as _pytest
+        with _pytest.raises(RuntimeError):
+            engine.dag_run(req, [rgn])
+    finally:
+        rgn.close()
