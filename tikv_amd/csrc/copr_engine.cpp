@@ -652,6 +652,9 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
 
   /* aggregation plan */
   pl->has_agg = true;
+  if (pl->stream_agg && sp.index_mode)
+    return SET_ERR(COPR_ERR_UNSUPPORTED,
+                   "stream agg over index scan not yet native");
   sp.mode = pl->stream_agg ? 3 : (pl->hash_agg ? 2 : 1);
   if (agg->n_aggs == 0 || agg->n_aggs > COPR_MAX_AGGS)
     return SET_ERR(COPR_ERR_UNSUPPORTED, "agg count out of range");
@@ -778,6 +781,9 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
                      "FIRST over int-key hash agg not yet native "
                      "(stream/bytes grouping supports it)");
   } else if (any_first) {
+    if (sp.index_mode)
+      return SET_ERR(COPR_ERR_UNSUPPORTED,
+                     "FIRST over index scan not yet native");
     /* simple agg with FIRST reroutes through a single-run stream pass
        (group col that never matches -> one NULL-key run) */
     sp.mode = 3;
