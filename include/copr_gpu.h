@@ -110,6 +110,34 @@ typedef struct CoprGenOut {   /* host buffers owned by the generator */
 copr_status copr_gen_region(const CoprGenSpec *, CoprGenOut *out);
 void        copr_gen_free(CoprGenOut *);
 
+/* ---- SST data-block ingestion ----
+ * Feed uncompressed RocksDB BlockBasedTable DATA blocks instead of
+ * per-KV arrays: each block is [entries: varint32 shared | varint32
+ * non_shared | varint32 value_len | key_delta | value]* followed by
+ * [restart offsets u32le x n][num_restarts u32le]; keys are InternalKeys
+ * (user_key + 8B (seq<<8|type) trailer, stripped on decode). The engine
+ * parses blocks ON DEVICE (one wavefront lane per restart interval) and
+ * materializes a resident region. Reference producer: engine_rocks
+ * iterators (engine_iterator.rs:12) behind SnapshotStore::scanner
+ * (txn/store.rs:431); format per RocksDB block_builder.cc/block.cc
+ * (public format; compression is the feeder's concern -- blocks arrive
+ * uncompressed here). */
+copr_status copr_region_create_blocks(copr_engine *,
+                                      const uint8_t *blocks,
+                                      const uint64_t *block_offs,
+                                      uint32_t n_blocks,
+                                      copr_region **out);
+
+/* fixture writer: pack a KV stream into data blocks (restart-interval
+ * prefix compression; ~target_block_bytes per block). Buffers owned by
+ * the generator allocator; free blocks with free() and offs with free() */
+copr_status copr_gen_blocks(const uint8_t *keys, const uint64_t *key_offs,
+                            const uint8_t *vals, const uint64_t *val_offs,
+                            uint64_t n_kv, uint32_t restart_interval,
+                            uint32_t target_block_bytes,
+                            uint8_t **blocks_out, uint64_t **block_offs_out,
+                            uint32_t *n_blocks_out);
+
 /* test/debug: copy a resident region back to host buffers (owned by the
  * engine allocator; free with copr_gen_free) */
 copr_status copr_region_dump(copr_engine *, copr_region *, CoprGenOut *out);

@@ -1413,6 +1413,77 @@ static bool run_pipeline(const CoprDagRequest *req,
 /* ================= C API ================= */
 extern "C" {
 
+/* ---- SST data-block decode (oracle restatement) ----
+ * RocksDB BlockBasedTable data block (block_builder.cc/block.cc public
+ * format; the crate TiKV pins is rust-rocksdb in Cargo.lock): entries of
+ * [varint32 shared][varint32 non_shared][varint32 value_len][key delta]
+ * [value], restart-point array u32le + count at the tail; keys are
+ * InternalKeys whose trailing 8 bytes ((seq<<8)|type) are stripped. */
+static bool blk_varint32(const uint8_t *p, size_t rem, uint32_t *v,
+                         size_t *n) {
+  uint32_t x = 0;
+  size_t i = 0;
+  int sh = 0;
+  while (i < rem && i < 5) {
+    uint8_t b = p[i++];
+    x |= (uint32_t)(b & 0x7F) << sh;
+    sh += 7;
+    if (!(b & 0x80)) { *v = x; *n = i; return true; }
+  }
+  return false;
+}
+
+extern "C" int orc_block_parse(const uint8_t *blocks,
+                               const uint64_t *block_offs, uint32_t n_blocks,
+                               OrcRegion *out) {
+  std::vector<uint8_t> keys, vals;
+  std::vector<uint64_t> koffs{0}, voffs{0};
+  for (uint32_t b = 0; b < n_blocks; b++) {
+    const uint8_t *blk = blocks + block_offs[b];
+    size_t blen = (size_t)(block_offs[b + 1] - block_offs[b]);
+    if (blen < 8) return -1;
+    uint32_t nr = (uint32_t)blk[blen - 4] | ((uint32_t)blk[blen - 3] << 8) |
+                  ((uint32_t)blk[blen - 2] << 16) |
+                  ((uint32_t)blk[blen - 1] << 24);
+    if (blen < 4 + (size_t)nr * 4) return -1;
+    size_t data_end = blen - 4 - (size_t)nr * 4;
+    std::string key;
+    size_t pos = 0;
+    while (pos < data_end) {
+      uint32_t shared, non_shared, vlen;
+      size_t n;
+      if (!blk_varint32(blk + pos, data_end - pos, &shared, &n)) return -1;
+      pos += n;
+      if (!blk_varint32(blk + pos, data_end - pos, &non_shared, &n)) return -1;
+      pos += n;
+      if (!blk_varint32(blk + pos, data_end - pos, &vlen, &n)) return -1;
+      pos += n;
+      if (pos + non_shared + vlen > data_end || shared > key.size()) return -1;
+      key.resize(shared);
+      key.append((const char *)(blk + pos), non_shared);
+      pos += non_shared;
+      if (key.size() < 8) return -1;
+      keys.insert(keys.end(), key.begin(), key.end() - 8);
+      koffs.push_back(keys.size());
+      vals.insert(vals.end(), blk + pos, blk + pos + vlen);
+      voffs.push_back(vals.size());
+      pos += vlen;
+    }
+    if (pos != data_end) return -1;
+  }
+  uint64_t n_kv = koffs.size() - 1;
+  out->keys = (uint8_t *)malloc(keys.size() ? keys.size() : 1);
+  memcpy(out->keys, keys.data(), keys.size());
+  out->vals = (uint8_t *)malloc(vals.size() ? vals.size() : 1);
+  memcpy(out->vals, vals.data(), vals.size());
+  out->key_offs = (uint64_t *)malloc(koffs.size() * 8);
+  memcpy(out->key_offs, koffs.data(), koffs.size() * 8);
+  out->val_offs = (uint64_t *)malloc(voffs.size() * 8);
+  memcpy(out->val_offs, voffs.data(), voffs.size() * 8);
+  out->n_kv = n_kv;
+  return 0;
+}
+
 int orc_dag_run(const CoprDagRequest *req,
                 const uint8_t *keys, const uint64_t *key_offs,
                 const uint8_t *vals, const uint64_t *val_offs,

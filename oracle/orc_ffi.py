@@ -53,6 +53,10 @@ def load_lib():
                                     C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
                                     C.c_uint64, C.c_uint64, C.POINTER(OrcRegion)]
     lib.orc_region_free.argtypes = [C.POINTER(OrcRegion)]
+    lib.orc_block_parse.restype = C.c_int
+    lib.orc_block_parse.argtypes = [C.POINTER(C.c_uint8),
+                                    C.POINTER(C.c_uint64), C.c_uint32,
+                                    C.POINTER(OrcRegion)]
     lib.orc_crc64_xz.restype = C.c_uint64
     lib.orc_crc64_xz.argtypes = [C.c_char_p, C.c_uint64]
     lib.orc_test_memcmp_encode.restype = C.c_uint64
@@ -108,6 +112,22 @@ def checksum(keys, key_offs, vals, val_offs, n_kv):
                           C.byref(cs), C.byref(kvs), C.byref(byts))
     assert st == 0
     return cs.value, kvs.value, byts.value
+
+
+def block_parse(blocks, block_offs, n_blocks):
+    """Oracle decode of RocksDB data blocks -> (keys, ko, vals, vo, n)."""
+    lib = load_lib()
+    out = OrcRegion()
+    st = lib.orc_block_parse(blocks, block_offs, n_blocks, C.byref(out))
+    if st != 0:
+        raise RuntimeError("orc_block_parse: %d" % st)
+    n = out.n_kv
+    koffs = [out.key_offs[i] for i in range(n + 1)]
+    voffs = [out.val_offs[i] for i in range(n + 1)]
+    keysb = C.string_at(out.keys, koffs[-1]) if koffs[-1] else b""
+    valsb = C.string_at(out.vals, voffs[-1]) if voffs[-1] else b""
+    lib.orc_region_free(C.byref(out))
+    return keysb, koffs, valsb, voffs, n
 
 
 def mvcc_filter(keys, key_offs, vals, val_offs, n_kv, read_ts):

@@ -10,5 +10,5 @@ from . import _ffi  # noqa: F401
 from .runner import (  # noqa: F401
     Col, Expr, DagSelect, Engine, Region, GenRegion,
     count_star, count_col, sum_col, avg_col, max_col, min_col, first_col, bit_op,
-    cmp_col_const, field_type,
+    cmp_col_const, field_type, gen_blocks,
 )

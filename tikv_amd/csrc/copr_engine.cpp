@@ -128,6 +128,53 @@ extern "C" copr_status copr_region_create(copr_engine *eng,
   return COPR_OK;
 }
 
+
+/* ---- SST data-block ingestion (copr_gpu.h) ---- */
+static copr_status region_from_dev(copr_engine *eng, DevRegion dr,
+                                   copr_region **out) {
+  copr_region *r = new copr_region();
+  r->eng = eng;
+  r->dev = dr;
+  r->h_key_offs.resize(dr.n_kv + 1);
+  r->h_val_offs.resize(dr.n_kv + 1);
+  hipError_t ce = hipMemcpy(r->h_key_offs.data(), dr.d_key_offs,
+                            (dr.n_kv + 1) * 8, hipMemcpyDeviceToHost);
+  if (ce == hipSuccess)
+    ce = hipMemcpy(r->h_val_offs.data(), dr.d_val_offs, (dr.n_kv + 1) * 8,
+                   hipMemcpyDeviceToHost);
+  if (ce != hipSuccess) {
+    copr_region_destroy(r);
+    return SET_ERR(COPR_ERR_INTERNAL, "region offs readback");
+  }
+  uint32_t mr = 0, mk = 0;
+  for (uint64_t i = 0; i < dr.n_kv; i++) {
+    uint64_t l = r->h_val_offs[i + 1] - r->h_val_offs[i];
+    if (l > mr) mr = (uint32_t)l;
+    uint64_t kl = r->h_key_offs[i + 1] - r->h_key_offs[i];
+    if (kl > mk) mk = (uint32_t)kl;
+  }
+  r->dev.max_row_bytes = mr;
+  r->dev.max_key_bytes = mk;
+  if (!getenv("COPR_NO_DIR")) dev_celldir_build(r->dev, eng->stream);
+  *out = r;
+  return COPR_OK;
+}
+
+extern "C" copr_status copr_region_create_blocks(copr_engine *eng,
+                                                 const uint8_t *blocks,
+                                                 const uint64_t *block_offs,
+                                                 uint32_t n_blocks,
+                                                 copr_region **out) {
+  if (!eng) return SET_ERR(COPR_ERR_INVALID_REQUEST, "null engine");
+  HIP_TRY(hipSetDevice(eng->device), "hipSetDevice");
+  DevRegion dr{};
+  int rc = dev_blocks_build(blocks, block_offs, n_blocks, &dr, eng->stream);
+  if (rc == -3) return SET_ERR(COPR_ERR_STORAGE, "malformed data block");
+  if (rc == -2) return SET_ERR(COPR_ERR_OOM, "block ingest alloc");
+  if (rc != 0) return SET_ERR(COPR_ERR_INTERNAL, "block ingest failed");
+  return region_from_dev(eng, dr, out);
+}
+
 extern "C" copr_status copr_region_create_mvcc(copr_engine *eng,
                                                const uint8_t *keys,
                                                const uint64_t *key_offs,

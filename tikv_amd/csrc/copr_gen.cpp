@@ -490,4 +490,72 @@ void copr_gen_free(CoprGenOut *out) {
   memset(out, 0, sizeof(*out));
 }
 
+
+/* ---- SST data-block fixture writer (RocksDB BlockBasedTable data block:
+ * prefix-compressed entries + restart array; block_builder.cc) ---- */
+static void put_varint32(std::vector<uint8_t> *out, uint32_t v) {
+  while (v >= 0x80) { out->push_back((uint8_t)(v | 0x80)); v >>= 7; }
+  out->push_back((uint8_t)v);
+}
+
+extern "C" copr_status copr_gen_blocks(
+    const uint8_t *keys, const uint64_t *key_offs, const uint8_t *vals,
+    const uint64_t *val_offs, uint64_t n_kv, uint32_t restart_interval,
+    uint32_t target_block_bytes, uint8_t **blocks_out,
+    uint64_t **block_offs_out, uint32_t *n_blocks_out) {
+  if (!restart_interval) restart_interval = 16;
+  if (!target_block_bytes) target_block_bytes = 4096;
+  std::vector<uint8_t> all;
+  std::vector<uint64_t> boffs{0};
+  std::vector<uint8_t> blk;
+  std::vector<uint32_t> restarts;
+  std::string prev_ikey;
+  uint32_t counter = 0;
+  auto flush = [&]() {
+    if (blk.empty()) return;
+    for (uint32_t r : restarts) {
+      blk.push_back((uint8_t)r); blk.push_back((uint8_t)(r >> 8));
+      blk.push_back((uint8_t)(r >> 16)); blk.push_back((uint8_t)(r >> 24));
+    }
+    uint32_t nr = (uint32_t)restarts.size();
+    blk.push_back((uint8_t)nr); blk.push_back((uint8_t)(nr >> 8));
+    blk.push_back((uint8_t)(nr >> 16)); blk.push_back((uint8_t)(nr >> 24));
+    all.insert(all.end(), blk.begin(), blk.end());
+    boffs.push_back(all.size());
+    blk.clear(); restarts.clear(); prev_ikey.clear(); counter = 0;
+  };
+  for (uint64_t i = 0; i < n_kv; i++) {
+    /* InternalKey = user key + 8B (seq<<8 | kTypeValue) little-endian */
+    std::string ikey((const char *)(keys + key_offs[i]),
+                     (size_t)(key_offs[i + 1] - key_offs[i]));
+    uint64_t trailer = ((n_kv - i) << 8) | 0x1;
+    for (int b = 0; b < 8; b++) ikey.push_back((char)(trailer >> (8 * b)));
+    uint32_t shared = 0;
+    if (counter % restart_interval == 0) {
+      restarts.push_back((uint32_t)blk.size());
+    } else {
+      size_t m = std::min(prev_ikey.size(), ikey.size());
+      while (shared < m && prev_ikey[shared] == ikey[shared]) shared++;
+    }
+    uint32_t vlen = (uint32_t)(val_offs[i + 1] - val_offs[i]);
+    put_varint32(&blk, shared);
+    put_varint32(&blk, (uint32_t)ikey.size() - shared);
+    put_varint32(&blk, vlen);
+    blk.insert(blk.end(), ikey.begin() + shared, ikey.end());
+    blk.insert(blk.end(), vals + val_offs[i], vals + val_offs[i + 1]);
+    prev_ikey = ikey;
+    counter++;
+    if (blk.size() >= target_block_bytes && i + 1 < n_kv) flush();
+  }
+  flush();
+  uint8_t *ab = (uint8_t *)malloc(all.size() ? all.size() : 1);
+  memcpy(ab, all.data(), all.size());
+  uint64_t *ob = (uint64_t *)malloc(boffs.size() * 8);
+  memcpy(ob, boffs.data(), boffs.size() * 8);
+  *blocks_out = ab;
+  *block_offs_out = ob;
+  *n_blocks_out = (uint32_t)(boffs.size() - 1);
+  return COPR_OK;
+}
+
 }  // extern "C"

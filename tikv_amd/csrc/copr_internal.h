@@ -236,6 +236,8 @@ int dev_bytes_agg(const ScanPlan &plan, const DevRegion &rgn, void *stream,
                   std::vector<SimpleAggAcc> *h_accs,
                   std::vector<uint64_t> *h_kofs, std::vector<uint32_t> *h_klen,
                   std::vector<uint8_t> *h_kst);
+int dev_blocks_build(const uint8_t *h_blocks, const uint64_t *h_block_offs,
+                     uint32_t n_blocks, DevRegion *out, void *stream);
 int dev_subregion_build(const DevRegion &src, const uint32_t *h_rows,
                         uint64_t m, DevRegion *out, void *stream);
 int dev_mvcc_build(const uint8_t *d_keys, const uint64_t *d_ko,

@@ -4054,6 +4054,239 @@ int dev_bytes_agg(const ScanPlan &plan, const DevRegion &rgn, void *stream,
   return (int)n_seg;
 }
 
+
+/* ---------------- SST data-block ingestion ----------------
+ * RocksDB BlockBasedTable data-block decode on device: restart intervals
+ * are independent (each starts with shared=0), so one LANE decodes one
+ * interval sequentially, reconstructing prefix-compressed InternalKeys in
+ * a per-lane LDS scratch line and stripping the 8-byte trailer. Two
+ * passes: sizes -> prefix sums -> materialize. */
+#define BLK_MAX_IKEY 128
+
+struct BlkInterval { uint64_t byte_start, byte_end; };
+
+__device__ static inline bool d_blk_varint32(const uint8_t *p, uint64_t rem,
+                                             uint32_t *v, uint32_t *n) {
+  uint32_t x = 0;
+  uint32_t i = 0;
+  int sh = 0;
+  while (i < rem && i < 5) {
+    uint8_t b = p[i++];
+    x |= (uint32_t)(b & 0x7F) << sh;
+    sh += 7;
+    if (!(b & 0x80)) { *v = x; *n = i; return true; }
+  }
+  return false;
+}
+
+template <bool WRITE>
+__global__ void __launch_bounds__(256)
+k_blk_parse(const uint8_t *__restrict__ blocks,
+            const BlkInterval *__restrict__ ivs, uint64_t n_ivs,
+            uint32_t *__restrict__ n_entries,
+            uint64_t *__restrict__ key_bytes,
+            uint64_t *__restrict__ val_bytes,
+            const uint64_t *__restrict__ ent_base,
+            const uint64_t *__restrict__ kb_base,
+            const uint64_t *__restrict__ vb_base,
+            uint8_t *__restrict__ out_keys, uint64_t *__restrict__ out_ko,
+            uint8_t *__restrict__ out_vals, uint64_t *__restrict__ out_vo,
+            unsigned int *__restrict__ err) {
+  extern __shared__ uint8_t lds[];
+  uint8_t *mykey = lds + (uint64_t)threadIdx.x * BLK_MAX_IKEY;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t iv = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       iv < n_ivs; iv += stride) {
+    uint64_t pos = ivs[iv].byte_start;
+    const uint64_t end = ivs[iv].byte_end;
+    uint32_t klen = 0;
+    uint32_t ents = 0;
+    uint64_t kb = 0, vb = 0;
+    uint64_t ko = WRITE ? kb_base[iv] : 0;
+    uint64_t vo = WRITE ? vb_base[iv] : 0;
+    uint64_t eb = WRITE ? ent_base[iv] : 0;
+    bool bad = false;
+    while (pos < end) {
+      uint32_t shared, non_shared, vlen, n;
+      if (!d_blk_varint32(blocks + pos, end - pos, &shared, &n)) { bad = true; break; }
+      pos += n;
+      if (!d_blk_varint32(blocks + pos, end - pos, &non_shared, &n)) { bad = true; break; }
+      pos += n;
+      if (!d_blk_varint32(blocks + pos, end - pos, &vlen, &n)) { bad = true; break; }
+      pos += n;
+      if (pos + non_shared + vlen > end || shared > klen ||
+          shared + non_shared > BLK_MAX_IKEY) { bad = true; break; }
+      for (uint32_t t = 0; t < non_shared; t++)
+        mykey[shared + t] = blocks[pos + t];
+      klen = shared + non_shared;
+      pos += non_shared;
+      if (klen < 8) { bad = true; break; }
+      if (WRITE) {
+        out_ko[eb + ents] = ko;
+        out_vo[eb + ents] = vo;
+        for (uint32_t t = 0; t < klen - 8; t++) out_keys[ko + t] = mykey[t];
+        ko += klen - 8;
+        for (uint32_t t = 0; t < vlen; t++) out_vals[vo + t] = blocks[pos + t];
+        vo += vlen;
+      }
+      kb += klen - 8;
+      vb += vlen;
+      ents++;
+      pos += vlen;
+    }
+    if (bad || pos != end) {
+      atomicOr(err, 1u);
+      if (!WRITE) { n_entries[iv] = 0; key_bytes[iv] = 0; val_bytes[iv] = 0; }
+      continue;
+    }
+    if (!WRITE) {
+      n_entries[iv] = ents;
+      key_bytes[iv] = kb;
+      val_bytes[iv] = vb;
+    }
+  }
+}
+
+__global__ static void k_blk_widen(const uint32_t *src, uint64_t *dst,
+                                   uint64_t n) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) dst[i] = src[i];
+}
+
+/* parse uncompressed data blocks into a DevRegion. Returns 0, -1 internal,
+ * -2 oom, -3 malformed block. h_blocks/h_offs are the HOST copies (restart
+ * arrays are read on host to build the interval table). */
+int dev_blocks_build(const uint8_t *h_blocks, const uint64_t *h_block_offs,
+                     uint32_t n_blocks, DevRegion *out, void *stream) {
+  hipStream_t s = (hipStream_t)stream;
+  *out = DevRegion{};
+  std::vector<BlkInterval> ivs;
+  for (uint32_t b = 0; b < n_blocks; b++) {
+    const uint8_t *blk = h_blocks + h_block_offs[b];
+    uint64_t blen = h_block_offs[b + 1] - h_block_offs[b];
+    if (blen < 8) return -3;
+    uint32_t nr = (uint32_t)blk[blen - 4] | ((uint32_t)blk[blen - 3] << 8) |
+                  ((uint32_t)blk[blen - 2] << 16) |
+                  ((uint32_t)blk[blen - 1] << 24);
+    if (blen < 4 + (uint64_t)nr * 4 || nr == 0) return -3;
+    uint64_t data_end = blen - 4 - (uint64_t)nr * 4;
+    const uint8_t *ra = blk + data_end;
+    uint64_t prev = UINT64_MAX;
+    for (uint32_t r = 0; r < nr; r++) {
+      uint32_t off = (uint32_t)ra[4 * r] | ((uint32_t)ra[4 * r + 1] << 8) |
+                     ((uint32_t)ra[4 * r + 2] << 16) |
+                     ((uint32_t)ra[4 * r + 3] << 24);
+      if (off >= data_end && !(off == 0 && data_end == 0)) return -3;
+      if (prev != UINT64_MAX) {
+        if (off <= prev) return -3;
+        ivs.push_back({h_block_offs[b] + prev, h_block_offs[b] + off});
+      }
+      prev = off;
+    }
+    ivs.push_back({h_block_offs[b] + prev, h_block_offs[b] + data_end});
+  }
+  uint64_t n_ivs = ivs.size();
+  uint64_t total = h_block_offs[n_blocks];
+  uint8_t *d_blocks = nullptr;
+  BlkInterval *d_ivs = nullptr;
+  uint32_t *d_ne = nullptr;
+  uint64_t *d_kb = nullptr, *d_vb = nullptr;
+  uint64_t *d_eb = nullptr, *d_kbb = nullptr, *d_vbb = nullptr;
+  uint64_t *d_ne64 = nullptr;
+  unsigned int *d_err = nullptr;
+  void *tmp = nullptr;
+  size_t tmpb = 0;
+  hipError_t e = hipSuccess;
+  auto fail = [&](int rc) {
+    hipFree(d_blocks); hipFree(d_ivs); hipFree(d_ne); hipFree(d_kb);
+    hipFree(d_vb); hipFree(d_eb); hipFree(d_kbb); hipFree(d_vbb);
+    hipFree(d_ne64); hipFree(d_err); hipFree(tmp);
+    hipFree(out->d_keys); hipFree(out->d_key_offs);
+    hipFree(out->d_vals); hipFree(out->d_val_offs);
+    *out = DevRegion{};
+    return rc;
+  };
+  if (e == hipSuccess) e = hipMalloc(&d_blocks, total + 2048);
+  if (e == hipSuccess)
+    e = hipMemcpyAsync(d_blocks, h_blocks, total, hipMemcpyHostToDevice, s);
+  if (e == hipSuccess) e = hipMalloc(&d_ivs, n_ivs * sizeof(BlkInterval) + 16);
+  if (e == hipSuccess)
+    e = hipMemcpyAsync(d_ivs, ivs.data(), n_ivs * sizeof(BlkInterval),
+                       hipMemcpyHostToDevice, s);
+  if (e == hipSuccess) e = hipMalloc(&d_ne, n_ivs * 4 + 8);
+  if (e == hipSuccess) e = hipMalloc(&d_ne64, (n_ivs + 1) * 8 + 8);
+  if (e == hipSuccess) e = hipMalloc(&d_kb, (n_ivs + 1) * 8 + 8);
+  if (e == hipSuccess) e = hipMalloc(&d_vb, (n_ivs + 1) * 8 + 8);
+  if (e == hipSuccess) e = hipMalloc(&d_eb, (n_ivs + 1) * 8 + 8);
+  if (e == hipSuccess) e = hipMalloc(&d_kbb, (n_ivs + 1) * 8 + 8);
+  if (e == hipSuccess) e = hipMalloc(&d_vbb, (n_ivs + 1) * 8 + 8);
+  if (e == hipSuccess) e = hipMalloc(&d_err, 4);
+  if (e == hipSuccess) e = hipMemsetAsync(d_err, 0, 4, s);
+  if (e != hipSuccess) return fail(-2);
+  uint32_t blocks_g = (uint32_t)((n_ivs + 255) / 256);
+  if (!blocks_g) blocks_g = 1;
+  uint32_t lds_b = 256 * BLK_MAX_IKEY;
+  hipLaunchKernelGGL((k_blk_parse<false>), dim3(blocks_g), dim3(256), lds_b,
+                     s, d_blocks, d_ivs, n_ivs, d_ne, d_kb, d_vb, nullptr,
+                     nullptr, nullptr, nullptr, nullptr, nullptr, nullptr,
+                     d_err);
+  unsigned int h_err = 0;
+  e = hipMemcpyAsync(&h_err, d_err, 4, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  if (e != hipSuccess) return fail(-1);
+  if (h_err) return fail(-3);
+  /* widen entry counts, then exclusive-scan entries/key-bytes/val-bytes */
+  hipLaunchKernelGGL(k_blk_widen, dim3(blocks_g), dim3(256), 0, s, d_ne,
+                     d_ne64, n_ivs);
+  hipMemsetAsync(d_ne64 + n_ivs, 0, 8, s);
+  hipMemsetAsync(d_kb + n_ivs, 0, 8, s);
+  hipMemsetAsync(d_vb + n_ivs, 0, 8, s);
+  hipcub::DeviceScan::ExclusiveSum(nullptr, tmpb, d_ne64, d_eb,
+                                   (int)(n_ivs + 1), s);
+  if (hipMalloc(&tmp, tmpb) != hipSuccess) return fail(-2);
+  hipcub::DeviceScan::ExclusiveSum(tmp, tmpb, d_ne64, d_eb, (int)(n_ivs + 1),
+                                   s);
+  hipcub::DeviceScan::ExclusiveSum(tmp, tmpb, d_kb, d_kbb, (int)(n_ivs + 1),
+                                   s);
+  hipcub::DeviceScan::ExclusiveSum(tmp, tmpb, d_vb, d_vbb, (int)(n_ivs + 1),
+                                   s);
+  uint64_t n_kv = 0, key_total = 0, val_total = 0;
+  e = hipMemcpyAsync(&n_kv, d_eb + n_ivs, 8, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess)
+    e = hipMemcpyAsync(&key_total, d_kbb + n_ivs, 8, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess)
+    e = hipMemcpyAsync(&val_total, d_vbb + n_ivs, 8, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  if (e != hipSuccess) return fail(-1);
+  out->n_kv = n_kv;
+  out->key_bytes = key_total;
+  out->val_bytes = val_total;
+  if (e == hipSuccess) e = hipMalloc(&out->d_keys, key_total + 2048);
+  if (e == hipSuccess) e = hipMalloc(&out->d_vals, val_total + 2048);
+  if (e == hipSuccess) e = hipMalloc(&out->d_key_offs, (n_kv + 1) * 8 + 2048);
+  if (e == hipSuccess) e = hipMalloc(&out->d_val_offs, (n_kv + 1) * 8 + 2048);
+  if (e != hipSuccess) return fail(-2);
+  hipLaunchKernelGGL((k_blk_parse<true>), dim3(blocks_g), dim3(256), lds_b,
+                     s, d_blocks, d_ivs, n_ivs, d_ne, d_kb, d_vb, d_eb,
+                     d_kbb, d_vbb, out->d_keys, out->d_key_offs, out->d_vals,
+                     out->d_val_offs, d_err);
+  hipError_t c1 = hipMemcpyAsync(out->d_key_offs + n_kv, &key_total, 8,
+                                 hipMemcpyHostToDevice, s);
+  hipError_t c2 = hipMemcpyAsync(out->d_val_offs + n_kv, &val_total, 8,
+                                 hipMemcpyHostToDevice, s);
+  e = hipStreamSynchronize(s);
+  hipFree(d_blocks); hipFree(d_ivs); hipFree(d_ne); hipFree(d_kb);
+  hipFree(d_vb); hipFree(d_eb); hipFree(d_kbb); hipFree(d_vbb);
+  hipFree(d_ne64); hipFree(d_err); hipFree(tmp);
+  if (e != hipSuccess || c1 != hipSuccess || c2 != hipSuccess) {
+    hipFree(out->d_keys); hipFree(out->d_key_offs);
+    hipFree(out->d_vals); hipFree(out->d_val_offs);
+    *out = DevRegion{};
+    return -1;
+  }
+  return 0;
+}
+
 /* gather the given rows (in order) of a region into a new DevRegion */
 __global__ static void k_sub_sizes(const uint64_t *ko, const uint64_t *vo,
                                    const uint32_t *rows, uint64_t m,

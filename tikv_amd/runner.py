@@ -251,6 +251,21 @@ def bit_op(func, offset):
     return (func, Expr().col(offset), field_type(F.TP_LONGLONG, F.FLAG_UNSIGNED))
 
 
+def gen_blocks(gen, restart_interval=16, target_block_bytes=4096):
+    """Pack a GenRegion's KV stream into RocksDB-format data blocks.
+    Returns (blocks_ptr, block_offs_array, n_blocks, keepalive)."""
+    lib = gen._lib
+    blocks = C.POINTER(C.c_uint8)()
+    offs = C.POINTER(C.c_uint64)()
+    n = C.c_uint32()
+    st = lib.copr_gen_blocks(gen.keys, gen.key_offs, gen.vals, gen.val_offs,
+                             gen.n_kv, restart_interval, target_block_bytes,
+                             C.byref(blocks), C.byref(offs), C.byref(n))
+    if st != 0:
+        raise RuntimeError("copr_gen_blocks: %d" % st)
+    return blocks, offs, n.value, (blocks, offs)
+
+
 def cmp_col_const(offset, sig, const, unsigned_const=False):
     return Expr().col(offset).const_int(const, unsigned_const).func(sig, 2)
 
@@ -331,6 +346,16 @@ class Engine:
                                           val_offs, n_kv, C.byref(r))
         if st != 0:
             raise RuntimeError("copr_region_create: %d (%s)" %
+                               (st, self._lib.copr_last_error().decode()))
+        return Region(self, r)
+
+    def region_blocks(self, blocks, block_offs, n_blocks):
+        """Region from uncompressed RocksDB data blocks (device parse)."""
+        r = C.c_void_p()
+        st = self._lib.copr_region_create_blocks(self._h, blocks, block_offs,
+                                                 n_blocks, C.byref(r))
+        if st != 0:
+            raise RuntimeError("copr_region_create_blocks: %d (%s)" %
                                (st, self._lib.copr_last_error().decode()))
         return Region(self, r)
 
