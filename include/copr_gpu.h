@@ -128,6 +128,16 @@ copr_status copr_region_create_blocks(copr_engine *,
                                       uint32_t n_blocks,
                                       copr_region **out);
 
+/* blocks carrying write-CF records: device block parse, then the device
+ * MVCC version filter at read_ts (the full SnapshotStore::scanner chain
+ * without per-KV host round trips) */
+copr_status copr_region_create_blocks_mvcc(copr_engine *,
+                                           const uint8_t *blocks,
+                                           const uint64_t *block_offs,
+                                           uint32_t n_blocks,
+                                           uint64_t read_ts,
+                                           copr_region **out);
+
 /* fixture writer: pack a KV stream into data blocks (restart-interval
  * prefix compression; ~target_block_bytes per block). Buffers owned by
  * the generator allocator; free blocks with free() and offs with free() */

@@ -140,3 +140,26 @@ def test_block_device_parity(engine):
             rgn_b.close()
     finally:
         g.close()
+
+
+@pytest.mark.gpu
+def test_block_mvcc_chain_gpu(engine):
+    """write-CF blocks -> device block parse -> device MVCC filter equals
+    the direct write-CF array path bit-for-bit."""
+    g = tikv_amd.GenRegion(config_index=1, n_rows=50001, table_id=1,
+                           row_format=3)
+    try:
+        blocks, offs, n, keep = tikv_amd.gen_blocks(g)
+        rgn_b = engine.region_blocks_mvcc(blocks, offs, n, 1000)
+        rgn_d = engine.region_mvcc(g, 1000)
+        try:
+            b = engine.dump_region(rgn_b)
+            d = engine.dump_region(rgn_d)
+            assert b[4] == d[4]
+            assert b[0] == d[0] and b[2] == d[2]
+            assert list(b[1]) == list(d[1]) and list(b[3]) == list(d[3])
+        finally:
+            rgn_b.close()
+            rgn_d.close()
+    finally:
+        g.close()

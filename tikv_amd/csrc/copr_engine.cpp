@@ -175,6 +175,28 @@ extern "C" copr_status copr_region_create_blocks(copr_engine *eng,
   return region_from_dev(eng, dr, out);
 }
 
+extern "C" copr_status copr_region_create_blocks_mvcc(
+    copr_engine *eng, const uint8_t *blocks, const uint64_t *block_offs,
+    uint32_t n_blocks, uint64_t read_ts, copr_region **out) {
+  if (!eng) return SET_ERR(COPR_ERR_INVALID_REQUEST, "null engine");
+  HIP_TRY(hipSetDevice(eng->device), "hipSetDevice");
+  DevRegion raw{};
+  int rc = dev_blocks_build(blocks, block_offs, n_blocks, &raw, eng->stream);
+  if (rc == -3) return SET_ERR(COPR_ERR_STORAGE, "malformed data block");
+  if (rc == -2) return SET_ERR(COPR_ERR_OOM, "block ingest alloc");
+  if (rc != 0) return SET_ERR(COPR_ERR_INTERNAL, "block ingest failed");
+  DevRegion vis{};
+  int unsup = 0;
+  rc = dev_mvcc_build(raw.d_keys, raw.d_key_offs, raw.d_vals, raw.d_val_offs,
+                      raw.n_kv, read_ts, &vis, &unsup, eng->stream);
+  hipFree(raw.d_keys); hipFree(raw.d_key_offs);
+  hipFree(raw.d_vals); hipFree(raw.d_val_offs);
+  if (rc == -3) return SET_ERR(COPR_ERR_UNSUPPORTED, "mvcc: default-CF value");
+  if (rc == -2) return SET_ERR(COPR_ERR_OOM, "mvcc build oom");
+  if (rc != 0) return SET_ERR(COPR_ERR_STORAGE, "malformed write-CF entry");
+  return region_from_dev(eng, vis, out);
+}
+
 extern "C" copr_status copr_region_create_mvcc(copr_engine *eng,
                                                const uint8_t *keys,
                                                const uint64_t *key_offs,

@@ -359,6 +359,16 @@ class Engine:
                                (st, self._lib.copr_last_error().decode()))
         return Region(self, r)
 
+    def region_blocks_mvcc(self, blocks, block_offs, n_blocks, read_ts):
+        r = C.c_void_p()
+        st = self._lib.copr_region_create_blocks_mvcc(
+            self._h, blocks, block_offs, n_blocks,
+            C.c_uint64(read_ts), C.byref(r))
+        if st != 0:
+            raise RuntimeError("copr_region_create_blocks_mvcc: %d (%s)" %
+                               (st, self._lib.copr_last_error().decode()))
+        return Region(self, r)
+
     def region_mvcc(self, gen: GenRegion, read_ts):
         """Build a visible-row region from a raw write-CF stream (the
         device MVCC version filter)."""
