@@ -138,6 +138,28 @@ copr_status copr_region_create_blocks_mvcc(copr_engine *,
                                            uint64_t read_ts,
                                            copr_region **out);
 
+/* compressed-block variants (RocksDB CompressionType per block:
+ * 0 = none, 4/5 = LZ4/LZ4HC, 7 = ZSTD; compress_format_version 2 framing
+ * = varint32 decompressed size + compressed payload). Decompression runs
+ * on host cores via the system liblz4/libzstd before device parse;
+ * snappy (1) is not present in this image and returns UNSUPPORTED. */
+copr_status copr_region_create_blocks_compressed(copr_engine *,
+                                                 const uint8_t *blocks,
+                                                 const uint64_t *block_offs,
+                                                 const uint8_t *types,
+                                                 uint32_t n_blocks,
+                                                 copr_region **out);
+
+/* host-side helpers (fixture + tests): compress / decompress a block set */
+copr_status copr_blocks_compress(const uint8_t *blocks,
+                                 const uint64_t *block_offs,
+                                 uint32_t n_blocks, uint8_t type,
+                                 uint8_t **out, uint64_t **out_offs);
+copr_status copr_blocks_decompress(const uint8_t *blocks,
+                                   const uint64_t *block_offs,
+                                   const uint8_t *types, uint32_t n_blocks,
+                                   uint8_t **out, uint64_t **out_offs);
+
 /* fixture writer: pack a KV stream into data blocks (restart-interval
  * prefix compression; ~target_block_bytes per block). Buffers owned by
  * the generator allocator; free blocks with free() and offs with free() */

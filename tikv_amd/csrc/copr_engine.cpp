@@ -129,6 +129,188 @@ extern "C" copr_status copr_region_create(copr_engine *eng,
 }
 
 
+
+/* ---- block compression (host side; RocksDB util/compression.h framing,
+ * compress_format_version 2: varint32 decompressed size prefix) ---- */
+#include <dlfcn.h>
+
+namespace {
+typedef int (*lz4_c_fn)(const char *, char *, int, int);
+typedef int (*lz4_d_fn)(const char *, char *, int, int);
+typedef size_t (*zstd_bound_fn)(size_t);
+typedef size_t (*zstd_c_fn)(void *, size_t, const void *, size_t, int);
+typedef size_t (*zstd_d_fn)(void *, size_t, const void *, size_t);
+typedef unsigned (*zstd_err_fn)(size_t);
+
+struct CompLibs {
+  lz4_c_fn lz4_compress = nullptr;
+  lz4_d_fn lz4_decompress = nullptr;
+  zstd_bound_fn zstd_bound = nullptr;
+  zstd_c_fn zstd_compress = nullptr;
+  zstd_d_fn zstd_decompress = nullptr;
+  zstd_err_fn zstd_iserr = nullptr;
+};
+
+static CompLibs *comp_libs() {
+  static CompLibs libs;
+  static bool tried = false;
+  if (!tried) {
+    tried = true;
+    if (void *l4 = dlopen("liblz4.so.1", RTLD_NOW)) {
+      libs.lz4_compress = (lz4_c_fn)dlsym(l4, "LZ4_compress_default");
+      libs.lz4_decompress = (lz4_d_fn)dlsym(l4, "LZ4_decompress_safe");
+    }
+    if (void *lz = dlopen("libzstd.so.1", RTLD_NOW)) {
+      libs.zstd_bound = (zstd_bound_fn)dlsym(lz, "ZSTD_compressBound");
+      libs.zstd_compress = (zstd_c_fn)dlsym(lz, "ZSTD_compress");
+      libs.zstd_decompress = (zstd_d_fn)dlsym(lz, "ZSTD_decompress");
+      libs.zstd_iserr = (zstd_err_fn)dlsym(lz, "ZSTD_isError");
+    }
+  }
+  return &libs;
+}
+
+static void put_var32(std::vector<uint8_t> *out, uint32_t v) {
+  while (v >= 0x80) { out->push_back((uint8_t)(v | 0x80)); v >>= 7; }
+  out->push_back((uint8_t)v);
+}
+
+static bool get_var32(const uint8_t *p, size_t rem, uint32_t *v, size_t *n) {
+  uint32_t x = 0;
+  size_t i = 0;
+  int sh = 0;
+  while (i < rem && i < 5) {
+    uint8_t b = p[i++];
+    x |= (uint32_t)(b & 0x7F) << sh;
+    sh += 7;
+    if (!(b & 0x80)) { *v = x; *n = i; return true; }
+  }
+  return false;
+}
+
+static copr_status decompress_blocks(const uint8_t *blocks,
+                                     const uint64_t *block_offs,
+                                     const uint8_t *types, uint32_t n_blocks,
+                                     std::vector<uint8_t> *out,
+                                     std::vector<uint64_t> *out_offs) {
+  CompLibs *L = comp_libs();
+  out_offs->assign(1, 0);
+  for (uint32_t b = 0; b < n_blocks; b++) {
+    const uint8_t *p = blocks + block_offs[b];
+    size_t len = (size_t)(block_offs[b + 1] - block_offs[b]);
+    uint8_t t = types ? types[b] : 0;
+    if (t == 0) {
+      out->insert(out->end(), p, p + len);
+    } else if (t == 4 || t == 5 || t == 7) {
+      uint32_t raw;
+      size_t n;
+      if (!get_var32(p, len, &raw, &n))
+        return SET_ERR(COPR_ERR_STORAGE, "bad compressed block header");
+      size_t base = out->size();
+      out->resize(base + raw);
+      if (t == 7) {
+        if (!L->zstd_decompress)
+          return SET_ERR(COPR_ERR_UNSUPPORTED, "libzstd unavailable");
+        size_t r = L->zstd_decompress(out->data() + base, raw, p + n,
+                                      len - n);
+        if (L->zstd_iserr(r) || r != raw)
+          return SET_ERR(COPR_ERR_STORAGE, "zstd decompress failed");
+      } else {
+        if (!L->lz4_decompress)
+          return SET_ERR(COPR_ERR_UNSUPPORTED, "liblz4 unavailable");
+        int r = L->lz4_decompress((const char *)(p + n),
+                                  (char *)(out->data() + base), (int)(len - n),
+                                  (int)raw);
+        if (r < 0 || (uint32_t)r != raw)
+          return SET_ERR(COPR_ERR_STORAGE, "lz4 decompress failed");
+      }
+    } else {
+      return SET_ERR(COPR_ERR_UNSUPPORTED,
+                     "compression type not available (snappy/zlib absent)");
+    }
+    out_offs->push_back(out->size());
+  }
+  return COPR_OK;
+}
+}  // namespace
+
+extern "C" copr_status copr_blocks_decompress(const uint8_t *blocks,
+                                              const uint64_t *block_offs,
+                                              const uint8_t *types,
+                                              uint32_t n_blocks,
+                                              uint8_t **out,
+                                              uint64_t **out_offs) {
+  std::vector<uint8_t> dec;
+  std::vector<uint64_t> offs;
+  copr_status st = decompress_blocks(blocks, block_offs, types, n_blocks,
+                                     &dec, &offs);
+  if (st != COPR_OK) return st;
+  *out = (uint8_t *)malloc(dec.size() ? dec.size() : 1);
+  memcpy(*out, dec.data(), dec.size());
+  *out_offs = (uint64_t *)malloc(offs.size() * 8);
+  memcpy(*out_offs, offs.data(), offs.size() * 8);
+  return COPR_OK;
+}
+
+extern "C" copr_status copr_blocks_compress(const uint8_t *blocks,
+                                            const uint64_t *block_offs,
+                                            uint32_t n_blocks, uint8_t type,
+                                            uint8_t **out,
+                                            uint64_t **out_offs) {
+  CompLibs *L = comp_libs();
+  std::vector<uint8_t> all;
+  std::vector<uint64_t> offs{0};
+  for (uint32_t b = 0; b < n_blocks; b++) {
+    const uint8_t *p = blocks + block_offs[b];
+    size_t len = (size_t)(block_offs[b + 1] - block_offs[b]);
+    std::vector<uint8_t> one;
+    put_var32(&one, (uint32_t)len);
+    if (type == 4 || type == 5) {
+      if (!L->lz4_compress)
+        return SET_ERR(COPR_ERR_UNSUPPORTED, "liblz4 unavailable");
+      size_t hdr = one.size();
+      one.resize(hdr + len + len / 255 + 16);
+      int r = L->lz4_compress((const char *)p, (char *)(one.data() + hdr),
+                              (int)len, (int)(one.size() - hdr));
+      if (r <= 0) return SET_ERR(COPR_ERR_INTERNAL, "lz4 compress failed");
+      one.resize(hdr + r);
+    } else if (type == 7) {
+      if (!L->zstd_compress)
+        return SET_ERR(COPR_ERR_UNSUPPORTED, "libzstd unavailable");
+      size_t hdr = one.size();
+      size_t bound = L->zstd_bound(len);
+      one.resize(hdr + bound);
+      size_t r = L->zstd_compress(one.data() + hdr, bound, p, len, 3);
+      if (L->zstd_iserr(r))
+        return SET_ERR(COPR_ERR_INTERNAL, "zstd compress failed");
+      one.resize(hdr + r);
+    } else {
+      return SET_ERR(COPR_ERR_UNSUPPORTED, "compression type");
+    }
+    all.insert(all.end(), one.begin(), one.end());
+    offs.push_back(all.size());
+  }
+  *out = (uint8_t *)malloc(all.size() ? all.size() : 1);
+  memcpy(*out, all.data(), all.size());
+  *out_offs = (uint64_t *)malloc(offs.size() * 8);
+  memcpy(*out_offs, offs.data(), offs.size() * 8);
+  return COPR_OK;
+}
+
+extern "C" copr_status copr_region_create_blocks_compressed(
+    copr_engine *eng, const uint8_t *blocks, const uint64_t *block_offs,
+    const uint8_t *types, uint32_t n_blocks, copr_region **out) {
+  if (!eng) return SET_ERR(COPR_ERR_INVALID_REQUEST, "null engine");
+  std::vector<uint8_t> dec;
+  std::vector<uint64_t> offs;
+  copr_status st = decompress_blocks(blocks, block_offs, types, n_blocks,
+                                     &dec, &offs);
+  if (st != COPR_OK) return st;
+  return copr_region_create_blocks(eng, dec.data(), offs.data(), n_blocks,
+                                   out);
+}
+
+
 /* ---- SST data-block ingestion (copr_gpu.h) ---- */
 static copr_status region_from_dev(copr_engine *eng, DevRegion dr,
                                    copr_region **out) {
