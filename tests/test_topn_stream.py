@@ -895,3 +895,28 @@ def test_rpn_overflow_errors_gpu(engine):
             engine.dag_run(req, [rgn])
     finally:
         rgn.close()
+
+
+@pytest.mark.gpu
+def test_first_over_hash_groups_gpu(engine):
+    """FIRST in a FastHash (int group) request routes through the sorted
+    pipeline; groups compare order-insensitively against the oracle."""
+    orc = _orc()
+    g = tikv_amd.GenRegion(config_index=0, n_rows=60001, table_id=5)
+    try:
+        rgn = engine.region(g)
+        try:
+            cols = [tikv_amd.Col(i) for i in range(1, 5)]
+            req = (tikv_amd.DagSelect(cols)
+                   .hash_agg([tikv_amd.first_col(1), tikv_amd.count_star(),
+                              tikv_amd.max_col(3)],
+                             tikv_amd.Expr().col(0)).build())
+            gd, gr, _ = engine.dag_run(req, [rgn])
+            od, orows = orc.dag_run(req, g.keys, g.key_offs, g.vals,
+                                    g.val_offs, g.n_kv)
+            assert gr == orows
+            assert set(split_rows(gd, 4)) == set(split_rows(od, 4))
+        finally:
+            rgn.close()
+    finally:
+        g.close()

@@ -484,6 +484,7 @@ struct HostPlan {
   bool hash_agg = false;
   bool stream_agg = false;
   bool simple_as_stream = false;   /* simple agg with FIRST: one-run stream */
+  bool hash_sorted = false;        /* FastHash + FIRST: sorted pipeline */
   CoprFieldType group_ft{};
   /* TopN (top_n_executor.rs): single int order-by column */
   bool has_topn = false;
@@ -1027,10 +1028,12 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
     sp.group_col_unsigned = (ci.ft.flag & COPR_FLAG_UNSIGNED) ? 1 : 0;
     pl->group_ft = ci.ft;
     pl->out_schema.push_back(ci.ft);
-    if (any_first && sp.mode == 2)
-      return SET_ERR(COPR_ERR_UNSUPPORTED,
-                     "FIRST over int-key hash agg not yet native "
-                     "(stream/bytes grouping supports it)");
+    if (any_first && sp.mode == 2) {
+      /* the atomic hash table cannot order rows; route through the
+         sorted-segmented pipeline (exact u64 key compare) */
+      sp.mode = 3;
+      pl->hash_sorted = true;
+    }
   } else if (any_first) {
     if (sp.index_mode)
       return SET_ERR(COPR_ERR_UNSUPPORTED,
@@ -1644,7 +1647,11 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
     std::vector<long long> h_gk;
     std::vector<uint8_t> h_gs;
     hipEventRecord(ev_a, eng->stream);
-    int n_seg = dev_stream_agg(sp, r->dev, eng->stream, &h_accs, &h_gk, &h_gs);
+    int n_seg = pl.hash_sorted
+                    ? dev_int_sorted_agg(sp, r->dev, eng->stream, &h_accs,
+                                         &h_gk, &h_gs)
+                    : dev_stream_agg(sp, r->dev, eng->stream, &h_accs, &h_gk,
+                                     &h_gs);
     hipEventRecord(ev_b, eng->stream);
     timed = true;
     if (n_seg == -3)

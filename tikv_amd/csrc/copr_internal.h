@@ -232,6 +232,10 @@ int dev_topn_select(const ScanPlan &plan, const DevRegion &rgn,
 int dev_ht_compact(const HashAggTable &ht, uint32_t tsize, int n_aggs,
                    void *stream, std::vector<long long> *h_keys,
                    std::vector<SimpleAggAcc> *h_accs);
+int dev_int_sorted_agg(const ScanPlan &plan, const DevRegion &rgn,
+                       void *stream, std::vector<SimpleAggAcc> *h_accs,
+                       std::vector<long long> *h_gk,
+                       std::vector<uint8_t> *h_gs);
 int dev_bytes_agg(const ScanPlan &plan, const DevRegion &rgn, void *stream,
                   std::vector<SimpleAggAcc> *h_accs,
                   std::vector<uint64_t> *h_kofs, std::vector<uint32_t> *h_klen,

@@ -3816,7 +3816,7 @@ __global__ static void k_bytes_boundary(const uint64_t *ch, const uint64_t *co,
                                         const uint32_t *cl, const uint8_t *cs,
                                         const uint32_t *perm,
                                         const uint8_t *vals, uint64_t m,
-                                        uint32_t *f) {
+                                        int int_keys, uint32_t *f) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= m) return;
   if (i == 0) { f[0] = 1; return; }
@@ -3824,14 +3824,23 @@ __global__ static void k_bytes_boundary(const uint64_t *ch, const uint64_t *co,
   uint32_t nf = 0;
   if (cs[a] != cs[b]) nf = 1;
   else if (cs[a] == 2) {
-    if (ch[a] != ch[b] || cl[a] != cl[b]) nf = 1;
-    else {
-      const uint8_t *pa = vals + co[a], *pb = vals + co[b];
-      for (uint32_t t = 0; t < cl[a]; t++)
-        if (pa[t] != pb[t]) { nf = 1; break; }
+    if (ch[a] != ch[b]) nf = 1;
+    else if (!int_keys) {
+      /* fingerprint tie: verify payload bytes (grouping stays exact) */
+      if (cl[a] != cl[b]) nf = 1;
+      else {
+        const uint8_t *pa = vals + co[a], *pb = vals + co[b];
+        for (uint32_t t = 0; t < cl[a]; t++)
+          if (pa[t] != pb[t]) { nf = 1; break; }
+      }
     }
   }
   f[i] = nf;
+}
+
+__global__ static void k_iota32(uint32_t *p, uint64_t n) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) p[i] = (uint32_t)i;
 }
 
 __global__ static void k_bytes_update(const uint64_t *co, const uint32_t *cl,
@@ -3997,7 +4006,7 @@ int dev_bytes_agg(const ScanPlan &plan, const DevRegion &rgn, void *stream,
   uint32_t mblocks = (uint32_t)((m + 255) / 256);
   if (hipMalloc(&segid, m * 4) != hipSuccess) { freeall(); return -2; }
   hipLaunchKernelGGL(k_bytes_boundary, dim3(mblocks), dim3(256), 0, s, ch, co,
-                     cl, cs, perm, rgn.d_vals, m, f32);
+                     cl, cs, perm, rgn.d_vals, m, /*int_keys=*/0, f32);
   size_t tb2 = 0;
   hipcub::DeviceScan::InclusiveSum(nullptr, tb2, f32, segid, (int)m, s);
   if (tb2 > tmpb) {
@@ -4285,6 +4294,152 @@ int dev_blocks_build(const uint8_t *h_blocks, const uint64_t *h_block_offs,
     return -1;
   }
   return 0;
+}
+
+
+/* int-key grouped aggregation through the SORTED pipeline: exact u64 key
+ * compare (no fingerprint verify). Used when a FastHash request carries
+ * FIRST, which the atomic-table path cannot order. Returns like
+ * dev_stream_agg; group keys come back in h_gk. */
+int dev_int_sorted_agg(const ScanPlan &plan, const DevRegion &rgn,
+                       void *stream, std::vector<SimpleAggAcc> *h_accs,
+                       std::vector<long long> *h_gk,
+                       std::vector<uint8_t> *h_gs) {
+  hipStream_t s = (hipStream_t)stream;
+  uint64_t n = rgn.n_kv;
+  if (!n) return 0;
+  uint8_t *st = nullptr, *as_ = nullptr, *cs = nullptr, *cas = nullptr;
+  uint8_t *rst = nullptr;
+  int64_t *gkey = nullptr, *av = nullptr, *ck = nullptr, *cav = nullptr;
+  uint64_t *rofs = nullptr, *pos = nullptr, *ch2 = nullptr;
+  uint32_t *cl0 = nullptr, *f32 = nullptr, *segid = nullptr;
+  uint32_t *idx = nullptr, *perm = nullptr, *rlen = nullptr;
+  unsigned int *d_err = nullptr;
+  SimpleAggAcc *accs = nullptr;
+  void *tmp = nullptr;
+  size_t tmpb = 0;
+  auto freeall = [&]() {
+    hipFree(st); hipFree(as_); hipFree(cs); hipFree(cas); hipFree(rst);
+    hipFree(gkey); hipFree(av); hipFree(ck); hipFree(cav); hipFree(rofs);
+    hipFree(pos); hipFree(ch2); hipFree(cl0); hipFree(f32); hipFree(segid);
+    hipFree(idx); hipFree(perm); hipFree(rlen); hipFree(d_err);
+    hipFree(accs); hipFree(tmp);
+  };
+  int na = plan.n_aggs;
+  hipError_t e = hipSuccess;
+  if (e == hipSuccess) e = hipMalloc(&st, n);
+  if (e == hipSuccess) e = hipMalloc(&gkey, n * 8);
+  if (e == hipSuccess) e = hipMalloc(&av, (uint64_t)na * n * 8 + 8);
+  if (e == hipSuccess) e = hipMalloc(&as_, (uint64_t)na * n + 8);
+  if (e == hipSuccess) e = hipMalloc(&d_err, 4);
+  if (e == hipSuccess) e = hipMemsetAsync(d_err, 0, 4, s);
+  if (e != hipSuccess) { freeall(); return -2; }
+  ExtractOut eo{st, gkey, av, as_, nullptr, nullptr, nullptr};
+  if (extract_launch(plan, rgn, eo, d_err, s)) { freeall(); return -1; }
+  unsigned int h_err = 0;
+  e = hipMemcpyAsync(&h_err, d_err, 4, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  if (e != hipSuccess) { freeall(); return -1; }
+  if (h_err) { freeall(); return -3; }
+
+  uint32_t blocks = (uint32_t)((n + 255) / 256);
+  if (e == hipSuccess) e = hipMalloc(&f32, n * 4 + 4);
+  if (e == hipSuccess) e = hipMalloc(&pos, n * 8 + 8);
+  if (e != hipSuccess) { freeall(); return -2; }
+  hipLaunchKernelGGL(k_st_keep_flags, dim3(blocks), dim3(256), 0, s, st, f32,
+                     n, 1, 2);
+  hipcub::DeviceScan::ExclusiveSum(nullptr, tmpb, f32, pos, (int)n, s);
+  if (hipMalloc(&tmp, tmpb) != hipSuccess) { freeall(); return -2; }
+  hipcub::DeviceScan::ExclusiveSum(tmp, tmpb, f32, pos, (int)n, s);
+  uint64_t m = 0;
+  uint32_t lf = 0;
+  e = hipMemcpyAsync(&m, pos + (n - 1), 8, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess)
+    e = hipMemcpyAsync(&lf, f32 + (n - 1), 4, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  if (e != hipSuccess) { freeall(); return -1; }
+  m += lf;
+  if (!m) { freeall(); return 0; }
+
+  if (e == hipSuccess) e = hipMalloc(&ck, m * 8);
+  if (e == hipSuccess) e = hipMalloc(&ch2, m * 8);
+  if (e == hipSuccess) e = hipMalloc(&cs, m);
+  if (e == hipSuccess) e = hipMalloc(&cl0, m * 4);
+  if (e == hipSuccess) e = hipMalloc(&idx, m * 4);
+  if (e == hipSuccess) e = hipMalloc(&perm, m * 4);
+  if (e == hipSuccess) e = hipMalloc(&cav, (uint64_t)na * m * 8 + 8);
+  if (e == hipSuccess) e = hipMalloc(&cas, (uint64_t)na * m + 8);
+  if (e != hipSuccess) { freeall(); return -2; }
+  hipMemsetAsync(cl0, 0, m * 4, s);
+  hipLaunchKernelGGL(k_compact_rows, dim3(blocks), dim3(256), 0, s, st, gkey,
+                     av, as_, pos, n, m, na, ck, cs, cav, cas,
+                     (uint32_t *)nullptr);
+  uint32_t mblocks = (uint32_t)((m + 255) / 256);
+  hipLaunchKernelGGL(k_iota32, dim3(mblocks), dim3(256), 0, s, idx, m);
+  size_t tb = 0;
+  hipcub::DeviceRadixSort::SortPairs(nullptr, tb, (const uint64_t *)ck, ch2,
+                                     idx, perm, (int)m, 0, 64, s);
+  if (tb > tmpb) {
+    hipFree(tmp); tmp = nullptr;
+    if (hipMalloc(&tmp, tb) != hipSuccess) { freeall(); return -2; }
+    tmpb = tb;
+  }
+  hipcub::DeviceRadixSort::SortPairs(tmp, tmpb, (const uint64_t *)ck, ch2,
+                                     idx, perm, (int)m, 0, 64, s);
+  if (hipMalloc(&segid, m * 4) != hipSuccess) { freeall(); return -2; }
+  hipLaunchKernelGGL(k_bytes_boundary, dim3(mblocks), dim3(256), 0, s,
+                     (const uint64_t *)ck, (const uint64_t *)ck, cl0, cs,
+                     perm, rgn.d_vals, m, /*int_keys=*/1, f32);
+  size_t tb2 = 0;
+  hipcub::DeviceScan::InclusiveSum(nullptr, tb2, f32, segid, (int)m, s);
+  if (tb2 > tmpb) {
+    hipFree(tmp); tmp = nullptr;
+    if (hipMalloc(&tmp, tb2) != hipSuccess) { freeall(); return -2; }
+    tmpb = tb2;
+  }
+  hipcub::DeviceScan::InclusiveSum(tmp, tmpb, f32, segid, (int)m, s);
+  uint32_t n_seg = 0;
+  e = hipMemcpyAsync(&n_seg, segid + (m - 1), 4, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  if (e != hipSuccess) { freeall(); return -1; }
+
+  if (e == hipSuccess)
+    e = hipMalloc(&accs, (uint64_t)n_seg * na * sizeof(SimpleAggAcc));
+  if (e == hipSuccess) e = hipMalloc(&rofs, (uint64_t)n_seg * 8);
+  if (e == hipSuccess) e = hipMalloc(&rlen, (uint64_t)n_seg * 4);
+  if (e == hipSuccess) e = hipMalloc(&rst, n_seg);
+  if (e == hipSuccess)
+    e = hipMemsetAsync(accs, 0, (uint64_t)n_seg * na * sizeof(SimpleAggAcc), s);
+  if (e != hipSuccess) { freeall(); return -2; }
+  hipLaunchKernelGGL(k_bytes_update, dim3(mblocks), dim3(256), 0, s,
+                     (const uint64_t *)ck, cl0, cs, cav, cas, perm, segid, m,
+                     plan, accs, rofs, rlen, rst);
+  bool has_first = false;
+  for (int a = 0; a < na; a++)
+    if (plan.aggs[a].kind == DAGG_FIRST) has_first = true;
+  if (has_first) {
+    hipLaunchKernelGGL(k_bytes_first_fix, dim3(mblocks), dim3(256), 0, s,
+                       cav, cas, perm, segid, m, plan, accs);
+    uint32_t sblocks = (uint32_t)((n_seg + 255) / 256);
+    if (sblocks)
+      hipLaunchKernelGGL(k_bytes_first_norm, dim3(sblocks), dim3(256), 0, s,
+                         (uint64_t)n_seg, plan, accs);
+  }
+  h_accs->resize((size_t)n_seg * na);
+  h_gk->resize(n_seg);
+  h_gs->resize(n_seg);
+  e = hipMemcpyAsync(h_accs->data(), accs,
+                     (uint64_t)n_seg * na * sizeof(SimpleAggAcc),
+                     hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess)
+    e = hipMemcpyAsync(h_gk->data(), rofs, (uint64_t)n_seg * 8,
+                       hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess)
+    e = hipMemcpyAsync(h_gs->data(), rst, n_seg, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  freeall();
+  if (e != hipSuccess) return -1;
+  return (int)n_seg;
 }
 
 /* gather the given rows (in order) of a region into a new DevRegion */
