@@ -2616,24 +2616,27 @@ k_crc64(const uint8_t *__restrict__ vals, const uint64_t *__restrict__ val_offs,
       uint32_t klen = (uint32_t)(key_offs[my_row + 1] - key_offs[my_row]);
       const uint8_t *vp = val_lds + vshift + (uint32_t)(val_offs[my_row] - vbase);
       uint32_t vlen = (uint32_t)(val_offs[my_row + 1] - val_offs[my_row]);
-      for (uint32_t i = 0; i < klen; i++)
-        crc = tab[(uint32_t)((crc ^ kp[i]) & 0xFF)] ^ (crc >> 8);
-      uint32_t i = 0;
-      for (; i + 8 <= vlen; i += 8) {
-        uint64_t x;
-        memcpy(&x, vp + i, 8);
-        crc ^= x;
-        crc = tab[7 * 256 + (uint32_t)(crc & 0xFF)] ^
-              tab[6 * 256 + (uint32_t)((crc >> 8) & 0xFF)] ^
-              tab[5 * 256 + (uint32_t)((crc >> 16) & 0xFF)] ^
-              tab[4 * 256 + (uint32_t)((crc >> 24) & 0xFF)] ^
-              tab[3 * 256 + (uint32_t)((crc >> 32) & 0xFF)] ^
-              tab[2 * 256 + (uint32_t)((crc >> 40) & 0xFF)] ^
-              tab[1 * 256 + (uint32_t)((crc >> 48) & 0xFF)] ^
-              tab[0 * 256 + (uint32_t)(crc >> 56)];
-      }
-      for (; i < vlen; i++)
-        crc = tab[(uint32_t)((crc ^ vp[i]) & 0xFF)] ^ (crc >> 8);
+      /* slice-by-8 over both streams; lds_win8 keeps every 8-byte load an
+         aligned ds_read_b64 pair (a raw misaligned b64 replays 64 cycles) */
+      auto crc8 = [&](const uint8_t *p, uint32_t len, uint64_t c) {
+        uint32_t i = 0;
+        for (; i + 8 <= len; i += 8) {
+          c ^= lds_win8(p + i);
+          c = tab[7 * 256 + (uint32_t)(c & 0xFF)] ^
+              tab[6 * 256 + (uint32_t)((c >> 8) & 0xFF)] ^
+              tab[5 * 256 + (uint32_t)((c >> 16) & 0xFF)] ^
+              tab[4 * 256 + (uint32_t)((c >> 24) & 0xFF)] ^
+              tab[3 * 256 + (uint32_t)((c >> 32) & 0xFF)] ^
+              tab[2 * 256 + (uint32_t)((c >> 40) & 0xFF)] ^
+              tab[1 * 256 + (uint32_t)((c >> 48) & 0xFF)] ^
+              tab[0 * 256 + (uint32_t)(c >> 56)];
+        }
+        for (; i < len; i++)
+          c = tab[(uint32_t)((c ^ p[i]) & 0xFF)] ^ (c >> 8);
+        return c;
+      };
+      crc = crc8(kp, klen, crc);
+      crc = crc8(vp, vlen, crc);
       acc ^= ~crc;
     }
   }
