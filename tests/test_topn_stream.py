@@ -920,3 +920,107 @@ def test_first_over_hash_groups_gpu(engine):
             rgn.close()
     finally:
         g.close()
+
+
+# ---- Real (f64) aggregates ---------------------------------------------
+import struct as _struct
+
+
+def cell_real(col_id, x):
+    bits = _struct.unpack("<Q", _struct.pack("<d", x))[0]
+    u = (bits | (1 << 63)) if not (bits >> 63) else (~bits) & (2**64 - 1)
+    return b"\x08" + var_i64(col_id) + b"\x05" + u.to_bytes(8, "big")
+
+
+def region_real(vals_list):
+    rows = []
+    keys = b""
+    vbytes = b""
+    vo = [0]
+    for i, x in enumerate(vals_list):
+        keys += row_key(i)
+        v = cell_int(1, i)
+        v += cell_null(2) if x is None else cell_real(2, x)
+        vbytes += v
+        vo.append(len(vbytes))
+    ko = [19 * i for i in range(len(vals_list) + 1)]
+    kb = (C.c_uint8 * len(keys)).from_buffer_copy(keys)
+    vb = (C.c_uint8 * len(vbytes)).from_buffer_copy(vbytes)
+    return (C.cast(kb, C.POINTER(C.c_uint8)), (C.c_uint64 * len(ko))(*ko),
+            C.cast(vb, C.POINTER(C.c_uint8)), (C.c_uint64 * len(vo))(*vo),
+            len(vals_list), (kb, vb))
+
+
+def dec_uint(data, pos):
+    assert data[pos] == 4
+    return int.from_bytes(data[pos + 1:pos + 9], "big"), pos + 9
+
+
+def dec_real(data, pos):
+    assert data[pos] == 5
+    u = int.from_bytes(data[pos + 1:pos + 9], "big")
+    if u >> 63:
+        u &= (1 << 63) - 1
+    else:
+        u = (~u) & (2**64 - 1)
+    return _struct.unpack("<d", _struct.pack("<Q", u))[0], pos + 9
+
+
+def test_real_aggs_oracle():
+    orc = _orc()
+    xs = [1.5, -2.25, None, 10.0, 0.5]
+    k, ko, v, vo, n, keep = region_real(xs)
+    cols = [tikv_amd.Col(1), tikv_amd.Col(2, tp=F.TP_DOUBLE)]
+    req = (tikv_amd.DagSelect(cols)
+           .simple_agg([tikv_amd.sum_real(1), tikv_amd.avg_real(1),
+                        tikv_amd.max_col(1, tp=F.TP_DOUBLE),
+                        tikv_amd.min_col(1, tp=F.TP_DOUBLE),
+                        tikv_amd.first_col(1, tp=F.TP_DOUBLE)]).build())
+    data, nrows = orc.dag_run(req, k, ko, v, vo, n)
+    assert nrows == 1
+    s, p = dec_real(data, 0)                      # sum
+    cnt, p = dec_uint(data, p)                    # avg count (UNSIGNED)
+    av, p = dec_real(data, p)                     # avg sum
+    mx, p = dec_real(data, p)
+    mn, p = dec_real(data, p)
+    fv, p = dec_real(data, p)
+    assert (s, cnt, av, mx, mn, fv) == (9.75, 4, 9.75, 10.0, -2.25, 1.5)
+    assert p == len(data)
+
+
+@pytest.mark.gpu
+def test_real_aggs_gpu_parity(engine):
+    import random
+    rng = random.Random(7)
+    xs = [None if rng.random() < 0.05 else rng.uniform(-1e6, 1e6)
+          for _ in range(30000)]
+    k, ko, v, vo, n, keep = region_real(xs)
+    orc = _orc()
+    rgn = engine.region_raw(k, ko, v, vo, n)
+    try:
+        cols = [tikv_amd.Col(1), tikv_amd.Col(2, tp=F.TP_DOUBLE)]
+        sel = tikv_amd.cmp_col_const(0, F.SIG_GE_INT, 1000)
+        req = (tikv_amd.DagSelect(cols).where(sel)
+               .simple_agg([tikv_amd.count_col(1),
+                            tikv_amd.sum_real(1),
+                            tikv_amd.max_col(1, tp=F.TP_DOUBLE),
+                            tikv_amd.min_col(1, tp=F.TP_DOUBLE),
+                            tikv_amd.first_col(1, tp=F.TP_DOUBLE)]).build())
+        gd, gr, _ = engine.dag_run(req, [rgn])
+        od, orows = orc.dag_run(req, k, ko, v, vo, n)
+        assert gr == orows == 1
+        gc, p = dec_int(gd, 0)
+        gs, p = dec_real(gd, p)
+        gmx, p = dec_real(gd, p)
+        gmn, p = dec_real(gd, p)
+        gf, p = dec_real(gd, p)
+        oc_, q = dec_int(od, 0)
+        os_, q = dec_real(od, q)
+        omx, q = dec_real(od, q)
+        omn, q = dec_real(od, q)
+        of_, q = dec_real(od, q)
+        # count / max / min / first: bit-exact; sum: parallel order -> ULPs
+        assert (gc, gmx, gmn, gf) == (oc_, omx, omn, of_)
+        assert abs(gs - os_) <= 1e-9 * max(1.0, abs(os_))
+    finally:
+        rgn.close()

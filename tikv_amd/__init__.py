@@ -9,6 +9,6 @@ Product layout:
 from . import _ffi  # noqa: F401
 from .runner import (  # noqa: F401
     Col, Expr, DagSelect, Engine, Region, GenRegion,
-    count_star, count_col, sum_col, avg_col, max_col, min_col, first_col, bit_op,
+    count_star, count_col, sum_col, avg_col, sum_real, avg_real, max_col, min_col, first_col, bit_op,
     cmp_col_const, field_type, gen_blocks,
 )

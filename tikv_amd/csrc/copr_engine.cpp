@@ -954,6 +954,9 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
         if (!ci) return SET_ERR(COPR_ERR_UNSUPPORTED, "sum/avg needs a column");
         if (et_int(ci->ft.tp)) {
           ds.kind = DAGG_SUM_INT;   /* exact int sum == Decimal rewrite sum */
+        } else if (ci->ft.tp == COPR_TP_DOUBLE) {
+          /* f64 sum: parallel order -> 1-ULP class (north_star budget) */
+          ds.kind = DAGG_SUM_REAL;
         } else if (ci->ft.tp == COPR_TP_NEWDECIMAL) {
           ds.kind = DAGG_SUM_DEC;
           ds.target_frac = ci->ft.decimal >= 0 ? ci->ft.decimal : 0;
@@ -965,14 +968,20 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
         oa.in_kind = ds.kind;
         break;
       case COPR_AGG_MAX: case COPR_AGG_MIN:
-        if (!ci || !et_int(ci->ft.tp))
-          return SET_ERR(COPR_ERR_UNSUPPORTED, "max/min native only over int");
-        ds.kind = ad.func == COPR_AGG_MAX ? DAGG_MAX_INT : DAGG_MIN_INT;
+        if (ci && ci->ft.tp == COPR_TP_DOUBLE) {
+          ds.kind = ad.func == COPR_AGG_MAX ? DAGG_MAX_REAL : DAGG_MIN_REAL;
+        } else if (ci && et_int(ci->ft.tp)) {
+          ds.kind = ad.func == COPR_AGG_MAX ? DAGG_MAX_INT : DAGG_MIN_INT;
+        } else {
+          return SET_ERR(COPR_ERR_UNSUPPORTED,
+                         "max/min native only over int/double");
+        }
         oa.in_kind = ds.kind;
         break;
       case COPR_AGG_FIRST:
-        if (!ci || !et_int(ci->ft.tp))
-          return SET_ERR(COPR_ERR_UNSUPPORTED, "first native only over int");
+        if (!ci || (!et_int(ci->ft.tp) && ci->ft.tp != COPR_TP_DOUBLE))
+          return SET_ERR(COPR_ERR_UNSUPPORTED,
+                         "first native only over int/double");
         ds.kind = DAGG_FIRST;
         oa.in_kind = DAGG_FIRST;
         break;
@@ -1111,6 +1120,13 @@ static void enc_datum_int(std::vector<uint8_t> *out, int64_t v, bool uns) {
   enc_cmp_u64(out, uns ? (uint64_t)v : ((uint64_t)v ^ 0x8000000000000000ull));
 }
 static void enc_datum_null(std::vector<uint8_t> *out) { out->push_back(0); }
+static void enc_datum_real(std::vector<uint8_t> *out, uint64_t bits) {
+  /* FLOAT datum: comparable f64 (convert.rs:16-22) */
+  out->push_back(5);
+  uint64_t u = (bits & 0x8000000000000000ull) ? ~bits
+                                              : (bits | 0x8000000000000000ull);
+  for (int i = 7; i >= 0; i--) out->push_back((uint8_t)(u >> (8 * i)));
+}
 static void enc_datum_dec_scaled(std::vector<uint8_t> *out, __int128 scaled, uint8_t frac) {
   prod::PDec d = prod::pdec_from_scaled_i128(scaled, frac);
   uint8_t prec, fr;
@@ -1145,13 +1161,22 @@ static void encode_agg_row(const HostPlan &pl, const SimpleAggAcc *accs,
       /* first row's value, NULL included (impl_first.rs) */
       bool uns = (pl.out_schema[oc].flag & COPR_FLAG_UNSIGNED) != 0;
       if (ac.cnt == 0 || ac.sum_hi) enc_datum_null(&(*cols_out)[oc]);
+      else if (pl.out_schema[oc].tp == COPR_TP_DOUBLE)
+        enc_datum_real(&(*cols_out)[oc], ac.sum_lo);
       else enc_datum_int(&(*cols_out)[oc], (int64_t)ac.sum_lo, uns);
       oc++;
     } else if (oa.func == COPR_AGG_MAX || oa.func == COPR_AGG_MIN) {
-      /* undo the fold transform (d_fold_xform): MIN folded max over ~biased */
+      /* undo the fold transform (d_fold_xform): MIN folded max over ~x */
       bool uns = (pl.out_schema[oc].flag & COPR_FLAG_UNSIGNED) != 0;
       if (ac.cnt == 0) enc_datum_null(&(*cols_out)[oc]);
-      else {
+      else if (oa.in_kind == DAGG_MAX_REAL || oa.in_kind == DAGG_MIN_REAL) {
+        unsigned long long b = ac.sum_lo;
+        if (oa.func == COPR_AGG_MIN) b = ~b;
+        uint64_t bits = (b & 0x8000000000000000ull)
+                            ? (b & 0x7FFFFFFFFFFFFFFFull)
+                            : ~b;
+        enc_datum_real(&(*cols_out)[oc], bits);
+      } else {
         unsigned long long b = ac.sum_lo;
         if (oa.func == COPR_AGG_MIN) b = ~b;
         bool col_uns = pl.sp.aggs[oa.dev_idx].col_unsigned != 0;
@@ -1173,7 +1198,9 @@ static void encode_agg_row(const HostPlan &pl, const SimpleAggAcc *accs,
       }
       std::vector<uint8_t> &sumcol = (*cols_out)[oc++];
       if (ac.cnt == 0) { enc_datum_null(&sumcol); }
-      else {
+      else if (oa.in_kind == DAGG_SUM_REAL) {
+        enc_datum_real(&sumcol, ac.sum_lo);   /* f64 bits live in sum_lo */
+      } else {
         __int128 s = i128_of(ac.sum_lo, ac.sum_hi);
         uint8_t frac = oa.in_kind == DAGG_SUM_DEC
                            ? (uint8_t)pl.sp.aggs[oa.dev_idx].target_frac : 0;

@@ -49,6 +49,11 @@ enum {
        BIT_OR / BIT_XOR: fold or/xor over v */
   DAGG_MAX_INT,
   DAGG_MIN_INT,
+  /* real (f64) max/min fold via the order-preserving bits transform
+     (sign-clear -> |SIGN, sign-set -> ~bits); values travel as f64 BITS
+     in the i64 channels */
+  DAGG_MAX_REAL,
+  DAGG_MIN_REAL,
   DAGG_BIT_AND,
   DAGG_BIT_OR,
   DAGG_BIT_XOR,
@@ -57,6 +62,9 @@ enum {
      single-run stream pass): acc.cnt = 1 once set, sum_lo = value,
      sum_hi = 1 when the first row's value is NULL. */
   DAGG_FIRST,
+  /* f64 sum: atomicAdd(double) on sum_lo's bits; parallel order makes it
+     1-ULP-class, within north_star's float budget */
+  DAGG_SUM_REAL,
 };
 
 /* flattened RPN node for the device predicate evaluator (mirrors the

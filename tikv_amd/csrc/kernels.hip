@@ -176,13 +176,15 @@ struct CellView {
   uint8_t flag;
   bool is_null;
   bool has_int;
-  int64_t ival;
+  int64_t ival;        /* also carries f64 BITS when has_real */
+  bool has_real;
   bool has_dec;
   int64_t dscaled; int32_t dfrac;
 };
 
 __device__ static inline void d_parse_datum(const uint8_t *p, uint32_t rem, CellView *cv) {
   cv->len = 0; cv->is_null = false; cv->has_int = false; cv->has_dec = false;
+  cv->has_real = false;
   if (rem == 0) return;
   uint8_t flag = p[0];
   cv->flag = flag;
@@ -201,7 +203,16 @@ __device__ static inline void d_parse_datum(const uint8_t *p, uint32_t rem, Cell
       cv->has_int = true;
       cv->ival = (int64_t)d_be_u64(pl);
       cv->len = 9; return;
-    case 5: case 7:  /* FLOAT / DURATION: 8B */
+    case 5: {  /* FLOAT: comparable f64 (convert.rs:16-22) -> raw bits */
+      if (prem < 8) return;
+      uint64_t u = d_be_u64(pl);
+      if (u & 0x8000000000000000ull) u &= 0x7FFFFFFFFFFFFFFFull;
+      else u = ~u;
+      cv->has_real = true;
+      cv->ival = (int64_t)u;
+      cv->len = 9; return;
+    }
+    case 7:  /* DURATION: 8B */
       if (prem < 8) return;
       cv->len = 9; return;
     case 8: {  /* VAR_INT */
@@ -348,6 +359,12 @@ __device__ static inline bool next_cell(const uint8_t *vp, uint32_t vlen,
 __device__ static inline unsigned long long d_fold_xform(int32_t kind,
                                                          int64_t v, bool uns) {
   unsigned long long b = (unsigned long long)v;
+  if (kind == DAGG_MAX_REAL || kind == DAGG_MIN_REAL) {
+    /* f64 bits -> order-preserving u64 (same map the comparable-f64 key
+       encode uses, convert.rs:16-22) */
+    b = (b & 0x8000000000000000ull) ? ~b : (b | 0x8000000000000000ull);
+    return kind == DAGG_MAX_REAL ? b : ~b;
+  }
   if (!uns) b ^= 0x8000000000000000ull;       /* order-preserving bias */
   switch (kind) {
     case DAGG_MAX_INT: return b;
@@ -582,7 +599,7 @@ __device__ static inline bool d_index_collect(const ScanPlan &plan,
         cols[a].iv = cell.ival;
         cols[a].has_dec = cell.has_dec;
         cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
-        if (!cell.is_null && !cell.has_int && !cell.has_dec) return false;
+        if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec) return false;
       }
     }
     pos += cell.len;
@@ -983,7 +1000,7 @@ k_scan_agg(ScanPlan plan,
                 cols[a].iv = cell.ival;
                 cols[a].has_dec = cell.has_dec;
                 cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
-                if (!cell.is_null && !cell.has_int && !cell.has_dec)
+                if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec)
                   parse_ok = false;
               } else parse_ok = false;
             }
@@ -1036,7 +1053,7 @@ k_scan_agg(ScanPlan plan,
               cols[a].iv = cell.ival;
               cols[a].has_dec = cell.has_dec;
               cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
-              if (!cell.is_null && !cell.has_int && !cell.has_dec) parse_ok = false;
+              if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec) parse_ok = false;
               found++;
             }
           }
@@ -1114,7 +1131,8 @@ k_scan_agg(ScanPlan plan,
             contribute = false;
           } else if (sp.kind == DAGG_COUNT_COL) {
             contribute = true;
-          } else if (sp.kind == DAGG_SUM_INT || d_is_fold(sp.kind)) {
+          } else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_REAL ||
+                     d_is_fold(sp.kind)) {
             contribute = true; v = cols[a].iv;
           } else {  /* SUM_DEC */
             int d = sp.target_frac - cols[a].dfr;
@@ -1132,19 +1150,26 @@ k_scan_agg(ScanPlan plan,
           if (IS_HASH) {
             if (acc_base) {
               atomicAdd(&acc_base[a].cnt, 1ull);
-              if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC)
+              if (sp.kind == DAGG_SUM_REAL)
+                atomicAdd((double *)&acc_base[a].sum_lo,
+                          __longlong_as_double(v));
+              else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC)
                 atomic_add_i128(&acc_base[a].sum_lo, &acc_base[a].sum_hi, v);
               else if (d_is_fold(sp.kind)) {
                 unsigned long long b = d_fold_xform(sp.kind, v, sp.col_unsigned);
                 if (d_is_xor(sp.kind)) atomicXor(&acc_base[a].sum_lo, b);
-                else if (sp.kind == DAGG_MAX_INT || sp.kind == DAGG_MIN_INT)
+                else if (sp.kind == DAGG_MAX_INT || sp.kind == DAGG_MIN_INT || sp.kind == DAGG_MAX_REAL || sp.kind == DAGG_MIN_REAL)
                   atomicMax(&acc_base[a].sum_lo, b);
                 else atomicOr(&acc_base[a].sum_lo, b);
               }
             }
           } else {
             l_cnt[a]++;
-            if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC) {
+            if (sp.kind == DAGG_SUM_REAL) {
+              l_lo[a] = (unsigned long long)__double_as_longlong(
+                  __longlong_as_double((long long)l_lo[a]) +
+                  __longlong_as_double(v));
+            } else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC) {
               unsigned long long old = l_lo[a];
               unsigned long long nv = old + (unsigned long long)v;
               l_hi[a] += (nv < old ? 1 : 0) + (v < 0 ? -1 : 0);
@@ -1152,7 +1177,7 @@ k_scan_agg(ScanPlan plan,
             } else if (d_is_fold(sp.kind)) {
               unsigned long long b = d_fold_xform(sp.kind, v, sp.col_unsigned);
               if (d_is_xor(sp.kind)) l_lo[a] ^= b;
-              else if (sp.kind == DAGG_MAX_INT || sp.kind == DAGG_MIN_INT)
+              else if (sp.kind == DAGG_MAX_INT || sp.kind == DAGG_MIN_INT || sp.kind == DAGG_MAX_REAL || sp.kind == DAGG_MIN_REAL)
                 l_lo[a] = l_lo[a] > b ? l_lo[a] : b;
               else l_lo[a] |= b;
             }
@@ -1192,7 +1217,7 @@ k_scan_agg(ScanPlan plan,
         if (d_is_fold(kind)) {
           if (la.sum_lo) {
             if (d_is_xor(kind)) atomicXor(&gacc[a].sum_lo, la.sum_lo);
-            else if (kind == DAGG_MAX_INT || kind == DAGG_MIN_INT)
+            else if (kind == DAGG_MAX_INT || kind == DAGG_MIN_INT || kind == DAGG_MAX_REAL || kind == DAGG_MIN_REAL)
               atomicMax(&gacc[a].sum_lo, la.sum_lo);
             else atomicOr(&gacc[a].sum_lo, la.sum_lo);
           }
@@ -1221,7 +1246,7 @@ k_scan_agg(ScanPlan plan,
           unsigned long long plo =
               (unsigned long long)__shfl_down((long long)lo, off, 64);
           if (d_is_xor(kind)) lo ^= plo;
-          else if (kind == DAGG_MAX_INT || kind == DAGG_MIN_INT)
+          else if (kind == DAGG_MAX_INT || kind == DAGG_MIN_INT || kind == DAGG_MAX_REAL || kind == DAGG_MIN_REAL)
             lo = lo > plo ? lo : plo;
           else lo |= plo;
         }
@@ -1229,10 +1254,22 @@ k_scan_agg(ScanPlan plan,
           if (c) atomicAdd(&simple_acc[a].cnt, c);
           if (lo) {
             if (d_is_xor(kind)) atomicXor(&simple_acc[a].sum_lo, lo);
-            else if (kind == DAGG_MAX_INT || kind == DAGG_MIN_INT)
+            else if (kind == DAGG_MAX_INT || kind == DAGG_MIN_INT || kind == DAGG_MAX_REAL || kind == DAGG_MIN_REAL)
               atomicMax(&simple_acc[a].sum_lo, lo);
             else atomicOr(&simple_acc[a].sum_lo, lo);
           }
+        }
+        continue;
+      }
+      if (kind == DAGG_SUM_REAL) {
+        double d = __longlong_as_double((long long)lo);
+        for (int off = 32; off > 0; off >>= 1) {
+          c += (unsigned long long)__shfl_down((long long)c, off, 64);
+          d += __shfl_down(d, off, 64);
+        }
+        if ((threadIdx.x & 63u) == 0) {
+          if (c) atomicAdd(&simple_acc[a].cnt, c);
+          if (d != 0.0) atomicAdd((double *)&simple_acc[a].sum_lo, d);
         }
         continue;
       }
@@ -1575,7 +1612,7 @@ k_scan_agg_pipe(ScanPlan plan,
                 cols[a].iv = cell.ival;
                 cols[a].has_dec = cell.has_dec;
                 cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
-                if (!cell.is_null && !cell.has_int && !cell.has_dec)
+                if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec)
                   parse_ok = false;
               } else parse_ok = false;
             }
@@ -1628,7 +1665,7 @@ k_scan_agg_pipe(ScanPlan plan,
               cols[a].iv = cell.ival;
               cols[a].has_dec = cell.has_dec;
               cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
-              if (!cell.is_null && !cell.has_int && !cell.has_dec) parse_ok = false;
+              if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec) parse_ok = false;
               found++;
             }
           }
@@ -1687,7 +1724,8 @@ k_scan_agg_pipe(ScanPlan plan,
             contribute = false;
           } else if (sp.kind == DAGG_COUNT_COL) {
             contribute = true;
-          } else if (sp.kind == DAGG_SUM_INT || d_is_fold(sp.kind)) {
+          } else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_REAL ||
+                     d_is_fold(sp.kind)) {
             contribute = true; v = cols[a].iv;
           } else {  /* SUM_DEC */
             int d = sp.target_frac - cols[a].dfr;
@@ -1705,19 +1743,26 @@ k_scan_agg_pipe(ScanPlan plan,
           if (IS_HASH) {
             if (acc_base) {
               atomicAdd(&acc_base[a].cnt, 1ull);
-              if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC)
+              if (sp.kind == DAGG_SUM_REAL)
+                atomicAdd((double *)&acc_base[a].sum_lo,
+                          __longlong_as_double(v));
+              else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC)
                 atomic_add_i128(&acc_base[a].sum_lo, &acc_base[a].sum_hi, v);
               else if (d_is_fold(sp.kind)) {
                 unsigned long long b = d_fold_xform(sp.kind, v, sp.col_unsigned);
                 if (d_is_xor(sp.kind)) atomicXor(&acc_base[a].sum_lo, b);
-                else if (sp.kind == DAGG_MAX_INT || sp.kind == DAGG_MIN_INT)
+                else if (sp.kind == DAGG_MAX_INT || sp.kind == DAGG_MIN_INT || sp.kind == DAGG_MAX_REAL || sp.kind == DAGG_MIN_REAL)
                   atomicMax(&acc_base[a].sum_lo, b);
                 else atomicOr(&acc_base[a].sum_lo, b);
               }
             }
           } else {
             l_cnt[a]++;
-            if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC) {
+            if (sp.kind == DAGG_SUM_REAL) {
+              l_lo[a] = (unsigned long long)__double_as_longlong(
+                  __longlong_as_double((long long)l_lo[a]) +
+                  __longlong_as_double(v));
+            } else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC) {
               unsigned long long old = l_lo[a];
               unsigned long long nv = old + (unsigned long long)v;
               l_hi[a] += (nv < old ? 1 : 0) + (v < 0 ? -1 : 0);
@@ -1725,7 +1770,7 @@ k_scan_agg_pipe(ScanPlan plan,
             } else if (d_is_fold(sp.kind)) {
               unsigned long long b = d_fold_xform(sp.kind, v, sp.col_unsigned);
               if (d_is_xor(sp.kind)) l_lo[a] ^= b;
-              else if (sp.kind == DAGG_MAX_INT || sp.kind == DAGG_MIN_INT)
+              else if (sp.kind == DAGG_MAX_INT || sp.kind == DAGG_MIN_INT || sp.kind == DAGG_MAX_REAL || sp.kind == DAGG_MIN_REAL)
                 l_lo[a] = l_lo[a] > b ? l_lo[a] : b;
               else l_lo[a] |= b;
             }
@@ -1751,7 +1796,7 @@ k_scan_agg_pipe(ScanPlan plan,
           unsigned long long plo =
               (unsigned long long)__shfl_down((long long)lo, off, 64);
           if (d_is_xor(kind)) lo ^= plo;
-          else if (kind == DAGG_MAX_INT || kind == DAGG_MIN_INT)
+          else if (kind == DAGG_MAX_INT || kind == DAGG_MIN_INT || kind == DAGG_MAX_REAL || kind == DAGG_MIN_REAL)
             lo = lo > plo ? lo : plo;
           else lo |= plo;
         }
@@ -1759,10 +1804,22 @@ k_scan_agg_pipe(ScanPlan plan,
           if (c) atomicAdd(&simple_acc[a].cnt, c);
           if (lo) {
             if (d_is_xor(kind)) atomicXor(&simple_acc[a].sum_lo, lo);
-            else if (kind == DAGG_MAX_INT || kind == DAGG_MIN_INT)
+            else if (kind == DAGG_MAX_INT || kind == DAGG_MIN_INT || kind == DAGG_MAX_REAL || kind == DAGG_MIN_REAL)
               atomicMax(&simple_acc[a].sum_lo, lo);
             else atomicOr(&simple_acc[a].sum_lo, lo);
           }
+        }
+        continue;
+      }
+      if (kind == DAGG_SUM_REAL) {
+        double d = __longlong_as_double((long long)lo);
+        for (int off = 32; off > 0; off >>= 1) {
+          c += (unsigned long long)__shfl_down((long long)c, off, 64);
+          d += __shfl_down(d, off, 64);
+        }
+        if ((threadIdx.x & 63u) == 0) {
+          if (c) atomicAdd(&simple_acc[a].cnt, c);
+          if (d != 0.0) atomicAdd((double *)&simple_acc[a].sum_lo, d);
         }
         continue;
       }
@@ -3190,7 +3247,7 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
                 cols[a].iv = cell.ival;
                 cols[a].has_dec = cell.has_dec;
                 cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
-                if (!cell.is_null && !cell.has_int && !cell.has_dec)
+                if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec)
                   parse_ok = false;
               } else parse_ok = false;
             }
@@ -3275,7 +3332,7 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
                 cols[a].iv = cell.ival;
                 cols[a].has_dec = cell.has_dec;
                 cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
-                if (!cell.is_null && !cell.has_int && !cell.has_dec)
+                if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec)
                   parse_ok = false;
                 found++;
               }
@@ -3317,7 +3374,8 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
           } else if (!cols[a].found || cols[a].null) {
           } else if (sp.kind == DAGG_COUNT_COL) {
             contribute = 1;
-          } else if (sp.kind == DAGG_SUM_INT || d_is_fold(sp.kind)) {
+          } else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_REAL ||
+                     d_is_fold(sp.kind)) {
             contribute = 1; v = cols[a].iv;
           } else {  /* SUM_DEC */
             int d = sp.target_frac - cols[a].dfr;
@@ -3402,12 +3460,14 @@ __global__ static void k_run_update(const int64_t *ck, const uint8_t *cs,
     if (!ct) continue;
     int64_t v = cav[(uint64_t)a * m + i];
     atomicAdd(&base[a].cnt, 1ull);
-    if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC)
+    if (sp.kind == DAGG_SUM_REAL)
+      atomicAdd((double *)&base[a].sum_lo, __longlong_as_double(v));
+    else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC)
       atomic_add_i128(&base[a].sum_lo, &base[a].sum_hi, v);
     else if (d_is_fold(sp.kind)) {
       unsigned long long b = d_fold_xform(sp.kind, v, sp.col_unsigned);
       if (d_is_xor(sp.kind)) atomicXor(&base[a].sum_lo, b);
-      else if (sp.kind == DAGG_MAX_INT || sp.kind == DAGG_MIN_INT)
+      else if (sp.kind == DAGG_MAX_INT || sp.kind == DAGG_MIN_INT || sp.kind == DAGG_MAX_REAL || sp.kind == DAGG_MIN_REAL)
         atomicMax(&base[a].sum_lo, b);
       else atomicOr(&base[a].sum_lo, b);
     }
@@ -3874,12 +3934,14 @@ __global__ static void k_bytes_update(const uint64_t *co, const uint32_t *cl,
     if (!cas[(uint64_t)a * m + a0]) continue;
     int64_t v = cav[(uint64_t)a * m + a0];
     atomicAdd(&base[a].cnt, 1ull);
-    if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC)
+    if (sp.kind == DAGG_SUM_REAL)
+      atomicAdd((double *)&base[a].sum_lo, __longlong_as_double(v));
+    else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC)
       atomic_add_i128(&base[a].sum_lo, &base[a].sum_hi, v);
     else if (d_is_fold(sp.kind)) {
       unsigned long long b = d_fold_xform(sp.kind, v, sp.col_unsigned);
       if (d_is_xor(sp.kind)) atomicXor(&base[a].sum_lo, b);
-      else if (sp.kind == DAGG_MAX_INT || sp.kind == DAGG_MIN_INT)
+      else if (sp.kind == DAGG_MAX_INT || sp.kind == DAGG_MIN_INT || sp.kind == DAGG_MAX_REAL || sp.kind == DAGG_MIN_REAL)
         atomicMax(&base[a].sum_lo, b);
       else atomicOr(&base[a].sum_lo, b);
     }

@@ -235,6 +235,14 @@ def avg_col(offset, decimal=-1):
     return (F.AGG_AVG, Expr().col(offset), field_type(F.TP_NEWDECIMAL, decimal=decimal))
 
 
+def sum_real(offset):
+    return (F.AGG_SUM, Expr().col(offset), field_type(F.TP_DOUBLE))
+
+
+def avg_real(offset):
+    return (F.AGG_AVG, Expr().col(offset), field_type(F.TP_DOUBLE))
+
+
 def max_col(offset, tp=F.TP_LONGLONG, flag=0):
     return (F.AGG_MAX, Expr().col(offset), field_type(tp, flag))
 
