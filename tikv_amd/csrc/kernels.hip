@@ -3361,6 +3361,7 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
       }
       #pragma unroll
       for (int a = 0; a < NAGGS; a++) {
+        if (a >= plan.n_aggs) break;   /* buffers are sized n_aggs */
         const DevAggSpec &sp = plan.aggs[a];
         uint8_t contribute = 0;
         int64_t v = 0;
@@ -3512,6 +3513,7 @@ static int extract_launch(const ScanPlan &plan, const DevRegion &rgn,
       break
   switch (plan.n_aggs) {
     XCASE(0); XCASE(1); XCASE(2); XCASE(3); XCASE(4);
+    XCASE(5); XCASE(6); XCASE(7);
     default: XCASE(COPR_MAX_AGGS); break;
   }
   #undef XCASE
