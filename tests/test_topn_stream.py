@@ -1024,3 +1024,95 @@ def test_real_aggs_gpu_parity(engine):
         assert abs(gs - os_) <= 1e-9 * max(1.0, abs(os_))
     finally:
         rgn.close()
+
+
+def test_real_aggs_grouped_oracle():
+    """real sums through grouped (stream) aggregation."""
+    orc = _orc()
+    rows = []
+    seq = [(1, 1.5), (1, 2.5), (2, -1.0), (2, None), (2, 4.0)]
+    for i, (g_, x) in enumerate(seq):
+        rows.append((g_, x))
+    keys = b""
+    vbytes = b""
+    vo = [0]
+    for i, (g_, x) in enumerate(rows):
+        keys += row_key(i)
+        v = cell_int(1, g_)
+        v += cell_null(2) if x is None else cell_real(2, x)
+        vbytes += v
+        vo.append(len(vbytes))
+    ko = [19 * i for i in range(len(rows) + 1)]
+    kb = (C.c_uint8 * len(keys)).from_buffer_copy(keys)
+    vb = (C.c_uint8 * len(vbytes)).from_buffer_copy(vbytes)
+    k = C.cast(kb, C.POINTER(C.c_uint8))
+    v = C.cast(vb, C.POINTER(C.c_uint8))
+    cols = [tikv_amd.Col(1), tikv_amd.Col(2, tp=F.TP_DOUBLE)]
+    req = (tikv_amd.DagSelect(cols)
+           .stream_agg([tikv_amd.sum_real(1),
+                        tikv_amd.max_col(1, tp=F.TP_DOUBLE)],
+                       tikv_amd.Expr().col(0)).build())
+    data, nrows = orc.dag_run(req, k, (C.c_uint64 * len(ko))(*ko), v,
+                              (C.c_uint64 * len(vo))(*vo), len(rows))
+    assert nrows == 2
+    s1, p = dec_real(data, 0)
+    m1, p = dec_real(data, p)
+    g1, p = dec_int(data, p)
+    s2, p = dec_real(data, p)
+    m2, p = dec_real(data, p)
+    g2, p = dec_int(data, p)
+    assert (s1, m1, g1) == (4.0, 2.5, 1)
+    assert (s2, m2, g2) == (3.0, 4.0, 2)
+
+
+@pytest.mark.gpu
+def test_real_aggs_grouped_gpu(engine):
+    import random
+    rng = random.Random(11)
+    rows = [(rng.randrange(8), rng.uniform(-100, 100)) for _ in range(20000)]
+    keys = b""
+    vbytes = b""
+    vo = [0]
+    for i, (g_, x) in enumerate(rows):
+        keys += row_key(i)
+        v = cell_int(1, g_) + cell_real(2, x)
+        vbytes += v
+        vo.append(len(vbytes))
+    ko = [19 * i for i in range(len(rows) + 1)]
+    kb = (C.c_uint8 * len(keys)).from_buffer_copy(keys)
+    vb = (C.c_uint8 * len(vbytes)).from_buffer_copy(vbytes)
+    k = C.cast(kb, C.POINTER(C.c_uint8))
+    v = C.cast(vb, C.POINTER(C.c_uint8))
+    koa = (C.c_uint64 * len(ko))(*ko)
+    voa = (C.c_uint64 * len(vo))(*vo)
+    orc = _orc()
+    rgn = engine.region_raw(k, koa, v, voa, len(rows))
+    try:
+        cols = [tikv_amd.Col(1), tikv_amd.Col(2, tp=F.TP_DOUBLE)]
+        req = (tikv_amd.DagSelect(cols)
+               .hash_agg([tikv_amd.count_star(), tikv_amd.sum_real(1),
+                          tikv_amd.max_col(1, tp=F.TP_DOUBLE)],
+                         tikv_amd.Expr().col(0)).build())
+        gd, gr, _ = engine.dag_run(req, [rgn])
+        od, orows = orc.dag_run(req, k, koa, v, voa, len(rows))
+        assert gr == orows == 8
+
+        def rows_of(data):
+            out = {}
+            p = 0
+            for _ in range(8):
+                cnt, p = dec_int(data, p)
+                s, p = dec_real(data, p)
+                m, p = dec_real(data, p)
+                g_, p = dec_int(data, p)
+                out[g_] = (cnt, s, m)
+            return out
+
+        gm, om = rows_of(gd), rows_of(od)
+        assert set(gm) == set(om)
+        for g_ in gm:
+            assert gm[g_][0] == om[g_][0]
+            assert gm[g_][2] == om[g_][2]
+            assert abs(gm[g_][1] - om[g_][1]) <= 1e-9 * max(1.0, abs(om[g_][1]))
+    finally:
+        rgn.close()
