@@ -1869,10 +1869,18 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
             sp.rows_per_tile = rows;
             sp.lds_bytes = (uint32_t)((uint64_t)rows * per_row + 96);
           }
-          sp.lds_agg_slots = 256;
-          sp.lds_agg_off = (sp.lds_bytes + 15u) & ~15u;
-          sp.lds_bytes = sp.lds_agg_off +
-                         sp.lds_agg_slots * (8 + (uint32_t)sp.n_aggs * (uint32_t)sizeof(SimpleAggAcc));
+          bool any_real_sum = false;
+          for (int a = 0; a < sp.n_aggs; a++)
+            if (sp.aggs[a].kind == DAGG_SUM_REAL) any_real_sum = true;
+          if (!any_real_sum) {
+            /* the LDS pre-agg table accumulates sums with integer
+               atomics; f64-bit sums go straight to the global table's
+               double atomics */
+            sp.lds_agg_slots = 256;
+            sp.lds_agg_off = (sp.lds_bytes + 15u) & ~15u;
+            sp.lds_bytes = sp.lds_agg_off +
+                           sp.lds_agg_slots * (8 + (uint32_t)sp.n_aggs * (uint32_t)sizeof(SimpleAggAcc));
+          }
           int le = dev_scan_launch(sp, regions[rg]->dev, nullptr, &ht, nullptr, eng->stream);
           if (le) { free_ht(); return SET_ERR(COPR_ERR_INTERNAL, "scan launch failed"); }
         }
