@@ -202,3 +202,75 @@ def test_mvcc_device_filter_parity(engine):
             rgn.close()
     finally:
         g.close()
+
+
+def _py_visible(entries_by_key, read_ts):
+    """Independent python model of forward.rs:440-515 + write.rs:425-442:
+    newest version with commit_ts <= read_ts decides; Put emits unless its
+    gc_fence is in (0, read_ts]; Delete hides; Lock/Rollback defer to the
+    next older version, except LastChange::NotExist which hides."""
+    out = {}
+    for uk, versions in entries_by_key.items():
+        for commit_ts, (tp, short, fence, last_change) in sorted(
+                versions, reverse=True):
+            if commit_ts > read_ts:
+                continue
+            if tp == "P":
+                if fence is not None and 0 < fence <= read_ts:
+                    break
+                out[uk] = short
+                break
+            if tp == "D":
+                break
+            # Lock / Rollback
+            if last_change is not None and last_change[0] == 0:
+                break  # LastChange::NotExist
+            continue
+        # fallthrough: no visible version
+    return out
+
+
+def test_mvcc_random_chains_vs_oracle():
+    """random multi-version chains: oracle == independent python model."""
+    import random
+    rng = random.Random(42)
+    for trial in range(20):
+        entries_by_key = {}
+        flat = []
+        n_keys = rng.randrange(1, 12)
+        for ki in range(n_keys):
+            uk = b"t" + bytes(7) + bytes([1]) + b"_r" + ki.to_bytes(8, "big")
+            n_vers = rng.randrange(1, 5)
+            used_ts = rng.sample(range(10, 3000), n_vers)
+            versions = []
+            for ts in used_ts:
+                tp = rng.choice(["P", "P", "P", "D", "L", "R"])
+                short = (b"v%d" % ts) if tp in ("P", "R") else None
+                if tp == "R":
+                    short = b"p"   # protected rollback payload
+                fence = None
+                if tp == "P" and rng.random() < 0.3:
+                    fence = rng.choice([0, 400, 5000])
+                    if fence == 0:
+                        fence = None
+                last_change = None
+                if tp in ("L", "R") and rng.random() < 0.3:
+                    last_change = (0, 1) if rng.random() < 0.5 else (5, 2)
+                versions.append((ts, (tp, short if tp == "P" else None,
+                                      fence, last_change)))
+                flat.append((uk, ts, tp, short, fence, last_change))
+            entries_by_key[uk] = versions
+        # write-CF stream: key asc, commit_ts desc
+        flat.sort(key=lambda e: (e[0], -e[1]))
+        ents = []
+        for uk, ts, tp, short, fence, last_change in flat:
+            ents.append((wkey(uk, ts),
+                         wval(tp, ts - 1, short_value=short, gc_fence=fence,
+                              last_change=last_change)))
+        read_ts = rng.choice([5, 500, 1500, 3500])
+        keys, ko, vals, vo, n = run_filter(ents, read_ts)
+        got = {}
+        for i in range(n):
+            got[keys[ko[i]:ko[i + 1]]] = vals[vo[i]:vo[i + 1]]
+        want = _py_visible(entries_by_key, read_ts)
+        assert got == want, (trial, read_ts)
