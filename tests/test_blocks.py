@@ -227,3 +227,52 @@ def test_block_compressed_device(engine):
             rgn.close()
     finally:
         g.close()
+
+
+def test_block_oracle_malformed():
+    """truncated / corrupt blocks are rejected, never misparsed."""
+    orc = _orc()
+    entries = [(b"tabc_r0001", b"v1"), (b"tabc_r0002", b"v2")]
+    blk = bytearray(hand_block(entries))
+    cases = [
+        bytes(blk[:-2]),                      # truncated restart count
+        bytes(blk[:3]),                       # truncated entry
+        blk[:-4] + (99).to_bytes(4, "little"),  # absurd restart count
+    ]
+    import pytest as _pytest
+    for bad in cases:
+        bb, oo, n, keep = as_bufs([bytes(bad)])
+        with _pytest.raises(RuntimeError):
+            orc.block_parse(bb, oo, n)
+    # shared > accumulated key length must be rejected
+    evil = bytearray()
+    evil += varint32(5) + varint32(1) + varint32(0) + b"k"   # shared=5, no prior
+    evil += (0).to_bytes(4, "little")
+    evil += (1).to_bytes(4, "little")
+    bb, oo, n, keep = as_bufs([bytes(evil)])
+    with _pytest.raises(RuntimeError):
+        orc.block_parse(bb, oo, n)
+
+
+def test_decompress_malformed():
+    """corrupt compressed payloads error cleanly."""
+    g = tikv_amd.GenRegion(config_index=1, n_rows=500, table_id=7)
+    try:
+        blocks, offs, cb, co, types, n, keep = _comp_bufs(g, 4)
+        lib = g._lib
+        # flip a byte in the first compressed block's payload
+        raw = bytearray(C.string_at(cb, co[n]))
+        raw[co[0] + 3] ^= 0xFF
+        rb = (C.c_uint8 * len(raw)).from_buffer_copy(bytes(raw))
+        db = C.POINTER(C.c_uint8)()
+        do = C.POINTER(C.c_uint64)()
+        st = lib.copr_blocks_decompress(C.cast(rb, C.POINTER(C.c_uint8)), co,
+                                        types, n, C.byref(db), C.byref(do))
+        # either an explicit error or (for lz4's tolerant cases) output that
+        # differs from the original -- it must never silently equal it
+        if st == 0:
+            assert C.string_at(db, offs[n]) != C.string_at(blocks, offs[n])
+        else:
+            assert st != 0
+    finally:
+        g.close()
