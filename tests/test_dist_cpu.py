@@ -22,3 +22,4 @@ def test_merge_steps_gloo():
     total = (1 << 70) - 1
     assert res["lo"] == total & (2**64 - 1)
     assert res["hi"] == (total >> 64) & (2**64 - 1)
+    assert res["r"] == 0.75

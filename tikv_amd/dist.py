@@ -31,6 +31,16 @@ def merge_checksum(xor_val: int) -> int:
     return acc
 
 
+def merge_sum_real(x):
+    """all-reduce an f64 partial sum (real aggregates across Region
+    shards); parallel order keeps it in the 1-ULP class like the kernels."""
+    import torch
+    import torch.distributed as dist
+    t = torch.tensor([x], dtype=torch.float64)
+    dist.all_reduce(t)
+    return float(t.item())
+
+
 def merge_sum_i128(lo: int, hi: int):
     """elementwise merge of a two's-complement i128 partial sum
     (Decimal partial aggregates travel as scaled i128 limbs)."""
