@@ -2190,7 +2190,7 @@ extern "C" copr_status copr_checksum(copr_engine *eng, copr_region *const *regio
   if (!eng) return SET_ERR(COPR_ERR_INVALID_REQUEST, "null engine");
   HIP_TRY(hipSetDevice(eng->device), "hipSetDevice");
   if (!eng->d_crc_tables) {
-    uint64_t tab[8 * 256];
+    uint64_t tab[16 * 256];
     build_crc_tables(tab);
     HIP_TRY(hipMalloc(&eng->d_crc_tables, sizeof(tab)), "crc tab alloc");
     HIP_TRY(hipMemcpy(eng->d_crc_tables, tab, sizeof(tab), hipMemcpyHostToDevice),

@@ -2640,6 +2640,7 @@ k_scan_project(ScanPlan plan,
 /* ---------------- CRC-64/XZ kernel (checksum.rs:105-114) ----------------
  * slice-by-8 tables staged in LDS (16 KiB); per-KV digest over key||value,
  * XOR-fold per wave then one atomicXor per block. */
+template <int NTAB>   /* 8 = slice-by-8; 16 = slice-by-16 (COPR_CRC16) */
 __global__ void __launch_bounds__(THREADS)
 k_crc64(const uint8_t *__restrict__ vals, const uint64_t *__restrict__ val_offs,
         const uint8_t *__restrict__ keys, const uint64_t *__restrict__ key_offs,
@@ -2677,6 +2678,30 @@ k_crc64(const uint8_t *__restrict__ vals, const uint64_t *__restrict__ val_offs,
          aligned ds_read_b64 pair (a raw misaligned b64 replays 64 cycles) */
       auto crc8 = [&](const uint8_t *p, uint32_t len, uint64_t c) {
         uint32_t i = 0;
+        if (NTAB >= 16) {
+          /* slice-by-16: half the dependent chain length; the 16 lookups
+             per iteration are independent given c */
+          for (; i + 16 <= len; i += 16) {
+            uint64_t a = c ^ lds_win8(p + i);
+            uint64_t b = lds_win8(p + i + 8);
+            c = tab[15 * 256 + (uint32_t)(a & 0xFF)] ^
+                tab[14 * 256 + (uint32_t)((a >> 8) & 0xFF)] ^
+                tab[13 * 256 + (uint32_t)((a >> 16) & 0xFF)] ^
+                tab[12 * 256 + (uint32_t)((a >> 24) & 0xFF)] ^
+                tab[11 * 256 + (uint32_t)((a >> 32) & 0xFF)] ^
+                tab[10 * 256 + (uint32_t)((a >> 40) & 0xFF)] ^
+                tab[9 * 256 + (uint32_t)((a >> 48) & 0xFF)] ^
+                tab[8 * 256 + (uint32_t)(a >> 56)] ^
+                tab[7 * 256 + (uint32_t)(b & 0xFF)] ^
+                tab[6 * 256 + (uint32_t)((b >> 8) & 0xFF)] ^
+                tab[5 * 256 + (uint32_t)((b >> 16) & 0xFF)] ^
+                tab[4 * 256 + (uint32_t)((b >> 24) & 0xFF)] ^
+                tab[3 * 256 + (uint32_t)((b >> 32) & 0xFF)] ^
+                tab[2 * 256 + (uint32_t)((b >> 40) & 0xFF)] ^
+                tab[1 * 256 + (uint32_t)((b >> 48) & 0xFF)] ^
+                tab[0 * 256 + (uint32_t)(b >> 56)];
+          }
+        }
         for (; i + 8 <= len; i += 8) {
           c ^= lds_win8(p + i);
           c = tab[7 * 256 + (uint32_t)(c & 0xFF)] ^
@@ -4637,25 +4662,34 @@ int dev_scan_launch(const ScanPlan &plan, const DevRegion &rgn,
 
 int dev_crc64_launch(const DevRegion &rgn, const uint64_t *d_tables,
                      unsigned long long *d_xor, void *stream) {
+  const bool s16 = getenv("COPR_CRC16") != nullptr;
+  const uint32_t tab_b = (s16 ? 16u : 8u) * 256u * 8u;
   uint32_t rpt = 256;
   uint32_t max_key = 64;
   uint64_t max_tile_val = (uint64_t)rpt * (rgn.max_row_bytes + 16) + 32;
   /* keep tables+keys+values <= ~52 KiB so >=3 blocks/CU stay resident —
      the per-row CRC chain is latency-bound and needs wave parallelism */
-  while (rpt > 32 && 16384 + rpt * max_key + max_tile_val > 52 * 1024) {
+  while (rpt > 32 && tab_b + rpt * max_key + max_tile_val > 52 * 1024) {
     rpt /= 2;
     max_tile_val = (uint64_t)rpt * (rgn.max_row_bytes + 16) + 32;
   }
-  while (rpt > 1 && 16384 + rpt * max_key + max_tile_val > 158 * 1024) {
+  while (rpt > 1 && tab_b + rpt * max_key + max_tile_val > 158 * 1024) {
     rpt /= 2;
     max_tile_val = (uint64_t)rpt * (rgn.max_row_bytes + 16) + 32;
   }
   uint32_t key_lds = rpt * max_key;
-  uint32_t lds_bytes = 16384 + key_lds + (uint32_t)max_tile_val;
+  uint32_t lds_bytes = tab_b + key_lds + (uint32_t)max_tile_val;
   uint64_t n_tiles = (rgn.n_kv + rpt - 1) / rpt;
   uint32_t grid = (uint32_t)(n_tiles < 4096 ? n_tiles : 4096);
   if (grid == 0) grid = 1;
-  hipLaunchKernelGGL(k_crc64, dim3(grid), dim3(THREADS), lds_bytes,
+  if (s16) {
+    hipLaunchKernelGGL((k_crc64<16>), dim3(grid), dim3(THREADS), lds_bytes,
+                       (hipStream_t)stream, rgn.d_vals, rgn.d_val_offs,
+                       rgn.d_keys, rgn.d_key_offs, rgn.n_kv, rpt, key_lds,
+                       d_tables, d_xor);
+    return (int)hipGetLastError();
+  }
+  hipLaunchKernelGGL((k_crc64<8>), dim3(grid), dim3(THREADS), lds_bytes,
                      (hipStream_t)stream,
                      rgn.d_vals, rgn.d_val_offs, rgn.d_keys, rgn.d_key_offs,
                      rgn.n_kv, rpt, key_lds, d_tables, d_xor);
