@@ -2648,11 +2648,12 @@ k_crc64(const uint8_t *__restrict__ vals, const uint64_t *__restrict__ val_offs,
         const uint64_t *__restrict__ g_tables,
         unsigned long long *__restrict__ out_xor) {
   extern __shared__ __attribute__((aligned(16))) uint8_t lds[];
-  uint64_t *tab = (uint64_t *)lds;                 /* 8*256*8 = 16 KiB */
-  uint8_t *key_lds = lds + 8 * 256 * 8;
+  uint64_t *tab = (uint64_t *)lds;                 /* NTAB*256*8 */
+  uint8_t *key_lds = lds + NTAB * 256 * 8;
   uint8_t *val_lds = key_lds + key_lds_bytes;
 
-  for (uint32_t i = threadIdx.x; i < 2048; i += blockDim.x) tab[i] = g_tables[i];
+  for (uint32_t i = threadIdx.x; i < NTAB * 256u; i += blockDim.x)
+    tab[i] = g_tables[i];
 
   const uint32_t rpt = rows_per_tile;
   const uint64_t n_tiles = (n_rows + rpt - 1) / rpt;
