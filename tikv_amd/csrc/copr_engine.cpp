@@ -2167,7 +2167,7 @@ extern "C" void copr_result_free(CoprSelectResult *r) {
 }
 
 /* ---------------- checksum ---------------- */
-static void build_crc_tables(uint64_t tab[8 * 256]) {
+static void build_crc_tables(uint64_t tab[16 * 256]) {
   const uint64_t POLY = 0x42F0E1EBA9EA3693ull;
   uint64_t rpoly = 0;
   for (int i = 0; i < 64; i++)
@@ -2177,7 +2177,8 @@ static void build_crc_tables(uint64_t tab[8 * 256]) {
     for (int j = 0; j < 8; j++) crc = (crc >> 1) ^ ((crc & 1) ? rpoly : 0);
     tab[i] = crc;
   }
-  for (int t = 1; t < 8; t++)
+  /* tables 0..7 serve slice-by-8; 8..15 extend to slice-by-16 */
+  for (int t = 1; t < 16; t++)
     for (int i = 0; i < 256; i++) {
       uint64_t prev = tab[(t - 1) * 256 + i];
       tab[t * 256 + i] = tab[prev & 0xFF] ^ (prev >> 8);
