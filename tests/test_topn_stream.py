@@ -1116,3 +1116,40 @@ def test_real_aggs_grouped_gpu(engine):
             assert abs(gm[g_][1] - om[g_][1]) <= 1e-9 * max(1.0, abs(om[g_][1]))
     finally:
         rgn.close()
+
+
+def test_unique_index_value_handle_oracle():
+    """old-format unique index: key holds only the column datums, the PK
+    int handle is the value's plain BE u64 (index_scan_executor.rs:416-422
+    -- NOT number::decode_i64, no sign flip)."""
+    orc = _orc()
+    # index keys: prefix 't'+tid+'_i'+idx + one INT datum; value = 8B handle
+    keys = b""
+    vals = b""
+    ko = [0]
+    vo = [0]
+    handles = [7, -3, 2**40]
+    for i, h in enumerate(handles):
+        k = (b"t" + (9).to_bytes(8, "big") + b"_i" + (1).to_bytes(8, "big")
+             + b"\x03" + (((100 + i) ^ (1 << 63)) & (2**64 - 1)).to_bytes(8, "big"))
+        keys += k
+        ko.append(len(keys))
+        vals += (h & (2**64 - 1)).to_bytes(8, "big")
+        vo.append(len(vals))
+    kb = (C.c_uint8 * len(keys)).from_buffer_copy(keys)
+    vb = (C.c_uint8 * len(vals)).from_buffer_copy(vals)
+    cols = [tikv_amd.Col(0), tikv_amd.Col(-1, pk_handle=True)]
+    req = (tikv_amd.DagSelect(cols, index=True).build())
+    data, nrows = orc.dag_run(req, C.cast(kb, C.POINTER(C.c_uint8)),
+                              (C.c_uint64 * len(ko))(*ko),
+                              C.cast(vb, C.POINTER(C.c_uint8)),
+                              (C.c_uint64 * len(vo))(*vo), len(handles))
+    assert nrows == 3
+    p = 0
+    got = []
+    for i in range(3):
+        cv, p = dec_int(data, p)       # index column datum
+        hv, p = dec_int(data, p)       # handle (decoded -> INT datum)
+        got.append((cv, hv))
+    assert got == [(100, 7), (101, -3), (102, 2**40)]
+    assert p == len(data)
