@@ -1153,3 +1153,39 @@ def test_unique_index_value_handle_oracle():
         got.append((cv, hv))
     assert got == [(100, 7), (101, -3), (102, 2**40)]
     assert p == len(data)
+
+
+def test_chunked_grouped_outputs_oracle():
+    """TypeChunk over grouped results: bytes group datums ride the var-size
+    column container; drains split into 1024-row chunks."""
+    orc = _orc()
+    # 2500 distinct int groups -> chunk lengths 1024, 1024, 452
+    rows = [{1: i, 2: 1} for i in range(2500)]
+    k, ko, v, vo, n, keep = region_of(rows)
+    cols = [tikv_amd.Col(1), tikv_amd.Col(2)]
+    req = (tikv_amd.DagSelect(cols)
+           .stream_agg([tikv_amd.count_star()], tikv_amd.Expr().col(0))
+           .chunked().build())
+    data, nrows = orc.dag_run(req, k, ko, v, vo, n)
+    assert nrows == 2500
+    lens = []
+    p = 0
+    while p < len(data):
+        # two fixed i64 columns per chunk (count, group), no nulls
+        ln = int.from_bytes(data[p:p + 4], "little")
+        lens.append(ln)
+        p += (8 + 8 * ln) * 2
+    assert lens == [1024, 1024, 452]
+    # bytes group keys through the chunk encoder
+    rows2 = [{1: 5, 2: b"aa"}, {1: 6, 2: b"aa"}, {1: 7, 2: None},
+             {1: 8, 2: b"zz"}]
+    k2, ko2, v2, vo2, n2, keep2 = region_of_mixed(rows2)
+    cols2 = [tikv_amd.Col(1), tikv_amd.Col(2, tp=F.TP_VARCHAR)]
+    req2 = (tikv_amd.DagSelect(cols2)
+            .hash_agg([tikv_amd.count_star(), tikv_amd.max_col(0)],
+                      tikv_amd.Expr().col(1)).chunked().build())
+    d2, nr2 = orc.dag_run(req2, k2, ko2, v2, vo2, n2)
+    assert nr2 == 3
+    got = decode_chunks(bytes(d2), ["i64", "i64", "bytes"])
+    assert sorted(got, key=lambda t: (t[2] is None, t[2])) == [
+        (2, 6, b"aa"), (1, 8, b"zz"), (1, 7, None)]
