@@ -187,7 +187,10 @@ typedef struct CoprDagRequest {
      (tidb_query_datatype/src/expr/ctx.rs:65-118) */
   uint64_t flags;                  /* Flag bitset */
   int32_t  div_precision_increment;
-  /* paging (runner.rs:92-126): 0 = disabled */
+  /* paging (runner.rs:92-126): 0 = disabled. Only valid with a single
+     region per request (the reference pages per coprocessor task = per
+     Region); copr_dag_run returns COPR_ERR_UNSUPPORTED for paging_size != 0
+     with n_regions > 1. */
   uint64_t paging_size;
   /* response encoding (tipb EncodeType, runner.rs:1188-1225):
      0 = TypeDefault (datum rows), 1 = TypeChunk (column-oriented chunks:

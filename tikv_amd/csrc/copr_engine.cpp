@@ -1490,6 +1490,10 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
   memset(out, 0, sizeof(*out));
   out->resume_row = UINT64_MAX;
   if (!eng) return SET_ERR(COPR_ERR_INVALID_REQUEST, "null engine");
+  if (req->paging_size && n_regions > 1)
+    return SET_ERR(COPR_ERR_UNSUPPORTED,
+                   "paging_size supports one region per request (the "
+                   "reference pages per coprocessor task = per Region)");
   HIP_TRY(hipSetDevice(eng->device), "hipSetDevice");
   HostPlan pl;
   copr_status st = build_plan(req, &pl);
@@ -1552,16 +1556,21 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
      reported in summaries[0].time_processed_ns (the ExecSummary surface the
      reference fills per executor slot, execute_stats.rs:43-76). bench.py's
      roofline leg reads this. */
-  hipEvent_t ev_a = nullptr, ev_b = nullptr;
-  hipEventCreate(&ev_a);
-  hipEventCreate(&ev_b);
+  struct EventPair {             /* RAII: every error return destroys both */
+    hipEvent_t a = nullptr, b = nullptr;
+    EventPair() { hipEventCreate(&a); hipEventCreate(&b); }
+    ~EventPair() {
+      if (a) hipEventDestroy(a);
+      if (b) hipEventDestroy(b);
+    }
+  } ev;
+  hipEvent_t ev_a = ev.a, ev_b = ev.b;
   float kernel_ms = 0.0f;
   bool timed = false;
 
   if (pl.has_topn) {
     /* ---- TopN: winner selection on device, then the regular project path
        over a gathered sub-region (rows already in output order) ---- */
-    hipEventDestroy(ev_a); hipEventDestroy(ev_b);
     if (n_regions != 1)
       return SET_ERR(COPR_ERR_UNSUPPORTED, "TopN supports one region per request");
     copr_region *r = regions[0];
@@ -2124,8 +2133,6 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
     hipEventSynchronize(ev_b);
     hipEventElapsedTime(&kernel_ms, ev_a, ev_b);
   }
-  hipEventDestroy(ev_a);
-  hipEventDestroy(ev_b);
   if (req->encode_type == 1) {
     if (chunk_rows.empty() && n_rows_out) {
       for (uint64_t left = n_rows_out; left;) {      /* drain: 1024/chunk */

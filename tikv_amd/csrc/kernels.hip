@@ -147,7 +147,7 @@ __device__ static inline uint32_t d_decimal_scaled(const uint8_t *p, uint32_t re
   };
   if (ld) {
     uint32_t w = rd_word(DIG2B[ld]) ^ mask;
-    if (w >= TEN_POW_D[ld + 1]) return 0;
+    if (w >= TEN_POW_D[ld]) return 0;   /* leading group: < 10^ld digits */
     acc = w;
   }
   for (uint32_t k = 0; k < iw; k++) {
@@ -4666,19 +4666,26 @@ int dev_crc64_launch(const DevRegion &rgn, const uint64_t *d_tables,
   const bool s16 = getenv("COPR_CRC16") != nullptr;
   const uint32_t tab_b = (s16 ? 16u : 8u) * 256u * 8u;
   uint32_t rpt = 256;
-  uint32_t max_key = 64;
+  /* key window sized from the region's real max key length, same formula as
+     the value window (stage_tile can round the staged range up by alignment
+     shift + 16B round-up, covered by the +16/+32 slack) */
+  uint64_t max_tile_key = (uint64_t)rpt * (rgn.max_key_bytes + 16) + 32;
   uint64_t max_tile_val = (uint64_t)rpt * (rgn.max_row_bytes + 16) + 32;
   /* keep tables+keys+values <= ~52 KiB so >=3 blocks/CU stay resident —
      the per-row CRC chain is latency-bound and needs wave parallelism */
-  while (rpt > 32 && tab_b + rpt * max_key + max_tile_val > 52 * 1024) {
+  while (rpt > 32 && tab_b + max_tile_key + max_tile_val > 52 * 1024) {
     rpt /= 2;
+    max_tile_key = (uint64_t)rpt * (rgn.max_key_bytes + 16) + 32;
     max_tile_val = (uint64_t)rpt * (rgn.max_row_bytes + 16) + 32;
   }
-  while (rpt > 1 && tab_b + rpt * max_key + max_tile_val > 158 * 1024) {
+  while (rpt > 1 && tab_b + max_tile_key + max_tile_val > 158 * 1024) {
     rpt /= 2;
+    max_tile_key = (uint64_t)rpt * (rgn.max_key_bytes + 16) + 32;
     max_tile_val = (uint64_t)rpt * (rgn.max_row_bytes + 16) + 32;
   }
-  uint32_t key_lds = rpt * max_key;
+  if (tab_b + max_tile_key + max_tile_val > 160 * 1024)
+    return -2;          /* region's rows cannot fit one per tile: loud error */
+  uint32_t key_lds = (uint32_t)max_tile_key;
   uint32_t lds_bytes = tab_b + key_lds + (uint32_t)max_tile_val;
   uint64_t n_tiles = (rgn.n_kv + rpt - 1) / rpt;
   uint32_t grid = (uint32_t)(n_tiles < 4096 ? n_tiles : 4096);

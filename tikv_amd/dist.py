@@ -34,9 +34,7 @@ def merge_checksum(xor_val: int) -> int:
 def merge_sum_real(x):
     """all-reduce an f64 partial sum (real aggregates across Region
     shards); parallel order keeps it in the 1-ULP class like the kernels."""
-    import torch
-    import torch.distributed as dist
-    t = torch.tensor([x], dtype=torch.float64)
+    t = torch.tensor([x], dtype=torch.float64, device=_dev())
     dist.all_reduce(t)
     return float(t.item())
 
