@@ -241,11 +241,15 @@ def bench_one(tikv_amd, F, eng, workload, n_rows, steps, warmup,
                          " 1 thread" % gen.n_kv}
 
     if dist and workload == "cfg2":
-        import torch
-        t = torch.tensor([cnt], dtype=torch.long,
-                         device="cuda" if dist.get_backend() == "nccl" else "cpu")
-        dist.all_reduce(t)
-        cnt = int(t.item())
+        if getattr(eng, "_comm_ranks", 0) == world:
+            # the engine's own RCCL merge over xGMI (copr_merge_count) — the
+            # data-path collective lives in libcopr.so, not torch
+            cnt = eng.merge_count(cnt)
+        else:
+            import torch
+            t = torch.tensor([cnt], dtype=torch.long)
+            dist.all_reduce(t)        # oversubscribed dev boxes: gloo
+            cnt = int(t.item())
 
     # timed region (each step already ends with a hipStreamSynchronize inside
     # copr_dag_run; torch sync covers any torch-side stream)
@@ -265,8 +269,7 @@ def bench_one(tikv_amd, F, eng, workload, n_rows, steps, warmup,
         dist.barrier()
     elapsed = time.perf_counter() - t0
     if dist:
-        t = torch.tensor([elapsed],
-                         device="cuda" if dist.get_backend() == "nccl" else "cpu")
+        t = torch.tensor([elapsed])   # timing plumbing rides the gloo group
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
@@ -347,10 +350,10 @@ def main():
     dist = None
     if world > 1:
         import torch.distributed as tdist
-        # RCCL needs one device per rank; oversubscribed test boxes use gloo
-        n_dev0 = torch.cuda.device_count() if have_cuda else 0
-        backend = "nccl" if have_cuda and world <= n_dev0 else "gloo"
-        tdist.init_process_group(backend=backend)
+        # gloo is BOOTSTRAP + timing plumbing only (comm-id broadcast,
+        # elapsed-max); the data-path collective is the engine's own RCCL
+        # merge (copr_comm.cpp), created below
+        tdist.init_process_group(backend="gloo")
         dist = tdist
 
     n_dev = torch.cuda.device_count() if have_cuda else 1
@@ -358,6 +361,13 @@ def main():
     eng = tikv_amd.Engine(dev)
     if have_cuda:
         torch.cuda.set_device(dev)
+    if dist and have_cuda and world <= n_dev:
+        # one GPU per rank: the engine's RCCL communicator over xGMI
+        obj = [tikv_amd.Engine.comm_id() if rank == 0 else None]
+        dist.broadcast_object_list(obj, src=0)
+        eng.comm_create(obj[0], world, rank)
+        eng._comm_ranks = world
+        log("rank %d: engine RCCL communicator up (world %d)" % (rank, world))
 
     primary = args.workload or "cfg2"
     # parity gate: oracle at full bench size is ~25-40 s of host CPU; the

@@ -85,6 +85,35 @@ copr_status copr_checksum(copr_engine *, copr_region *const *regions,
                           uint32_t n_regions, uint64_t *checksum,
                           uint64_t *total_kvs, uint64_t *total_bytes);
 
+/* ---- multi-GPU partial-aggregate merge (RCCL over xGMI) ----
+ * The path shards by Region across GPUs with no data-path collective
+ * (TiDB's own per-Region coprocessor fan-out, endpoint.rs:238-248); the ONLY
+ * exchange is the final partial-aggregate merge. These calls
+ * wrap an RCCL communicator owned by the engine: one process per GPU, the
+ * 128-byte id produced by rank 0's copr_comm_id and distributed out-of-band
+ * (the Rust shim's gRPC / the bench's rendezvous — plumbing, like NCCL's own
+ * bootstrap). Payloads are KB-scale and latency-bound. All merge calls are
+ * collective: every rank in the communicator must call them in the same
+ * order. */
+#define COPR_COMM_ID_BYTES 128
+copr_status copr_comm_id(uint8_t out[COPR_COMM_ID_BYTES]);
+copr_status copr_comm_create(copr_engine *,
+                             const uint8_t id[COPR_COMM_ID_BYTES],
+                             int n_ranks, int rank);
+void        copr_comm_destroy(copr_engine *);
+/* count(*)/count(col) final merge: sum of per-rank u64 counts */
+copr_status copr_merge_count(copr_engine *, uint64_t *inout);
+/* CRC64 running XOR is order-independent (checksum.rs:78-87); RCCL has no
+ * XOR reduce op, so this allgathers world u64s and folds */
+copr_status copr_merge_checksum(copr_engine *, uint64_t *inout);
+/* exact two's-complement i128 partial-sum merge (Decimal/int sums travel
+ * as scaled i128 limbs): elementwise allgather + host fold
+ * with carries */
+copr_status copr_merge_sum_i128(copr_engine *, uint64_t *lo, uint64_t *hi);
+/* f64 partial sums (Real aggregates): allreduce; parallel order keeps the
+ * 1-ULP class */
+copr_status copr_merge_sum_f64(copr_engine *, double *inout);
+
 /* ---- synthetic region generator (fixture factory) ----
  * Host-side (OpenMP) generator of reference-format regions, mirroring
  * test_coprocessor's fixture store (test_coprocessor/src/store.rs:83-91:

@@ -151,5 +151,21 @@ def load_lib():
     lib.copr_gen_region.restype = C.c_int
     lib.copr_gen_region.argtypes = [C.POINTER(CoprGenSpec), C.POINTER(CoprGenOut)]
     lib.copr_gen_free.argtypes = [C.POINTER(CoprGenOut)]
+    # RCCL merge surface (copr_comm.cpp)
+    lib.copr_comm_id.restype = C.c_int
+    lib.copr_comm_id.argtypes = [C.POINTER(C.c_uint8)]
+    lib.copr_comm_create.restype = C.c_int
+    lib.copr_comm_create.argtypes = [C.c_void_p, C.POINTER(C.c_uint8),
+                                     C.c_int, C.c_int]
+    lib.copr_comm_destroy.argtypes = [C.c_void_p]
+    lib.copr_merge_count.restype = C.c_int
+    lib.copr_merge_count.argtypes = [C.c_void_p, C.POINTER(C.c_uint64)]
+    lib.copr_merge_checksum.restype = C.c_int
+    lib.copr_merge_checksum.argtypes = [C.c_void_p, C.POINTER(C.c_uint64)]
+    lib.copr_merge_sum_i128.restype = C.c_int
+    lib.copr_merge_sum_i128.argtypes = [C.c_void_p, C.POINTER(C.c_uint64),
+                                        C.POINTER(C.c_uint64)]
+    lib.copr_merge_sum_f64.restype = C.c_int
+    lib.copr_merge_sum_f64.argtypes = [C.c_void_p, C.POINTER(C.c_double)]
     _lib = lib
     return lib

@@ -25,6 +25,9 @@ using namespace copr;
 
 static thread_local std::string g_err;
 extern "C" const char *copr_last_error(void) { return g_err.c_str(); }
+namespace copr {
+std::string &tls_err() { return g_err; }   /* shared with copr_comm.cpp */
+}
 
 #define SET_ERR(st, msg) (g_err = (msg), (st))
 #define HIP_TRY(expr, what)                                                  \
@@ -37,23 +40,8 @@ extern "C" const char *copr_last_error(void) { return g_err.c_str(); }
     }                                                                        \
   } while (0)
 
-struct copr_engine {
-  int device = 0;
-  hipStream_t stream = nullptr;
-  uint64_t *d_crc_tables = nullptr;   /* 8*256 u64, built lazily */
-  /* internal channel for the TopN sub-region project: column OFFSET the
-     order expression decoded in place (response encodes it decoded,
-     lazy_column.rs:165,242); -1 = none. dec2 = the original request's
-     filter column when distinct from the order column. */
-  int dec_col_off = -1;
-  int dec2_col_off = -1;
-};
-
-struct copr_region {
-  copr_engine *eng = nullptr;
-  DevRegion dev;
-  std::vector<uint64_t> h_key_offs, h_val_offs;   /* host copies for encode */
-};
+/* struct copr_engine / copr_region now live in copr_internal.h (shared with
+   copr_comm.cpp, the RCCL merge module). */
 
 /* ---------------- engine ---------------- */
 extern "C" copr_status copr_engine_create(int hip_device, copr_engine **out) {
@@ -77,6 +65,7 @@ extern "C" copr_status copr_engine_create(int hip_device, copr_engine **out) {
 
 extern "C" void copr_engine_destroy(copr_engine *eng) {
   if (!eng) return;
+  copr::comm_free(eng);
   if (eng->d_crc_tables) hipFree(eng->d_crc_tables);
   if (eng->stream) hipStreamDestroy(eng->stream);
   delete eng;

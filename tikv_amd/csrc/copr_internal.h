@@ -4,11 +4,16 @@
 
 #include <stdint.h>
 #include <stddef.h>
+#include <string>
 #include <vector>
 
 #include "../../include/copr_types.h"   /* scalar sig / tp enums */
 
 namespace copr {
+
+/* thread-local error string (defined in copr_engine.cpp; feeds
+   copr_last_error()) */
+std::string &tls_err();
 
 /* device-resident region */
 struct DevRegion {
@@ -258,4 +263,30 @@ int dev_mvcc_build(const uint8_t *d_keys, const uint64_t *d_ko,
                    void *stream);
 
 }  // namespace copr
+
+/* ---- engine / region handles (C-ABI opaque types) ---- */
+struct copr_engine {
+  int device = 0;
+  hipStream_t stream = nullptr;
+  uint64_t *d_crc_tables = nullptr;   /* 8*256 u64, built lazily */
+  /* internal channel for the TopN sub-region project: column OFFSET the
+     order expression decoded in place (response encodes it decoded,
+     lazy_column.rs:165,242); -1 = none. dec2 = the original request's
+     filter column when distinct from the order column. */
+  int dec_col_off = -1;
+  int dec2_col_off = -1;
+  /* RCCL communicator state (copr_comm.cpp); null until copr_comm_create */
+  void *comm_state = nullptr;
+};
+
+struct copr_region {
+  copr_engine *eng = nullptr;
+  copr::DevRegion dev;
+  std::vector<uint64_t> h_key_offs, h_val_offs;   /* host copies for encode */
+};
+
+namespace copr {
+/* frees the engine's RCCL state (no-op when absent); copr_comm.cpp */
+void comm_free(copr_engine *);
+}
 #endif

@@ -420,6 +420,63 @@ class Engine:
             return data, n_rows, kernel_ns, resume
         return data, n_rows, kernel_ns
 
+    # ---- RCCL merge surface (the engine's only collective).
+    # The 128-byte id comes from rank 0's comm_id() and is distributed
+    # out-of-band (the bench uses a gloo broadcast as the bootstrap channel,
+    # mirroring NCCL's own bootstrap-over-sockets).
+    @staticmethod
+    def comm_id():
+        lib = F.load_lib()
+        buf = (C.c_uint8 * 128)()
+        st = lib.copr_comm_id(buf)
+        if st != 0:
+            raise RuntimeError("copr_comm_id: %d (%s)" %
+                               (st, lib.copr_last_error().decode()))
+        return bytes(buf)
+
+    def comm_create(self, comm_id: bytes, n_ranks: int, rank: int):
+        buf = (C.c_uint8 * 128).from_buffer_copy(comm_id)
+        st = self._lib.copr_comm_create(self._h, buf, n_ranks, rank)
+        if st != 0:
+            raise RuntimeError("copr_comm_create: %d (%s)" %
+                               (st, self._lib.copr_last_error().decode()))
+
+    def comm_destroy(self):
+        self._lib.copr_comm_destroy(self._h)
+
+    def merge_count(self, count: int) -> int:
+        v = C.c_uint64(count)
+        st = self._lib.copr_merge_count(self._h, C.byref(v))
+        if st != 0:
+            raise RuntimeError("copr_merge_count: %d (%s)" %
+                               (st, self._lib.copr_last_error().decode()))
+        return v.value
+
+    def merge_checksum(self, xor_val: int) -> int:
+        v = C.c_uint64(xor_val)
+        st = self._lib.copr_merge_checksum(self._h, C.byref(v))
+        if st != 0:
+            raise RuntimeError("copr_merge_checksum: %d (%s)" %
+                               (st, self._lib.copr_last_error().decode()))
+        return v.value
+
+    def merge_sum_i128(self, lo: int, hi: int):
+        lov = C.c_uint64(lo)
+        hiv = C.c_uint64(hi)
+        st = self._lib.copr_merge_sum_i128(self._h, C.byref(lov), C.byref(hiv))
+        if st != 0:
+            raise RuntimeError("copr_merge_sum_i128: %d (%s)" %
+                               (st, self._lib.copr_last_error().decode()))
+        return lov.value, hiv.value
+
+    def merge_sum_f64(self, x: float) -> float:
+        v = C.c_double(x)
+        st = self._lib.copr_merge_sum_f64(self._h, C.byref(v))
+        if st != 0:
+            raise RuntimeError("copr_merge_sum_f64: %d (%s)" %
+                               (st, self._lib.copr_last_error().decode()))
+        return v.value
+
     def checksum(self, regions):
         arr = (C.c_void_p * len(regions))(*[r._h for r in regions])
         cs = C.c_uint64()
