@@ -606,6 +606,9 @@ static void wire_celldir(ScanPlan *sp, const DevRegion &dev) {
   sp->dir_plane = nullptr;
   sp->celldir = nullptr;
   sp->celldir_n = 0;
+  sp->n_dir_slabs = 0;
+  sp->dirslab_f = sp->dirslab_f2 = sp->dirslab_g = -1;
+  for (int a = 0; a < COPR_MAX_AGGS; a++) sp->dirslab_a[a] = -1;
   if (sp->index_mode || !dev.d_celldir) return;
   if (sp->has_filter && sp->filter_col_id >= 1 && sp->filter_col_id <= 16)
     sp->dir_plane = dev.d_celldir +
@@ -624,6 +627,26 @@ static void wire_celldir(ScanPlan *sp, const DevRegion &dev) {
         (sp->aggs[a].col_id < 1 || sp->aggs[a].col_id > 16))
       ok = false;
   if (ok) { sp->celldir = dev.d_celldir; sp->celldir_n = dev.n_kv; }
+
+  /* assign staged dir slabs (pipe kernel): one per DISTINCT needed plane,
+     slab 0 = the filter plane (FASTFC contract) */
+  if (!ok && !sp->dir_plane) return;
+  auto slab_of = [&](int64_t col) -> int32_t {
+    if (col < 1 || col > 16) return -1;
+    const uint8_t *p = dev.d_celldir + (uint64_t)(col - 1) * dev.n_kv;
+    for (int32_t k = 0; k < sp->n_dir_slabs; k++)
+      if (sp->dir_planes_staged[k] == p) return k;
+    if (sp->n_dir_slabs == COPR_MAX_DIR_SLABS) return -1;
+    sp->dir_planes_staged[sp->n_dir_slabs] = p;
+    return sp->n_dir_slabs++;
+  };
+  if (sp->has_filter) sp->dirslab_f = slab_of(sp->filter_col_id);
+  if (sp->filter2_on) sp->dirslab_f2 = slab_of(sp->filter2_col_id);
+  if (sp->mode == 2 || sp->group_col_id != 0)
+    sp->dirslab_g = slab_of(sp->group_col_id);
+  for (int a = 0; a < sp->n_aggs; a++)
+    if (sp->aggs[a].kind != DAGG_COUNT_ROWS)
+      sp->dirslab_a[a] = slab_of(sp->aggs[a].col_id);
 }
 
 
@@ -1049,7 +1072,8 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
 /* tiling: pick rows_per_tile from the region's max row size so the staged
  * tile fits the LDS budget (Guideline: 256-thread blocks; <=64 KiB tile
  * keeps >=2 blocks/CU of occupancy on the 160 KiB LDS). */
-static void pick_tiling(const DevRegion &rgn, ScanPlan *sp, bool force_nopipe = false) {
+static void pick_tiling(const DevRegion &rgn, ScanPlan *sp,
+                        bool force_nopipe = false, uint32_t extra_lds = 0) {
   sp->diag_stage_only = getenv("COPR_DIAG_STAGE_ONLY") ? 1 : 0;
   sp->use_pipe = 0;
   sp->glds_nt = getenv("COPR_GLDS_NT") ? 1 : 0;
@@ -1057,9 +1081,14 @@ static void pick_tiling(const DevRegion &rgn, ScanPlan *sp, bool force_nopipe = 
 
   if (!force_nopipe && !getenv("COPR_NO_PIPE")) {
     /* glds double-buffer pipeline: 2 x (offs slab + vals slab [+ dir
-       slab]), 1 KiB granular. >= 2 blocks/CU needs <= ~78 KiB. */
-    uint32_t ds = (sp->dir_plane && !getenv("COPR_NO_DIR_SLAB")) ? 1024u : 0u;
+       slabs]), 1 KiB granular. >= 2 blocks/CU needs <= ~78 KiB.
+       extra_lds (e.g. the hash pre-agg table) eats into the budget. */
+    uint32_t nslabs = getenv("COPR_NO_DIR_SLAB") ? 0u
+                      : (sp->n_dir_slabs ? (uint32_t)sp->n_dir_slabs
+                                         : (sp->dir_plane ? 1u : 0u));
+    uint32_t ds = nslabs * 1024u;
     uint32_t budget = 76 * 1024 + 2 * ds;
+    budget = budget > extra_lds ? budget - extra_lds : 0;
     if (const char *e = getenv("COPR_PIPE_LDS_BUDGET")) budget = (uint32_t)atoi(e);
     uint32_t rows = 1024;
     if (const char *e = getenv("COPR_ROWS_PER_TILE")) rows = (uint32_t)atoi(e);
@@ -1856,10 +1885,21 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
           ScanPlan sp = pl.sp;
           sp.table_size = tsize;
           wire_celldir(&sp, regions[rg]->dev);
-          /* hash mode runs the single-buffer kernel with a per-block LDS
-             pre-aggregation table appended after the tile region */
-          pick_tiling(regions[rg]->dev, &sp, /*force_nopipe=*/true);
-          if (sp.lds_bytes > 70 * 1024) {
+          bool any_real_sum = false;
+          for (int a = 0; a < sp.n_aggs; a++)
+            if (sp.aggs[a].kind == DAGG_SUM_REAL) any_real_sum = true;
+          /* the LDS pre-agg table accumulates sums with integer atomics;
+             f64-bit sums go straight to the global table's double atomics */
+          uint32_t slots = any_real_sum ? 0u : 256u;
+          uint32_t table_b =
+              slots ? (16u + slots * (8 + (uint32_t)sp.n_aggs *
+                                              (uint32_t)sizeof(SimpleAggAcc)))
+                    : 0u;
+          /* hash mode prefers the glds-pipelined kernel (DMA-staged dir
+             planes + overlap; r01's single-buffer kernel measured 47%
+             wave-parked); the table rides after the two buffers */
+          pick_tiling(regions[rg]->dev, &sp, /*force_nopipe=*/false, table_b);
+          if (!sp.use_pipe && sp.lds_bytes + table_b > 70 * 1024) {
             /* shrink the tile so table + tile keep >=2 blocks/CU */
             uint32_t per_row = regions[rg]->dev.max_row_bytes + 1;
             uint32_t rows = sp.rows_per_tile;
@@ -1867,17 +1907,11 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
             sp.rows_per_tile = rows;
             sp.lds_bytes = (uint32_t)((uint64_t)rows * per_row + 96);
           }
-          bool any_real_sum = false;
-          for (int a = 0; a < sp.n_aggs; a++)
-            if (sp.aggs[a].kind == DAGG_SUM_REAL) any_real_sum = true;
-          if (!any_real_sum) {
-            /* the LDS pre-agg table accumulates sums with integer
-               atomics; f64-bit sums go straight to the global table's
-               double atomics */
-            sp.lds_agg_slots = 256;
+          if (slots) {
+            sp.lds_agg_slots = slots;
             sp.lds_agg_off = (sp.lds_bytes + 15u) & ~15u;
             sp.lds_bytes = sp.lds_agg_off +
-                           sp.lds_agg_slots * (8 + (uint32_t)sp.n_aggs * (uint32_t)sizeof(SimpleAggAcc));
+                           slots * (8 + (uint32_t)sp.n_aggs * (uint32_t)sizeof(SimpleAggAcc));
           }
           int le = dev_scan_launch(sp, regions[rg]->dev, nullptr, &ht, nullptr, eng->stream);
           if (le) { free_ht(); return SET_ERR(COPR_ERR_INTERNAL, "scan launch failed"); }

@@ -180,11 +180,18 @@ struct ScanPlan {
   int32_t glds_nt;               /* nt (aux=2) on the values stream */
   uint32_t offs_slab;
   uint32_t vals_slab;
-  /* 1 KiB per buffer staging the tile's slice of the filter column's
-     directory plane (dir_plane != null): the per-row dir byte is the head
-     of the parse dependency chain, so it rides the DMA instead of being a
-     per-row global load */
+  /* dir_slab = TOTAL bytes per buffer staging tile slices of needed
+     directory planes (n_dir_slabs x 1 KiB): the per-row dir bytes head the
+     parse dependency chain, so they ride the DMA instead of being per-row
+     random global loads. Slab 0 is the filter plane (FASTFC reads it).
+     dirslab_* = slab index per consumer, -1 = fall back to a global
+     celldir load. */
   uint32_t dir_slab;
+#define COPR_MAX_DIR_SLABS 6
+  int32_t n_dir_slabs;
+  const uint8_t *dir_planes_staged[COPR_MAX_DIR_SLABS];
+  int32_t dirslab_f, dirslab_f2, dirslab_g;
+  int32_t dirslab_a[COPR_MAX_AGGS];
   /* hash agg: per-block LDS pre-aggregation table (0 = disabled).
      Low-cardinality GROUP BY otherwise serializes on a handful of global
      atomic addresses. */

@@ -1334,6 +1334,24 @@ k_scan_agg_pipe(ScanPlan plan,
 
   bool any_parse_err = false;
 
+  /* per-block LDS pre-aggregation table (hash mode, after the two DMA
+     buffers): cuts global atomic contention ~rows/slots-fold for
+     low-cardinality GROUP BY (same structure as k_scan_agg's) */
+  long long *lkeys = nullptr;
+  SimpleAggAcc *laccs = nullptr;
+  const uint32_t LSLOTS = IS_HASH ? plan.lds_agg_slots : 0;
+  if (IS_HASH && LSLOTS) {
+    lkeys = (long long *)(lds + plan.lds_agg_off);
+    laccs = (SimpleAggAcc *)(lkeys + LSLOTS);
+    for (uint32_t s = threadIdx.x; s < LSLOTS; s += blockDim.x) {
+      lkeys[s] = (long long)0x8000000000000000ll;
+      #pragma unroll
+      for (int a = 0; a < NAGGS; a++)
+        laccs[s * NAGGS + a] = SimpleAggAcc{0, 0, 0};
+    }
+    __syncthreads();
+  }
+
   /* issue the HBM->LDS DMA for a tile whose bounds (gb, ge) are ALREADY in
      registers — no loads between glds issues, so no compiler vmcnt lands in
      the middle of the queue */
@@ -1370,13 +1388,26 @@ k_scan_agg_pipe(ScanPlan plan,
                                          (uint32_t *)(bv + off), 16, 0, 0);
       }
     }
-    if (DS && wave == nwaves - 1) {
-      /* tile's directory slice: rpt <= 1024 bytes = one 1 KiB chunk
-         (row0 is rpt-aligned and the plane has tail slack) */
-      const uint8_t *dsrc = plan.dir_plane + row0;
-      __builtin_amdgcn_global_load_lds((const uint32_t *)(dsrc + lane * 16u),
-                                       (uint32_t *)(b + DOFF + lane * 16u),
-                                       16, 0, 0);
+    /* tile slices of the needed directory planes: rpt <= 1024 bytes each =
+       one 1 KiB chunk per slab (row0 is rpt-aligned, planes have tail
+       slack). Slab k staged by wave (nwaves-1-k)%nwaves so slab 0 keeps
+       its historical wave. */
+    if (DS) {
+      if (plan.n_dir_slabs > 0) {
+        for (int32_t k2 = 0; k2 < plan.n_dir_slabs; k2++) {
+          if (wave != (nwaves - 1u - ((uint32_t)k2 % nwaves))) continue;
+          const uint8_t *dsrc = plan.dir_planes_staged[k2] + row0;
+          __builtin_amdgcn_global_load_lds(
+              (const uint32_t *)(dsrc + lane * 16u),
+              (uint32_t *)(b + DOFF + (uint32_t)k2 * 1024u + lane * 16u),
+              16, 0, 0);
+        }
+      } else if (wave == nwaves - 1) {
+        const uint8_t *dsrc = plan.dir_plane + row0;
+        __builtin_amdgcn_global_load_lds((const uint32_t *)(dsrc + lane * 16u),
+                                         (uint32_t *)(b + DOFF + lane * 16u),
+                                         16, 0, 0);
+      }
     }
   };
   auto bounds_of = [&](uint64_t tile, uint64_t *gb, uint64_t *ge) {
@@ -1550,19 +1581,26 @@ k_scan_agg_pipe(ScanPlan plan,
         if (plan.celldir && vp[0] != 128) {
           const uint8_t *db = plan.celldir;
           const uint64_t dn = plan.celldir_n;
+          /* dir bytes come from the DMA-staged tile slices when assigned
+             (they head the parse dependency chain; a per-row random global
+             load here parks the wave ~L2/HBM latency per row) */
+          const uint8_t *b_dirs = b + DOFF;
+          auto dload = [&](int32_t slab, int64_t cid) -> uint32_t {
+            return (DS && slab >= 0)
+                       ? (uint32_t)b_dirs[(uint32_t)slab * 1024u + r]
+                       : (uint32_t)db[(uint64_t)(cid - 1) * dn + my_row];
+          };
           uint32_t d_f = 0xFFu, d_f2 = 0xFFu, d_g = 0xFFu, d_a[NAGGS];
           bool seq = false;
-          if (plan.has_filter)
-            d_f = db[(uint64_t)(plan.filter_col_id - 1) * dn + my_row];
+          if (plan.has_filter) d_f = dload(plan.dirslab_f, plan.filter_col_id);
           if (plan.filter2_on)
-            d_f2 = db[(uint64_t)(plan.filter2_col_id - 1) * dn + my_row];
-          if (IS_HASH)
-            d_g = db[(uint64_t)(plan.group_col_id - 1) * dn + my_row];
+            d_f2 = dload(plan.dirslab_f2, plan.filter2_col_id);
+          if (IS_HASH) d_g = dload(plan.dirslab_g, plan.group_col_id);
           #pragma unroll
           for (int a = 0; a < NAGGS; a++) {
             d_a[a] = 0xFFu;
             if (plan.aggs[a].kind != DAGG_COUNT_ROWS)
-              d_a[a] = db[(uint64_t)(plan.aggs[a].col_id - 1) * dn + my_row];
+              d_a[a] = dload(plan.dirslab_a[a], plan.aggs[a].col_id);
           }
           seq = (d_f == 0xFEu) | (d_f2 == 0xFEu) | (d_g == 0xFEu);
           #pragma unroll
@@ -1692,24 +1730,43 @@ k_scan_agg_pipe(ScanPlan plan,
           } else {
             uint64_t h = (uint64_t)grp_v * 0x9E3779B97F4A7C15ull;
             h ^= h >> 29;
-            uint32_t mask = plan.table_size - 1u;
-            uint32_t slot = (uint32_t)(h & mask);
             const unsigned long long EMPTY = 0x8000000000000000ull;
-            for (uint32_t probe = 0; ; probe++) {
-              if (probe > mask) { atomicOr(ht.error, 1u); break; }
-              unsigned long long curk =
-                  atomicCAS((unsigned long long *)&ht.keys[slot], EMPTY,
-                            (unsigned long long)grp_v);
-              if (curk == EMPTY) {
-                atomicAdd(ht.n_groups, 1ull);
-                acc_base = ht.accs + (uint64_t)slot * NAGGS;
-                break;
+            bool in_lds = false;
+            if (LSLOTS) {
+              /* per-block LDS table first; fall through to global if full */
+              uint32_t lmask = LSLOTS - 1u;
+              uint32_t slot = (uint32_t)(h & lmask);
+              for (uint32_t probe = 0; probe <= lmask / 2; probe++) {
+                unsigned long long curk =
+                    atomicCAS((unsigned long long *)&lkeys[slot], EMPTY,
+                              (unsigned long long)grp_v);
+                if (curk == EMPTY || curk == (unsigned long long)grp_v) {
+                  acc_base = laccs + (uint64_t)slot * NAGGS;
+                  in_lds = true;
+                  break;
+                }
+                slot = (slot + 1) & lmask;
               }
-              if (curk == (unsigned long long)grp_v) {
-                acc_base = ht.accs + (uint64_t)slot * NAGGS;
-                break;
+            }
+            if (!in_lds) {
+              uint32_t mask = plan.table_size - 1u;
+              uint32_t slot = (uint32_t)(h & mask);
+              for (uint32_t probe = 0; ; probe++) {
+                if (probe > mask) { atomicOr(ht.error, 1u); break; }
+                unsigned long long curk =
+                    atomicCAS((unsigned long long *)&ht.keys[slot], EMPTY,
+                              (unsigned long long)grp_v);
+                if (curk == EMPTY) {
+                  atomicAdd(ht.n_groups, 1ull);
+                  acc_base = ht.accs + (uint64_t)slot * NAGGS;
+                  break;
+                }
+                if (curk == (unsigned long long)grp_v) {
+                  acc_base = ht.accs + (uint64_t)slot * NAGGS;
+                  break;
+                }
+                slot = (slot + 1) & mask;
               }
-              slot = (slot + 1) & mask;
             }
           }
         }
@@ -1781,6 +1838,52 @@ k_scan_agg_pipe(ScanPlan plan,
     }  /* FASTFC else */
     }  /* !diag */
     bsel ^= 1;
+  }
+
+  /* flush the block's LDS pre-agg table into the global table */
+  if (IS_HASH && LSLOTS) {
+    __syncthreads();
+    const unsigned long long EMPTY = 0x8000000000000000ull;
+    uint32_t mask = plan.table_size - 1u;
+    for (uint32_t s = threadIdx.x; s < LSLOTS; s += blockDim.x) {
+      long long key = lkeys[s];
+      if (key == (long long)0x8000000000000000ll) continue;
+      uint64_t h = (uint64_t)key * 0x9E3779B97F4A7C15ull;
+      h ^= h >> 29;
+      uint32_t slot = (uint32_t)(h & mask);
+      SimpleAggAcc *gacc = nullptr;
+      for (uint32_t probe = 0; ; probe++) {
+        if (probe > mask) { atomicOr(ht.error, 1u); break; }
+        unsigned long long curk =
+            atomicCAS((unsigned long long *)&ht.keys[slot], EMPTY,
+                      (unsigned long long)key);
+        if (curk == EMPTY) { atomicAdd(ht.n_groups, 1ull); gacc = ht.accs + (uint64_t)slot * NAGGS; break; }
+        if (curk == (unsigned long long)key) { gacc = ht.accs + (uint64_t)slot * NAGGS; break; }
+        slot = (slot + 1) & mask;
+      }
+      if (!gacc) continue;
+      #pragma unroll
+      for (int a = 0; a < NAGGS; a++) {
+        const SimpleAggAcc &la = laccs[s * NAGGS + a];
+        const int32_t kind = plan.aggs[a].kind;
+        if (la.cnt) atomicAdd(&gacc[a].cnt, la.cnt);
+        if (d_is_fold(kind)) {
+          if (la.sum_lo) {
+            if (d_is_xor(kind)) atomicXor(&gacc[a].sum_lo, la.sum_lo);
+            else if (kind == DAGG_MAX_INT || kind == DAGG_MIN_INT || kind == DAGG_MAX_REAL || kind == DAGG_MIN_REAL)
+              atomicMax(&gacc[a].sum_lo, la.sum_lo);
+            else atomicOr(&gacc[a].sum_lo, la.sum_lo);
+          }
+          continue;
+        }
+        if (la.sum_lo | la.sum_hi) {
+          unsigned long long old = atomicAdd(&gacc[a].sum_lo, la.sum_lo);
+          long long carry = (old + la.sum_lo < old) ? 1 : 0;
+          long long hi_add = (long long)la.sum_hi + carry;
+          if (hi_add) atomicAdd(&gacc[a].sum_hi, (unsigned long long)hi_add);
+        }
+      }
+    }
   }
 
   if (!IS_HASH) {
@@ -2637,9 +2740,103 @@ k_scan_project(ScanPlan plan,
   if (any_parse_err) atomicOr(po.error + 1, 1u);
 }
 
-/* ---------------- CRC-64/XZ kernel (checksum.rs:105-114) ----------------
- * slice-by-8 tables staged in LDS (16 KiB); per-KV digest over key||value,
- * XOR-fold per wave then one atomicXor per block. */
+/* ---------------- CRC-64/XZ kernels (checksum.rs:105-114) ----------------
+ * Per-KV digest over key||value, XOR-fold per wave then one atomicXor per
+ * block (the running XOR is order-independent, checksum.rs:78-87).
+ *
+ * k_crc64_reg (default): one lane per KV, row bytes loaded from GLOBAL
+ * memory into 8xu64 register chunks — no tile staging, no barriers. LDS
+ * holds only the 16 KiB slice-by-8 tables, so 8 blocks/CU keep 8 waves/SIMD
+ * of independent CRC chains in flight. The r01 tiled kernel measured 80%
+ * SQ_WAIT_ANY (wave-parked) at 3 blocks/CU: the per-lane chain of
+ * LDS-window + 8 table reads per 8 bytes has ~50-cycle dependent latency
+ * per step and the stage/parse barrier phases serialized on top
+ * (profiles/r02_cfg4_*). More independent chains per SIMD is the fix, not
+ * more LDS bandwidth. */
+__device__ static inline uint64_t d_crc64_stream(const uint8_t *__restrict__ base,
+                                                 uint64_t b0, uint64_t b1,
+                                                 uint64_t crc,
+                                                 const uint64_t *__restrict__ tab) {
+  uint64_t len = b1 - b0;
+  if (!len) return crc;
+  /* aligned u64 stream with a funnel shift; region buffers carry +2 KiB
+     tail slack so word over-reads past b1 are in-bounds */
+  const uint64_t *q = (const uint64_t *)(base + (b0 & ~7ull));
+  uint32_t sh = (uint32_t)(b0 & 7) * 8u;
+  uint64_t prev = q[0];
+  uint64_t wi = 1;
+  uint64_t n8 = len >> 3;
+  auto step8 = [&](uint64_t cur) {
+    crc ^= cur;
+    crc = tab[7 * 256 + (uint32_t)(crc & 0xFF)] ^
+          tab[6 * 256 + (uint32_t)((crc >> 8) & 0xFF)] ^
+          tab[5 * 256 + (uint32_t)((crc >> 16) & 0xFF)] ^
+          tab[4 * 256 + (uint32_t)((crc >> 24) & 0xFF)] ^
+          tab[3 * 256 + (uint32_t)((crc >> 32) & 0xFF)] ^
+          tab[2 * 256 + (uint32_t)((crc >> 40) & 0xFF)] ^
+          tab[1 * 256 + (uint32_t)((crc >> 48) & 0xFF)] ^
+          tab[0 * 256 + (uint32_t)(crc >> 56)];
+  };
+  while (n8 >= 8) {
+    /* 64-byte register chunk: 8 independent loads issue together, one
+       wait, then 8 dependent table steps consume from registers */
+    uint64_t w[8];
+    #pragma unroll
+    for (int j = 0; j < 8; j++) w[j] = q[wi + j];
+    #pragma unroll
+    for (int j = 0; j < 8; j++) {
+      uint64_t cur = sh ? ((prev >> sh) | (w[j] << (64 - sh))) : prev;
+      prev = w[j];
+      step8(cur);
+    }
+    wi += 8;
+    n8 -= 8;
+  }
+  while (n8) {
+    uint64_t w = q[wi++];
+    uint64_t cur = sh ? ((prev >> sh) | (w << (64 - sh))) : prev;
+    prev = w;
+    step8(cur);
+    n8--;
+  }
+  uint32_t tail = (uint32_t)(len & 7u);
+  if (tail) {
+    uint64_t w = q[wi];
+    uint64_t cur = sh ? ((prev >> sh) | (w << (64 - sh))) : prev;
+    for (uint32_t t = 0; t < tail; t++) {
+      crc = tab[(uint32_t)((crc ^ cur) & 0xFF)] ^ (crc >> 8);
+      cur >>= 8;
+    }
+  }
+  return crc;
+}
+
+__global__ void __launch_bounds__(THREADS, 8)
+k_crc64_reg(const uint8_t *__restrict__ vals,
+            const uint64_t *__restrict__ val_offs,
+            const uint8_t *__restrict__ keys,
+            const uint64_t *__restrict__ key_offs, uint64_t n_rows,
+            const uint64_t *__restrict__ g_tables,
+            unsigned long long *__restrict__ out_xor) {
+  __shared__ uint64_t tab[8 * 256];
+  for (uint32_t i = threadIdx.x; i < 8 * 256u; i += blockDim.x)
+    tab[i] = g_tables[i];
+  __syncthreads();
+  unsigned long long acc = 0;
+  for (uint64_t row = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       row < n_rows; row += (uint64_t)gridDim.x * blockDim.x) {
+    uint64_t crc = ~0ull;
+    crc = d_crc64_stream(keys, key_offs[row], key_offs[row + 1], crc, tab);
+    crc = d_crc64_stream(vals, val_offs[row], val_offs[row + 1], crc, tab);
+    acc ^= ~crc;
+  }
+  for (int off = 32; off > 0; off >>= 1)
+    acc ^= (unsigned long long)__shfl_down((long long)acc, off, 64);
+  if ((threadIdx.x & 63u) == 0 && acc) atomicXor(out_xor, acc);
+}
+
+/* tiled LDS-staged variant (r01; COPR_CRC_TILE / COPR_CRC16 opt-in):
+ * slice-by-8/16 tables + key/value tiles staged in LDS. */
 template <int NTAB>   /* 8 = slice-by-8; 16 = slice-by-16 (COPR_CRC16) */
 __global__ void __launch_bounds__(THREADS)
 k_crc64(const uint8_t *__restrict__ vals, const uint64_t *__restrict__ val_offs,
@@ -4664,6 +4861,16 @@ int dev_scan_launch(const ScanPlan &plan, const DevRegion &rgn,
 int dev_crc64_launch(const DevRegion &rgn, const uint64_t *d_tables,
                      unsigned long long *d_xor, void *stream) {
   const bool s16 = getenv("COPR_CRC16") != nullptr;
+  if (!s16 && !getenv("COPR_CRC_TILE")) {
+    /* default: register-streamed kernel (no staging; 8 blocks/CU) */
+    uint64_t n_blk = (rgn.n_kv + THREADS - 1) / THREADS;
+    uint32_t grid = (uint32_t)(n_blk < 8192 ? n_blk : 8192);
+    if (grid == 0) grid = 1;
+    hipLaunchKernelGGL(k_crc64_reg, dim3(grid), dim3(THREADS), 0,
+                       (hipStream_t)stream, rgn.d_vals, rgn.d_val_offs,
+                       rgn.d_keys, rgn.d_key_offs, rgn.n_kv, d_tables, d_xor);
+    return (int)hipGetLastError();
+  }
   const uint32_t tab_b = (s16 ? 16u : 8u) * 256u * 8u;
   uint32_t rpt = 256;
   /* key window sized from the region's real max key length, same formula as
