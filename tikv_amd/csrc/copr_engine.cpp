@@ -1106,7 +1106,9 @@ static void pick_tiling(const DevRegion &rgn, ScanPlan *sp,
       sp->offs_slab = os;
       sp->vals_slab = vs;
       sp->dir_slab = ds;
-      sp->lds_bytes = 2 * (os + vs + ds);
+      /* +16: the branchless varint window may read past the last staged
+         byte of the second buffer */
+      sp->lds_bytes = 2 * (os + vs + ds) + 16;
       return;
     }
   }

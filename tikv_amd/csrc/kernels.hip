@@ -74,8 +74,8 @@ __device__ static inline uint64_t lds_win8(const uint8_t *p) {
 
 /* ---------------- row-v1 parse from LDS ----------------
  * datum split lengths per flag: datum.rs:1117-1155; varint: number.rs:445-520 */
-__device__ static inline bool d_var_u64(const uint8_t *p, uint32_t rem,
-                                        uint64_t *v, uint32_t *n) {
+__device__ static inline bool d_var_u64_slow(const uint8_t *p, uint32_t rem,
+                                             uint64_t *v, uint32_t *n) {
   uint64_t val = 0;
   uint32_t i = 0;
   int shift = 0;
@@ -91,6 +91,31 @@ __device__ static inline bool d_var_u64(const uint8_t *p, uint32_t rem,
     *v = val; *n = 10; return true;
   }
   return false;
+}
+
+__device__ static inline bool d_var_u64(const uint8_t *p, uint32_t rem,
+                                        uint64_t *v, uint32_t *n) {
+  /* branchless window path for <=8-byte varints (the byte-at-a-time loop
+     costs ~6 divergent instructions per byte; this is ~20 uniform ones).
+     The 8-byte LDS window may read past the varint inside the staged slab
+     (slabs carry >=16 B slack). */
+  if (rem >= 8) {
+    uint64_t m = lds_win8(p);
+    uint64_t stops = ~m & 0x8080808080808080ull;
+    if (stops) {
+      uint32_t nb = ((uint32_t)__ffsll((long long)stops) >> 3) + 1u; /* 1..8 */
+      uint64_t val = (m & 0x7f) | ((m >> 8) & 0x7f) << 7 |
+                     ((m >> 16) & 0x7f) << 14 | ((m >> 24) & 0x7f) << 21 |
+                     ((m >> 32) & 0x7f) << 28 | ((m >> 40) & 0x7f) << 35 |
+                     ((m >> 48) & 0x7f) << 42 | ((m >> 56) & 0x7f) << 49;
+      val &= (1ull << (7u * nb)) - 1ull;      /* 7*8=56 < 64 */
+      *v = val;
+      *n = nb;
+      return true;
+    }
+    /* 9-10-byte varints: rare, take the loop */
+  }
+  return d_var_u64_slow(p, rem, v, n);
 }
 __device__ static inline bool d_var_i64(const uint8_t *p, uint32_t rem,
                                         int64_t *v, uint32_t *n) {
@@ -2777,20 +2802,37 @@ __device__ static inline uint64_t d_crc64_stream(const uint8_t *__restrict__ bas
           tab[1 * 256 + (uint32_t)((crc >> 48) & 0xFF)] ^
           tab[0 * 256 + (uint32_t)(crc >> 56)];
   };
-  while (n8 >= 8) {
-    /* 64-byte register chunk: 8 independent loads issue together, one
-       wait, then 8 dependent table steps consume from registers */
+  if (n8 >= 8) {
+    /* software pipeline over 64-byte register chunks: chunk k+1's 8
+       independent loads are in flight while the dependent table chain
+       consumes chunk k (the unpipelined form measured 54% SQ_WAIT_ANY —
+       one HBM latency exposed per chunk) */
     uint64_t w[8];
     #pragma unroll
     for (int j = 0; j < 8; j++) w[j] = q[wi + j];
+    wi += 8;
+    n8 -= 8;
+    while (n8 >= 8) {
+      uint64_t w2[8];
+      #pragma unroll
+      for (int j = 0; j < 8; j++) w2[j] = q[wi + j];
+      #pragma unroll
+      for (int j = 0; j < 8; j++) {
+        uint64_t cur = sh ? ((prev >> sh) | (w[j] << (64 - sh))) : prev;
+        prev = w[j];
+        step8(cur);
+      }
+      #pragma unroll
+      for (int j = 0; j < 8; j++) w[j] = w2[j];
+      wi += 8;
+      n8 -= 8;
+    }
     #pragma unroll
     for (int j = 0; j < 8; j++) {
       uint64_t cur = sh ? ((prev >> sh) | (w[j] << (64 - sh))) : prev;
       prev = w[j];
       step8(cur);
     }
-    wi += 8;
-    n8 -= 8;
   }
   while (n8) {
     uint64_t w = q[wi++];
