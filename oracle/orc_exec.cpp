@@ -396,31 +396,129 @@ struct ScanState {
   uint64_t next_kv = 0;                         /* scan cursor */
 };
 
-/* IndexScan process_kv_pair (index_scan_executor.rs:373-560, old-collation
- * local path): key = 't'||tid||'_i'||index_id|| comparable datums ||
- * [int-handle datum]; columns are POSITIONAL raw comparable datum slices
- * (extract_columns_from_datum_format :504-517); the int handle comes from
- * the key tail for non-unique indexes (decode_int_handle_from_key :460-481)
- * or the 8-byte BE value for unique ones (:417-422). Values longer than 9
- * bytes (new encoding) are out of scope here. */
+/* Index VALUE layout split (index_scan_executor.rs:322-371,700-885):
+ *   old (len <= 9): [8B BE handle][flag] for unique / '0' or empty for
+ *     non-unique (:336-345);
+ *   new (len > 9): TailLen | [VersionFlag(125) Version] | Options | tail;
+ *     options = [127 len u16le CHandle] (common handle — outside the
+ *     int-handle subset), [126 pid 8B], [128.. restore-data row-v2 to the
+ *     segment end]; handle = first 8 BE bytes of the last tail_len bytes
+ *     when tail_len >= 8 (decode_int_handle_from_value :416-422 applied to
+ *     build_operations' tail :794-835). V4 restore data (version 0) makes
+ *     the restore row the source of ALL index columns (:903-907); V5 is
+ *     skipped for int columns (need_restored_data false, :639-698). */
+static bool split_index_value(const uint8_t *v, size_t n, uint64_t *handle,
+                              bool *has_handle, const uint8_t **restore,
+                              size_t *rlen) {
+  *has_handle = false;
+  *restore = nullptr;
+  *rlen = 0;
+  if (n <= 9) {
+    if (n >= 8) {
+      uint64_t u = 0;
+      for (int b = 0; b < 8; b++) u = (u << 8) | v[b];   /* plain BE u64 */
+      *handle = u;
+      *has_handle = true;
+    }
+    return true;
+  }
+  size_t tail_len = v[0];
+  if (tail_len >= n) return false;
+  int version = 0;
+  size_t opt = 1;
+  if ((tail_len == 0 || tail_len == 1) && v[1] == 125) {
+    version = v[2];
+    opt = 3;
+  }
+  if (n < opt + tail_len) return false;
+  size_t opt_end = n - tail_len;
+  while (opt < opt_end) {
+    uint8_t f = v[opt];
+    if (f == 127) return false;          /* common handle: not int-handle */
+    if (f == 126) {
+      if (opt + 9 > opt_end) return false;
+      opt += 9;
+      continue;
+    }
+    if (f == 128) {
+      if (version == 0) {
+        *restore = v + opt;
+        *rlen = opt_end - opt;
+      }
+      opt = opt_end;
+      break;
+    }
+    return false;
+  }
+  if (tail_len >= 8) {
+    uint64_t u = 0;
+    for (int b = 0; b < 8; b++) u = (u << 8) | v[n - tail_len + b];
+    *handle = u;
+    *has_handle = true;
+  }
+  return true;
+}
+
+/* IndexScan process_kv_pair (index_scan_executor.rs:373-560): key =
+ * 't'||tid||'_i'||index_id|| comparable datums || [int-handle datum];
+ * columns are POSITIONAL raw comparable datum slices
+ * (extract_columns_from_datum_format :504-517) unless a V4 restore row
+ * overrides them; the int handle comes from the key tail for non-unique
+ * indexes (decode_int_handle_from_key :460-481) or the value for unique
+ * ones (old 8-byte form :417-422, new TailLen form :833-835). */
 static bool scan_process_kv_index(ScanState &st, const uint8_t *key, size_t klen,
                                   const uint8_t *val, size_t vlen, Batch *batch) {
   if (klen < 19 || key[0] != 't' || key[9] != '_' || key[10] != 'i')
     FAIL("not an index key");
-  if (vlen > 9) FAIL("new index value encoding not supported");
+  uint64_t vhandle = 0;
+  bool v_has_handle = false;
+  const uint8_t *restore = nullptr;
+  size_t rlen = 0;
+  if (!split_index_value(val, vlen, &vhandle, &v_has_handle, &restore, &rlen))
+    FAIL("bad index value layout");
   const uint8_t *p = key + 19;
   size_t rem = klen - 19;
   size_t ncols = st.cols.size();
   size_t n_idx_cols = 0;
   for (auto &c : st.cols) if (!c.pk_handle) n_idx_cols++;
-  size_t ci = 0;
-  for (size_t i = 0; i < ncols && ci < n_idx_cols; i++) {
-    if (st.cols[i].pk_handle) continue;
-    size_t dlen;
-    if (!split_datum(p, rem, &dlen)) FAIL("bad index key datum");
-    batch->cols[i].raw_vals.emplace_back((const char *)p, dlen);
-    p += dlen; rem -= dlen;
-    ci++;
+
+  if (restore) {
+    /* V4 restore data: every index column from the restore row-v2, keyed
+       by the reference column id (:903-907, :483-501); key datums are
+       skipped positionally to reach a key-form handle */
+    RowSliceV2 rs;
+    if (!row_v2_parse(restore, rlen, &rs)) FAIL("bad restore-data row");
+    for (size_t i = 0, ci = 0; i < ncols && ci < n_idx_cols; i++) {
+      if (st.cols[i].pk_handle) continue;
+      uint32_t s, e;
+      if (row_v2_find(rs, st.cols[i].column_id, &s, &e)) {
+        std::vector<uint8_t> d;
+        if (!row_v2_cell_to_v1_datum(rs.values + s, e - s, st.cols[i].ft.tp,
+                                     st.cols[i].ft.flag, &d))
+          FAIL("bad restore-data cell");
+        batch->cols[i].raw_vals.emplace_back((const char *)d.data(), d.size());
+      } else if (row_v2_is_null(rs, st.cols[i].column_id)) {
+        batch->cols[i].raw_vals.emplace_back(1, '\0');   /* NULL datum */
+      } else {
+        FAIL("restore-data row missing column");
+      }
+      ci++;
+    }
+    for (size_t ci = 0; ci < n_idx_cols && rem > 0; ci++) {
+      size_t dlen;
+      if (!split_datum(p, rem, &dlen)) FAIL("bad index key datum");
+      p += dlen; rem -= dlen;
+    }
+  } else {
+    size_t ci = 0;
+    for (size_t i = 0; i < ncols && ci < n_idx_cols; i++) {
+      if (st.cols[i].pk_handle) continue;
+      size_t dlen;
+      if (!split_datum(p, rem, &dlen)) FAIL("bad index key datum");
+      batch->cols[i].raw_vals.emplace_back((const char *)p, dlen);
+      p += dlen; rem -= dlen;
+      ci++;
+    }
   }
   for (size_t i = 0; i < ncols; i++) {
     if (!st.cols[i].pk_handle) continue;
@@ -429,11 +527,10 @@ static bool scan_process_kv_index(ScanState &st, const uint8_t *key, size_t klen
       if (rem < 9 || (p[0] != 3 && p[0] != 4)) FAIL("bad index handle datum");
       handle = p[0] == 3 ? decode_comparable_i64(p + 1)
                          : (int64_t)decode_comparable_u64(p + 1);
+    } else if (v_has_handle) {
+      handle = (int64_t)vhandle;
     } else {
-      if (vlen < 8) FAIL("unique index value too short for handle");
-      uint64_t u = 0;
-      for (int b = 0; b < 8; b++) u = (u << 8) | val[b];   /* plain BE u64 */
-      handle = (int64_t)u;
+      FAIL("index entry carries no int handle");
     }
     batch->cols[i].dec.push_int(handle);
   }

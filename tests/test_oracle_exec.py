@@ -388,6 +388,57 @@ def test_oracle_index_scan():
     assert data == d_int(9) + d_int(1) + d_int(12345)
 
 
+def v2_row(cells):
+    """tiny row-v2 (small layout) of (id, int) cells, ids ascending."""
+    body = b""
+    ends = []
+    for _, v in cells:
+        if 0 <= v <= 0xFF:
+            body += bytes([v])
+        elif 0 <= v <= 0xFFFF:
+            body += v.to_bytes(2, "little")
+        elif 0 <= v <= 0xFFFFFFFF:
+            body += v.to_bytes(4, "little")
+        else:
+            body += (v & (2**64 - 1)).to_bytes(8, "little")
+        ends.append(len(body))
+    out = bytes([128, 0, len(cells), 0, 0, 0])
+    out += bytes(cid for cid, _ in cells)
+    for e in ends:
+        out += e.to_bytes(2, "little")
+    return out + body
+
+
+def test_oracle_index_value_layouts():
+    """new index value layouts (index_scan_executor.rs:322-371):
+    unique v0 + V4 restore row; non-unique v1 + partition-id option."""
+    cols = [tikv_amd.Col(1), tikv_amd.Col(2), tikv_amd.Col(-1, pk_handle=True)]
+    req = tikv_amd.DagSelect(cols, index=True).build()
+
+    # unique new-format v0: [TailLen=8] || restore row-v2 || BE handle;
+    # columns come from the RESTORE row (V4), not the key datums
+    val = bytes([8]) + v2_row([(1, 9), (2, 77)]) + (4242).to_bytes(8, "big")
+    rows = [(idx_key(1, 1, [d_int(9), d_int(77)]), val)]
+    data, n = run_oracle(req, make_region(rows))
+    assert n == 1
+    assert data == d_int(9) + d_int(77) + d_int(4242)
+
+    # non-unique new-format v1: [0][125][1][126][pid 8B]; handle from key,
+    # V5 restore absent, columns from key datums
+    val = bytes([0, 125, 1, 126]) + (1).to_bytes(8, "big")
+    rows = [(idx_key(1, 1, [d_int(3), d_int(4)], handle=17), val)]
+    data, n = run_oracle(req, make_region(rows))
+    assert n == 1
+    assert data == d_int(3) + d_int(4) + d_int(17)
+
+    # corrupted tail_len is a loud storage error
+    import pytest
+    bad = bytes([200]) + bytes(10)
+    with pytest.raises(Exception):
+        run_oracle(req, make_region([(idx_key(1, 1, [d_int(1), d_int(2)]),
+                                      bad)]))
+
+
 def test_oracle_index_agg():
     cols = [tikv_amd.Col(1), tikv_amd.Col(2), tikv_amd.Col(-1, pk_handle=True)]
     rows = []

@@ -359,3 +359,45 @@ def test_cfg5_index_simple_agg_parity(engine):
         assert o == g
     finally:
         gen.close()
+
+
+@pytest.mark.parametrize("layout", [1, 2, 3])
+def test_index_value_layouts_parity(engine, layout):
+    """Unique-index VALUE handles and the new TailLen|Options layouts
+    (index_scan_executor.rs:322-371,416-422): 1 = unique old 8B BE value
+    handle, 2 = unique new-format v0 with a V4 restore-data row (the
+    reference's test_new_collation_unique_int_handle_index shape), 3 =
+    non-unique new-format v1 with a partition-id option segment."""
+    gen = tikv_amd.GenRegion(config_index=4, n_rows=60000, table_id=1,
+                             n_cols=layout)
+    try:
+        cols = [tikv_amd.Col(1), tikv_amd.Col(2),
+                tikv_amd.Col(-1, pk_handle=True)]
+        sel = tikv_amd.cmp_col_const(1, F.SIG_GE_INT, 0)   # amount >= 0
+        req = (tikv_amd.DagSelect(cols, index=True).where(sel)
+               .hash_agg([tikv_amd.count_star(), tikv_amd.sum_col(1),
+                          tikv_amd.max_col(2), tikv_amd.min_col(2)],
+                         tikv_amd.Expr().col(0)).build())
+        (o, on), (g, gn) = run_both(req, gen, engine)
+        assert on == gn
+        assert sorted(split_rows(o, 5)) == sorted(split_rows(g, 5))
+    finally:
+        gen.close()
+
+
+@pytest.mark.parametrize("layout", [1, 2])
+def test_index_value_handle_simple_agg(engine, layout):
+    """sum/min/max over the VALUE-decoded handle column itself."""
+    gen = tikv_amd.GenRegion(config_index=4, n_rows=50000, table_id=1,
+                             n_cols=layout)
+    try:
+        cols = [tikv_amd.Col(1), tikv_amd.Col(2),
+                tikv_amd.Col(-1, pk_handle=True)]
+        req = (tikv_amd.DagSelect(cols, index=True)
+               .simple_agg([tikv_amd.count_star(), tikv_amd.sum_col(2),
+                            tikv_amd.min_col(2), tikv_amd.max_col(2)]).build())
+        (o, on), (g, gn) = run_both(req, gen, engine)
+        assert on == gn == 1
+        assert o == g
+    finally:
+        gen.close()

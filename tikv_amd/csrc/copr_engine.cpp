@@ -797,7 +797,10 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
     sp.index_mode = 1;
     int pos = 0;
     for (auto &ci : pl->cols)
-      if (!ci.pk_handle) ci.column_id = pos++;
+      if (!ci.pk_handle) {
+        if (pos < COPR_MAX_OUT_COLS) sp.index_real_ids[pos] = ci.column_id;
+        ci.column_id = pos++;
+      }
     sp.index_n_cols = pos;
     for (auto &ci : pl->cols)
       if (ci.pk_handle) ci.column_id = pos;

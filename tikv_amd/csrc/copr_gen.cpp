@@ -396,7 +396,17 @@ copr_status copr_gen_region(const CoprGenSpec *spec, CoprGenOut *out) {
 
   uint64_t *key_offs = (uint64_t *)malloc((n + 1) * sizeof(uint64_t));
   uint64_t *val_offs = (uint64_t *)malloc((n + 1) * sizeof(uint64_t));
-  uint64_t klen = spec->config_index == 4 ? 46 : 19;
+  /* cfg5 index layout variants (spec->n_cols): 0 = non-unique old (handle
+     datum in key, value '0'); 1 = unique old (no key handle, value = 8B BE
+     handle, index_scan_executor.rs:336-345); 2 = unique new-format version
+     0 with V4 restore data (value = [TailLen=8]||row-v2(ids 1,2)||BE
+     handle — the shape of the reference's own
+     test_new_collation_unique_int_handle_index :1826-1870); 3 = non-unique
+     new-format version 1 with a partition-id option (value =
+     [0][125][1][126][pid 8B], handle from key). */
+  uint32_t il = spec->config_index == 4 ? (uint32_t)spec->n_cols : 0;
+  uint64_t klen = spec->config_index == 4 ? ((il == 1 || il == 2) ? 37 : 46)
+                                          : 19;
   uint8_t *keys = (uint8_t *)malloc(n * klen + 1);
   if (!key_offs || !val_offs || !keys) return COPR_ERR_OOM;
   for (uint64_t i = 0; i <= n; i++) key_offs[i] = i * klen;
@@ -427,25 +437,74 @@ copr_status copr_gen_region(const CoprGenSpec *spec, CoprGenOut *out) {
       int64_t handle = (int64_t)(spec->first_handle + i);
       rng.seed(seed, (uint64_t)handle);
       if (spec->config_index == 4) {
-        /* cfg5: TPCC-order-line-shaped secondary index entry (index_id 1,
-           non-unique): 't'||tid||'_i'||BE(1^S)|| INT datums (ol_w_d i64 in
-           [0,3000), amount i64 +-1e6) || INT handle
+        /* cfg5: TPCC-order-line-shaped secondary index entry (index_id 1):
+           't'||tid||'_i'||BE(1)|| INT datums (ol_w_d i64 in [0,3000),
+           amount i64 +-1e6) [|| INT handle datum for non-unique layouts]
            (index_scan_executor.rs key format; encode_index_seek_key,
            table.rs:229-235) */
-        uint8_t *k = keys + i * 46;
+        int64_t c1 = (int64_t)(rng.next() % 3000);
+        int64_t c2 = uniform_pm(rng, 1000000);
+        uint8_t *k = keys + i * klen;
         k[0] = 't';
         enc_cmp_i64(k + 1, spec->table_id);
         k[9] = '_'; k[10] = 'i';
         enc_cmp_i64(k + 11, 1);
         k[19] = 3;  /* INT_FLAG */
-        enc_cmp_i64(k + 20, (int64_t)(rng.next() % 3000));
+        enc_cmp_i64(k + 20, c1);
         k[28] = 3;
-        enc_cmp_i64(k + 29, uniform_pm(rng, 1000000));
-        k[37] = 3;
-        enc_cmp_i64(k + 38, handle);
-      } else {
-        enc_row_key(spec->table_id, handle, keys + i * 19);
+        enc_cmp_i64(k + 29, c2);
+        if (il == 0 || il == 3) {
+          k[37] = 3;
+          enc_cmp_i64(k + 38, handle);
+        }
+        row.clear();
+        switch (il) {
+          case 0:
+            row.push_back('0');          /* old-encoding non-unique value */
+            break;
+          case 1:                        /* unique old: 8B BE handle */
+            for (int b = 7; b >= 0; b--)
+              row.push_back((char)(uint8_t)((uint64_t)handle >> (8 * b)));
+            break;
+          case 2: {                      /* unique new v0 + V4 restore row */
+            std::string v2;
+            uint8_t nn = 2;
+            std::string cells;
+            uint16_t ends[2];
+            v2_int_cell(cells, c1); ends[0] = (uint16_t)cells.size();
+            v2_int_cell(cells, c2); ends[1] = (uint16_t)cells.size();
+            v2.push_back((char)128); v2.push_back(0);
+            v2.push_back((char)nn); v2.push_back(0);
+            v2.push_back(0); v2.push_back(0);
+            v2.push_back(1); v2.push_back(2);          /* ids */
+            for (int j = 0; j < 2; j++) {
+              v2.push_back((char)(uint8_t)ends[j]);
+              v2.push_back((char)(uint8_t)(ends[j] >> 8));
+            }
+            v2 += cells;
+            row.push_back(8);            /* TailLen = 8 (handle only) */
+            row += v2;                   /* restore data (flag byte 128) */
+            for (int b = 7; b >= 0; b--)
+              row.push_back((char)(uint8_t)((uint64_t)handle >> (8 * b)));
+            break;
+          }
+          case 3:                        /* non-unique new v1 + pid option */
+            row.push_back(0);            /* TailLen = 0 */
+            row.push_back(125);          /* INDEX_VALUE_VERSION_FLAG */
+            row.push_back(1);            /* version 1 */
+            row.push_back(126);          /* INDEX_VALUE_PARTITION_ID_FLAG */
+            for (int b = 7; b >= 0; b--)
+              row.push_back((char)(uint8_t)((uint64_t)spec->table_id >> (8 * b)));
+            break;
+          default:
+            ok = false;
+        }
+        if (!ok) break;
+        chunk_sizes[t][i - lo] = (uint32_t)row.size();
+        buf += row;
+        continue;
       }
+      enc_row_key(spec->table_id, handle, keys + i * 19);
       row.clear();
       bool wok = spec->row_format == 2 ? write_row_v2(cfg, rng, row)
                                        : write_row(cfg, rng, row);

@@ -147,6 +147,15 @@ struct ScanPlan {
      index_n_cols). col ids in this plan are positions. */
   int32_t index_mode;
   int32_t index_n_cols;
+  /* index mode: the ORIGINAL value stream (unique-index value handles,
+     index_scan_executor.rs:416-422, and the new TailLen|Options layouts
+     :322-371). Loaded lazily — only for rows whose key has no trailing
+     handle datum or whose value is new-format. */
+  const uint8_t *aux_vals;
+  const uint64_t *aux_val_offs;
+  /* index position -> the reference column_id (restore-data row lookups,
+     RestoreData::V4: extract_columns_from_row_format :483-501) */
+  int64_t index_real_ids[COPR_MAX_OUT_COLS];
   int32_t n_aggs;
   DevAggSpec aggs[COPR_MAX_AGGS];
 

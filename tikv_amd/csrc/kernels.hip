@@ -580,9 +580,72 @@ __device__ static inline bool d_v2_collect(const ScanPlan &plan,
  * local path): vp = the whole index KEY; comparable datums start at byte 19;
  * cell "ids" are positions 0..n-1, with the trailing int-handle datum at
  * position index_n_cols. */
+/* Index VALUE layouts (index_scan_executor.rs:322-371,416-422,700-885):
+ *   old (len <= 9): [8B BE handle]['flag'] for unique, or '0'/empty for
+ *     non-unique;
+ *   new (len > 9): TailLen | [VersionFlag Version] | Options | tail,
+ *     options = [127 CHandle-len u16le CHandle] [126 pid 8B]
+ *               [128... restore-data row-v2 to the segment end];
+ *     handle = first 8 BE bytes of the tail when TailLen >= 8
+ *     (decode_int_handle_from_value :416-422 on build_operations' tail).
+ * Returns: 1 ok, 0 malformed. *restore/-len = V4 restore-data row (version
+ * 0: columns must be read from this row-v2, :903-907); V5 restore data is
+ * skipped (int columns are never restored — need_restored_data is false
+ * for them, :639-698). Common-handle options are out of the int-handle
+ * native subset -> malformed (loud). */
+__device__ static inline bool d_index_value_split(const uint8_t *v, uint32_t n,
+                                                  int64_t *handle, bool *has_handle,
+                                                  const uint8_t **restore,
+                                                  uint32_t *restore_len) {
+  *has_handle = false;
+  *restore = nullptr;
+  *restore_len = 0;
+  if (n <= 9) {                    /* old encoding (:336-345) */
+    if (n >= 8) {
+      *handle = (int64_t)d_be_u64(v);   /* plain BE u64, NOT comparable */
+      *has_handle = true;
+    }
+    return true;                   /* '0' / empty / short: no handle */
+  }
+  uint32_t tail_len = v[0];
+  if (tail_len >= n) return false;
+  int version = 0;
+  uint32_t opt = 1;
+  if ((tail_len == 0 || tail_len == 1) && v[1] == 125 /*VERSION_FLAG*/) {
+    version = v[2];
+    opt = 3;
+  }
+  if (n < opt + tail_len) return false;
+  uint32_t opt_end = n - tail_len;
+  while (opt < opt_end) {
+    uint8_t f = v[opt];
+    if (f == 127) return false;          /* common handle: not int-handle */
+    if (f == 126) {                      /* partition id segment */
+      if (opt + 9 > opt_end) return false;
+      opt += 9;
+      continue;
+    }
+    if (f == 128) {                      /* restore data = row-v2 to end */
+      if (version == 0) {
+        *restore = v + opt;
+        *restore_len = opt_end - opt;
+      }                                  /* V5: ints never restored; skip */
+      opt = opt_end;
+      break;
+    }
+    return false;                        /* unknown segment */
+  }
+  if (tail_len >= 8) {
+    *handle = (int64_t)d_be_u64(v + n - tail_len);
+    *has_handle = true;
+  }
+  return true;
+}
+
 template <int NAGGS, bool IS_HASH>
 __device__ static inline bool d_index_collect(const ScanPlan &plan,
                                               const uint8_t *vp, uint32_t vlen,
+                                              uint64_t my_row,
                                               bool *filt_found, bool *filt_null,
                                               int64_t *filt_v, bool *grp_found,
                                               bool *grp_null, int64_t *grp_v,
@@ -590,14 +653,28 @@ __device__ static inline bool d_index_collect(const ScanPlan &plan,
                                               int64_t *handle_out,
                                               bool *handle_found) {
   if (vlen < 19 || vp[0] != 't' || vp[9] != '_' || vp[10] != 'i') return false;
-  uint32_t pos = 19;
-  int64_t ci = 0;
+  /* the VALUE decides the layout; load it lazily (non-unique old-format
+     rows keep the handle in the key and usually need no value read, but
+     new-format non-unique values carry restore data that overrides the key
+     columns, so the split must run before the key walk) */
+  const uint8_t *ival = nullptr;
+  uint32_t ivlen = 0;
+  int64_t vhandle = 0;
+  bool v_has_handle = false;
+  const uint8_t *restore = nullptr;
+  uint32_t restore_len = 0;
+  if (plan.aux_vals) {
+    uint64_t o0 = plan.aux_val_offs[my_row], o1 = plan.aux_val_offs[my_row + 1];
+    ival = plan.aux_vals + o0;
+    ivlen = (uint32_t)(o1 - o0);
+    if (!d_index_value_split(ival, ivlen, &vhandle, &v_has_handle,
+                             &restore, &restore_len))
+      return false;
+  }
   *handle_found = false;
-  while (pos < vlen) {
-    CellView cell;
-    d_parse_datum(vp + pos, vlen - pos, &cell);
-    if (cell.len == 0) return false;
-    const int64_t cell_id = ci;
+
+  /* per-cell consumer shared by the key walk and the restore-row path */
+  auto match_cell = [&](int64_t cell_id, const CellView &cell) -> bool {
     if (cell_id == (int64_t)plan.index_n_cols) {   /* trailing handle datum */
       if (!cell.has_int) return false;
       *handle_out = cell.ival;
@@ -624,11 +701,67 @@ __device__ static inline bool d_index_collect(const ScanPlan &plan,
         cols[a].iv = cell.ival;
         cols[a].has_dec = cell.has_dec;
         cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
-        if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec) return false;
+        if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec)
+          return false;
       }
     }
-    pos += cell.len;
-    ci++;
+    return true;
+  };
+
+  if (restore) {
+    /* V4 restore data: ALL index columns come from the restore row-v2,
+       keyed by the reference column ids (extract_columns_from_row_format
+       :483-501); the key datums are not consulted */
+    V2Row r;
+    if (!d_v2_parse(restore, restore_len, &r)) return false;
+    for (int32_t p = 0; p < plan.index_n_cols; p++) {
+      uint32_t s, e;
+      int st = d_v2_find(r, plan.index_real_ids[p], &s, &e);
+      if (st < 0) return false;          /* missing column (:496-497) */
+      CellView cell{};
+      if (st == 0) {
+        cell.is_null = true;
+      } else {
+        /* int columns only in the native index subset; the consumer's
+           signedness drives the v2 sign extension */
+        int64_t iv;
+        bool uns = false;
+        if (plan.has_filter && p == plan.filter_col_id)
+          uns = plan.filter_col_unsigned != 0;
+        else if (IS_HASH && p == plan.group_col_id)
+          uns = plan.group_col_unsigned != 0;
+        else
+          for (int a = 0; a < NAGGS; a++)
+            if (plan.aggs[a].kind != DAGG_COUNT_ROWS &&
+                p == plan.aggs[a].col_id) {
+              uns = plan.aggs[a].col_unsigned != 0;
+              break;
+            }
+        if (!d_v2_int(r.vals + s, e - s, uns, &iv)) return false;
+        cell.has_int = true;
+        cell.ival = iv;
+      }
+      if (!match_cell(p, cell)) return false;
+    }
+  } else {
+    uint32_t pos = 19;
+    int64_t ci = 0;
+    while (pos < vlen) {
+      CellView cell;
+      d_parse_datum(vp + pos, vlen - pos, &cell);
+      if (cell.len == 0) return false;
+      if (!match_cell(ci, cell)) return false;
+      pos += cell.len;
+      ci++;
+    }
+  }
+
+  if (!*handle_found && v_has_handle) {
+    /* unique index: PK int handle lives in the value (:553-562) */
+    CellView cell{};
+    cell.has_int = true;
+    cell.ival = vhandle;
+    if (!match_cell((int64_t)plan.index_n_cols, cell)) return false;
   }
   return true;
 }
@@ -951,7 +1084,7 @@ k_scan_agg(ScanPlan plan,
 
       if (plan.index_mode) {
         int64_t hval = 0; bool hfound = false;
-        parse_ok = d_index_collect<NAGGS, IS_HASH>(plan, vp, vlen,
+        parse_ok = d_index_collect<NAGGS, IS_HASH>(plan, vp, vlen, my_row,
                                                    &filt_found, &filt_null,
                                                    &filt_v, &grp_found,
                                                    &grp_null, &grp_v, cols,
@@ -1594,7 +1727,7 @@ k_scan_agg_pipe(ScanPlan plan,
 
       if (plan.index_mode) {
         int64_t hval = 0; bool hfound = false;
-        parse_ok = d_index_collect<NAGGS, IS_HASH>(plan, vp, vlen,
+        parse_ok = d_index_collect<NAGGS, IS_HASH>(plan, vp, vlen, my_row,
                                                    &filt_found, &filt_null,
                                                    &filt_v, &grp_found,
                                                    &grp_null, &grp_v, cols,
@@ -4883,17 +5016,22 @@ int dev_scan_launch(const ScanPlan &plan, const DevRegion &rgn,
   if (grid == 0) grid = 1;
   hipStream_t s = (hipStream_t)stream;
   DevRegion r2 = rgn;
+  ScanPlan p2 = plan;
   if (plan.index_mode) {
-    /* index scans parse the KEY stream: swap it into the streamed slot */
+    /* index scans parse the KEY stream: swap it into the streamed slot;
+       the original VALUE stream rides along for unique-index handles and
+       new-format layouts (d_index_value_split) */
     r2.d_vals = rgn.d_keys;
     r2.d_val_offs = rgn.d_key_offs;
     r2.val_bytes = rgn.key_bytes;
     r2.max_row_bytes = rgn.max_key_bytes;
+    p2.aux_vals = rgn.d_vals;
+    p2.aux_val_offs = rgn.d_val_offs;
   }
   if (plan.mode == 1)
-    return launch_agg<false>(plan, r2, d_simple, HashAggTable{}, s, grid);
+    return launch_agg<false>(p2, r2, d_simple, HashAggTable{}, s, grid);
   if (plan.mode == 2)
-    return launch_agg<true>(plan, r2, nullptr, *ht, s, grid);
+    return launch_agg<true>(p2, r2, nullptr, *ht, s, grid);
   hipLaunchKernelGGL(k_scan_project, dim3(grid), dim3(THREADS), plan.lds_bytes,
                      s, plan, rgn.d_vals, rgn.d_val_offs, rgn.d_keys,
                      rgn.d_key_offs, rgn.n_kv, *po);
