@@ -68,6 +68,21 @@ copr_status copr_region_create_mvcc(copr_engine *,
                                     const uint8_t *vals, const uint64_t *val_offs,
                                     uint64_t n_kv, uint64_t read_ts,
                                     copr_region **out);
+/* MVCC variant with the DEFAULT CF beside the write CF: Puts whose write
+ * record carries no short value (>255 B values, write.rs:296) resolve
+ * their row bytes from the default-CF stream at key =
+ * memcomparable(user_key)||BE(~start_ts) (forward.rs:433-515
+ * load_data_from_default_cf). dkeys/dvals must be sorted like RocksDB
+ * stores them (user_key asc, start_ts desc); a Put whose default entry is
+ * absent is COPR_ERR_STORAGE (the reference treats it as corruption). */
+copr_status copr_region_create_mvcc_with_default(copr_engine *,
+                                    const uint8_t *keys, const uint64_t *key_offs,
+                                    const uint8_t *vals, const uint64_t *val_offs,
+                                    uint64_t n_kv,
+                                    const uint8_t *dkeys, const uint64_t *dkey_offs,
+                                    const uint8_t *dvals, const uint64_t *dval_offs,
+                                    uint64_t n_default,
+                                    uint64_t read_ts, copr_region **out);
 void        copr_region_destroy(copr_region *);
 uint64_t    copr_region_num_kv(const copr_region *);
 

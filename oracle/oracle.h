@@ -74,6 +74,15 @@ typedef struct OrcRegion {
 int  orc_mvcc_filter(const uint8_t *keys, const uint64_t *key_offs,
                      const uint8_t *vals, const uint64_t *val_offs,
                      uint64_t n_kv, uint64_t read_ts, OrcRegion *out);
+/* with the DEFAULT CF beside the write CF (forward.rs:433-515
+ * load_data_from_default_cf): Puts without a short value resolve from
+ * dkeys at memcomparable(user_key)||BE(~start_ts). */
+int  orc_mvcc_filter2(const uint8_t *keys, const uint64_t *key_offs,
+                      const uint8_t *vals, const uint64_t *val_offs,
+                      uint64_t n_kv,
+                      const uint8_t *dkeys, const uint64_t *dkey_offs,
+                      const uint8_t *dvals, const uint64_t *dval_offs,
+                      uint64_t n_default, uint64_t read_ts, OrcRegion *out);
 void orc_region_free(OrcRegion *);
 
 #ifdef __cplusplus

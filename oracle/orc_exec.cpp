@@ -1694,7 +1694,8 @@ int orc_test_dec_from_i64_encode(int64_t v, uint8_t *out) {
  * types.rs:721-731) -> next older version. gc_fence unsupported. */
 static bool parse_write_rec(const uint8_t *v, size_t len, char *type,
                             const uint8_t **sv, size_t *sv_len,
-                            uint64_t *gc_fence, int *lc_not_exist) {
+                            uint64_t *gc_fence, int *lc_not_exist,
+                            uint64_t *start_ts = nullptr) {
   *sv = nullptr; *sv_len = 0; *gc_fence = 0; *lc_not_exist = 0;
   if (len < 1) return false;
   char t = (char)v[0];
@@ -1703,6 +1704,7 @@ static bool parse_write_rec(const uint8_t *v, size_t len, char *type,
   size_t p = 1;
   uint64_t sts; size_t n;
   if (!decode_var_u64(v + p, len - p, &sts, &n)) return false;
+  if (start_ts) *start_ts = sts;
   p += n;
   while (p < len) {
     uint8_t tag = v[p++];
@@ -1751,11 +1753,41 @@ static bool parse_write_rec(const uint8_t *v, size_t len, char *type,
   return true;
 }
 
-extern "C" int orc_mvcc_filter(const uint8_t *keys, const uint64_t *key_offs,
-                               const uint8_t *vals, const uint64_t *val_offs,
-                               uint64_t n_kv, uint64_t read_ts, OrcRegion *out) {
+static int mvcc_filter_impl(const uint8_t *keys, const uint64_t *key_offs,
+                            const uint8_t *vals, const uint64_t *val_offs,
+                            uint64_t n_kv,
+                            const uint8_t *dkeys, const uint64_t *dkey_offs,
+                            const uint8_t *dvals, const uint64_t *dval_offs,
+                            uint64_t n_default,
+                            uint64_t read_ts, OrcRegion *out) {
   std::vector<uint8_t> okeys, ovals;
   std::vector<uint64_t> okoffs{0}, ovoffs{0};
+  /* default-CF lookup at memcomparable(user_key)||BE(~start_ts)
+     (forward.rs:433-515 load_data_from_default_cf) */
+  auto default_find = [&](const uint8_t *u, size_t ulen,
+                          uint64_t start_ts) -> int64_t {
+    uint64_t tsd = ~start_ts;
+    uint8_t suffix[8];
+    for (int b = 0; b < 8; b++) suffix[b] = (uint8_t)(tsd >> (8 * (7 - b)));
+    uint64_t lo = 0, hi = n_default;
+    while (lo < hi) {
+      uint64_t mid = (lo + hi) >> 1;
+      const uint8_t *d = dkeys + dkey_offs[mid];
+      size_t dlen = (size_t)(dkey_offs[mid + 1] - dkey_offs[mid]);
+      size_t tot = ulen + 8;
+      int c = 0;
+      size_t n = tot < dlen ? tot : dlen;
+      for (size_t i2 = 0; i2 < n && c == 0; i2++) {
+        uint8_t ub = i2 < ulen ? u[i2] : suffix[i2 - ulen];
+        if (ub != d[i2]) c = ub < d[i2] ? -1 : 1;
+      }
+      if (c == 0) c = tot == dlen ? 0 : (tot < dlen ? -1 : 1);
+      if (c == 0) return (int64_t)mid;
+      if (c < 0) hi = mid;
+      else lo = mid + 1;
+    }
+    return -1;
+  };
   auto uenc = [&](uint64_t i, size_t *len) -> const uint8_t * {
     size_t kl = (size_t)(key_offs[i + 1] - key_offs[i]);
     if (kl < 9) return nullptr;
@@ -1782,15 +1814,22 @@ extern "C" int orc_mvcc_filter(const uint8_t *keys, const uint64_t *key_offs,
       uint64_t ts = commit_ts(j);
       if (ts > read_ts) { j++; continue; }
       char type; const uint8_t *sv; size_t svl; uint64_t fence; int lc_ne;
+      uint64_t start_ts = 0;
       if (!parse_write_rec(vals + val_offs[j],
                            (size_t)(val_offs[j + 1] - val_offs[j]),
-                           &type, &sv, &svl, &fence, &lc_ne))
+                           &type, &sv, &svl, &fence, &lc_ne, &start_ts))
         return 1;
       /* gc fence pointing within read_ts: key invisible
          (write.rs:425-442 + forward.rs:444-446 break None) */
       if (fence != 0 && fence <= read_ts) break;
       if (type == 'P') {
-        if (!sv) return 2;                     /* default-CF value */
+        if (!sv) {
+          if (!dkeys) return 2;                /* no default stream: loud */
+          int64_t m = default_find(u, ulen, start_ts);
+          if (m < 0) return 1;                 /* corruption: DEFAULT_NOT_FOUND */
+          sv = dvals + dval_offs[m];
+          svl = (size_t)(dval_offs[m + 1] - dval_offs[m]);
+        }
         std::vector<uint8_t> raw;
         if (!memcmp_decode(u, ulen, &raw)) return 1;
         okeys.insert(okeys.end(), raw.begin(), raw.end());
@@ -1823,6 +1862,25 @@ extern "C" int orc_mvcc_filter(const uint8_t *keys, const uint64_t *key_offs,
   out->val_offs = (uint64_t *)malloc((n + 1) * 8);
   memcpy(out->val_offs, ovoffs.data(), (n + 1) * 8);
   return 0;
+}
+
+extern "C" int orc_mvcc_filter(const uint8_t *keys, const uint64_t *key_offs,
+                               const uint8_t *vals, const uint64_t *val_offs,
+                               uint64_t n_kv, uint64_t read_ts, OrcRegion *out) {
+  return mvcc_filter_impl(keys, key_offs, vals, val_offs, n_kv,
+                          nullptr, nullptr, nullptr, nullptr, 0, read_ts, out);
+}
+
+extern "C" int orc_mvcc_filter2(const uint8_t *keys, const uint64_t *key_offs,
+                                const uint8_t *vals, const uint64_t *val_offs,
+                                uint64_t n_kv,
+                                const uint8_t *dkeys, const uint64_t *dkey_offs,
+                                const uint8_t *dvals, const uint64_t *dval_offs,
+                                uint64_t n_default,
+                                uint64_t read_ts, OrcRegion *out) {
+  return mvcc_filter_impl(keys, key_offs, vals, val_offs, n_kv,
+                          dkeys, dkey_offs, dvals, dval_offs, n_default,
+                          read_ts, out);
 }
 
 extern "C" void orc_region_free(OrcRegion *r) {

@@ -52,6 +52,13 @@ def load_lib():
     lib.orc_mvcc_filter.argtypes = [C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
                                     C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
                                     C.c_uint64, C.c_uint64, C.POINTER(OrcRegion)]
+    lib.orc_mvcc_filter2.restype = C.c_int
+    lib.orc_mvcc_filter2.argtypes = [
+        C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
+        C.POINTER(C.c_uint8), C.POINTER(C.c_uint64), C.c_uint64,
+        C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
+        C.POINTER(C.c_uint8), C.POINTER(C.c_uint64), C.c_uint64,
+        C.c_uint64, C.POINTER(OrcRegion)]
     lib.orc_region_free.argtypes = [C.POINTER(OrcRegion)]
     lib.orc_block_parse.restype = C.c_int
     lib.orc_block_parse.argtypes = [C.POINTER(C.c_uint8),
@@ -130,12 +137,20 @@ def block_parse(blocks, block_offs, n_blocks):
     return keysb, koffs, valsb, voffs, n
 
 
-def mvcc_filter(keys, key_offs, vals, val_offs, n_kv, read_ts):
-    """Run the oracle MVCC filter; returns (keys, key_offs, vals, val_offs, n)."""
+def mvcc_filter(keys, key_offs, vals, val_offs, n_kv, read_ts,
+                default_cf=None):
+    """Run the oracle MVCC filter; returns (keys, key_offs, vals, val_offs,
+    n). default_cf = (dkeys, dkey_offs, dvals, dval_offs, n) resolves Puts
+    without short values (forward.rs:433-515)."""
     lib = load_lib()
     out = OrcRegion()
-    st = lib.orc_mvcc_filter(keys, key_offs, vals, val_offs, n_kv, read_ts,
-                             C.byref(out))
+    if default_cf is not None:
+        dk, dko, dv, dvo, dn = default_cf
+        st = lib.orc_mvcc_filter2(keys, key_offs, vals, val_offs, n_kv,
+                                  dk, dko, dv, dvo, dn, read_ts, C.byref(out))
+    else:
+        st = lib.orc_mvcc_filter(keys, key_offs, vals, val_offs, n_kv,
+                                 read_ts, C.byref(out))
     if st != 0:
         raise RuntimeError("orc_mvcc_filter: %d" % st)
     n = out.n_kv

@@ -274,3 +274,115 @@ def test_mvcc_random_chains_vs_oracle():
             got[keys[ko[i]:ko[i + 1]]] = vals[vo[i]:vo[i + 1]]
         want = _py_visible(entries_by_key, read_ts)
         assert got == want, (trial, read_ts)
+
+
+# ---------- default-CF lookup (forward.rs:433-515; write.rs:296) ----------
+def dkey(user_raw, start_ts):
+    """default-CF key = memcomparable(user_key) || BE(~start_ts) — the same
+    append_ts encoding as write keys but with the START ts."""
+    return memcmp_enc(user_raw) + ((~start_ts) & (2**64 - 1)).to_bytes(8, "big")
+
+
+def test_default_cf_lookup_oracle():
+    long_a = b"A" * 300            # >255 B: cannot be a short value
+    long_b = b"B" * 400
+    wents = [(wkey(UK_A, 100), wval("P", 99)),          # no short value
+             (wkey(UK_B, 200), wval("P", 150))]
+    dents = [(dkey(UK_A, 99), long_a), (dkey(UK_B, 150), long_b)]
+    wents.sort(key=lambda e: e[0])
+    dents.sort(key=lambda e: e[0])
+    orc = _orc()
+    k, ko, v, vo, n, keep = make_arrays(wents)
+    dk, dko, dv, dvo, dn, keep2 = make_arrays(dents)
+    keys, kof, vals, vof, nv = orc.mvcc_filter(
+        k, ko, v, vo, n, 1000, default_cf=(dk, dko, dv, dvo, dn))
+    assert nv == 2
+    got = {keys[kof[i]:kof[i + 1]]: vals[vof[i]:vof[i + 1]] for i in range(nv)}
+    assert got == {UK_A: long_a, UK_B: long_b}
+
+    # mixed: one short value, one default-CF value
+    wents = [(wkey(UK_A, 100), wval("P", 99)),
+             (wkey(UK_B, 200), wval("P", 150, b"short"))]
+    wents.sort(key=lambda e: e[0])
+    k, ko, v, vo, n, keep = make_arrays(wents)
+    keys, kof, vals, vof, nv = orc.mvcc_filter(
+        k, ko, v, vo, n, 1000, default_cf=(dk, dko, dv, dvo, dn))
+    got = {keys[kof[i]:kof[i + 1]]: vals[vof[i]:vof[i + 1]] for i in range(nv)}
+    assert got == {UK_A: long_a, UK_B: b"short"}
+
+    # missing default entry = corruption (DEFAULT_NOT_FOUND), loud
+    wents = [(wkey(UK_A, 100), wval("P", 77))]          # start_ts 77 absent
+    k, ko, v, vo, n, keep = make_arrays(wents)
+    import pytest
+    with pytest.raises(RuntimeError):
+        orc.mvcc_filter(k, ko, v, vo, n, 1000,
+                        default_cf=(dk, dko, dv, dvo, dn))
+    # and with NO default stream it stays loudly unsupported
+    with pytest.raises(RuntimeError):
+        orc.mvcc_filter(k, ko, v, vo, n, 1000)
+
+
+@pytest.mark.gpu
+def test_default_cf_lookup_device(engine):
+    """device MVCC filter with default-CF stream == oracle, and the
+    resulting region (long row-v1 values) scans correctly end to end."""
+    import tikv_amd
+
+    def zig(v):
+        return (v << 1) if v >= 0 else (((-v) << 1) - 1)
+
+    def row_v1(cells):
+        """[VAR_INT flag, zigzag col_id, VAR_INT flag, zigzag value]* — the
+        encode_row layout (codec/table.rs:166-184)."""
+        out = bytearray()
+        for cid, val in cells:
+            out.append(8)
+            out += var_u64(zig(cid))
+            out.append(8)
+            out += var_u64(zig(val))
+        return bytes(out)
+
+    rng = __import__("random").Random(7)
+    wents, dents = [], []
+    expect_count = 0
+    for h in range(2000):
+        uk = b"t" + bytes(7) + bytes([1]) + b"_r" + h.to_bytes(8, "big")
+        val = row_v1([(1, h), (2, rng.randrange(0, 100))])
+        # pad with a big bytes cell so the value exceeds 255 B for half
+        if h % 2 == 0:
+            pad = bytes([97 + (h % 26)]) * 300
+            val += bytes([8]) + var_u64(zig(3)) + bytes([2]) + \
+                var_u64(zig(len(pad))) + pad
+            wents.append((wkey(uk, 100 + h), wval("P", 99)))
+            dents.append((dkey(uk, 99), val))
+        else:
+            wents.append((wkey(uk, 100 + h), wval("P", 99, val)))
+        expect_count += 1
+    wents.sort(key=lambda e: e[0])
+    dents.sort(key=lambda e: e[0])
+    k, ko, v, vo, n, keep = make_arrays(wents)
+    dk, dko, dv, dvo, dn, keep2 = make_arrays(dents)
+    rgn = engine.region_mvcc_with_default(k, ko, v, vo, n,
+                                          dk, dko, dv, dvo, dn, 1 << 40)
+    try:
+        from tikv_amd import _ffi as F
+        cols = [tikv_amd.Col(1), tikv_amd.Col(2)]
+        sel = tikv_amd.cmp_col_const(1, F.SIG_GE_INT, 0)
+        req = (tikv_amd.DagSelect(cols).where(sel)
+               .simple_agg([tikv_amd.count_star(), tikv_amd.sum_col(0)]).build())
+        g_data, g_n, _ = engine.dag_run(req, [rgn])
+    finally:
+        rgn.close()
+    # oracle over the same streams
+    orc = _orc()
+    okeys, okof, ovals, ovof, onv = orc.mvcc_filter(
+        k, ko, v, vo, n, 1 << 40, default_cf=(dk, dko, dv, dvo, dn))
+    assert onv == expect_count
+    o_data, o_n = orc.dag_run(req, C.cast(
+        (C.c_uint8 * len(okeys)).from_buffer_copy(okeys), C.POINTER(C.c_uint8)),
+        (C.c_uint64 * len(okof))(*okof),
+        C.cast((C.c_uint8 * len(ovals)).from_buffer_copy(ovals),
+               C.POINTER(C.c_uint8)),
+        (C.c_uint64 * len(ovof))(*ovof), onv)
+    assert o_n == g_n == 1
+    assert o_data == g_data

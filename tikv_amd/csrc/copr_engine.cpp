@@ -359,7 +359,8 @@ extern "C" copr_status copr_region_create_blocks_mvcc(
   DevRegion vis{};
   int unsup = 0;
   rc = dev_mvcc_build(raw.d_keys, raw.d_key_offs, raw.d_vals, raw.d_val_offs,
-                      raw.n_kv, read_ts, &vis, &unsup, eng->stream);
+                      raw.n_kv, nullptr, nullptr, nullptr, nullptr, 0,
+                      read_ts, &vis, &unsup, eng->stream);
   hipFree(raw.d_keys); hipFree(raw.d_key_offs);
   hipFree(raw.d_vals); hipFree(raw.d_val_offs);
   if (rc == -3) return SET_ERR(COPR_ERR_UNSUPPORTED, "mvcc: default-CF value");
@@ -368,18 +369,26 @@ extern "C" copr_status copr_region_create_blocks_mvcc(
   return region_from_dev(eng, vis, out);
 }
 
-extern "C" copr_status copr_region_create_mvcc(copr_engine *eng,
-                                               const uint8_t *keys,
-                                               const uint64_t *key_offs,
-                                               const uint8_t *vals,
-                                               const uint64_t *val_offs,
-                                               uint64_t n_kv, uint64_t read_ts,
-                                               copr_region **out) {
+static copr_status region_create_mvcc_impl(copr_engine *eng,
+                                           const uint8_t *keys,
+                                           const uint64_t *key_offs,
+                                           const uint8_t *vals,
+                                           const uint64_t *val_offs,
+                                           uint64_t n_kv,
+                                           const uint8_t *dkeys,
+                                           const uint64_t *dkey_offs,
+                                           const uint8_t *dvals,
+                                           const uint64_t *dval_offs,
+                                           uint64_t n_default,
+                                           uint64_t read_ts,
+                                           copr_region **out) {
   if (!eng) return SET_ERR(COPR_ERR_INVALID_REQUEST, "null engine");
   HIP_TRY(hipSetDevice(eng->device), "hipSetDevice");
-  /* upload raw write-CF arrays (temporary) */
+  /* upload raw write-CF (+ optional default-CF) arrays (temporary) */
   uint8_t *rk = nullptr, *rv = nullptr;
   uint64_t *rko = nullptr, *rvo = nullptr;
+  uint8_t *dk = nullptr, *dv = nullptr;
+  uint64_t *dko = nullptr, *dvo = nullptr;
   hipError_t e = hipSuccess;
   auto up = [&](void **dst, const void *srcp, uint64_t bytes) {
     if (e != hipSuccess) return;
@@ -390,7 +399,16 @@ extern "C" copr_status copr_region_create_mvcc(copr_engine *eng,
   up((void **)&rko, key_offs, (n_kv + 1) * 8);
   up((void **)&rv, vals, val_offs[n_kv]);
   up((void **)&rvo, val_offs, (n_kv + 1) * 8);
-  auto free_raw = [&]() { hipFree(rk); hipFree(rko); hipFree(rv); hipFree(rvo); };
+  if (dkeys) {
+    up((void **)&dk, dkeys, dkey_offs[n_default]);
+    up((void **)&dko, dkey_offs, (n_default + 1) * 8);
+    up((void **)&dv, dvals, dval_offs[n_default]);
+    up((void **)&dvo, dval_offs, (n_default + 1) * 8);
+  }
+  auto free_raw = [&]() {
+    hipFree(rk); hipFree(rko); hipFree(rv); hipFree(rvo);
+    hipFree(dk); hipFree(dko); hipFree(dv); hipFree(dvo);
+  };
   if (e != hipSuccess) {
     free_raw();
     return SET_ERR(e == hipErrorOutOfMemory ? COPR_ERR_OOM : COPR_ERR_INTERNAL,
@@ -398,8 +416,8 @@ extern "C" copr_status copr_region_create_mvcc(copr_engine *eng,
   }
   DevRegion dr{};
   int unsup = 0;
-  int rc = dev_mvcc_build(rk, rko, rv, rvo, n_kv, read_ts, &dr, &unsup,
-                          eng->stream);
+  int rc = dev_mvcc_build(rk, rko, rv, rvo, n_kv, dk, dko, dv, dvo, n_default,
+                          read_ts, &dr, &unsup, eng->stream);
   free_raw();
   if (rc == -3) return SET_ERR(COPR_ERR_UNSUPPORTED, "mvcc: default-CF value");
   if (rc == -2) return SET_ERR(COPR_ERR_OOM, "mvcc build oom");
@@ -421,6 +439,31 @@ extern "C" copr_status copr_region_create_mvcc(copr_engine *eng,
   if (!getenv("COPR_NO_DIR")) dev_celldir_build(r->dev, eng->stream);
   *out = r;
   return COPR_OK;
+}
+
+extern "C" copr_status copr_region_create_mvcc(copr_engine *eng,
+                                               const uint8_t *keys,
+                                               const uint64_t *key_offs,
+                                               const uint8_t *vals,
+                                               const uint64_t *val_offs,
+                                               uint64_t n_kv, uint64_t read_ts,
+                                               copr_region **out) {
+  return region_create_mvcc_impl(eng, keys, key_offs, vals, val_offs, n_kv,
+                                 nullptr, nullptr, nullptr, nullptr, 0,
+                                 read_ts, out);
+}
+
+extern "C" copr_status copr_region_create_mvcc_with_default(
+    copr_engine *eng, const uint8_t *keys, const uint64_t *key_offs,
+    const uint8_t *vals, const uint64_t *val_offs, uint64_t n_kv,
+    const uint8_t *dkeys, const uint64_t *dkey_offs, const uint8_t *dvals,
+    const uint64_t *dval_offs, uint64_t n_default, uint64_t read_ts,
+    copr_region **out) {
+  if (!dkeys || !dkey_offs || !dvals || !dval_offs)
+    return SET_ERR(COPR_ERR_INVALID_REQUEST, "null default-CF stream");
+  return region_create_mvcc_impl(eng, keys, key_offs, vals, val_offs, n_kv,
+                                 dkeys, dkey_offs, dvals, dval_offs,
+                                 n_default, read_ts, out);
 }
 
 extern "C" copr_status copr_region_dump(copr_engine *eng, copr_region *r,

@@ -273,8 +273,12 @@ int dev_blocks_build(const uint8_t *h_blocks, const uint64_t *h_block_offs,
                      uint32_t n_blocks, DevRegion *out, void *stream);
 int dev_subregion_build(const DevRegion &src, const uint32_t *h_rows,
                         uint64_t m, DevRegion *out, void *stream);
+/* dd_* = the DEFAULT-CF stream (sorted user_key asc / start_ts desc), or
+ * null: a Put without a short value then fails loudly (unsupported). */
 int dev_mvcc_build(const uint8_t *d_keys, const uint64_t *d_ko,
                    const uint8_t *d_vals, const uint64_t *d_vo, uint64_t n,
+                   const uint8_t *dd_keys, const uint64_t *dd_ko,
+                   const uint8_t *dd_vals, const uint64_t *dd_vo, uint64_t dn,
                    uint64_t read_ts, DevRegion *out, int *unsupported,
                    void *stream);
 

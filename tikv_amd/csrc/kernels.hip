@@ -3116,7 +3116,8 @@ k_crc64(const uint8_t *__restrict__ vals, const uint64_t *__restrict__ val_offs,
 __device__ static inline bool d_parse_write_rec(const uint8_t *v, uint32_t len,
                                                 char *type, uint32_t *sv_off,
                                                 uint32_t *sv_len,
-                                                uint64_t *fence, int *lc_ne) {
+                                                uint64_t *fence, int *lc_ne,
+                                                uint64_t *start_ts = nullptr) {
   *sv_off = 0; *sv_len = 0; *fence = 0; *lc_ne = 0;
   if (len < 1) return false;
   char t = (char)v[0];
@@ -3125,6 +3126,7 @@ __device__ static inline bool d_parse_write_rec(const uint8_t *v, uint32_t len,
   uint32_t p = 1;
   uint64_t sts; uint32_t n;
   if (!d_var_u64(v + p, len - p, &sts, &n)) return false;
+  if (start_ts) *start_ts = sts;
   p += n;
   while (p < len) {
     uint8_t tag = v[p++];
@@ -3173,15 +3175,55 @@ __device__ static inline bool d_ukey_eq(const uint8_t *a, uint32_t alen,
   return true;
 }
 
+/* lexicographic compare of (ukey||BE(~ts)) against a default-CF key */
+__device__ static inline int d_dkey_cmp(const uint8_t *u, uint32_t ulen,
+                                        uint64_t ts_desc_be,
+                                        const uint8_t *d, uint32_t dlen) {
+  uint32_t tot = ulen + 8;
+  uint32_t n = tot < dlen ? tot : dlen;
+  for (uint32_t i = 0; i < n; i++) {
+    uint8_t ub = i < ulen ? u[i]
+                          : (uint8_t)(ts_desc_be >> (8 * (7 - (i - ulen))));
+    if (ub != d[i]) return ub < d[i] ? -1 : 1;
+  }
+  return tot == dlen ? 0 : (tot < dlen ? -1 : 1);
+}
+
+/* default-CF lookup (forward.rs:433-515 load_data_from_default_cf /
+ * near_loadData; write.rs:296 short-value-vs-default split): the value of a
+ * Put without a short value lives at key = memcomparable(user_key) ||
+ * BE(~start_ts) in the default CF. Returns the entry index or UINT64_MAX. */
+__device__ static inline uint64_t d_default_cf_find(
+    const uint8_t *__restrict__ dkeys, const uint64_t *__restrict__ dko,
+    uint64_t dn, const uint8_t *u, uint32_t ulen, uint64_t start_ts) {
+  uint64_t ts_desc_be = ~start_ts;       /* compared as BE bytes */
+  uint64_t lo = 0, hi = dn;
+  while (lo < hi) {
+    uint64_t mid = (lo + hi) >> 1;
+    const uint8_t *d = dkeys + dko[mid];
+    uint32_t dlen = (uint32_t)(dko[mid + 1] - dko[mid]);
+    int c = d_dkey_cmp(u, ulen, ts_desc_be, d, dlen);
+    if (c == 0) return mid;
+    if (c < 0) hi = mid;
+    else lo = mid + 1;
+  }
+  return UINT64_MAX;
+}
+
 __global__ void __launch_bounds__(THREADS)
 k_mvcc_flags(const uint8_t *__restrict__ keys, const uint64_t *__restrict__ ko,
              const uint8_t *__restrict__ vals, const uint64_t *__restrict__ vo,
              uint64_t n, uint64_t read_ts,
+             const uint8_t *__restrict__ dkeys,
+             const uint64_t *__restrict__ dko,
+             const uint64_t *__restrict__ dvo, uint64_t dn,
+             uint64_t *__restrict__ dref,
              uint8_t *__restrict__ vis, uint32_t *__restrict__ ksz,
              uint32_t *__restrict__ vsz, unsigned int *__restrict__ err) {
   for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (uint64_t)gridDim.x * blockDim.x) {
     vis[i] = 0; ksz[i] = 0; vsz[i] = 0;
+    if (dref) dref[i] = UINT64_MAX;
     uint32_t klen = (uint32_t)(ko[i + 1] - ko[i]);
     if (klen < 17 || ((klen - 8) % 9) != 0) { atomicOr(err, 1u); continue; }
     const uint8_t *k = keys + ko[i];
@@ -3189,8 +3231,10 @@ k_mvcc_flags(const uint8_t *__restrict__ keys, const uint64_t *__restrict__ ko,
     uint64_t ts = ~d_be_u64(k + ulen);
     if (ts > read_ts) continue;
     char type; uint32_t sv_off, sv_len; uint64_t fence; int lc_ne;
+    uint64_t start_ts = 0;
     if (!d_parse_write_rec(vals + vo[i], (uint32_t)(vo[i + 1] - vo[i]),
-                           &type, &sv_off, &sv_len, &fence, &lc_ne)) {
+                           &type, &sv_off, &sv_len, &fence, &lc_ne,
+                           &start_ts)) {
       atomicOr(err, 1u);
       continue;
     }
@@ -3218,7 +3262,16 @@ k_mvcc_flags(const uint8_t *__restrict__ keys, const uint64_t *__restrict__ ko,
       if (lnj) { emit = false; break; }
     }
     if (!emit) continue;
-    if (sv_len == 0 && sv_off == 0) { atomicOr(err, 2u); continue; } /* default CF */
+    uint32_t out_vlen = sv_len;
+    if (sv_len == 0 && sv_off == 0) {
+      /* no short value: the row value lives in the default CF at
+         user_key||start_ts (forward.rs:433-515; write.rs:296) */
+      if (!dkeys) { atomicOr(err, 2u); continue; }   /* no stream: loud */
+      uint64_t m = d_default_cf_find(dkeys, dko, dn, k, ulen, start_ts);
+      if (m == UINT64_MAX) { atomicOr(err, 1u); continue; }  /* corruption */
+      dref[i] = m;
+      out_vlen = (uint32_t)(dvo[m + 1] - dvo[m]);
+    }
     /* decoded user key length from the memcomparable groups */
     uint32_t groups = ulen / 9;
     uint8_t marker = k[ulen - 1];
@@ -3226,7 +3279,7 @@ k_mvcc_flags(const uint8_t *__restrict__ keys, const uint64_t *__restrict__ ko,
     if (pad > 8) { atomicOr(err, 1u); continue; }
     vis[i] = 1;
     ksz[i] = (groups - 1) * 8 + (8 - pad);
-    vsz[i] = sv_len;
+    vsz[i] = out_vlen;
   }
 }
 
@@ -3247,6 +3300,9 @@ __global__ void __launch_bounds__(THREADS)
 k_mvcc_gather(const uint8_t *__restrict__ keys, const uint64_t *__restrict__ ko,
               const uint8_t *__restrict__ vals, const uint64_t *__restrict__ vo,
               uint64_t n,
+              const uint8_t *__restrict__ dvals,
+              const uint64_t *__restrict__ dvo,
+              const uint64_t *__restrict__ dref,
               const uint8_t *__restrict__ vis, const uint32_t *__restrict__ ksz,
               const uint32_t *__restrict__ vsz,
               const uint64_t *__restrict__ idx_sc,
@@ -3273,16 +3329,27 @@ k_mvcc_gather(const uint8_t *__restrict__ keys, const uint64_t *__restrict__ ko,
     char type; uint32_t sv_off, sv_len; uint64_t fence; int lc_ne;
     d_parse_write_rec(vals + vo[i], (uint32_t)(vo[i + 1] - vo[i]),
                       &type, &sv_off, &sv_len, &fence, &lc_ne);
-    const uint8_t *sv = vals + vo[i] + sv_off;
-    for (uint32_t b = 0; b < sv_len; b++) out_vals[vout + b] = sv[b];
+    const uint8_t *sv;
+    uint32_t out_vlen;
+    if (dref && dref[i] != UINT64_MAX) {     /* default-CF value */
+      uint64_t m = dref[i];
+      sv = dvals + dvo[m];
+      out_vlen = (uint32_t)(dvo[m + 1] - dvo[m]);
+    } else {
+      sv = vals + vo[i] + sv_off;
+      out_vlen = sv_len;
+    }
+    for (uint32_t b = 0; b < out_vlen; b++) out_vals[vout + b] = sv[b];
     out_ko[pos] = kout;
     out_vo[pos] = vout;
-    atomicMax(max_row, (unsigned long long)sv_len);
+    atomicMax(max_row, (unsigned long long)out_vlen);
   }
 }
 
 int dev_mvcc_build(const uint8_t *d_keys, const uint64_t *d_ko,
                    const uint8_t *d_vals, const uint64_t *d_vo, uint64_t n,
+                   const uint8_t *dd_keys, const uint64_t *dd_ko,
+                   const uint8_t *dd_vals, const uint64_t *dd_vo, uint64_t dn,
                    uint64_t read_ts, DevRegion *out, int *unsupported,
                    void *stream) {
   *unsupported = 0;
@@ -3292,6 +3359,7 @@ int dev_mvcc_build(const uint8_t *d_keys, const uint64_t *d_ko,
   unsigned int *err = nullptr;
   uint64_t *v64 = nullptr, *k64 = nullptr, *s64 = nullptr;
   uint64_t *idx_sc = nullptr, *kb_sc = nullptr, *vb_sc = nullptr;
+  uint64_t *dref = nullptr;
   unsigned long long *d_maxrow = nullptr;
   void *tmp = nullptr;
   size_t tmp_bytes = 0;
@@ -3300,7 +3368,7 @@ int dev_mvcc_build(const uint8_t *d_keys, const uint64_t *d_ko,
     hipFree(vis); hipFree(ksz); hipFree(vsz); hipFree(err);
     hipFree(v64); hipFree(k64); hipFree(s64);
     hipFree(idx_sc); hipFree(kb_sc); hipFree(vb_sc);
-    hipFree(d_maxrow); hipFree(tmp);
+    hipFree(dref); hipFree(d_maxrow); hipFree(tmp);
   };
   uint64_t na = n ? n : 1;
   if (e == hipSuccess) e = hipMalloc(&vis, na);
@@ -3313,6 +3381,7 @@ int dev_mvcc_build(const uint8_t *d_keys, const uint64_t *d_ko,
   if (e == hipSuccess) e = hipMalloc(&idx_sc, na * 8);
   if (e == hipSuccess) e = hipMalloc(&kb_sc, na * 8);
   if (e == hipSuccess) e = hipMalloc(&vb_sc, na * 8);
+  if (e == hipSuccess && dd_keys) e = hipMalloc(&dref, na * 8);
   if (e == hipSuccess) e = hipMalloc(&d_maxrow, 8);
   if (e != hipSuccess) { freeall(); return -2; }
   hipMemsetAsync(err, 0, 4, s);
@@ -3321,7 +3390,8 @@ int dev_mvcc_build(const uint8_t *d_keys, const uint64_t *d_ko,
                                  ? ((n + THREADS - 1) / THREADS) : 8192);
   if (grid == 0) grid = 1;
   hipLaunchKernelGGL(k_mvcc_flags, dim3(grid), dim3(THREADS), 0, s,
-                     d_keys, d_ko, d_vals, d_vo, n, read_ts, vis, ksz, vsz, err);
+                     d_keys, d_ko, d_vals, d_vo, n, read_ts,
+                     dd_keys, dd_ko, dd_vo, dn, dref, vis, ksz, vsz, err);
   hipLaunchKernelGGL(k_mvcc_widen, dim3(grid), dim3(THREADS), 0, s,
                      vis, ksz, vsz, n, v64, k64, s64);
   unsigned int h_err = 0;
@@ -3360,7 +3430,8 @@ int dev_mvcc_build(const uint8_t *d_keys, const uint64_t *d_ko,
     return -2;
   }
   hipLaunchKernelGGL(k_mvcc_gather, dim3(grid), dim3(THREADS), 0, s,
-                     d_keys, d_ko, d_vals, d_vo, n, vis, ksz, vsz,
+                     d_keys, d_ko, d_vals, d_vo, n, dd_vals, dd_vo, dref,
+                     vis, ksz, vsz,
                      idx_sc, kb_sc, vb_sc, o_keys, o_ko, o_vals, o_vo, d_maxrow);
   hipMemcpyAsync(o_ko + n_vis, &ktot, 8, hipMemcpyHostToDevice, s);
   hipMemcpyAsync(o_vo + n_vis, &vtot, 8, hipMemcpyHostToDevice, s);

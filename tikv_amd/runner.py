@@ -367,6 +367,21 @@ class Engine:
                                (st, self._lib.copr_last_error().decode()))
         return Region(self, r)
 
+    def region_mvcc_with_default(self, keys, key_offs, vals, val_offs, n_kv,
+                                 dkeys, dkey_offs, dvals, dval_offs,
+                                 n_default, read_ts):
+        """MVCC region with the default CF beside the write CF: Puts without
+        short values resolve from the default stream (forward.rs:433-515)."""
+        r = C.c_void_p()
+        st = self._lib.copr_region_create_mvcc_with_default(
+            self._h, keys, key_offs, vals, val_offs, n_kv,
+            dkeys, dkey_offs, dvals, dval_offs, n_default,
+            C.c_uint64(read_ts), C.byref(r))
+        if st != 0:
+            raise RuntimeError("copr_region_create_mvcc_with_default: %d (%s)" %
+                               (st, self._lib.copr_last_error().decode()))
+        return Region(self, r)
+
     def region_blocks_mvcc(self, blocks, block_offs, n_blocks, read_ts):
         r = C.c_void_p()
         st = self._lib.copr_region_create_blocks_mvcc(
