@@ -1189,3 +1189,71 @@ def test_chunked_grouped_outputs_oracle():
     got = decode_chunks(bytes(d2), ["i64", "i64", "bytes"])
     assert sorted(got, key=lambda t: (t[2] is None, t[2])) == [
         (2, 6, b"aa"), (1, 8, b"zz"), (1, 7, None)]
+
+
+# ---- Real comparers (impl_compare.rs:66-160 Real path) -----------------
+def test_real_filter_oracle():
+    orc = _orc()
+    xs = [1.5, -2.25, None, 10.0, 0.5, 3.0]
+    k, ko, v, vo, n, keep = region_real(xs)
+    cols = [tikv_amd.Col(1), tikv_amd.Col(2, tp=F.TP_DOUBLE)]
+    sel = (tikv_amd.Expr().col(1).const_real(1.0)
+           .func(F.SIG_GT_REAL, 2))
+    req = (tikv_amd.DagSelect(cols).where(sel)
+           .simple_agg([tikv_amd.count_star()]).build())
+    data, nrows = orc.dag_run(req, k, ko, v, vo, n)
+    assert nrows == 1
+    assert data[0] == 3 and int.from_bytes(data[1:9], "big") ^ (1 << 63) == 3
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("sig,name", [
+    (F.SIG_LT_REAL, "lt"), (F.SIG_LE_REAL, "le"), (F.SIG_GT_REAL, "gt"),
+    (F.SIG_GE_REAL, "ge"), (F.SIG_EQ_REAL, "eq"), (F.SIG_NE_REAL, "ne")])
+def test_real_filter_gpu_parity(engine, sig, name):
+    """Real comparer fast path on device, vs the oracle, incl. NULLs and a
+    sum over the surviving rows."""
+    import random
+    rng = random.Random(11)
+    xs = [None if rng.random() < 0.05 else rng.uniform(-100.0, 100.0)
+          for _ in range(40000)]
+    xs[17] = 25.5            # exact-match row for EQ
+    k, ko, v, vo, n, keep = region_real(xs)
+    orc = _orc()
+    rgn = engine.region_raw(k, ko, v, vo, n)
+    try:
+        cols = [tikv_amd.Col(1), tikv_amd.Col(2, tp=F.TP_DOUBLE)]
+        sel = tikv_amd.Expr().col(1).const_real(25.5).func(sig, 2)
+        req = (tikv_amd.DagSelect(cols).where(sel)
+               .simple_agg([tikv_amd.count_star(),
+                            tikv_amd.sum_real(1)]).build())
+        gd, gr, _ = engine.dag_run(req, [rgn])
+        od, orows = orc.dag_run(req, k, ko, v, vo, n)
+        assert orows == gr == 1
+        # count bit-exact; f64 sum within the 1-ULP-class budget
+        assert od[0:9] == gd[0:9]
+        os_, _ = dec_real(od, 9)
+        gs_, _ = dec_real(gd, 9)
+        assert os_ == gs_ or abs(os_ - gs_) <= 1e-9 * max(1.0, abs(os_))
+    finally:
+        rgn.close()
+
+
+@pytest.mark.gpu
+def test_real_filter_project_gpu(engine):
+    """project mode with a Real predicate: the filter column outputs in
+    DECODED datum form (lazy_column.rs:165,242)."""
+    xs = [1.5, -2.25, None, 10.0, 0.5]
+    k, ko, v, vo, n, keep = region_real(xs)
+    orc = _orc()
+    rgn = engine.region_raw(k, ko, v, vo, n)
+    try:
+        cols = [tikv_amd.Col(1), tikv_amd.Col(2, tp=F.TP_DOUBLE)]
+        sel = tikv_amd.Expr().col(1).const_real(0.0).func(F.SIG_GE_REAL, 2)
+        req = tikv_amd.DagSelect(cols).where(sel).build()
+        gd, gr, _ = engine.dag_run(req, [rgn])
+        od, orows = orc.dag_run(req, k, ko, v, vo, n)
+        assert orows == gr == 3
+        assert od == gd
+    finally:
+        rgn.close()

@@ -580,6 +580,7 @@ static copr_status match_filter(const CoprExpr &cond, HostPlan &pl, ScanPlan *sp
   const CoprExprNode &a = cond.nodes[0], &b = cond.nodes[1], &f = cond.nodes[2];
   if (f.kind != COPR_EXPR_SCALAR_FUNC || f.n_args != 2) return COPR_ERR_UNSUPPORTED;
   int cmp;
+  bool is_real = false;
   switch (f.sig) {
     case COPR_SIG_LT_INT: cmp = CMP_LT; break;
     case COPR_SIG_LE_INT: cmp = CMP_LE; break;
@@ -587,17 +588,25 @@ static copr_status match_filter(const CoprExpr &cond, HostPlan &pl, ScanPlan *sp
     case COPR_SIG_GE_INT: cmp = CMP_GE; break;
     case COPR_SIG_EQ_INT: cmp = CMP_EQ; break;
     case COPR_SIG_NE_INT: cmp = CMP_NE; break;
+    /* Real comparers (impl_compare.rs:66-160 Real path) */
+    case COPR_SIG_LT_REAL: cmp = CMP_LT; is_real = true; break;
+    case COPR_SIG_LE_REAL: cmp = CMP_LE; is_real = true; break;
+    case COPR_SIG_GT_REAL: cmp = CMP_GT; is_real = true; break;
+    case COPR_SIG_GE_REAL: cmp = CMP_GE; is_real = true; break;
+    case COPR_SIG_EQ_REAL: cmp = CMP_EQ; is_real = true; break;
+    case COPR_SIG_NE_REAL: cmp = CMP_NE; is_real = true; break;
     default: return COPR_ERR_UNSUPPORTED;
   }
+  auto const_ok = [&](const CoprExprNode &n) {
+    if (n.kind == COPR_EXPR_CONST_NULL) return true;
+    if (is_real) return n.kind == COPR_EXPR_CONST_REAL;
+    return n.kind == COPR_EXPR_CONST_INT || n.kind == COPR_EXPR_CONST_UINT;
+  };
   const CoprExprNode *colref, *konst;
   bool swapped;
-  if (a.kind == COPR_EXPR_COLUMN_REF &&
-      (b.kind == COPR_EXPR_CONST_INT || b.kind == COPR_EXPR_CONST_UINT ||
-       b.kind == COPR_EXPR_CONST_NULL)) {
+  if (a.kind == COPR_EXPR_COLUMN_REF && const_ok(b)) {
     colref = &a; konst = &b; swapped = false;
-  } else if (b.kind == COPR_EXPR_COLUMN_REF &&
-             (a.kind == COPR_EXPR_CONST_INT || a.kind == COPR_EXPR_CONST_UINT ||
-              a.kind == COPR_EXPR_CONST_NULL)) {
+  } else if (b.kind == COPR_EXPR_COLUMN_REF && const_ok(a)) {
     colref = &b; konst = &a; swapped = true;
   } else {
     return COPR_ERR_UNSUPPORTED;
@@ -605,8 +614,14 @@ static copr_status match_filter(const CoprExpr &cond, HostPlan &pl, ScanPlan *sp
   size_t off = (size_t)colref->i64_val;
   if (off >= pl.cols.size()) return COPR_ERR_INVALID_REQUEST;
   const CoprColumnInfo &ci = pl.cols[off];
-  if (!et_int(ci.ft.tp)) return COPR_ERR_UNSUPPORTED;
+  if (is_real) {
+    if (ci.ft.tp != COPR_TP_DOUBLE && ci.ft.tp != COPR_TP_FLOAT)
+      return COPR_ERR_UNSUPPORTED;
+  } else if (!et_int(ci.ft.tp)) {
+    return COPR_ERR_UNSUPPORTED;
+  }
   if (ci.pk_handle && !pl.sp.index_mode) return COPR_ERR_UNSUPPORTED;
+  if (is_real && pl.sp.index_mode) return COPR_ERR_UNSUPPORTED;
   pl.filter_col_offset = (int)off;
   if (swapped) {
     /* const OP col  ==  col flip(OP) const */
@@ -620,21 +635,39 @@ static copr_status match_filter(const CoprExpr &cond, HostPlan &pl, ScanPlan *sp
   }
   sp->has_filter = 1;
   sp->filter_col_id = ci.column_id;
+  sp->filter_is_real = is_real ? 1 : 0;
   /* default fill for a row missing this column (scan fills defaults before
      the predicate decodes the column) */
   if (ci.default_val && ci.default_val_len) {
-    int64_t dv;
-    int r = host_decode_int_datum(ci.default_val, ci.default_val_len, &dv);
-    if (r < 0) return COPR_ERR_INVALID_REQUEST;
-    sp->filter_missing_null = r == 1 ? 1 : 0;
-    sp->filter_missing_val = dv;
+    if (is_real) {
+      /* FLOAT datum default: flag 5 + 8B comparable f64 -> IEEE bits */
+      if (ci.default_val_len != 9 || ci.default_val[0] != 5)
+        return COPR_ERR_INVALID_REQUEST;
+      uint64_t u = 0;
+      for (int b2 = 0; b2 < 8; b2++) u = (u << 8) | ci.default_val[1 + b2];
+      sp->filter_missing_val =
+          (int64_t)((u & 0x8000000000000000ull) ? (u ^ 0x8000000000000000ull)
+                                                : ~u);
+      sp->filter_missing_null = 0;
+    } else {
+      int64_t dv;
+      int r = host_decode_int_datum(ci.default_val, ci.default_val_len, &dv);
+      if (r < 0) return COPR_ERR_INVALID_REQUEST;
+      sp->filter_missing_null = r == 1 ? 1 : 0;
+      sp->filter_missing_val = dv;
+    }
   } else {
     /* NULLable: missing -> NULL. (A missing NOT NULL column is a data
        error in the reference; rows cannot legally lack it.) */
     sp->filter_missing_null = 1;
   }
   sp->filter_cmp = cmp;
-  sp->filter_const = konst->i64_val;
+  if (is_real && konst->kind == COPR_EXPR_CONST_REAL) {
+    double dv = konst->f64_val;
+    memcpy(&sp->filter_const, &dv, 8);           /* f64 BITS */
+  } else {
+    sp->filter_const = konst->i64_val;
+  }
   sp->filter_col_unsigned = (ci.ft.flag & COPR_FLAG_UNSIGNED) ? 1 : 0;
   sp->filter_const_unsigned =
       (konst->kind == COPR_EXPR_CONST_UINT || (konst->ft.flag & COPR_FLAG_UNSIGNED)) ? 1 : 0;
@@ -889,6 +922,7 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
           }
           sp.filter2_on = 1;
           sp.filter2_col_id = s2.filter_col_id;
+          sp.filter2_is_real = s2.filter_is_real;
           sp.filter2_cmp = s2.filter_cmp;
           sp.filter2_const = s2.filter_const;
           sp.filter2_col_unsigned = s2.filter_col_unsigned;
@@ -2148,9 +2182,13 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
           if ((int)off == pl.dec2_col_offset && pl.sp.dec2_col_id) {
             uint8_t st8 = h_d2state[i];
             bool uns = (pl.cols[off].ft.flag & COPR_FLAG_UNSIGNED) != 0;
-            if (st8 == 0) enc_datum_int(&resp, h_d2vals[i], uns);
-            else if (st8 == 1) resp.push_back(0);
+            bool fr = pl.sp.filter2_is_real != 0;
+            if (st8 == 0) {
+              if (fr) enc_datum_real(&resp, (uint64_t)h_d2vals[i]);
+              else enc_datum_int(&resp, h_d2vals[i], uns);
+            } else if (st8 == 1) resp.push_back(0);
             else if (pl.sp.dec2_missing_null) resp.push_back(0);
+            else if (fr) enc_datum_real(&resp, (uint64_t)pl.sp.dec2_missing_val);
             else enc_datum_int(&resp, pl.sp.dec2_missing_val, uns);
             continue;
           }
@@ -2160,9 +2198,13 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
                Decoded encode, lazy_column.rs:165,242; vector.rs:372-383) */
             uint8_t st8 = h_fstate[i];
             bool uns = (pl.cols[off].ft.flag & COPR_FLAG_UNSIGNED) != 0;
-            if (st8 == 0) enc_datum_int(&resp, h_fvals[i], uns);
-            else if (st8 == 1) resp.push_back(0);
+            bool fr = pl.sp.filter_is_real != 0;
+            if (st8 == 0) {
+              if (fr) enc_datum_real(&resp, (uint64_t)h_fvals[i]);
+              else enc_datum_int(&resp, h_fvals[i], uns);
+            } else if (st8 == 1) resp.push_back(0);
             else if (pl.sp.filter_missing_null) resp.push_back(0);
+            else if (fr) enc_datum_real(&resp, (uint64_t)pl.sp.filter_missing_val);
             else enc_datum_int(&resp, pl.sp.filter_missing_val, uns);
             continue;
           }

@@ -99,10 +99,14 @@ struct ScanPlan {
   int32_t has_filter;
   int64_t filter_col_id;
   int32_t filter_cmp;
-  int64_t filter_const;
+  int64_t filter_const;          /* f64 BITS when filter_is_real */
   int32_t filter_col_unsigned;
   int32_t filter_const_unsigned;
   int32_t filter_const_null;     /* NULL const: predicate never true */
+  /* Real comparers (impl_compare.rs:66-160 Real path): the filter channel
+     carries f64 BITS through the i64 value slots */
+  int32_t filter_is_real;
+  int32_t filter2_is_real;
   /* value a row takes when the filter column is absent: the scan default
      fill (table_scan_executor.rs:456-483) decoded */
   int32_t filter_missing_null;   /* 1 => NULL */

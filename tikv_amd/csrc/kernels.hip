@@ -103,7 +103,9 @@ __device__ static inline bool d_var_u64(const uint8_t *p, uint32_t rem,
     uint64_t m = lds_win8(p);
     uint64_t stops = ~m & 0x8080808080808080ull;
     if (stops) {
-      uint32_t nb = ((uint32_t)__ffsll((long long)stops) >> 3) + 1u; /* 1..8 */
+      /* stop at byte k sets bit 8k+7; __ffsll is 1-based -> 8k+8; >>3 = k+1
+         = the varint's byte count */
+      uint32_t nb = (uint32_t)__ffsll((long long)stops) >> 3;   /* 1..8 */
       uint64_t val = (m & 0x7f) | ((m >> 8) & 0x7f) << 7 |
                      ((m >> 16) & 0x7f) << 14 | ((m >> 24) & 0x7f) << 21 |
                      ((m >> 32) & 0x7f) << 28 | ((m >> 40) & 0x7f) << 35 |
@@ -524,13 +526,24 @@ __device__ static inline bool d_v2_collect(const ScanPlan &plan,
                                            int64_t *f2_v = nullptr) {
   V2Row r;
   if (!d_v2_parse(vp, vlen, &r)) return false;
+  /* v2 Float/Double cells keep the v1 FLOAT payload (compat_v1.rs:68-72):
+     8-byte BE comparable f64 -> IEEE bits for the real channels */
+  auto v2_real_bits = [&](uint32_t s, uint32_t e, int64_t *out) -> bool {
+    if (e - s != 8) return false;
+    uint64_t u = d_be_u64(r.vals + s);
+    *out = (int64_t)((u & 0x8000000000000000ull)
+                         ? (u ^ 0x8000000000000000ull) : ~u);
+    return true;
+  };
   if (plan.has_filter) {
     uint32_t s, e;
     int st = d_v2_find(r, plan.filter_col_id, &s, &e);
     if (st >= 0) {
       *filt_found = true;
       if (st == 0) *filt_null = true;
-      else if (!d_v2_int(r.vals + s, e - s, plan.filter_col_unsigned, filt_v))
+      else if (plan.filter_is_real) {
+        if (!v2_real_bits(s, e, filt_v)) return false;
+      } else if (!d_v2_int(r.vals + s, e - s, plan.filter_col_unsigned, filt_v))
         return false;
     }
   }
@@ -540,7 +553,9 @@ __device__ static inline bool d_v2_collect(const ScanPlan &plan,
     if (st >= 0) {
       *f2_found = true;
       if (st == 0) *f2_null = true;
-      else if (!d_v2_int(r.vals + s, e - s, plan.filter2_col_unsigned, f2_v))
+      else if (plan.filter2_is_real) {
+        if (!v2_real_bits(s, e, f2_v)) return false;
+      } else if (!d_v2_int(r.vals + s, e - s, plan.filter2_col_unsigned, f2_v))
         return false;
     }
   }
@@ -567,6 +582,14 @@ __device__ static inline bool d_v2_collect(const ScanPlan &plan,
       int64_t sc; int32_t fr;
       if (!d_decimal_scaled(r.vals + s, e - s, &sc, &fr)) return false;
       cols[a].has_dec = true; cols[a].dsc = sc; cols[a].dfr = fr;
+    } else if (sp.kind == DAGG_SUM_REAL || sp.kind == DAGG_MAX_REAL ||
+               sp.kind == DAGG_MIN_REAL) {
+      /* v2 Float/Double cell keeps the v1 FLOAT datum payload verbatim
+         (compat_v1.rs:68-72): 8-byte BE comparable f64 -> IEEE bits */
+      if (e - s != 8) return false;
+      uint64_t u = d_be_u64(r.vals + s);
+      cols[a].iv = (int64_t)((u & 0x8000000000000000ull)
+                                 ? (u ^ 0x8000000000000000ull) : ~u);
     } else if (sp.kind == DAGG_SUM_INT || d_is_fold(sp.kind)) {
       if (!d_v2_int(r.vals + s, e - s, sp.col_unsigned, &cols[a].iv)) return false;
     }
@@ -788,6 +811,15 @@ __device__ static inline bool d_cmp_res(int32_t kind, int ord) {
   }
 }
 
+/* Real comparer (impl_compare.rs:66-160 Real path): values travel as f64
+ * BITS in the i64 channels; Real is guaranteed non-NaN in the reference
+ * (Real::new rejects NaN), so plain f64 ordering is total */
+__device__ static inline int d_cmp_real(int64_t lbits, int64_t rbits) {
+  double l = __longlong_as_double((long long)lbits);
+  double r = __longlong_as_double((long long)rbits);
+  return l < r ? -1 : l > r ? 1 : 0;
+}
+
 /* evaluate the plan's filter for one row's decoded filter-column state */
 __device__ static inline bool d_filter_keep(const ScanPlan &plan, bool found,
                                             bool is_null, int64_t v) {
@@ -799,8 +831,11 @@ __device__ static inline bool d_filter_keep(const ScanPlan &plan, bool found,
   }
   if (plan.filter_const_null || is_null) return false;
   return d_cmp_res(plan.filter_cmp,
-                   d_cmp_int(v, plan.filter_const,
-                             plan.filter_col_unsigned, plan.filter_const_unsigned));
+                   plan.filter_is_real
+                       ? d_cmp_real(v, plan.filter_const)
+                       : d_cmp_int(v, plan.filter_const,
+                                   plan.filter_col_unsigned,
+                                   plan.filter_const_unsigned));
 }
 
 
@@ -814,8 +849,11 @@ __device__ static inline bool d_filter2_keep(const ScanPlan &plan,
   }
   if (plan.filter2_const_null || is_null) return false;
   return d_cmp_res(plan.filter2_cmp,
-                   d_cmp_int(v, plan.filter2_const, plan.filter2_col_unsigned,
-                             plan.filter2_const_unsigned));
+                   plan.filter2_is_real
+                       ? d_cmp_real(v, plan.filter2_const)
+                       : d_cmp_int(v, plan.filter2_const,
+                                   plan.filter2_col_unsigned,
+                                   plan.filter2_const_unsigned));
 }
 
 /* evaluate the plan's RPN predicate over the two captured columns.
@@ -1122,7 +1160,7 @@ k_scan_agg(ScanPlan plan,
                   cid == plan.filter_col_id) {
                 filt_found = true;
                 if (cell.is_null) filt_null = true;
-                else if (cell.has_int) filt_v = cell.ival;
+                else if (plan.filter_is_real ? cell.has_real : cell.has_int) filt_v = cell.ival;
                 else parse_ok = false;
               } else parse_ok = false;
             }
@@ -1132,7 +1170,7 @@ k_scan_agg(ScanPlan plan,
                   cid == plan.filter2_col_id) {
                 f2_found = true;
                 if (cell.is_null) f2_null = true;
-                else if (cell.has_int) f2_v = cell.ival;
+                else if (plan.filter2_is_real ? cell.has_real : cell.has_int) f2_v = cell.ival;
                 else parse_ok = false;
               } else parse_ok = false;
             }
@@ -1184,14 +1222,14 @@ k_scan_agg(ScanPlan plan,
           if (plan.has_filter && !filt_found && cell_id == plan.filter_col_id) {
             filt_found = true;
             if (cell.is_null) filt_null = true;
-            else if (cell.has_int) filt_v = cell.ival;
+            else if (plan.filter_is_real ? cell.has_real : cell.has_int) filt_v = cell.ival;
             else parse_ok = false;
             found++;
           }
           if (plan.filter2_on && !f2_found && cell_id == plan.filter2_col_id) {
             f2_found = true;
             if (cell.is_null) f2_null = true;
-            else if (cell.has_int) f2_v = cell.ival;
+            else if (plan.filter2_is_real ? cell.has_real : cell.has_int) f2_v = cell.ival;
             else parse_ok = false;
             found++;
           }
@@ -1696,7 +1734,7 @@ k_scan_agg_pipe(ScanPlan plan,
               if (!next_cell(vp, vlen, &pos, &cid, &cell_off, &cell)) { ok = false; break; }
               if (cid == FCID) {
                 if (cell.is_null) fnull = true;
-                else if (cell.has_int) fv = cell.ival;
+                else if (plan.filter_is_real ? cell.has_real : cell.has_int) fv = cell.ival;
                 else ok = false;
                 found = true;
                 break;
@@ -1772,7 +1810,7 @@ k_scan_agg_pipe(ScanPlan plan,
                   cid == plan.filter_col_id) {
                 filt_found = true;
                 if (cell.is_null) filt_null = true;
-                else if (cell.has_int) filt_v = cell.ival;
+                else if (plan.filter_is_real ? cell.has_real : cell.has_int) filt_v = cell.ival;
                 else parse_ok = false;
               } else parse_ok = false;
             }
@@ -1782,7 +1820,7 @@ k_scan_agg_pipe(ScanPlan plan,
                   cid == plan.filter2_col_id) {
                 f2_found = true;
                 if (cell.is_null) f2_null = true;
-                else if (cell.has_int) f2_v = cell.ival;
+                else if (plan.filter2_is_real ? cell.has_real : cell.has_int) f2_v = cell.ival;
                 else parse_ok = false;
               } else parse_ok = false;
             }
@@ -1834,14 +1872,14 @@ k_scan_agg_pipe(ScanPlan plan,
           if (plan.has_filter && !filt_found && cell_id == plan.filter_col_id) {
             filt_found = true;
             if (cell.is_null) filt_null = true;
-            else if (cell.has_int) filt_v = cell.ival;
+            else if (plan.filter_is_real ? cell.has_real : cell.has_int) filt_v = cell.ival;
             else parse_ok = false;
             found++;
           }
           if (plan.filter2_on && !f2_found && cell_id == plan.filter2_col_id) {
             f2_found = true;
             if (cell.is_null) f2_null = true;
-            else if (cell.has_int) f2_v = cell.ival;
+            else if (plan.filter2_is_real ? cell.has_real : cell.has_int) f2_v = cell.ival;
             else parse_ok = false;
             found++;
           }
@@ -2266,7 +2304,7 @@ k_scan_fc_pipe3(ScanPlan plan,
             if (!next_cell(vp, vlen, &pos, &cid, &cell_off, &cell)) { ok = false; break; }
             if (cid == FCID) {
               if (cell.is_null) fnull = true;
-              else if (cell.has_int) fv = cell.ival;
+              else if (plan.filter_is_real ? cell.has_real : cell.has_int) fv = cell.ival;
               else ok = false;
               found = true;
               break;
@@ -2594,7 +2632,7 @@ k_scan_fc_ring(ScanPlan plan,
               if (!next_cell(vp, vlen, &posb, &cid, &cell_off, &cell)) { ok = false; break; }
               if (cid == FCID) {
                 if (cell.is_null) fnull = true;
-                else if (cell.has_int) fv = cell.ival;
+                else if (plan.filter_is_real ? cell.has_real : cell.has_int) fv = cell.ival;
                 else ok = false;
                 found = true;
                 break;
@@ -2685,7 +2723,7 @@ k_scan_fc_direct(ScanPlan plan,
               if (!next_cell(vp, vlen, &pos, &cid, &cell_off, &cell)) { ok = false; break; }
               if (cid == FCID) {
                 if (cell.is_null) fnull = true;
-                else if (cell.has_int) fv = cell.ival;
+                else if (plan.filter_is_real ? cell.has_real : cell.has_int) fv = cell.ival;
                 else ok = false;
                 found = true;
                 break;
@@ -2740,7 +2778,7 @@ k_scan_fc_direct(ScanPlan plan,
             if (!next_cell(vp, vlen, &pos, &cid, &cell_off, &cell)) { ok = false; break; }
             if (cid == FCID) {
               if (cell.is_null) fnull = true;
-              else if (cell.has_int) fv = cell.ival;
+              else if (plan.filter_is_real ? cell.has_real : cell.has_int) fv = cell.ival;
               else ok = false;
               found = true;
               break;
@@ -2842,7 +2880,7 @@ k_scan_project(ScanPlan plan,
         if (plan.has_filter && !filt_found && cell_id == plan.filter_col_id) {
           filt_found = true;
           if (cell.is_null) filt_null = true;
-          else if (cell.has_int) filt_v = cell.ival;
+          else if (plan.filter_is_real ? cell.has_real : cell.has_int) filt_v = cell.ival;
           else parse_ok = false;
           found++;
         }
@@ -3480,7 +3518,7 @@ static int launch_agg_pipe(const ScanPlan &plan, const DevRegion &rgn,
                            hipStream_t s, uint32_t grid) {
   if (!IS_HASH && !plan.index_mode && plan.n_aggs == 1 &&
       plan.aggs[0].kind == DAGG_COUNT_ROWS && plan.has_filter &&
-      !plan.filter2_on && !plan.rpn_on) {
+      !plan.filter2_on && !plan.rpn_on && !plan.filter_is_real) {
     if (getenv("COPR_DIRECT")) {
       uint64_t n_blk = (rgn.n_kv + THREADS - 1) / THREADS;
       uint32_t dgrid = (uint32_t)(n_blk < 8192 ? n_blk : 8192);
@@ -3673,7 +3711,7 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
                   cid == plan.filter_col_id) {
                 filt_found = true;
                 if (cell.is_null) filt_null = true;
-                else if (cell.has_int) filt_v = cell.ival;
+                else if (plan.filter_is_real ? cell.has_real : cell.has_int) filt_v = cell.ival;
                 else parse_ok = false;
               } else parse_ok = false;
             }
@@ -3683,7 +3721,7 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
                   cid == plan.filter2_col_id) {
                 f2_found = true;
                 if (cell.is_null) f2_null = true;
-                else if (cell.has_int) f2_v = cell.ival;
+                else if (plan.filter2_is_real ? cell.has_real : cell.has_int) f2_v = cell.ival;
                 else parse_ok = false;
               } else parse_ok = false;
             }
@@ -3765,7 +3803,7 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
                 cell_id == plan.filter_col_id) {
               filt_found = true;
               if (cell.is_null) filt_null = true;
-              else if (cell.has_int) filt_v = cell.ival;
+              else if (plan.filter_is_real ? cell.has_real : cell.has_int) filt_v = cell.ival;
               else parse_ok = false;
               found++;
             }
@@ -3773,7 +3811,7 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
                 cell_id == plan.filter2_col_id) {
               f2_found = true;
               if (cell.is_null) f2_null = true;
-              else if (cell.has_int) f2_v = cell.ival;
+              else if (plan.filter2_is_real ? cell.has_real : cell.has_int) f2_v = cell.ival;
               else parse_ok = false;
               found++;
             }

@@ -49,6 +49,14 @@ class Expr:
         self.nodes.append(n)
         return self
 
+    def const_real(self, v):
+        n = F.CoprExprNode()
+        n.kind = F.EXPR_CONST_REAL
+        n.f64_val = float(v)
+        n.ft = field_type(F.TP_DOUBLE)
+        self.nodes.append(n)
+        return self
+
     def const_null(self):
         n = F.CoprExprNode()
         n.kind = F.EXPR_CONST_NULL
