@@ -2949,6 +2949,7 @@ k_scan_project(ScanPlan plan,
  * per step and the stage/parse barrier phases serialized on top
  * (profiles/r02_cfg4_*). More independent chains per SIMD is the fix, not
  * more LDS bandwidth. */
+template <bool PREFETCH>
 __device__ static inline uint64_t d_crc64_stream(const uint8_t *__restrict__ base,
                                                  uint64_t b0, uint64_t b1,
                                                  uint64_t crc,
@@ -2973,11 +2974,12 @@ __device__ static inline uint64_t d_crc64_stream(const uint8_t *__restrict__ bas
           tab[1 * 256 + (uint32_t)((crc >> 48) & 0xFF)] ^
           tab[0 * 256 + (uint32_t)(crc >> 56)];
   };
-  if (n8 >= 8) {
+  if (PREFETCH && n8 >= 8) {
     /* software pipeline over 64-byte register chunks: chunk k+1's 8
        independent loads are in flight while the dependent table chain
        consumes chunk k (the unpipelined form measured 54% SQ_WAIT_ANY —
-       one HBM latency exposed per chunk) */
+       one HBM latency exposed per chunk). Costs ~32 VGPRs: pair with
+       __launch_bounds__(256, 4), NOT 8 (at 8 it spills 52 B/lane). */
     uint64_t w[8];
     #pragma unroll
     for (int j = 0; j < 8; j++) w[j] = q[wi + j];
@@ -3004,6 +3006,20 @@ __device__ static inline uint64_t d_crc64_stream(const uint8_t *__restrict__ bas
       prev = w[j];
       step8(cur);
     }
+  } else {
+    while (n8 >= 8) {
+      uint64_t w[8];
+      #pragma unroll
+      for (int j = 0; j < 8; j++) w[j] = q[wi + j];
+      #pragma unroll
+      for (int j = 0; j < 8; j++) {
+        uint64_t cur = sh ? ((prev >> sh) | (w[j] << (64 - sh))) : prev;
+        prev = w[j];
+        step8(cur);
+      }
+      wi += 8;
+      n8 -= 8;
+    }
   }
   while (n8) {
     uint64_t w = q[wi++];
@@ -3024,7 +3040,9 @@ __device__ static inline uint64_t d_crc64_stream(const uint8_t *__restrict__ bas
   return crc;
 }
 
-__global__ void __launch_bounds__(THREADS, 8)
+/* prefetch variant: 4 waves/SIMD (128 VGPRs, no spill), pipeline hides the
+ * per-chunk HBM latency inside the lane */
+__global__ void __launch_bounds__(THREADS, 4)
 k_crc64_reg(const uint8_t *__restrict__ vals,
             const uint64_t *__restrict__ val_offs,
             const uint8_t *__restrict__ keys,
@@ -3039,8 +3057,33 @@ k_crc64_reg(const uint8_t *__restrict__ vals,
   for (uint64_t row = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
        row < n_rows; row += (uint64_t)gridDim.x * blockDim.x) {
     uint64_t crc = ~0ull;
-    crc = d_crc64_stream(keys, key_offs[row], key_offs[row + 1], crc, tab);
-    crc = d_crc64_stream(vals, val_offs[row], val_offs[row + 1], crc, tab);
+    crc = d_crc64_stream<true>(keys, key_offs[row], key_offs[row + 1], crc, tab);
+    crc = d_crc64_stream<true>(vals, val_offs[row], val_offs[row + 1], crc, tab);
+    acc ^= ~crc;
+  }
+  for (int off = 32; off > 0; off >>= 1)
+    acc ^= (unsigned long long)__shfl_down((long long)acc, off, 64);
+  if ((threadIdx.x & 63u) == 0 && acc) atomicXor(out_xor, acc);
+}
+
+/* no-prefetch variant: 8 waves/SIMD of independent chains (COPR_CRC_NP) */
+__global__ void __launch_bounds__(THREADS, 8)
+k_crc64_reg_np(const uint8_t *__restrict__ vals,
+               const uint64_t *__restrict__ val_offs,
+               const uint8_t *__restrict__ keys,
+               const uint64_t *__restrict__ key_offs, uint64_t n_rows,
+               const uint64_t *__restrict__ g_tables,
+               unsigned long long *__restrict__ out_xor) {
+  __shared__ uint64_t tab[8 * 256];
+  for (uint32_t i = threadIdx.x; i < 8 * 256u; i += blockDim.x)
+    tab[i] = g_tables[i];
+  __syncthreads();
+  unsigned long long acc = 0;
+  for (uint64_t row = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       row < n_rows; row += (uint64_t)gridDim.x * blockDim.x) {
+    uint64_t crc = ~0ull;
+    crc = d_crc64_stream<false>(keys, key_offs[row], key_offs[row + 1], crc, tab);
+    crc = d_crc64_stream<false>(vals, val_offs[row], val_offs[row + 1], crc, tab);
     acc ^= ~crc;
   }
   for (int off = 32; off > 0; off >>= 1)
@@ -5151,13 +5194,18 @@ int dev_crc64_launch(const DevRegion &rgn, const uint64_t *d_tables,
                      unsigned long long *d_xor, void *stream) {
   const bool s16 = getenv("COPR_CRC16") != nullptr;
   if (!s16 && !getenv("COPR_CRC_TILE")) {
-    /* default: register-streamed kernel (no staging; 8 blocks/CU) */
+    /* default: register-streamed kernel (no staging) */
     uint64_t n_blk = (rgn.n_kv + THREADS - 1) / THREADS;
     uint32_t grid = (uint32_t)(n_blk < 8192 ? n_blk : 8192);
     if (grid == 0) grid = 1;
-    hipLaunchKernelGGL(k_crc64_reg, dim3(grid), dim3(THREADS), 0,
-                       (hipStream_t)stream, rgn.d_vals, rgn.d_val_offs,
-                       rgn.d_keys, rgn.d_key_offs, rgn.n_kv, d_tables, d_xor);
+    if (getenv("COPR_CRC_NP"))
+      hipLaunchKernelGGL(k_crc64_reg_np, dim3(grid), dim3(THREADS), 0,
+                         (hipStream_t)stream, rgn.d_vals, rgn.d_val_offs,
+                         rgn.d_keys, rgn.d_key_offs, rgn.n_kv, d_tables, d_xor);
+    else
+      hipLaunchKernelGGL(k_crc64_reg, dim3(grid), dim3(THREADS), 0,
+                         (hipStream_t)stream, rgn.d_vals, rgn.d_val_offs,
+                         rgn.d_keys, rgn.d_key_offs, rgn.n_kv, d_tables, d_xor);
     return (int)hipGetLastError();
   }
   const uint32_t tab_b = (s16 ? 16u : 8u) * 256u * 8u;
