@@ -157,6 +157,10 @@ struct ScanPlan {
      handle datum or whose value is new-format. */
   const uint8_t *aux_vals;
   const uint64_t *aux_val_offs;
+  /* max value length in the region: <= 9 means every value is old-format
+     (no restore data possible), so the value is only read for rows whose
+     key carries no trailing handle */
+  uint32_t aux_max_vlen;
   /* index position -> the reference column_id (restore-data row lookups,
      RestoreData::V4: extract_columns_from_row_format :483-501) */
   int64_t index_real_ids[COPR_MAX_OUT_COLS];

@@ -676,23 +676,25 @@ __device__ static inline bool d_index_collect(const ScanPlan &plan,
                                               int64_t *handle_out,
                                               bool *handle_found) {
   if (vlen < 19 || vp[0] != 't' || vp[9] != '_' || vp[10] != 'i') return false;
-  /* the VALUE decides the layout; load it lazily (non-unique old-format
-     rows keep the handle in the key and usually need no value read, but
-     new-format non-unique values carry restore data that overrides the key
-     columns, so the split must run before the key walk) */
-  const uint8_t *ival = nullptr;
-  uint32_t ivlen = 0;
+  /* The VALUE decides the layout. New-format values (len > 9) can carry
+     restore data that OVERRIDES the key columns, so they must be split
+     before the key walk — but only regions that contain such values pay
+     for it (aux_max_vlen > 9). Old-format regions read the value lazily,
+     only for rows whose key carries no trailing handle. */
   int64_t vhandle = 0;
   bool v_has_handle = false;
   const uint8_t *restore = nullptr;
   uint32_t restore_len = 0;
-  if (plan.aux_vals) {
+  bool value_split_done = false;
+  auto split_value = [&]() -> bool {
     uint64_t o0 = plan.aux_val_offs[my_row], o1 = plan.aux_val_offs[my_row + 1];
-    ival = plan.aux_vals + o0;
-    ivlen = (uint32_t)(o1 - o0);
-    if (!d_index_value_split(ival, ivlen, &vhandle, &v_has_handle,
-                             &restore, &restore_len))
-      return false;
+    value_split_done = true;
+    return d_index_value_split(plan.aux_vals + o0, (uint32_t)(o1 - o0),
+                               &vhandle, &v_has_handle, &restore,
+                               &restore_len);
+  };
+  if (plan.aux_vals && plan.aux_max_vlen > 9) {
+    if (!split_value()) return false;
   }
   *handle_found = false;
 
@@ -779,12 +781,15 @@ __device__ static inline bool d_index_collect(const ScanPlan &plan,
     }
   }
 
-  if (!*handle_found && v_has_handle) {
-    /* unique index: PK int handle lives in the value (:553-562) */
-    CellView cell{};
-    cell.has_int = true;
-    cell.ival = vhandle;
-    if (!match_cell((int64_t)plan.index_n_cols, cell)) return false;
+  if (!*handle_found) {
+    if (!value_split_done && plan.aux_vals && !split_value()) return false;
+    if (v_has_handle) {
+      /* unique index: PK int handle lives in the value (:553-562) */
+      CellView cell{};
+      cell.has_int = true;
+      cell.ival = vhandle;
+      if (!match_cell((int64_t)plan.index_n_cols, cell)) return false;
+    }
   }
   return true;
 }
@@ -5179,6 +5184,7 @@ int dev_scan_launch(const ScanPlan &plan, const DevRegion &rgn,
     r2.max_row_bytes = rgn.max_key_bytes;
     p2.aux_vals = rgn.d_vals;
     p2.aux_val_offs = rgn.d_val_offs;
+    p2.aux_max_vlen = rgn.max_row_bytes;
   }
   if (plan.mode == 1)
     return launch_agg<false>(p2, r2, d_simple, HashAggTable{}, s, grid);
