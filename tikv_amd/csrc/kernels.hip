@@ -1193,6 +1193,23 @@ k_scan_agg(ScanPlan plan,
             for (int a = 0; a < NAGGS; a++) {
               if (plan.aggs[a].kind == DAGG_COUNT_ROWS || d_a[a] == 0xFFu)
                 continue;
+              if (IS_HASH && grp_found && !grp_null && parse_ok &&
+                  plan.aggs[a].col_id == plan.group_col_id &&
+                  plan.aggs[a].kind != DAGG_SUM_DEC &&
+                  plan.aggs[a].kind != DAGG_SUM_REAL) {
+                /* agg over the GROUP column: reuse the parsed int value
+                   (avg/sum BY the same column parses the cell once) */
+                cols[a].found = true;
+                cols[a].null = false;
+                cols[a].iv = grp_v;
+                continue;
+              }
+              if (IS_HASH && grp_found && grp_null && parse_ok &&
+                  plan.aggs[a].col_id == plan.group_col_id) {
+                cols[a].found = true;
+                cols[a].null = true;
+                continue;
+              }
               pos = d_a[a];
               if (next_cell(vp, vlen, &pos, &cid, &coff, &cell) &&
                   cid == plan.aggs[a].col_id) {
@@ -1843,6 +1860,23 @@ k_scan_agg_pipe(ScanPlan plan,
             for (int a = 0; a < NAGGS; a++) {
               if (plan.aggs[a].kind == DAGG_COUNT_ROWS || d_a[a] == 0xFFu)
                 continue;
+              if (IS_HASH && grp_found && !grp_null && parse_ok &&
+                  plan.aggs[a].col_id == plan.group_col_id &&
+                  plan.aggs[a].kind != DAGG_SUM_DEC &&
+                  plan.aggs[a].kind != DAGG_SUM_REAL) {
+                /* agg over the GROUP column: reuse the parsed int value
+                   (avg/sum BY the same column parses the cell once) */
+                cols[a].found = true;
+                cols[a].null = false;
+                cols[a].iv = grp_v;
+                continue;
+              }
+              if (IS_HASH && grp_found && grp_null && parse_ok &&
+                  plan.aggs[a].col_id == plan.group_col_id) {
+                cols[a].found = true;
+                cols[a].null = true;
+                continue;
+              }
               pos = d_a[a];
               if (next_cell(vp, vlen, &pos, &cid, &coff, &cell) &&
                   cid == plan.aggs[a].col_id) {
@@ -3054,6 +3088,32 @@ k_crc64_reg(const uint8_t *__restrict__ vals,
             const uint64_t *__restrict__ key_offs, uint64_t n_rows,
             const uint64_t *__restrict__ g_tables,
             unsigned long long *__restrict__ out_xor) {
+  __shared__ uint64_t tab[8 * 256];
+  for (uint32_t i = threadIdx.x; i < 8 * 256u; i += blockDim.x)
+    tab[i] = g_tables[i];
+  __syncthreads();
+  unsigned long long acc = 0;
+  for (uint64_t row = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       row < n_rows; row += (uint64_t)gridDim.x * blockDim.x) {
+    uint64_t crc = ~0ull;
+    crc = d_crc64_stream<true>(keys, key_offs[row], key_offs[row + 1], crc, tab);
+    crc = d_crc64_stream<true>(vals, val_offs[row], val_offs[row + 1], crc, tab);
+    acc ^= ~crc;
+  }
+  for (int off = 32; off > 0; off >>= 1)
+    acc ^= (unsigned long long)__shfl_down((long long)acc, off, 64);
+  if ((threadIdx.x & 63u) == 0 && acc) atomicXor(out_xor, acc);
+}
+
+/* prefetch at 6 waves/SIMD (80 VGPRs; COPR_CRC_PF6): more chains in
+ * flight if the pipeline fits the tighter budget */
+__global__ void __launch_bounds__(THREADS, 6)
+k_crc64_reg_pf6(const uint8_t *__restrict__ vals,
+                const uint64_t *__restrict__ val_offs,
+                const uint8_t *__restrict__ keys,
+                const uint64_t *__restrict__ key_offs, uint64_t n_rows,
+                const uint64_t *__restrict__ g_tables,
+                unsigned long long *__restrict__ out_xor) {
   __shared__ uint64_t tab[8 * 256];
   for (uint32_t i = threadIdx.x; i < 8 * 256u; i += blockDim.x)
     tab[i] = g_tables[i];
@@ -5206,6 +5266,10 @@ int dev_crc64_launch(const DevRegion &rgn, const uint64_t *d_tables,
     if (grid == 0) grid = 1;
     if (getenv("COPR_CRC_NP"))
       hipLaunchKernelGGL(k_crc64_reg_np, dim3(grid), dim3(THREADS), 0,
+                         (hipStream_t)stream, rgn.d_vals, rgn.d_val_offs,
+                         rgn.d_keys, rgn.d_key_offs, rgn.n_kv, d_tables, d_xor);
+    else if (getenv("COPR_CRC_PF6"))
+      hipLaunchKernelGGL(k_crc64_reg_pf6, dim3(grid), dim3(THREADS), 0,
                          (hipStream_t)stream, rgn.d_vals, rgn.d_val_offs,
                          rgn.d_keys, rgn.d_key_offs, rgn.n_kv, d_tables, d_xor);
     else
