@@ -226,11 +226,16 @@ struct SimpleAggAcc {
 /* hash-agg table (device): parallel arrays.
  * keys[]: EMPTY sentinel = INT64_MIN bias — the real INT64_MIN key and the
  * NULL key get dedicated accumulator blocks (reserved[0]=int64_min,
- * reserved[1]=null). */
+ * reserved[1]=null).
+ * ext/rsvd_ext: the two HIGH limbs extending each acc's sum to 256 bits
+ * (wide Decimal sums, decimal.rs:927-942 word_buf range; values up to 38
+ * digits, sums up to ~77). Narrow adds ripple carries in lazily. */
 struct HashAggTable {
   long long *keys;               /* [table_size] */
   SimpleAggAcc *accs;            /* [table_size * n_aggs] */
   SimpleAggAcc *reserved;        /* [2 * n_aggs] */
+  unsigned long long *ext;       /* [table_size * n_aggs * 2] or null */
+  unsigned long long *rsvd_ext;  /* [2 * n_aggs * 2] or null */
   unsigned long long *rsvd_seen; /* [2]: row counts for the 2 reserved keys */
   unsigned int *error;           /* [0]=table full, [1]=parse error */
   unsigned long long *n_groups;  /* occupied slot count */

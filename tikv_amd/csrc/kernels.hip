@@ -197,6 +197,68 @@ __device__ static inline uint32_t d_decimal_scaled(const uint8_t *p, uint32_t re
   return need;
 }
 
+/* wide decimal payload -> scaled i128 (prec <= 38); the same read_decimal
+ * restatement as d_decimal_scaled with 128-bit accumulation
+ * (decimal.rs:2204-2289). Returns consumed bytes, 0 = malformed/too wide. */
+__device__ static inline uint32_t d_decimal_scaled128(const uint8_t *p,
+                                                      uint32_t rem,
+                                                      __int128 *scaled,
+                                                      int32_t *frac) {
+  if (rem < 3) return 0;
+  uint32_t prec = p[0], fr = p[1];
+  if (prec < fr || prec > 38) return 0;
+  uint32_t int_cnt = prec - fr;
+  uint32_t iw = int_cnt / 9, ld = int_cnt - iw * 9;
+  uint32_t fw = fr / 9, td = fr - fw * 9;
+  uint32_t need = 2 + iw * 4 + DIG2B[ld] + fw * 4 + DIG2B[td];
+  if (rem < need) return 0;
+  const uint8_t *q = p + 2;
+  uint32_t mask = (q[0] & 0x80) ? 0u : 0xFFFFFFFFu;
+  bool neg = mask != 0;
+  bool first = true;
+  unsigned __int128 acc = 0;
+  auto rd_word = [&](uint32_t size) -> uint32_t {
+    uint8_t b0 = q[0];
+    if (first) { b0 ^= 0x80; first = false; }
+    uint32_t r;
+    switch (size) {
+      case 1: r = (uint32_t)(int32_t)(int8_t)b0; break;
+      case 2: r = (uint32_t)(((int32_t)(int8_t)b0 << 8) + (int32_t)q[1]); break;
+      case 3: r = (b0 & 128) ? ((255u << 24) | ((uint32_t)b0 << 16) |
+                                ((uint32_t)q[1] << 8) | q[2])
+                             : (((uint32_t)b0 << 16) | ((uint32_t)q[1] << 8) | q[2]);
+              break;
+      default: r = (uint32_t)(((int32_t)(int8_t)b0 << 24) + ((int32_t)q[1] << 16) +
+                              ((int32_t)q[2] << 8) + (int32_t)q[3]); break;
+    }
+    q += size;
+    return r;
+  };
+  if (ld) {
+    uint32_t w = rd_word(DIG2B[ld]) ^ mask;
+    if (w >= TEN_POW_D[ld]) return 0;
+    acc = w;
+  }
+  for (uint32_t k = 0; k < iw; k++) {
+    uint32_t w = rd_word(4) ^ mask;
+    if (w > 999999999u) return 0;
+    acc = acc * 1000000000u + w;
+  }
+  for (uint32_t k = 0; k < fw; k++) {
+    uint32_t w = rd_word(4) ^ mask;
+    if (w > 999999999u) return 0;
+    acc = acc * 1000000000u + w;
+  }
+  if (td) {
+    uint32_t w = rd_word(DIG2B[td]) ^ mask;
+    if (w >= TEN_POW_D[td]) return 0;
+    acc = acc * TEN_POW_D[td] + w;
+  }
+  *scaled = neg ? -(__int128)acc : (__int128)acc;
+  *frac = (int32_t)fr;
+  return need;
+}
+
 /* one datum (flag+payload) view */
 struct CellView {
   uint32_t len;        /* full datum length incl flag; 0 = error */
@@ -207,11 +269,15 @@ struct CellView {
   bool has_real;
   bool has_dec;
   int64_t dscaled; int32_t dfrac;
+  /* wide decimal (19..38 digits): payload pointer for a deferred
+     d_decimal_scaled128 parse at the contribute site; null otherwise */
+  const uint8_t *dwide;
+  uint32_t dwide_rem;
 };
 
 __device__ static inline void d_parse_datum(const uint8_t *p, uint32_t rem, CellView *cv) {
   cv->len = 0; cv->is_null = false; cv->has_int = false; cv->has_dec = false;
-  cv->has_real = false;
+  cv->has_real = false; cv->dwide = nullptr;
   if (rem == 0) return;
   uint8_t flag = p[0];
   cv->flag = flag;
@@ -283,6 +349,9 @@ __device__ static inline void d_parse_datum(const uint8_t *p, uint32_t rem, Cell
         uint32_t fwc = frc / 9, tdg = frc - fwc * 9;
         uint32_t need = 2 + iw * 4 + DIG2B[ldg] + fwc * 4 + DIG2B[tdg];
         if (prem < need) return;
+        /* 19..38-digit decimals: defer the 128-bit parse to the
+           contribute site (sums), keep the walk moving */
+        if (prec <= 38) { cv->dwide = pl; cv->dwide_rem = prem; }
         cv->len = 1 + need; return;
       }
       cv->has_dec = true; cv->dscaled = sc; cv->dfrac = fr;
@@ -407,7 +476,8 @@ __device__ static inline bool d_is_fold(int32_t kind) {
 __device__ static inline bool d_is_xor(int32_t kind) { return kind == DAGG_BIT_XOR; }
 
 
-struct AggColView { bool found, null, has_dec; int64_t iv, dsc; int32_t dfr; };
+struct AggColView { bool found, null, has_dec; int64_t iv, dsc; int32_t dfr;
+                    const uint8_t *dwide; uint32_t dwrem; };
 
 /* ---------------- row v2 (codec/row/v2/row_slice.rs:76-168) ---------------- */
 struct V2Row {
@@ -726,6 +796,7 @@ __device__ static inline bool d_index_collect(const ScanPlan &plan,
         cols[a].iv = cell.ival;
         cols[a].has_dec = cell.has_dec;
         cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
+        cols[a].dwide = cell.dwide; cols[a].dwrem = cell.dwide_rem;
         if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec)
           return false;
       }
@@ -1122,7 +1193,7 @@ k_scan_agg(ScanPlan plan,
       bool grp_found = false, grp_null = false; int64_t grp_v = 0;
       AggColView cols[NAGGS];
       #pragma unroll
-      for (int a = 0; a < NAGGS; a++) cols[a] = {false, false, false, 0, 0, 0};
+      for (int a = 0; a < NAGGS; a++) cols[a] = {false, false, false, 0, 0, 0, nullptr, 0};
       int found = 0;
 
       if (plan.index_mode) {
@@ -1218,6 +1289,7 @@ k_scan_agg(ScanPlan plan,
                 cols[a].iv = cell.ival;
                 cols[a].has_dec = cell.has_dec;
                 cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
+                cols[a].dwide = cell.dwide; cols[a].dwrem = cell.dwide_rem;
                 if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec)
                   parse_ok = false;
               } else parse_ok = false;
@@ -1271,6 +1343,7 @@ k_scan_agg(ScanPlan plan,
               cols[a].iv = cell.ival;
               cols[a].has_dec = cell.has_dec;
               cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
+              cols[a].dwide = cell.dwide; cols[a].dwrem = cell.dwide_rem;
               if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec) parse_ok = false;
               found++;
             }
@@ -1782,7 +1855,7 @@ k_scan_agg_pipe(ScanPlan plan,
       bool grp_found = false, grp_null = false; int64_t grp_v = 0;
       AggColView cols[NAGGS];
       #pragma unroll
-      for (int a = 0; a < NAGGS; a++) cols[a] = {false, false, false, 0, 0, 0};
+      for (int a = 0; a < NAGGS; a++) cols[a] = {false, false, false, 0, 0, 0, nullptr, 0};
       int found = 0;
 
       if (plan.index_mode) {
@@ -1885,6 +1958,7 @@ k_scan_agg_pipe(ScanPlan plan,
                 cols[a].iv = cell.ival;
                 cols[a].has_dec = cell.has_dec;
                 cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
+                cols[a].dwide = cell.dwide; cols[a].dwrem = cell.dwide_rem;
                 if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec)
                   parse_ok = false;
               } else parse_ok = false;
@@ -1938,6 +2012,7 @@ k_scan_agg_pipe(ScanPlan plan,
               cols[a].iv = cell.ival;
               cols[a].has_dec = cell.has_dec;
               cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
+              cols[a].dwide = cell.dwide; cols[a].dwrem = cell.dwide_rem;
               if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec) parse_ok = false;
               found++;
             }
@@ -3786,7 +3861,7 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
       bool grp_found = false, grp_null = false; int64_t grp_v = 0;
       AggColView cols[NAGGS > 0 ? NAGGS : 1];
       #pragma unroll
-      for (int a = 0; a < NAGGS; a++) cols[a] = {false, false, false, 0, 0, 0};
+      for (int a = 0; a < NAGGS; a++) cols[a] = {false, false, false, 0, 0, 0, nullptr, 0};
 
       if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
         bool dir_done = false;
@@ -3862,6 +3937,7 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
                 cols[a].iv = cell.ival;
                 cols[a].has_dec = cell.has_dec;
                 cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
+                cols[a].dwide = cell.dwide; cols[a].dwrem = cell.dwide_rem;
                 if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec)
                   parse_ok = false;
               } else parse_ok = false;
@@ -3947,6 +4023,7 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
                 cols[a].iv = cell.ival;
                 cols[a].has_dec = cell.has_dec;
                 cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
+                cols[a].dwide = cell.dwide; cols[a].dwrem = cell.dwide_rem;
                 if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec)
                   parse_ok = false;
                 found++;
