@@ -1236,6 +1236,21 @@ static void enc_datum_dec_scaled(std::vector<uint8_t> *out, __int128 scaled, uin
   out->push_back(6);
   out->insert(out->end(), tmp, tmp + n);
 }
+/* wide sums: 256-bit scaled value (see atomic_add_i256). Returns false on
+   word_buf overflow (reference Res::Overflow -> request error). */
+static bool enc_datum_dec_scaled256(std::vector<uint8_t> *out,
+                                    const uint64_t limbs[4], uint8_t frac) {
+  bool ovf = false;
+  prod::PDec d = prod::pdec_from_scaled_i256(limbs, frac, &ovf);
+  if (ovf) return false;
+  uint8_t prec, fr;
+  prod::pdec_prec_and_frac(d, &prec, &fr);
+  uint8_t tmp[48];
+  size_t n = prod::pdec_encode(d, prec, fr, tmp);
+  out->push_back(6);
+  out->insert(out->end(), tmp, tmp + n);
+  return true;
+}
 
 static inline __int128 i128_of(unsigned long long lo, unsigned long long hi) {
   return ((__int128)(long long)hi << 64) | (__int128)lo;
@@ -1249,7 +1264,9 @@ struct GroupRow {
 
 static void encode_agg_row(const HostPlan &pl, const SimpleAggAcc *accs,
                            bool has_group, bool group_null, int64_t group_key,
-                           std::vector<std::vector<uint8_t>> *cols_out) {
+                           std::vector<std::vector<uint8_t>> *cols_out,
+                           const unsigned long long *ext = nullptr,
+                           bool *enc_err = nullptr) {
   size_t oc = 0;
   for (auto &oa : pl.out_aggs) {
     const SimpleAggAcc &ac = accs[oa.dev_idx];
@@ -1304,7 +1321,16 @@ static void encode_agg_row(const HostPlan &pl, const SimpleAggAcc *accs,
         __int128 s = i128_of(ac.sum_lo, ac.sum_hi);
         uint8_t frac = oa.in_kind == DAGG_SUM_DEC
                            ? (uint8_t)pl.sp.aggs[oa.dev_idx].target_frac : 0;
-        enc_datum_dec_scaled(&sumcol, s, frac);
+        uint64_t sgn = s < 0 ? ~0ull : 0ull;
+        if (ext && (ext[oa.dev_idx * 2] != sgn || ext[oa.dev_idx * 2 + 1] != sgn)) {
+          /* the sum spilled past i128: encode the 256-bit value */
+          uint64_t limbs[4] = {ac.sum_lo, ac.sum_hi, ext[oa.dev_idx * 2],
+                               ext[oa.dev_idx * 2 + 1]};
+          if (!enc_datum_dec_scaled256(&sumcol, limbs, frac) && enc_err)
+            *enc_err = true;
+        } else {
+          enc_datum_dec_scaled(&sumcol, s, frac);
+        }
       }
     }
   }
@@ -1887,31 +1913,53 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
     if (pl.sp.mode == 1) {
       /* ---- simple agg: one accumulator set across all regions ---- */
       SimpleAggAcc *d_acc = nullptr;
+      unsigned long long *d_ext = nullptr;
       HIP_TRY(hipMalloc(&d_acc, sizeof(SimpleAggAcc) * (COPR_MAX_AGGS + 1)), "acc alloc");
       HIP_TRY(hipMemsetAsync(d_acc, 0, sizeof(SimpleAggAcc) * (COPR_MAX_AGGS + 1),
                              eng->stream), "acc memset");
+      bool any_dec = false;
+      for (int a = 0; a < pl.sp.n_aggs; a++)
+        if (pl.sp.aggs[a].kind == DAGG_SUM_DEC || pl.sp.aggs[a].kind == DAGG_SUM_INT)
+          any_dec = true;
+      if (any_dec) {
+        if (hipMalloc(&d_ext, COPR_MAX_AGGS * 16) != hipSuccess) {
+          hipFree(d_acc);
+          return SET_ERR(COPR_ERR_OOM, "acc ext alloc");
+        }
+        hipMemsetAsync(d_ext, 0, COPR_MAX_AGGS * 16, eng->stream);
+      }
       hipEventRecord(ev_a, eng->stream);
       for (uint32_t rg = 0; rg < n_regions; rg++) {
         ScanPlan sp = pl.sp;
+        sp.simple_ext = d_ext;
         wire_celldir(&sp, regions[rg]->dev);
         pick_tiling(regions[rg]->dev, &sp);
         int e = dev_scan_launch(sp, regions[rg]->dev, d_acc, nullptr, nullptr, eng->stream);
-        if (e) { hipFree(d_acc); return SET_ERR(COPR_ERR_INTERNAL, "scan launch failed"); }
+        if (e) { hipFree(d_acc); hipFree(d_ext); return SET_ERR(COPR_ERR_INTERNAL, "scan launch failed"); }
       }
       hipEventRecord(ev_b, eng->stream);
       timed = true;
       SimpleAggAcc h_acc[COPR_MAX_AGGS + 1];
+      unsigned long long h_ext[COPR_MAX_AGGS * 2] = {0};
       hipError_t ce = hipMemcpyAsync(h_acc, d_acc, sizeof(h_acc), hipMemcpyDeviceToHost,
                                      eng->stream);
+      if (ce == hipSuccess && d_ext)
+        ce = hipMemcpyAsync(h_ext, d_ext, COPR_MAX_AGGS * 16,
+                            hipMemcpyDeviceToHost, eng->stream);
       if (ce == hipSuccess) ce = hipStreamSynchronize(eng->stream);
       hipFree(d_acc);
+      hipFree(d_ext);
       if (ce != hipSuccess) return SET_ERR(COPR_ERR_INTERNAL, hipGetErrorString(ce));
       if (h_acc[COPR_MAX_AGGS].cnt & 1)
         return SET_ERR(COPR_ERR_STORAGE, "row parse error on device");
       /* encode the single result row */
       size_t n_out_cols = pl.out_schema.size();
       std::vector<std::vector<uint8_t>> cols(n_out_cols);
-      encode_agg_row(pl, h_acc, false, false, 0, &cols);
+      bool enc_err = false;
+      encode_agg_row(pl, h_acc, false, false, 0, &cols,
+                     any_dec ? h_ext : nullptr, &enc_err);
+      if (enc_err)
+        return SET_ERR(COPR_ERR_STORAGE, "decimal sum overflow (Res::Overflow)");
       for (uint32_t oo = 0; oo < req->n_output_offsets; oo++) {
         uint32_t off = req->output_offsets[oo];
         if (off >= n_out_cols) return SET_ERR(COPR_ERR_INVALID_REQUEST, "bad output offset");
@@ -1931,14 +1979,22 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
         hipError_t e = hipSuccess;
         size_t keys_b = (size_t)tsize * 8;
         size_t accs_b = (size_t)tsize * pl.sp.n_aggs * sizeof(SimpleAggAcc);
+        bool any_dec = false;
+        for (int a = 0; a < pl.sp.n_aggs; a++)
+          if (pl.sp.aggs[a].kind == DAGG_SUM_DEC) any_dec = true;
         if (e == hipSuccess) e = hipMalloc((void **)&ht.keys, keys_b);
         if (e == hipSuccess) e = hipMalloc((void **)&ht.accs, accs_b);
         if (e == hipSuccess) e = hipMalloc((void **)&ht.reserved, 2 * pl.sp.n_aggs * sizeof(SimpleAggAcc));
+        if (e == hipSuccess && any_dec)
+          e = hipMalloc((void **)&ht.ext, (size_t)tsize * pl.sp.n_aggs * 16);
+        if (e == hipSuccess && any_dec)
+          e = hipMalloc((void **)&ht.rsvd_ext, 2 * pl.sp.n_aggs * 16);
         if (e == hipSuccess) e = hipMalloc((void **)&ht.rsvd_seen, 2 * 8);
         if (e == hipSuccess) e = hipMalloc((void **)&ht.error, 8);
         if (e == hipSuccess) e = hipMalloc((void **)&ht.n_groups, 8);
         auto free_ht = [&]() {
           hipFree(ht.keys); hipFree(ht.accs); hipFree(ht.reserved);
+          hipFree(ht.ext); hipFree(ht.rsvd_ext);
           hipFree(ht.rsvd_seen); hipFree(ht.error); hipFree(ht.n_groups);
         };
         if (e != hipSuccess) { free_ht(); return SET_ERR(COPR_ERR_OOM, "hash table alloc"); }
@@ -1955,6 +2011,10 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
           std::vector<long long> fill(tsize, (long long)0x8000000000000000ll);
           e = hipMemcpyAsync(ht.keys, fill.data(), keys_b, hipMemcpyHostToDevice, eng->stream);
           if (e == hipSuccess) e = hipMemsetAsync(ht.accs, 0, accs_b, eng->stream);
+          if (e == hipSuccess && ht.ext)
+            e = hipMemsetAsync(ht.ext, 0, (size_t)tsize * pl.sp.n_aggs * 16, eng->stream);
+          if (e == hipSuccess && ht.rsvd_ext)
+            e = hipMemsetAsync(ht.rsvd_ext, 0, 2 * pl.sp.n_aggs * 16, eng->stream);
           if (e == hipSuccess) e = hipMemsetAsync(ht.reserved, 0, 2 * pl.sp.n_aggs * sizeof(SimpleAggAcc), eng->stream);
           if (e == hipSuccess) e = hipMemsetAsync(ht.rsvd_seen, 0, 16, eng->stream);
           if (e == hipSuccess) e = hipMemsetAsync(ht.error, 0, 8, eng->stream);
@@ -2018,20 +2078,28 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
         std::vector<long long> h_keys;
         std::vector<SimpleAggAcc> h_accs;
         std::vector<SimpleAggAcc> h_rsvd(2 * pl.sp.n_aggs);
+        std::vector<unsigned long long> h_ext;
+        std::vector<unsigned long long> h_rsvd_ext(2 * pl.sp.n_aggs * 2, 0);
         int ng = dev_ht_compact(ht, tsize, pl.sp.n_aggs, eng->stream,
-                                &h_keys, &h_accs);
+                                &h_keys, &h_accs,
+                                any_dec ? &h_ext : nullptr);
         ce = hipMemcpy(h_rsvd.data(), ht.reserved,
                        2 * pl.sp.n_aggs * sizeof(SimpleAggAcc),
                        hipMemcpyDeviceToHost);
+        if (ce == hipSuccess && ht.rsvd_ext)
+          ce = hipMemcpy(h_rsvd_ext.data(), ht.rsvd_ext,
+                         2 * pl.sp.n_aggs * 16, hipMemcpyDeviceToHost);
         free_ht();
         if (ng == -2) return SET_ERR(COPR_ERR_OOM, "table compact alloc");
         if (ng < 0 || ce != hipSuccess)
           return SET_ERR(COPR_ERR_INTERNAL, "table compact failed");
         /* encode rows: occupied slots + reserved groups */
         size_t n_out_cols = pl.out_schema.size();
-        auto emit_group = [&](const SimpleAggAcc *accs, bool gnull, int64_t gkey) {
+        bool enc_err = false;
+        auto emit_group = [&](const SimpleAggAcc *accs, bool gnull,
+                              int64_t gkey, const unsigned long long *gext) {
           std::vector<std::vector<uint8_t>> cols(n_out_cols);
-          encode_agg_row(pl, accs, true, gnull, gkey, &cols);
+          encode_agg_row(pl, accs, true, gnull, gkey, &cols, gext, &enc_err);
           for (uint32_t oo = 0; oo < req->n_output_offsets; oo++) {
             uint32_t off = req->output_offsets[oo];
             if (off < n_out_cols)
@@ -2040,9 +2108,17 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
           n_rows_out++;
         };
         for (size_t s = 0; s < h_keys.size(); s++)
-          emit_group(&h_accs[s * pl.sp.n_aggs], false, h_keys[s]);
-        if (h_rsvd_seen[0]) emit_group(&h_rsvd[0], false, INT64_MIN);
-        if (h_rsvd_seen[1]) emit_group(&h_rsvd[pl.sp.n_aggs], true, 0);
+          emit_group(&h_accs[s * pl.sp.n_aggs], false, h_keys[s],
+                     any_dec ? &h_ext[s * pl.sp.n_aggs * 2] : nullptr);
+        if (h_rsvd_seen[0])
+          emit_group(&h_rsvd[0], false, INT64_MIN,
+                     any_dec ? &h_rsvd_ext[0] : nullptr);
+        if (h_rsvd_seen[1])
+          emit_group(&h_rsvd[pl.sp.n_aggs], true, 0,
+                     any_dec ? &h_rsvd_ext[pl.sp.n_aggs * 2] : nullptr);
+        if (enc_err)
+          return SET_ERR(COPR_ERR_STORAGE,
+                         "decimal sum overflow (Res::Overflow)");
         break;
       }
     }

@@ -214,6 +214,9 @@ struct ScanPlan {
      atomic addresses. */
   uint32_t lds_agg_slots;        /* power of two */
   uint32_t lds_agg_off;          /* byte offset of the table in LDS */
+  /* simple agg (mode 1): the two HIGH sum limbs per agg ([n_aggs*2]),
+     extending decimal/int sums to 256 bits (wide Decimal sums) */
+  unsigned long long *simple_ext;
 };
 
 /* simple-agg accumulators (device buffer, one per agg) */
@@ -273,7 +276,8 @@ int dev_topn_select(const ScanPlan &plan, const DevRegion &rgn,
                     std::vector<uint32_t> *winners);
 int dev_ht_compact(const HashAggTable &ht, uint32_t tsize, int n_aggs,
                    void *stream, std::vector<long long> *h_keys,
-                   std::vector<SimpleAggAcc> *h_accs);
+                   std::vector<SimpleAggAcc> *h_accs,
+                   std::vector<unsigned long long> *h_ext = nullptr);
 int dev_int_sorted_agg(const ScanPlan &plan, const DevRegion &rgn,
                        void *stream, std::vector<SimpleAggAcc> *h_accs,
                        std::vector<long long> *h_gk,

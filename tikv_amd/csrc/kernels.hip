@@ -1061,6 +1061,36 @@ __device__ static inline void atomic_add_i128(unsigned long long *lo,
   if (hi_add) atomicAdd(hi, (unsigned long long)hi_add);
 }
 
+/* 256-bit signed accumulate: acc = (ext[1] ext[0] hi lo) two's complement;
+ * the i128 addend sign-extends into the ext limbs, carries ripple through
+ * the returned old values (each limb's add is commutative, carries are
+ * explicit later adds, so concurrent updates compose). Wide Decimal sums
+ * (decimal.rs:927-942 word_buf range: 38-digit values, 65+-digit sums). */
+__device__ static inline void atomic_add_i256(unsigned long long *lo,
+                                              unsigned long long *hi,
+                                              unsigned long long *ext,
+                                              __int128 x) {
+  unsigned long long x0 = (unsigned long long)(unsigned __int128)x;
+  unsigned long long x1 = (unsigned long long)((unsigned __int128)x >> 64);
+  unsigned long long sign = x < 0 ? ~0ull : 0ull;
+  unsigned long long o0 = atomicAdd(lo, x0);
+  unsigned long long c = (o0 + x0 < o0) ? 1ull : 0ull;
+  unsigned long long a1 = x1 + c;
+  unsigned long long w1 = (a1 < x1) ? 1ull : 0ull;     /* addend wrapped */
+  unsigned long long o1 = atomicAdd(hi, a1);
+  c = w1 | ((o1 + a1 < o1) ? 1ull : 0ull);
+  unsigned long long a2 = sign + c;
+  unsigned long long w2 = (a2 < sign) ? 1ull : 0ull;
+  if (a2) {
+    unsigned long long o2 = atomicAdd(&ext[0], a2);
+    c = w2 | ((o2 + a2 < o2) ? 1ull : 0ull);
+  } else {
+    c = w2;
+  }
+  unsigned long long a3 = sign + c;
+  if (a3) atomicAdd(&ext[1], a3);
+}
+
 /* ---------------- fused scan + filter + aggregate ----------------
  * NAGGS is a compile-time bound so all per-row/per-lane state stays in
  * registers. IS_HASH selects grouped aggregation. NLOADS = staged uint4
@@ -1361,19 +1391,42 @@ k_scan_agg(ScanPlan plan,
                      ? true
                      : (ke ? (any_parse_err = true, false) : false)) {
         SimpleAggAcc *acc_base = nullptr;
+        unsigned long long *ext_base = nullptr;
+        /* conservative wide-decimal pre-pass: rows that may need a 256-bit
+           add must take a GLOBAL slot (the LDS table is 128-bit) */
+        bool row_wide = false;
+        if (IS_HASH) {
+          #pragma unroll
+          for (int a = 0; a < NAGGS; a++) {
+            if (plan.aggs[a].kind != DAGG_SUM_DEC || !cols[a].found ||
+                cols[a].null)
+              continue;
+            if (cols[a].dwide) { row_wide = true; continue; }
+            if (!cols[a].has_dec) continue;
+            int d = plan.aggs[a].target_frac - cols[a].dfr;
+            if (d > 18) { row_wide = true; continue; }
+            if (d > 0) {
+              int64_t lim = (int64_t)(0x7FFFFFFFFFFFFFFFll);
+              for (int t = 0; t < d; t++) lim /= 10;
+              if (cols[a].dsc > lim || cols[a].dsc < -lim) row_wide = true;
+            }
+          }
+        }
         if (IS_HASH) {
           if (!grp_found || grp_null) {
             atomicAdd(&ht.rsvd_seen[1], 1ull);
             acc_base = ht.reserved + 1 * NAGGS;
+            if (ht.rsvd_ext) ext_base = ht.rsvd_ext + 1 * NAGGS * 2;
           } else if (grp_v == (long long)0x8000000000000000ll) {
             atomicAdd(&ht.rsvd_seen[0], 1ull);
             acc_base = ht.reserved + 0 * NAGGS;
+            if (ht.rsvd_ext) ext_base = ht.rsvd_ext + 0 * NAGGS * 2;
           } else {
             const unsigned long long EMPTY = 0x8000000000000000ull;
             uint64_t h = (uint64_t)grp_v * 0x9E3779B97F4A7C15ull;
             h ^= h >> 29;
             bool in_lds = false;
-            if (LSLOTS) {
+            if (LSLOTS && !row_wide) {
               /* per-block LDS table first; fall through to global if full */
               uint32_t lmask = LSLOTS - 1u;
               uint32_t slot = (uint32_t)(h & lmask);
@@ -1400,10 +1453,12 @@ k_scan_agg(ScanPlan plan,
                 if (curk == EMPTY) {
                   atomicAdd(ht.n_groups, 1ull);
                   acc_base = ht.accs + (uint64_t)slot * NAGGS;
+                  if (ht.ext) ext_base = ht.ext + (uint64_t)slot * NAGGS * 2;
                   break;
                 }
                 if (curk == (unsigned long long)grp_v) {
                   acc_base = ht.accs + (uint64_t)slot * NAGGS;
+                  if (ht.ext) ext_base = ht.ext + (uint64_t)slot * NAGGS * 2;
                   break;
                 }
                 slot = (slot + 1) & mask;
@@ -1415,6 +1470,8 @@ k_scan_agg(ScanPlan plan,
         for (int a = 0; a < NAGGS; a++) {
           const DevAggSpec &sp = plan.aggs[a];
           bool contribute;
+          bool dec_wide = false;       /* value does not fit a scaled i64 */
+          __int128 vw = 0;
           int64_t v = 0;
           if (sp.kind == DAGG_COUNT_ROWS) {
             contribute = true;
@@ -1425,16 +1482,43 @@ k_scan_agg(ScanPlan plan,
           } else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_REAL ||
                      d_is_fold(sp.kind)) {
             contribute = true; v = cols[a].iv;
-          } else {  /* SUM_DEC */
-            int d = sp.target_frac - cols[a].dfr;
-            if (!cols[a].has_dec || d < 0 || d > 18) {
+          } else {  /* SUM_DEC: values to 38 digits via the i128 parse;
+                         a scaled value past i64 goes to the 256-bit path */
+            __int128 sv = 0;
+            int32_t fr = 0;
+            bool okd = true;
+            if (cols[a].has_dec) {
+              sv = cols[a].dsc;
+              fr = cols[a].dfr;
+            } else if (cols[a].dwide) {
+              if (!d_decimal_scaled128(cols[a].dwide, cols[a].dwrem, &sv, &fr))
+                okd = false;
+            } else {
+              okd = false;
+            }
+            int d = okd ? sp.target_frac - fr : -1;
+            if (!okd || d < 0 || d > 38) {
               any_parse_err = true;
               contribute = false;
             } else {
-              int64_t scale = 1;
-              for (int t = 0; t < d; t++) scale *= 10;
-              v = cols[a].dsc * scale;
-              contribute = true;
+              const __int128 LIM =
+                  (__int128)(((unsigned __int128)~(unsigned __int128)0) >> 1) / 10;
+              bool ovf = false;
+              for (int t = 0; t < d; t++) {
+                if (sv > LIM || sv < -LIM) { ovf = true; break; }
+                sv *= 10;
+              }
+              if (ovf) {
+                any_parse_err = true;       /* > 38-digit scaled: loud */
+                contribute = false;
+              } else if (sv >= (__int128)INT64_MIN && sv <= (__int128)INT64_MAX) {
+                v = (int64_t)sv;
+                contribute = true;
+              } else {
+                vw = sv;
+                dec_wide = true;
+                contribute = true;
+              }
             }
           }
           if (!contribute) continue;
@@ -1444,8 +1528,17 @@ k_scan_agg(ScanPlan plan,
               if (sp.kind == DAGG_SUM_REAL)
                 atomicAdd((double *)&acc_base[a].sum_lo,
                           __longlong_as_double(v));
-              else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC)
-                atomic_add_i128(&acc_base[a].sum_lo, &acc_base[a].sum_hi, v);
+              else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC) {
+                if (dec_wide) {
+                  if (ext_base)
+                    atomic_add_i256(&acc_base[a].sum_lo, &acc_base[a].sum_hi,
+                                    ext_base + a * 2, vw);
+                  else
+                    any_parse_err = true;   /* no ext buffers: loud */
+                } else {
+                  atomic_add_i128(&acc_base[a].sum_lo, &acc_base[a].sum_hi, v);
+                }
+              }
               else if (d_is_fold(sp.kind)) {
                 unsigned long long b = d_fold_xform(sp.kind, v, sp.col_unsigned);
                 if (d_is_xor(sp.kind)) atomicXor(&acc_base[a].sum_lo, b);
@@ -1461,10 +1554,20 @@ k_scan_agg(ScanPlan plan,
                   __longlong_as_double((long long)l_lo[a]) +
                   __longlong_as_double(v));
             } else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC) {
-              unsigned long long old = l_lo[a];
-              unsigned long long nv = old + (unsigned long long)v;
-              l_hi[a] += (nv < old ? 1 : 0) + (v < 0 ? -1 : 0);
-              l_lo[a] = nv;
+              if (dec_wide) {
+                /* wide values go straight to the 256-bit global acc; the
+                   per-lane i128 partial keeps only i64-fitting values */
+                if (plan.simple_ext)
+                  atomic_add_i256(&simple_acc[a].sum_lo, &simple_acc[a].sum_hi,
+                                  plan.simple_ext + a * 2, vw);
+                else
+                  any_parse_err = true;
+              } else {
+                unsigned long long old = l_lo[a];
+                unsigned long long nv = old + (unsigned long long)v;
+                l_hi[a] += (nv < old ? 1 : 0) + (v < 0 ? -1 : 0);
+                l_lo[a] = nv;
+              }
             } else if (d_is_fold(sp.kind)) {
               unsigned long long b = d_fold_xform(sp.kind, v, sp.col_unsigned);
               if (d_is_xor(sp.kind)) l_lo[a] ^= b;
@@ -1490,6 +1593,7 @@ k_scan_agg(ScanPlan plan,
       h ^= h >> 29;
       uint32_t slot = (uint32_t)(h & mask);
       SimpleAggAcc *gacc = nullptr;
+      unsigned long long *gext = nullptr;
       for (uint32_t probe = 0; ; probe++) {
         if (probe > mask) { atomicOr(ht.error, 1u); break; }
         unsigned long long curk =
@@ -1500,6 +1604,7 @@ k_scan_agg(ScanPlan plan,
         slot = (slot + 1) & mask;
       }
       if (!gacc) continue;
+      if (ht.ext) gext = ht.ext + (uint64_t)slot * NAGGS * 2;
       #pragma unroll
       for (int a = 0; a < NAGGS; a++) {
         const SimpleAggAcc &la = laccs[s * NAGGS + a];
@@ -1515,10 +1620,17 @@ k_scan_agg(ScanPlan plan,
           continue;
         }
         if (la.sum_lo | la.sum_hi) {
-          unsigned long long old = atomicAdd(&gacc[a].sum_lo, la.sum_lo);
-          long long carry = (old + la.sum_lo < old) ? 1 : 0;
-          long long hi_add = (long long)la.sum_hi + carry;
-          if (hi_add) atomicAdd(&gacc[a].sum_hi, (unsigned long long)hi_add);
+          __int128 part = (__int128)(
+              ((unsigned __int128)la.sum_hi << 64) | la.sum_lo);
+          if (gext)
+            atomic_add_i256(&gacc[a].sum_lo, &gacc[a].sum_hi, gext + a * 2,
+                            part);
+          else {
+            unsigned long long old = atomicAdd(&gacc[a].sum_lo, la.sum_lo);
+            long long carry = (old + la.sum_lo < old) ? 1 : 0;
+            long long hi_add = (long long)la.sum_hi + carry;
+            if (hi_add) atomicAdd(&gacc[a].sum_hi, (unsigned long long)hi_add);
+          }
         }
       }
     }
@@ -1575,10 +1687,18 @@ k_scan_agg(ScanPlan plan,
       if ((threadIdx.x & 63u) == 0) {
         if (c) atomicAdd(&simple_acc[a].cnt, c);
         if (lo | (unsigned long long)hi) {
-          unsigned long long old = atomicAdd(&simple_acc[a].sum_lo, lo);
-          long long carry = (old + lo < old) ? 1 : 0;
-          long long hi_add = hi + carry;
-          if (hi_add) atomicAdd(&simple_acc[a].sum_hi, (unsigned long long)hi_add);
+          __int128 part = (__int128)(
+              ((unsigned __int128)(unsigned long long)hi << 64) | lo);
+          if (plan.simple_ext)
+            atomic_add_i256(&simple_acc[a].sum_lo, &simple_acc[a].sum_hi,
+                            plan.simple_ext + a * 2, part);
+          else {
+            unsigned long long old = atomicAdd(&simple_acc[a].sum_lo, lo);
+            long long carry = (old + lo < old) ? 1 : 0;
+            long long hi_add = hi + carry;
+            if (hi_add)
+              atomicAdd(&simple_acc[a].sum_hi, (unsigned long long)hi_add);
+          }
         }
       }
     }
@@ -2030,19 +2150,42 @@ k_scan_agg_pipe(ScanPlan plan,
                      ? true
                      : (ke ? (any_parse_err = true, false) : false)) {
         SimpleAggAcc *acc_base = nullptr;
+        unsigned long long *ext_base = nullptr;
+        /* conservative wide-decimal pre-pass: rows that may need a 256-bit
+           add must take a GLOBAL slot (the LDS table is 128-bit) */
+        bool row_wide = false;
+        if (IS_HASH) {
+          #pragma unroll
+          for (int a = 0; a < NAGGS; a++) {
+            if (plan.aggs[a].kind != DAGG_SUM_DEC || !cols[a].found ||
+                cols[a].null)
+              continue;
+            if (cols[a].dwide) { row_wide = true; continue; }
+            if (!cols[a].has_dec) continue;
+            int d = plan.aggs[a].target_frac - cols[a].dfr;
+            if (d > 18) { row_wide = true; continue; }
+            if (d > 0) {
+              int64_t lim = (int64_t)(0x7FFFFFFFFFFFFFFFll);
+              for (int t = 0; t < d; t++) lim /= 10;
+              if (cols[a].dsc > lim || cols[a].dsc < -lim) row_wide = true;
+            }
+          }
+        }
         if (IS_HASH) {
           if (!grp_found || grp_null) {
             atomicAdd(&ht.rsvd_seen[1], 1ull);
             acc_base = ht.reserved + 1 * NAGGS;
+            if (ht.rsvd_ext) ext_base = ht.rsvd_ext + 1 * NAGGS * 2;
           } else if (grp_v == (long long)0x8000000000000000ll) {
             atomicAdd(&ht.rsvd_seen[0], 1ull);
             acc_base = ht.reserved + 0 * NAGGS;
+            if (ht.rsvd_ext) ext_base = ht.rsvd_ext + 0 * NAGGS * 2;
           } else {
             uint64_t h = (uint64_t)grp_v * 0x9E3779B97F4A7C15ull;
             h ^= h >> 29;
             const unsigned long long EMPTY = 0x8000000000000000ull;
             bool in_lds = false;
-            if (LSLOTS) {
+            if (LSLOTS && !row_wide) {
               /* per-block LDS table first; fall through to global if full */
               uint32_t lmask = LSLOTS - 1u;
               uint32_t slot = (uint32_t)(h & lmask);
@@ -2069,10 +2212,12 @@ k_scan_agg_pipe(ScanPlan plan,
                 if (curk == EMPTY) {
                   atomicAdd(ht.n_groups, 1ull);
                   acc_base = ht.accs + (uint64_t)slot * NAGGS;
+                  if (ht.ext) ext_base = ht.ext + (uint64_t)slot * NAGGS * 2;
                   break;
                 }
                 if (curk == (unsigned long long)grp_v) {
                   acc_base = ht.accs + (uint64_t)slot * NAGGS;
+                  if (ht.ext) ext_base = ht.ext + (uint64_t)slot * NAGGS * 2;
                   break;
                 }
                 slot = (slot + 1) & mask;
@@ -2084,6 +2229,8 @@ k_scan_agg_pipe(ScanPlan plan,
         for (int a = 0; a < NAGGS; a++) {
           const DevAggSpec &sp = plan.aggs[a];
           bool contribute;
+          bool dec_wide = false;       /* value does not fit a scaled i64 */
+          __int128 vw = 0;
           int64_t v = 0;
           if (sp.kind == DAGG_COUNT_ROWS) {
             contribute = true;
@@ -2094,16 +2241,43 @@ k_scan_agg_pipe(ScanPlan plan,
           } else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_REAL ||
                      d_is_fold(sp.kind)) {
             contribute = true; v = cols[a].iv;
-          } else {  /* SUM_DEC */
-            int d = sp.target_frac - cols[a].dfr;
-            if (!cols[a].has_dec || d < 0 || d > 18) {
+          } else {  /* SUM_DEC: values to 38 digits via the i128 parse;
+                         a scaled value past i64 goes to the 256-bit path */
+            __int128 sv = 0;
+            int32_t fr = 0;
+            bool okd = true;
+            if (cols[a].has_dec) {
+              sv = cols[a].dsc;
+              fr = cols[a].dfr;
+            } else if (cols[a].dwide) {
+              if (!d_decimal_scaled128(cols[a].dwide, cols[a].dwrem, &sv, &fr))
+                okd = false;
+            } else {
+              okd = false;
+            }
+            int d = okd ? sp.target_frac - fr : -1;
+            if (!okd || d < 0 || d > 38) {
               any_parse_err = true;
               contribute = false;
             } else {
-              int64_t scale = 1;
-              for (int t = 0; t < d; t++) scale *= 10;
-              v = cols[a].dsc * scale;
-              contribute = true;
+              const __int128 LIM =
+                  (__int128)(((unsigned __int128)~(unsigned __int128)0) >> 1) / 10;
+              bool ovf = false;
+              for (int t = 0; t < d; t++) {
+                if (sv > LIM || sv < -LIM) { ovf = true; break; }
+                sv *= 10;
+              }
+              if (ovf) {
+                any_parse_err = true;       /* > 38-digit scaled: loud */
+                contribute = false;
+              } else if (sv >= (__int128)INT64_MIN && sv <= (__int128)INT64_MAX) {
+                v = (int64_t)sv;
+                contribute = true;
+              } else {
+                vw = sv;
+                dec_wide = true;
+                contribute = true;
+              }
             }
           }
           if (!contribute) continue;
@@ -2113,8 +2287,17 @@ k_scan_agg_pipe(ScanPlan plan,
               if (sp.kind == DAGG_SUM_REAL)
                 atomicAdd((double *)&acc_base[a].sum_lo,
                           __longlong_as_double(v));
-              else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC)
-                atomic_add_i128(&acc_base[a].sum_lo, &acc_base[a].sum_hi, v);
+              else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC) {
+                if (dec_wide) {
+                  if (ext_base)
+                    atomic_add_i256(&acc_base[a].sum_lo, &acc_base[a].sum_hi,
+                                    ext_base + a * 2, vw);
+                  else
+                    any_parse_err = true;   /* no ext buffers: loud */
+                } else {
+                  atomic_add_i128(&acc_base[a].sum_lo, &acc_base[a].sum_hi, v);
+                }
+              }
               else if (d_is_fold(sp.kind)) {
                 unsigned long long b = d_fold_xform(sp.kind, v, sp.col_unsigned);
                 if (d_is_xor(sp.kind)) atomicXor(&acc_base[a].sum_lo, b);
@@ -2130,10 +2313,20 @@ k_scan_agg_pipe(ScanPlan plan,
                   __longlong_as_double((long long)l_lo[a]) +
                   __longlong_as_double(v));
             } else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC) {
-              unsigned long long old = l_lo[a];
-              unsigned long long nv = old + (unsigned long long)v;
-              l_hi[a] += (nv < old ? 1 : 0) + (v < 0 ? -1 : 0);
-              l_lo[a] = nv;
+              if (dec_wide) {
+                /* wide values go straight to the 256-bit global acc; the
+                   per-lane i128 partial keeps only i64-fitting values */
+                if (plan.simple_ext)
+                  atomic_add_i256(&simple_acc[a].sum_lo, &simple_acc[a].sum_hi,
+                                  plan.simple_ext + a * 2, vw);
+                else
+                  any_parse_err = true;
+              } else {
+                unsigned long long old = l_lo[a];
+                unsigned long long nv = old + (unsigned long long)v;
+                l_hi[a] += (nv < old ? 1 : 0) + (v < 0 ? -1 : 0);
+                l_lo[a] = nv;
+              }
             } else if (d_is_fold(sp.kind)) {
               unsigned long long b = d_fold_xform(sp.kind, v, sp.col_unsigned);
               if (d_is_xor(sp.kind)) l_lo[a] ^= b;
@@ -2162,6 +2355,7 @@ k_scan_agg_pipe(ScanPlan plan,
       h ^= h >> 29;
       uint32_t slot = (uint32_t)(h & mask);
       SimpleAggAcc *gacc = nullptr;
+      unsigned long long *gext = nullptr;
       for (uint32_t probe = 0; ; probe++) {
         if (probe > mask) { atomicOr(ht.error, 1u); break; }
         unsigned long long curk =
@@ -2172,6 +2366,7 @@ k_scan_agg_pipe(ScanPlan plan,
         slot = (slot + 1) & mask;
       }
       if (!gacc) continue;
+      if (ht.ext) gext = ht.ext + (uint64_t)slot * NAGGS * 2;
       #pragma unroll
       for (int a = 0; a < NAGGS; a++) {
         const SimpleAggAcc &la = laccs[s * NAGGS + a];
@@ -2187,10 +2382,17 @@ k_scan_agg_pipe(ScanPlan plan,
           continue;
         }
         if (la.sum_lo | la.sum_hi) {
-          unsigned long long old = atomicAdd(&gacc[a].sum_lo, la.sum_lo);
-          long long carry = (old + la.sum_lo < old) ? 1 : 0;
-          long long hi_add = (long long)la.sum_hi + carry;
-          if (hi_add) atomicAdd(&gacc[a].sum_hi, (unsigned long long)hi_add);
+          __int128 part = (__int128)(
+              ((unsigned __int128)la.sum_hi << 64) | la.sum_lo);
+          if (gext)
+            atomic_add_i256(&gacc[a].sum_lo, &gacc[a].sum_hi, gext + a * 2,
+                            part);
+          else {
+            unsigned long long old = atomicAdd(&gacc[a].sum_lo, la.sum_lo);
+            long long carry = (old + la.sum_lo < old) ? 1 : 0;
+            long long hi_add = (long long)la.sum_hi + carry;
+            if (hi_add) atomicAdd(&gacc[a].sum_hi, (unsigned long long)hi_add);
+          }
         }
       }
     }
@@ -2247,10 +2449,18 @@ k_scan_agg_pipe(ScanPlan plan,
       if ((threadIdx.x & 63u) == 0) {
         if (c) atomicAdd(&simple_acc[a].cnt, c);
         if (lo | (unsigned long long)hi) {
-          unsigned long long old = atomicAdd(&simple_acc[a].sum_lo, lo);
-          long long carry = (old + lo < old) ? 1 : 0;
-          long long hi_add = hi + carry;
-          if (hi_add) atomicAdd(&simple_acc[a].sum_hi, (unsigned long long)hi_add);
+          __int128 part = (__int128)(
+              ((unsigned __int128)(unsigned long long)hi << 64) | lo);
+          if (plan.simple_ext)
+            atomic_add_i256(&simple_acc[a].sum_lo, &simple_acc[a].sum_hi,
+                            plan.simple_ext + a * 2, part);
+          else {
+            unsigned long long old = atomicAdd(&simple_acc[a].sum_lo, lo);
+            long long carry = (old + lo < old) ? 1 : 0;
+            long long hi_add = hi + carry;
+            if (hi_add)
+              atomicAdd(&simple_acc[a].sum_hi, (unsigned long long)hi_add);
+          }
         }
       }
     }
@@ -3088,7 +3298,8 @@ __device__ static inline uint64_t d_crc64_stream(const uint8_t *__restrict__ bas
           tab[1 * 256 + (uint32_t)((crc >> 48) & 0xFF)] ^
           tab[0 * 256 + (uint32_t)(crc >> 56)];
   };
-  if (PREFETCH && n8 >= 8) {
+  if constexpr (PREFETCH) {
+  if (n8 >= 8) {
     /* software pipeline over 64-byte register chunks: chunk k+1's 8
        independent loads are in flight while the dependent table chain
        consumes chunk k (the unpipelined form measured 54% SQ_WAIT_ANY —
@@ -3120,6 +3331,7 @@ __device__ static inline uint64_t d_crc64_stream(const uint8_t *__restrict__ bas
       prev = w[j];
       step8(cur);
     }
+  }
   } else {
     while (n8 >= 8) {
       uint64_t w[8];
@@ -4479,30 +4691,41 @@ __global__ static void k_ht_flags(const long long *keys, uint32_t tsize,
 }
 
 __global__ static void k_ht_gather(const long long *keys,
-                                   const SimpleAggAcc *accs, uint32_t tsize,
+                                   const SimpleAggAcc *accs,
+                                   const unsigned long long *ext,
+                                   uint32_t tsize,
                                    const uint64_t *pos, int n_aggs,
-                                   long long *ck, SimpleAggAcc *caccs) {
+                                   long long *ck, SimpleAggAcc *caccs,
+                                   unsigned long long *cext) {
   uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= tsize || keys[i] == (long long)0x8000000000000000ll) return;
   uint64_t p = pos[i];
   ck[p] = keys[i];
-  for (int a = 0; a < n_aggs; a++)
+  for (int a = 0; a < n_aggs; a++) {
     caccs[p * n_aggs + a] = accs[(uint64_t)i * n_aggs + a];
+    if (ext) {
+      cext[(p * n_aggs + a) * 2] = ext[((uint64_t)i * n_aggs + a) * 2];
+      cext[(p * n_aggs + a) * 2 + 1] = ext[((uint64_t)i * n_aggs + a) * 2 + 1];
+    }
+  }
 }
 
 /* returns n_groups (>=0) with host vectors filled, or -1/-2 */
 int dev_ht_compact(const HashAggTable &ht, uint32_t tsize, int n_aggs,
                    void *stream, std::vector<long long> *h_keys,
-                   std::vector<SimpleAggAcc> *h_accs) {
+                   std::vector<SimpleAggAcc> *h_accs,
+                   std::vector<unsigned long long> *h_ext) {
   hipStream_t s = (hipStream_t)stream;
   uint32_t *f = nullptr;
   uint64_t *pos = nullptr;
   long long *ck = nullptr;
   SimpleAggAcc *caccs = nullptr;
+  unsigned long long *cext = nullptr;
   void *tmp = nullptr;
   size_t tmpb = 0;
   auto freeall = [&]() {
-    hipFree(f); hipFree(pos); hipFree(ck); hipFree(caccs); hipFree(tmp);
+    hipFree(f); hipFree(pos); hipFree(ck); hipFree(caccs); hipFree(cext);
+    hipFree(tmp);
   };
   hipError_t e = hipSuccess;
   if (e == hipSuccess) e = hipMalloc(&f, (uint64_t)tsize * 4 + 4);
@@ -4524,17 +4747,24 @@ int dev_ht_compact(const HashAggTable &ht, uint32_t tsize, int n_aggs,
   g += lf;
   h_keys->resize(g);
   h_accs->resize(g * n_aggs);
+  if (h_ext) h_ext->assign(g * n_aggs * 2, 0ull);
   if (g) {
     if (e == hipSuccess) e = hipMalloc(&ck, g * 8);
     if (e == hipSuccess)
       e = hipMalloc(&caccs, g * n_aggs * sizeof(SimpleAggAcc));
+    if (e == hipSuccess && ht.ext && h_ext)
+      e = hipMalloc(&cext, g * n_aggs * 16);
     if (e != hipSuccess) { freeall(); return -2; }
     hipLaunchKernelGGL(k_ht_gather, dim3(blocks), dim3(256), 0, s, ht.keys,
-                       ht.accs, tsize, pos, n_aggs, ck, caccs);
+                       ht.accs, (h_ext ? ht.ext : nullptr), tsize, pos,
+                       n_aggs, ck, caccs, cext);
     e = hipMemcpyAsync(h_keys->data(), ck, g * 8, hipMemcpyDeviceToHost, s);
     if (e == hipSuccess)
       e = hipMemcpyAsync(h_accs->data(), caccs,
                          g * n_aggs * sizeof(SimpleAggAcc),
+                         hipMemcpyDeviceToHost, s);
+    if (e == hipSuccess && cext)
+      e = hipMemcpyAsync(h_ext->data(), cext, g * n_aggs * 16,
                          hipMemcpyDeviceToHost, s);
     if (e == hipSuccess) e = hipStreamSynchronize(s);
   }

@@ -48,6 +48,70 @@ PDec pdec_from_scaled_i128(__int128 scaled, uint8_t frac) {
   return d;
 }
 
+namespace {
+/* 256-bit unsigned helper for the wide-sum path: long division by a small
+ * base extracts the decimal digits the word_buf wants */
+struct U256 {
+  uint64_t l[4];
+  bool is_zero() const { return !(l[0] | l[1] | l[2] | l[3]); }
+  uint32_t divmod(uint32_t div) {
+    unsigned __int128 rem = 0;
+    for (int i = 3; i >= 0; i--) {
+      unsigned __int128 cur = (rem << 64) | l[i];
+      l[i] = (uint64_t)(cur / div);
+      rem = cur % div;
+    }
+    return (uint32_t)rem;
+  }
+};
+}  // namespace
+
+PDec pdec_from_scaled_i256(const uint64_t limbs[4], uint8_t frac, bool *ovf) {
+  if (ovf) *ovf = false;
+  uint64_t sgn = (limbs[3] >> 63) ? ~0ull : 0ull;
+  if (limbs[2] == sgn && limbs[3] == sgn) {   /* fits i128 */
+    __int128 v =
+        (__int128)(((unsigned __int128)limbs[1] << 64) | limbs[0]);
+    return pdec_from_scaled_i128(v, frac);
+  }
+  PDec d;
+  d.negative = sgn != 0;
+  U256 mag{{limbs[0], limbs[1], limbs[2], limbs[3]}};
+  if (d.negative) {                            /* two's-complement negate */
+    unsigned __int128 c = 1;
+    for (int i = 0; i < 4; i++) {
+      unsigned __int128 t = (unsigned __int128)(uint64_t)~mag.l[i] + c;
+      mag.l[i] = (uint64_t)t;
+      c = t >> 64;
+    }
+  }
+  int frac_words = (frac + DIGITS_PER_WORD - 1) / DIGITS_PER_WORD;
+  uint32_t fw[9] = {0};
+  for (int w = frac_words - 1; w >= 0; w--) {
+    int dig = frac - w * DIGITS_PER_WORD;
+    if (dig > DIGITS_PER_WORD) dig = DIGITS_PER_WORD;
+    uint32_t chunk = mag.divmod(TEN_POW[dig]);
+    fw[w] = chunk * TEN_POW[DIGITS_PER_WORD - dig];
+  }
+  uint32_t iw[9] = {0};
+  int int_words = 0;
+  while (!mag.is_zero()) {
+    if (int_words + frac_words >= 9) {
+      /* beyond word_buf capacity (81 digits): the reference's Res::Overflow
+         — callers surface it as an error */
+      if (ovf) *ovf = true;
+      return d;
+    }
+    iw[int_words++] = mag.divmod(WORD_BASE);
+  }
+  if (int_words == 0) int_words = 1;
+  d.int_cnt = (uint8_t)(int_words * DIGITS_PER_WORD);
+  d.frac_cnt = frac;
+  for (int i = 0; i < int_words; i++) d.word_buf[i] = iw[int_words - 1 - i];
+  for (int w = 0; w < frac_words; w++) d.word_buf[int_words + w] = fw[w];
+  return d;
+}
+
 static uint8_t count_leading_zeroes(uint8_t i, uint32_t word) {
   uint8_t c = 0;
   while (TEN_POW[i] > word) { i--; c++; }

@@ -28,6 +28,12 @@ struct PDec {
  * Exact for |scaled| < 10^81 (word_buf capacity); callers stay far below. */
 PDec pdec_from_scaled_i128(__int128 scaled, uint8_t frac);
 
+/* 256-bit variant (wide Decimal sums): limbs = little-endian two's
+ * complement. *ovf set when the magnitude exceeds the word_buf capacity
+ * (the reference's Res::Overflow). */
+PDec pdec_from_scaled_i256(const uint64_t limbs[4], uint8_t frac,
+                           bool *ovf = nullptr);
+
 /* least (prec, frac) encoding this value completely (prec_and_frac) */
 void pdec_prec_and_frac(const PDec &d, uint8_t *prec, uint8_t *frac);
 
