@@ -1257,3 +1257,82 @@ def test_real_filter_project_gpu(engine):
         assert od == gd
     finally:
         rgn.close()
+
+
+# ---- selection over >2 distinct columns (selection_executor.rs:86) -----
+def _four_col_region(n=40000):
+    import random
+    rng = random.Random(23)
+    rows = []
+    for i in range(n):
+        rows.append(b"".join(
+            cell_int(c, rng.randrange(-1000, 1000)) for c in range(1, 5)))
+    keys = b"".join(row_key(i) for i in range(n))
+    ko = [19 * i for i in range(n + 1)]
+    vals = b"".join(rows)
+    vo = [0]
+    for r in rows:
+        vo.append(vo[-1] + len(r))
+    kb = (C.c_uint8 * len(keys)).from_buffer_copy(keys)
+    vb = (C.c_uint8 * len(vals)).from_buffer_copy(vals)
+    return (C.cast(kb, C.POINTER(C.c_uint8)), (C.c_uint64 * len(ko))(*ko),
+            C.cast(vb, C.POINTER(C.c_uint8)), (C.c_uint64 * len(vo))(*vo),
+            n, (kb, vb))
+
+
+def _four_col_req(n_conds):
+    cols = [tikv_amd.Col(i) for i in range(1, 5)]
+    sels = [tikv_amd.cmp_col_const(0, F.SIG_GT_INT, -500),
+            tikv_amd.cmp_col_const(1, F.SIG_LT_INT, 500),
+            tikv_amd.cmp_col_const(2, F.SIG_NE_INT, 0),
+            tikv_amd.cmp_col_const(3, F.SIG_GE_INT, -900)][:n_conds]
+    return (tikv_amd.DagSelect(cols).where(*sels)
+            .simple_agg([tikv_amd.count_star(), tikv_amd.sum_col(0),
+                         tikv_amd.max_col(3)]).build())
+
+
+def test_four_col_selection_oracle():
+    orc = _orc()
+    k, ko, v, vo, n, keep = _four_col_region(5000)
+    for nc in (3, 4):
+        data, nrows = orc.dag_run(_four_col_req(nc), k, ko, v, vo, n)
+        assert nrows == 1
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("n_conds", [3, 4])
+def test_four_col_selection_gpu_parity(engine, n_conds):
+    """ANDed conjuncts over 3-4 DISTINCT columns ride the capture-only
+    pseudo-agg channels (DAGG_XCAP)."""
+    orc = _orc()
+    k, ko, v, vo, n, keep = _four_col_region()
+    req = _four_col_req(n_conds)
+    od, orows = orc.dag_run(req, k, ko, v, vo, n)
+    rgn = engine.region_raw(k, ko, v, vo, n)
+    try:
+        gd, gr, _ = engine.dag_run(req, [rgn])
+    finally:
+        rgn.close()
+    assert orows == gr == 1
+    assert od == gd
+
+
+@pytest.mark.gpu
+def test_four_col_selection_hash_gpu(engine):
+    orc = _orc()
+    k, ko, v, vo, n, keep = _four_col_region()
+    cols = [tikv_amd.Col(i) for i in range(1, 5)]
+    sels = [tikv_amd.cmp_col_const(1, F.SIG_GT_INT, -800),
+            tikv_amd.cmp_col_const(2, F.SIG_LT_INT, 800),
+            tikv_amd.cmp_col_const(3, F.SIG_NE_INT, 7)]
+    req = (tikv_amd.DagSelect(cols).where(*sels)
+           .hash_agg([tikv_amd.count_star(), tikv_amd.sum_col(1)],
+                     tikv_amd.Expr().col(0)).build())
+    od, orows = orc.dag_run(req, k, ko, v, vo, n)
+    rgn = engine.region_raw(k, ko, v, vo, n)
+    try:
+        gd, gr, _ = engine.dag_run(req, [rgn])
+    finally:
+        rgn.close()
+    assert orows == gr
+    assert sorted(split_rows(od, 3)) == sorted(split_rows(gd, 3))

@@ -733,7 +733,7 @@ static void wire_celldir(ScanPlan *sp, const DevRegion &dev) {
 static copr_status translate_rpn(const CoprExecutor &ex, HostPlan *pl,
                                  ScanPlan *sp) {
   struct Slot { int64_t col_id; int off; };
-  Slot slots[2];
+  Slot slots[4];
   int n_slots = 0;
   int n = 0;
   int depth = 0;
@@ -760,7 +760,7 @@ static copr_status translate_rpn(const CoprExecutor &ex, HostPlan *pl,
           for (int t = 0; t < n_slots; t++)
             if (slots[t].col_id == ci.column_id) s = t;
           if (s < 0) {
-            if (n_slots == 2) return COPR_ERR_UNSUPPORTED;
+            if (n_slots == 4) return COPR_ERR_UNSUPPORTED;
             slots[n_slots] = {ci.column_id, (int)off};
             s = n_slots++;
           }
@@ -843,7 +843,7 @@ static copr_status translate_rpn(const CoprExecutor &ex, HostPlan *pl,
     if (st != COPR_OK) return st;
     pl->filter_col_offset = slots[0].off;
   }
-  if (n_slots == 2) {
+  if (n_slots >= 2) {
     const CoprColumnInfo &ci = pl->cols[slots[1].off];
     sp->filter2_on = 1;
     sp->filter2_col_id = ci.column_id;
@@ -852,6 +852,27 @@ static copr_status translate_rpn(const CoprExecutor &ex, HostPlan *pl,
                           &sp->filter2_missing_val);
     if (st != COPR_OK) return st;
     pl->filter2_col_offset = slots[1].off;
+  }
+  /* channels 2..3: capture-only pseudo-agg slots (DAGG_XCAP). The agg
+     node appends its own aggs AFTER the selection is planned, so reserve
+     the TAIL slots; dev_idx for real aggs starts at 0 and xcap rides the
+     high indices once n_aggs is final (patched in build_plan's agg pass:
+     see xcap relocation below). Here we stage them at the front of
+     sp->aggs beyond n_aggs=0 and record indices. */
+  for (int k = 0; k + 2 < n_slots; k++) {
+    const CoprColumnInfo &ci = pl->cols[slots[2 + k].off];
+    DevAggSpec ds{};
+    ds.kind = DAGG_XCAP;
+    ds.col_id = ci.column_id;
+    ds.col_unsigned = (ci.ft.flag & COPR_FLAG_UNSIGNED) ? 1 : 0;
+    int idx = sp->n_aggs++;
+    if (idx >= COPR_MAX_AGGS) return COPR_ERR_UNSUPPORTED;
+    sp->aggs[idx] = ds;
+    sp->xcap_idx[sp->n_xcap] = idx;
+    copr_status st = fill(ci, &sp->xcap_missing_null[sp->n_xcap],
+                          &sp->xcap_missing_val[sp->n_xcap]);
+    if (st != COPR_OK) return st;
+    sp->n_xcap++;
   }
   sp->rpn_on = 1;
   sp->rpn_n = n;
@@ -982,6 +1003,10 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
   if (!agg) {
     if (sp.index_mode)
       return SET_ERR(COPR_ERR_UNSUPPORTED, "index scan project not yet native");
+    if (sp.n_xcap && !pl->has_topn)
+      return SET_ERR(COPR_ERR_UNSUPPORTED,
+                     "plain project supports predicates over <=2 distinct "
+                     "columns (the project kernel has no capture slots)");
     sp.mode = 0;
     sp.n_out = (int32_t)pl->cols.size();
     if (sp.n_out > COPR_MAX_OUT_COLS)
@@ -1002,6 +1027,12 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
   sp.mode = pl->stream_agg ? 3 : (pl->hash_agg ? 2 : 1);
   if (agg->n_aggs == 0 || agg->n_aggs > COPR_MAX_AGGS)
     return SET_ERR(COPR_ERR_UNSUPPORTED, "agg count out of range");
+  /* the selection pass may have staged capture-only XCAP slots; real aggs
+     take the low dev indices, xcap relocates to the tail afterwards */
+  DevAggSpec xcap_saved[2];
+  int n_xcap_saved = sp.n_xcap;
+  for (int k = 0; k < n_xcap_saved; k++)
+    xcap_saved[k] = sp.aggs[sp.xcap_idx[k]];
   sp.n_aggs = 0;
   for (uint32_t a = 0; a < agg->n_aggs; a++) {
     const CoprAggDef &ad = agg->aggs[a];
@@ -1101,6 +1132,13 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
       pl->out_schema.push_back(cnt_ft);
     }
     pl->out_schema.push_back(ad.out_ft);
+  }
+  /* relocate staged XCAP capture slots to the tail dev indices */
+  for (int k = 0; k < n_xcap_saved; k++) {
+    if (sp.n_aggs >= COPR_MAX_AGGS)
+      return SET_ERR(COPR_ERR_UNSUPPORTED, "agg+predicate column budget");
+    sp.xcap_idx[k] = sp.n_aggs;
+    sp.aggs[sp.n_aggs++] = xcap_saved[k];
   }
   bool any_first = false;
   for (int a = 0; a < sp.n_aggs; a++)

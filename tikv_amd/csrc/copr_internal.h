@@ -70,6 +70,9 @@ enum {
   /* f64 sum: atomicAdd(double) on sum_lo's bits; parallel order makes it
      1-ULP-class, within north_star's float budget */
   DAGG_SUM_REAL,
+  /* capture-only pseudo-agg: the cell value feeds an RPN capture channel
+     (selection over >2 distinct columns); never accumulates or emits */
+  DAGG_XCAP,
 };
 
 /* flattened RPN node for the device predicate evaluator (mirrors the
@@ -131,6 +134,14 @@ struct ScanPlan {
   int32_t rpn_on;
   int32_t rpn_n;
   DevRpnNode rpn[COPR_MAX_RPN];
+  /* RPN capture channels beyond the two filter channels (selection over
+     up to 4 distinct columns, selection_executor.rs:86 ANDs any number):
+     channel 2+k's value rides the capture-only pseudo-agg slot
+     xcap_idx[k] (kind DAGG_XCAP) */
+  int32_t n_xcap;
+  int32_t xcap_idx[2];
+  int32_t xcap_missing_null[2];
+  int64_t xcap_missing_val[2];
   /* 1 = keep every row but still export filt_vals/filt_state: used for a
      column an upstream expression decoded in place (e.g. the TopN order
      column) so the response encodes it in DECODED form

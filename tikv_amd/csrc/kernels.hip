@@ -660,7 +660,8 @@ __device__ static inline bool d_v2_collect(const ScanPlan &plan,
       uint64_t u = d_be_u64(r.vals + s);
       cols[a].iv = (int64_t)((u & 0x8000000000000000ull)
                                  ? (u ^ 0x8000000000000000ull) : ~u);
-    } else if (sp.kind == DAGG_SUM_INT || d_is_fold(sp.kind)) {
+    } else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_XCAP ||
+               d_is_fold(sp.kind)) {
       if (!d_v2_int(r.vals + s, e - s, sp.col_unsigned, &cols[a].iv)) return false;
     }
     /* COUNT_COL: found/null is all that matters */
@@ -797,7 +798,8 @@ __device__ static inline bool d_index_collect(const ScanPlan &plan,
         cols[a].has_dec = cell.has_dec;
         cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
         cols[a].dwide = cell.dwide; cols[a].dwrem = cell.dwide_rem;
-        if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec)
+        if (!cell.is_null && !cell.has_int && !cell.has_real &&
+            !cell.has_dec && !cell.dwide)
           return false;
       }
     }
@@ -940,7 +942,8 @@ __device__ static inline bool d_rpn_keep(const ScanPlan &plan,
                                          bool f1_found, bool f1_null,
                                          int64_t f1_v, bool f2_found,
                                          bool f2_null, int64_t f2_v,
-                                         bool *err) {
+                                         bool *err,
+                                         const AggColView *xcols = nullptr) {
   int64_t sv[4];
   uint8_t sn[4], su[4];
   int sp = 0;
@@ -953,12 +956,32 @@ __device__ static inline bool d_rpn_keep(const ScanPlan &plan,
     if (plan.filter2_missing_null) f2_null = true;
     else { f2_v = plan.filter2_missing_val; f2_null = false; }
   }
+  /* channels 2..3 ride capture-only pseudo-agg slots */
+  int64_t x_v[2] = {0, 0};
+  bool x_null[2] = {false, false};
+  for (int k = 0; k < plan.n_xcap; k++) {
+    if (!xcols) { *err = true; return false; }
+    const AggColView &c = xcols[plan.xcap_idx[k]];
+    if (!c.found) {
+      if (plan.xcap_missing_null[k]) x_null[k] = true;
+      else x_v[k] = plan.xcap_missing_val[k];
+    } else if (c.null) {
+      x_null[k] = true;
+    } else {
+      x_v[k] = c.iv;
+    }
+  }
   for (int i = 0; i < plan.rpn_n; i++) {
     const DevRpnNode &nd = plan.rpn[i];
     if (nd.kind == 0) {
       if (sp >= 4) { *err = true; return false; }
-      sv[sp] = nd.slot ? f2_v : f1_v;
-      sn[sp] = nd.slot ? (f2_null ? 1 : 0) : (f1_null ? 1 : 0);
+      if (nd.slot >= 2) {
+        sv[sp] = x_v[nd.slot - 2];
+        sn[sp] = x_null[nd.slot - 2] ? 1 : 0;
+      } else {
+        sv[sp] = nd.slot ? f2_v : f1_v;
+        sn[sp] = nd.slot ? (f2_null ? 1 : 0) : (f1_null ? 1 : 0);
+      }
       su[sp] = (uint8_t)nd.uns;
       sp++;
     } else if (nd.kind == 1) {
@@ -1043,10 +1066,11 @@ __device__ static inline bool d_rpn_keep(const ScanPlan &plan,
 __device__ static inline bool d_keep2(const ScanPlan &plan, bool f1_found,
                                       bool f1_null, int64_t f1_v,
                                       bool f2_found, bool f2_null,
-                                      int64_t f2_v, bool *err) {
+                                      int64_t f2_v, bool *err,
+                                      const AggColView *xcols = nullptr) {
   if (plan.rpn_on)
     return d_rpn_keep(plan, f1_found, f1_null, f1_v, f2_found, f2_null, f2_v,
-                      err);
+                      err, xcols);
   return d_filter_keep(plan, f1_found, f1_null, f1_v) &&
          d_filter2_keep(plan, f2_found, f2_null, f2_v);
 }
@@ -1320,7 +1344,8 @@ k_scan_agg(ScanPlan plan,
                 cols[a].has_dec = cell.has_dec;
                 cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
                 cols[a].dwide = cell.dwide; cols[a].dwrem = cell.dwide_rem;
-                if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec)
+                if (!cell.is_null && !cell.has_int && !cell.has_real &&
+                    !cell.has_dec && !cell.dwide)
                   parse_ok = false;
               } else parse_ok = false;
             }
@@ -1374,7 +1399,7 @@ k_scan_agg(ScanPlan plan,
               cols[a].has_dec = cell.has_dec;
               cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
               cols[a].dwide = cell.dwide; cols[a].dwrem = cell.dwide_rem;
-              if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec) parse_ok = false;
+              if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec && !cell.dwide) parse_ok = false;
               found++;
             }
           }
@@ -1387,7 +1412,7 @@ k_scan_agg(ScanPlan plan,
         any_parse_err = true;
       } else if (bool ke = false;
                  d_keep2(plan, filt_found, filt_null, filt_v, f2_found,
-                         f2_null, f2_v, &ke)
+                         f2_null, f2_v, &ke, cols)
                      ? true
                      : (ke ? (any_parse_err = true, false) : false)) {
         SimpleAggAcc *acc_base = nullptr;
@@ -1473,7 +1498,9 @@ k_scan_agg(ScanPlan plan,
           bool dec_wide = false;       /* value does not fit a scaled i64 */
           __int128 vw = 0;
           int64_t v = 0;
-          if (sp.kind == DAGG_COUNT_ROWS) {
+          if (sp.kind == DAGG_XCAP) {
+            continue;                    /* capture-only channel */
+          } else if (sp.kind == DAGG_COUNT_ROWS) {
             contribute = true;
           } else if (!cols[a].found || cols[a].null) {
             contribute = false;
@@ -1529,12 +1556,14 @@ k_scan_agg(ScanPlan plan,
                 atomicAdd((double *)&acc_base[a].sum_lo,
                           __longlong_as_double(v));
               else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC) {
-                if (dec_wide) {
-                  if (ext_base)
-                    atomic_add_i256(&acc_base[a].sum_lo, &acc_base[a].sum_hi,
-                                    ext_base + a * 2, vw);
-                  else
-                    any_parse_err = true;   /* no ext buffers: loud */
+                /* a 256-bit acc must see EVERY add as i256 — a narrow
+                   negative i128 add would not borrow from the ext limbs */
+                if (ext_base) {
+                  atomic_add_i256(&acc_base[a].sum_lo, &acc_base[a].sum_hi,
+                                  ext_base + a * 2,
+                                  dec_wide ? vw : (__int128)v);
+                } else if (dec_wide) {
+                  any_parse_err = true;     /* no ext buffers: loud */
                 } else {
                   atomic_add_i128(&acc_base[a].sum_lo, &acc_base[a].sum_hi, v);
                 }
@@ -2079,7 +2108,8 @@ k_scan_agg_pipe(ScanPlan plan,
                 cols[a].has_dec = cell.has_dec;
                 cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
                 cols[a].dwide = cell.dwide; cols[a].dwrem = cell.dwide_rem;
-                if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec)
+                if (!cell.is_null && !cell.has_int && !cell.has_real &&
+                    !cell.has_dec && !cell.dwide)
                   parse_ok = false;
               } else parse_ok = false;
             }
@@ -2133,7 +2163,7 @@ k_scan_agg_pipe(ScanPlan plan,
               cols[a].has_dec = cell.has_dec;
               cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
               cols[a].dwide = cell.dwide; cols[a].dwrem = cell.dwide_rem;
-              if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec) parse_ok = false;
+              if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec && !cell.dwide) parse_ok = false;
               found++;
             }
           }
@@ -2146,7 +2176,7 @@ k_scan_agg_pipe(ScanPlan plan,
         any_parse_err = true;
       } else if (bool ke = false;
                  d_keep2(plan, filt_found, filt_null, filt_v, f2_found,
-                         f2_null, f2_v, &ke)
+                         f2_null, f2_v, &ke, cols)
                      ? true
                      : (ke ? (any_parse_err = true, false) : false)) {
         SimpleAggAcc *acc_base = nullptr;
@@ -2232,7 +2262,9 @@ k_scan_agg_pipe(ScanPlan plan,
           bool dec_wide = false;       /* value does not fit a scaled i64 */
           __int128 vw = 0;
           int64_t v = 0;
-          if (sp.kind == DAGG_COUNT_ROWS) {
+          if (sp.kind == DAGG_XCAP) {
+            continue;                    /* capture-only channel */
+          } else if (sp.kind == DAGG_COUNT_ROWS) {
             contribute = true;
           } else if (!cols[a].found || cols[a].null) {
             contribute = false;
@@ -2288,12 +2320,14 @@ k_scan_agg_pipe(ScanPlan plan,
                 atomicAdd((double *)&acc_base[a].sum_lo,
                           __longlong_as_double(v));
               else if (sp.kind == DAGG_SUM_INT || sp.kind == DAGG_SUM_DEC) {
-                if (dec_wide) {
-                  if (ext_base)
-                    atomic_add_i256(&acc_base[a].sum_lo, &acc_base[a].sum_hi,
-                                    ext_base + a * 2, vw);
-                  else
-                    any_parse_err = true;   /* no ext buffers: loud */
+                /* a 256-bit acc must see EVERY add as i256 — a narrow
+                   negative i128 add would not borrow from the ext limbs */
+                if (ext_base) {
+                  atomic_add_i256(&acc_base[a].sum_lo, &acc_base[a].sum_hi,
+                                  ext_base + a * 2,
+                                  dec_wide ? vw : (__int128)v);
+                } else if (dec_wide) {
+                  any_parse_err = true;     /* no ext buffers: loud */
                 } else {
                   atomic_add_i128(&acc_base[a].sum_lo, &acc_base[a].sum_hi, v);
                 }
@@ -4150,7 +4184,8 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
                 cols[a].has_dec = cell.has_dec;
                 cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
                 cols[a].dwide = cell.dwide; cols[a].dwrem = cell.dwide_rem;
-                if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec)
+                if (!cell.is_null && !cell.has_int && !cell.has_real &&
+                    !cell.has_dec && !cell.dwide)
                   parse_ok = false;
               } else parse_ok = false;
             }
@@ -4236,7 +4271,8 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
                 cols[a].has_dec = cell.has_dec;
                 cols[a].dsc = cell.dscaled; cols[a].dfr = cell.dfrac;
                 cols[a].dwide = cell.dwide; cols[a].dwrem = cell.dwide_rem;
-                if (!cell.is_null && !cell.has_int && !cell.has_real && !cell.has_dec)
+                if (!cell.is_null && !cell.has_int && !cell.has_real &&
+                    !cell.has_dec && !cell.dwide)
                   parse_ok = false;
                 found++;
               }
@@ -4251,7 +4287,7 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
         any_err = true;
       } else if (bool ke = false;
                  d_keep2(plan, filt_found, filt_null, filt_v, f2_found,
-                         f2_null, f2_v, &ke)
+                         f2_null, f2_v, &ke, cols)
                      ? true
                      : (ke ? (any_err = true, false) : false)) {
         s = (grp_found && !grp_null) ? 2 : 1;
@@ -4270,7 +4306,9 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
         uint8_t contribute = 0;
         int64_t v = 0;
         if (s) {
-          if (sp.kind == DAGG_FIRST) {
+          if (sp.kind == DAGG_XCAP) {
+            /* capture-only channel: no contribution downstream */
+          } else if (sp.kind == DAGG_FIRST) {
             /* first row's value, NULL (or missing->NULL) included */
             contribute = (cols[a].found && !cols[a].null) ? 1 : 2;
             v = cols[a].iv;
