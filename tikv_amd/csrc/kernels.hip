@@ -3307,11 +3307,13 @@ k_scan_project(ScanPlan plan,
  * per step and the stage/parse barrier phases serialized on top
  * (profiles/r02_cfg4_*). More independent chains per SIMD is the fix, not
  * more LDS bandwidth. */
+/* always_inline: with several kernels (different launch-bounds budgets)
+ * calling this, hipcc otherwise OUTLINES it once sized for the tightest
+ * caller — measured as 44 B/lane scratch spill and a ~2x CRC slowdown */
 template <bool PREFETCH>
-__device__ static inline uint64_t d_crc64_stream(const uint8_t *__restrict__ base,
-                                                 uint64_t b0, uint64_t b1,
-                                                 uint64_t crc,
-                                                 const uint64_t *__restrict__ tab) {
+__device__ __attribute__((always_inline)) static inline uint64_t
+d_crc64_stream(const uint8_t *__restrict__ base, uint64_t b0, uint64_t b1,
+               uint64_t crc, const uint64_t *__restrict__ tab) {
   uint64_t len = b1 - b0;
   if (!len) return crc;
   /* aligned u64 stream with a funnel shift; region buffers carry +2 KiB
