@@ -401,3 +401,52 @@ def test_index_value_handle_simple_agg(engine, layout):
         assert o == g
     finally:
         gen.close()
+
+
+def test_index_project_parity(engine):
+    """plain index-scan project: positional raw datum spans + handle
+    (key-form and value-form), vs the oracle."""
+    for layout in (0, 1, 2):
+        gen = tikv_amd.GenRegion(config_index=4, n_rows=40000, table_id=1,
+                                 n_cols=layout)
+        try:
+            cols = [tikv_amd.Col(1), tikv_amd.Col(2),
+                    tikv_amd.Col(-1, pk_handle=True)]
+            req = tikv_amd.DagSelect(cols, index=True).build()
+            (o, on), (g, gn) = run_both(req, gen, engine)
+            assert on == gn == 40000, layout
+            assert o == g, layout
+        finally:
+            gen.close()
+
+
+def test_index_project_filtered_limit(engine):
+    gen = tikv_amd.GenRegion(config_index=4, n_rows=60000, table_id=1)
+    try:
+        cols = [tikv_amd.Col(1), tikv_amd.Col(2),
+                tikv_amd.Col(-1, pk_handle=True)]
+        sel = tikv_amd.cmp_col_const(1, F.SIG_GE_INT, 0)
+        req = (tikv_amd.DagSelect(cols, index=True).where(sel)
+               .limit(5000).build())
+        (o, on), (g, gn) = run_both(req, gen, engine)
+        assert on == gn == 5000
+        assert o == g
+    finally:
+        gen.close()
+
+
+def test_index_topn_parity(engine):
+    """TopN over an index scan (top_n_executor.rs over
+    BatchIndexScanExecutor): winners + order, vs the oracle."""
+    for desc in (False, True):
+        gen = tikv_amd.GenRegion(config_index=4, n_rows=50000, table_id=1)
+        try:
+            cols = [tikv_amd.Col(1), tikv_amd.Col(2),
+                    tikv_amd.Col(-1, pk_handle=True)]
+            req = (tikv_amd.DagSelect(cols, index=True)
+                   .topn(tikv_amd.Expr().col(1), 37, desc=desc).build())
+            (o, on), (g, gn) = run_both(req, gen, engine)
+            assert on == gn == 37, desc
+            assert o == g, desc
+        finally:
+            gen.close()

@@ -94,18 +94,21 @@ def region(rows):
             len(rows), (kb, vb))
 
 
+# cell fracs <= the sum's target frac (the engine scales values UP to the
+# output type's frac; a cell with more fraction digits than the target is a
+# loud error by design)
 WIDE_VALS = [
-    (10**37 - 1, 2),            # 37 digits
+    (10**37 - 1, 4),            # 37 digits
     (-(10**30 + 12345), 2),
-    (98765432109876543210987654321, 4),   # 29 digits, frac 4 -> scale up
+    (98765432109876543210987654321, 4),   # 29 digits
     (5, 2),                     # tiny narrow among wide
-    (-(10**19), 2),
+    (-(10**19), 0),
 ]
 
 
 def build_req(group=False):
-    cols = [tikv_amd.Col(1), tikv_amd.Col(2, tp=F.TP_NEWDECIMAL, decimal=2)]
-    aggs = [tikv_amd.count_star(), tikv_amd.sum_col(1, decimal=2)]
+    cols = [tikv_amd.Col(1), tikv_amd.Col(2, tp=F.TP_NEWDECIMAL, decimal=4)]
+    aggs = [tikv_amd.count_star(), tikv_amd.sum_col(1, decimal=4)]
     if group:
         return tikv_amd.DagSelect(cols).hash_agg(
             aggs, tikv_amd.Expr().col(0)).build()

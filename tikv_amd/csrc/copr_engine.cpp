@@ -968,8 +968,6 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
         break;
       case COPR_EXEC_TOPN: {
         if (pl->has_topn) return SET_ERR(COPR_ERR_UNSUPPORTED, "one TopN supported");
-        if (sp.index_mode)
-          return SET_ERR(COPR_ERR_UNSUPPORTED, "TopN over index scan not yet native");
         if (ex.n_order_by != 1)
           return SET_ERR(COPR_ERR_UNSUPPORTED, "one order-by expression supported");
         const CoprExpr &oe = ex.order_by[0];
@@ -1001,8 +999,9 @@ static copr_status build_plan(const CoprDagRequest *req, HostPlan *pl) {
   if (agg && pl->has_topn)
     return SET_ERR(COPR_ERR_UNSUPPORTED, "TopN with aggregation unsupported");
   if (!agg) {
-    if (sp.index_mode)
-      return SET_ERR(COPR_ERR_UNSUPPORTED, "index scan project not yet native");
+    if (sp.index_mode && sp.rpn_on)
+      return SET_ERR(COPR_ERR_UNSUPPORTED,
+                     "index project supports cmp(col,const) predicates only");
     if (sp.n_xcap && !pl->has_topn)
       return SET_ERR(COPR_ERR_UNSUPPORTED,
                      "plain project supports predicates over <=2 distinct "
@@ -2230,12 +2229,18 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
       if (ce != hipSuccess) return SET_ERR(COPR_ERR_INTERNAL, hipGetErrorString(ce));
       if (h_err[1]) return SET_ERR(COPR_ERR_STORAGE, "row parse error on device");
 
-      /* pull value bytes for kept rows (whole buffer if small) */
+      /* pull the SPAN-SOURCE bytes for kept rows (whole buffer if small):
+         index project spans reference the KEY stream */
+      const bool idxp = pl.sp.index_mode != 0;
+      const uint8_t *d_span_src = idxp ? r->dev.d_keys : r->dev.d_vals;
+      const uint64_t span_bytes = idxp ? r->dev.key_bytes : r->dev.val_bytes;
+      const std::vector<uint64_t> &span_offs =
+          idxp ? r->h_key_offs : r->h_val_offs;
       std::vector<uint8_t> h_vals;
-      bool whole = r->dev.val_bytes <= (256u << 20);
+      bool whole = span_bytes <= (256u << 20);
       if (whole) {
-        h_vals.resize(r->dev.val_bytes);
-        HIP_TRY(hipMemcpy(h_vals.data(), r->dev.d_vals, r->dev.val_bytes,
+        h_vals.resize(span_bytes);
+        HIP_TRY(hipMemcpy(h_vals.data(), d_span_src, span_bytes,
                           hipMemcpyDeviceToHost), "vals D2H");
       }
       /* paging (runner.rs:917-943): walk the reference's batch ladder
@@ -2279,9 +2284,9 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
         const uint8_t *vbase = nullptr;
         if (whole) vbase = h_vals.data();
         else {
-          uint64_t vo = r->h_val_offs[i], vl = r->h_val_offs[i + 1] - vo;
+          uint64_t vo = span_offs[i], vl = span_offs[i + 1] - vo;
           rowbuf.resize(vl);
-          hipMemcpy(rowbuf.data(), r->dev.d_vals + vo, vl, hipMemcpyDeviceToHost);
+          hipMemcpy(rowbuf.data(), d_span_src + vo, vl, hipMemcpyDeviceToHost);
           vbase = rowbuf.data() - vo;
         }
         for (uint32_t oo = 0; oo < req->n_output_offsets; oo++) {
@@ -2338,9 +2343,9 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
             resp.push_back(0);                 /* explicit v2 NULL cell */
           } else {
             uint64_t goff = cp >> 20;
-            uint64_t rs = r->h_val_offs[i];
-            uint64_t rl = r->h_val_offs[i + 1] - rs;
-            if (rl > 1 && vbase[rs] == 128) {
+            uint64_t rs = span_offs[i];
+            uint64_t rl = span_offs[i + 1] - rs;
+            if (!idxp && rl > 1 && vbase[rs] == 128) {
               /* v2 row: raw payload -> datum re-encode */
               const CoprColumnInfo &ci = pl.cols[off];
               if (!e_v2_cell_to_datum(ci.ft.tp, ci.ft.flag, vbase + goff,

@@ -3184,12 +3184,80 @@ k_scan_project(ScanPlan plan,
       bool filt_found = false, filt_null = false; int64_t filt_v = 0;
       bool d2_found = false, d2_null = false;
       int64_t d2_v = 0;
+      int64_t idx_handle = 0;
       unsigned long long cell_pack[COPR_MAX_OUT_COLS];
       for (int j = 0; j < plan.n_out; j++) cell_pack[j] = 0xFFFFFull;
       int needed = (plan.has_filter ? 1 : 0) + (plan.dec2_col_id ? 1 : 0) +
                    plan.n_out;
       int found = 0;
 
+      if (plan.index_mode) {
+        /* index project: vp IS the index key (the launcher streams the key
+           stream); out cols are POSITIONAL raw datum spans in the key
+           (extract_columns_from_datum_format :504-517); the handle comes
+           from the trailing key datum or the value
+           (d_index_value_split). V4 restore-data rows are not native in
+           project mode (columns would live in the value's row-v2). */
+        if (vlen < 19 || vp[0] != 't' || vp[9] != '_' || vp[10] != 'i') {
+          parse_ok = false;
+        } else {
+          bool hfound = false;
+          uint32_t pos = 19;
+          int64_t ci = 0;
+          while (pos < vlen && parse_ok) {
+            CellView cell;
+            d_parse_datum(vp + pos, vlen - pos, &cell);
+            if (!cell.len) { parse_ok = false; break; }
+            if (ci == (int64_t)plan.index_n_cols) {
+              if (!cell.has_int) { parse_ok = false; break; }
+              idx_handle = cell.ival;
+              hfound = true;
+            }
+            for (int j = 0; j < plan.n_out; j++) {
+              if (plan.out_is_handle[j]) continue;
+              if (plan.out_col_ids[j] == ci) {
+                uint64_t goff = val_offs[my_row] + pos;
+                if (cell.len >= 0xFFFFEu) { parse_ok = false; break; }
+                cell_pack[j] = (goff << 20) | cell.len;
+              }
+            }
+            if (plan.has_filter && !filt_found && ci == plan.filter_col_id) {
+              filt_found = true;
+              if (cell.is_null) filt_null = true;
+              else if (cell.has_int) filt_v = cell.ival;
+              else parse_ok = false;
+            }
+            if (plan.dec2_col_id && !d2_found && ci == plan.dec2_col_id) {
+              d2_found = true;
+              if (cell.is_null) d2_null = true;
+              else if (cell.has_int) d2_v = cell.ival;
+              else parse_ok = false;
+            }
+            pos += cell.len;
+            ci++;
+          }
+          if (parse_ok && !hfound && plan.aux_vals) {
+            uint64_t o0 = plan.aux_val_offs[my_row];
+            uint64_t o1 = plan.aux_val_offs[my_row + 1];
+            int64_t vh = 0;
+            bool has_h = false;
+            const uint8_t *restore = nullptr;
+            uint32_t rl = 0;
+            if (!d_index_value_split(plan.aux_vals + o0, (uint32_t)(o1 - o0),
+                                     &vh, &has_h, &restore, &rl) ||
+                restore)
+              parse_ok = false;
+            else if (has_h) {
+              idx_handle = vh;
+              hfound = true;
+            }
+          }
+          bool need_handle = false;
+          for (int j = 0; j < plan.n_out; j++)
+            if (plan.out_is_handle[j]) need_handle = true;
+          if (parse_ok && need_handle && !hfound) parse_ok = false;
+        }
+      } else
       if (vlen > 1 && vp[0] == 128) {
         /* row v2: direct column lookup; value cells are RAW payloads, the
            host re-encodes them as datums (compat_v1.rs:28-126). 0xFFFFE
@@ -3280,9 +3348,13 @@ k_scan_project(ScanPlan plan,
         if (keep) {
           for (int j = 0; j < plan.n_out; j++) {
             if (plan.out_is_handle[j]) {
-              const uint8_t *kp = keys + key_offs[my_row];
-              uint64_t h = d_be_u64(kp + 11) ^ 0x8000000000000000ull;
-              po.handles[my_row] = (long long)h;
+              if (plan.index_mode) {
+                po.handles[my_row] = idx_handle;
+              } else {
+                const uint8_t *kp = keys + key_offs[my_row];
+                uint64_t h = d_be_u64(kp + 11) ^ 0x8000000000000000ull;
+                po.handles[my_row] = (long long)h;
+              }
             } else {
               po.cells[my_row * plan.n_out + j] = cell_pack[j];
             }
@@ -4111,6 +4183,16 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
       #pragma unroll
       for (int a = 0; a < NAGGS; a++) cols[a] = {false, false, false, 0, 0, 0, nullptr, 0};
 
+      if (plan.index_mode) {
+        /* vp IS the index key (extract_launch streams the key slab) */
+        int64_t hval = 0;
+        bool hfound = false;
+        parse_ok = !GBYTES &&
+                   d_index_collect<(NAGGS > 0 ? NAGGS : 1), true>(
+                       plan, vp, vlen, my_row, &filt_found, &filt_null,
+                       &filt_v, &grp_found, &grp_null, &grp_v, cols, &hval,
+                       &hfound);
+      } else
       if (!(vlen == 0 || (vlen == 1 && vp[0] == 0))) {
         bool dir_done = false;
         if (plan.celldir && vp[0] != 128) {
@@ -4449,11 +4531,26 @@ static int extract_launch(const ScanPlan &plan, const DevRegion &rgn,
   uint64_t n_tiles = (rgn.n_kv + plan.rows_per_tile - 1) / plan.rows_per_tile;
   uint32_t grid = (uint32_t)(n_tiles < 4096 ? n_tiles : 4096);
   if (grid == 0) grid = 1;
+  /* index mode streams the KEY slab; the value stream rides the aux
+     pointers (unique-index handles) */
+  ScanPlan p2 = plan;
+  DevRegion r2 = rgn;
+  if (plan.index_mode) {
+    r2.d_vals = rgn.d_keys;
+    r2.d_val_offs = rgn.d_key_offs;
+    r2.val_bytes = rgn.key_bytes;
+    r2.max_row_bytes = rgn.max_key_bytes;
+    p2.aux_vals = rgn.d_vals;
+    p2.aux_val_offs = rgn.d_val_offs;
+    p2.aux_max_vlen = rgn.max_row_bytes;
+  }
+  const ScanPlan &plan_ = p2;
+  const DevRegion &rgn_ = r2;
   #define XCASE(NA)                                                           \
     case NA:                                                                  \
       hipLaunchKernelGGL((k_scan_extract<NA>), dim3(grid), dim3(256),         \
-                         plan.lds_bytes, s, plan, rgn.d_vals, rgn.d_val_offs, \
-                         rgn.n_kv, eo, d_err);                                \
+                         plan_.lds_bytes, s, plan_, rgn_.d_vals,              \
+                         rgn_.d_val_offs, rgn_.n_kv, eo, d_err);              \
       break
   switch (plan.n_aggs) {
     XCASE(0); XCASE(1); XCASE(2); XCASE(3); XCASE(4);
@@ -5597,8 +5694,10 @@ int dev_scan_launch(const ScanPlan &plan, const DevRegion &rgn,
     return launch_agg<false>(p2, r2, d_simple, HashAggTable{}, s, grid);
   if (plan.mode == 2)
     return launch_agg<true>(p2, r2, nullptr, *ht, s, grid);
+  /* project: index mode streams the KEY slab (r2 swap) and reads values
+     only through the aux pointers (value handles) */
   hipLaunchKernelGGL(k_scan_project, dim3(grid), dim3(THREADS), plan.lds_bytes,
-                     s, plan, rgn.d_vals, rgn.d_val_offs, rgn.d_keys,
+                     s, p2, r2.d_vals, r2.d_val_offs, rgn.d_keys,
                      rgn.d_key_offs, rgn.n_kv, *po);
   return (int)hipGetLastError();
 }
