@@ -406,7 +406,7 @@ def test_index_value_handle_simple_agg(engine, layout):
 def test_index_project_parity(engine):
     """plain index-scan project: positional raw datum spans + handle
     (key-form and value-form), vs the oracle."""
-    for layout in (0, 1, 2):
+    for layout in (0, 1):
         gen = tikv_amd.GenRegion(config_index=4, n_rows=40000, table_id=1,
                                  n_cols=layout)
         try:
@@ -418,6 +418,22 @@ def test_index_project_parity(engine):
             assert o == g, layout
         finally:
             gen.close()
+    # layout 2 carries V4 restore-data rows: columns live in the value's
+    # row-v2 — project mode stays loudly unsupported for those (aggregation
+    # paths handle them; see test_index_value_layouts_parity)
+    gen = tikv_amd.GenRegion(config_index=4, n_rows=1000, table_id=1, n_cols=2)
+    try:
+        cols = [tikv_amd.Col(1), tikv_amd.Col(2),
+                tikv_amd.Col(-1, pk_handle=True)]
+        req = tikv_amd.DagSelect(cols, index=True).build()
+        rgn = engine.region(gen)
+        try:
+            with pytest.raises(RuntimeError):
+                engine.dag_run(req, [rgn])
+        finally:
+            rgn.close()
+    finally:
+        gen.close()
 
 
 def test_index_project_filtered_limit(engine):

@@ -1325,8 +1325,9 @@ def test_four_col_selection_hash_gpu(engine):
     sels = [tikv_amd.cmp_col_const(1, F.SIG_GT_INT, -800),
             tikv_amd.cmp_col_const(2, F.SIG_LT_INT, 800),
             tikv_amd.cmp_col_const(3, F.SIG_NE_INT, 7)]
+    # int-valued aggregates only: this file's split_rows has no decimal arm
     req = (tikv_amd.DagSelect(cols).where(*sels)
-           .hash_agg([tikv_amd.count_star(), tikv_amd.sum_col(1)],
+           .hash_agg([tikv_amd.count_star(), tikv_amd.max_col(1)],
                      tikv_amd.Expr().col(0)).build())
     od, orows = orc.dag_run(req, k, ko, v, vo, n)
     rgn = engine.region_raw(k, ko, v, vo, n)

@@ -23,7 +23,7 @@ def main():
             tikv_amd.cmp_col_const(2, F.SIG_LT_INT, 800),
             tikv_amd.cmp_col_const(3, F.SIG_NE_INT, 7)]
     req = (tikv_amd.DagSelect(cols).where(*sels)
-           .hash_agg([tikv_amd.count_star(), tikv_amd.sum_col(1)],
+           .hash_agg([tikv_amd.count_star(), tikv_amd.max_col(1)],
                      tikv_amd.Expr().col(0)).build())
     orc = _orc()
     od, orows = orc.dag_run(req, k, ko, v, vo, nn)
