@@ -67,6 +67,12 @@ extern "C" void copr_engine_destroy(copr_engine *eng) {
   if (!eng) return;
   copr::comm_free(eng);
   if (eng->d_crc_tables) hipFree(eng->d_crc_tables);
+  {
+    copr::HashAggTable &c = eng->ht_cache;
+    hipFree(c.keys); hipFree(c.accs); hipFree(c.reserved);
+    hipFree(c.ext); hipFree(c.rsvd_ext);
+    hipFree(c.rsvd_seen); hipFree(c.error); hipFree(c.n_groups);
+  }
   if (eng->stream) hipStreamDestroy(eng->stream);
   delete eng;
 }
@@ -2011,43 +2017,55 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
          The device-side compaction keeps the readback proportional to
          n_groups when the table does grow. */
       uint32_t tsize = 1u << 16;
+      bool any_dec = false;
+      for (int a = 0; a < pl.sp.n_aggs; a++)
+        if (pl.sp.aggs[a].kind == DAGG_SUM_DEC) any_dec = true;
+      auto ht_cache_drop = [&]() {
+        HashAggTable &c = eng->ht_cache;
+        hipFree(c.keys); hipFree(c.accs); hipFree(c.reserved);
+        hipFree(c.ext); hipFree(c.rsvd_ext);
+        hipFree(c.rsvd_seen); hipFree(c.error); hipFree(c.n_groups);
+        c = HashAggTable{};
+        eng->ht_cache_tsize = 0;
+      };
       for (int attempt = 0; attempt < 4; attempt++) {
-        HashAggTable ht{};
-        hipError_t e = hipSuccess;
-        size_t keys_b = (size_t)tsize * 8;
-        size_t accs_b = (size_t)tsize * pl.sp.n_aggs * sizeof(SimpleAggAcc);
-        bool any_dec = false;
-        for (int a = 0; a < pl.sp.n_aggs; a++)
-          if (pl.sp.aggs[a].kind == DAGG_SUM_DEC) any_dec = true;
-        if (e == hipSuccess) e = hipMalloc((void **)&ht.keys, keys_b);
-        if (e == hipSuccess) e = hipMalloc((void **)&ht.accs, accs_b);
-        if (e == hipSuccess) e = hipMalloc((void **)&ht.reserved, 2 * pl.sp.n_aggs * sizeof(SimpleAggAcc));
-        if (e == hipSuccess && any_dec)
-          e = hipMalloc((void **)&ht.ext, (size_t)tsize * pl.sp.n_aggs * 16);
-        if (e == hipSuccess && any_dec)
-          e = hipMalloc((void **)&ht.rsvd_ext, 2 * pl.sp.n_aggs * 16);
-        if (e == hipSuccess) e = hipMalloc((void **)&ht.rsvd_seen, 2 * 8);
-        if (e == hipSuccess) e = hipMalloc((void **)&ht.error, 8);
-        if (e == hipSuccess) e = hipMalloc((void **)&ht.n_groups, 8);
-        auto free_ht = [&]() {
-          hipFree(ht.keys); hipFree(ht.accs); hipFree(ht.reserved);
-          hipFree(ht.ext); hipFree(ht.rsvd_ext);
-          hipFree(ht.rsvd_seen); hipFree(ht.error); hipFree(ht.n_groups);
-        };
-        if (e != hipSuccess) { free_ht(); return SET_ERR(COPR_ERR_OOM, "hash table alloc"); }
-        /* init keys to EMPTY (0x80 bytes = 0x8080.. != EMPTY!) — memset per
-           8-byte word needed: EMPTY = INT64_MIN = 0x8000000000000000; byte
-           pattern not uniform, so use a small fill kernel via hipMemsetD32?
-           Simplest portable: fill on host once for the first attempt sizes
-           is expensive; use hipMemset of 0xFF then treat EMPTY=-1? -1 is a
-           plausible key. Instead fill with hipMemsetD32 pattern:
-           lo=0x00000000, hi=0x80000000 via two strided memsets is not
-           available — do a device-side fill with hipMemcpy of a host
-           pattern buffer (tsize*8 <= 1 GiB worst case). */
+        /* the table buffers persist on the engine across requests: the
+           per-request hipMalloc/hipFree set measured ~3 ms at cfg3 scale */
+        if (eng->ht_cache_tsize != tsize ||
+            eng->ht_cache_naggs != pl.sp.n_aggs ||
+            eng->ht_cache_ext != any_dec || !eng->ht_cache.keys) {
+          ht_cache_drop();
+          HashAggTable c{};
+          hipError_t e = hipSuccess;
+          if (e == hipSuccess) e = hipMalloc((void **)&c.keys, (size_t)tsize * 8);
+          if (e == hipSuccess) e = hipMalloc((void **)&c.accs, (size_t)tsize * pl.sp.n_aggs * sizeof(SimpleAggAcc));
+          if (e == hipSuccess) e = hipMalloc((void **)&c.reserved, 2 * pl.sp.n_aggs * sizeof(SimpleAggAcc));
+          if (e == hipSuccess && any_dec)
+            e = hipMalloc((void **)&c.ext, (size_t)tsize * pl.sp.n_aggs * 16);
+          if (e == hipSuccess && any_dec)
+            e = hipMalloc((void **)&c.rsvd_ext, 2 * pl.sp.n_aggs * 16);
+          if (e == hipSuccess) e = hipMalloc((void **)&c.rsvd_seen, 2 * 8);
+          if (e == hipSuccess) e = hipMalloc((void **)&c.error, 8);
+          if (e == hipSuccess) e = hipMalloc((void **)&c.n_groups, 8);
+          if (e != hipSuccess) {
+            eng->ht_cache = c;
+            ht_cache_drop();
+            return SET_ERR(COPR_ERR_OOM, "hash table alloc");
+          }
+          eng->ht_cache = c;
+          eng->ht_cache_tsize = tsize;
+          eng->ht_cache_naggs = pl.sp.n_aggs;
+          eng->ht_cache_ext = any_dec;
+        }
+        HashAggTable ht = eng->ht_cache;
+        auto free_ht = [&]() { ht_cache_drop(); };
+        /* re-init for this request (async; EMPTY = INT64_MIN needs a fill
+           kernel, the byte pattern is not memset-able) */
         {
-          std::vector<long long> fill(tsize, (long long)0x8000000000000000ll);
-          e = hipMemcpyAsync(ht.keys, fill.data(), keys_b, hipMemcpyHostToDevice, eng->stream);
-          if (e == hipSuccess) e = hipMemsetAsync(ht.accs, 0, accs_b, eng->stream);
+          hipError_t e = hipSuccess;
+          dev_fill_keys(ht.keys, tsize, eng->stream);
+          e = hipGetLastError();
+          if (e == hipSuccess) e = hipMemsetAsync(ht.accs, 0, (size_t)tsize * pl.sp.n_aggs * sizeof(SimpleAggAcc), eng->stream);
           if (e == hipSuccess && ht.ext)
             e = hipMemsetAsync(ht.ext, 0, (size_t)tsize * pl.sp.n_aggs * 16, eng->stream);
           if (e == hipSuccess && ht.rsvd_ext)
@@ -2056,7 +2074,6 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
           if (e == hipSuccess) e = hipMemsetAsync(ht.rsvd_seen, 0, 16, eng->stream);
           if (e == hipSuccess) e = hipMemsetAsync(ht.error, 0, 8, eng->stream);
           if (e == hipSuccess) e = hipMemsetAsync(ht.n_groups, 0, 8, eng->stream);
-          if (e == hipSuccess) e = hipStreamSynchronize(eng->stream);
           if (e != hipSuccess) { free_ht(); return SET_ERR(COPR_ERR_INTERNAL, "ht init"); }
         }
         hipEventRecord(ev_a, eng->stream);
@@ -2126,7 +2143,7 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
         if (ce == hipSuccess && ht.rsvd_ext)
           ce = hipMemcpy(h_rsvd_ext.data(), ht.rsvd_ext,
                          2 * pl.sp.n_aggs * 16, hipMemcpyDeviceToHost);
-        free_ht();
+        /* buffers stay cached on the engine for the next request */
         if (ng == -2) return SET_ERR(COPR_ERR_OOM, "table compact alloc");
         if (ng < 0 || ce != hipSuccess)
           return SET_ERR(COPR_ERR_INTERNAL, "table compact failed");

@@ -285,6 +285,8 @@ int dev_stream_agg(const ScanPlan &plan, const DevRegion &rgn, void *stream,
 int dev_topn_select(const ScanPlan &plan, const DevRegion &rgn,
                     uint64_t topn_n, int desc, void *stream,
                     std::vector<uint32_t> *winners);
+/* fill keys[] with the EMPTY sentinel (INT64_MIN) */
+void dev_fill_keys(long long *keys, uint64_t n, void *stream);
 int dev_ht_compact(const HashAggTable &ht, uint32_t tsize, int n_aggs,
                    void *stream, std::vector<long long> *h_keys,
                    std::vector<SimpleAggAcc> *h_accs,
@@ -325,6 +327,13 @@ struct copr_engine {
   int dec2_col_off = -1;
   /* RCCL communicator state (copr_comm.cpp); null until copr_comm_create */
   void *comm_state = nullptr;
+  /* hash-agg table cache: the per-request hipMalloc/hipFree set measured
+     ~3 ms/step at cfg3 scale; buffers persist across dag_run calls and
+     re-init with async fills */
+  copr::HashAggTable ht_cache{};
+  uint32_t ht_cache_tsize = 0;
+  int ht_cache_naggs = 0;
+  bool ht_cache_ext = false;
 };
 
 struct copr_region {

@@ -4820,6 +4820,17 @@ int dev_topn_select(const ScanPlan &plan, const DevRegion &rgn,
 /* compact a global hash table's occupied slots (keys + accumulator rows)
  * into dense arrays so the host readback scales with n_groups, not with
  * the table size */
+__global__ static void k_fill_keys(long long *keys, uint64_t n) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) keys[i] = (long long)0x8000000000000000ll;
+}
+
+void dev_fill_keys(long long *keys, uint64_t n, void *stream) {
+  uint32_t blocks = (uint32_t)((n + 255) / 256);
+  hipLaunchKernelGGL(k_fill_keys, dim3(blocks), dim3(256), 0,
+                     (hipStream_t)stream, keys, n);
+}
+
 __global__ static void k_ht_flags(const long long *keys, uint32_t tsize,
                                   uint32_t *f) {
   uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
