@@ -943,7 +943,8 @@ __device__ static inline bool d_rpn_keep(const ScanPlan &plan,
                                          int64_t f1_v, bool f2_found,
                                          bool f2_null, int64_t f2_v,
                                          bool *err,
-                                         const AggColView *xcols = nullptr) {
+                                         int64_t xv0 = 0, uint8_t xn0 = 0,
+                                         int64_t xv1 = 0, uint8_t xn1 = 0) {
   int64_t sv[4];
   uint8_t sn[4], su[4];
   int sp = 0;
@@ -956,28 +957,16 @@ __device__ static inline bool d_rpn_keep(const ScanPlan &plan,
     if (plan.filter2_missing_null) f2_null = true;
     else { f2_v = plan.filter2_missing_val; f2_null = false; }
   }
-  /* channels 2..3 ride capture-only pseudo-agg slots */
-  int64_t x_v[2] = {0, 0};
-  bool x_null[2] = {false, false};
-  for (int k = 0; k < plan.n_xcap; k++) {
-    if (!xcols) { *err = true; return false; }
-    const AggColView &c = xcols[plan.xcap_idx[k]];
-    if (!c.found) {
-      if (plan.xcap_missing_null[k]) x_null[k] = true;
-      else x_v[k] = plan.xcap_missing_val[k];
-    } else if (c.null) {
-      x_null[k] = true;
-    } else {
-      x_v[k] = c.iv;
-    }
-  }
+
   for (int i = 0; i < plan.rpn_n; i++) {
     const DevRpnNode &nd = plan.rpn[i];
     if (nd.kind == 0) {
       if (sp >= 4) { *err = true; return false; }
       if (nd.slot >= 2) {
-        sv[sp] = x_v[nd.slot - 2];
-        sn[sp] = x_null[nd.slot - 2] ? 1 : 0;
+        /* scalars, never an indexed array: a memory-addressable capture
+           array spilled every instantiation of the callers */
+        sv[sp] = nd.slot == 2 ? xv0 : xv1;
+        sn[sp] = nd.slot == 2 ? xn0 : xn1;
       } else {
         sv[sp] = nd.slot ? f2_v : f1_v;
         sn[sp] = nd.slot ? (f2_null ? 1 : 0) : (f1_null ? 1 : 0);
@@ -1067,12 +1056,47 @@ __device__ static inline bool d_keep2(const ScanPlan &plan, bool f1_found,
                                       bool f1_null, int64_t f1_v,
                                       bool f2_found, bool f2_null,
                                       int64_t f2_v, bool *err,
-                                      const AggColView *xcols = nullptr) {
+                                      int64_t xv0 = 0, uint8_t xn0 = 0,
+                                      int64_t xv1 = 0, uint8_t xn1 = 0) {
   if (plan.rpn_on)
     return d_rpn_keep(plan, f1_found, f1_null, f1_v, f2_found, f2_null, f2_v,
-                      err, xcols);
+                      err, xv0, xn0, xv1, xn1);
   return d_filter_keep(plan, f1_found, f1_null, f1_v) &&
          d_filter2_keep(plan, f2_found, f2_null, f2_v);
+}
+
+/* pull the xcap channels (2..3) out of the capture slots with STATIC
+ * indexing only (an indexed read of cols[] would force the whole per-lane
+ * array to scratch) and apply the missing fill */
+template <int NAGGS>
+__device__ static inline void d_xcap_extract(const ScanPlan &plan,
+                                             const AggColView (&cols)[NAGGS],
+                                             int64_t *xv0, uint8_t *xn0,
+                                             int64_t *xv1, uint8_t *xn1) {
+  int64_t v0 = 0, v1 = 0;
+  uint8_t n0 = 0, n1 = 0;
+  #pragma unroll
+  for (int a = 0; a < NAGGS; a++) {
+    bool m0 = plan.n_xcap > 0 && plan.xcap_idx[0] == a;
+    bool m1 = plan.n_xcap > 1 && plan.xcap_idx[1] == a;
+    if (m0 | m1) {
+      int64_t v;
+      uint8_t nu = 0;
+      int k = m0 ? 0 : 1;
+      if (!cols[a].found) {
+        if (plan.xcap_missing_null[k]) nu = 1;
+        v = plan.xcap_missing_val[k];
+      } else if (cols[a].null) {
+        nu = 1;
+        v = 0;
+      } else {
+        v = cols[a].iv;
+      }
+      if (m0) { v0 = v; n0 = nu; }
+      if (m1) { v1 = v; n1 = nu; }
+    }
+  }
+  *xv0 = v0; *xn0 = n0; *xv1 = v1; *xn1 = n1;
 }
 
 /* 128-bit signed accumulate via two u64 atomics (carry trick) */
@@ -1408,11 +1432,15 @@ k_scan_agg(ScanPlan plan,
         }  /* v1/v2 */
       }
 
+      int64_t xv0_ = 0, xv1_ = 0;
+      uint8_t xn0_ = 0, xn1_ = 0;
+      if (plan.n_xcap)
+        d_xcap_extract(plan, cols, &xv0_, &xn0_, &xv1_, &xn1_);
       if (!parse_ok) {
         any_parse_err = true;
       } else if (bool ke = false;
                  d_keep2(plan, filt_found, filt_null, filt_v, f2_found,
-                         f2_null, f2_v, &ke, cols)
+                         f2_null, f2_v, &ke, xv0_, xn0_, xv1_, xn1_)
                      ? true
                      : (ke ? (any_parse_err = true, false) : false)) {
         SimpleAggAcc *acc_base = nullptr;
@@ -2172,11 +2200,15 @@ k_scan_agg_pipe(ScanPlan plan,
         }  /* v1/v2 */
       }
 
+      int64_t xv0_ = 0, xv1_ = 0;
+      uint8_t xn0_ = 0, xn1_ = 0;
+      if (plan.n_xcap)
+        d_xcap_extract(plan, cols, &xv0_, &xn0_, &xv1_, &xn1_);
       if (!parse_ok) {
         any_parse_err = true;
       } else if (bool ke = false;
                  d_keep2(plan, filt_found, filt_null, filt_v, f2_found,
-                         f2_null, f2_v, &ke, cols)
+                         f2_null, f2_v, &ke, xv0_, xn0_, xv1_, xn1_)
                      ? true
                      : (ke ? (any_parse_err = true, false) : false)) {
         SimpleAggAcc *acc_base = nullptr;
@@ -4367,11 +4399,15 @@ k_scan_extract(ScanPlan plan, const uint8_t *__restrict__ vals,
       }
 
       uint8_t s = 0;
+      int64_t xv0_ = 0, xv1_ = 0;
+      uint8_t xn0_ = 0, xn1_ = 0;
+      if (plan.n_xcap)
+        d_xcap_extract(plan, cols, &xv0_, &xn0_, &xv1_, &xn1_);
       if (!parse_ok) {
         any_err = true;
       } else if (bool ke = false;
                  d_keep2(plan, filt_found, filt_null, filt_v, f2_found,
-                         f2_null, f2_v, &ke, cols)
+                         f2_null, f2_v, &ke, xv0_, xn0_, xv1_, xn1_)
                      ? true
                      : (ke ? (any_err = true, false) : false)) {
         s = (grp_found && !grp_null) ? 2 : 1;
