@@ -3506,6 +3506,92 @@ d_crc64_stream(const uint8_t *__restrict__ base, uint64_t b0, uint64_t b1,
   return crc;
 }
 
+__device__ __attribute__((always_inline)) static inline void
+d_crc_tab8(uint64_t &crc, uint64_t cur, const uint64_t *__restrict__ tab) {
+  crc ^= cur;
+  crc = tab[7 * 256 + (uint32_t)(crc & 0xFF)] ^
+        tab[6 * 256 + (uint32_t)((crc >> 8) & 0xFF)] ^
+        tab[5 * 256 + (uint32_t)((crc >> 16) & 0xFF)] ^
+        tab[4 * 256 + (uint32_t)((crc >> 24) & 0xFF)] ^
+        tab[3 * 256 + (uint32_t)((crc >> 32) & 0xFF)] ^
+        tab[2 * 256 + (uint32_t)((crc >> 40) & 0xFF)] ^
+        tab[1 * 256 + (uint32_t)((crc >> 48) & 0xFF)] ^
+        tab[0 * 256 + (uint32_t)(crc >> 56)];
+}
+
+/* TWO-ROW interleaved variant (COPR_CRC_X2): each lane advances two
+ * independent CRC chains step-for-step, so one chain's 8 dependent table
+ * lookups issue under the other's latency — the single-chain kernel is
+ * chain-latency-bound. Keys (tail-sized) run serially; the
+ * value streams interleave in 64 B blocks and finish on the single-row
+ * pipeline. */
+__global__ void __launch_bounds__(THREADS, 4)
+k_crc64_reg_x2(const uint8_t *__restrict__ vals,
+               const uint64_t *__restrict__ val_offs,
+               const uint8_t *__restrict__ keys,
+               const uint64_t *__restrict__ key_offs, uint64_t n_rows,
+               const uint64_t *__restrict__ g_tables,
+               unsigned long long *__restrict__ out_xor) {
+  __shared__ uint64_t tab[8 * 256];
+  for (uint32_t i = threadIdx.x; i < 8 * 256u; i += blockDim.x)
+    tab[i] = g_tables[i];
+  __syncthreads();
+  unsigned long long acc = 0;
+  const uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t ra = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       ra < n_rows; ra += 2 * stride) {
+    uint64_t rb = ra + stride;
+    const bool has_b = rb < n_rows;
+    uint64_t ca = ~0ull, cb = ~0ull;
+    ca = d_crc64_stream<false>(keys, key_offs[ra], key_offs[ra + 1], ca, tab);
+    if (has_b)
+      cb = d_crc64_stream<false>(keys, key_offs[rb], key_offs[rb + 1], cb, tab);
+    uint64_t a0 = val_offs[ra], a1 = val_offs[ra + 1];
+    uint64_t b0 = has_b ? val_offs[rb] : 0;
+    uint64_t b1 = has_b ? val_offs[rb + 1] : 0;
+    const uint64_t *qa = (const uint64_t *)(vals + (a0 & ~7ull));
+    const uint64_t *qb = (const uint64_t *)(vals + (b0 & ~7ull));
+    uint32_t sha = (uint32_t)(a0 & 7) * 8u;
+    uint32_t shb = (uint32_t)(b0 & 7) * 8u;
+    uint64_t n8a = (a1 - a0) >> 3, n8b = has_b ? ((b1 - b0) >> 3) : 0;
+    uint64_t ka = 0, kb = 0;       /* 8-byte steps consumed */
+    if (n8a >= 8 && n8b >= 8) {
+      uint64_t preva = qa[0], prevb = qb[0];
+      uint64_t wia = 1, wib = 1;
+      while (n8a - ka >= 8 && n8b - kb >= 8) {
+        uint64_t wa[8], wb[8];
+        #pragma unroll
+        for (int j = 0; j < 8; j++) wa[j] = qa[wia + j];
+        #pragma unroll
+        for (int j = 0; j < 8; j++) wb[j] = qb[wib + j];
+        #pragma unroll
+        for (int j = 0; j < 8; j++) {
+          uint64_t cura = sha ? ((preva >> sha) | (wa[j] << (64 - sha)))
+                              : preva;
+          uint64_t curb = shb ? ((prevb >> shb) | (wb[j] << (64 - shb)))
+                              : prevb;
+          preva = wa[j];
+          prevb = wb[j];
+          d_crc_tab8(ca, cura, tab);
+          d_crc_tab8(cb, curb, tab);
+        }
+        wia += 8; wib += 8; ka += 8; kb += 8;
+      }
+    }
+    /* remainders (and the whole row when the partner is short) on the
+       single-row pipeline; the resume offset keeps the 8-byte phase */
+    ca = d_crc64_stream<true>(vals, a0 + ka * 8, a1, ca, tab);
+    acc ^= ~ca;
+    if (has_b) {
+      cb = d_crc64_stream<true>(vals, b0 + kb * 8, b1, cb, tab);
+      acc ^= ~cb;
+    }
+  }
+  for (int off = 32; off > 0; off >>= 1)
+    acc ^= (unsigned long long)__shfl_down((long long)acc, off, 64);
+  if ((threadIdx.x & 63u) == 0 && acc) atomicXor(out_xor, acc);
+}
+
 /* prefetch variant: 4 waves/SIMD (128 VGPRs, no spill), pipeline hides the
  * per-chunk HBM latency inside the lane */
 __global__ void __launch_bounds__(THREADS, 4)
@@ -5759,6 +5845,10 @@ int dev_crc64_launch(const DevRegion &rgn, const uint64_t *d_tables,
     if (grid == 0) grid = 1;
     if (getenv("COPR_CRC_NP"))
       hipLaunchKernelGGL(k_crc64_reg_np, dim3(grid), dim3(THREADS), 0,
+                         (hipStream_t)stream, rgn.d_vals, rgn.d_val_offs,
+                         rgn.d_keys, rgn.d_key_offs, rgn.n_kv, d_tables, d_xor);
+    else if (getenv("COPR_CRC_X2"))
+      hipLaunchKernelGGL(k_crc64_reg_x2, dim3(grid), dim3(THREADS), 0,
                          (hipStream_t)stream, rgn.d_vals, rgn.d_val_offs,
                          rgn.d_keys, rgn.d_key_offs, rgn.n_kv, d_tables, d_xor);
     else if (getenv("COPR_CRC_PF6"))
