@@ -2087,10 +2087,22 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
           /* the LDS pre-agg table accumulates sums with integer atomics;
              f64-bit sums go straight to the global table's double atomics */
           uint32_t slots = any_real_sum ? 0u : 256u;
+          if (const char *e2 = getenv("COPR_LDS_SLOTS"))
+            slots = any_real_sum ? 0u : (uint32_t)atoi(e2);
+          if (slots & (slots - 1)) {          /* probe masks need pow2 */
+            uint32_t p2 = 1;
+            while (p2 * 2 <= slots) p2 *= 2;
+            slots = p2;
+          }
           uint32_t table_b =
               slots ? (16u + slots * (8 + (uint32_t)sp.n_aggs *
                                               (uint32_t)sizeof(SimpleAggAcc)))
                     : 0u;
+          while (slots && table_b > 100u * 1024u) {
+            slots /= 2;
+            table_b = 16u + slots * (8 + (uint32_t)sp.n_aggs *
+                                             (uint32_t)sizeof(SimpleAggAcc));
+          }
           /* hash mode prefers the glds-pipelined kernel (DMA-staged dir
              planes + overlap; r01's single-buffer kernel measured 47%
              wave-parked); the table rides after the two buffers */

@@ -150,45 +150,50 @@ __device__ static inline uint32_t d_decimal_scaled(const uint8_t *p, uint32_t re
   uint32_t fw = fr / 9, td = fr - fw * 9;
   uint32_t need = 2 + iw * 4 + DIG2B[ld] + fw * 4 + DIG2B[td];
   if (rem < need) return 0;
-  const uint8_t *q = p + 2;
-  uint32_t mask = (q[0] & 0x80) ? 0u : 0xFFFFFFFFu;
+  /* payload <= 16 bytes (prec <= 18): two 8-byte windows, fields extracted
+     by shifts — the byte-pointer walk serialized the groups and its
+     per-width switch diverged lanes with differing digit counts (decimal
+     parse was ~1/3 of the cfg3 kernel, profiles/r04_cfg3_attribution) */
+  uint64_t w0 = lds_win8(p + 2);
+  uint64_t w1 = lds_win8(p + 10);
+  uint32_t mask = (uint32_t)(w0 & 0x80) ? 0u : 0xFFFFFFFFu;
+  w0 ^= 0x80ull;                          /* flip the first payload byte */
   bool neg = mask != 0;
-  bool first = true;
   uint64_t acc = 0;
-  auto rd_word = [&](uint32_t size) -> uint32_t {
-    uint8_t b0 = q[0];
-    if (first) { b0 ^= 0x80; first = false; }
-    uint32_t r;
-    switch (size) {
-      case 1: r = (uint32_t)(int32_t)(int8_t)b0; break;
-      case 2: r = (uint32_t)(((int32_t)(int8_t)b0 << 8) + (int32_t)q[1]); break;
-      case 3: r = (b0 & 128) ? ((255u << 24) | ((uint32_t)b0 << 16) |
-                                ((uint32_t)q[1] << 8) | q[2])
-                             : (((uint32_t)b0 << 16) | ((uint32_t)q[1] << 8) | q[2]);
-              break;
-      default: r = (uint32_t)(((int32_t)(int8_t)b0 << 24) + ((int32_t)q[1] << 16) +
-                              ((int32_t)q[2] << 8) + (int32_t)q[3]); break;
+  uint32_t off = 0;
+  auto field = [&](uint32_t z) -> uint32_t {
+    uint32_t v = 0;
+    #pragma unroll
+    for (uint32_t j = 0; j < 4; j++) {
+      if (j < z) {
+        uint32_t bo = off + j;
+        uint32_t b = bo < 8 ? (uint32_t)(w0 >> (8 * bo))
+                            : (uint32_t)(w1 >> (8 * (bo - 8)));
+        v = (v << 8) | (b & 0xFFu);
+      }
     }
-    q += size;
-    return r;
+    off += z;
+    int32_t sx = (int32_t)(v << (8 * (4 - z)));  /* sign-extend field */
+    sx >>= 8 * (4 - z);
+    return (uint32_t)sx ^ mask;
   };
   if (ld) {
-    uint32_t w = rd_word(DIG2B[ld]) ^ mask;
+    uint32_t w = field(DIG2B[ld]);
     if (w >= TEN_POW_D[ld]) return 0;   /* leading group: < 10^ld digits */
     acc = w;
   }
   for (uint32_t k = 0; k < iw; k++) {
-    uint32_t w = rd_word(4) ^ mask;
+    uint32_t w = field(4);
     if (w > 999999999u) return 0;
     acc = acc * 1000000000ull + w;
   }
   for (uint32_t k = 0; k < fw; k++) {
-    uint32_t w = rd_word(4) ^ mask;
+    uint32_t w = field(4);
     if (w > 999999999u) return 0;
     acc = acc * 1000000000ull + w;
   }
   if (td) {
-    uint32_t w = rd_word(DIG2B[td]) ^ mask;
+    uint32_t w = field(DIG2B[td]);
     if (w >= TEN_POW_D[td]) return 0;
     acc = acc * TEN_POW_D[td] + w;
   }
