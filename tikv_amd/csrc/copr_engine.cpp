@@ -1721,6 +1721,8 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
   /* TypeChunk segmentation (one chunk per executor batch; drains in
      1024-row chunks) */
   std::vector<uint64_t> chunk_rows;
+  /* the device chunk encoder produced the final TypeChunk bytes directly */
+  bool dev_chunked = false;
   /* HIP-event timing of the scan kernel(s), on the engine's own stream —
      reported in summaries[0].time_processed_ns (the ExecSummary surface the
      reference fills per executor slot, execute_stats.rs:43-76). bench.py's
@@ -2220,6 +2222,128 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
       pick_tiling(r->dev, &sp);
       int le = dev_scan_launch(sp, r->dev, nullptr, nullptr, &po, eng->stream);
       if (le) { free_po(); return SET_ERR(COPR_ERR_INTERNAL, "scan launch failed"); }
+
+      /* ---- device TypeChunk fast path (kernels.hip dev_chunk_encode):
+         single region, table-mode project, every output column
+         chunk-encodes as an 8-byte fixed value. Skips the span-stream +
+         cell-directory D2H and the serial host datum->chunk re-encode.
+         Any row the device cannot encode falls back to the host path
+         below (bit-identical either way — tests compare both). */
+      if (req->encode_type == 1 && n_regions == 1 && !pl.sp.index_mode) {
+        const char *dc_env = getenv("COPR_DEV_CHUNK");
+        bool dc_on = !(dc_env && dc_env[0] == '0');
+        std::vector<copr::ChunkColSpec> specs(req->n_output_offsets);
+        bool eligible = dc_on && req->n_output_offsets > 0;
+        for (uint32_t oo = 0; eligible && oo < req->n_output_offsets; oo++) {
+          uint32_t off = req->output_offsets[oo];
+          if (off >= (uint32_t)pl.sp.n_out || off >= pl.out_schema.size() ||
+              off >= pl.cols.size()) { eligible = false; break; }
+          copr::ChunkColSpec &cs = specs[oo];
+          memset(&cs, 0, sizeof(cs));
+          switch (pl.out_schema[off].tp) {
+            case COPR_TP_TINY: case COPR_TP_SHORT: case COPR_TP_INT24:
+            case COPR_TP_LONG: case COPR_TP_LONGLONG: case COPR_TP_YEAR:
+              cs.is_real = 0; break;
+            case COPR_TP_DOUBLE: cs.is_real = 1; break;
+            case COPR_TP_DURATION: cs.is_real = 2; break;
+            default: eligible = false;
+          }
+          if (!eligible) break;
+          if (pl.sp.out_is_handle[off]) {
+            cs.kind = 1;
+          } else if ((int)off == pl.dec2_col_offset && pl.sp.dec2_col_id) {
+            cs.kind = 3;
+            cs.missing_null = pl.sp.dec2_missing_null;
+            cs.missing_val = pl.sp.dec2_missing_val;
+          } else if ((int)off == pl.filter_col_offset && pl.sp.has_filter) {
+            cs.kind = 2;
+            cs.missing_null = pl.sp.filter_missing_null;
+            cs.missing_val = pl.sp.filter_missing_val;
+          } else {
+            cs.kind = 0;
+            cs.j = (int32_t)off;
+            const CoprColumnInfo &ci = pl.cols[off];
+            cs.uns = (ci.ft.flag & COPR_FLAG_UNSIGNED) ? 1 : 0;
+            if (ci.default_val && ci.default_val_len) {
+              int64_t dv;
+              int dr = host_decode_int_datum(ci.default_val,
+                                             ci.default_val_len, &dv);
+              if (dr < 0) { eligible = false; break; }
+              cs.missing_null = dr == 1 ? 1 : 0;
+              cs.missing_val = dv;
+            } else if (!(ci.ft.flag & COPR_FLAG_NOT_NULL)) {
+              cs.missing_null = 1;
+            } else {
+              /* NOT NULL without default: a missing cell must raise the
+                 reference's error -> host path */
+              eligible = false;
+            }
+          }
+        }
+        if (eligible) {
+          std::vector<uint8_t> dk(n);
+          unsigned int derr[2] = {0, 0};
+          hipError_t de = hipMemcpyAsync(dk.data(), po.keep, n,
+                                         hipMemcpyDeviceToHost, eng->stream);
+          if (de == hipSuccess)
+            de = hipMemcpyAsync(derr, po.error, 8, hipMemcpyDeviceToHost,
+                                eng->stream);
+          if (de == hipSuccess) de = hipStreamSynchronize(eng->stream);
+          if (de != hipSuccess) {
+            free_po();
+            return SET_ERR(COPR_ERR_INTERNAL, hipGetErrorString(de));
+          }
+          if (derr[1]) {
+            free_po();
+            return SET_ERR(COPR_ERR_STORAGE, "row parse error on device");
+          }
+          /* paging + chunk ladder over the keep flags (same walk as the
+             host path below; runner.rs:917-943) */
+          uint64_t scan_end_l = n, resume_l = UINT64_MAX;
+          if (req->paging_size) {
+            uint64_t bs = 32, pos = 0, outcnt = 0;
+            while (pos < n) {
+              uint64_t e2 = std::min(n, pos + bs);
+              for (uint64_t i = pos; i < e2; i++) outcnt += dk[i] ? 1 : 0;
+              pos = e2;
+              if (outcnt >= req->paging_size) break;
+              if (bs < 1024) bs *= 2;
+            }
+            if (pos < n) { scan_end_l = pos; resume_l = pos; }
+          }
+          std::vector<uint64_t> cr_l;
+          {
+            uint64_t bs = 32, pos2 = 0, limit_left =
+                pl.limit == UINT64_MAX ? UINT64_MAX : pl.limit - n_rows_out;
+            while (pos2 < scan_end_l && limit_left) {
+              uint64_t e2 = std::min(scan_end_l, pos2 + bs);
+              uint64_t cnt2 = 0;
+              for (uint64_t i = pos2; i < e2 && cnt2 < limit_left; i++)
+                cnt2 += dk[i] ? 1 : 0;
+              if (cnt2) cr_l.push_back(cnt2);
+              if (limit_left != UINT64_MAX) limit_left -= cnt2;
+              pos2 = e2;
+              if (bs < 1024) bs *= 2;
+            }
+          }
+          std::vector<uint8_t> chunk_bytes;
+          int rc = copr::dev_chunk_encode(po, r->dev, scan_end_l,
+                                          specs.data(),
+                                          (int)req->n_output_offsets,
+                                          pl.sp.n_out, cr_l, eng->stream,
+                                          &chunk_bytes);
+          if (rc == 0) {
+            free_po();
+            resp.insert(resp.end(), chunk_bytes.begin(), chunk_bytes.end());
+            for (uint64_t cn : cr_l) n_rows_out += cn;
+            out->resume_row = resume_l;
+            dev_chunked = true;
+            continue;
+          }
+          /* -3 (a row needs host decode) or alloc/hip trouble: fall
+             through to the host path, which recomputes everything */
+        }
+      }
       std::vector<uint8_t> h_keep(n);
       std::vector<unsigned long long> h_cells((size_t)n * pl.sp.n_out);
       std::vector<long long> h_handles;
@@ -2394,7 +2518,7 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
     hipEventSynchronize(ev_b);
     hipEventElapsedTime(&kernel_ms, ev_a, ev_b);
   }
-  if (req->encode_type == 1) {
+  if (req->encode_type == 1 && !dev_chunked) {
     if (chunk_rows.empty() && n_rows_out) {
       for (uint64_t left = n_rows_out; left;) {      /* drain: 1024/chunk */
         uint64_t b = left < 1024 ? left : 1024;

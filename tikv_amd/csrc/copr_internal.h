@@ -305,6 +305,24 @@ int dev_subregion_build(const DevRegion &src, const uint32_t *h_rows,
                         uint64_t m, DevRegion *out, void *stream);
 /* dd_* = the DEFAULT-CF stream (sorted user_key asc / start_ts desc), or
  * null: a Put without a short value then fails loudly (unsupported). */
+/* device TypeChunk encode (project mode, fixed-8 int/real columns only):
+ * per output column, how the chunk value is sourced. Mirrors the host
+ * datum-row encode loop + e_chunk_append_datum round trip (chunk data is
+ * the 8-byte LE of the decoded value either way). */
+struct ChunkColSpec {
+  int32_t kind;         /* 0 cell[j], 1 handle, 2 filter chan, 3 dec2 chan */
+  int32_t j;            /* kind 0: index into po.cells row */
+  int32_t uns;          /* kind 0: column UNSIGNED (v2 raw-payload decode) */
+  int32_t is_real;      /* kind 0: DOUBLE column (v2 rows unsupported->err) */
+  int32_t missing_null; /* kind 0/2/3: missing column -> NULL */
+  int64_t missing_val;  /* else the decoded default (int or f64 bits) */
+};
+/* 0 ok; -1 hip error; -2 oom; -3 a row needs the host path (fall back) */
+int dev_chunk_encode(const ProjectOut &po, const DevRegion &rgn,
+                     uint64_t scan_end, const ChunkColSpec *h_specs,
+                     int n_cols, int n_out,
+                     const std::vector<uint64_t> &chunk_rows, void *stream,
+                     std::vector<uint8_t> *out_resp);
 int dev_mvcc_build(const uint8_t *d_keys, const uint64_t *d_ko,
                    const uint8_t *d_vals, const uint64_t *d_vo, uint64_t n,
                    const uint8_t *dd_keys, const uint64_t *dd_ko,

@@ -4122,6 +4122,282 @@ int dev_mvcc_build(const uint8_t *d_keys, const uint64_t *d_ko,
   return 0;
 }
 
+/* ---------------- device TypeChunk encode (project, fixed-8 columns) ----
+ * chunk/column.rs:41-71,1052-1071 wire layout built ON DEVICE for output
+ * schemas whose columns all chunk-encode as 8-byte fixed values (ints,
+ * DOUBLE, DURATION). The host keeps the batch-ladder chunk segmentation
+ * (from the keep flags) and patches the 8-byte [len][null_cnt] headers;
+ * kernels decode each kept row's datums straight out of the resident
+ * value stream and write the data sections + null bitmaps in place —
+ * no span-stream D2H, no serial re-encode. Rows the device cannot encode
+ * (v2 DURATION cells, decimal/bytes datums under an int schema, short v2
+ * real payloads) raise the err flag and the caller falls back to the
+ * host path, which reproduces the reference behaviour exactly. */
+__global__ static void __launch_bounds__(THREADS)
+k_u8_widen(const uint8_t *__restrict__ f, uint64_t n,
+           uint64_t *__restrict__ out) {
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (uint64_t)gridDim.x * blockDim.x)
+    out[i] = f[i];
+}
+
+__device__ static inline uint32_t d_chunk_of(
+    const uint64_t *__restrict__ prefix, uint32_t n_chunks, uint64_t pos) {
+  uint32_t lo = 0, hi = n_chunks;
+  while (lo + 1 < hi) {
+    uint32_t mid = (lo + hi) >> 1;
+    if (prefix[mid] <= pos) lo = mid;
+    else hi = mid;
+  }
+  return lo;
+}
+
+__global__ static void __launch_bounds__(THREADS)
+k_chunk_decode(const uint8_t *__restrict__ keep,
+               const uint64_t *__restrict__ out_pos, uint64_t scan_end,
+               uint64_t total_rows,
+               const unsigned long long *__restrict__ cells, int n_out,
+               const uint8_t *__restrict__ vals,
+               const uint64_t *__restrict__ val_offs,
+               const long long *__restrict__ filt_vals,
+               const uint8_t *__restrict__ filt_state,
+               const long long *__restrict__ dec2_vals,
+               const uint8_t *__restrict__ dec2_state,
+               const long long *__restrict__ handles,
+               const ChunkColSpec *__restrict__ specs, int n_cols,
+               const uint64_t *__restrict__ chunk_prefix, uint32_t n_chunks,
+               long long *__restrict__ tmp_vals,
+               uint8_t *__restrict__ tmp_null,
+               unsigned int *__restrict__ null_cnt,
+               unsigned int *__restrict__ err) {
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < scan_end; i += (uint64_t)gridDim.x * blockDim.x) {
+    if (!keep[i]) continue;
+    uint64_t pos = out_pos[i];
+    if (pos >= total_rows) continue;             /* beyond LIMIT */
+    uint32_t ck = d_chunk_of(chunk_prefix, n_chunks, pos);
+    uint64_t rs = val_offs[i];
+    bool v2row = (val_offs[i + 1] - rs) > 1 && vals[rs] == 128;
+    for (int c = 0; c < n_cols; c++) {
+      const ChunkColSpec sp = specs[c];
+      long long v = 0;
+      uint8_t nul = 0;
+      if (sp.kind == 1) {
+        v = handles[i];
+      } else if (sp.kind == 2 || sp.kind == 3) {
+        uint8_t st = sp.kind == 2 ? filt_state[i] : dec2_state[i];
+        if (st == 0) v = sp.kind == 2 ? filt_vals[i] : dec2_vals[i];
+        else if (st == 1 || sp.missing_null) nul = 1;
+        else v = sp.missing_val;
+      } else {
+        unsigned long long cp = cells[(uint64_t)i * n_out + sp.j];
+        uint32_t clen = (uint32_t)(cp & 0xFFFFFu);
+        if (clen == 0xFFFFFu) {                  /* missing column */
+          if (sp.missing_null) nul = 1;
+          else v = sp.missing_val;
+        } else if (clen == 0xFFFFEu) {           /* explicit v2 NULL */
+          nul = 1;
+        } else {
+          const uint8_t *p = vals + (cp >> 20);
+          if (v2row) {
+            /* raw v2 payload (compat_v1.rs:28-126 -> chunk decode) */
+            if (sp.is_real == 1) {
+              /* DOUBLE: payload is the comparable f64 (flag-5 transform) */
+              if (clen != 8) { atomicOr(err, 1u); continue; }
+              uint64_t u = 0;
+              for (int b = 0; b < 8; b++) u = (u << 8) | p[b];
+              if (u & 0x8000000000000000ull) u &= 0x7FFFFFFFFFFFFFFFull;
+              else u = ~u;
+              v = (long long)u;
+            } else if (sp.is_real == 2) {        /* duration etc: host */
+              atomicOr(err, 1u); continue;
+            } else if (sp.uns) {
+              if (clen != 1 && clen != 2 && clen != 4 && clen != 8) {
+                atomicOr(err, 1u); continue;
+              }
+              uint64_t u = 0;
+              for (uint32_t b = 0; b < clen; b++)
+                u |= (uint64_t)p[b] << (8 * b);
+              v = (long long)u;
+            } else {
+              int64_t iv;
+              if (!d_v2_int(p, clen, false, &iv)) { atomicOr(err, 1u); continue; }
+              v = iv;
+            }
+          } else {
+            CellView cell;
+            d_parse_datum(p, clen, &cell);
+            if (!cell.len) { atomicOr(err, 1u); continue; }
+            if (cell.is_null) nul = 1;
+            else if (cell.has_int || cell.has_real) v = cell.ival;
+            else { atomicOr(err, 1u); continue; } /* decimal/bytes: host */
+          }
+        }
+      }
+      tmp_vals[(uint64_t)c * total_rows + pos] = v;
+      tmp_null[(uint64_t)c * total_rows + pos] = nul;
+      if (nul)
+        atomicAdd(&null_cnt[(uint64_t)ck * n_cols + c], 1u);
+    }
+  }
+}
+
+__global__ static void __launch_bounds__(THREADS)
+k_chunk_fill(uint64_t total_rows, int n_cols,
+             const long long *__restrict__ tmp_vals,
+             const uint8_t *__restrict__ tmp_null,
+             const uint64_t *__restrict__ chunk_prefix, uint32_t n_chunks,
+             const uint64_t *__restrict__ data_off,  /* [n_chunks*n_cols] */
+             const uint64_t *__restrict__ bmp_off,   /* ~0ull = no bitmap */
+             uint8_t *__restrict__ outbuf) {
+  for (uint64_t pos = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       pos < total_rows; pos += (uint64_t)gridDim.x * blockDim.x) {
+    uint32_t ck = d_chunk_of(chunk_prefix, n_chunks, pos);
+    uint64_t r = pos - chunk_prefix[ck];
+    uint64_t clen = (ck + 1 < n_chunks ? chunk_prefix[ck + 1]
+                                       : total_rows) - chunk_prefix[ck];
+    for (int c = 0; c < n_cols; c++) {
+      uint8_t nul = tmp_null[(uint64_t)c * total_rows + pos];
+      uint64_t uv = nul ? 0ull
+                        : (uint64_t)tmp_vals[(uint64_t)c * total_rows + pos];
+      uint64_t doff = data_off[(uint64_t)ck * n_cols + c] + 8 * r;
+      #pragma unroll
+      for (int b = 0; b < 8; b++) outbuf[doff + b] = (uint8_t)(uv >> (8 * b));
+      /* null bitmap: the (r%8==0) thread composes each byte — exactly one
+         writer per byte, plain byte stores, no RMW against the data bytes */
+      uint64_t bo = bmp_off[(uint64_t)ck * n_cols + c];
+      if (bo != ~0ull && (r & 7) == 0) {
+        uint8_t byte = 0;
+        for (uint64_t k = 0; k < 8 && r + k < clen; k++)
+          if (!tmp_null[(uint64_t)c * total_rows + pos + k])
+            byte |= (uint8_t)(1u << k);
+        outbuf[bo + (r >> 3)] = byte;
+      }
+    }
+  }
+}
+
+int dev_chunk_encode(const ProjectOut &po, const DevRegion &rgn,
+                     uint64_t scan_end, const ChunkColSpec *h_specs,
+                     int n_cols, int n_out,
+                     const std::vector<uint64_t> &chunk_rows, void *stream,
+                     std::vector<uint8_t> *out_resp) {
+  hipStream_t s = (hipStream_t)stream;
+  uint32_t n_chunks = (uint32_t)chunk_rows.size();
+  std::vector<uint64_t> prefix(n_chunks + 1, 0);
+  for (uint32_t k = 0; k < n_chunks; k++)
+    prefix[k + 1] = prefix[k] + chunk_rows[k];
+  uint64_t total = prefix[n_chunks];
+  out_resp->clear();
+  if (!total) return 0;
+  uint64_t *d_outpos = nullptr, *d_prefix = nullptr;
+  uint64_t *d_doff = nullptr, *d_boff = nullptr, *k64 = nullptr;
+  long long *d_tmpv = nullptr;
+  uint8_t *d_tmpn = nullptr, *d_out = nullptr;
+  unsigned int *d_nullcnt = nullptr, *d_err = nullptr;
+  ChunkColSpec *d_specs = nullptr;
+  void *tmp = nullptr;
+  size_t tmpb = 0;
+  auto freeall = [&]() {
+    hipFree(d_outpos); hipFree(d_prefix); hipFree(d_doff); hipFree(d_boff);
+    hipFree(k64); hipFree(d_tmpv); hipFree(d_tmpn); hipFree(d_out);
+    hipFree(d_nullcnt); hipFree(d_err); hipFree(d_specs); hipFree(tmp);
+  };
+  hipError_t e = hipSuccess;
+  uint64_t na = scan_end ? scan_end : 1;
+  uint64_t nm = (uint64_t)n_chunks * n_cols;
+  if (e == hipSuccess) e = hipMalloc(&d_outpos, na * 8);
+  if (e == hipSuccess) e = hipMalloc(&k64, na * 8);
+  if (e == hipSuccess) e = hipMalloc(&d_prefix, (n_chunks + 1) * 8);
+  if (e == hipSuccess) e = hipMalloc(&d_tmpv, (uint64_t)n_cols * total * 8);
+  if (e == hipSuccess) e = hipMalloc(&d_tmpn, (uint64_t)n_cols * total);
+  if (e == hipSuccess) e = hipMalloc(&d_nullcnt, nm * 4);
+  if (e == hipSuccess) e = hipMalloc(&d_err, 4);
+  if (e == hipSuccess) e = hipMalloc(&d_specs, n_cols * sizeof(ChunkColSpec));
+  if (e != hipSuccess) { freeall(); return -2; }
+  hipMemcpyAsync(d_prefix, prefix.data(), (n_chunks + 1) * 8,
+                 hipMemcpyHostToDevice, s);
+  hipMemcpyAsync(d_specs, h_specs, n_cols * sizeof(ChunkColSpec),
+                 hipMemcpyHostToDevice, s);
+  hipMemsetAsync(d_nullcnt, 0, nm * 4, s);
+  hipMemsetAsync(d_err, 0, 4, s);
+  uint32_t grid = (uint32_t)(((scan_end + THREADS - 1) / THREADS) < 8192
+                                 ? ((scan_end + THREADS - 1) / THREADS)
+                                 : 8192);
+  if (!grid) grid = 1;
+  hipLaunchKernelGGL(k_u8_widen, dim3(grid), dim3(THREADS), 0, s,
+                     po.keep, scan_end, k64);
+  hipcub::DeviceScan::ExclusiveSum(nullptr, tmpb, k64, d_outpos,
+                                   (int)scan_end, s);
+  if (hipMalloc(&tmp, tmpb ? tmpb : 16) != hipSuccess) { freeall(); return -2; }
+  hipcub::DeviceScan::ExclusiveSum(tmp, tmpb, k64, d_outpos, (int)scan_end, s);
+  hipLaunchKernelGGL(k_chunk_decode, dim3(grid), dim3(THREADS), 0, s,
+                     po.keep, d_outpos, scan_end, total, po.cells, n_out,
+                     rgn.d_vals, rgn.d_val_offs, po.filt_vals, po.filt_state,
+                     po.dec2_vals, po.dec2_state, po.handles, d_specs, n_cols,
+                     d_prefix, n_chunks, d_tmpv, d_tmpn, d_nullcnt, d_err);
+  std::vector<unsigned int> h_nullcnt(nm);
+  unsigned int h_err = 0;
+  e = hipMemcpyAsync(h_nullcnt.data(), d_nullcnt, nm * 4,
+                     hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess)
+    e = hipMemcpyAsync(&h_err, d_err, 4, hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  if (e != hipSuccess) { freeall(); return -1; }
+  if (h_err) { freeall(); return -3; }
+  /* section layout (column.rs flush order): per chunk per col
+     [len u32][null_cnt u32][bitmap if null_cnt>0][data 8*len] */
+  std::vector<uint64_t> h_doff(nm), h_boff(nm, ~0ull);
+  uint64_t off = 0;
+  for (uint32_t k = 0; k < n_chunks; k++) {
+    uint64_t len = chunk_rows[k];
+    for (int c = 0; c < n_cols; c++) {
+      off += 8;
+      if (h_nullcnt[(uint64_t)k * n_cols + c] > 0) {
+        h_boff[(uint64_t)k * n_cols + c] = off;
+        off += (len + 7) / 8;
+      }
+      h_doff[(uint64_t)k * n_cols + c] = off;
+      off += 8 * len;
+    }
+  }
+  uint64_t total_bytes = off;
+  if (e == hipSuccess) e = hipMalloc(&d_doff, nm * 8);
+  if (e == hipSuccess) e = hipMalloc(&d_boff, nm * 8);
+  if (e == hipSuccess) e = hipMalloc(&d_out, total_bytes);
+  if (e != hipSuccess) { freeall(); return -2; }
+  hipMemcpyAsync(d_doff, h_doff.data(), nm * 8, hipMemcpyHostToDevice, s);
+  hipMemcpyAsync(d_boff, h_boff.data(), nm * 8, hipMemcpyHostToDevice, s);
+  uint32_t grid2 = (uint32_t)(((total + THREADS - 1) / THREADS) < 8192
+                                  ? ((total + THREADS - 1) / THREADS) : 8192);
+  if (!grid2) grid2 = 1;
+  hipLaunchKernelGGL(k_chunk_fill, dim3(grid2), dim3(THREADS), 0, s,
+                     total, n_cols, d_tmpv, d_tmpn, d_prefix, n_chunks,
+                     d_doff, d_boff, d_out);
+  out_resp->resize(total_bytes);
+  e = hipMemcpyAsync(out_resp->data(), d_out, total_bytes,
+                     hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  freeall();
+  if (e != hipSuccess) { out_resp->clear(); return -1; }
+  /* patch the per-(chunk,col) headers */
+  for (uint32_t k = 0; k < n_chunks; k++) {
+    uint64_t len = chunk_rows[k];
+    for (int c = 0; c < n_cols; c++) {
+      uint64_t bmb = h_boff[(uint64_t)k * n_cols + c] != ~0ull
+                         ? (len + 7) / 8 : 0;
+      uint8_t *hw = out_resp->data() +
+                    (h_doff[(uint64_t)k * n_cols + c] - bmb - 8);
+      uint32_t nc2 = h_nullcnt[(uint64_t)k * n_cols + c];
+      hw[0] = (uint8_t)len;  hw[1] = (uint8_t)(len >> 8);
+      hw[2] = (uint8_t)(len >> 16); hw[3] = (uint8_t)(len >> 24);
+      hw[4] = (uint8_t)nc2;  hw[5] = (uint8_t)(nc2 >> 8);
+      hw[6] = (uint8_t)(nc2 >> 16); hw[7] = (uint8_t)(nc2 >> 24);
+    }
+  }
+  return 0;
+}
+
 /* ---------------- launch wrappers ---------------- */
 template <bool IS_HASH, int NLOADS>
 static int launch_agg2(const ScanPlan &plan, const DevRegion &rgn,
