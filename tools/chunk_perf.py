@@ -31,7 +31,8 @@ def main():
             tikv_amd.Col(2, tp=F.TP_NEWDECIMAL, decimal=2),
             tikv_amd.Col(3, tp=F.TP_VARCHAR)]
     out = []
-    for name, thr in (("sel~5pct", -9 * 10**8), ("sel~50pct", 0)):
+    # config_index=2 col1 is the group key, uniform over [0, 64)
+    for name, thr in (("sel~5pct", 3), ("sel~50pct", 32)):
         sel = tikv_amd.cmp_col_const(0, F.SIG_LT_INT, thr)
         req = (tikv_amd.DagSelect(cols).where(sel).output([0])
                .chunked().build())
