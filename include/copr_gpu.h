@@ -204,6 +204,29 @@ copr_status copr_blocks_decompress(const uint8_t *blocks,
                                    const uint8_t *types, uint32_t n_blocks,
                                    uint8_t **out, uint64_t **out_offs);
 
+/* ---- whole-SST ingestion (the file layer above the block layer) ----
+ * One complete BlockBasedTable file: the engine parses the footer
+ * (format_version 1..5, 53-byte form: [checksum_type u8][metaindex
+ * BlockHandle][index BlockHandle][padding][version u32le][magic u64le]),
+ * walks the index block (values = plain BlockHandles, the kBinarySearch
+ * shape TiKV writes), verifies each block's RocksDB-masked crc32c
+ * trailer when checksum_type == 1 (the TiKV default; xxHash kinds are
+ * accepted unverified), decompresses per the per-block trailer type
+ * byte (0/4/5/7), and feeds the device block parser. Legacy (v0) and
+ * v6+ footers, and delta-encoded index values, are rejected loudly.
+ * Reference: RocksDB format.cc / block_based_table_reader.cc (public
+ * format) consumed by TiKV via rust-rocksdb (engine_iterator.rs:12). */
+copr_status copr_region_create_sst(copr_engine *,
+                                   const uint8_t *file, uint64_t file_len,
+                                   copr_region **out);
+
+/* SST carrying write-CF records: file walk + device block parse + the
+ * device MVCC version filter at read_ts */
+copr_status copr_region_create_sst_mvcc(copr_engine *,
+                                        const uint8_t *file,
+                                        uint64_t file_len, uint64_t read_ts,
+                                        copr_region **out);
+
 /* fixture writer: pack a KV stream into data blocks (restart-interval
  * prefix compression; ~target_block_bytes per block). Buffers owned by
  * the generator allocator; free blocks with free() and offs with free() */

@@ -85,6 +85,11 @@ int  orc_mvcc_filter2(const uint8_t *keys, const uint64_t *key_offs,
                       uint64_t n_default, uint64_t read_ts, OrcRegion *out);
 void orc_region_free(OrcRegion *);
 
+/* ---- whole-SST walk (BlockBasedTable footer + index + per-block
+ * trailer restatement; RocksDB format.cc public format). 0 ok,
+ * -1 malformed/checksum mismatch, -2 unsupported footer/index shape. */
+int orc_sst_parse(const uint8_t *file, uint64_t file_len, OrcRegion *out);
+
 #ifdef __cplusplus
 }
 #endif

@@ -20,6 +20,7 @@
 
 #include <cstring>
 #include <cstdio>
+#include <dlfcn.h>
 #include <unordered_map>
 #include <map>
 #include <vector>
@@ -1887,6 +1888,151 @@ extern "C" void orc_region_free(OrcRegion *r) {
   if (!r) return;
   free(r->keys); free(r->key_offs); free(r->vals); free(r->val_offs);
   memset(r, 0, sizeof(*r));
+}
+
+/* ---- whole-SST walk (oracle restatement of the BlockBasedTable file
+ * layer: RocksDB format.cc / block_based_table_reader.cc public format;
+ * TiKV consumes SSTs via rust-rocksdb, engine_iterator.rs:12).
+ * Footer (format_version 1..5): last 53 bytes =
+ *   [checksum_type u8][metaindex handle][index handle][pad to 40]
+ *   [format_version u32le][magic u64le], magic 0x88e241b785f4cff7.
+ * Per-block trailer: [compression u8][checksum u32le]; checksum_type 1 =
+ * masked crc32c(contents || compression byte) (util/crc32c.h mask).
+ * Index values are plain BlockHandles (varint64 offset + size). */
+
+static uint32_t o_crc32c(const uint8_t *p, size_t n) {
+  uint32_t c = 0xFFFFFFFFu;
+  for (size_t i = 0; i < n; i++) {
+    c ^= p[i];
+    for (int k = 0; k < 8; k++) c = (c >> 1) ^ ((c & 1) ? 0x82F63B78u : 0);
+  }
+  return c ^ 0xFFFFFFFFu;
+}
+
+static bool o_var64(const uint8_t *p, size_t rem, uint64_t *v, size_t *n) {
+  uint64_t x = 0;
+  size_t i = 0;
+  int sh = 0;
+  while (i < rem && i < 10) {
+    uint8_t b = p[i++];
+    x |= (uint64_t)(b & 0x7F) << sh;
+    sh += 7;
+    if (!(b & 0x80)) { *v = x; *n = i; return true; }
+  }
+  return false;
+}
+
+static bool o_decompress(uint8_t type, const uint8_t *p, size_t len,
+                         std::vector<uint8_t> *out) {
+  /* compress_format_version 2: varint32 raw size + payload */
+  uint32_t raw;
+  size_t n;
+  if (!blk_varint32(p, len, &raw, &n)) return false;
+  size_t base = out->size();
+  out->resize(base + raw);
+  static void *l4 = dlopen("liblz4.so.1", RTLD_NOW);
+  static void *lz = dlopen("libzstd.so.1", RTLD_NOW);
+  if (type == 4 || type == 5) {
+    typedef int (*fn)(const char *, char *, int, int);
+    static fn d = l4 ? (fn)dlsym(l4, "LZ4_decompress_safe") : nullptr;
+    if (!d) return false;
+    int r = d((const char *)(p + n), (char *)(out->data() + base),
+              (int)(len - n), (int)raw);
+    return r >= 0 && (uint32_t)r == raw;
+  }
+  if (type == 7) {
+    typedef size_t (*fn)(void *, size_t, const void *, size_t);
+    typedef unsigned (*efn)(size_t);
+    static fn d = lz ? (fn)dlsym(lz, "ZSTD_decompress") : nullptr;
+    static efn ie = lz ? (efn)dlsym(lz, "ZSTD_isError") : nullptr;
+    if (!d || !ie) return false;
+    size_t r = d(out->data() + base, raw, p + n, len - n);
+    return !ie(r) && r == raw;
+  }
+  return false;
+}
+
+/* -1 malformed/checksum, -2 unsupported shape, 0 ok */
+extern "C" int orc_sst_parse(const uint8_t *f, uint64_t len, OrcRegion *out) {
+  if (len < 53) return -1;
+  uint64_t magic = 0;
+  for (int i = 7; i >= 0; i--) magic = (magic << 8) | f[len - 8 + i];
+  if (magic != 0x88e241b785f4cff7ull) return -1;
+  uint32_t ver = (uint32_t)f[len - 12] | ((uint32_t)f[len - 11] << 8) |
+                 ((uint32_t)f[len - 10] << 16) | ((uint32_t)f[len - 9] << 24);
+  if (ver < 1 || ver > 5) return -2;
+  const uint8_t *fp = f + len - 53;
+  uint8_t cks = fp[0];
+  uint64_t h[4];
+  size_t pos = 1, n;
+  for (int i = 0; i < 4; i++) {
+    if (!o_var64(fp + pos, 41 - pos, &h[i], &n)) return -1;
+    pos += n;
+  }
+  uint64_t i_off = h[2], i_sz = h[3];
+  auto trailer_ok = [&](uint64_t off, uint64_t sz, uint8_t *type) {
+    if (off + sz + 5 > len) return false;
+    *type = f[off + sz];
+    if (cks == 1) {
+      uint32_t stored = (uint32_t)f[off + sz + 1] |
+                        ((uint32_t)f[off + sz + 2] << 8) |
+                        ((uint32_t)f[off + sz + 3] << 16) |
+                        ((uint32_t)f[off + sz + 4] << 24);
+      uint32_t c = o_crc32c(f + off, sz + 1);
+      if (((c >> 15) | (c << 17)) + 0xa282ead8u != stored) return false;
+    }
+    return true;
+  };
+  uint8_t itype;
+  if (!trailer_ok(i_off, i_sz, &itype)) return -1;
+  std::vector<uint8_t> ibuf;
+  const uint8_t *ib = f + i_off;
+  size_t iblen = (size_t)i_sz;
+  if (itype != 0) {
+    if (!o_decompress(itype, f + i_off, (size_t)i_sz, &ibuf)) return -1;
+    ib = ibuf.data();
+    iblen = ibuf.size();
+  }
+  if (iblen < 8) return -1;
+  uint32_t nr = (uint32_t)ib[iblen - 4] | ((uint32_t)ib[iblen - 3] << 8) |
+                ((uint32_t)ib[iblen - 2] << 16) |
+                ((uint32_t)ib[iblen - 1] << 24);
+  if (iblen < 4 + (size_t)nr * 4) return -1;
+  size_t dend = iblen - 4 - (size_t)nr * 4;
+  std::vector<uint8_t> cat;
+  std::vector<uint64_t> offs{0};
+  std::string key;
+  pos = 0;
+  while (pos < dend) {
+    uint32_t sh, ns, vl;
+    if (!blk_varint32(ib + pos, dend - pos, &sh, &n)) return -1;
+    pos += n;
+    if (!blk_varint32(ib + pos, dend - pos, &ns, &n)) return -1;
+    pos += n;
+    if (!blk_varint32(ib + pos, dend - pos, &vl, &n)) return -1;
+    pos += n;
+    if (pos + ns + vl > dend || sh > key.size()) return -1;
+    key.resize(sh);
+    key.append((const char *)(ib + pos), ns);
+    pos += ns;
+    uint64_t b_off, b_sz;
+    size_t n2;
+    if (!o_var64(ib + pos, vl, &b_off, &n) ||
+        !o_var64(ib + pos + n, vl - n, &b_sz, &n2))
+      return -2;
+    pos += vl;
+    uint8_t bt;
+    if (!trailer_ok(b_off, b_sz, &bt)) return -1;
+    if (bt == 0) {
+      cat.insert(cat.end(), f + b_off, f + b_off + b_sz);
+    } else {
+      if (!o_decompress(bt, f + b_off, (size_t)b_sz, &cat)) return -1;
+    }
+    offs.push_back(cat.size());
+  }
+  if (pos != dend || offs.size() < 2) return -1;
+  return orc_block_parse(cat.data(), offs.data(), (uint32_t)(offs.size() - 1),
+                         out);
 }
 
 }  // extern "C"

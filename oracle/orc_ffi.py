@@ -137,6 +137,29 @@ def block_parse(blocks, block_offs, n_blocks):
     return keysb, koffs, valsb, voffs, n
 
 
+def sst_parse(file_bytes):
+    """Oracle walk of a whole BlockBasedTable SST (footer + index +
+    crc32c + decompress + block decode) -> (keys, ko, vals, vo, n)."""
+    lib = load_lib()
+    lib.orc_sst_parse.restype = C.c_int
+    lib.orc_sst_parse.argtypes = [C.POINTER(C.c_uint8), C.c_uint64,
+                                  C.POINTER(OrcRegion)]
+    buf = (C.c_uint8 * max(len(file_bytes), 1)).from_buffer_copy(
+        file_bytes or b"\0")
+    out = OrcRegion()
+    st = lib.orc_sst_parse(C.cast(buf, C.POINTER(C.c_uint8)),
+                           len(file_bytes), C.byref(out))
+    if st != 0:
+        raise RuntimeError("orc_sst_parse: %d" % st)
+    n = out.n_kv
+    koffs = [out.key_offs[i] for i in range(n + 1)]
+    voffs = [out.val_offs[i] for i in range(n + 1)]
+    keysb = C.string_at(out.keys, koffs[-1]) if koffs[-1] else b""
+    valsb = C.string_at(out.vals, voffs[-1]) if voffs[-1] else b""
+    lib.orc_region_free(C.byref(out))
+    return keysb, koffs, valsb, voffs, n
+
+
 def mvcc_filter(keys, key_offs, vals, val_offs, n_kv, read_ts,
                 default_cf=None):
     """Run the oracle MVCC filter; returns (keys, key_offs, vals, val_offs,

@@ -142,6 +142,13 @@ def load_lib():
         C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
         C.POINTER(C.c_uint8), C.POINTER(C.c_uint64), C.c_uint64,
         C.c_uint64, C.POINTER(C.c_void_p)]
+    lib.copr_region_create_sst.restype = C.c_int
+    lib.copr_region_create_sst.argtypes = [
+        C.c_void_p, C.POINTER(C.c_uint8), C.c_uint64, C.POINTER(C.c_void_p)]
+    lib.copr_region_create_sst_mvcc.restype = C.c_int
+    lib.copr_region_create_sst_mvcc.argtypes = [
+        C.c_void_p, C.POINTER(C.c_uint8), C.c_uint64, C.c_uint64,
+        C.POINTER(C.c_void_p)]
     lib.copr_region_dump.restype = C.c_int
     lib.copr_region_dump.argtypes = [C.c_void_p, C.c_void_p, C.POINTER(CoprGenOut)]
     lib.copr_region_num_kv.restype = C.c_uint64

@@ -375,6 +375,34 @@ class Engine:
                                (st, self._lib.copr_last_error().decode()))
         return Region(self, r)
 
+    def region_sst(self, file_bytes):
+        """Region from one whole BlockBasedTable SST file: footer + index
+        walk + crc32c verify + per-block decompression on host, block
+        parse on device."""
+        buf = (C.c_uint8 * max(len(file_bytes), 1)).from_buffer_copy(
+            file_bytes or b"\0")
+        r = C.c_void_p()
+        st = self._lib.copr_region_create_sst(
+            self._h, C.cast(buf, C.POINTER(C.c_uint8)), len(file_bytes),
+            C.byref(r))
+        if st != 0:
+            raise RuntimeError("copr_region_create_sst: %d (%s)" %
+                               (st, self._lib.copr_last_error().decode()))
+        return Region(self, r)
+
+    def region_sst_mvcc(self, file_bytes, read_ts):
+        """SST of write-CF records: file walk + device MVCC filter."""
+        buf = (C.c_uint8 * max(len(file_bytes), 1)).from_buffer_copy(
+            file_bytes or b"\0")
+        r = C.c_void_p()
+        st = self._lib.copr_region_create_sst_mvcc(
+            self._h, C.cast(buf, C.POINTER(C.c_uint8)), len(file_bytes),
+            read_ts, C.byref(r))
+        if st != 0:
+            raise RuntimeError("copr_region_create_sst_mvcc: %d (%s)" %
+                               (st, self._lib.copr_last_error().decode()))
+        return Region(self, r)
+
     def region_mvcc_with_default(self, keys, key_offs, vals, val_offs, n_kv,
                                  dkeys, dkey_offs, dvals, dval_offs,
                                  n_default, read_ts):
