@@ -152,6 +152,40 @@ def test_dev_chunk_v2_rows(engine):
 
 
 @pytest.mark.gpu
+def test_dev_chunk_index_project(engine):
+    """chunked project over an INDEX scan: spans reference the key
+    stream; device == host == oracle on key-form and value-form handle
+    layouts."""
+    orc = _orc()
+    for layout in (0, 1):
+        g = tikv_amd.GenRegion(config_index=4, n_rows=40000, table_id=1,
+                               n_cols=layout)
+        try:
+            rgn = engine.region(g)
+            try:
+                cols = [tikv_amd.Col(1), tikv_amd.Col(2),
+                        tikv_amd.Col(-1, pk_handle=True)]
+                req = (tikv_amd.DagSelect(cols, index=True)
+                       .chunked().build())
+                os.environ.pop("COPR_DEV_CHUNK", None)
+                d_dev, n_dev, _ = engine.dag_run(req, [rgn])
+                os.environ["COPR_DEV_CHUNK"] = "0"
+                try:
+                    d_host, n_host, _ = engine.dag_run(req, [rgn])
+                finally:
+                    del os.environ["COPR_DEV_CHUNK"]
+                assert (n_dev, d_dev) == (n_host, d_host), layout
+                od, orows = orc.dag_run(req, g.keys, g.key_offs, g.vals,
+                                        g.val_offs, g.n_kv)
+                assert n_dev == orows == 40000, layout
+                assert d_dev == od, layout
+            finally:
+                rgn.close()
+        finally:
+            g.close()
+
+
+@pytest.mark.gpu
 def test_dev_chunk_fallback_varbytes(engine):
     """a varbytes output column is ineligible for the device encoder; the
     silent host fallback must still match the oracle bit-for-bit."""

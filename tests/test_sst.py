@@ -131,6 +131,12 @@ def two_block_kvs():
     return blocks, kvs
 
 
+def test_crc32c_known_answer():
+    """pins the fixture's crc32c (and, via every round-trip test below,
+    the engine's and the oracle's) to the published Castagnoli KAT."""
+    assert crc32c(b"123456789") == 0xE3069283
+
+
 # ---- CPU: oracle walk -------------------------------------------------
 def test_sst_oracle_roundtrip():
     orc = _orc()

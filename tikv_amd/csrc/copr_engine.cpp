@@ -2458,12 +2458,12 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
       if (le) { free_po(); return SET_ERR(COPR_ERR_INTERNAL, "scan launch failed"); }
 
       /* ---- device TypeChunk fast path (kernels.hip dev_chunk_encode):
-         single region, table-mode project, every output column
+         single region, table or index project, every output column
          chunk-encodes as an 8-byte fixed value. Skips the span-stream +
          cell-directory D2H and the serial host datum->chunk re-encode.
          Any row the device cannot encode falls back to the host path
          below (bit-identical either way — tests compare both). */
-      if (req->encode_type == 1 && n_regions == 1 && !pl.sp.index_mode) {
+      if (req->encode_type == 1 && n_regions == 1) {
         const char *dc_env = getenv("COPR_DEV_CHUNK");
         bool dc_on = !(dc_env && dc_env[0] == '0');
         std::vector<copr::ChunkColSpec> specs(req->n_output_offsets);
@@ -2561,7 +2561,8 @@ extern "C" copr_status copr_dag_run(copr_engine *eng, const CoprDagRequest *req,
             }
           }
           std::vector<uint8_t> chunk_bytes;
-          int rc = copr::dev_chunk_encode(po, r->dev, scan_end_l,
+          int rc = copr::dev_chunk_encode(po, r->dev,
+                                          pl.sp.index_mode != 0, scan_end_l,
                                           specs.data(),
                                           (int)req->n_output_offsets,
                                           pl.sp.n_out, cr_l, eng->stream,

@@ -317,8 +317,9 @@ struct ChunkColSpec {
   int32_t missing_null; /* kind 0/2/3: missing column -> NULL */
   int64_t missing_val;  /* else the decoded default (int or f64 bits) */
 };
-/* 0 ok; -1 hip error; -2 oom; -3 a row needs the host path (fall back) */
-int dev_chunk_encode(const ProjectOut &po, const DevRegion &rgn,
+/* 0 ok; -1 hip error; -2 oom; -3 a row needs the host path (fall back).
+ * idxp: project over an index scan — spans reference the key stream. */
+int dev_chunk_encode(const ProjectOut &po, const DevRegion &rgn, int idxp,
                      uint64_t scan_end, const ChunkColSpec *h_specs,
                      int n_cols, int n_out,
                      const std::vector<uint64_t> &chunk_rows, void *stream,

@@ -4277,12 +4277,16 @@ k_chunk_fill(uint64_t total_rows, int n_cols,
   }
 }
 
-int dev_chunk_encode(const ProjectOut &po, const DevRegion &rgn,
+int dev_chunk_encode(const ProjectOut &po, const DevRegion &rgn, int idxp,
                      uint64_t scan_end, const ChunkColSpec *h_specs,
                      int n_cols, int n_out,
                      const std::vector<uint64_t> &chunk_rows, void *stream,
                      std::vector<uint8_t> *out_resp) {
   hipStream_t s = (hipStream_t)stream;
+  /* index project spans reference the KEY stream; keys start with 't',
+     never 128, so the v2-row sniff in k_chunk_decode stays inert */
+  const uint8_t *d_span = idxp ? rgn.d_keys : rgn.d_vals;
+  const uint64_t *d_span_offs = idxp ? rgn.d_key_offs : rgn.d_val_offs;
   uint32_t n_chunks = (uint32_t)chunk_rows.size();
   std::vector<uint64_t> prefix(n_chunks + 1, 0);
   for (uint32_t k = 0; k < n_chunks; k++)
@@ -4333,7 +4337,7 @@ int dev_chunk_encode(const ProjectOut &po, const DevRegion &rgn,
   hipcub::DeviceScan::ExclusiveSum(tmp, tmpb, k64, d_outpos, (int)scan_end, s);
   hipLaunchKernelGGL(k_chunk_decode, dim3(grid), dim3(THREADS), 0, s,
                      po.keep, d_outpos, scan_end, total, po.cells, n_out,
-                     rgn.d_vals, rgn.d_val_offs, po.filt_vals, po.filt_state,
+                     d_span, d_span_offs, po.filt_vals, po.filt_state,
                      po.dec2_vals, po.dec2_state, po.handles, d_specs, n_cols,
                      d_prefix, n_chunks, d_tmpv, d_tmpn, d_nullcnt, d_err);
   std::vector<unsigned int> h_nullcnt(nm);
