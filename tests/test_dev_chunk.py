@@ -186,6 +186,22 @@ def test_dev_chunk_index_project(engine):
 
 
 @pytest.mark.gpu
+def test_dev_chunk_topn(engine):
+    """chunked TopN: the sub-region project routes its decoded order
+    column through the filter channel; all-int outputs take the device
+    encoder. device == host == oracle."""
+    raw = mixed_int_rows(500)
+    cols = [tikv_amd.Col(1), tikv_amd.Col(2, default_val=b"\x08" + T.var_i64(7)),
+            tikv_amd.Col(3), tikv_amd.Col(4, flag=F.FLAG_UNSIGNED)]
+    for desc in (False, True):
+        req = (tikv_amd.DagSelect(cols)
+               .topn(tikv_amd.Expr().col(0), 63, desc=desc)
+               .chunked().build())
+        data, nrows = run_three_ways(engine, req, raw)
+        assert nrows == 63, desc
+
+
+@pytest.mark.gpu
 def test_dev_chunk_fallback_varbytes(engine):
     """a varbytes output column is ineligible for the device encoder; the
     silent host fallback must still match the oracle bit-for-bit."""
