@@ -3653,6 +3653,126 @@ k_crc64_reg_pf6(const uint8_t *__restrict__ vals,
   if ((threadIdx.x & 63u) == 0 && acc) atomicXor(out_xor, acc);
 }
 
+/* slice-by-16 register stream (COPR_CRC_S16): one dependent table chain
+ * per 16 bytes — HALF the chain steps of slice-8 — using the upper eight
+ * tables the host already builds (tables 8..15 extend the slice-8 set).
+ * 32 KiB of tables in LDS pair with 4 blocks/CU (128 of 160 KiB); the
+ * r01 tiled kernel could not afford that because its LDS also held the
+ * row tiles, which is what the COPR_CRC16 dead-end measured. */
+__device__ __attribute__((always_inline)) static inline void
+d_crc_tab16(uint64_t &crc, uint64_t lo, uint64_t hi,
+            const uint64_t *__restrict__ tab) {
+  lo ^= crc;
+  crc = tab[15 * 256 + (uint32_t)(lo & 0xFF)] ^
+        tab[14 * 256 + (uint32_t)((lo >> 8) & 0xFF)] ^
+        tab[13 * 256 + (uint32_t)((lo >> 16) & 0xFF)] ^
+        tab[12 * 256 + (uint32_t)((lo >> 24) & 0xFF)] ^
+        tab[11 * 256 + (uint32_t)((lo >> 32) & 0xFF)] ^
+        tab[10 * 256 + (uint32_t)((lo >> 40) & 0xFF)] ^
+        tab[9 * 256 + (uint32_t)((lo >> 48) & 0xFF)] ^
+        tab[8 * 256 + (uint32_t)(lo >> 56)] ^
+        tab[7 * 256 + (uint32_t)(hi & 0xFF)] ^
+        tab[6 * 256 + (uint32_t)((hi >> 8) & 0xFF)] ^
+        tab[5 * 256 + (uint32_t)((hi >> 16) & 0xFF)] ^
+        tab[4 * 256 + (uint32_t)((hi >> 24) & 0xFF)] ^
+        tab[3 * 256 + (uint32_t)((hi >> 32) & 0xFF)] ^
+        tab[2 * 256 + (uint32_t)((hi >> 40) & 0xFF)] ^
+        tab[1 * 256 + (uint32_t)((hi >> 48) & 0xFF)] ^
+        tab[0 * 256 + (uint32_t)(hi >> 56)];
+}
+
+__device__ __attribute__((always_inline)) static inline uint64_t
+d_crc64_stream16(const uint8_t *__restrict__ base, uint64_t b0, uint64_t b1,
+                 uint64_t crc, const uint64_t *__restrict__ tab) {
+  uint64_t len = b1 - b0;
+  if (!len) return crc;
+  const uint64_t *q = (const uint64_t *)(base + (b0 & ~7ull));
+  uint32_t sh = (uint32_t)(b0 & 7) * 8u;
+  uint64_t prev = q[0];
+  uint64_t wi = 1;
+  uint64_t n8 = len >> 3;
+  auto cur_of = [&](uint64_t w) {
+    uint64_t cur = sh ? ((prev >> sh) | (w << (64 - sh))) : prev;
+    prev = w;
+    return cur;
+  };
+  if (n8 >= 8) {
+    /* same 64-byte software pipeline as the slice-8 kernel, consumed in
+       16-byte table steps */
+    uint64_t w[8];
+    #pragma unroll
+    for (int j = 0; j < 8; j++) w[j] = q[wi + j];
+    wi += 8;
+    n8 -= 8;
+    while (n8 >= 8) {
+      uint64_t w2[8];
+      #pragma unroll
+      for (int j = 0; j < 8; j++) w2[j] = q[wi + j];
+      #pragma unroll
+      for (int j = 0; j < 8; j += 2) {
+        uint64_t lo = cur_of(w[j]);
+        uint64_t hi = cur_of(w[j + 1]);
+        d_crc_tab16(crc, lo, hi, tab);
+      }
+      #pragma unroll
+      for (int j = 0; j < 8; j++) w[j] = w2[j];
+      wi += 8;
+      n8 -= 8;
+    }
+    #pragma unroll
+    for (int j = 0; j < 8; j += 2) {
+      uint64_t lo = cur_of(w[j]);
+      uint64_t hi = cur_of(w[j + 1]);
+      d_crc_tab16(crc, lo, hi, tab);
+    }
+  }
+  while (n8 >= 2) {
+    uint64_t lo = cur_of(q[wi]);
+    uint64_t hi = cur_of(q[wi + 1]);
+    wi += 2;
+    d_crc_tab16(crc, lo, hi, tab);
+    n8 -= 2;
+  }
+  if (n8) {
+    d_crc_tab8(crc, cur_of(q[wi]), tab);
+    wi++;
+  }
+  uint32_t tail = (uint32_t)(len & 7u);
+  if (tail) {
+    uint64_t w = q[wi];
+    uint64_t cur = sh ? ((prev >> sh) | (w << (64 - sh))) : prev;
+    for (uint32_t t = 0; t < tail; t++) {
+      crc = tab[(uint32_t)((crc ^ cur) & 0xFF)] ^ (crc >> 8);
+      cur >>= 8;
+    }
+  }
+  return crc;
+}
+
+__global__ void __launch_bounds__(THREADS, 4)
+k_crc64_reg_s16(const uint8_t *__restrict__ vals,
+                const uint64_t *__restrict__ val_offs,
+                const uint8_t *__restrict__ keys,
+                const uint64_t *__restrict__ key_offs, uint64_t n_rows,
+                const uint64_t *__restrict__ g_tables,
+                unsigned long long *__restrict__ out_xor) {
+  __shared__ uint64_t tab[16 * 256];
+  for (uint32_t i = threadIdx.x; i < 16 * 256u; i += blockDim.x)
+    tab[i] = g_tables[i];
+  __syncthreads();
+  unsigned long long acc = 0;
+  for (uint64_t row = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       row < n_rows; row += (uint64_t)gridDim.x * blockDim.x) {
+    uint64_t crc = ~0ull;
+    crc = d_crc64_stream16(keys, key_offs[row], key_offs[row + 1], crc, tab);
+    crc = d_crc64_stream16(vals, val_offs[row], val_offs[row + 1], crc, tab);
+    acc ^= ~crc;
+  }
+  for (int off = 32; off > 0; off >>= 1)
+    acc ^= (unsigned long long)__shfl_down((long long)acc, off, 64);
+  if ((threadIdx.x & 63u) == 0 && acc) atomicXor(out_xor, acc);
+}
+
 /* no-prefetch variant: 8 waves/SIMD of independent chains (COPR_CRC_NP) */
 __global__ void __launch_bounds__(THREADS, 8)
 k_crc64_reg_np(const uint8_t *__restrict__ vals,
@@ -6134,6 +6254,14 @@ int dev_crc64_launch(const DevRegion &rgn, const uint64_t *d_tables,
     if (grid == 0) grid = 1;
     if (getenv("COPR_CRC_NP"))
       hipLaunchKernelGGL(k_crc64_reg_np, dim3(grid), dim3(THREADS), 0,
+                         (hipStream_t)stream, rgn.d_vals, rgn.d_val_offs,
+                         rgn.d_keys, rgn.d_key_offs, rgn.n_kv, d_tables, d_xor);
+    else if (getenv("COPR_CRC_S8"))
+      hipLaunchKernelGGL(k_crc64_reg, dim3(grid), dim3(THREADS), 0,
+                         (hipStream_t)stream, rgn.d_vals, rgn.d_val_offs,
+                         rgn.d_keys, rgn.d_key_offs, rgn.n_kv, d_tables, d_xor);
+    else if (getenv("COPR_CRC_S16"))
+      hipLaunchKernelGGL(k_crc64_reg_s16, dim3(grid), dim3(THREADS), 0,
                          (hipStream_t)stream, rgn.d_vals, rgn.d_val_offs,
                          rgn.d_keys, rgn.d_key_offs, rgn.n_kv, d_tables, d_xor);
     else if (getenv("COPR_CRC_X2"))
