@@ -159,6 +159,51 @@ def test_sst_oracle_compressed(ctype):
     assert got == kvs
 
 
+def test_sst_oracle_mixed_compression():
+    """per-block compression types vary within one file (RocksDB picks
+    per level); the walk honours each block's own trailer byte."""
+    orc = _orc()
+    blocks, kvs = two_block_kvs()
+    out = bytearray()
+    handles = []
+    for i, blk in enumerate(blocks):
+        ctype = (0, 7)[i % 2]
+        payload = blk if ctype == 0 else _compress_payload(blk, ctype)
+        handles.append((len(out), len(payload)))
+        out += payload
+        out.append(ctype)
+        out += crc_mask(crc32c(payload + bytes([ctype]))).to_bytes(4, "little")
+    # reuse write_sst's meta/index/footer by rebuilding around the data:
+    # simplest correct path -- write a fresh file with the same blocks but
+    # patch in our mixed payloads is equivalent to building it directly
+    m_off = len(out)
+    meta = (0).to_bytes(4, "little") + (1).to_bytes(4, "little")
+    out += meta + b"\0" + crc_mask(crc32c(meta + b"\0")).to_bytes(4, "little")
+    idx = bytearray()
+    restarts = []
+    for i, (off, sz) in enumerate(handles):
+        key = b"idx%08d" % i + (0).to_bytes(8, "little")
+        val = varint64(off) + varint64(sz)
+        restarts.append(len(idx))
+        idx += varint32(0) + varint32(len(key)) + varint32(len(val))
+        idx += key + val
+    for rr in restarts:
+        idx += rr.to_bytes(4, "little")
+    idx += len(restarts).to_bytes(4, "little")
+    i_off = len(out)
+    out += idx + b"\0" + crc_mask(crc32c(bytes(idx) + b"\0")) \
+        .to_bytes(4, "little")
+    hb = (varint64(m_off) + varint64(len(meta)) +
+          varint64(i_off) + varint64(len(idx)))
+    out += bytes([1]) + hb + b"\0" * (40 - len(hb))
+    out += (2).to_bytes(4, "little")
+    out += (0x88E241B785F4CFF7).to_bytes(8, "little")
+    keys, ko, vals, vo, n = orc.sst_parse(bytes(out))
+    assert n == len(kvs)
+    got = [(keys[ko[i]:ko[i + 1]], vals[vo[i]:vo[i + 1]]) for i in range(n)]
+    assert got == kvs
+
+
 def test_sst_oracle_rejects():
     orc = _orc()
     blocks, kvs = two_block_kvs()
