@@ -116,6 +116,19 @@ def test_dev_chunk_limit_and_paging(engine):
 
 
 @pytest.mark.gpu
+def test_dev_chunk_paging_plus_limit(engine):
+    """paging and LIMIT together: the ladder stops at the paging
+    boundary AND caps at the remaining limit; resume rows agree."""
+    raw = mixed_int_rows(400)
+    sel = tikv_amd.cmp_col_const(0, F.SIG_GT_INT, -40)
+    for limit, page in ((30, 100), (200, 21), (37, 37)):
+        req = (tikv_amd.DagSelect(int_cols()).where(sel).limit(limit)
+               .paging(page).chunked().build())
+        data, nrows = run_three_ways(engine, req, raw)
+        assert nrows <= limit
+
+
+@pytest.mark.gpu
 def test_dev_chunk_double_col(engine):
     rows = []
     for i in range(80):

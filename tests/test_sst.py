@@ -244,6 +244,14 @@ def test_sst_engine_error_paths_cpu():
     assert st != 0
 
 
+def test_sst_oracle_single_block():
+    orc = _orc()
+    e1 = [(b"tabc_r0001", b"only")]
+    sst = write_sst([hand_block(e1, 1)])
+    keys, ko, vals, vo, n = orc.sst_parse(sst)
+    assert n == 1 and vals == b"only"
+
+
 # ---- GPU: engine ingestion parity -------------------------------------
 @pytest.mark.gpu
 def test_sst_device_matches_direct(engine):
@@ -270,6 +278,36 @@ def test_sst_device_matches_direct(engine):
         bad = write_sst(blist, corrupt_block_byte=10)
         with pytest.raises(RuntimeError):
             engine.region_sst(bad)
+    finally:
+        g.close()
+
+
+@pytest.mark.gpu
+def test_sst_query_end_to_end(engine):
+    """ingest an LZ4 SST, run the cfg2-shaped scan+filter+count against
+    it, compare with the oracle over the same KVs."""
+    orc = _orc()
+    g = tikv_amd.GenRegion(config_index=1, n_rows=30001, table_id=1)
+    try:
+        blocks, offs, n, keep = tikv_amd.gen_blocks(g)
+        blist = [C.string_at(
+            C.cast(C.addressof(blocks.contents) + offs[i],
+                   C.POINTER(C.c_uint8)), offs[i + 1] - offs[i])
+            for i in range(n)]
+        sst = write_sst(blist, compression=4)
+        cols = [tikv_amd.Col(i) for i in range(1, 17)]
+        sel = tikv_amd.cmp_col_const(3, tikv_amd._ffi.SIG_LT_INT,
+                                     -800_000_000)
+        req = (tikv_amd.DagSelect(cols).where(sel)
+               .simple_agg([tikv_amd.count_star()]).build())
+        od, on = orc.dag_run(req, g.keys, g.key_offs, g.vals, g.val_offs,
+                             g.n_kv)
+        rgn = engine.region_sst(sst)
+        try:
+            gd, gn, _ = engine.dag_run(req, [rgn])
+        finally:
+            rgn.close()
+        assert (gn, gd) == (on, od)
     finally:
         g.close()
 
